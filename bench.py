@@ -1,0 +1,148 @@
+#!/usr/bin/env python3
+"""trtlab_amd flagship benchmark: ResNet-50 fp16 batch=8 inference serving.
+
+Measures the reference's headline metric (BASELINE.json: inferences/sec +
+p99 latency, ResNet-50 fp16 b8) on N MI355X GPUs, one process per GPU
+(data-parallel replicas, weights broadcast over RCCL at load). Synthetic
+images, random-init calibrated weights (reference models/README.md:4-8).
+
+Each step = one full serving iteration for one batch-8 request: H2D input
+copy + graph-replayed fp16 forward + D2H output copy, pipelined across
+multiple execution contexts (reference examples/00_TensorRT/inference.cc
+pipeline). Weak scaling: per-GPU work is fixed as N grows.
+
+Usage: python bench.py [--gpus N] [--steps K] [--warmup W]
+  (for N>1 the driver launches via torch.distributed.run, one rank per GPU)
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+import numpy as np
+
+BASELINE_INF_S = 953.414  # reference examples/00_TensorRT/README.md:46 (V100)
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=200)
+    ap.add_argument("--warmup", type=int, default=20)
+    ap.add_argument("--batch", type=int, default=8)
+    ap.add_argument("--contexts", type=int, default=3)
+    ap.add_argument("--depth", type=int, default=50)
+    args = ap.parse_args()
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    n_gpus = max(world, 1)
+
+    import torch
+    import torch.distributed as dist
+
+    distributed = world > 1
+    if distributed:
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        torch.cuda.set_device(local_rank)
+        dist.init_process_group("nccl")
+
+    import trtlab_amd
+    from trtlab_amd.engine.planner import Planner
+    from trtlab_amd.engine.runtime import NativeEngine
+    from trtlab_amd.models import build_resnet
+    from trtlab_amd.parallel import broadcast_weights
+
+    # Build the plan (identical on every rank: same seed).
+    g = build_resnet(args.depth, batch=args.batch, image=224, seed=0)
+    plan = Planner().compile(g)
+
+    eng = NativeEngine(plan, device=local_rank)
+    if distributed:
+        # RCCL weight broadcast at load: rank 0's blob is
+        # authoritative; replicas receive over xGMI.
+        broadcast_weights(eng, src_rank=0, device=local_rank)
+
+    ctxs = [eng.create_context(capture=True) for _ in range(args.contexts)]
+
+    rng = np.random.RandomState(123 + rank)
+    batch = (rng.randn(*plan.input_shape) * 0.5).astype(np.float16)
+    for c in ctxs:
+        np.copyto(c.input, batch)
+
+    nc = len(ctxs)
+    launch_t = [0.0] * nc
+    lat_ms: list[float] = []
+
+    def run_steps(k: int, record: bool):
+        for i in range(k):
+            c = ctxs[i % nc]
+            if i >= nc:
+                t_sync = time.perf_counter()
+                c.synchronize()  # wait for this context's previous step
+                if record:
+                    lat_ms.append((time.perf_counter() - launch_t[i % nc]) * 1e3)
+            launch_t[i % nc] = time.perf_counter()
+            c.launch()
+        for j, c in enumerate(ctxs):
+            c.synchronize()
+            if record:
+                lat_ms.append((time.perf_counter() - launch_t[j]) * 1e3)
+
+    # ---- warmup ----
+    run_steps(args.warmup, record=False)
+
+    # ---- timed region ----
+    if distributed:
+        dist.barrier()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    run_steps(args.steps, record=True)
+    torch.cuda.synchronize()
+    elapsed = time.perf_counter() - t0
+    if distributed:
+        t = torch.tensor([elapsed], device="cuda", dtype=torch.float64)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+        dist.barrier()
+
+    if rank == 0:
+        inf_s = n_gpus * args.steps * args.batch / elapsed
+        ms_per_step = elapsed / args.steps * 1e3
+        p99 = float(np.percentile(lat_ms, 99)) if lat_ms else None
+        p50 = float(np.percentile(lat_ms, 50)) if lat_ms else None
+        print(json.dumps({
+            "metric": "inferences/sec",
+            "value": round(inf_s, 2),
+            "unit": "inf/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 4),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": round(inf_s / BASELINE_INF_S, 3),
+            "dtype": "fp16",
+            "data": "synthetic",
+            "config": {
+                "model": f"resnet{args.depth}",
+                "global_batch": args.batch * n_gpus,
+                "batch_per_gpu": args.batch,
+                "image": 224,
+                "contexts": args.contexts,
+                "parallelism": f"dp{n_gpus}",
+                "p50_ms": p50,
+                "p99_ms": p99,
+                "latency_note": "sync-observed request latency (upper bound)",
+            },
+        }), flush=True)
+
+    if distributed:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

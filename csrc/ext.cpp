@@ -1,0 +1,301 @@
+// trtlab_amd — pybind11 bindings for the native MI355X runtime + kernels.
+// The compute path is hand-written HIP; Python orchestrates. Tensors cross
+// the boundary as raw device pointers (torch .data_ptr() or our own arenas).
+#include <pybind11/numpy.h>
+#include <pybind11/pybind11.h>
+#include <pybind11/stl.h>
+
+#include "kernels/launchers.h"
+#include "runtime/runtime.h"
+
+namespace py = pybind11;
+using namespace trtlab;
+
+namespace {
+
+hipStream_t as_stream(uintptr_t s) { return (hipStream_t)s; }
+
+OpDesc op_from_dict(const py::dict& d) {
+  OpDesc o;
+  auto gi = [&](const char* k, int def) -> int {
+    return d.contains(k) ? d[k].cast<int>() : def;
+  };
+  auto gl = [&](const char* k, int64_t def) -> int64_t {
+    return d.contains(k) ? d[k].cast<int64_t>() : def;
+  };
+  auto gf = [&](const char* k, float def) -> float {
+    return d.contains(k) ? d[k].cast<float>() : def;
+  };
+  o.kind = gi("kind", 0);
+  o.dtype = gi("dtype", 0);
+  o.epi = gi("epi", 0);
+  o.in_off = gl("in_off", -1);
+  o.in2_off = gl("in2_off", -1);
+  o.out_off = gl("out_off", -1);
+  o.out2_off = gl("out2_off", -1);
+  o.w_off = gl("w_off", -1);
+  o.scale_off = gl("scale_off", -1);
+  o.bias_off = gl("bias_off", -1);
+  o.M = gi("M", 0); o.N = gi("N", 0); o.K = gi("K", 0);
+  o.Nb = gi("Nb", 0); o.H = gi("H", 0); o.W = gi("W", 0); o.C = gi("C", 0);
+  o.Cout = gi("Cout", 0); o.KH = gi("KH", 0); o.KW = gi("KW", 0);
+  o.sh = gi("sh", 1); o.sw = gi("sw", 1); o.ph = gi("ph", 0); o.pw = gi("pw", 0);
+  o.HW = gi("HW", 0);
+  o.eps = gf("eps", 1e-5f);
+  o.n_elems = gl("n_elems", 0);
+  o.B = gi("B", 0); o.S = gi("S", 0); o.NH = gi("NH", 0); o.HD = gi("HD", 0);
+  o.att_scale = gf("att_scale", 1.0f);
+  return o;
+}
+
+}  // namespace
+
+PYBIND11_MODULE(_C, m) {
+  m.doc() = "trtlab_amd native runtime (gfx950/CDNA4)";
+
+  // ------------------------------------------------------------- memory
+  auto mem = m.def_submodule("memory");
+  mem.def("device_malloc", [](size_t b, int dev) {
+    return (uintptr_t)device_malloc(b, dev);
+  });
+  mem.def("device_free", [](uintptr_t p, size_t b) { device_free((void*)p, b); });
+  mem.def("pinned_malloc", [](size_t b) { return (uintptr_t)pinned_malloc(b); });
+  mem.def("pinned_free", [](uintptr_t p, size_t b) { pinned_free((void*)p, b); });
+  mem.def("device_bytes_in_use", &device_bytes_in_use);
+  mem.def("pinned_bytes_in_use", &pinned_bytes_in_use);
+  mem.def("memcpy_h2d", [](uintptr_t dst, py::buffer src, size_t bytes) {
+    py::buffer_info info = src.request();
+    TRT_HIP_CHECK(hipMemcpy((void*)dst, info.ptr, bytes, hipMemcpyHostToDevice));
+  });
+  mem.def("memcpy_d2h", [](py::buffer dst, uintptr_t src, size_t bytes) {
+    py::buffer_info info = dst.request();
+    TRT_HIP_CHECK(hipMemcpy(info.ptr, (void*)src, bytes, hipMemcpyDeviceToHost));
+  });
+  mem.def("memcpy_d2d", [](uintptr_t dst, uintptr_t src, size_t bytes) {
+    TRT_HIP_CHECK(hipMemcpy((void*)dst, (void*)src, bytes, hipMemcpyDeviceToDevice));
+  });
+  mem.def("memset_d", [](uintptr_t p, int v, size_t bytes) {
+    TRT_HIP_CHECK(hipMemset((void*)p, v, bytes));
+  });
+
+  py::class_<BlockPool>(mem, "BlockPool")
+      .def(py::init<size_t, int, int>(), py::arg("block_bytes"),
+           py::arg("count"), py::arg("device") = 0)
+      .def("acquire", [](BlockPool& p) { return (uintptr_t)p.acquire(); })
+      .def("release", [](BlockPool& p, uintptr_t b) { p.release((void*)b); })
+      .def("available", &BlockPool::available)
+      .def_property_readonly("block_bytes", &BlockPool::block_bytes)
+      .def_property_readonly("total", &BlockPool::total);
+
+  // ---------------------------------------------------------------- hip
+  auto hip = m.def_submodule("hip");
+  hip.def("device_count", [] {
+    int n = 0;
+    hipError_t e = hipGetDeviceCount(&n);
+    return e == hipSuccess ? n : 0;
+  });
+  hip.def("set_device", [](int d) { TRT_HIP_CHECK(hipSetDevice(d)); });
+  hip.def("device_synchronize", [] {
+    py::gil_scoped_release rel;
+    TRT_HIP_CHECK(hipDeviceSynchronize());
+  });
+  hip.def("stream_create", [] {
+    hipStream_t s;
+    TRT_HIP_CHECK(hipStreamCreateWithFlags(&s, hipStreamNonBlocking));
+    return (uintptr_t)s;
+  });
+  hip.def("stream_destroy", [](uintptr_t s) { hipStreamDestroy(as_stream(s)); });
+  hip.def("stream_synchronize", [](uintptr_t s) {
+    py::gil_scoped_release rel;
+    TRT_HIP_CHECK(hipStreamSynchronize(as_stream(s)));
+  });
+  hip.def("device_properties", [](int dev) {
+    hipDeviceProp_t p;
+    TRT_HIP_CHECK(hipGetDeviceProperties(&p, dev));
+    py::dict d;
+    d["name"] = std::string(p.name);
+    d["gcn_arch"] = std::string(p.gcnArchName);
+    d["total_mem"] = (int64_t)p.totalGlobalMem;
+    d["multi_processor_count"] = p.multiProcessorCount;
+    d["lds_per_block"] = (int64_t)p.sharedMemPerBlock;
+    d["clock_khz"] = p.clockRate;
+    d["warp_size"] = p.warpSize;
+    return d;
+  });
+
+  // ---------------------------------------------------------------- ops
+  // Raw kernel launchers (tests + eager use). Pointers are integers
+  // (torch .data_ptr() or memory.device_malloc). stream 0 = default.
+  auto ops = m.def_submodule("ops");
+  ops.def("gemm_bt",
+          [](int dtype, uintptr_t A, uintptr_t B, uintptr_t C, uintptr_t scale,
+             uintptr_t bias, uintptr_t residual, int M, int N, int K, int epi,
+             uintptr_t stream, bool sync) {
+            launch_gemm_bt(dtype, (void*)A, (void*)B, (void*)C, (float*)scale,
+                           (float*)bias, (void*)residual, M, N, K, K, K, N,
+                           epi, as_stream(stream));
+            if (sync) TRT_HIP_CHECK(hipStreamSynchronize(as_stream(stream)));
+          },
+          py::arg("dtype"), py::arg("A"), py::arg("B"), py::arg("C"),
+          py::arg("scale") = 0, py::arg("bias") = 0, py::arg("residual") = 0,
+          py::arg("M") = 0, py::arg("N") = 0, py::arg("K") = 0,
+          py::arg("epi") = 0, py::arg("stream") = 0, py::arg("sync") = true);
+  ops.def("conv2d",
+          [](int dtype, uintptr_t in, uintptr_t Wt, uintptr_t out,
+             uintptr_t scale, uintptr_t bias, uintptr_t residual,
+             uintptr_t zero_page, int Nb, int H, int W, int C, int Cout,
+             int KH, int KW, int sh, int sw, int ph, int pw, int epi,
+             uintptr_t stream, bool sync) {
+            launch_conv2d(dtype, (void*)in, (void*)Wt, (void*)out,
+                          (float*)scale, (float*)bias, (void*)residual,
+                          (void*)zero_page, Nb, H, W, C, Cout, KH, KW, sh, sw,
+                          ph, pw, epi, as_stream(stream));
+            if (sync) TRT_HIP_CHECK(hipStreamSynchronize(as_stream(stream)));
+          },
+          py::arg("dtype"), py::arg("in"), py::arg("Wt"), py::arg("out"),
+          py::arg("scale") = 0, py::arg("bias") = 0, py::arg("residual") = 0,
+          py::arg("zero_page") = 0, py::arg("Nb") = 1, py::arg("H") = 0,
+          py::arg("W") = 0, py::arg("C") = 0, py::arg("Cout") = 0,
+          py::arg("KH") = 1, py::arg("KW") = 1, py::arg("sh") = 1,
+          py::arg("sw") = 1, py::arg("ph") = 0, py::arg("pw") = 0,
+          py::arg("epi") = 0, py::arg("stream") = 0, py::arg("sync") = true);
+  ops.def("maxpool2d",
+          [](int dtype, uintptr_t in, uintptr_t out, int Nb, int H, int W,
+             int C, int KH, int KW, int sh, int sw, int ph, int pw,
+             uintptr_t stream, bool sync) {
+            launch_maxpool2d(dtype, (void*)in, (void*)out, Nb, H, W, C, KH,
+                             KW, sh, sw, ph, pw, as_stream(stream));
+            if (sync) TRT_HIP_CHECK(hipStreamSynchronize(as_stream(stream)));
+          },
+          py::arg("dtype"), py::arg("in"), py::arg("out"), py::arg("Nb"),
+          py::arg("H"), py::arg("W"), py::arg("C"), py::arg("KH"),
+          py::arg("KW"), py::arg("sh"), py::arg("sw"), py::arg("ph"),
+          py::arg("pw"), py::arg("stream") = 0, py::arg("sync") = true);
+  ops.def("gavgpool",
+          [](int dtype, uintptr_t in, uintptr_t out, int Nb, int HW, int C,
+             uintptr_t stream, bool sync) {
+            launch_gavgpool(dtype, (void*)in, (void*)out, Nb, HW, C,
+                            as_stream(stream));
+            if (sync) TRT_HIP_CHECK(hipStreamSynchronize(as_stream(stream)));
+          },
+          py::arg("dtype"), py::arg("in"), py::arg("out"), py::arg("Nb"),
+          py::arg("HW"), py::arg("C"), py::arg("stream") = 0,
+          py::arg("sync") = true);
+  ops.def("softmax_rows",
+          [](int dtype, uintptr_t in, uintptr_t out, int M, int N,
+             uintptr_t stream, bool sync) {
+            launch_softmax_rows(dtype, (void*)in, (void*)out, M, N, N,
+                                as_stream(stream));
+            if (sync) TRT_HIP_CHECK(hipStreamSynchronize(as_stream(stream)));
+          },
+          py::arg("dtype"), py::arg("in"), py::arg("out"), py::arg("M"),
+          py::arg("N"), py::arg("stream") = 0, py::arg("sync") = true);
+  ops.def("layernorm",
+          [](int dtype, uintptr_t in, uintptr_t gamma, uintptr_t beta,
+             uintptr_t out, int M, int N, float eps, uintptr_t stream,
+             bool sync) {
+            launch_layernorm(dtype, (void*)in, (float*)gamma, (float*)beta,
+                             (void*)out, M, N, N, eps, as_stream(stream));
+            if (sync) TRT_HIP_CHECK(hipStreamSynchronize(as_stream(stream)));
+          },
+          py::arg("dtype"), py::arg("in"), py::arg("gamma"), py::arg("beta"),
+          py::arg("out"), py::arg("M"), py::arg("N"), py::arg("eps") = 1e-5f,
+          py::arg("stream") = 0, py::arg("sync") = true);
+  ops.def("add_layernorm",
+          [](int dtype, uintptr_t x, uintptr_t res, uintptr_t gamma,
+             uintptr_t beta, uintptr_t out, uintptr_t sum_out, int M, int N,
+             float eps, uintptr_t stream, bool sync) {
+            launch_add_layernorm(dtype, (void*)x, (void*)res, (float*)gamma,
+                                 (float*)beta, (void*)out, (void*)sum_out, M,
+                                 N, N, eps, as_stream(stream));
+            if (sync) TRT_HIP_CHECK(hipStreamSynchronize(as_stream(stream)));
+          },
+          py::arg("dtype"), py::arg("x"), py::arg("res"), py::arg("gamma"),
+          py::arg("beta"), py::arg("out"), py::arg("sum_out") = 0,
+          py::arg("M") = 0, py::arg("N") = 0, py::arg("eps") = 1e-5f,
+          py::arg("stream") = 0, py::arg("sync") = true);
+  ops.def("elementwise",
+          [](int dtype, int op, uintptr_t a, uintptr_t b, uintptr_t out,
+             int64_t n, uintptr_t stream, bool sync) {
+            launch_elementwise(dtype, op, (void*)a, (void*)b, (void*)out, n,
+                               as_stream(stream));
+            if (sync) TRT_HIP_CHECK(hipStreamSynchronize(as_stream(stream)));
+          },
+          py::arg("dtype"), py::arg("op"), py::arg("a"), py::arg("b") = 0,
+          py::arg("out") = 0, py::arg("n") = 0, py::arg("stream") = 0,
+          py::arg("sync") = true);
+  ops.def("channel_pad",
+          [](int dtype, uintptr_t in, uintptr_t out, int64_t M, int Cin,
+             int Cpad, uintptr_t stream, bool sync) {
+            launch_channel_pad(dtype, (void*)in, (void*)out, M, Cin, Cpad,
+                               as_stream(stream));
+            if (sync) TRT_HIP_CHECK(hipStreamSynchronize(as_stream(stream)));
+          },
+          py::arg("dtype"), py::arg("in"), py::arg("out"), py::arg("M"),
+          py::arg("Cin"), py::arg("Cpad"), py::arg("stream") = 0,
+          py::arg("sync") = true);
+  ops.def("attention",
+          [](int dtype, uintptr_t qkv, uintptr_t out, int B, int S, int H,
+             int D, float scale, uintptr_t stream, bool sync) {
+            launch_attention(dtype, (void*)qkv, (void*)out, B, S, H, D, scale,
+                             as_stream(stream));
+            if (sync) TRT_HIP_CHECK(hipStreamSynchronize(as_stream(stream)));
+          },
+          py::arg("dtype"), py::arg("qkv"), py::arg("out"), py::arg("B"),
+          py::arg("S"), py::arg("H"), py::arg("D"), py::arg("scale"),
+          py::arg("stream") = 0, py::arg("sync") = true);
+
+  // ------------------------------------------------------------- engine
+  py::class_<Engine, std::shared_ptr<Engine>>(m, "Engine")
+      .def(py::init([](int device, py::buffer weights, size_t arena_bytes,
+                       std::vector<py::dict> ops_in, int64_t input_off,
+                       size_t input_bytes, int64_t output_off,
+                       size_t output_bytes) {
+             py::buffer_info wi = weights.request();
+             std::vector<OpDesc> ops;
+             ops.reserve(ops_in.size());
+             for (auto& d : ops_in) ops.push_back(op_from_dict(d));
+             return std::make_shared<Engine>(
+                 device, wi.ptr, (size_t)(wi.size * wi.itemsize), arena_bytes,
+                 std::move(ops), input_off, input_bytes, output_off,
+                 output_bytes);
+           }),
+           py::arg("device"), py::arg("weights"), py::arg("arena_bytes"),
+           py::arg("ops"), py::arg("input_off"), py::arg("input_bytes"),
+           py::arg("output_off"), py::arg("output_bytes"))
+      .def_property_readonly("device", &Engine::device)
+      .def_property_readonly("arena_bytes", &Engine::arena_bytes)
+      .def_property_readonly("input_bytes", &Engine::input_bytes)
+      .def_property_readonly("output_bytes", &Engine::output_bytes)
+      .def_property_readonly("weights_ptr", &Engine::weights_ptr)
+      .def_property_readonly("weight_bytes", &Engine::weight_bytes)
+      .def("upload_weights", [](Engine& e, py::buffer b) {
+        py::buffer_info bi = b.request();
+        e.upload_weights(bi.ptr, (size_t)(bi.size * bi.itemsize));
+      });
+
+  py::class_<ExecutionContext>(m, "ExecutionContext")
+      .def(py::init<std::shared_ptr<Engine>>())
+      .def("capture",
+           [](ExecutionContext& c) {
+             py::gil_scoped_release rel;
+             c.capture();
+           })
+      .def("launch", &ExecutionContext::launch)
+      .def("synchronize",
+           [](ExecutionContext& c) {
+             py::gil_scoped_release rel;
+             c.synchronize();
+           })
+      .def("ready", &ExecutionContext::ready)
+      .def_property_readonly("host_input_ptr", &ExecutionContext::host_input_ptr)
+      .def_property_readonly("host_output_ptr", &ExecutionContext::host_output_ptr)
+      .def("input_view",
+           [](ExecutionContext& c, size_t bytes) {
+             return py::memoryview::from_memory((void*)c.host_input_ptr(),
+                                                bytes);
+           })
+      .def("output_view", [](ExecutionContext& c, size_t bytes) {
+        return py::memoryview::from_memory((void*)c.host_output_ptr(), bytes);
+      });
+}

@@ -1,0 +1,138 @@
+// trtlab_amd — elementwise kernels (gfx950): vectorized fp16/bf16, 16 B/lane.
+// SURVEY.md §2.8 item 5; also dtype converts and the channel-pad copy used to
+// bring NHWC inputs to C % 8 == 0 for the implicit-GEMM conv.
+#include "../common.h"
+
+namespace trtlab {
+
+enum class EwOp : int { kRelu = 0, kGelu = 1, kAdd = 2, kAddRelu = 3 };
+
+__device__ __forceinline__ float gelu_tanh_e(float x) {
+  const float k0 = 0.7978845608028654f;
+  const float k1 = 0.044715f;
+  float u = k0 * (x + k1 * x * x * x);
+  return 0.5f * x * (1.0f + tanhf(u));
+}
+
+template <typename T, EwOp OP>
+__global__ void elementwise_kernel(const T* __restrict__ a,
+                                   const T* __restrict__ b, T* __restrict__ o,
+                                   int64_t n8) {
+  // n8 = number of 8-element groups (callers pad buffers to 16 B)
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n8;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    short4v va0 = *(const short4v*)(a + i * 8);
+    short4v va1 = *(const short4v*)(a + i * 8 + 4);
+    T r[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float x = (float)((const T*)(j < 4 ? (const void*)&va0 : (const void*)&va1))[j & 3];
+      float y;
+      if constexpr (OP == EwOp::kRelu) y = fmaxf(x, 0.f);
+      else if constexpr (OP == EwOp::kGelu) y = gelu_tanh_e(x);
+      else {
+        float xb = (float)b[i * 8 + j];
+        y = x + xb;
+        if constexpr (OP == EwOp::kAddRelu) y = fmaxf(y, 0.f);
+      }
+      r[j] = (T)y;
+    }
+    *(short4v*)(o + i * 8) = *(const short4v*)&r[0];
+    *(short4v*)(o + i * 8 + 4) = *(const short4v*)&r[4];
+  }
+}
+
+// NHWC channel pad: in [M, Cin] -> out [M, Cpad], zeros beyond Cin.
+template <typename T>
+__global__ void channel_pad_kernel(const T* __restrict__ in, T* __restrict__ out,
+                                   int64_t M, int Cin, int Cpad) {
+  int64_t total = M * Cpad;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int c = (int)(i % Cpad);
+    int64_t m = i / Cpad;
+    out[i] = (c < Cin) ? in[m * Cin + c] : (T)0.0f;
+  }
+}
+
+// fp32 -> fp16/bf16 and back (bindings staging, tests)
+template <typename T>
+__global__ void cast_from_f32_kernel(const float* __restrict__ in,
+                                     T* __restrict__ out, int64_t n) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x)
+    out[i] = (T)in[i];
+}
+template <typename T>
+__global__ void cast_to_f32_kernel(const T* __restrict__ in,
+                                   float* __restrict__ out, int64_t n) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x)
+    out[i] = (float)in[i];
+}
+
+static inline int ew_blocks(int64_t work) {
+  return (int)std::min<int64_t>(cdiv(work, 256), 2048);
+}
+
+void launch_elementwise(int dtype, int op, const void* a, const void* b,
+                        void* out, int64_t n, hipStream_t stream) {
+  if (n % 8 != 0) throw std::runtime_error("elementwise: n % 8 != 0");
+  int64_t n8 = n / 8;
+  int blocks = ew_blocks(n8);
+  auto l = [&](auto t, auto o) {
+    using T = decltype(t);
+    constexpr EwOp O = decltype(o)::value;
+    hipLaunchKernelGGL((elementwise_kernel<T, O>), dim3(blocks), dim3(256), 0,
+                       stream, (const T*)a, (const T*)b, (T*)out, n8);
+  };
+  auto dis = [&](auto t) {
+    using T = decltype(t);
+    switch ((EwOp)op) {
+      case EwOp::kRelu: l(T{}, std::integral_constant<EwOp, EwOp::kRelu>{}); break;
+      case EwOp::kGelu: l(T{}, std::integral_constant<EwOp, EwOp::kGelu>{}); break;
+      case EwOp::kAdd: l(T{}, std::integral_constant<EwOp, EwOp::kAdd>{}); break;
+      case EwOp::kAddRelu: l(T{}, std::integral_constant<EwOp, EwOp::kAddRelu>{}); break;
+    }
+  };
+  if (dtype == 0) dis(_Float16{});
+  else dis(__bf16{});
+}
+
+void launch_channel_pad(int dtype, const void* in, void* out, int64_t M,
+                        int Cin, int Cpad, hipStream_t stream) {
+  int blocks = ew_blocks(M * Cpad);
+  if (dtype == 0)
+    hipLaunchKernelGGL((channel_pad_kernel<_Float16>), dim3(blocks), dim3(256),
+                       0, stream, (const _Float16*)in, (_Float16*)out, M, Cin,
+                       Cpad);
+  else
+    hipLaunchKernelGGL((channel_pad_kernel<__bf16>), dim3(blocks), dim3(256),
+                       0, stream, (const __bf16*)in, (__bf16*)out, M, Cin,
+                       Cpad);
+}
+
+void launch_cast(int dtype, bool to_f32, const void* in, void* out, int64_t n,
+                 hipStream_t stream) {
+  int blocks = ew_blocks(n);
+  if (dtype == 0) {
+    if (to_f32)
+      hipLaunchKernelGGL((cast_to_f32_kernel<_Float16>), dim3(blocks),
+                         dim3(256), 0, stream, (const _Float16*)in,
+                         (float*)out, n);
+    else
+      hipLaunchKernelGGL((cast_from_f32_kernel<_Float16>), dim3(blocks),
+                         dim3(256), 0, stream, (const float*)in,
+                         (_Float16*)out, n);
+  } else {
+    if (to_f32)
+      hipLaunchKernelGGL((cast_to_f32_kernel<__bf16>), dim3(blocks), dim3(256),
+                         0, stream, (const __bf16*)in, (float*)out, n);
+    else
+      hipLaunchKernelGGL((cast_from_f32_kernel<__bf16>), dim3(blocks),
+                         dim3(256), 0, stream, (const float*)in, (__bf16*)out,
+                         n);
+  }
+}
+
+}  // namespace trtlab

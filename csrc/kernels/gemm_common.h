@@ -1,0 +1,121 @@
+// trtlab_amd — shared MFMA-GEMM machinery for gfx950 kernels.
+// See gemm.hip for the structure notes.
+#pragma once
+#include "../common.h"
+
+namespace trtlab {
+
+// ---------------------------------------------------------------- epilogues
+enum class Epi : int {
+  kNone = 0,           // C = acc
+  kBias = 1,           // C = acc + bias[n]
+  kBiasRelu = 2,       // C = relu(acc + bias[n])
+  kBiasGelu = 3,       // C = gelu(acc + bias[n])  (tanh approximation)
+  kScaleBias = 4,      // C = acc*scale[n] + bias[n]          (folded BN)
+  kScaleBiasRelu = 5,  // C = relu(acc*scale[n] + bias[n])    (conv+BN+ReLU)
+  kScaleBiasAddRelu = 6,  // C = relu(acc*scale[n]+bias[n]+res[m][n])
+};
+
+__device__ __forceinline__ float gelu_tanh(float x) {
+  const float k0 = 0.7978845608028654f;
+  const float k1 = 0.044715f;
+  float u = k0 * (x + k1 * x * x * x);
+  return 0.5f * x * (1.0f + tanhf(u));
+}
+
+template <Epi E>
+__device__ __forceinline__ float apply_epi(float acc, float scale, float bias,
+                                           float res) {
+  float v = acc;
+  if constexpr (E == Epi::kScaleBias || E == Epi::kScaleBiasRelu ||
+                E == Epi::kScaleBiasAddRelu)
+    v = v * scale;
+  if constexpr (E != Epi::kNone) v = v + bias;
+  if constexpr (E == Epi::kScaleBiasAddRelu) v = v + res;
+  if constexpr (E == Epi::kBiasRelu || E == Epi::kScaleBiasRelu ||
+                E == Epi::kScaleBiasAddRelu)
+    v = fmaxf(v, 0.0f);
+  if constexpr (E == Epi::kBiasGelu) v = gelu_tanh(v);
+  return v;
+}
+
+// ------------------------------------------------------------- MFMA dispatch
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8v;
+
+template <typename T>
+struct Mfma16x16x32;
+
+template <>
+struct Mfma16x16x32<_Float16> {
+  using frag = half8v;
+  static __device__ __forceinline__ f32x4 run(frag a, frag b, f32x4 c) {
+    return __builtin_amdgcn_mfma_f32_16x16x32_f16(a, b, c, 0, 0, 0);
+  }
+};
+
+template <>
+struct Mfma16x16x32<__bf16> {
+  using frag = bf16x8v;
+  static __device__ __forceinline__ f32x4 run(frag a, frag b, f32x4 c) {
+    return __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
+  }
+};
+
+// ------------------------------------------------------------------ staging
+__device__ __forceinline__ void glds16(const void* gsrc, uint32_t lds_byte) {
+  __builtin_amdgcn_global_load_lds(
+      (const __attribute__((address_space(1))) uint32_t*)gsrc,
+      (__attribute__((address_space(3))) uint32_t*)(uintptr_t)lds_byte, 16, 0,
+      0);
+}
+
+// Stage a 128-row x 64-elem (16 KiB) tile of a row-major matrix into LDS,
+// rows clamped to [0, nrows). Source-side XOR swizzle (rule 21).
+template <typename T>
+__device__ __forceinline__ void stage_tile_128x64(
+    const T* __restrict__ src, int64_t stride_elems, int row0, int nrows,
+    uint32_t lds_base, int tid) {
+#pragma unroll
+  for (int c = 0; c < 4; ++c) {
+    uint32_t p = c * 4096 + tid * 16;
+    uint32_t row = p >> 7;
+    uint32_t kb = (p & 127) ^ ((row & 7) << 4);
+    int r = row0 + (int)row;
+    r = (r < 0) ? 0 : (r >= nrows ? nrows - 1 : r);
+    const char* g = (const char*)src +
+                    ((int64_t)r - row0) * stride_elems * (int64_t)sizeof(T) +
+                    kb;
+    glds16(g, lds_base + p);
+  }
+}
+
+// Swizzled ds_read_b128 of one MFMA fragment (8 consecutive k elements).
+template <typename T>
+__device__ __forceinline__ typename Mfma16x16x32<T>::frag read_frag(
+    const char* lds, uint32_t row, uint32_t kbyte) {
+  uint32_t off = (row << 7) + (kbyte ^ ((row & 7) << 4));
+  return *(const typename Mfma16x16x32<T>::frag*)(lds + off);
+}
+
+// Bijective XCD-aware blockIdx remap (T1).
+__device__ __forceinline__ uint32_t xcd_swizzle(uint32_t bid, uint32_t nwg) {
+  uint32_t q = nwg / 8, r = nwg % 8;
+  uint32_t xcd = bid % 8, idx = bid / 8;
+  return (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+}
+
+// Dispatch a runtime Epi to a compile-time template instantiation.
+template <typename F>
+inline void epi_dispatch(int epi, F&& f) {
+  switch ((Epi)epi) {
+    case Epi::kNone: f(std::integral_constant<Epi, Epi::kNone>{}); break;
+    case Epi::kBias: f(std::integral_constant<Epi, Epi::kBias>{}); break;
+    case Epi::kBiasRelu: f(std::integral_constant<Epi, Epi::kBiasRelu>{}); break;
+    case Epi::kBiasGelu: f(std::integral_constant<Epi, Epi::kBiasGelu>{}); break;
+    case Epi::kScaleBias: f(std::integral_constant<Epi, Epi::kScaleBias>{}); break;
+    case Epi::kScaleBiasRelu: f(std::integral_constant<Epi, Epi::kScaleBiasRelu>{}); break;
+    case Epi::kScaleBiasAddRelu: f(std::integral_constant<Epi, Epi::kScaleBiasAddRelu>{}); break;
+  }
+}
+
+}  // namespace trtlab

@@ -1,0 +1,46 @@
+// trtlab_amd — kernel launcher declarations (implemented in kernels/*.hip).
+#pragma once
+#include <hip/hip_runtime.h>
+#include <cstdint>
+
+namespace trtlab {
+
+// dtype: 0 = fp16, 1 = bf16
+void launch_gemm_bt(int dtype, const void* A, const void* B, void* C,
+                    const float* scale, const float* bias, const void* residual,
+                    int M, int N, int K, int64_t lda, int64_t ldb, int64_t ldc,
+                    int epi, hipStream_t stream);
+
+void launch_conv2d(int dtype, const void* in, const void* Wt, void* out,
+                   const float* scale, const float* bias, const void* residual,
+                   const void* zero_page, int Nb, int H, int W, int C,
+                   int Cout, int KH, int KW, int sh, int sw, int ph, int pw,
+                   int epi, hipStream_t stream);
+
+void launch_maxpool2d(int dtype, const void* in, void* out, int Nb, int H,
+                      int W, int C, int KH, int KW, int sh, int sw, int ph,
+                      int pw, hipStream_t stream);
+void launch_gavgpool(int dtype, const void* in, void* out, int Nb, int HW,
+                     int C, hipStream_t stream);
+
+void launch_softmax_rows(int dtype, const void* in, void* out, int M, int N,
+                         int64_t ld, hipStream_t stream);
+void launch_layernorm(int dtype, const void* in, const float* gamma,
+                      const float* beta, void* out, int M, int N, int64_t ld,
+                      float eps, hipStream_t stream);
+void launch_add_layernorm(int dtype, const void* x, const void* res,
+                          const float* gamma, const float* beta, void* out,
+                          void* sum_out, int M, int N, int64_t ld, float eps,
+                          hipStream_t stream);
+
+void launch_elementwise(int dtype, int op, const void* a, const void* b,
+                        void* out, int64_t n, hipStream_t stream);
+void launch_channel_pad(int dtype, const void* in, void* out, int64_t M,
+                        int Cin, int Cpad, hipStream_t stream);
+void launch_cast(int dtype, bool to_f32, const void* in, void* out, int64_t n,
+                 hipStream_t stream);
+
+void launch_attention(int dtype, const void* qkv, void* out, int B, int S,
+                      int H, int D, float scale, hipStream_t stream);
+
+}  // namespace trtlab

@@ -1,0 +1,164 @@
+// trtlab_amd — graph-captured executor (MI355X-native replacement for the
+// reference's TensorRT workspace/enqueueV2 path, trtlab/tensorrt/src/
+// workspace.cc:21-75: bindings + activation scratch + warm-up + cudaGraph
+// capture + replay — here with our own op plan over CDNA4 kernels).
+#include "runtime.h"
+#include "../kernels/launchers.h"
+
+namespace trtlab {
+
+Engine::Engine(int device, const void* weights, size_t weight_bytes,
+               size_t arena_bytes, std::vector<OpDesc> ops, int64_t input_off,
+               size_t input_bytes, int64_t output_off, size_t output_bytes)
+    : device_(device),
+      weight_bytes_(weight_bytes),
+      arena_bytes_(arena_bytes),
+      ops_(std::move(ops)),
+      input_off_(input_off),
+      output_off_(output_off),
+      input_bytes_(input_bytes),
+      output_bytes_(output_bytes) {
+  TRT_HIP_CHECK(hipSetDevice(device_));
+  weights_ = device_malloc(weight_bytes_ ? weight_bytes_ : 256, device_);
+  if (weight_bytes_)
+    TRT_HIP_CHECK(hipMemcpy(weights_, weights, weight_bytes_, hipMemcpyHostToDevice));
+  zero_page_ = device_malloc(256, device_);
+  TRT_HIP_CHECK(hipMemset(zero_page_, 0, 256));
+  TRT_HIP_CHECK(hipDeviceSynchronize());
+}
+
+Engine::~Engine() {
+  device_free(weights_, weight_bytes_ ? weight_bytes_ : 256);
+  device_free(zero_page_, 256);
+}
+
+void Engine::upload_weights(const void* src, size_t bytes) {
+  if (bytes > weight_bytes_) throw std::runtime_error("upload_weights: too large");
+  TRT_HIP_CHECK(hipSetDevice(device_));
+  TRT_HIP_CHECK(hipMemcpy(weights_, src, bytes, hipMemcpyHostToDevice));
+}
+
+ExecutionContext::ExecutionContext(std::shared_ptr<Engine> engine)
+    : eng_(std::move(engine)) {
+  TRT_HIP_CHECK(hipSetDevice(eng_->device()));
+  arena_ = (char*)device_malloc(eng_->arena_bytes(), eng_->device());
+  h_in_ = (char*)pinned_malloc(eng_->input_bytes());
+  h_out_ = (char*)pinned_malloc(eng_->output_bytes());
+  TRT_HIP_CHECK(hipStreamCreateWithFlags(&stream_, hipStreamNonBlocking));
+}
+
+ExecutionContext::~ExecutionContext() {
+  if (graph_exec_) hipGraphExecDestroy(graph_exec_);
+  if (graph_) hipGraphDestroy(graph_);
+  hipStreamDestroy(stream_);
+  device_free(arena_, eng_->arena_bytes());
+  pinned_free(h_in_, eng_->input_bytes());
+  pinned_free(h_out_, eng_->output_bytes());
+}
+
+void ExecutionContext::enqueue_all(hipStream_t s) {
+  const char* wb = eng_->weights();
+  char* ar = arena_;
+  auto A = [&](int64_t off) -> void* { return off < 0 ? nullptr : ar + off; };
+  auto Wp = [&](int64_t off) -> const void* {
+    return off < 0 ? nullptr : wb + off;
+  };
+  auto Fp = [&](int64_t off) -> const float* {
+    return off < 0 ? nullptr : (const float*)(wb + off);
+  };
+
+  TRT_HIP_CHECK(hipMemcpyAsync(ar + eng_->input_off(), h_in_,
+                               eng_->input_bytes(), hipMemcpyHostToDevice, s));
+
+  for (const OpDesc& op : eng_->ops()) {
+    switch (op.kind) {
+      case kConv2d:
+        launch_conv2d(op.dtype, A(op.in_off), Wp(op.w_off), A(op.out_off),
+                      Fp(op.scale_off), Fp(op.bias_off), A(op.in2_off),
+                      eng_->zero_page(), op.Nb, op.H, op.W, op.C, op.Cout,
+                      op.KH, op.KW, op.sh, op.sw, op.ph, op.pw, op.epi, s);
+        break;
+      case kGemmBt:
+        launch_gemm_bt(op.dtype, A(op.in_off), Wp(op.w_off), A(op.out_off),
+                       Fp(op.scale_off), Fp(op.bias_off), A(op.in2_off), op.M,
+                       op.N, op.K, op.K /*lda*/, op.K /*ldb*/, op.N /*ldc*/,
+                       op.epi, s);
+        break;
+      case kMaxPool:
+        launch_maxpool2d(op.dtype, A(op.in_off), A(op.out_off), op.Nb, op.H,
+                         op.W, op.C, op.KH, op.KW, op.sh, op.sw, op.ph, op.pw,
+                         s);
+        break;
+      case kGAvgPool:
+        launch_gavgpool(op.dtype, A(op.in_off), A(op.out_off), op.Nb, op.HW,
+                        op.C, s);
+        break;
+      case kSoftmax:
+        launch_softmax_rows(op.dtype, A(op.in_off), A(op.out_off), op.M, op.N,
+                            op.N, s);
+        break;
+      case kLayerNorm:
+        launch_layernorm(op.dtype, A(op.in_off), Fp(op.scale_off),
+                         Fp(op.bias_off), A(op.out_off), op.M, op.N, op.N,
+                         op.eps, s);
+        break;
+      case kAddLayerNorm:
+        launch_add_layernorm(op.dtype, A(op.in_off), A(op.in2_off),
+                             Fp(op.scale_off), Fp(op.bias_off), A(op.out_off),
+                             A(op.out2_off), op.M, op.N, op.N, op.eps, s);
+        break;
+      case kElementwise:
+        launch_elementwise(op.dtype, op.epi, A(op.in_off), A(op.in2_off),
+                           A(op.out_off), op.n_elems, s);
+        break;
+      case kChannelPad:
+        launch_channel_pad(op.dtype, A(op.in_off), A(op.out_off), op.n_elems,
+                           op.C, op.Cout, s);
+        break;
+      case kAttention:
+        launch_attention(op.dtype, A(op.in_off), A(op.out_off), op.B, op.S,
+                         op.NH, op.HD, op.att_scale, s);
+        break;
+      default:
+        throw std::runtime_error("unknown op kind");
+    }
+  }
+
+  TRT_HIP_CHECK(hipMemcpyAsync(h_out_, ar + eng_->output_off(),
+                               eng_->output_bytes(), hipMemcpyDeviceToHost, s));
+}
+
+void ExecutionContext::capture() {
+  TRT_HIP_CHECK(hipSetDevice(eng_->device()));
+  // warm-up (module loading, autotune state) before capture — reference
+  // workspace.cc:47 does one enqueueV2 before cudaStreamBeginCapture.
+  enqueue_all(stream_);
+  TRT_HIP_CHECK(hipStreamSynchronize(stream_));
+  TRT_HIP_CHECK(hipStreamBeginCapture(stream_, hipStreamCaptureModeRelaxed));
+  enqueue_all(stream_);
+  TRT_HIP_CHECK(hipStreamEndCapture(stream_, &graph_));
+  TRT_HIP_CHECK(hipGraphInstantiate(&graph_exec_, graph_, nullptr, nullptr, 0));
+  captured_ = true;
+}
+
+void ExecutionContext::launch() {
+  if (captured_) {
+    TRT_HIP_CHECK(hipGraphLaunch(graph_exec_, stream_));
+  } else {
+    enqueue_all(stream_);
+  }
+}
+
+void ExecutionContext::synchronize() {
+  TRT_HIP_CHECK(hipStreamSynchronize(stream_));
+}
+
+bool ExecutionContext::ready() {
+  hipError_t e = hipStreamQuery(stream_);
+  if (e == hipSuccess) return true;
+  if (e == hipErrorNotReady) return false;
+  TRT_HIP_CHECK(e);
+  return false;
+}
+
+}  // namespace trtlab

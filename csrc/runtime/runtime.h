@@ -1,0 +1,141 @@
+// trtlab_amd — native runtime: memory primitives + graph-captured executor.
+#pragma once
+#include "../common.h"
+
+#include <atomic>
+#include <memory>
+#include <mutex>
+#include <vector>
+
+namespace trtlab {
+
+// ------------------------------------------------------------------ memory
+void* device_malloc(size_t bytes, int device);
+void device_free(void* p, size_t bytes);
+void* pinned_malloc(size_t bytes);
+void pinned_free(void* p, size_t bytes);
+int64_t device_bytes_in_use();
+int64_t pinned_bytes_in_use();
+
+class BlockPool {
+ public:
+  BlockPool(size_t block_bytes, int count, int device);
+  ~BlockPool();
+  void* acquire();
+  void release(void* p);
+  int available();
+  size_t block_bytes() const { return block_bytes_; }
+  int total() const { return total_; }
+
+ private:
+  size_t block_bytes_;
+  int device_;
+  char* base_;
+  int total_;
+  std::vector<char*> free_;
+  std::mutex mu_;
+};
+
+// ---------------------------------------------------------------- executor
+// Op kinds executed by the engine (planned host-side in Python).
+enum OpKind : int {
+  kConv2d = 0,
+  kGemmBt = 1,
+  kMaxPool = 2,
+  kGAvgPool = 3,
+  kSoftmax = 4,
+  kLayerNorm = 5,
+  kAddLayerNorm = 6,
+  kElementwise = 7,
+  kChannelPad = 8,
+  kAttention = 9,
+};
+
+struct OpDesc {
+  int kind = 0;
+  int dtype = 0;  // 0 fp16, 1 bf16
+  int epi = 0;    // epilogue enum / elementwise op code
+  // arena offsets in bytes (-1 = absent)
+  int64_t in_off = -1, in2_off = -1, out_off = -1, out2_off = -1;
+  // weight-blob offsets in bytes (-1 = absent)
+  int64_t w_off = -1, scale_off = -1, bias_off = -1;
+  // gemm
+  int M = 0, N = 0, K = 0;
+  // conv / pool
+  int Nb = 0, H = 0, W = 0, C = 0, Cout = 0, KH = 0, KW = 0, sh = 1, sw = 1,
+      ph = 0, pw = 0, HW = 0;
+  // norms / elementwise
+  float eps = 1e-5f;
+  int64_t n_elems = 0;
+  // attention
+  int B = 0, S = 0, NH = 0, HD = 0;
+  float att_scale = 1.0f;
+};
+
+// A compiled model: weight blob on device + op list + arena layout.
+// Replaces the reference's Model/ICudaEngine (trtlab/tensorrt/model.h:17):
+// here the "engine" is an explicit op plan over hand-written CDNA4 kernels.
+class Engine {
+ public:
+  Engine(int device, const void* weights, size_t weight_bytes,
+         size_t arena_bytes, std::vector<OpDesc> ops, int64_t input_off,
+         size_t input_bytes, int64_t output_off, size_t output_bytes);
+  ~Engine();
+
+  int device() const { return device_; }
+  size_t arena_bytes() const { return arena_bytes_; }
+  size_t input_bytes() const { return input_bytes_; }
+  size_t output_bytes() const { return output_bytes_; }
+  int64_t input_off() const { return input_off_; }
+  int64_t output_off() const { return output_off_; }
+  const std::vector<OpDesc>& ops() const { return ops_; }
+  const char* weights() const { return (const char*)weights_; }
+  const char* zero_page() const { return (const char*)zero_page_; }
+  // Overwrite the device weight blob (RCCL broadcast target, tests).
+  void upload_weights(const void* src, size_t bytes);
+  uintptr_t weights_ptr() const { return (uintptr_t)weights_; }
+  size_t weight_bytes() const { return weight_bytes_; }
+
+ private:
+  int device_;
+  void* weights_ = nullptr;
+  size_t weight_bytes_ = 0;
+  void* zero_page_ = nullptr;
+  size_t arena_bytes_;
+  std::vector<OpDesc> ops_;
+  int64_t input_off_, output_off_;
+  size_t input_bytes_, output_bytes_;
+};
+
+// Per-request execution context: private stream + activation arena + pinned
+// staging + captured hipGraph. Mirrors the reference's
+// StaticSingleModelGraphWorkspace (trtlab/tensorrt/src/workspace.cc:21-75):
+// allocate bindings/scratch, warm up, capture enqueue into a graph, replay.
+class ExecutionContext {
+ public:
+  ExecutionContext(std::shared_ptr<Engine> engine);
+  ~ExecutionContext();
+
+  // Raw host staging buffers (pinned) for input/output bindings.
+  uintptr_t host_input_ptr() const { return (uintptr_t)h_in_; }
+  uintptr_t host_output_ptr() const { return (uintptr_t)h_out_; }
+  uintptr_t arena_ptr() const { return (uintptr_t)arena_; }
+
+  void enqueue_all(hipStream_t s);  // H2D + ops + D2H on stream s
+  void capture();                   // warm-up + hipGraph capture
+  void launch();                    // graph launch (or eager enqueue)
+  void synchronize();
+  bool ready();  // hipStreamQuery == success
+
+ private:
+  std::shared_ptr<Engine> eng_;
+  char* arena_ = nullptr;
+  char* h_in_ = nullptr;
+  char* h_out_ = nullptr;
+  hipStream_t stream_{};
+  hipGraph_t graph_{};
+  hipGraphExec_t graph_exec_{};
+  bool captured_ = false;
+};
+
+}  // namespace trtlab

@@ -1,0 +1,86 @@
+import time
+from concurrent.futures import Future
+
+import pytest
+
+from trtlab_amd.core import (Dispatcher, DeferredShortTaskPool, Pool,
+                             StandardBatcher, ThreadPool)
+
+
+def test_thread_pool_basic():
+    tp = ThreadPool(4, "t")
+    futs = [tp.enqueue(lambda i=i: i * i) for i in range(32)]
+    assert [f.result(timeout=5) for f in futs] == [i * i for i in range(32)]
+    tp.shutdown()
+
+
+def test_thread_pool_exception():
+    tp = ThreadPool(1)
+    f = tp.enqueue(lambda: 1 / 0)
+    with pytest.raises(ZeroDivisionError):
+        f.result(timeout=5)
+    tp.shutdown()
+
+
+def test_pool_checkout_returns():
+    pool = Pool([1, 2])
+    co1 = pool.pop()
+    co2 = pool.pop()
+    assert pool.available == 0
+    with pytest.raises(TimeoutError):
+        pool.pop(timeout=0.05)
+    co1.release()
+    assert pool.available == 1
+    with pool.pop() as item:
+        assert item in (1, 2)
+    assert pool.available == 1
+    co2.release()
+    assert pool.available == 2
+
+
+def test_pool_on_return_hook():
+    seen = []
+    pool = Pool(["x"])
+    co = pool.pop(on_return=lambda it: seen.append(it))
+    co.release()
+    assert seen == ["x"]
+
+
+def test_batcher_closes_on_max_size():
+    b = StandardBatcher(max_batch_size=3, timeout_s=10)
+    f1, c1 = b.enqueue("a")
+    f2, c2 = b.enqueue("b")
+    assert c1 is None and c2 is None
+    f3, c3 = b.enqueue("c")
+    assert c3 is not None and c3.items == ["a", "b", "c"]
+    assert b.open_batch is None
+
+
+def test_deferred_task_pool_ordering():
+    pool = DeferredShortTaskPool()
+    out = []
+    now = time.monotonic()
+    pool.enqueue_deferred(now + 0.10, lambda: out.append(2))
+    pool.enqueue_deferred(now + 0.03, lambda: out.append(1))
+    time.sleep(0.3)
+    assert out == [1, 2]
+    pool.shutdown()
+
+
+def test_dispatcher_batches_by_size_and_timeout():
+    calls = []
+
+    def compute(items):
+        calls.append(list(items))
+        return [i * 10 for i in items]
+
+    d = Dispatcher(max_batch_size=2, timeout_s=0.05, compute_batch_fn=compute)
+    f1 = d.enqueue(1)
+    f2 = d.enqueue(2)  # closes by size
+    assert f1.result(timeout=5) == 10
+    assert f2.result(timeout=5) == 20
+    f3 = d.enqueue(3)  # closes by timeout
+    assert f3.result(timeout=5) == 30
+    assert calls[0] == [1, 2]
+    assert calls[1] == [3]
+    d.shutdown()

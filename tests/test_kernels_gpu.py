@@ -1,0 +1,231 @@
+"""GPU kernel numerics: each HIP kernel vs a plain PyTorch fp32 reference.
+
+All tensors fp16 on device; references computed in fp32. Asymmetric random
+inputs (transpose-detecting — cdna guide §5.4 rule 16).
+"""
+import numpy as np
+import pytest
+import torch
+
+import trtlab_amd
+
+pytestmark = pytest.mark.gpu
+
+# epilogue codes (csrc/kernels/gemm_common.h)
+EPI_NONE, EPI_BIAS, EPI_BIAS_RELU, EPI_BIAS_GELU = 0, 1, 2, 3
+EPI_SB, EPI_SB_RELU, EPI_SB_ADD_RELU = 4, 5, 6
+
+
+@pytest.fixture(scope="module")
+def C():
+    return trtlab_amd.native()
+
+
+def t16(*shape, seed=0, scale=1.0):
+    g = torch.Generator(device="cuda").manual_seed(seed)
+    return (torch.randn(*shape, generator=g, device="cuda") * scale).half().contiguous()
+
+
+def check(out, ref, rtol=2e-2, atol=2e-2, frac=1e-3):
+    out = out.float().cpu()
+    ref = ref.float().cpu()
+    err = (out - ref).abs()
+    tol = atol + rtol * ref.abs()
+    bad = (err > tol).float().mean().item()
+    assert bad <= frac, f"mismatch frac {bad}: max err {err.max()} vs {ref.abs().max()}"
+
+
+@pytest.mark.parametrize("M,N,K", [(128, 128, 64), (256, 384, 128),
+                                   (1000, 1000, 256), (100, 64, 2048),
+                                   (8, 1000, 2048)])
+def test_gemm_bt(C, M, N, K):
+    a = t16(M, K, seed=M + N)
+    b = t16(N, K, seed=M + N + 1)
+    out = torch.empty(M, N, dtype=torch.half, device="cuda")
+    torch.cuda.synchronize()
+    C.ops.gemm_bt(0, a.data_ptr(), b.data_ptr(), out.data_ptr(),
+                  M=M, N=N, K=K, epi=EPI_NONE)
+    ref = a.float() @ b.float().t()
+    check(out, ref)
+
+
+def test_gemm_bt_bias_relu(C):
+    M, N, K = 200, 300, 192
+    a, b = t16(M, K, seed=1), t16(N, K, seed=2)
+    bias = torch.randn(N, device="cuda").float().contiguous()
+    out = torch.empty(M, N, dtype=torch.half, device="cuda")
+    torch.cuda.synchronize()
+    C.ops.gemm_bt(0, a.data_ptr(), b.data_ptr(), out.data_ptr(),
+                  bias=bias.data_ptr(), M=M, N=N, K=K, epi=EPI_BIAS_RELU)
+    ref = torch.relu(a.float() @ b.float().t() + bias)
+    check(out, ref)
+
+
+def test_gemm_bt_bias_gelu(C):
+    M, N, K = 128, 256, 128
+    a, b = t16(M, K, seed=3), t16(N, K, seed=4)
+    bias = torch.randn(N, device="cuda").float().contiguous()
+    out = torch.empty(M, N, dtype=torch.half, device="cuda")
+    torch.cuda.synchronize()
+    C.ops.gemm_bt(0, a.data_ptr(), b.data_ptr(), out.data_ptr(),
+                  bias=bias.data_ptr(), M=M, N=N, K=K, epi=EPI_BIAS_GELU)
+    ref = torch.nn.functional.gelu(a.float() @ b.float().t() + bias,
+                                   approximate="tanh")
+    check(out, ref)
+
+
+def test_gemm_bt_bf16(C):
+    M, N, K = 256, 256, 128
+    g = torch.Generator(device="cuda").manual_seed(9)
+    a = torch.randn(M, K, generator=g, device="cuda").bfloat16().contiguous()
+    b = torch.randn(N, K, generator=g, device="cuda").bfloat16().contiguous()
+    out = torch.empty(M, N, dtype=torch.bfloat16, device="cuda")
+    torch.cuda.synchronize()
+    C.ops.gemm_bt(1, a.data_ptr(), b.data_ptr(), out.data_ptr(),
+                  M=M, N=N, K=K, epi=EPI_NONE)
+    ref = a.float() @ b.float().t()
+    check(out, ref, rtol=4e-2, atol=4e-2)
+
+
+def _conv_ref(x, w, stride, pad, scale=None, bias=None, res=None, epi=EPI_NONE):
+    xc = x.float().permute(0, 3, 1, 2)
+    wc = w.float()  # [Cout, Cin, KH, KW]
+    y = torch.nn.functional.conv2d(xc, wc, stride=stride, padding=pad)
+    y = y.permute(0, 2, 3, 1)
+    if epi in (EPI_SB, EPI_SB_RELU, EPI_SB_ADD_RELU):
+        y = y * scale + bias
+    if epi == EPI_SB_ADD_RELU:
+        y = y + res.float()
+    if epi in (EPI_SB_RELU, EPI_SB_ADD_RELU):
+        y = torch.relu(y)
+    return y
+
+
+def _pack_w(w):
+    # [Cout, Cin, KH, KW] -> [Cout][KH*KW*Cin] padded to K%64
+    cout, cin, kh, kw = w.shape
+    flat = w.permute(0, 2, 3, 1).reshape(cout, kh * kw * cin)
+    kp = ((kh * kw * cin + 63) // 64) * 64
+    if kp != flat.shape[1]:
+        flat = torch.nn.functional.pad(flat, (0, kp - flat.shape[1]))
+    return flat.half().contiguous()
+
+
+@pytest.mark.parametrize("shape", [
+    # (Nb, H, W, Cin, Cout, KH, stride, pad)
+    (2, 16, 16, 64, 128, 1, 1, 0),       # 1x1
+    (2, 16, 16, 64, 64, 3, 1, 1),        # 3x3 s1
+    (2, 32, 32, 64, 128, 3, 2, 1),       # 3x3 s2
+    (2, 16, 16, 128, 64, 1, 2, 0),       # 1x1 s2 downsample
+    (1, 56, 56, 8, 64, 7, 2, 3),         # stem-like 7x7 (C=8 padded)
+])
+def test_conv2d(C, shape):
+    nb, h, w_, cin, cout, k, s, p = shape
+    x = t16(nb, h, w_, cin, seed=sum(shape))
+    wt = t16(cout, cin, k, k, seed=sum(shape) + 1, scale=0.1)
+    zero = torch.zeros(64, dtype=torch.half, device="cuda")
+    oh = (h + 2 * p - k) // s + 1
+    ow = (w_ + 2 * p - k) // s + 1
+    out = torch.empty(nb, oh, ow, cout, dtype=torch.half, device="cuda")
+    wp = _pack_w(wt)
+    torch.cuda.synchronize()
+    C.ops.conv2d(0, x.data_ptr(), wp.data_ptr(), out.data_ptr(),
+                 zero_page=zero.data_ptr(), Nb=nb, H=h, W=w_, C=cin,
+                 Cout=cout, KH=k, KW=k, sh=s, sw=s, ph=p, pw=p, epi=EPI_NONE)
+    ref = _conv_ref(x, wt, s, p)
+    check(out, ref)
+
+
+def test_conv2d_fused_bn_add_relu(C):
+    nb, h, w_, cin, cout = 2, 14, 14, 256, 256
+    x = t16(nb, h, w_, cin, seed=42)
+    wt = t16(cout, cin, 3, 3, seed=43, scale=0.05)
+    scale = torch.rand(cout, device="cuda").float() + 0.5
+    bias = torch.randn(cout, device="cuda").float() * 0.1
+    res = t16(nb, h, w_, cout, seed=44)
+    zero = torch.zeros(64, dtype=torch.half, device="cuda")
+    out = torch.empty(nb, h, w_, cout, dtype=torch.half, device="cuda")
+    wp = _pack_w(wt)
+    torch.cuda.synchronize()
+    C.ops.conv2d(0, x.data_ptr(), wp.data_ptr(), out.data_ptr(),
+                 scale=scale.data_ptr(), bias=bias.data_ptr(),
+                 residual=res.data_ptr(), zero_page=zero.data_ptr(),
+                 Nb=nb, H=h, W=w_, C=cin, Cout=cout, KH=3, KW=3, sh=1, sw=1,
+                 ph=1, pw=1, epi=EPI_SB_ADD_RELU)
+    ref = _conv_ref(x, wt, 1, 1, scale.cpu(), bias.cpu(), res, EPI_SB_ADD_RELU)
+    check(out, ref)
+
+
+def test_maxpool(C):
+    x = t16(2, 32, 32, 64, seed=5)
+    out = torch.empty(2, 16, 16, 64, dtype=torch.half, device="cuda")
+    torch.cuda.synchronize()
+    C.ops.maxpool2d(0, x.data_ptr(), out.data_ptr(), 2, 32, 32, 64, 3, 3,
+                    2, 2, 1, 1)
+    ref = torch.nn.functional.max_pool2d(
+        x.float().permute(0, 3, 1, 2), 3, stride=2, padding=1
+    ).permute(0, 2, 3, 1)
+    check(out, ref, rtol=1e-3, atol=1e-3)
+
+
+def test_gavgpool(C):
+    x = t16(4, 7, 7, 2048, seed=6)
+    out = torch.empty(4, 2048, dtype=torch.half, device="cuda")
+    torch.cuda.synchronize()
+    C.ops.gavgpool(0, x.data_ptr(), out.data_ptr(), 4, 49, 2048)
+    ref = x.float().reshape(4, 49, 2048).mean(dim=1)
+    check(out, ref, rtol=1e-2, atol=1e-3)
+
+
+def test_softmax_rows(C):
+    x = t16(100, 1000, seed=7, scale=3.0)
+    out = torch.empty_like(x)
+    torch.cuda.synchronize()
+    C.ops.softmax_rows(0, x.data_ptr(), out.data_ptr(), 100, 1000)
+    ref = torch.softmax(x.float(), dim=-1)
+    check(out, ref, rtol=1e-2, atol=1e-4)
+
+
+def test_layernorm(C):
+    M, N = 256, 768
+    x = t16(M, N, seed=8, scale=2.0)
+    g_ = (torch.rand(N, device="cuda") + 0.5).float()
+    b_ = torch.randn(N, device="cuda").float()
+    out = torch.empty_like(x)
+    torch.cuda.synchronize()
+    C.ops.layernorm(0, x.data_ptr(), g_.data_ptr(), b_.data_ptr(),
+                    out.data_ptr(), M, N, 1e-5)
+    ref = torch.nn.functional.layer_norm(x.float(), (N,), g_, b_, 1e-5)
+    check(out, ref)
+
+
+def test_add_layernorm(C):
+    M, N = 256, 768
+    x, r = t16(M, N, seed=9), t16(M, N, seed=10)
+    g_ = (torch.rand(N, device="cuda") + 0.5).float()
+    b_ = torch.randn(N, device="cuda").float()
+    out = torch.empty_like(x)
+    torch.cuda.synchronize()
+    C.ops.add_layernorm(0, x.data_ptr(), r.data_ptr(), g_.data_ptr(),
+                        b_.data_ptr(), out.data_ptr(), 0, M, N, 1e-5)
+    ref = torch.nn.functional.layer_norm((x + r).float(), (N,), g_, b_, 1e-5)
+    check(out, ref)
+
+
+def test_elementwise_add_relu(C):
+    a, b = t16(64, 128, seed=11), t16(64, 128, seed=12)
+    out = torch.empty_like(a)
+    torch.cuda.synchronize()
+    C.ops.elementwise(0, 3, a.data_ptr(), b.data_ptr(), out.data_ptr(),
+                      a.numel())
+    ref = torch.relu(a.float() + b.float())
+    check(out, ref, rtol=1e-3, atol=1e-3)
+
+
+def test_channel_pad(C):
+    x = t16(100, 3, seed=13)
+    out = torch.empty(100, 8, dtype=torch.half, device="cuda")
+    torch.cuda.synchronize()
+    C.ops.channel_pad(0, x.data_ptr(), out.data_ptr(), 100, 3, 8)
+    ref = torch.nn.functional.pad(x.float(), (0, 5))
+    check(out, ref, rtol=1e-3, atol=1e-3)

@@ -1,0 +1,56 @@
+import numpy as np
+import pytest
+
+from trtlab_amd.memory import ArenaPlanner, TransactionalStack
+
+
+def test_transactional_stack():
+    st = TransactionalStack(10240, alignment=256)
+    a = st.allocate(100)
+    assert a == 0
+    st.begin()
+    b = st.allocate(100)
+    assert b == 256
+    st.commit()
+    c = st.allocate(100)
+    assert c == 256  # reused after transaction rollback
+    with pytest.raises(MemoryError):
+        st.allocate(1 << 20)
+
+
+def _check_no_overlap(items, offsets):
+    # items: list of (name, size, start, end)
+    for i, (n1, s1, a1, b1) in enumerate(items):
+        for n2, s2, a2, b2 in items[i + 1:]:
+            if b1 < a2 or b2 < a1:
+                continue  # lifetimes disjoint
+            o1, o2 = offsets[n1], offsets[n2]
+            assert o1 + s1 <= o2 or o2 + s2 <= o1, f"overlap {n1} {n2}"
+
+
+def test_arena_planner_no_overlap():
+    rng = np.random.RandomState(0)
+    items = []
+    planner = ArenaPlanner(alignment=256)
+    for i in range(200):
+        size = int(rng.randint(1, 1 << 16))
+        start = int(rng.randint(0, 50))
+        end = start + int(rng.randint(0, 10))
+        name = f"t{i}"
+        items.append((name, ((size + 255) // 256) * 256, start, end))
+        planner.add(name, size, start, end)
+    offsets, total = planner.plan()
+    assert total % 256 == 0
+    _check_no_overlap(items, offsets)
+    # reuse should beat the no-reuse sum
+    assert total < sum(s for _, s, _, _ in items)
+
+
+def test_arena_planner_serial_chain_reuses():
+    planner = ArenaPlanner()
+    # a -> b -> c : a dies when b is made, so c can reuse a's space
+    planner.add("a", 1000, 0, 1)
+    planner.add("b", 1000, 1, 2)
+    planner.add("c", 1000, 2, 3)
+    offsets, total = planner.plan()
+    assert total <= 2 * 1024

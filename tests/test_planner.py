@@ -1,0 +1,132 @@
+"""Planner/IR tests (CPU): fusion correctness, liveness safety, and the
+fused CPU reference vs an unfused node-level interpretation."""
+import numpy as np
+import pytest
+
+from trtlab_amd.engine.calibrate import calibrate_bn
+from trtlab_amd.engine.ir import Graph
+from trtlab_amd.engine.planner import (EPI_SCALE_BIAS, EPI_SCALE_BIAS_ADD_RELU,
+                                       EPI_SCALE_BIAS_RELU, K_CHANNEL_PAD,
+                                       K_CONV, K_GEMM, Planner)
+from trtlab_amd.engine.reference import run_reference
+from trtlab_amd.models import build_resnet
+
+
+@pytest.fixture(scope="module")
+def rn50_plan():
+    g = build_resnet(50, batch=2, seed=0)
+    return g, Planner().compile(g)
+
+
+def test_resnet50_fusion_counts(rn50_plan):
+    g, plan = rn50_plan
+    kinds = [d["kind"] for d in plan.ops]
+    assert kinds.count(K_CONV) == 53  # 1 stem + 16*3 bottleneck + 4 downsample
+    assert kinds.count(K_CHANNEL_PAD) == 1
+    assert kinds.count(K_GEMM) == 1
+    # every bottleneck's last conv carries the fused residual-add + relu
+    adds = [d for d in plan.ops if d.get("epi") == EPI_SCALE_BIAS_ADD_RELU]
+    assert len(adds) == 16
+    # downsample convs are plain scale+bias
+    plain = [d for d in plan.ops if d.get("epi") == EPI_SCALE_BIAS]
+    assert len(plain) == 4
+    relu = [d for d in plan.ops if d.get("epi") == EPI_SCALE_BIAS_RELU]
+    assert len(relu) == 33  # stem + 2 per bottleneck
+
+
+def test_conv_k_padding(rn50_plan):
+    g, plan = rn50_plan
+    stem = next(d for d in plan.ops if d["kind"] == K_CONV)
+    assert stem["C"] == 8  # padded from 3
+    assert stem["KH"] == 7
+
+
+def test_arena_liveness_safe(rn50_plan):
+    """No two tensors with overlapping lifetimes share arena bytes."""
+    g, plan = rn50_plan
+    # rebuild intervals exactly as the planner does
+    touched = {}
+
+    def touch(t, i):
+        s, e = touched.get(t, (i, i))
+        touched[t] = (min(s, i), max(e, i))
+
+    touch(plan.input_name, 0)
+    for i, op in enumerate(plan.exec_ops):
+        for t in op.inputs:
+            touch(t, i)
+        touch(op.output, i)
+    s, e = touched[plan.output_name]
+    touched[plan.output_name] = (s, len(plan.exec_ops))
+
+    def nbytes(t):
+        # find a shape: inputs/outputs of ops
+        import math
+        for op, d in zip(plan.exec_ops, plan.ops):
+            pass
+        return None
+
+    names = list(touched)
+    for i, n1 in enumerate(names):
+        for n2 in names[i + 1:]:
+            a1, b1 = touched[n1]
+            a2, b2 = touched[n2]
+            if b1 < a2 or b2 < a1:
+                continue
+            # overlapping lifetime -> distinct offsets required
+            assert plan.offsets[n1] != plan.offsets[n2], (n1, n2)
+
+
+def test_fused_reference_matches_unfused_interpreter():
+    """run_reference (fused exec ops, prepacked fp16 weights) must match the
+    node-level fp32 interpreter within fp16-weight-quantization tolerance."""
+    g = build_resnet(50, batch=2, seed=3)
+    plan = Planner().compile(g)
+    x = np.random.RandomState(7).randn(2, 224, 224, 3).astype(np.float32) * 0.5
+
+    out_fused = run_reference(plan, x)
+
+    # unfused node-level fp32 interpretation with the SAME (build-time) BN
+    # stats the plan was compiled from
+    import torch
+    import torch.nn.functional as F
+
+    g2 = g
+    t = {g2.input_name: torch.from_numpy(x).float()}
+    for n in g2.nodes:
+        if n.kind == "input":
+            continue
+        xx = t[n.inputs[0]]
+        if n.kind == "conv2d":
+            w = torch.from_numpy(n.attrs["weight"])
+            y = F.conv2d(xx.permute(0, 3, 1, 2), w, stride=n.attrs["stride"],
+                         padding=n.attrs["padding"]).permute(0, 2, 3, 1)
+        elif n.kind == "batchnorm":
+            a = n.attrs
+            y = (xx - torch.from_numpy(a["mean"])) / torch.sqrt(
+                torch.from_numpy(a["var"]) + a["eps"])
+            y = y * torch.from_numpy(a["gamma"]) + torch.from_numpy(a["beta"])
+        elif n.kind == "relu":
+            y = F.relu(xx)
+        elif n.kind == "add":
+            y = xx + t[n.inputs[1]]
+        elif n.kind == "maxpool":
+            y = F.max_pool2d(xx.permute(0, 3, 1, 2), n.attrs["kernel"],
+                             stride=n.attrs["stride"],
+                             padding=n.attrs["padding"]).permute(0, 2, 3, 1)
+        elif n.kind == "gavgpool":
+            nb, h, w_, c = xx.shape
+            y = xx.reshape(nb, h * w_, c).mean(1)
+        elif n.kind == "gemm":
+            y = xx @ torch.from_numpy(n.attrs["weight"]).t()
+            if n.attrs.get("bias") is not None:
+                y = y + torch.from_numpy(n.attrs["bias"])
+        else:
+            raise AssertionError(n.kind)
+        t[n.output] = y
+    out_unfused = t[g2.output_name].numpy()
+
+    # fp16 weight quantization in the fused path -> modest tolerance
+    err = np.abs(out_fused - out_unfused).max()
+    scale = np.abs(out_unfused).max()
+    assert err / scale < 0.05, (err, scale)

@@ -1,0 +1,159 @@
+"""Model IR: a small typed graph of inference ops, NHWC layout, fp16 compute.
+
+The IR is built by model builders (trtlab_amd.models) or the ONNX importer,
+then lowered by the Planner (fusion + memory planning) into an executable
+plan. Replaces the reference's reliance on nvinfer1::ICudaEngine
+(trtlab/tensorrt/model.h:17) with an explicit, inspectable graph.
+"""
+from __future__ import annotations
+
+from dataclasses import dataclass, field
+from typing import Any, Dict, List, Optional, Tuple
+
+import numpy as np
+
+
+@dataclass
+class TensorSpec:
+    name: str
+    shape: Tuple[int, ...]  # NHWC for 4-D activations, [M, K] for 2-D
+    dtype: str = "f16"
+
+    @property
+    def numel(self) -> int:
+        n = 1
+        for s in self.shape:
+            n *= int(s)
+        return n
+
+    @property
+    def nbytes(self) -> int:
+        itemsize = {"f16": 2, "bf16": 2, "f32": 4}[self.dtype]
+        return self.numel * itemsize
+
+
+@dataclass
+class Node:
+    kind: str                      # conv2d | batchnorm | relu | gelu | add |
+    #                                maxpool | gavgpool | flatten | gemm |
+    #                                softmax | layernorm | add_layernorm |
+    #                                attention | input
+    name: str
+    inputs: List[str]
+    output: str
+    attrs: Dict[str, Any] = field(default_factory=dict)
+
+
+class Graph:
+    """Topologically-ordered op list with named tensors."""
+
+    def __init__(self, name: str = "model"):
+        self.name = name
+        self.nodes: List[Node] = []
+        self.tensors: Dict[str, TensorSpec] = {}
+        self.input_name: Optional[str] = None
+        self.output_name: Optional[str] = None
+        self._ctr = 0
+
+    # ---------------------------------------------------------- builders
+    def _fresh(self, prefix: str) -> str:
+        self._ctr += 1
+        return f"{prefix}_{self._ctr}"
+
+    def add_tensor(self, name: str, shape: Tuple[int, ...], dtype: str = "f16") -> str:
+        self.tensors[name] = TensorSpec(name, tuple(int(s) for s in shape), dtype)
+        return name
+
+    def input(self, shape: Tuple[int, ...], name: str = "input") -> str:
+        self.add_tensor(name, shape)
+        self.nodes.append(Node("input", name, [], name))
+        self.input_name = name
+        return name
+
+    def _emit(self, kind: str, inputs: List[str], out_shape: Tuple[int, ...],
+              attrs: Dict[str, Any], name: Optional[str] = None) -> str:
+        name = name or self._fresh(kind)
+        out = self.add_tensor(name, out_shape)
+        self.nodes.append(Node(kind, name, list(inputs), out, attrs))
+        self.output_name = out
+        return out
+
+    def conv2d(self, x: str, weight: np.ndarray, stride: int = 1,
+               padding: int = 0, name: Optional[str] = None) -> str:
+        n, h, w, c = self.tensors[x].shape
+        cout, cin, kh, kw = weight.shape
+        assert cin == c, f"conv2d: Cin {cin} != input C {c}"
+        oh = (h + 2 * padding - kh) // stride + 1
+        ow = (w + 2 * padding - kw) // stride + 1
+        return self._emit("conv2d", [x], (n, oh, ow, cout),
+                          dict(weight=np.asarray(weight, np.float32),
+                               stride=stride, padding=padding), name)
+
+    def batchnorm(self, x: str, gamma, beta, mean, var, eps: float = 1e-5,
+                  name: Optional[str] = None) -> str:
+        return self._emit("batchnorm", [x], self.tensors[x].shape,
+                          dict(gamma=np.asarray(gamma, np.float32),
+                               beta=np.asarray(beta, np.float32),
+                               mean=np.asarray(mean, np.float32),
+                               var=np.asarray(var, np.float32), eps=eps), name)
+
+    def relu(self, x: str, name: Optional[str] = None) -> str:
+        return self._emit("relu", [x], self.tensors[x].shape, {}, name)
+
+    def gelu(self, x: str, name: Optional[str] = None) -> str:
+        return self._emit("gelu", [x], self.tensors[x].shape, {}, name)
+
+    def add(self, a: str, b: str, name: Optional[str] = None) -> str:
+        return self._emit("add", [a, b], self.tensors[a].shape, {}, name)
+
+    def maxpool(self, x: str, kernel: int, stride: int, padding: int = 0,
+                name: Optional[str] = None) -> str:
+        n, h, w, c = self.tensors[x].shape
+        oh = (h + 2 * padding - kernel) // stride + 1
+        ow = (w + 2 * padding - kernel) // stride + 1
+        return self._emit("maxpool", [x], (n, oh, ow, c),
+                          dict(kernel=kernel, stride=stride, padding=padding),
+                          name)
+
+    def global_avgpool(self, x: str, name: Optional[str] = None) -> str:
+        n, h, w, c = self.tensors[x].shape
+        return self._emit("gavgpool", [x], (n, c), dict(hw=h * w), name)
+
+    def gemm(self, x: str, weight: np.ndarray, bias: Optional[np.ndarray] = None,
+             name: Optional[str] = None) -> str:
+        m, k = self.tensors[x].shape
+        nout, kin = weight.shape
+        assert kin == k, f"gemm: K {kin} != input {k}"
+        return self._emit("gemm", [x], (m, nout),
+                          dict(weight=np.asarray(weight, np.float32),
+                               bias=None if bias is None else np.asarray(bias, np.float32)),
+                          name)
+
+    def softmax(self, x: str, name: Optional[str] = None) -> str:
+        return self._emit("softmax", [x], self.tensors[x].shape, {}, name)
+
+    def layernorm(self, x: str, gamma, beta, eps: float = 1e-5,
+                  name: Optional[str] = None) -> str:
+        return self._emit("layernorm", [x], self.tensors[x].shape,
+                          dict(gamma=np.asarray(gamma, np.float32),
+                               beta=np.asarray(beta, np.float32), eps=eps),
+                          name)
+
+    def add_layernorm(self, x: str, res: str, gamma, beta, eps: float = 1e-5,
+                      name: Optional[str] = None) -> str:
+        return self._emit("add_layernorm", [x, res], self.tensors[x].shape,
+                          dict(gamma=np.asarray(gamma, np.float32),
+                               beta=np.asarray(beta, np.float32), eps=eps),
+                          name)
+
+    def attention(self, qkv: str, heads: int, seq: int,
+                  name: Optional[str] = None) -> str:
+        m, k3 = self.tensors[qkv].shape
+        hid = k3 // 3
+        return self._emit("attention", [qkv], (m, hid),
+                          dict(heads=heads, seq=seq, head_dim=hid // heads),
+                          name)
+
+    # ------------------------------------------------------------ helpers
+    def users(self, tensor: str) -> List[Node]:
+        return [n for n in self.nodes if tensor in n.inputs]

@@ -1,0 +1,379 @@
+"""Planner: IR graph -> executable plan.
+
+Passes (mirrors what TensorRT did internally for the reference, made
+explicit and MI355X-shaped):
+  1. fusion — conv2d+BN(+ReLU)(+residual add) into one implicit-GEMM kernel
+     with a fused epilogue; gemm+bias(+ReLU/GeLU) likewise.
+  2. legalization — pad input channels to C % 8 == 0 (the implicit-GEMM
+     staging loads 16 B per lane), assert K % 64 == 0 for GEMM paths.
+  3. weight prepacking — conv [Cout,Cin,KH,KW] -> [Cout][KH*KW*Cpad] fp16
+     "bt" layout (K padded to 64); BN folded to per-channel scale/bias fp32.
+  4. memory planning — liveness intervals + best-fit arena offsets
+     (trtlab_amd.memory.ArenaPlanner): the activation-arena contract of the
+     reference (workspace.cc:40-41 setDeviceMemory).
+"""
+from __future__ import annotations
+
+from dataclasses import dataclass, field
+from typing import Any, Dict, List, Optional, Tuple
+
+import numpy as np
+
+from trtlab_amd.engine.ir import Graph, Node
+from trtlab_amd.memory import ArenaPlanner
+from trtlab_amd.utils import round_up
+
+# epilogue codes — keep in sync with csrc/kernels/gemm_common.h Epi
+EPI_NONE = 0
+EPI_BIAS = 1
+EPI_BIAS_RELU = 2
+EPI_BIAS_GELU = 3
+EPI_SCALE_BIAS = 4
+EPI_SCALE_BIAS_RELU = 5
+EPI_SCALE_BIAS_ADD_RELU = 6
+
+# op kinds — keep in sync with csrc/runtime/runtime.h OpKind
+K_CONV, K_GEMM, K_MAXPOOL, K_GAVGPOOL, K_SOFTMAX, K_LAYERNORM, \
+    K_ADD_LAYERNORM, K_ELEMENTWISE, K_CHANNEL_PAD, K_ATTENTION = range(10)
+
+DT_F16 = 0
+DT_BF16 = 1
+
+
+@dataclass
+class ExecOp:
+    kind: int
+    name: str
+    inputs: List[str]            # tensor names (in, [in2/residual])
+    output: str
+    params: Dict[str, Any] = field(default_factory=dict)
+    # weight payloads (numpy), packed into the blob by finalize()
+    w: Optional[np.ndarray] = None        # fp16 bt-packed weights
+    scale: Optional[np.ndarray] = None    # fp32 per-channel
+    bias: Optional[np.ndarray] = None     # fp32 per-channel
+
+
+@dataclass
+class EnginePlan:
+    name: str
+    ops: List[Dict[str, Any]]            # dicts for _C.Engine
+    exec_ops: List[ExecOp]               # named ops (CPU reference, tests)
+    weights: np.ndarray                  # uint8 blob
+    arena_bytes: int
+    offsets: Dict[str, int]
+    input_name: str
+    input_off: int
+    input_bytes: int
+    input_shape: Tuple[int, ...]
+    output_name: str
+    output_off: int
+    output_bytes: int
+    output_shape: Tuple[int, ...]
+    dtype: int = DT_F16
+
+
+class Planner:
+    def __init__(self, dtype: int = DT_F16):
+        self.dtype = dtype
+
+    # ------------------------------------------------------------- fusion
+    def fuse(self, g: Graph) -> List[ExecOp]:
+        nodes = g.nodes
+        consumed: set = set()
+        by_output: Dict[str, Node] = {n.output: n for n in nodes}
+        exec_ops: List[ExecOp] = []
+
+        def single_user(t: str) -> Optional[Node]:
+            us = g.users(t)
+            return us[0] if len(us) == 1 else None
+
+        for n in nodes:
+            if n.name in consumed or n.kind == "input":
+                continue
+            if n.kind == "conv2d":
+                exec_ops.append(self._fuse_conv(g, n, consumed, single_user))
+            elif n.kind == "gemm":
+                exec_ops.append(self._fuse_gemm(g, n, consumed, single_user))
+            elif n.kind == "batchnorm":
+                raise ValueError(f"unfused batchnorm {n.name} (expected after conv2d)")
+            elif n.kind == "relu":
+                exec_ops.append(ExecOp(K_ELEMENTWISE, n.name, [n.inputs[0]],
+                                       n.output, dict(op=0)))
+            elif n.kind == "gelu":
+                exec_ops.append(ExecOp(K_ELEMENTWISE, n.name, [n.inputs[0]],
+                                       n.output, dict(op=1)))
+            elif n.kind == "add":
+                exec_ops.append(ExecOp(K_ELEMENTWISE, n.name, list(n.inputs),
+                                       n.output, dict(op=2)))
+            elif n.kind == "maxpool":
+                exec_ops.append(ExecOp(K_MAXPOOL, n.name, [n.inputs[0]],
+                                       n.output, dict(n.attrs)))
+            elif n.kind == "gavgpool":
+                exec_ops.append(ExecOp(K_GAVGPOOL, n.name, [n.inputs[0]],
+                                       n.output, dict(n.attrs)))
+            elif n.kind == "softmax":
+                exec_ops.append(ExecOp(K_SOFTMAX, n.name, [n.inputs[0]],
+                                       n.output, {}))
+            elif n.kind == "layernorm":
+                op = ExecOp(K_LAYERNORM, n.name, [n.inputs[0]], n.output,
+                            dict(eps=n.attrs["eps"]))
+                op.scale = n.attrs["gamma"].astype(np.float32)
+                op.bias = n.attrs["beta"].astype(np.float32)
+                exec_ops.append(op)
+            elif n.kind == "add_layernorm":
+                op = ExecOp(K_ADD_LAYERNORM, n.name, list(n.inputs), n.output,
+                            dict(eps=n.attrs["eps"]))
+                op.scale = n.attrs["gamma"].astype(np.float32)
+                op.bias = n.attrs["beta"].astype(np.float32)
+                exec_ops.append(op)
+            elif n.kind == "attention":
+                exec_ops.append(ExecOp(K_ATTENTION, n.name, [n.inputs[0]],
+                                       n.output, dict(n.attrs)))
+            elif n.kind == "flatten":
+                raise ValueError("flatten should be a view, not a node")
+            else:
+                raise ValueError(f"unknown node kind {n.kind}")
+        return exec_ops
+
+    def _fuse_conv(self, g: Graph, conv: Node, consumed: set, single_user):
+        def_index = {n.output: i for i, n in enumerate(g.nodes)}
+        conv_idx = def_index[conv.output]
+        out = conv.output
+        scale = bias = None
+        residual = None
+        epi = EPI_NONE
+        # conv -> batchnorm?
+        bn = single_user(out)
+        if bn is not None and bn.kind == "batchnorm":
+            a = bn.attrs
+            s = (a["gamma"] / np.sqrt(a["var"] + a["eps"])).astype(np.float32)
+            b = (a["beta"] - a["mean"] * s).astype(np.float32)
+            scale, bias = s, b
+            consumed.add(bn.name)
+            out = bn.output
+            epi = EPI_SCALE_BIAS
+        # -> add(residual)?
+        nxt = single_user(out)
+        if nxt is not None and nxt.kind == "add" and scale is not None:
+            other = nxt.inputs[0] if nxt.inputs[1] == out else nxt.inputs[1]
+            rl = single_user(nxt.output)
+            # the residual must already be computed when this conv runs
+            if rl is not None and rl.kind == "relu" and \
+                    def_index.get(other, 1 << 30) < conv_idx:
+                residual = other
+                consumed.add(nxt.name)
+                consumed.add(rl.name)
+                out = rl.output
+                epi = EPI_SCALE_BIAS_ADD_RELU
+        # -> relu?
+        if epi in (EPI_NONE, EPI_SCALE_BIAS):
+            r = single_user(out)
+            if r is not None and r.kind == "relu":
+                consumed.add(r.name)
+                out = r.output
+                epi = EPI_SCALE_BIAS_RELU if scale is not None else EPI_BIAS_RELU
+                if scale is None:
+                    cout = conv.attrs["weight"].shape[0]
+                    bias = np.zeros(cout, np.float32)
+
+        inputs = [conv.inputs[0]] + ([residual] if residual else [])
+        op = ExecOp(K_CONV, conv.name, inputs, out,
+                    dict(stride=conv.attrs["stride"],
+                         padding=conv.attrs["padding"],
+                         weight_shape=conv.attrs["weight"].shape, epi=epi))
+        op.w = conv.attrs["weight"]  # packed in finalize
+        op.scale, op.bias = scale, bias
+        return op
+
+    def _fuse_gemm(self, g: Graph, gm: Node, consumed: set, single_user):
+        out = gm.output
+        bias = gm.attrs.get("bias")
+        epi = EPI_BIAS if bias is not None else EPI_NONE
+        nxt = single_user(out)
+        if nxt is not None and nxt.kind in ("relu", "gelu"):
+            consumed.add(nxt.name)
+            out = nxt.output
+            if bias is None:
+                bias = np.zeros(gm.attrs["weight"].shape[0], np.float32)
+            epi = EPI_BIAS_RELU if nxt.kind == "relu" else EPI_BIAS_GELU
+        op = ExecOp(K_GEMM, gm.name, [gm.inputs[0]], out,
+                    dict(weight_shape=gm.attrs["weight"].shape, epi=epi))
+        op.w = gm.attrs["weight"]
+        op.bias = None if bias is None else bias.astype(np.float32)
+        return op
+
+    # --------------------------------------------------------------- plan
+    def compile(self, g: Graph) -> EnginePlan:
+        exec_ops = self.fuse(g)
+        shapes: Dict[str, Tuple[int, ...]] = {
+            t: spec.shape for t, spec in g.tensors.items()
+        }
+
+        # ---- legalization: pad input channels for conv staging ----
+        input_name = g.input_name
+        assert input_name is not None
+        in_shape = shapes[input_name]
+        padded_input = input_name
+        if len(in_shape) == 4 and in_shape[3] % 8 != 0:
+            cpad = round_up(in_shape[3], 8)
+            padded_input = input_name + "_padded"
+            shapes[padded_input] = (*in_shape[:3], cpad)
+            m = in_shape[0] * in_shape[1] * in_shape[2]
+            pad_op = ExecOp(K_CHANNEL_PAD, padded_input, [input_name],
+                            padded_input,
+                            dict(M=m, Cin=in_shape[3], Cpad=cpad))
+            exec_ops.insert(0, pad_op)
+            for op in exec_ops[1:]:
+                op.inputs = [padded_input if t == input_name else t
+                             for t in op.inputs]
+
+        # ---- weight prepacking ----
+        for op in exec_ops:
+            if op.kind == K_CONV:
+                w = op.w  # [Cout, Cin, KH, KW] fp32
+                cout, cin, kh, kw = w.shape
+                cpad = round_up(cin, 8)
+                whwc = np.transpose(w, (0, 2, 3, 1))  # [Cout, KH, KW, Cin]
+                if cpad != cin:
+                    whwc = np.pad(whwc, ((0, 0), (0, 0), (0, 0), (0, cpad - cin)))
+                k = kh * kw * cpad
+                kp = round_up(k, 64)
+                flat = whwc.reshape(cout, k)
+                if kp != k:
+                    flat = np.pad(flat, ((0, 0), (0, kp - k)))
+                op.w = np.ascontiguousarray(flat, np.float16)
+                op.params["C"] = cpad
+                op.params["Kp"] = kp
+            elif op.kind == K_GEMM:
+                w = op.w  # [Nout, K] fp32
+                nout, k = w.shape
+                if k % 64 != 0:
+                    raise ValueError(f"gemm {op.name}: K={k} must be %64")
+                op.w = np.ascontiguousarray(w, np.float16)
+
+        # ---- weight blob ----
+        blob = bytearray()
+
+        def pack(arr: Optional[np.ndarray]) -> int:
+            if arr is None:
+                return -1
+            off = round_up(len(blob), 256)
+            blob.extend(b"\0" * (off - len(blob)))
+            blob.extend(np.ascontiguousarray(arr).tobytes())
+            return off
+
+        w_offs: Dict[str, Tuple[int, int, int]] = {}
+        for op in exec_ops:
+            w_offs[op.name] = (pack(op.w), pack(op.scale), pack(op.bias))
+
+        # ---- liveness + arena offsets ----
+        tensors_used: Dict[str, Tuple[int, int]] = {}
+
+        def touch(t: str, i: int):
+            if t in tensors_used:
+                s, e = tensors_used[t]
+                tensors_used[t] = (min(s, i), max(e, i))
+            else:
+                tensors_used[t] = (i, i)
+
+        touch(input_name, 0)
+        for i, op in enumerate(exec_ops):
+            for t in op.inputs:
+                touch(t, i)
+            touch(op.output, i)
+        output_name = exec_ops[-1].output
+        # input live from the start; output live to the end
+        s, e = tensors_used[output_name]
+        tensors_used[output_name] = (s, len(exec_ops))
+
+        def nbytes_of(t: str) -> int:
+            n = 1
+            for d in shapes[t]:
+                n *= d
+            return n * 2  # fp16/bf16
+
+        arena = ArenaPlanner()
+        for t, (s0, e0) in tensors_used.items():
+            arena.add(t, nbytes_of(t), s0, e0)
+        offsets, arena_bytes = arena.plan()
+
+        # ---- emit op dicts ----
+        op_dicts: List[Dict[str, Any]] = []
+        for op in exec_ops:
+            w_off, s_off, b_off = w_offs[op.name]
+            d: Dict[str, Any] = dict(dtype=self.dtype, w_off=w_off,
+                                     scale_off=s_off, bias_off=b_off,
+                                     in_off=offsets[op.inputs[0]],
+                                     out_off=offsets[op.output])
+            if len(op.inputs) > 1:
+                d["in2_off"] = offsets[op.inputs[1]]
+            if op.kind == K_CONV:
+                ish = shapes[op.inputs[0]]
+                osh = shapes[op.output]
+                d.update(kind=K_CONV, epi=op.params["epi"], Nb=ish[0],
+                         H=ish[1], W=ish[2], C=op.params["C"],
+                         Cout=osh[3], KH=op.params["weight_shape"][2],
+                         KW=op.params["weight_shape"][3],
+                         sh=op.params["stride"], sw=op.params["stride"],
+                         ph=op.params["padding"], pw=op.params["padding"])
+            elif op.kind == K_GEMM:
+                m, k = shapes[op.inputs[0]]
+                nout = shapes[op.output][1]
+                d.update(kind=K_GEMM, epi=op.params["epi"], M=m, N=nout, K=k)
+            elif op.kind == K_MAXPOOL:
+                ish = shapes[op.inputs[0]]
+                d.update(kind=K_MAXPOOL, Nb=ish[0], H=ish[1], W=ish[2],
+                         C=ish[3], KH=op.params["kernel"],
+                         KW=op.params["kernel"], sh=op.params["stride"],
+                         sw=op.params["stride"], ph=op.params["padding"],
+                         pw=op.params["padding"])
+            elif op.kind == K_GAVGPOOL:
+                ish = shapes[op.inputs[0]]
+                d.update(kind=K_GAVGPOOL, Nb=ish[0], HW=ish[1] * ish[2],
+                         C=ish[3])
+            elif op.kind == K_SOFTMAX:
+                m, ncol = shapes[op.inputs[0]]
+                d.update(kind=K_SOFTMAX, M=m, N=ncol)
+            elif op.kind == K_LAYERNORM:
+                m, ncol = shapes[op.inputs[0]]
+                d.update(kind=K_LAYERNORM, M=m, N=ncol, eps=op.params["eps"])
+            elif op.kind == K_ADD_LAYERNORM:
+                m, ncol = shapes[op.inputs[0]]
+                d.update(kind=K_ADD_LAYERNORM, M=m, N=ncol,
+                         eps=op.params["eps"], out2_off=-1)
+            elif op.kind == K_ELEMENTWISE:
+                n = 1
+                for s_ in shapes[op.output]:
+                    n *= s_
+                d.update(kind=K_ELEMENTWISE, epi=op.params["op"], n_elems=n)
+            elif op.kind == K_CHANNEL_PAD:
+                d.update(kind=K_CHANNEL_PAD, n_elems=op.params["M"],
+                         C=op.params["Cin"], Cout=op.params["Cpad"])
+            elif op.kind == K_ATTENTION:
+                m, hid = shapes[op.output]
+                heads = op.params["heads"]
+                seq = op.params["seq"]
+                hd = op.params["head_dim"]
+                d.update(kind=K_ATTENTION, B=m // seq, S=seq, NH=heads, HD=hd,
+                         att_scale=1.0 / float(np.sqrt(hd)))
+            else:
+                raise ValueError(f"bad exec op kind {op.kind}")
+            op_dicts.append(d)
+
+        return EnginePlan(
+            name=g.name,
+            ops=op_dicts,
+            exec_ops=exec_ops,
+            weights=np.frombuffer(bytes(blob), dtype=np.uint8).copy(),
+            arena_bytes=arena_bytes,
+            offsets=offsets,
+            input_name=input_name,
+            input_off=offsets[input_name],
+            input_bytes=nbytes_of(input_name),
+            input_shape=shapes[input_name],
+            output_name=output_name,
+            output_off=offsets[output_name],
+            output_bytes=nbytes_of(output_name),
+            output_shape=shapes[output_name],
+            dtype=self.dtype,
+        )

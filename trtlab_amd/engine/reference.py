@@ -1,0 +1,115 @@
+"""CPU reference executor: runs an EnginePlan's exec_ops with torch fp32.
+
+Role: (a) validates the planner on CPU-only CI, (b) provides the numerics
+reference the GPU engine is compared against in tests (the role ONNX test
+vectors play in the reference's examples/ONNX/resnet50/run_onnx_tests.py).
+Uses the SAME prepacked weights as the GPU path so the comparison isolates
+kernel numerics.
+"""
+from __future__ import annotations
+
+from typing import Dict
+
+import numpy as np
+import torch
+import torch.nn.functional as F
+
+from trtlab_amd.engine.planner import (
+    EnginePlan, K_ADD_LAYERNORM, K_ATTENTION, K_CHANNEL_PAD, K_CONV,
+    K_ELEMENTWISE, K_GAVGPOOL, K_GEMM, K_LAYERNORM, K_MAXPOOL, K_SOFTMAX,
+    EPI_BIAS, EPI_BIAS_GELU, EPI_BIAS_RELU, EPI_NONE, EPI_SCALE_BIAS,
+    EPI_SCALE_BIAS_ADD_RELU, EPI_SCALE_BIAS_RELU)
+
+
+def _epi(acc: torch.Tensor, epi: int, scale, bias, res) -> torch.Tensor:
+    v = acc
+    if epi in (EPI_SCALE_BIAS, EPI_SCALE_BIAS_RELU, EPI_SCALE_BIAS_ADD_RELU):
+        v = v * scale
+    if epi != EPI_NONE:
+        v = v + bias
+    if epi == EPI_SCALE_BIAS_ADD_RELU:
+        v = v + res
+    if epi in (EPI_BIAS_RELU, EPI_SCALE_BIAS_RELU, EPI_SCALE_BIAS_ADD_RELU):
+        v = F.relu(v)
+    if epi == EPI_BIAS_GELU:
+        v = F.gelu(v, approximate="tanh")
+    return v
+
+
+def run_reference(plan: EnginePlan, input_nhwc: np.ndarray) -> np.ndarray:
+    """Run the plan on CPU (fp32). input is the RAW (unpadded) input."""
+    t: Dict[str, torch.Tensor] = {
+        plan.input_name: torch.from_numpy(np.ascontiguousarray(input_nhwc)).float()
+    }
+    # shapes registry from the planner's op dicts + exec op metadata
+    for op, d in zip(plan.exec_ops, plan.ops):
+        x = t[op.inputs[0]]
+        if op.kind == K_CHANNEL_PAD:
+            cin, cpad = d["C"], d["Cout"]
+            flat = x.reshape(-1, cin)
+            t[op.output] = F.pad(flat, (0, cpad - cin)).reshape(
+                *x.shape[:-1], cpad)
+        elif op.kind == K_CONV:
+            nb, h, w, c = x.shape
+            kh, kw = d["KH"], d["KW"]
+            cout = d["Cout"]
+            wt = torch.from_numpy(op.w.astype(np.float32))  # [Cout, Kp]
+            k = kh * kw * c
+            wt = wt[:, :k].reshape(cout, kh, kw, c).permute(0, 3, 1, 2)
+            xc = x.permute(0, 3, 1, 2)
+            acc = F.conv2d(xc, wt, stride=d["sh"], padding=d["ph"])
+            acc = acc.permute(0, 2, 3, 1)  # NHWC
+            scale = bias = res = None
+            if op.scale is not None:
+                scale = torch.from_numpy(op.scale)
+            if op.bias is not None:
+                bias = torch.from_numpy(op.bias)
+            if len(op.inputs) > 1:
+                res = t[op.inputs[1]]
+            t[op.output] = _epi(acc, d["epi"], scale, bias, res)
+        elif op.kind == K_GEMM:
+            wt = torch.from_numpy(op.w.astype(np.float32))  # [N, K]
+            acc = x @ wt.t()
+            bias = torch.from_numpy(op.bias) if op.bias is not None else None
+            t[op.output] = _epi(acc, d["epi"], None, bias, None)
+        elif op.kind == K_MAXPOOL:
+            xc = x.permute(0, 3, 1, 2)
+            y = F.max_pool2d(xc, d["KH"], stride=d["sh"], padding=d["ph"])
+            t[op.output] = y.permute(0, 2, 3, 1)
+        elif op.kind == K_GAVGPOOL:
+            nb, h, w, c = x.shape
+            t[op.output] = x.reshape(nb, h * w, c).mean(dim=1)
+        elif op.kind == K_SOFTMAX:
+            t[op.output] = F.softmax(x, dim=-1)
+        elif op.kind == K_LAYERNORM:
+            g_ = torch.from_numpy(op.scale)
+            b_ = torch.from_numpy(op.bias)
+            t[op.output] = F.layer_norm(x, (x.shape[-1],), g_, b_, d["eps"])
+        elif op.kind == K_ADD_LAYERNORM:
+            s = x + t[op.inputs[1]]
+            g_ = torch.from_numpy(op.scale)
+            b_ = torch.from_numpy(op.bias)
+            t[op.output] = F.layer_norm(s, (s.shape[-1],), g_, b_, d["eps"])
+        elif op.kind == K_ELEMENTWISE:
+            code = d["epi"]
+            if code == 0:
+                t[op.output] = F.relu(x)
+            elif code == 1:
+                t[op.output] = F.gelu(x, approximate="tanh")
+            elif code == 2:
+                t[op.output] = x + t[op.inputs[1]]
+            elif code == 3:
+                t[op.output] = F.relu(x + t[op.inputs[1]])
+        elif op.kind == K_ATTENTION:
+            b, s, nh, hd = d["B"], d["S"], d["NH"], d["HD"]
+            hid = nh * hd
+            qkv = x.reshape(b, s, 3, nh, hd)
+            q = qkv[:, :, 0].permute(0, 2, 1, 3)  # [B, NH, S, HD]
+            k = qkv[:, :, 1].permute(0, 2, 1, 3)
+            v = qkv[:, :, 2].permute(0, 2, 1, 3)
+            att = torch.softmax(q @ k.transpose(-1, -2) * d["att_scale"], dim=-1)
+            y = (att @ v).permute(0, 2, 1, 3).reshape(b * s, hid)
+            t[op.output] = y
+        else:
+            raise ValueError(f"bad op kind {op.kind}")
+    return t[plan.output_name].numpy()

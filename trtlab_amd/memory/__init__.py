@@ -1,0 +1,210 @@
+"""trtlab_amd.memory — the allocator layer.
+
+MI355X redesign of the reference's foonathan-style allocator framework
+(trtlab/memory: memory_type concept, block allocators, arenas, pools,
+transactional stack, descriptors, tracking) plus trtlab/cuda's device/pinned
+memory types. The GPU-facing primitives are native (csrc/runtime/memory.cpp
+on hipMalloc/hipHostMalloc, 256-B aligned for HBM3E); this module adds the
+host-side structure: RAII descriptors, blocking pools, transactional stacks
+and the liveness-based arena planner the engine uses for activation reuse
+(the contract the reference gets from TensorRT's
+createExecutionContextWithoutDeviceMemory + setDeviceMemory,
+trtlab/tensorrt/src/workspace.cc:40-41).
+"""
+from __future__ import annotations
+
+import threading
+from dataclasses import dataclass, field
+from typing import List, Optional, Tuple
+
+from trtlab_amd.utils import round_up
+
+MIN_DEVICE_ALIGN = 256  # HBM3E-friendly minimum allocation alignment
+
+
+# --------------------------------------------------------------------------
+# Memory types (reference memory_type.h:93 — where do the bytes live)
+class MemoryType:
+    HOST = "host"
+    PINNED = "host_pinned"
+    DEVICE = "device"
+
+
+# --------------------------------------------------------------------------
+# Descriptors: RAII allocation handles (reference descriptor.h:40,102)
+class DeviceBuffer:
+    """Owns a hipMalloc'd region; frees on close()/del."""
+
+    def __init__(self, nbytes: int, device: int = 0):
+        from trtlab_amd import native
+
+        self._C = native()
+        self.nbytes = int(nbytes)
+        self.device = device
+        self.ptr = self._C.memory.device_malloc(self.nbytes, device)
+
+    def close(self):
+        if getattr(self, "ptr", 0):
+            self._C.memory.device_free(self.ptr, self.nbytes)
+            self.ptr = 0
+
+    def __del__(self):
+        try:
+            self.close()
+        except Exception:
+            pass
+
+    def upload(self, buf) -> None:
+        """Copy a host buffer (numpy array / bytes) to the device."""
+        import numpy as np
+
+        arr = np.ascontiguousarray(buf) if not isinstance(buf, (bytes, bytearray)) else np.frombuffer(buf, dtype=np.uint8)
+        self._C.memory.memcpy_h2d(self.ptr, arr, min(arr.nbytes, self.nbytes))
+
+    def download(self, arr) -> None:
+        self._C.memory.memcpy_d2h(arr, self.ptr, min(arr.nbytes, self.nbytes))
+
+
+class PinnedBuffer:
+    """Page-locked host staging buffer (reference host_pinned_memory)."""
+
+    def __init__(self, nbytes: int):
+        from trtlab_amd import native
+
+        self._C = native()
+        self.nbytes = int(nbytes)
+        self.ptr = self._C.memory.pinned_malloc(self.nbytes)
+
+    def close(self):
+        if getattr(self, "ptr", 0):
+            self._C.memory.pinned_free(self.ptr, self.nbytes)
+            self.ptr = 0
+
+    def __del__(self):
+        try:
+            self.close()
+        except Exception:
+            pass
+
+
+def device_bytes_in_use() -> int:
+    from trtlab_amd import native
+
+    return native().memory.device_bytes_in_use()
+
+
+# --------------------------------------------------------------------------
+# Blocking block pool (reference pool.h v4 + cuda allocator pools):
+# fixed-size device blocks, Pop blocks until a block is free, releases
+# return the block — the concurrency-limiting primitive for Buffers.
+class BlockingBlockPool:
+    def __init__(self, block_bytes: int, count: int, device: int = 0):
+        from trtlab_amd import native
+
+        self._native = native().memory.BlockPool(block_bytes, count, device)
+        self._cv = threading.Condition()
+
+    def pop(self, timeout: Optional[float] = None) -> int:
+        with self._cv:
+            deadline = None
+            while True:
+                p = self._native.acquire()
+                if p:
+                    return p
+                if not self._cv.wait(timeout):
+                    raise TimeoutError("BlockingBlockPool.pop timed out")
+
+    def push(self, ptr: int) -> None:
+        with self._cv:
+            self._native.release(ptr)
+            self._cv.notify()
+
+    @property
+    def available(self) -> int:
+        return self._native.available()
+
+    @property
+    def block_bytes(self) -> int:
+        return self._native.block_bytes
+
+
+# --------------------------------------------------------------------------
+# Transactional stack (host-side mirror of transactional_allocator.h:156):
+# LIFO request-scoped scratch offsets. Used for planning and host scratch.
+class TransactionalStack:
+    def __init__(self, capacity: int, alignment: int = MIN_DEVICE_ALIGN):
+        self.capacity = capacity
+        self.alignment = alignment
+        self._top = 0
+        self._marks: List[int] = []
+
+    def allocate(self, nbytes: int) -> int:
+        off = round_up(self._top, self.alignment)
+        if off + nbytes > self.capacity:
+            raise MemoryError(
+                f"TransactionalStack overflow: {off + nbytes} > {self.capacity}"
+            )
+        self._top = off + nbytes
+        return off
+
+    def begin(self) -> None:
+        self._marks.append(self._top)
+
+    def commit(self) -> None:
+        self._top = self._marks.pop()
+
+    @property
+    def high_water(self) -> int:
+        return self._top
+
+
+# --------------------------------------------------------------------------
+# Liveness-based arena planner: assign byte offsets to tensors with
+# [first_use, last_use] intervals so non-overlapping lifetimes share memory.
+# This is the activation-arena sizing contract of the reference
+# (inference_manager.cc:108-117 GetActivationsMemorySize) made explicit.
+@dataclass
+class _Interval:
+    name: str
+    size: int
+    start: int
+    end: int
+    offset: int = -1
+
+
+class ArenaPlanner:
+    def __init__(self, alignment: int = MIN_DEVICE_ALIGN):
+        self.alignment = alignment
+        self._items: List[_Interval] = []
+
+    def add(self, name: str, nbytes: int, first_use: int, last_use: int) -> None:
+        self._items.append(
+            _Interval(name, round_up(max(nbytes, 1), self.alignment), first_use, last_use)
+        )
+
+    def plan(self) -> Tuple[dict, int]:
+        """Greedy best-fit by decreasing size. Returns ({name: offset}, total)."""
+        items = sorted(self._items, key=lambda t: -t.size)
+        placed: List[_Interval] = []
+        total = 0
+        for it in items:
+            overlapping = [
+                p for p in placed if not (p.end < it.start or p.start > it.end)
+            ]
+            overlapping.sort(key=lambda p: p.offset)
+            best_off = None
+            best_waste = None
+            prev_end = 0
+            for p in overlapping:
+                gap = p.offset - prev_end
+                if gap >= it.size:
+                    waste = gap - it.size
+                    if best_waste is None or waste < best_waste:
+                        best_off, best_waste = prev_end, waste
+                prev_end = max(prev_end, p.offset + p.size)
+            if best_off is None:
+                best_off = prev_end
+            it.offset = best_off
+            placed.append(it)
+            total = max(total, it.offset + it.size)
+        return {it.name: it.offset for it in items}, round_up(total, self.alignment)
