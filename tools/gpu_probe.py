@@ -1,6 +1,11 @@
 """One-shot GPU diagnostics: device info, MFMA fragment-layout probe,
 kernel spot checks. Run on the GPU box; prints everything needed to debug a
 wrong fragment-layout guess without a second round trip."""
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
 import numpy as np
 import torch
 

@@ -72,11 +72,13 @@ class NativeContext:
         return self._out_view
 
     def infer(self, batch: Optional[np.ndarray] = None) -> np.ndarray:
+        """Synchronous convenience path; returns a COPY of the output (the
+        zero-copy `.output` view is only valid while this context lives)."""
         if batch is not None:
             np.copyto(self._in_view, batch.astype(np.float16, copy=False))
         self.ctx.launch()
         self.ctx.synchronize()
-        return self._out_view
+        return np.array(self._out_view, copy=True)
 
     def launch(self):
         self.ctx.launch()
