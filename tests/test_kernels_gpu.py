@@ -152,7 +152,7 @@ def test_conv2d_fused_bn_add_relu(C):
                  residual=res.data_ptr(), zero_page=zero.data_ptr(),
                  Nb=nb, H=h, W=w_, C=cin, Cout=cout, KH=3, KW=3, sh=1, sw=1,
                  ph=1, pw=1, epi=EPI_SB_ADD_RELU)
-    ref = _conv_ref(x, wt, 1, 1, scale.cpu(), bias.cpu(), res, EPI_SB_ADD_RELU)
+    ref = _conv_ref(x, wt, 1, 1, scale, bias, res, EPI_SB_ADD_RELU)
     check(out, ref)
 
 
