@@ -5,10 +5,12 @@
 #include <hip/hip_runtime.h>
 #include <hip/hip_fp16.h>
 #include <hip/hip_bf16.h>
+#include <algorithm>
 #include <cstdint>
 #include <cstdio>
 #include <stdexcept>
 #include <string>
+#include <type_traits>
 
 #define TRT_HIP_CHECK(expr)                                                    \
   do {                                                                         \

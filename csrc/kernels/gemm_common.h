@@ -69,14 +69,15 @@ __device__ __forceinline__ void glds16(const void* gsrc, uint32_t lds_byte) {
       0);
 }
 
-// Stage a 128-row x 64-elem (16 KiB) tile of a row-major matrix into LDS,
-// rows clamped to [0, nrows). Source-side XOR swizzle (rule 21).
-template <typename T>
-__device__ __forceinline__ void stage_tile_128x64(
-    const T* __restrict__ src, int64_t stride_elems, int row0, int nrows,
-    uint32_t lds_base, int tid) {
+// Stage a ROWS x 64-elem tile of a row-major matrix into LDS, rows clamped
+// to [0, nrows). Source-side XOR swizzle (rule 21). ROWS in {64, 128}.
+template <typename T, int ROWS>
+__device__ __forceinline__ void stage_tile(const T* __restrict__ src,
+                                           int64_t stride_elems, int row0,
+                                           int nrows, uint32_t lds_base,
+                                           int tid) {
 #pragma unroll
-  for (int c = 0; c < 4; ++c) {
+  for (int c = 0; c < ROWS / 32; ++c) {
     uint32_t p = c * 4096 + tid * 16;
     uint32_t row = p >> 7;
     uint32_t kb = (p & 127) ^ ((row & 7) << 4);
@@ -95,6 +96,98 @@ __device__ __forceinline__ typename Mfma16x16x32<T>::frag read_frag(
     const char* lds, uint32_t row, uint32_t kbyte) {
   uint32_t off = (row << 7) + (kbyte ^ ((row & 7) << 4));
   return *(const typename Mfma16x16x32<T>::frag*)(lds + off);
+}
+
+// ---------------------------------------------------------- tiled compute
+// Shared MFMA core for gemm/conv: 4 waves as 2x2, wave tile (BM/2)x(BN/2),
+// fragments of 16x16x32, BK=64 (two MFMA k-steps per staged tile).
+template <typename T, int BM, int BN>
+__device__ __forceinline__ void mfma_tile(const char* As, const char* Bs,
+                                          int lane, int wr, int wc,
+                                          f32x4 (&acc)[BM / 32][BN / 32]) {
+  using MF = Mfma16x16x32<T>;
+  constexpr int MFr = BM / 32, NFr = BN / 32;
+#pragma unroll
+  for (int ks = 0; ks < 2; ++ks) {
+    typename MF::frag af[MFr], bf[NFr];
+    uint32_t kbyte = ks * 64 + ((lane >> 4) << 4);
+#pragma unroll
+    for (int f = 0; f < MFr; ++f)
+      af[f] = read_frag<T>(As, wr * (BM / 2) + f * 16 + (lane & 15), kbyte);
+#pragma unroll
+    for (int f = 0; f < NFr; ++f)
+      bf[f] = read_frag<T>(Bs, wc * (BN / 2) + f * 16 + (lane & 15), kbyte);
+#pragma unroll
+    for (int i = 0; i < MFr; ++i)
+#pragma unroll
+      for (int j = 0; j < NFr; ++j)
+        acc[i][j] = MF::run(af[i], bf[j], acc[i][j]);
+  }
+}
+
+// Shared predicated epilogue store. D mapping for 16x16x32 MFMA:
+// col = lane&15, row = (lane>>4)*4 + r.
+template <typename T, Epi E, int BM, int BN>
+__device__ __forceinline__ void store_epilogue(
+    f32x4 (&acc)[BM / 32][BN / 32], T* __restrict__ C, int64_t ldc, int m0,
+    int n0, int M, int N, const float* __restrict__ scale,
+    const float* __restrict__ bias, const T* __restrict__ residual, int lane,
+    int wr, int wc) {
+  constexpr int MFr = BM / 32, NFr = BN / 32;
+#pragma unroll
+  for (int i = 0; i < MFr; ++i) {
+#pragma unroll
+    for (int j = 0; j < NFr; ++j) {
+      int col = n0 + wc * (BN / 2) + j * 16 + (lane & 15);
+      if (col >= N) continue;
+      float sc = 1.0f, bi = 0.0f;
+      if constexpr (E == Epi::kScaleBias || E == Epi::kScaleBiasRelu ||
+                    E == Epi::kScaleBiasAddRelu)
+        sc = scale[col];
+      if constexpr (E != Epi::kNone) bi = bias[col];
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = m0 + wr * (BM / 2) + i * 16 + ((lane >> 4) << 2) + r;
+        if (row >= M) continue;
+        float res = 0.0f;
+        if constexpr (E == Epi::kScaleBiasAddRelu)
+          res = (float)residual[(int64_t)row * ldc + col];
+        float v = apply_epi<E>(acc[i][j][r], sc, bi, res);
+        C[(int64_t)row * ldc + col] = (T)v;
+      }
+    }
+  }
+}
+
+// Host-side tile-config choice: prefer the config that fills the chip
+// (>=512 workgroups) at the highest tile utilization; otherwise maximize
+// parallelism x utilization. Small deep-layer shapes (ResNet stage 4/5 at
+// batch 8) need 64x64 tiles to reach enough workgroups.
+struct TileCfg { int bm; int bn; };
+inline TileCfg pick_tile(int M, int N) {
+  const TileCfg cands[] = {{128, 128}, {128, 64}, {64, 128}, {64, 64}};
+  TileCfg best = cands[0];
+  double best_score = -1.0;
+  for (const auto& c : cands) {
+    long tm = cdiv(M, c.bm), tn = cdiv(N, c.bn);
+    double par = (double)std::min<long>(tm * tn, 512) / 512.0;
+    double util = (double)M * N / ((double)tm * c.bm * tn * c.bn);
+    double score = par * util;
+    if (score > best_score + 1e-9) {
+      best_score = score;
+      best = c;
+    }
+  }
+  return best;
+}
+
+// Dispatch a runtime TileCfg to a compile-time <BM, BN> template call.
+template <typename F>
+inline void tile_dispatch(TileCfg c, F&& f) {
+  if (c.bm == 128 && c.bn == 128) f(std::integral_constant<int, 128>{}, std::integral_constant<int, 128>{});
+  else if (c.bm == 128) f(std::integral_constant<int, 128>{}, std::integral_constant<int, 64>{});
+  else if (c.bn == 128) f(std::integral_constant<int, 64>{}, std::integral_constant<int, 128>{});
+  else f(std::integral_constant<int, 64>{}, std::integral_constant<int, 64>{});
 }
 
 // Bijective XCD-aware blockIdx remap (T1).

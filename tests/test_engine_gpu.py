@@ -105,3 +105,22 @@ def test_infer_runner_pipeline(rn50_small):
     for o in outs[1:]:
         assert np.array_equal(o, outs[0])
     mgr.shutdown()
+
+
+def test_engine_bert():
+    from trtlab_amd.engine.planner import Planner
+    from trtlab_amd.engine.reference import run_reference
+    from trtlab_amd.engine.runtime import NativeEngine
+    from trtlab_amd.models import build_bert
+
+    g = build_bert(batch=2, seq=128, layers=2, seed=0)
+    plan = Planner().compile(g)
+    eng = NativeEngine(plan)
+    ctx = eng.create_context(capture=True)
+    x = np.random.RandomState(9).randn(*plan.input_shape).astype(np.float32)
+    out = ctx.infer(x).astype(np.float32)
+    ref = run_reference(plan, x)
+    err = np.abs(out - ref).max()
+    scale = max(np.abs(ref).max(), 1e-6)
+    assert np.isfinite(out).all()
+    assert err / scale < 0.08, (err, scale)

@@ -229,3 +229,19 @@ def test_channel_pad(C):
     C.ops.channel_pad(0, x.data_ptr(), out.data_ptr(), 100, 3, 8)
     ref = torch.nn.functional.pad(x.float(), (0, 5))
     check(out, ref, rtol=1e-3, atol=1e-3)
+
+
+def test_attention(C):
+    B, S, H, D = 2, 128, 12, 64
+    hid = H * D
+    qkv = t16(B * S, 3 * hid, seed=77)
+    out = torch.empty(B * S, hid, dtype=torch.half, device="cuda")
+    torch.cuda.synchronize()
+    C.ops.attention(0, qkv.data_ptr(), out.data_ptr(), B, S, H, D,
+                    1.0 / (D ** 0.5))
+    q = qkv.float().reshape(B, S, 3, H, D)
+    att = torch.softmax(
+        (q[:, :, 0].permute(0, 2, 1, 3) @ q[:, :, 1].permute(0, 2, 1, 3).transpose(-1, -2))
+        / (D ** 0.5), dim=-1)
+    ref = (att @ q[:, :, 2].permute(0, 2, 1, 3)).permute(0, 2, 1, 3).reshape(B * S, hid)
+    check(out, ref)

@@ -1,0 +1,49 @@
+"""BERT-base encoder IR builder (seq=128, fp16), random weights.
+
+BASELINE config 5: "BERT-base seq=128 fp16 from ONNX (attention/GEMM MFMA
+path, hipGraph-captured forward)". The graph takes pre-embedded hidden
+states [B*S, hidden] as input (embedding gather is planned; the encoder
+stack — QKV/attention/projection/FFN/LayerNorm — is the compute path the
+baseline measures).
+"""
+from __future__ import annotations
+
+import numpy as np
+
+from trtlab_amd.engine.ir import Graph
+
+
+def build_bert(batch: int = 8, seq: int = 128, hidden: int = 768,
+               layers: int = 12, heads: int = 12, seed: int = 0,
+               intermediate: int | None = None) -> Graph:
+    inter = intermediate or hidden * 4
+    rng = np.random.RandomState(seed)
+
+    def w(nout, nin, scale=None):
+        s = scale or np.sqrt(1.0 / nin)
+        return (rng.randn(nout, nin) * s).astype(np.float32)
+
+    def b(n, scale=0.02):
+        return (rng.randn(n) * scale).astype(np.float32)
+
+    def ln(n):
+        return (rng.uniform(0.9, 1.1, n).astype(np.float32),
+                (rng.randn(n) * 0.02).astype(np.float32))
+
+    g = Graph(f"bert_base_s{seq}_b{batch}")
+    m = batch * seq
+    x = g.input((m, hidden), name="hidden_in")
+    h = x
+    for li in range(layers):
+        qkv = g.gemm(h, w(3 * hidden, hidden), b(3 * hidden),
+                     name=f"l{li}_qkv")
+        att = g.attention(qkv, heads=heads, seq=seq, name=f"l{li}_att")
+        proj = g.gemm(att, w(hidden, hidden), b(hidden), name=f"l{li}_proj")
+        ga, ba = ln(hidden)
+        h1 = g.add_layernorm(proj, h, ga, ba, name=f"l{li}_ln1")
+        ff1 = g.gemm(h1, w(inter, hidden), b(inter), name=f"l{li}_ff1")
+        ff1 = g.gelu(ff1, name=f"l{li}_gelu")
+        ff2 = g.gemm(ff1, w(hidden, inter), b(hidden), name=f"l{li}_ff2")
+        gb, bb = ln(hidden)
+        h = g.add_layernorm(ff2, h1, gb, bb, name=f"l{li}_ln2")
+    return g

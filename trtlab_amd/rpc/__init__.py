@@ -1,0 +1,12 @@
+"""trtlab_amd.rpc — asynchronous gRPC microservice framework.
+
+MI355X-native redesign of trtlab/nvrpc: Server /
+AsyncService / unary+streaming+batching lifecycles / executors / clients,
+built on grpc.aio completion-queue machinery. Messages are defined with
+dynamic protobuf descriptors (no protoc needed offline) in rpc.proto.
+"""
+from trtlab_amd.rpc.proto import (EchoRequest, EchoResponse, InferRequest,  # noqa: F401
+                                  InferResponse, HealthRequest, HealthResponse)
+from trtlab_amd.rpc.server import (AsyncService, BatchingService, Server,  # noqa: F401
+                                   StreamingService, UnaryService)
+from trtlab_amd.rpc.client import AsyncClient, SyncClient, siege  # noqa: F401

@@ -1,0 +1,144 @@
+"""RPC clients: sync + async unary/streaming, and the siege constant-rate
+load generator (reference client/client_unary.h, client_streaming*.h,
+02_TensorRT_GRPC/src/siege.cc:226-258)."""
+from __future__ import annotations
+
+import asyncio
+import threading
+import time
+from concurrent.futures import Future
+from typing import Any, AsyncIterator, Callable, List, Optional
+
+import grpc
+import grpc.aio
+
+
+class SyncClient:
+    """Blocking unary client over a shared channel."""
+
+    def __init__(self, target: str):
+        self.channel = grpc.insecure_channel(target)
+
+    def call(self, service: str, method: str, request, resp_cls,
+             timeout: Optional[float] = None):
+        fn = self.channel.unary_unary(
+            f"/{service}/{method}",
+            request_serializer=lambda m: m.SerializeToString(),
+            response_deserializer=resp_cls.FromString)
+        return fn(request, timeout=timeout)
+
+    def close(self):
+        self.channel.close()
+
+
+class AsyncClient:
+    """Async unary/streaming client with its own event-loop thread, so
+    synchronous code gets future-based Infer (reference client::Executor
+    (client/executor.h:39) + ClientUnary's async_compute futures)."""
+
+    def __init__(self, target: str):
+        self.target = target
+        self._loop = asyncio.new_event_loop()
+        self._thread = threading.Thread(target=self._run, daemon=True,
+                                        name="rpc-client")
+        self._ready = threading.Event()
+        self._thread.start()
+        self._ready.wait(timeout=10)
+
+    def _run(self):
+        asyncio.set_event_loop(self._loop)
+        self.channel = grpc.aio.insecure_channel(self.target)
+        self._ready.set()
+        self._loop.run_forever()
+
+    def call(self, service: str, method: str, request, resp_cls,
+             timeout: Optional[float] = None) -> Future:
+        async def do():
+            fn = self.channel.unary_unary(
+                f"/{service}/{method}",
+                request_serializer=lambda m: m.SerializeToString(),
+                response_deserializer=resp_cls.FromString)
+            return await fn(request, timeout=timeout)
+
+        return asyncio.run_coroutine_threadsafe(do(), self._loop)
+
+    def stream(self, service: str, method: str, requests: List[Any],
+               resp_cls) -> Future:
+        """Bidirectional stream: send all requests, collect all responses."""
+        async def do():
+            fn = self.channel.stream_stream(
+                f"/{service}/{method}",
+                request_serializer=lambda m: m.SerializeToString(),
+                response_deserializer=resp_cls.FromString)
+
+            async def gen():
+                for r in requests:
+                    yield r
+
+            return [resp async for resp in fn(gen())]
+
+        return asyncio.run_coroutine_threadsafe(do(), self._loop)
+
+    def close(self):
+        async def _close():
+            await self.channel.close()
+
+        try:
+            asyncio.run_coroutine_threadsafe(_close(), self._loop).result(5)
+        except Exception:
+            pass
+        self._loop.call_soon_threadsafe(self._loop.stop)
+        self._thread.join(timeout=5)
+
+
+def siege(target: str, service: str, method: str, make_request: Callable[[int], Any],
+          resp_cls, rate_hz: float = 100.0, duration_s: float = 5.0,
+          max_outstanding: int = 950) -> dict:
+    """Constant-rate load generator with an outstanding-request cap
+    (reference siege.cc: max 950 outstanding, constant-rate issue loop).
+    Returns latency/throughput stats."""
+    import numpy as np
+
+    client = AsyncClient(target)
+    sem = threading.Semaphore(max_outstanding)
+    lat: List[float] = []
+    lock = threading.Lock()
+    errors = [0]
+    issued = 0
+    t0 = time.monotonic()
+    period = 1.0 / rate_hz
+    futs = []
+    while time.monotonic() - t0 < duration_s:
+        target_t = t0 + issued * period
+        now = time.monotonic()
+        if now < target_t:
+            time.sleep(target_t - now)
+        sem.acquire()
+        start = time.monotonic()
+        f = client.call(service, method, make_request(issued), resp_cls,
+                        timeout=30)
+
+        def done(fut, start=start):
+            with lock:
+                if fut.exception():
+                    errors[0] += 1
+                else:
+                    lat.append(time.monotonic() - start)
+            sem.release()
+
+        f.add_done_callback(done)
+        futs.append(f)
+        issued += 1
+    for f in futs:
+        try:
+            f.result(timeout=60)
+        except Exception:
+            pass
+    elapsed = time.monotonic() - t0
+    client.close()
+    lat_ms = np.array(sorted(lat)) * 1e3 if lat else np.array([0.0])
+    return dict(issued=issued, completed=len(lat), errors=errors[0],
+                seconds=elapsed, rate=len(lat) / elapsed,
+                p50_ms=float(np.percentile(lat_ms, 50)),
+                p90_ms=float(np.percentile(lat_ms, 90)),
+                p99_ms=float(np.percentile(lat_ms, 99)))
