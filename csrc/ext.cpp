@@ -290,6 +290,7 @@ PYBIND11_MODULE(_C, m) {
       .def("ready", &ExecutionContext::ready)
       .def_property_readonly("host_input_ptr", &ExecutionContext::host_input_ptr)
       .def_property_readonly("host_output_ptr", &ExecutionContext::host_output_ptr)
+      .def_property_readonly("arena_ptr", &ExecutionContext::arena_ptr)
       .def("input_view",
            [](ExecutionContext& c, size_t bytes) {
              return py::memoryview::from_memory((void*)c.host_input_ptr(),

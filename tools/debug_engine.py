@@ -1,0 +1,85 @@
+"""Localize engine-vs-reference mismatches: run the native engine eagerly,
+then read every intermediate tensor back from the activation arena and
+compare against the CPU fp32 reference. Tensors whose arena slot is later
+reused are skipped (their bytes are overwritten by design).
+
+Usage (on the GPU box):  python tools/debug_engine.py [bert|resnet]
+"""
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+import numpy as np
+
+import trtlab_amd
+from trtlab_amd.engine.planner import Planner
+from trtlab_amd.engine.reference import run_reference
+from trtlab_amd.engine.runtime import NativeEngine
+
+
+def build_plan(which: str):
+    if which == "bert":
+        from trtlab_amd.models import build_bert
+
+        return Planner().compile(build_bert(batch=2, seq=128, layers=2, seed=0))
+    from trtlab_amd.models import build_resnet
+
+    return Planner().compile(build_resnet(50, batch=2, image=64, seed=0))
+
+
+def main():
+    which = sys.argv[1] if len(sys.argv) > 1 else "bert"
+    plan = build_plan(which)
+    C = trtlab_amd.native()
+
+    x = np.random.RandomState(9).randn(*plan.input_shape).astype(np.float32)
+    cpu = run_reference(plan, x, return_all=True)
+
+    eng = NativeEngine(plan)
+    ctx = eng.create_context(capture=False)
+    ctx.infer(x)
+
+    # def order of tensors = exec op order
+    def_order = {plan.input_name: -1}
+    for i, op in enumerate(plan.exec_ops):
+        def_order[op.output] = i
+
+    def overlaps(a, b):
+        o1, s1 = a
+        o2, s2 = b
+        return not (o1 + s1 <= o2 or o2 + s2 <= o1)
+
+    def nbytes(t):
+        n = 1
+        for d in plan.shapes[t]:
+            n *= d
+        return n * 2
+
+    spans = {t: (plan.offsets[t], nbytes(t)) for t in def_order}
+    arena = ctx.ctx.arena_ptr
+
+    print(f"{'op':4} {'tensor':24} {'shape':20} {'rel_err':>10}  note")
+    for i, op in enumerate(plan.exec_ops):
+        t = op.output
+        # skip if a later-defined tensor overwrites this slot
+        clobbered = any(
+            def_order[u] > i and overlaps(spans[t], spans[u])
+            for u in def_order if u != t)
+        if clobbered:
+            continue
+        shape = plan.shapes[t]
+        buf = np.empty(int(np.prod(shape)), dtype=np.float16)
+        C.memory.memcpy_d2h(buf, arena + plan.offsets[t], buf.nbytes)
+        got = buf.reshape(shape).astype(np.float32)
+        want = cpu[t]
+        scale = max(np.abs(want).max(), 1e-6)
+        rel = np.abs(got - want).max() / scale
+        note = "<-- FIRST BAD" if rel > 0.05 else ""
+        print(f"{i:4} {t[:24]:24} {str(shape):20} {rel:10.5f}  {note}")
+        if rel > 0.05 and "--stop" in sys.argv:
+            break
+
+
+if __name__ == "__main__":
+    main()

@@ -70,6 +70,7 @@ class EnginePlan:
     output_bytes: int
     output_shape: Tuple[int, ...]
     dtype: int = DT_F16
+    shapes: Dict[str, Tuple[int, ...]] = field(default_factory=dict)
 
 
 class Planner:
@@ -376,4 +377,5 @@ class Planner:
             output_bytes=nbytes_of(output_name),
             output_shape=shapes[output_name],
             dtype=self.dtype,
+            shapes=dict(shapes),
         )

@@ -36,8 +36,10 @@ def _epi(acc: torch.Tensor, epi: int, scale, bias, res) -> torch.Tensor:
     return v
 
 
-def run_reference(plan: EnginePlan, input_nhwc: np.ndarray) -> np.ndarray:
-    """Run the plan on CPU (fp32). input is the RAW (unpadded) input."""
+def run_reference(plan: EnginePlan, input_nhwc: np.ndarray,
+                  return_all: bool = False):
+    """Run the plan on CPU (fp32). input is the RAW (unpadded) input.
+    With return_all=True, returns the dict of ALL tensors (debugging)."""
     t: Dict[str, torch.Tensor] = {
         plan.input_name: torch.from_numpy(np.ascontiguousarray(input_nhwc)).float()
     }
@@ -112,4 +114,6 @@ def run_reference(plan: EnginePlan, input_nhwc: np.ndarray) -> np.ndarray:
             t[op.output] = y
         else:
             raise ValueError(f"bad op kind {op.kind}")
+    if return_all:
+        return {k: v.numpy() for k, v in t.items()}
     return t[plan.output_name].numpy()
