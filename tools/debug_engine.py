@@ -19,13 +19,14 @@ from trtlab_amd.engine.runtime import NativeEngine
 
 
 def build_plan(which: str):
+    planner = Planner(reuse=False)  # disjoint slots: every tensor readable
     if which == "bert":
         from trtlab_amd.models import build_bert
 
-        return Planner().compile(build_bert(batch=2, seq=128, layers=2, seed=0))
+        return planner.compile(build_bert(batch=2, seq=128, layers=2, seed=0))
     from trtlab_amd.models import build_resnet
 
-    return Planner().compile(build_resnet(50, batch=2, image=64, seed=0))
+    return planner.compile(build_resnet(50, batch=2, image=64, seed=0))
 
 
 def main():
@@ -74,11 +75,17 @@ def main():
         got = buf.reshape(shape).astype(np.float32)
         want = cpu[t]
         scale = max(np.abs(want).max(), 1e-6)
-        rel = np.abs(got - want).max() / scale
-        note = "<-- FIRST BAD" if rel > 0.05 else ""
+        nbad_nan = int(np.isnan(got).sum())
+        rel = np.nanmax(np.abs(got - want)) / scale
+        note = ""
+        if nbad_nan:
+            idx = np.argwhere(np.isnan(got))[0]
+            note = f"NaNs={nbad_nan} first@{tuple(idx)}"
+        elif rel > 0.05:
+            idx = np.unravel_index(np.argmax(np.abs(got - want)), got.shape)
+            note = (f"<-- BAD worst@{tuple(idx)} got={got[idx]:.4f} "
+                    f"want={want[idx]:.4f}")
         print(f"{i:4} {t[:24]:24} {str(shape):20} {rel:10.5f}  {note}")
-        if rel > 0.05 and "--stop" in sys.argv:
-            break
 
 
 if __name__ == "__main__":

@@ -74,8 +74,9 @@ class EnginePlan:
 
 
 class Planner:
-    def __init__(self, dtype: int = DT_F16):
+    def __init__(self, dtype: int = DT_F16, reuse: bool = True):
         self.dtype = dtype
+        self.reuse = reuse  # False: disjoint arena slots (debugging)
 
     # ------------------------------------------------------------- fusion
     def fuse(self, g: Graph) -> List[ExecOp]:
@@ -295,7 +296,10 @@ class Planner:
 
         arena = ArenaPlanner()
         for t, (s0, e0) in tensors_used.items():
-            arena.add(t, nbytes_of(t), s0, e0)
+            if self.reuse:
+                arena.add(t, nbytes_of(t), s0, e0)
+            else:
+                arena.add(t, nbytes_of(t), 0, len(exec_ops))
         offsets, arena_bytes = arena.plan()
 
         # ---- emit op dicts ----
