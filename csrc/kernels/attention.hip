@@ -146,8 +146,12 @@ __global__ __launch_bounds__(256) void attention_kernel(
       }
     }
   }
-  // P and Vt are read by the same wave that wrote P; Vt was barriered above.
-  // ds ordering within a wave is handled by the compiler's lgkmcnt waits.
+  // Force completion of the P ds_writes before the PV ds_reads. A bare
+  // same-wave write->read *should* be ordered by the compiler's lgkmcnt
+  // bookkeeping, but was observed to return stale P for some blocks on
+  // real data (one (b,h) block NaN per launch); the barrier makes the
+  // ordering explicit and costs nothing at this kernel's size.
+  __syncthreads();
 
   // ---- PV: out_tile[32 rows][64 d] = P[32][128] @ Vt^T ----
   f32x4 oacc[2][4];

@@ -76,3 +76,19 @@ report("engine att  vs torch(fp16qkv)", att_eng, ref)
 report("standalone  vs torch(fp16qkv)", att_alone, ref)
 report("engine      vs standalone    ", att_eng, att_alone)
 report("cpu_ref att vs torch(fp16qkv)", cpu["l0_att"], ref)
+
+# determinism check: run the standalone kernel 10x, compare runs
+print("\ndeterminism check (10 runs):")
+outs = []
+for it in range(10):
+    o = torch.empty(B * S, hid, dtype=torch.half, device="cuda")
+    torch.cuda.synchronize()
+    C.ops.attention(0, qkv_t.data_ptr(), o.data_ptr(), B, S, H, D,
+                    float(1.0 / np.sqrt(D)))
+    outs.append(o.float().cpu().numpy())
+ref0 = outs[0]
+for it, o in enumerate(outs[1:], 1):
+    same = np.array_equal(o, ref0, equal_nan=True)
+    nans = int(np.isnan(o).sum())
+    err = np.nanmax(np.abs(o - ref))
+    print(f"  run{it}: identical_to_run0={same} nans={nans} max_vs_torch={err:.4f}")
