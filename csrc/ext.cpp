@@ -89,6 +89,10 @@ PYBIND11_MODULE(_C, m) {
 
   // ---------------------------------------------------------------- hip
   auto hip = m.def_submodule("hip");
+  hip.def("last_error", [] {
+    hipError_t e = hipGetLastError();
+    return std::string(hipGetErrorString(e));
+  });
   hip.def("device_count", [] {
     int n = 0;
     hipError_t e = hipGetDeviceCount(&n);
