@@ -48,12 +48,12 @@ __global__ __launch_bounds__(256) void attention_kernel(
   };
 
   // ---- stage Q, K (swizzled 16-B chunks), Vt (transposed scatter) ----
-  // Q/K: 128 rows x 4 chunks of 16 B -> 512 chunks, 2 per thread each.
+  // Q/K: 128 rows x 128 B/row -> 1024 16-B chunks each, 4 per thread.
 #pragma unroll
-  for (int c = 0; c < 2; ++c) {
-    int idx = c * 256 + tid;
-    int row = idx >> 2;            // 0..127
-    int cb = (idx & 3) * 16;       // byte offset within 128-B row
+  for (int c = 0; c < 4; ++c) {
+    int idx = c * 256 + tid;       // 0..1023
+    int row = idx >> 3;            // 0..127
+    int cb = (idx & 7) * 16;       // byte offset within 128-B row
     const T* src = base + (int64_t)row * row_stride + qoff + cb / 2;
     *(short8v*)(Qs + swz(row, cb)) = *(const short8v*)src;
     const T* ksrc = base + (int64_t)row * row_stride + koff + cb / 2;
@@ -196,6 +196,22 @@ __global__ __launch_bounds__(256) void attention_kernel(
       }
     }
   }
+}
+
+// Debug probe: each block writes its view of the launch parameters.
+__global__ void attention_probe_kernel(int* dbg, int B, int S, int H, int D) {
+  if (threadIdx.x == 0) {
+    int bh = blockIdx.x;
+    int* p = dbg + bh * 8;
+    p[0] = B; p[1] = S; p[2] = H; p[3] = D;
+    p[4] = (int)gridDim.x; p[5] = bh; p[6] = bh / H; p[7] = bh % H;
+  }
+}
+
+void launch_attention_probe(int* dbg, int B, int S, int H, int D,
+                            hipStream_t stream) {
+  hipLaunchKernelGGL(attention_probe_kernel, dim3(B * H), dim3(256), 0,
+                     stream, dbg, B, S, H, D);
 }
 
 void launch_attention(int dtype, const void* qkv, void* out, int B, int S,
