@@ -33,7 +33,8 @@ def main():
     ap.add_argument("--warmup", type=int, default=20)
     ap.add_argument("--batch", type=int, default=8)
     ap.add_argument("--contexts", type=int, default=3)
-    ap.add_argument("--depth", type=int, default=50)
+    ap.add_argument("--model", default="resnet50",
+                    choices=["resnet50", "resnet101", "resnet152", "bert"])
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -53,11 +54,17 @@ def main():
     import trtlab_amd
     from trtlab_amd.engine.planner import Planner
     from trtlab_amd.engine.runtime import NativeEngine
-    from trtlab_amd.models import build_resnet
+    from trtlab_amd.models import build_bert, build_resnet
     from trtlab_amd.parallel import broadcast_weights
 
     # Build the plan (identical on every rank: same seed).
-    g = build_resnet(args.depth, batch=args.batch, image=224, seed=0)
+    if args.model == "bert":
+        g = build_bert(batch=args.batch, seq=128, layers=12, seed=0)
+        cfg_extra = {"seq_len": 128, "hidden": 768, "layers": 12}
+    else:
+        depth = int(args.model.replace("resnet", ""))
+        g = build_resnet(depth, batch=args.batch, image=224, seed=0)
+        cfg_extra = {"image": 224}
     plan = Planner().compile(g)
 
     eng = NativeEngine(plan, device=local_rank)
@@ -128,10 +135,10 @@ def main():
             "dtype": "fp16",
             "data": "synthetic",
             "config": {
-                "model": f"resnet{args.depth}",
+                "model": args.model,
                 "global_batch": args.batch * n_gpus,
                 "batch_per_gpu": args.batch,
-                "image": 224,
+                **cfg_extra,
                 "contexts": args.contexts,
                 "parallelism": f"dp{n_gpus}",
                 "p50_ms": p50,
