@@ -134,26 +134,27 @@ PYBIND11_MODULE(_C, m) {
   ops.def("gemm_bt",
           [](int dtype, uintptr_t A, uintptr_t B, uintptr_t C, uintptr_t scale,
              uintptr_t bias, uintptr_t residual, int M, int N, int K, int epi,
-             uintptr_t stream, bool sync) {
+             uintptr_t stream, bool sync, int tile) {
             launch_gemm_bt(dtype, (void*)A, (void*)B, (void*)C, (float*)scale,
                            (float*)bias, (void*)residual, M, N, K, K, K, N,
-                           epi, as_stream(stream));
+                           epi, as_stream(stream), tile);
             if (sync) TRT_HIP_CHECK(hipStreamSynchronize(as_stream(stream)));
           },
           py::arg("dtype"), py::arg("A"), py::arg("B"), py::arg("C"),
           py::arg("scale") = 0, py::arg("bias") = 0, py::arg("residual") = 0,
           py::arg("M") = 0, py::arg("N") = 0, py::arg("K") = 0,
-          py::arg("epi") = 0, py::arg("stream") = 0, py::arg("sync") = true);
+          py::arg("epi") = 0, py::arg("stream") = 0, py::arg("sync") = true,
+          py::arg("tile") = 0);
   ops.def("conv2d",
           [](int dtype, uintptr_t in, uintptr_t Wt, uintptr_t out,
              uintptr_t scale, uintptr_t bias, uintptr_t residual,
              uintptr_t zero_page, int Nb, int H, int W, int C, int Cout,
              int KH, int KW, int sh, int sw, int ph, int pw, int epi,
-             uintptr_t stream, bool sync) {
+             uintptr_t stream, bool sync, int tile) {
             launch_conv2d(dtype, (void*)in, (void*)Wt, (void*)out,
                           (float*)scale, (float*)bias, (void*)residual,
                           (void*)zero_page, Nb, H, W, C, Cout, KH, KW, sh, sw,
-                          ph, pw, epi, as_stream(stream));
+                          ph, pw, epi, as_stream(stream), tile);
             if (sync) TRT_HIP_CHECK(hipStreamSynchronize(as_stream(stream)));
           },
           py::arg("dtype"), py::arg("in"), py::arg("Wt"), py::arg("out"),
@@ -162,7 +163,8 @@ PYBIND11_MODULE(_C, m) {
           py::arg("W") = 0, py::arg("C") = 0, py::arg("Cout") = 0,
           py::arg("KH") = 1, py::arg("KW") = 1, py::arg("sh") = 1,
           py::arg("sw") = 1, py::arg("ph") = 0, py::arg("pw") = 0,
-          py::arg("epi") = 0, py::arg("stream") = 0, py::arg("sync") = true);
+          py::arg("epi") = 0, py::arg("stream") = 0, py::arg("sync") = true,
+          py::arg("tile") = 0);
   ops.def("maxpool2d",
           [](int dtype, uintptr_t in, uintptr_t out, int Nb, int H, int W,
              int C, int KH, int KW, int sh, int sw, int ph, int pw,

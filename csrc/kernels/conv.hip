@@ -128,8 +128,9 @@ template <typename T>
 static void launch_conv2d_t(const void* in, const void* Wt, void* out,
                             const float* scale, const float* bias,
                             const void* residual, const void* zero_page,
-                            const ConvParams& p, int epi, hipStream_t stream) {
-  TileCfg cfg = pick_tile(p.M, p.Cout);
+                            const ConvParams& p, int epi, hipStream_t stream,
+                            int tile) {
+  TileCfg cfg = tile ? tile_from_code(tile) : pick_tile(p.M, p.Cout);
   int tiles_m = (int)cdiv(p.M, cfg.bm);
   int tiles_n = (int)cdiv(p.Cout, cfg.bn);
   dim3 grid(tiles_m * tiles_n);
@@ -151,7 +152,7 @@ void launch_conv2d(int dtype, const void* in, const void* Wt, void* out,
                    const float* scale, const float* bias, const void* residual,
                    const void* zero_page, int Nb, int H, int W, int C,
                    int Cout, int KH, int KW, int sh, int sw, int ph, int pw,
-                   int epi, hipStream_t stream) {
+                   int epi, hipStream_t stream, int tile) {
   ConvParams p;
   p.Nb = Nb; p.H = H; p.W = W; p.C = C;
   p.Cout = Cout; p.KH = KH; p.KW = KW;
@@ -168,10 +169,10 @@ void launch_conv2d(int dtype, const void* in, const void* Wt, void* out,
   p.d_kw = make_fastdiv((uint32_t)KW);
   if (dtype == 0)
     launch_conv2d_t<_Float16>(in, Wt, out, scale, bias, residual, zero_page, p,
-                              epi, stream);
+                              epi, stream, tile);
   else
     launch_conv2d_t<__bf16>(in, Wt, out, scale, bias, residual, zero_page, p,
-                            epi, stream);
+                            epi, stream, tile);
 }
 
 }  // namespace trtlab

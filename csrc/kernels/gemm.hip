@@ -81,8 +81,8 @@ static void launch_gemm_bt_t(const void* A, const void* B, void* C,
                              const float* scale, const float* bias,
                              const void* residual, int M, int N, int K,
                              int64_t lda, int64_t ldb, int64_t ldc, int epi,
-                             hipStream_t stream) {
-  TileCfg cfg = pick_tile(M, N);
+                             hipStream_t stream, int tile) {
+  TileCfg cfg = tile ? tile_from_code(tile) : pick_tile(M, N);
   int tiles_m = (int)cdiv(M, cfg.bm);
   int tiles_n = (int)cdiv(N, cfg.bn);
   dim3 grid(tiles_m * tiles_n);
@@ -103,14 +103,14 @@ void launch_gemm_bt(int dtype,  // 0 = fp16, 1 = bf16
                     const void* A, const void* B, void* C, const float* scale,
                     const float* bias, const void* residual, int M, int N,
                     int K, int64_t lda, int64_t ldb, int64_t ldc, int epi,
-                    hipStream_t stream) {
+                    hipStream_t stream, int tile) {
   if (K % 64 != 0) throw std::runtime_error("gemm_bt: K must be a multiple of 64");
   if (dtype == 0)
     launch_gemm_bt_t<_Float16>(A, B, C, scale, bias, residual, M, N, K, lda,
-                               ldb, ldc, epi, stream);
+                               ldb, ldc, epi, stream, tile);
   else
     launch_gemm_bt_t<__bf16>(A, B, C, scale, bias, residual, M, N, K, lda, ldb,
-                             ldc, epi, stream);
+                             ldc, epi, stream, tile);
 }
 
 }  // namespace trtlab

@@ -181,6 +181,16 @@ inline TileCfg pick_tile(int M, int N) {
   return best;
 }
 
+// Explicit tile override codes: 1=(128,128) 2=(128,64) 3=(64,128) 4=(64,64).
+inline TileCfg tile_from_code(int code) {
+  switch (code) {
+    case 1: return {128, 128};
+    case 2: return {128, 64};
+    case 3: return {64, 128};
+    default: return {64, 64};
+  }
+}
+
 // Dispatch a runtime TileCfg to a compile-time <BM, BN> template call.
 template <typename F>
 inline void tile_dispatch(TileCfg c, F&& f) {

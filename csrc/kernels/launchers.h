@@ -9,13 +9,13 @@ namespace trtlab {
 void launch_gemm_bt(int dtype, const void* A, const void* B, void* C,
                     const float* scale, const float* bias, const void* residual,
                     int M, int N, int K, int64_t lda, int64_t ldb, int64_t ldc,
-                    int epi, hipStream_t stream);
+                    int epi, hipStream_t stream, int tile = 0);
 
 void launch_conv2d(int dtype, const void* in, const void* Wt, void* out,
                    const float* scale, const float* bias, const void* residual,
                    const void* zero_page, int Nb, int H, int W, int C,
                    int Cout, int KH, int KW, int sh, int sw, int ph, int pw,
-                   int epi, hipStream_t stream);
+                   int epi, hipStream_t stream, int tile = 0);
 
 void launch_maxpool2d(int dtype, const void* in, void* out, int Nb, int H,
                       int W, int C, int KH, int KW, int sh, int sw, int ph,

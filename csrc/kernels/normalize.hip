@@ -1,6 +1,10 @@
 // trtlab_amd — row softmax and layernorm for gfx950 (wave64 shuffle
-// reductions, fp32 accumulation, vectorized fp16/bf16 I/O).
+// reductions, fp32 accumulation, vectorized 16-B fp16/bf16 I/O per G13:
+// scalar bf16/fp16 loads cost ~2-2.5x on memory-bound kernels).
 // Covers SURVEY.md §2.8 items 6 (softmax) and 8 (layernorm).
+//
+// Layout: one wave per row, 4 waves per block. Each lane owns 8-element
+// chunks (chunk c = lane + 64*p); rows up to 2048 elements (N % 8 == 0).
 #include "../common.h"
 
 namespace trtlab {
@@ -17,118 +21,137 @@ __device__ __forceinline__ float wave_reduce_sum(float v) {
   return v;
 }
 
-// One wave per row; rows [M][ld], N valid columns; out = softmax(row).
-// N <= 64*VMAX (VMAX=32 -> N<=2048 per row held in registers).
+template <typename T>
+__device__ __forceinline__ void load8(const T* p, float* dst) {
+  short8v raw = *(const short8v*)p;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) dst[j] = (float)((const T*)&raw)[j];
+}
+
+template <typename T>
+__device__ __forceinline__ void store8(T* p, const float* src) {
+  T out[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) out[j] = (T)src[j];
+  *(short8v*)p = *(const short8v*)out;
+}
+
+constexpr int kMaxChunks = 4;  // 4 chunks/lane * 64 lanes * 8 elems = 2048
+
 template <typename T>
 __global__ void softmax_rows_kernel(const T* __restrict__ in,
                                     T* __restrict__ out, int M, int N,
                                     int64_t ld) {
-  constexpr int VMAX = 32;
   int lane = threadIdx.x & 63;
   int row = (blockIdx.x * (blockDim.x >> 6)) + (threadIdx.x >> 6);
   if (row >= M) return;
   const T* src = in + (int64_t)row * ld;
   T* dst = out + (int64_t)row * ld;
-  float v[VMAX];
-  int cnt = 0;
+  float v[kMaxChunks][8];
+  int nc = 0;
   float m = -3.0e38f;
-  for (int i = lane; i < N; i += 64) {
-    v[cnt] = (float)src[i];
-    m = fmaxf(m, v[cnt]);
-    ++cnt;
+  for (int c = lane; c * 8 < N; c += 64, ++nc) {
+    load8(src + c * 8, v[nc]);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) m = fmaxf(m, v[nc][j]);
   }
   m = wave_reduce_max(m);
   float s = 0.f;
-  for (int c = 0; c < cnt; ++c) {
-    v[c] = __expf(v[c] - m);
-    s += v[c];
-  }
+  for (int q = 0; q < nc; ++q)
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      v[q][j] = __expf(v[q][j] - m);
+      s += v[q][j];
+    }
   s = wave_reduce_sum(s);
   float inv = 1.0f / s;
-  cnt = 0;
-  for (int i = lane; i < N; i += 64) dst[i] = (T)(v[cnt++] * inv);
+  nc = 0;
+  for (int c = lane; c * 8 < N; c += 64, ++nc) {
+#pragma unroll
+    for (int j = 0; j < 8; ++j) v[nc][j] *= inv;
+    store8(dst + c * 8, v[nc]);
+  }
 }
 
-// One wave per row layernorm: out = (x - mean) / sqrt(var + eps) * gamma + beta
+// layernorm core shared by the plain and residual-add variants.
+template <typename T, bool ADD>
+__device__ __forceinline__ void layernorm_row(
+    const T* __restrict__ src, const T* __restrict__ res,
+    const float* __restrict__ gamma, const float* __restrict__ beta,
+    T* __restrict__ dst, T* __restrict__ sum_out, int N, float eps,
+    int lane) {
+  float v[kMaxChunks][8];
+  int nc = 0;
+  float s = 0.f;
+  for (int c = lane; c * 8 < N; c += 64, ++nc) {
+    load8(src + c * 8, v[nc]);
+    if constexpr (ADD) {
+      float r[8];
+      load8(res + c * 8, r);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) v[nc][j] += r[j];
+    }
+#pragma unroll
+    for (int j = 0; j < 8; ++j) s += v[nc][j];
+  }
+  if (ADD && sum_out) {
+    nc = 0;
+    for (int c = lane; c * 8 < N; c += 64, ++nc) store8(sum_out + c * 8, v[nc]);
+  }
+  s = wave_reduce_sum(s);
+  float mean = s / (float)N;
+  float q = 0.f;
+  for (int p = 0; p < nc; ++p)
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float d = v[p][j] - mean;
+      q += d * d;
+    }
+  q = wave_reduce_sum(q);
+  float rstd = rsqrtf(q / (float)N + eps);
+  nc = 0;
+  for (int c = lane; c * 8 < N; c += 64, ++nc) {
+    float g8[8], b8[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      g8[j] = gamma[c * 8 + j];
+      b8[j] = beta[c * 8 + j];
+    }
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      v[nc][j] = (v[nc][j] - mean) * rstd * g8[j] + b8[j];
+    store8(dst + c * 8, v[nc]);
+  }
+}
+
 template <typename T>
 __global__ void layernorm_kernel(const T* __restrict__ in,
                                  const float* __restrict__ gamma,
                                  const float* __restrict__ beta,
                                  T* __restrict__ out, int M, int N, int64_t ld,
                                  float eps) {
-  constexpr int VMAX = 32;
   int lane = threadIdx.x & 63;
   int row = (blockIdx.x * (blockDim.x >> 6)) + (threadIdx.x >> 6);
   if (row >= M) return;
-  const T* src = in + (int64_t)row * ld;
-  T* dst = out + (int64_t)row * ld;
-  float v[VMAX];
-  int cnt = 0;
-  float s = 0.f;
-  for (int i = lane; i < N; i += 64) {
-    v[cnt] = (float)src[i];
-    s += v[cnt];
-    ++cnt;
-  }
-  s = wave_reduce_sum(s);
-  float mean = s / (float)N;
-  float q = 0.f;
-  for (int c = 0; c < cnt; ++c) {
-    float d = v[c] - mean;
-    q += d * d;
-  }
-  q = wave_reduce_sum(q);
-  float rstd = rsqrtf(q / (float)N + eps);
-  cnt = 0;
-  for (int i = lane; i < N; i += 64) {
-    float y = (v[cnt++] - mean) * rstd * gamma[i] + beta[i];
-    dst[i] = (T)y;
-  }
+  layernorm_row<T, false>(in + (int64_t)row * ld, nullptr, gamma, beta,
+                          out + (int64_t)row * ld, nullptr, N, eps, lane);
 }
 
-// Residual-add + layernorm fused (transformer block epilogue):
-// out = LN(x + res), also writes the sum if sum_out != nullptr.
 template <typename T>
 __global__ void add_layernorm_kernel(const T* __restrict__ x,
                                      const T* __restrict__ res,
                                      const float* __restrict__ gamma,
                                      const float* __restrict__ beta,
-                                     T* __restrict__ out, T* __restrict__ sum_out,
-                                     int M, int N, int64_t ld, float eps) {
-  constexpr int VMAX = 32;
+                                     T* __restrict__ out,
+                                     T* __restrict__ sum_out, int M, int N,
+                                     int64_t ld, float eps) {
   int lane = threadIdx.x & 63;
   int row = (blockIdx.x * (blockDim.x >> 6)) + (threadIdx.x >> 6);
   if (row >= M) return;
-  const T* src = x + (int64_t)row * ld;
-  const T* rsc = res + (int64_t)row * ld;
-  T* dst = out + (int64_t)row * ld;
-  float v[VMAX];
-  int cnt = 0;
-  float s = 0.f;
-  for (int i = lane; i < N; i += 64) {
-    v[cnt] = (float)src[i] + (float)rsc[i];
-    s += v[cnt];
-    ++cnt;
-  }
-  if (sum_out) {
-    T* so = sum_out + (int64_t)row * ld;
-    cnt = 0;
-    for (int i = lane; i < N; i += 64) so[i] = (T)v[cnt++];
-  }
-  s = wave_reduce_sum(s);
-  float mean = s / (float)N;
-  float q = 0.f;
-  for (int c = 0; c < cnt; ++c) {
-    float d = v[c] - mean;
-    q += d * d;
-  }
-  q = wave_reduce_sum(q);
-  float rstd = rsqrtf(q / (float)N + eps);
-  cnt = 0;
-  for (int i = lane; i < N; i += 64) {
-    float y = (v[cnt++] - mean) * rstd * gamma[i] + beta[i];
-    dst[i] = (T)y;
-  }
+  layernorm_row<T, true>(x + (int64_t)row * ld, res + (int64_t)row * ld,
+                         gamma, beta, out + (int64_t)row * ld,
+                         sum_out ? sum_out + (int64_t)row * ld : nullptr, N,
+                         eps, lane);
 }
 
 static inline dim3 rows_grid(int M) { return dim3((unsigned)cdiv(M, 4)); }
@@ -136,6 +159,7 @@ static inline dim3 rows_grid(int M) { return dim3((unsigned)cdiv(M, 4)); }
 void launch_softmax_rows(int dtype, const void* in, void* out, int M, int N,
                          int64_t ld, hipStream_t stream) {
   if (N > 2048) throw std::runtime_error("softmax_rows: N > 2048 unsupported");
+  if (N % 8 != 0) throw std::runtime_error("softmax_rows: N % 8 != 0");
   if (dtype == 0)
     hipLaunchKernelGGL((softmax_rows_kernel<_Float16>), rows_grid(M), dim3(256),
                        0, stream, (const _Float16*)in, (_Float16*)out, M, N, ld);
@@ -147,7 +171,7 @@ void launch_softmax_rows(int dtype, const void* in, void* out, int M, int N,
 void launch_layernorm(int dtype, const void* in, const float* gamma,
                       const float* beta, void* out, int M, int N, int64_t ld,
                       float eps, hipStream_t stream) {
-  if (N > 2048) throw std::runtime_error("layernorm: N > 2048 unsupported");
+  if (N > 2048 || N % 8 != 0) throw std::runtime_error("layernorm: bad N");
   if (dtype == 0)
     hipLaunchKernelGGL((layernorm_kernel<_Float16>), rows_grid(M), dim3(256), 0,
                        stream, (const _Float16*)in, gamma, beta, (_Float16*)out,
@@ -162,7 +186,7 @@ void launch_add_layernorm(int dtype, const void* x, const void* res,
                           const float* gamma, const float* beta, void* out,
                           void* sum_out, int M, int N, int64_t ld, float eps,
                           hipStream_t stream) {
-  if (N > 2048) throw std::runtime_error("add_layernorm: N > 2048 unsupported");
+  if (N > 2048 || N % 8 != 0) throw std::runtime_error("add_layernorm: bad N");
   if (dtype == 0)
     hipLaunchKernelGGL((add_layernorm_kernel<_Float16>), rows_grid(M),
                        dim3(256), 0, stream, (const _Float16*)x,
