@@ -48,6 +48,19 @@ OpDesc op_from_dict(const py::dict& d) {
   return o;
 }
 
+// Lazily-grown scratch for the raw ops.* test path (serial use only; the
+// engine allocates per-context scratch for concurrent replay).
+void* test_scratch(size_t need) {
+  static void* p = nullptr;
+  static size_t cap = 0;
+  if (need > cap) {
+    if (p) hipFree(p);
+    TRT_HIP_CHECK(hipMalloc(&p, need));
+    cap = need;
+  }
+  return need ? p : nullptr;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(_C, m) {
@@ -137,7 +150,8 @@ PYBIND11_MODULE(_C, m) {
              uintptr_t stream, bool sync, int tile) {
             launch_gemm_bt(dtype, (void*)A, (void*)B, (void*)C, (float*)scale,
                            (float*)bias, (void*)residual, M, N, K, K, K, N,
-                           epi, as_stream(stream), tile);
+                           epi, as_stream(stream), tile,
+                           test_scratch(gemm_scratch_bytes(M, N, K)));
             if (sync) TRT_HIP_CHECK(hipStreamSynchronize(as_stream(stream)));
           },
           py::arg("dtype"), py::arg("A"), py::arg("B"), py::arg("C"),
@@ -154,7 +168,9 @@ PYBIND11_MODULE(_C, m) {
             launch_conv2d(dtype, (void*)in, (void*)Wt, (void*)out,
                           (float*)scale, (float*)bias, (void*)residual,
                           (void*)zero_page, Nb, H, W, C, Cout, KH, KW, sh, sw,
-                          ph, pw, epi, as_stream(stream), tile);
+                          ph, pw, epi, as_stream(stream), tile,
+                          test_scratch(conv_scratch_bytes(
+                              Nb, H, W, C, Cout, KH, KW, sh, sw, ph, pw)));
             if (sync) TRT_HIP_CHECK(hipStreamSynchronize(as_stream(stream)));
           },
           py::arg("dtype"), py::arg("in"), py::arg("Wt"), py::arg("out"),

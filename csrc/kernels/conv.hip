@@ -69,18 +69,26 @@ __device__ __forceinline__ void stage_conv_a(
   }
 }
 
-template <typename T, Epi E, int BM, int BN>
+void launch_splitk_reduce(int dtype, const float* scratch, void* C,
+                          const float* scale, const float* bias,
+                          const void* residual, int M, int N, int64_t ldc,
+                          int tiles_m, int tiles_n, int splitk, int bm,
+                          int bn, int epi, hipStream_t stream);
+
+template <typename T, Epi E, int BM, int BN, bool SPLIT>
 __global__ __launch_bounds__(256) void conv_igemm_kernel(
     const T* __restrict__ in, const T* __restrict__ Wt, T* __restrict__ out,
     const float* __restrict__ scale, const float* __restrict__ bias,
     const T* __restrict__ residual, const T* __restrict__ zero_page,
-    const ConvParams p, int tiles_n) {
+    const ConvParams p, int tiles_n, float* __restrict__ scratch, int splitk,
+    int ktper) {
   constexpr int kABytes = BM * 128;
   constexpr int kBuf = (BM + BN) * 128;
 
   uint32_t bid = xcd_swizzle(blockIdx.x, gridDim.x);
-  int m0 = (int)(bid / tiles_n) * BM;
-  int n0 = (int)(bid % tiles_n) * BN;
+  uint32_t tile = SPLIT ? bid / splitk : bid;
+  int m0 = (int)(tile / tiles_n) * BM;
+  int n0 = (int)(tile % tiles_n) * BN;
 
   __shared__ __attribute__((aligned(16))) char smem[2 * kBuf];
   uint32_t lds0 = (uint32_t)(uintptr_t)&smem[0];
@@ -98,16 +106,22 @@ __global__ __launch_bounds__(256) void conv_igemm_kernel(
     for (int j = 0; j < BN / 32; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
   const int ktiles = p.K >> 6;
+  int kt0 = 0, kt1 = ktiles;
+  if constexpr (SPLIT) {
+    int slice = bid % splitk;
+    kt0 = slice * ktper;
+    kt1 = min(ktiles, kt0 + ktper);
+  }
 
-  stage_conv_a<T, BM>(in, zero_page, p, m0, 0, lds0, tid);
-  stage_tile<T, BN>(Wt + (int64_t)n0 * p.K, p.K, n0, p.Cout, lds0 + kABytes,
-                    tid);
+  stage_conv_a<T, BM>(in, zero_page, p, m0, kt0 * 64, lds0, tid);
+  stage_tile<T, BN>(Wt + (int64_t)n0 * p.K + kt0 * 64, p.K, n0, p.Cout,
+                    lds0 + kABytes, tid);
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   __syncthreads();
 
   int cur = 0;
-  for (int t = 0; t < ktiles; ++t) {
-    if (t + 1 < ktiles) {
+  for (int t = kt0; t < kt1; ++t) {
+    if (t + 1 < kt1) {
       uint32_t nb = lds0 + (cur ^ 1) * kBuf;
       stage_conv_a<T, BM>(in, zero_page, p, m0, (t + 1) * 64, nb, tid);
       stage_tile<T, BN>(Wt + (int64_t)n0 * p.K + (t + 1) * 64, p.K, n0,
@@ -120,8 +134,26 @@ __global__ __launch_bounds__(256) void conv_igemm_kernel(
     cur ^= 1;
   }
 
-  store_epilogue<T, E, BM, BN>(acc, out, p.Cout, m0, n0, p.M, p.Cout, scale,
-                               bias, residual, lane, wr, wc);
+  if constexpr (SPLIT) {
+    store_splitk<T, BM, BN>(acc, scratch + (int64_t)bid * BM * BN, lane, wr,
+                            wc);
+  } else {
+    store_epilogue<T, E, BM, BN>(acc, out, p.Cout, m0, n0, p.M, p.Cout, scale,
+                                 bias, residual, lane, wr, wc);
+  }
+}
+
+size_t conv_scratch_bytes(int Nb, int H, int W, int C, int Cout, int KH,
+                          int KW, int sh, int sw, int ph, int pw) {
+  int OH = (H + 2 * ph - KH) / sh + 1;
+  int OW = (W + 2 * pw - KW) / sw + 1;
+  int M = Nb * OH * OW;
+  int K = (int)round_up(KH * KW * C, 64);
+  TileCfg cfg = pick_tile(M, Cout);
+  long tiles = cdiv(M, cfg.bm) * cdiv(Cout, cfg.bn);
+  int splitk = pick_splitk(tiles, K >> 6);
+  if (splitk == 1) return 0;
+  return (size_t)tiles * splitk * cfg.bm * cfg.bn * 4;
 }
 
 template <typename T>
@@ -129,21 +161,43 @@ static void launch_conv2d_t(const void* in, const void* Wt, void* out,
                             const float* scale, const float* bias,
                             const void* residual, const void* zero_page,
                             const ConvParams& p, int epi, hipStream_t stream,
-                            int tile) {
+                            int tile, float* scratch) {
   TileCfg cfg = tile ? tile_from_code(tile) : pick_tile(p.M, p.Cout);
   int tiles_m = (int)cdiv(p.M, cfg.bm);
   int tiles_n = (int)cdiv(p.Cout, cfg.bn);
-  dim3 grid(tiles_m * tiles_n);
+  long tiles = (long)tiles_m * tiles_n;
+  int ktiles = p.K >> 6;
+  int splitk = (!tile && scratch) ? pick_splitk(tiles, ktiles) : 1;
   dim3 block(256);
+  int dtype = std::is_same<T, _Float16>::value ? 0 : 1;
+  if (splitk > 1) {
+    int ktper = (int)cdiv(ktiles, splitk);
+    dim3 grid((unsigned)(tiles * splitk));
+    tile_dispatch(cfg, [&](auto bm, auto bn) {
+      constexpr int BM = decltype(bm)::value;
+      constexpr int BN = decltype(bn)::value;
+      hipLaunchKernelGGL((conv_igemm_kernel<T, Epi::kNone, BM, BN, true>),
+                         grid, block, 0, stream, (const T*)in, (const T*)Wt,
+                         (T*)out, scale, bias, (const T*)residual,
+                         (const T*)zero_page, p, tiles_n, scratch, splitk,
+                         ktper);
+    });
+    launch_splitk_reduce(dtype, scratch, out, scale, bias, residual, p.M,
+                         p.Cout, p.Cout, tiles_m, tiles_n, splitk, cfg.bm,
+                         cfg.bn, epi, stream);
+    return;
+  }
+  dim3 grid((unsigned)tiles);
   epi_dispatch(epi, [&](auto e) {
     constexpr Epi EE = decltype(e)::value;
     tile_dispatch(cfg, [&](auto bm, auto bn) {
       constexpr int BM = decltype(bm)::value;
       constexpr int BN = decltype(bn)::value;
-      hipLaunchKernelGGL((conv_igemm_kernel<T, EE, BM, BN>), grid, block, 0,
-                         stream, (const T*)in, (const T*)Wt, (T*)out, scale,
-                         bias, (const T*)residual, (const T*)zero_page, p,
-                         tiles_n);
+      hipLaunchKernelGGL((conv_igemm_kernel<T, EE, BM, BN, false>), grid,
+                         block, 0, stream, (const T*)in, (const T*)Wt,
+                         (T*)out, scale, bias, (const T*)residual,
+                         (const T*)zero_page, p, tiles_n, (float*)nullptr, 1,
+                         ktiles);
     });
   });
 }
@@ -152,7 +206,7 @@ void launch_conv2d(int dtype, const void* in, const void* Wt, void* out,
                    const float* scale, const float* bias, const void* residual,
                    const void* zero_page, int Nb, int H, int W, int C,
                    int Cout, int KH, int KW, int sh, int sw, int ph, int pw,
-                   int epi, hipStream_t stream, int tile) {
+                   int epi, hipStream_t stream, int tile, void* scratch) {
   ConvParams p;
   p.Nb = Nb; p.H = H; p.W = W; p.C = C;
   p.Cout = Cout; p.KH = KH; p.KW = KW;
@@ -169,10 +223,10 @@ void launch_conv2d(int dtype, const void* in, const void* Wt, void* out,
   p.d_kw = make_fastdiv((uint32_t)KW);
   if (dtype == 0)
     launch_conv2d_t<_Float16>(in, Wt, out, scale, bias, residual, zero_page, p,
-                              epi, stream, tile);
+                              epi, stream, tile, (float*)scratch);
   else
     launch_conv2d_t<__bf16>(in, Wt, out, scale, bias, residual, zero_page, p,
-                            epi, stream, tile);
+                            epi, stream, tile, (float*)scratch);
 }
 
 }  // namespace trtlab

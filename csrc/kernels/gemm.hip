@@ -13,7 +13,10 @@
 //     double-buffered; one vmcnt(0) + barrier per K-tile.
 //   - LDS XOR swizzle ((row&7)<<4) applied on the SOURCE address and
 //     re-applied on every ds_read_b128 (both-sides-or-neither, rule 21).
-//   - XCD-aware bijective blockIdx swizzle (T1).
+//   - XCD-aware bijective blockIdx swizzle (T1); a split tile's K-slices
+//     stay adjacent (same XCD) per the split-K guidance.
+//   - split-K for K-heavy grid-starved shapes: fp32 slabs + a deterministic
+//     reduce kernel (splitk.hip) that applies the epilogue.
 //
 // Replaces the TensorRT-internal GEMM path of the reference
 // (trtlab/tensorrt/src/workspace.cc:47 enqueueV2 — opaque engine kernels).
@@ -21,18 +24,34 @@
 
 namespace trtlab {
 
-template <typename T, Epi E, int BM, int BN>
+void launch_splitk_reduce(int dtype, const float* scratch, void* C,
+                          const float* scale, const float* bias,
+                          const void* residual, int M, int N, int64_t ldc,
+                          int tiles_m, int tiles_n, int splitk, int bm,
+                          int bn, int epi, hipStream_t stream);
+
+template <typename T, Epi E, int BM, int BN, bool SPLIT>
 __global__ __launch_bounds__(256) void gemm_bt_kernel(
     const T* __restrict__ A, const T* __restrict__ B, T* __restrict__ C,
     const float* __restrict__ scale, const float* __restrict__ bias,
     const T* __restrict__ residual, int M, int N, int K, int64_t lda,
-    int64_t ldb, int64_t ldc, int tiles_n) {
+    int64_t ldb, int64_t ldc, int tiles_n, float* __restrict__ scratch,
+    int splitk, int ktper) {
   constexpr int kABytes = BM * 128;
   constexpr int kBuf = (BM + BN) * 128;
 
   uint32_t bid = xcd_swizzle(blockIdx.x, gridDim.x);
-  int m0 = (int)(bid / tiles_n) * BM;
-  int n0 = (int)(bid % tiles_n) * BN;
+  uint32_t tile = SPLIT ? bid / splitk : bid;
+  int m0 = (int)(tile / tiles_n) * BM;
+  int n0 = (int)(tile % tiles_n) * BN;
+
+  const int ktiles = K >> 6;  // K % 64 == 0 (host asserts)
+  int kt0 = 0, kt1 = ktiles;
+  if constexpr (SPLIT) {
+    int slice = bid % splitk;
+    kt0 = slice * ktper;
+    kt1 = min(ktiles, kt0 + ktper);
+  }
 
   __shared__ __attribute__((aligned(16))) char smem[2 * kBuf];
   uint32_t lds0 = (uint32_t)(uintptr_t)&smem[0];
@@ -49,16 +68,15 @@ __global__ __launch_bounds__(256) void gemm_bt_kernel(
 #pragma unroll
     for (int j = 0; j < BN / 32; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
-  const int ktiles = K >> 6;  // K % 64 == 0 (host asserts)
-
-  stage_tile<T, BM>(A + (int64_t)m0 * lda, lda, m0, M, lds0, tid);
-  stage_tile<T, BN>(B + (int64_t)n0 * ldb, ldb, n0, N, lds0 + kABytes, tid);
+  stage_tile<T, BM>(A + (int64_t)m0 * lda + kt0 * 64, lda, m0, M, lds0, tid);
+  stage_tile<T, BN>(B + (int64_t)n0 * ldb + kt0 * 64, ldb, n0, N,
+                    lds0 + kABytes, tid);
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   __syncthreads();
 
   int cur = 0;
-  for (int t = 0; t < ktiles; ++t) {
-    if (t + 1 < ktiles) {
+  for (int t = kt0; t < kt1; ++t) {
+    if (t + 1 < kt1) {
       uint32_t nb = lds0 + (cur ^ 1) * kBuf;
       stage_tile<T, BM>(A + (int64_t)m0 * lda + (t + 1) * 64, lda, m0, M, nb,
                         tid);
@@ -72,8 +90,21 @@ __global__ __launch_bounds__(256) void gemm_bt_kernel(
     cur ^= 1;
   }
 
-  store_epilogue<T, E, BM, BN>(acc, C, ldc, m0, n0, M, N, scale, bias,
-                               residual, lane, wr, wc);
+  if constexpr (SPLIT) {
+    store_splitk<T, BM, BN>(acc, scratch + (int64_t)bid * BM * BN, lane, wr,
+                            wc);
+  } else {
+    store_epilogue<T, E, BM, BN>(acc, C, ldc, m0, n0, M, N, scale, bias,
+                                 residual, lane, wr, wc);
+  }
+}
+
+size_t gemm_scratch_bytes(int M, int N, int K) {
+  TileCfg cfg = pick_tile(M, N);
+  long tiles = cdiv(M, cfg.bm) * cdiv(N, cfg.bn);
+  int splitk = pick_splitk(tiles, K >> 6);
+  if (splitk == 1) return 0;
+  return (size_t)tiles * splitk * cfg.bm * cfg.bn * 4;
 }
 
 template <typename T>
@@ -81,20 +112,41 @@ static void launch_gemm_bt_t(const void* A, const void* B, void* C,
                              const float* scale, const float* bias,
                              const void* residual, int M, int N, int K,
                              int64_t lda, int64_t ldb, int64_t ldc, int epi,
-                             hipStream_t stream, int tile) {
+                             hipStream_t stream, int tile, float* scratch) {
   TileCfg cfg = tile ? tile_from_code(tile) : pick_tile(M, N);
   int tiles_m = (int)cdiv(M, cfg.bm);
   int tiles_n = (int)cdiv(N, cfg.bn);
-  dim3 grid(tiles_m * tiles_n);
+  long tiles = (long)tiles_m * tiles_n;
+  int ktiles = K >> 6;
+  int splitk = (!tile && scratch) ? pick_splitk(tiles, ktiles) : 1;
   dim3 block(256);
+  int dtype = std::is_same<T, _Float16>::value ? 0 : 1;
+  if (splitk > 1) {
+    int ktper = (int)cdiv(ktiles, splitk);
+    dim3 grid((unsigned)(tiles * splitk));
+    tile_dispatch(cfg, [&](auto bm, auto bn) {
+      constexpr int BM = decltype(bm)::value;
+      constexpr int BN = decltype(bn)::value;
+      hipLaunchKernelGGL((gemm_bt_kernel<T, Epi::kNone, BM, BN, true>), grid,
+                         block, 0, stream, (const T*)A, (const T*)B, (T*)C,
+                         scale, bias, (const T*)residual, M, N, K, lda, ldb,
+                         ldc, tiles_n, scratch, splitk, ktper);
+    });
+    launch_splitk_reduce(dtype, scratch, C, scale, bias, residual, M, N, ldc,
+                         tiles_m, tiles_n, splitk, cfg.bm, cfg.bn, epi,
+                         stream);
+    return;
+  }
+  dim3 grid((unsigned)tiles);
   epi_dispatch(epi, [&](auto e) {
     constexpr Epi EE = decltype(e)::value;
     tile_dispatch(cfg, [&](auto bm, auto bn) {
       constexpr int BM = decltype(bm)::value;
       constexpr int BN = decltype(bn)::value;
-      hipLaunchKernelGGL((gemm_bt_kernel<T, EE, BM, BN>), grid, block, 0,
-                         stream, (const T*)A, (const T*)B, (T*)C, scale, bias,
-                         (const T*)residual, M, N, K, lda, ldb, ldc, tiles_n);
+      hipLaunchKernelGGL((gemm_bt_kernel<T, EE, BM, BN, false>), grid, block,
+                         0, stream, (const T*)A, (const T*)B, (T*)C, scale,
+                         bias, (const T*)residual, M, N, K, lda, ldb, ldc,
+                         tiles_n, (float*)nullptr, 1, ktiles);
     });
   });
 }
@@ -103,14 +155,14 @@ void launch_gemm_bt(int dtype,  // 0 = fp16, 1 = bf16
                     const void* A, const void* B, void* C, const float* scale,
                     const float* bias, const void* residual, int M, int N,
                     int K, int64_t lda, int64_t ldb, int64_t ldc, int epi,
-                    hipStream_t stream, int tile) {
+                    hipStream_t stream, int tile, void* scratch) {
   if (K % 64 != 0) throw std::runtime_error("gemm_bt: K must be a multiple of 64");
   if (dtype == 0)
     launch_gemm_bt_t<_Float16>(A, B, C, scale, bias, residual, M, N, K, lda,
-                               ldb, ldc, epi, stream, tile);
+                               ldb, ldc, epi, stream, tile, (float*)scratch);
   else
     launch_gemm_bt_t<__bf16>(A, B, C, scale, bias, residual, M, N, K, lda, ldb,
-                             ldc, epi, stream, tile);
+                             ldc, epi, stream, tile, (float*)scratch);
 }
 
 }  // namespace trtlab

@@ -159,6 +159,33 @@ __device__ __forceinline__ void store_epilogue(
   }
 }
 
+// Split-K slab store: this (tile, slice)'s [BM][BN] fp32 partial sums.
+template <typename T, int BM, int BN>
+__device__ __forceinline__ void store_splitk(f32x4 (&acc)[BM / 32][BN / 32],
+                                             float* __restrict__ slab,
+                                             int lane, int wr, int wc) {
+#pragma unroll
+  for (int i = 0; i < BM / 32; ++i)
+#pragma unroll
+    for (int j = 0; j < BN / 32; ++j) {
+      int col = wc * (BN / 2) + j * 16 + (lane & 15);
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = wr * (BM / 2) + i * 16 + ((lane >> 4) << 2) + r;
+        slab[row * BN + col] = acc[i][j][r];
+      }
+    }
+}
+
+// Split-K decision: split the K loop across slices until the grid fills the
+// chip (256 CUs), keeping >= 2 K-tiles per slice. Returns 1 = no split.
+inline int pick_splitk(long blocks, int ktiles) {
+  int splitk = 1;
+  while (blocks * splitk < 256 && ktiles / (splitk * 2) >= 2 && splitk < 8)
+    splitk *= 2;
+  return splitk;
+}
+
 // Host-side tile-config choice: prefer the config that fills the chip
 // (>=512 workgroups) at the highest tile utilization; otherwise maximize
 // parallelism x utilization. Small deep-layer shapes (ResNet stage 4/5 at

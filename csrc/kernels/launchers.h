@@ -9,13 +9,18 @@ namespace trtlab {
 void launch_gemm_bt(int dtype, const void* A, const void* B, void* C,
                     const float* scale, const float* bias, const void* residual,
                     int M, int N, int K, int64_t lda, int64_t ldb, int64_t ldc,
-                    int epi, hipStream_t stream, int tile = 0);
+                    int epi, hipStream_t stream, int tile = 0,
+                    void* scratch = nullptr);
+size_t gemm_scratch_bytes(int M, int N, int K);
 
 void launch_conv2d(int dtype, const void* in, const void* Wt, void* out,
                    const float* scale, const float* bias, const void* residual,
                    const void* zero_page, int Nb, int H, int W, int C,
                    int Cout, int KH, int KW, int sh, int sw, int ph, int pw,
-                   int epi, hipStream_t stream, int tile = 0);
+                   int epi, hipStream_t stream, int tile = 0,
+                   void* scratch = nullptr);
+size_t conv_scratch_bytes(int Nb, int H, int W, int C, int Cout, int KH,
+                          int KW, int sh, int sw, int ph, int pw);
 
 void launch_maxpool2d(int dtype, const void* in, void* out, int Nb, int H,
                       int W, int C, int KH, int KW, int sh, int sw, int ph,

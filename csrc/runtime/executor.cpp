@@ -24,6 +24,17 @@ Engine::Engine(int device, const void* weights, size_t weight_bytes,
     TRT_HIP_CHECK(hipMemcpy(weights_, weights, weight_bytes_, hipMemcpyHostToDevice));
   zero_page_ = device_malloc(256, device_);
   TRT_HIP_CHECK(hipMemset(zero_page_, 0, 256));
+  // Split-K slab workspace: sized to the worst op in the plan.
+  for (const OpDesc& op : ops_) {
+    if (op.kind == kConv2d)
+      scratch_bytes_ = std::max(
+          scratch_bytes_,
+          conv_scratch_bytes(op.Nb, op.H, op.W, op.C, op.Cout, op.KH, op.KW,
+                             op.sh, op.sw, op.ph, op.pw));
+    else if (op.kind == kGemmBt)
+      scratch_bytes_ =
+          std::max(scratch_bytes_, gemm_scratch_bytes(op.M, op.N, op.K));
+  }
   TRT_HIP_CHECK(hipDeviceSynchronize());
 }
 
@@ -42,6 +53,8 @@ ExecutionContext::ExecutionContext(std::shared_ptr<Engine> engine)
     : eng_(std::move(engine)) {
   TRT_HIP_CHECK(hipSetDevice(eng_->device()));
   arena_ = (char*)device_malloc(eng_->arena_bytes(), eng_->device());
+  if (eng_->scratch_bytes())
+    scratch_ = (char*)device_malloc(eng_->scratch_bytes(), eng_->device());
   h_in_ = (char*)pinned_malloc(eng_->input_bytes());
   h_out_ = (char*)pinned_malloc(eng_->output_bytes());
   TRT_HIP_CHECK(hipStreamCreateWithFlags(&stream_, hipStreamNonBlocking));
@@ -52,6 +65,7 @@ ExecutionContext::~ExecutionContext() {
   if (graph_) hipGraphDestroy(graph_);
   hipStreamDestroy(stream_);
   device_free(arena_, eng_->arena_bytes());
+  if (scratch_) device_free(scratch_, eng_->scratch_bytes());
   pinned_free(h_in_, eng_->input_bytes());
   pinned_free(h_out_, eng_->output_bytes());
 }
@@ -76,13 +90,14 @@ void ExecutionContext::enqueue_all(hipStream_t s) {
         launch_conv2d(op.dtype, A(op.in_off), Wp(op.w_off), A(op.out_off),
                       Fp(op.scale_off), Fp(op.bias_off), A(op.in2_off),
                       eng_->zero_page(), op.Nb, op.H, op.W, op.C, op.Cout,
-                      op.KH, op.KW, op.sh, op.sw, op.ph, op.pw, op.epi, s);
+                      op.KH, op.KW, op.sh, op.sw, op.ph, op.pw, op.epi, s, 0,
+                      scratch_);
         break;
       case kGemmBt:
         launch_gemm_bt(op.dtype, A(op.in_off), Wp(op.w_off), A(op.out_off),
                        Fp(op.scale_off), Fp(op.bias_off), A(op.in2_off), op.M,
                        op.N, op.K, op.K /*lda*/, op.K /*ldb*/, op.N /*ldc*/,
-                       op.epi, s);
+                       op.epi, s, 0, scratch_);
         break;
       case kMaxPool:
         launch_maxpool2d(op.dtype, A(op.in_off), A(op.out_off), op.Nb, op.H,

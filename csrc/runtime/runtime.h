@@ -84,6 +84,7 @@ class Engine {
 
   int device() const { return device_; }
   size_t arena_bytes() const { return arena_bytes_; }
+  size_t scratch_bytes() const { return scratch_bytes_; }
   size_t input_bytes() const { return input_bytes_; }
   size_t output_bytes() const { return output_bytes_; }
   int64_t input_off() const { return input_off_; }
@@ -102,6 +103,7 @@ class Engine {
   size_t weight_bytes_ = 0;
   void* zero_page_ = nullptr;
   size_t arena_bytes_;
+  size_t scratch_bytes_ = 0;  // split-K slab workspace (max over ops)
   std::vector<OpDesc> ops_;
   int64_t input_off_, output_off_;
   size_t input_bytes_, output_bytes_;
@@ -130,6 +132,7 @@ class ExecutionContext {
  private:
   std::shared_ptr<Engine> eng_;
   char* arena_ = nullptr;
+  char* scratch_ = nullptr;
   char* h_in_ = nullptr;
   char* h_out_ = nullptr;
   hipStream_t stream_{};
