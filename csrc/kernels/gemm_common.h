@@ -177,11 +177,13 @@ __device__ __forceinline__ void store_splitk(f32x4 (&acc)[BM / 32][BN / 32],
     }
 }
 
-// Split-K decision: split the K loop across slices until the grid fills the
-// chip (256 CUs), keeping >= 2 K-tiles per slice. Returns 1 = no split.
+// Split-K decision: only for severely grid-starved, K-heavy shapes (the
+// slab round-trip + reduce launch costs ~8-10 us, so marginal cases lose —
+// measured in tools/tune_tiles.py). Returns 1 = no split.
 inline int pick_splitk(long blocks, int ktiles) {
+  if (blocks > 96 || ktiles < 32) return 1;
   int splitk = 1;
-  while (blocks * splitk < 256 && ktiles / (splitk * 2) >= 2 && splitk < 8)
+  while (blocks * splitk < 256 && ktiles / (splitk * 2) >= 4 && splitk < 8)
     splitk *= 2;
   return splitk;
 }
