@@ -1,0 +1,110 @@
+"""Inference service tests: CPU path uses a fake manager (the reference's
+echo-backend pattern); the GPU test runs the real engine behind gRPC."""
+import threading
+from concurrent.futures import Future
+
+import numpy as np
+import pytest
+
+from trtlab_amd.rpc import InferRequest, InferResponse, SyncClient
+from trtlab_amd.rpc.server import Server
+from trtlab_amd.rpc.service import InferenceResources, InferenceService
+
+
+class _FakePlan:
+    input_shape = (2, 4)
+    output_shape = (2, 4)
+
+
+class _FakeEngine:
+    plan = _FakePlan()
+
+
+class _FakeRunner:
+    """Doubles the input asynchronously (stands in for InferRunner)."""
+
+    def infer(self, batch):
+        fut = Future()
+
+        def run():
+            fut.set_result((batch.astype(np.float32) * 2).astype(np.float16))
+
+        threading.Thread(target=run, daemon=True).start()
+        return fut
+
+
+class _FakeManager:
+    def infer_runner(self, name):
+        return _FakeRunner()
+
+    def get_model(self, name):
+        return _FakeEngine()
+
+
+@pytest.fixture()
+def service_server():
+    server = Server("127.0.0.1:0")
+    svc = InferenceService(InferenceResources(_FakeManager()))
+    server.register_service(svc)
+    server.register_service(svc.health_service)
+    server.async_start()
+    yield server
+    server.shutdown()
+
+
+def test_inference_service_roundtrip(service_server):
+    c = SyncClient(f"127.0.0.1:{service_server.port}")
+    x = np.arange(8, dtype=np.float16).reshape(2, 4)
+    resp = c.call("trtlab.Inference", "Compute",
+                  InferRequest(model="m", input=x.tobytes(), shape=[2, 4],
+                               dtype="f16", batch_id=5),
+                  InferResponse, timeout=10)
+    out = np.frombuffer(resp.output, dtype=np.float16).reshape(2, 4)
+    assert np.allclose(out, x * 2)
+    assert resp.batch_id == 5
+    assert resp.compute_ms >= 0
+    assert resp.request_ms >= resp.compute_ms
+    c.close()
+
+
+def test_metrics_observe():
+    from trtlab_amd.utils.metrics import Metrics
+
+    m = Metrics(port=0)  # not started: observe() only
+    m.observe(2.0, 3.0)
+    m.observe(1.0, 50.0)  # queueing-dominated: load ratio 50 -> last bucket
+
+
+@pytest.mark.gpu
+def test_inference_service_gpu_end_to_end():
+    from trtlab_amd.engine.planner import Planner
+    from trtlab_amd.engine.reference import run_reference
+    from trtlab_amd.engine.runtime import InferenceManager
+    from trtlab_amd.models import build_resnet
+
+    g = build_resnet(50, batch=2, image=64, seed=0)
+    plan = Planner().compile(g)
+    mgr = InferenceManager(max_contexts=2)
+    mgr.register_model("rn50", plan)
+    mgr.allocate_resources()
+
+    server = Server("127.0.0.1:0")
+    svc = InferenceService(InferenceResources(mgr))
+    server.register_service(svc)
+    server.async_start()
+    try:
+        c = SyncClient(f"127.0.0.1:{server.port}")
+        x = (np.random.RandomState(3).randn(*plan.input_shape) * 0.5).astype(np.float16)
+        resp = c.call("trtlab.Inference", "Compute",
+                      InferRequest(model="rn50", input=x.tobytes(),
+                                   shape=list(plan.input_shape), dtype="f16"),
+                      InferResponse, timeout=120)
+        out = np.frombuffer(resp.output, dtype=np.float16).reshape(
+            tuple(resp.shape)).astype(np.float32)
+        ref = run_reference(plan, x.astype(np.float32))
+        err = np.abs(out - ref).max() / max(np.abs(ref).max(), 1e-6)
+        assert err < 0.08, err
+        c.close()
+    finally:
+        server.shutdown()
+        mgr.shutdown()

@@ -1,0 +1,73 @@
+"""Inference gRPC service: wires an InferenceManager/InferRunner into the
+RPC layer — the reference's 02_TensorRT_GRPC server (FlowersContext::
+ExecuteRPC, server.cc:150-183) with its compute/request duration split and
+prometheus metrics, rebuilt for the native MI355X engine.
+"""
+from __future__ import annotations
+
+import asyncio
+import time
+from typing import Dict, Optional
+
+import numpy as np
+
+from trtlab_amd.core import Resources
+from trtlab_amd.rpc.proto import (HealthRequest, HealthResponse, InferRequest,
+                                  InferResponse)
+from trtlab_amd.rpc.server import AsyncService
+from trtlab_amd.utils import log
+
+
+class InferenceResources(Resources):
+    """Resources handle injected into RPC contexts (reference server.cc:287:
+    FlowersResources{InferenceManager})."""
+
+    def __init__(self, manager):
+        self.manager = manager  # engine.runtime.InferenceManager
+        self.runners: Dict[str, object] = {}
+
+    def runner(self, model: str):
+        if model not in self.runners:
+            self.runners[model] = self.manager.infer_runner(model)
+        return self.runners[model]
+
+
+class InferenceService(AsyncService):
+    """`trtlab.Inference/Compute` unary service. The RPC handler hops off
+    the event loop into the InferRunner's pre/hip/post pipeline (reference:
+    never block the CQ thread — server.cc:155 cuda_pool.enqueue) and
+    reports compute vs request ms (server.cc:175-177)."""
+
+    def __init__(self, resources: InferenceResources, metrics=None):
+        super().__init__("trtlab.Inference", resources)
+        self.metrics = metrics
+        self.register_unary("Compute", self._compute, InferRequest,
+                            InferResponse)
+
+        async def health(request, context, _r):
+            return HealthResponse(ready=True, status="serving")
+
+        hs = AsyncService("trtlab.Health")
+        hs.register_unary("Check", health, HealthRequest, HealthResponse)
+        self.health_service = hs
+
+    async def _compute(self, request: InferRequest, context, resources):
+        t_start = time.monotonic()
+        runner = resources.runner(request.model)
+        plan = resources.manager.get_model(request.model).plan
+        dtype = np.float16 if request.dtype in ("", "f16", "float16") else np.dtype(request.dtype)
+        batch = np.frombuffer(request.input, dtype=dtype)
+        shape = tuple(request.shape) or plan.input_shape
+        batch = batch.reshape(shape)
+
+        t_compute = time.monotonic()
+        fut = runner.infer(batch)
+        out = await asyncio.wrap_future(fut)
+        compute_ms = (time.monotonic() - t_compute) * 1e3
+        request_ms = (time.monotonic() - t_start) * 1e3
+        if self.metrics:
+            self.metrics.observe(compute_ms, request_ms)
+        return InferResponse(
+            output=out.tobytes(), shape=list(out.shape), dtype="f16",
+            batch_id=request.batch_id, compute_ms=compute_ms,
+            request_ms=request_ms)
