@@ -1,0 +1,303 @@
+"""ONNX model import/export for the trtlab_amd IR.
+
+Import: parses an .onnx file (protobuf wire, onnx_wire.py) into a Graph —
+the role the reference delegates to TensorRT's ONNX parser
+(examples/ONNX/resnet50/build.py). Supported op set (ResNet-family inference
+graphs): Conv, BatchNormalization, Relu, Add, MaxPool, GlobalAveragePool,
+Flatten, Gemm, MatMul, Softmax, LayerNormalization, Gelu.
+
+Layout note: ONNX activations are NCHW; the IR is NHWC. Op semantics here
+are layout-independent (weights are imported in their canonical ONNX shape
+[Cout, Cin, KH, KW] and the engine re-packs), so the importer only rewrites
+the input ValueInfo shape NCHW -> NHWC.
+
+Export: serializes an IR graph back to ONNX bytes (round-trip tested;
+also a plan-inspection artifact).
+"""
+from __future__ import annotations
+
+import struct
+from typing import Dict, List, Optional
+
+import numpy as np
+
+from trtlab_amd.engine import onnx_wire as w
+from trtlab_amd.engine.ir import Graph
+
+# onnx.proto3 field numbers
+_MODEL_GRAPH = 7
+_GRAPH_NODE, _GRAPH_NAME, _GRAPH_INIT = 1, 2, 5
+_GRAPH_INPUT, _GRAPH_OUTPUT = 11, 12
+_NODE_INPUT, _NODE_OUTPUT, _NODE_NAME, _NODE_OPTYPE, _NODE_ATTR = 1, 2, 3, 4, 5
+_ATTR_NAME, _ATTR_F, _ATTR_I, _ATTR_INTS = 1, 2, 3, 8
+_T_DIMS, _T_DTYPE, _T_FLOAT_DATA, _T_INT64_DATA, _T_NAME, _T_RAW = 1, 2, 4, 7, 8, 9
+_VI_NAME, _VI_TYPE = 1, 2
+_TP_TENSOR = 1
+_TT_ELEM, _TT_SHAPE = 1, 2
+_TS_DIM = 1
+_TD_VALUE = 1
+
+_F32, _F16, _I64 = 1, 10, 7
+
+
+# -------------------------------------------------------------------- load
+def _parse_tensor(buf: bytes) -> tuple[str, np.ndarray]:
+    d = w.fields_dict(buf)
+    dims = [w.varint_to_sint64(v) for v in d.get(_T_DIMS, [])]
+    dtype = d.get(_T_DTYPE, [_F32])[0]
+    name = d.get(_T_NAME, [b""])[0].decode()
+    if _T_RAW in d:
+        raw = d[_T_RAW][0]
+        np_dt = {_F32: np.float32, _F16: np.float16, _I64: np.int64}[dtype]
+        arr = np.frombuffer(raw, dtype=np_dt)
+    elif _T_FLOAT_DATA in d:
+        arr = np.array([struct.unpack("<f", struct.pack("<I", v))[0]
+                        if isinstance(v, int) else v
+                        for v in d[_T_FLOAT_DATA]], dtype=np.float32)
+    elif _T_INT64_DATA in d:
+        arr = np.array([w.varint_to_sint64(v) for v in d[_T_INT64_DATA]],
+                       dtype=np.int64)
+    else:
+        arr = np.zeros(0, np.float32)
+    return name, arr.reshape(dims) if dims else arr
+
+
+def _parse_attrs(bufs: List[bytes]) -> Dict[str, object]:
+    out: Dict[str, object] = {}
+    for b in bufs:
+        d = w.fields_dict(b)
+        name = d[_ATTR_NAME][0].decode()
+        if _ATTR_INTS in d:
+            vals = []
+            for v in d[_ATTR_INTS]:
+                if isinstance(v, bytes):  # packed
+                    vals.extend(w.decode_packed_varints(v))
+                else:
+                    vals.append(w.varint_to_sint64(v))
+            out[name] = vals
+        elif _ATTR_I in d:
+            out[name] = w.varint_to_sint64(d[_ATTR_I][0])
+        elif _ATTR_F in d:
+            out[name] = struct.unpack("<f", struct.pack("<i", d[_ATTR_F][0]))[0]
+    return out
+
+
+def _parse_vi_shape(buf: bytes) -> tuple[str, List[int]]:
+    d = w.fields_dict(buf)
+    name = d[_VI_NAME][0].decode()
+    dims: List[int] = []
+    if _VI_TYPE in d:
+        tp = w.fields_dict(d[_VI_TYPE][0])
+        if _TP_TENSOR in tp:
+            tt = w.fields_dict(tp[_TP_TENSOR][0])
+            if _TT_SHAPE in tt:
+                for dim_buf in w.fields_dict(tt[_TT_SHAPE][0]).get(_TS_DIM, []):
+                    dd = w.fields_dict(dim_buf)
+                    dims.append(w.varint_to_sint64(dd.get(_TD_VALUE, [0])[0]))
+    return name, dims
+
+
+def import_onnx(data: bytes, batch: Optional[int] = None,
+                name: str = "onnx_model") -> Graph:
+    model = w.fields_dict(data)
+    graph_buf = model[_MODEL_GRAPH][0]
+    gd = w.fields_dict(graph_buf)
+
+    inits: Dict[str, np.ndarray] = {}
+    for t in gd.get(_GRAPH_INIT, []):
+        nm, arr = _parse_tensor(t)
+        inits[nm] = arr
+
+    g = Graph(name)
+    # graph input (the one without an initializer)
+    input_name = None
+    for vi in gd.get(_GRAPH_INPUT, []):
+        nm, dims = _parse_vi_shape(vi)
+        if nm in inits:
+            continue
+        n, c, h, ww = dims  # NCHW
+        if batch:
+            n = batch
+        input_name = g.input((n, h, ww, c), name=nm)  # IR is NHWC
+    assert input_name is not None, "no graph input found"
+
+    # name remapping: ONNX tensor name -> IR tensor name
+    remap: Dict[str, str] = {input_name: input_name}
+
+    for nbuf in gd.get(_GRAPH_NODE, []):
+        nd = w.fields_dict(nbuf)
+        op = nd[_NODE_OPTYPE][0].decode()
+        ins = [b.decode() for b in nd.get(_NODE_INPUT, [])]
+        outs = [b.decode() for b in nd.get(_NODE_OUTPUT, [])]
+        attrs = _parse_attrs(nd.get(_NODE_ATTR, []))
+        x = remap.get(ins[0], ins[0]) if ins else None
+
+        if op == "Conv":
+            wt = inits[ins[1]].astype(np.float32)
+            strides = attrs.get("strides", [1, 1])
+            pads = attrs.get("pads", [0, 0, 0, 0])
+            out = g.conv2d(x, wt, stride=int(strides[0]),
+                           padding=int(pads[0]))
+            if len(ins) > 2:  # conv bias -> fold into a batchnorm-less bias
+                b = inits[ins[2]].astype(np.float32)
+                cout = wt.shape[0]
+                out = g.batchnorm(out, gamma=np.ones(cout, np.float32),
+                                  beta=b, mean=np.zeros(cout, np.float32),
+                                  var=np.ones(cout, np.float32) - 1e-5)
+        elif op == "BatchNormalization":
+            gamma, beta, mean, var = (inits[ins[k]].astype(np.float32)
+                                      for k in (1, 2, 3, 4))
+            out = g.batchnorm(x, gamma=gamma, beta=beta, mean=mean, var=var,
+                              eps=float(attrs.get("epsilon", 1e-5)))
+        elif op == "Relu":
+            out = g.relu(x)
+        elif op == "Gelu":
+            out = g.gelu(x)
+        elif op == "Add":
+            out = g.add(x, remap.get(ins[1], ins[1]))
+        elif op == "MaxPool":
+            ks = attrs.get("kernel_shape", [2, 2])
+            strides = attrs.get("strides", [1, 1])
+            pads = attrs.get("pads", [0, 0, 0, 0])
+            out = g.maxpool(x, kernel=int(ks[0]), stride=int(strides[0]),
+                            padding=int(pads[0]))
+        elif op == "GlobalAveragePool":
+            out = g.global_avgpool(x)
+        elif op == "Flatten":
+            out = x  # GAP output is already [N, C] in the IR
+        elif op in ("Gemm", "MatMul"):
+            wt = inits[ins[1]].astype(np.float32)
+            if op == "MatMul" or not attrs.get("transB", 0):
+                wt = wt.T  # IR gemm takes [out, in]
+            bias = inits[ins[2]].astype(np.float32) if len(ins) > 2 else None
+            out = g.gemm(x, np.ascontiguousarray(wt), bias)
+        elif op == "Softmax":
+            out = g.softmax(x)
+        elif op == "LayerNormalization":
+            out = g.layernorm(x, inits[ins[1]].astype(np.float32),
+                              inits[ins[2]].astype(np.float32),
+                              eps=float(attrs.get("epsilon", 1e-5)))
+        else:
+            raise ValueError(f"ONNX op {op} not supported by the importer")
+        remap[outs[0]] = out
+    return g
+
+
+def load_onnx(path: str, batch: Optional[int] = None) -> Graph:
+    with open(path, "rb") as f:
+        return import_onnx(f.read(), batch=batch)
+
+
+# ------------------------------------------------------------------ export
+def _tensor_bytes(name: str, arr: np.ndarray) -> bytes:
+    out = b"".join(w.f_varint(_T_DIMS, d) for d in arr.shape)
+    out += w.f_varint(_T_DTYPE, _F32)
+    out += w.f_string(_T_NAME, name)
+    out += w.f_bytes(_T_RAW, np.ascontiguousarray(arr, np.float32).tobytes())
+    return out
+
+
+def _attr_ints(name: str, vals: List[int]) -> bytes:
+    body = w.f_string(_ATTR_NAME, name)
+    for v in vals:
+        body += w.f_varint(_ATTR_INTS, v)
+    body += w.f_varint(20, 7)  # AttributeProto.type = INTS
+    return w.f_bytes(_NODE_ATTR, body)
+
+
+def _attr_f(name: str, v: float) -> bytes:
+    body = w.f_string(_ATTR_NAME, name) + w.f_float(_ATTR_F, v)
+    body += w.f_varint(20, 1)  # FLOAT
+    return w.f_bytes(_NODE_ATTR, body)
+
+
+def _attr_i(name: str, v: int) -> bytes:
+    body = w.f_string(_ATTR_NAME, name) + w.f_varint(_ATTR_I, v)
+    body += w.f_varint(20, 2)  # INT
+    return w.f_bytes(_NODE_ATTR, body)
+
+
+def _node(op: str, ins: List[str], outs: List[str], *attrs: bytes) -> bytes:
+    body = b"".join(w.f_string(_NODE_INPUT, i) for i in ins)
+    body += b"".join(w.f_string(_NODE_OUTPUT, o) for o in outs)
+    body += w.f_string(_NODE_OPTYPE, op)
+    body += b"".join(attrs)
+    return w.f_bytes(_GRAPH_NODE, body)
+
+
+def _value_info(name: str, dims: List[int]) -> bytes:
+    dim_bufs = b"".join(
+        w.f_bytes(_TS_DIM, w.f_varint(_TD_VALUE, d)) for d in dims)
+    shape = w.f_bytes(_TT_SHAPE, dim_bufs)
+    tt = w.f_varint(_TT_ELEM, _F32) + shape
+    tp = w.f_bytes(_TP_TENSOR, tt)
+    return w.f_string(_VI_NAME, name) + w.f_bytes(_VI_TYPE, tp)
+
+
+def export_onnx(g: Graph) -> bytes:
+    """Serialize the (pre-fusion) IR graph to ONNX bytes."""
+    nodes = b""
+    inits = b""
+    init_ct = 0
+
+    def add_init(arr: np.ndarray) -> str:
+        nonlocal inits, init_ct
+        nm = f"w{init_ct}"
+        init_ct += 1
+        inits += w.f_bytes(_GRAPH_INIT, _tensor_bytes(nm, arr))
+        return nm
+
+    for n in g.nodes:
+        if n.kind == "input":
+            continue
+        if n.kind == "conv2d":
+            wn = add_init(n.attrs["weight"])
+            p = n.attrs["padding"]
+            nodes += _node("Conv", [n.inputs[0], wn], [n.output],
+                           _attr_ints("strides", [n.attrs["stride"]] * 2),
+                           _attr_ints("pads", [p] * 4),
+                           _attr_ints("kernel_shape",
+                                      list(n.attrs["weight"].shape[2:])))
+        elif n.kind == "batchnorm":
+            a = n.attrs
+            names = [add_init(a[k]) for k in ("gamma", "beta", "mean", "var")]
+            nodes += _node("BatchNormalization", [n.inputs[0]] + names,
+                           [n.output], _attr_f("epsilon", a["eps"]))
+        elif n.kind == "relu":
+            nodes += _node("Relu", [n.inputs[0]], [n.output])
+        elif n.kind == "gelu":
+            nodes += _node("Gelu", [n.inputs[0]], [n.output])
+        elif n.kind == "add":
+            nodes += _node("Add", list(n.inputs), [n.output])
+        elif n.kind == "maxpool":
+            a = n.attrs
+            nodes += _node("MaxPool", [n.inputs[0]], [n.output],
+                           _attr_ints("kernel_shape", [a["kernel"]] * 2),
+                           _attr_ints("strides", [a["stride"]] * 2),
+                           _attr_ints("pads", [a["padding"]] * 4))
+        elif n.kind == "gavgpool":
+            nodes += _node("GlobalAveragePool", [n.inputs[0]], [n.output])
+        elif n.kind == "gemm":
+            wn = add_init(n.attrs["weight"])
+            ins = [n.inputs[0], wn]
+            if n.attrs.get("bias") is not None:
+                ins.append(add_init(n.attrs["bias"]))
+            nodes += _node("Gemm", ins, [n.output], _attr_i("transB", 1))
+        elif n.kind == "softmax":
+            nodes += _node("Softmax", [n.inputs[0]], [n.output])
+        elif n.kind == "layernorm":
+            a = n.attrs
+            nodes += _node("LayerNormalization",
+                           [n.inputs[0], add_init(a["gamma"]),
+                            add_init(a["beta"])], [n.output],
+                           _attr_f("epsilon", a["eps"]))
+        else:
+            raise ValueError(f"export: unsupported node kind {n.kind}")
+
+    nb, h, ww, c = (list(g.tensors[g.input_name].shape) + [0, 0, 0, 0])[:4]
+    in_vi = w.f_bytes(_GRAPH_INPUT, _value_info(g.input_name, [nb, c, h, ww]))
+    out_vi = w.f_bytes(_GRAPH_OUTPUT, _value_info(g.output_name, []))
+    graph = nodes + w.f_string(_GRAPH_NAME, g.name) + inits + in_vi + out_vi
+    model = w.f_varint(1, 8)  # ir_version
+    model += w.f_bytes(_MODEL_GRAPH, graph)
+    return model
