@@ -45,6 +45,8 @@ OpDesc op_from_dict(const py::dict& d) {
   o.n_elems = gl("n_elems", 0);
   o.B = gi("B", 0); o.S = gi("S", 0); o.NH = gi("NH", 0); o.HD = gi("HD", 0);
   o.att_scale = gf("att_scale", 1.0f);
+  o.res_scale = gf("res_scale", 1.0f);
+  o.q_scale = gf("q_scale", 1.0f);
   return o;
 }
 
@@ -147,10 +149,10 @@ PYBIND11_MODULE(_C, m) {
   ops.def("gemm_bt",
           [](int dtype, uintptr_t A, uintptr_t B, uintptr_t C, uintptr_t scale,
              uintptr_t bias, uintptr_t residual, int M, int N, int K, int epi,
-             uintptr_t stream, bool sync, int tile) {
+             uintptr_t stream, bool sync, int tile, float res_scale) {
             launch_gemm_bt(dtype, (void*)A, (void*)B, (void*)C, (float*)scale,
-                           (float*)bias, (void*)residual, M, N, K, K, K, N,
-                           epi, as_stream(stream), tile,
+                           (float*)bias, (void*)residual, res_scale, M, N, K,
+                           K, K, N, epi, as_stream(stream), tile,
                            test_scratch(gemm_scratch_bytes(M, N, K)));
             if (sync) TRT_HIP_CHECK(hipStreamSynchronize(as_stream(stream)));
           },
@@ -158,19 +160,20 @@ PYBIND11_MODULE(_C, m) {
           py::arg("scale") = 0, py::arg("bias") = 0, py::arg("residual") = 0,
           py::arg("M") = 0, py::arg("N") = 0, py::arg("K") = 0,
           py::arg("epi") = 0, py::arg("stream") = 0, py::arg("sync") = true,
-          py::arg("tile") = 0);
+          py::arg("tile") = 0, py::arg("res_scale") = 1.0f);
   ops.def("conv2d",
           [](int dtype, uintptr_t in, uintptr_t Wt, uintptr_t out,
              uintptr_t scale, uintptr_t bias, uintptr_t residual,
              uintptr_t zero_page, int Nb, int H, int W, int C, int Cout,
              int KH, int KW, int sh, int sw, int ph, int pw, int epi,
-             uintptr_t stream, bool sync, int tile) {
+             uintptr_t stream, bool sync, int tile, float res_scale) {
             launch_conv2d(dtype, (void*)in, (void*)Wt, (void*)out,
                           (float*)scale, (float*)bias, (void*)residual,
                           (void*)zero_page, Nb, H, W, C, Cout, KH, KW, sh, sw,
                           ph, pw, epi, as_stream(stream), tile,
                           test_scratch(conv_scratch_bytes(
-                              Nb, H, W, C, Cout, KH, KW, sh, sw, ph, pw)));
+                              Nb, H, W, C, Cout, KH, KW, sh, sw, ph, pw)),
+                          res_scale);
             if (sync) TRT_HIP_CHECK(hipStreamSynchronize(as_stream(stream)));
           },
           py::arg("dtype"), py::arg("in"), py::arg("Wt"), py::arg("out"),
@@ -180,7 +183,7 @@ PYBIND11_MODULE(_C, m) {
           py::arg("KH") = 1, py::arg("KW") = 1, py::arg("sh") = 1,
           py::arg("sw") = 1, py::arg("ph") = 0, py::arg("pw") = 0,
           py::arg("epi") = 0, py::arg("stream") = 0, py::arg("sync") = true,
-          py::arg("tile") = 0);
+          py::arg("tile") = 0, py::arg("res_scale") = 1.0f);
   ops.def("maxpool2d",
           [](int dtype, uintptr_t in, uintptr_t out, int Nb, int H, int W,
              int C, int KH, int KW, int sh, int sw, int ph, int pw,
@@ -256,6 +259,22 @@ PYBIND11_MODULE(_C, m) {
           py::arg("dtype"), py::arg("in"), py::arg("out"), py::arg("M"),
           py::arg("Cin"), py::arg("Cpad"), py::arg("stream") = 0,
           py::arg("sync") = true);
+  ops.def("quantize",
+          [](uintptr_t in, uintptr_t out, int64_t n, float scale,
+             uintptr_t stream, bool sync) {
+            launch_quantize((void*)in, (void*)out, n, scale, as_stream(stream));
+            if (sync) TRT_HIP_CHECK(hipStreamSynchronize(as_stream(stream)));
+          },
+          py::arg("in"), py::arg("out"), py::arg("n"), py::arg("scale"),
+          py::arg("stream") = 0, py::arg("sync") = true);
+  ops.def("dequant",
+          [](uintptr_t in, uintptr_t out, int64_t n, float scale,
+             uintptr_t stream, bool sync) {
+            launch_dequant((void*)in, (void*)out, n, scale, as_stream(stream));
+            if (sync) TRT_HIP_CHECK(hipStreamSynchronize(as_stream(stream)));
+          },
+          py::arg("in"), py::arg("out"), py::arg("n"), py::arg("scale"),
+          py::arg("stream") = 0, py::arg("sync") = true);
   ops.def("attention",
           [](int dtype, uintptr_t qkv, uintptr_t out, int B, int S, int H,
              int D, float scale, uintptr_t stream, bool sync) {

@@ -26,8 +26,9 @@ struct ConvParams {
   int OH, OW;             // output spatial
   int sh, sw, ph, pw;     // stride / padding
   int M;                  // Nb*OH*OW
-  int K;                  // KH*KW*C rounded up to %64
+  int K;                  // KH*KW*C rounded up to the K-tile (64/128 elems)
   int Kreal;              // KH*KW*C
+  float res_scale;        // int8 residual dequant ratio (s_res/s_out)
   FastDiv d_ohw, d_ow, d_c, d_kw;
 };
 
@@ -47,7 +48,7 @@ __device__ __forceinline__ void stage_conv_a(
     uint32_t rem = (uint32_t)m - n * (uint32_t)(p.OH * p.OW);
     uint32_t oh = fdiv(rem, p.d_ow);
     uint32_t ow = rem - oh * (uint32_t)p.OW;
-    int k = k0 + (int)(kb >> 1);  // element index along K (2-byte elems)
+    int k = k0 + (int)(kb / sizeof(T));  // element index along K
     const char* src;
     if (k >= p.Kreal) {
       src = (const char*)zero_page;
@@ -71,9 +72,9 @@ __device__ __forceinline__ void stage_conv_a(
 
 void launch_splitk_reduce(int dtype, const float* scratch, void* C,
                           const float* scale, const float* bias,
-                          const void* residual, int M, int N, int64_t ldc,
-                          int tiles_m, int tiles_n, int splitk, int bm,
-                          int bn, int epi, hipStream_t stream);
+                          const void* residual, float res_scale, int M, int N,
+                          int64_t ldc, int tiles_m, int tiles_n, int splitk,
+                          int bm, int bn, int epi, hipStream_t stream);
 
 template <typename T, Epi E, int BM, int BN, bool SPLIT>
 __global__ __launch_bounds__(256) void conv_igemm_kernel(
@@ -99,13 +100,14 @@ __global__ __launch_bounds__(256) void conv_igemm_kernel(
   const int wr = wave >> 1;
   const int wc = wave & 1;
 
-  f32x4 acc[BM / 32][BN / 32];
+  typename Mfma16x16x32<T>::accv acc[BM / 32][BN / 32];
 #pragma unroll
   for (int i = 0; i < BM / 32; ++i)
 #pragma unroll
-    for (int j = 0; j < BN / 32; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+    for (int j = 0; j < BN / 32; ++j) acc[i][j] = {0, 0, 0, 0};
 
-  const int ktiles = p.K >> 6;
+  constexpr int KT = kTileElems<T>;
+  const int ktiles = p.K / KT;
   int kt0 = 0, kt1 = ktiles;
   if constexpr (SPLIT) {
     int slice = bid % splitk;
@@ -113,8 +115,8 @@ __global__ __launch_bounds__(256) void conv_igemm_kernel(
     kt1 = min(ktiles, kt0 + ktper);
   }
 
-  stage_conv_a<T, BM>(in, zero_page, p, m0, kt0 * 64, lds0, tid);
-  stage_tile<T, BN>(Wt + (int64_t)n0 * p.K + kt0 * 64, p.K, n0, p.Cout,
+  stage_conv_a<T, BM>(in, zero_page, p, m0, kt0 * KT, lds0, tid);
+  stage_tile<T, BN>(Wt + (int64_t)n0 * p.K + kt0 * KT, p.K, n0, p.Cout,
                     lds0 + kABytes, tid);
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   __syncthreads();
@@ -123,8 +125,8 @@ __global__ __launch_bounds__(256) void conv_igemm_kernel(
   for (int t = kt0; t < kt1; ++t) {
     if (t + 1 < kt1) {
       uint32_t nb = lds0 + (cur ^ 1) * kBuf;
-      stage_conv_a<T, BM>(in, zero_page, p, m0, (t + 1) * 64, nb, tid);
-      stage_tile<T, BN>(Wt + (int64_t)n0 * p.K + (t + 1) * 64, p.K, n0,
+      stage_conv_a<T, BM>(in, zero_page, p, m0, (t + 1) * KT, nb, tid);
+      stage_tile<T, BN>(Wt + (int64_t)n0 * p.K + (t + 1) * KT, p.K, n0,
                         p.Cout, nb + kABytes, tid);
     }
     const char* As = &smem[cur * kBuf];
@@ -139,7 +141,7 @@ __global__ __launch_bounds__(256) void conv_igemm_kernel(
                             wc);
   } else {
     store_epilogue<T, E, BM, BN>(acc, out, p.Cout, m0, n0, p.M, p.Cout, scale,
-                                 bias, residual, lane, wr, wc);
+                                 bias, residual, p.res_scale, lane, wr, wc);
   }
 }
 
@@ -166,10 +168,11 @@ static void launch_conv2d_t(const void* in, const void* Wt, void* out,
   int tiles_m = (int)cdiv(p.M, cfg.bm);
   int tiles_n = (int)cdiv(p.Cout, cfg.bn);
   long tiles = (long)tiles_m * tiles_n;
-  int ktiles = p.K >> 6;
+  int ktiles = p.K / kTileElems<T>;
   int splitk = (!tile && scratch) ? pick_splitk(tiles, ktiles) : 1;
   dim3 block(256);
-  int dtype = std::is_same<T, _Float16>::value ? 0 : 1;
+  int dtype = std::is_same<T, _Float16>::value ? 0
+              : (std::is_same<T, __bf16>::value ? 1 : 2);
   if (splitk > 1) {
     int ktper = (int)cdiv(ktiles, splitk);
     dim3 grid((unsigned)(tiles * splitk));
@@ -182,9 +185,9 @@ static void launch_conv2d_t(const void* in, const void* Wt, void* out,
                          (const T*)zero_page, p, tiles_n, scratch, splitk,
                          ktper);
     });
-    launch_splitk_reduce(dtype, scratch, out, scale, bias, residual, p.M,
-                         p.Cout, p.Cout, tiles_m, tiles_n, splitk, cfg.bm,
-                         cfg.bn, epi, stream);
+    launch_splitk_reduce(dtype, scratch, out, scale, bias, residual,
+                         p.res_scale, p.M, p.Cout, p.Cout, tiles_m, tiles_n,
+                         splitk, cfg.bm, cfg.bn, epi, stream);
     return;
   }
   dim3 grid((unsigned)tiles);
@@ -206,8 +209,10 @@ void launch_conv2d(int dtype, const void* in, const void* Wt, void* out,
                    const float* scale, const float* bias, const void* residual,
                    const void* zero_page, int Nb, int H, int W, int C,
                    int Cout, int KH, int KW, int sh, int sw, int ph, int pw,
-                   int epi, hipStream_t stream, int tile, void* scratch) {
+                   int epi, hipStream_t stream, int tile, void* scratch,
+                   float res_scale) {
   ConvParams p;
+  p.res_scale = res_scale;
   p.Nb = Nb; p.H = H; p.W = W; p.C = C;
   p.Cout = Cout; p.KH = KH; p.KW = KW;
   p.sh = sh; p.sw = sw; p.ph = ph; p.pw = pw;
@@ -215,8 +220,12 @@ void launch_conv2d(int dtype, const void* in, const void* Wt, void* out,
   p.OW = (W + 2 * pw - KW) / sw + 1;
   p.M = Nb * p.OH * p.OW;
   p.Kreal = KH * KW * C;
-  p.K = (int)round_up(p.Kreal, 64);
-  if (C % 8 != 0) throw std::runtime_error("conv2d: C must be a multiple of 8 (pad input channels)");
+  int kt = dtype == 2 ? 128 : 64;
+  p.K = (int)round_up(p.Kreal, kt);
+  int cmin = dtype == 2 ? 16 : 8;  // one 16-B glds chunk per pixel minimum
+  if (C % cmin != 0)
+    throw std::runtime_error("conv2d: C must be a multiple of 16 B / elem "
+                             "size (pad input channels)");
   p.d_ohw = make_fastdiv((uint32_t)(p.OH * p.OW));
   p.d_ow = make_fastdiv((uint32_t)p.OW);
   p.d_c = make_fastdiv((uint32_t)C);
@@ -224,8 +233,11 @@ void launch_conv2d(int dtype, const void* in, const void* Wt, void* out,
   if (dtype == 0)
     launch_conv2d_t<_Float16>(in, Wt, out, scale, bias, residual, zero_page, p,
                               epi, stream, tile, (float*)scratch);
-  else
+  else if (dtype == 1)
     launch_conv2d_t<__bf16>(in, Wt, out, scale, bias, residual, zero_page, p,
+                            epi, stream, tile, (float*)scratch);
+  else
+    launch_conv2d_t<int8_t>(in, Wt, out, scale, bias, residual, zero_page, p,
                             epi, stream, tile, (float*)scratch);
 }
 

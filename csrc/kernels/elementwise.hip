@@ -106,10 +106,67 @@ void launch_channel_pad(int dtype, const void* in, void* out, int64_t M,
     hipLaunchKernelGGL((channel_pad_kernel<_Float16>), dim3(blocks), dim3(256),
                        0, stream, (const _Float16*)in, (_Float16*)out, M, Cin,
                        Cpad);
-  else
+  else if (dtype == 1)
     hipLaunchKernelGGL((channel_pad_kernel<__bf16>), dim3(blocks), dim3(256),
                        0, stream, (const __bf16*)in, (__bf16*)out, M, Cin,
                        Cpad);
+  else
+    hipLaunchKernelGGL((channel_pad_kernel<int8_t>), dim3(blocks), dim3(256),
+                       0, stream, (const int8_t*)in, (int8_t*)out, M, Cin,
+                       Cpad);
+}
+
+// fp16 -> int8 symmetric quantization (and inverse), vectorized 8-wide.
+__global__ void quantize_kernel(const _Float16* __restrict__ in,
+                                int8_t* __restrict__ out, int64_t n8,
+                                float inv_scale) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n8;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    short4v v0 = *(const short4v*)(in + i * 8);
+    short4v v1 = *(const short4v*)(in + i * 8 + 4);
+    int8_t r[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float x = (float)((const _Float16*)(j < 4 ? (const void*)&v0
+                                                : (const void*)&v1))[j & 3];
+      float q = rintf(x * inv_scale);
+      q = fminf(fmaxf(q, -127.f), 127.f);
+      r[j] = (int8_t)q;
+    }
+    *(uint2*)(out + i * 8) = *(const uint2*)r;
+  }
+}
+
+__global__ void dequant_kernel(const int8_t* __restrict__ in,
+                               _Float16* __restrict__ out, int64_t n8,
+                               float scale) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n8;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int8_t r[8];
+    *(uint2*)r = *(const uint2*)(in + i * 8);
+    _Float16 o[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) o[j] = (_Float16)((float)r[j] * scale);
+    *(short4v*)(out + i * 8) = *(const short4v*)&o[0];
+    *(short4v*)(out + i * 8 + 4) = *(const short4v*)&o[4];
+  }
+}
+
+void launch_quantize(const void* in_f16, void* out_i8, int64_t n, float scale,
+                     hipStream_t stream) {
+  if (n % 8 != 0) throw std::runtime_error("quantize: n % 8 != 0");
+  int blocks = ew_blocks(n / 8);
+  hipLaunchKernelGGL(quantize_kernel, dim3(blocks), dim3(256), 0, stream,
+                     (const _Float16*)in_f16, (int8_t*)out_i8, n / 8,
+                     1.0f / scale);
+}
+
+void launch_dequant(const void* in_i8, void* out_f16, int64_t n, float scale,
+                    hipStream_t stream) {
+  if (n % 8 != 0) throw std::runtime_error("dequant: n % 8 != 0");
+  int blocks = ew_blocks(n / 8);
+  hipLaunchKernelGGL(dequant_kernel, dim3(blocks), dim3(256), 0, stream,
+                     (const int8_t*)in_i8, (_Float16*)out_f16, n / 8, scale);
 }
 
 void launch_cast(int dtype, bool to_f32, const void* in, void* out, int64_t n,

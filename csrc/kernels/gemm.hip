@@ -26,17 +26,17 @@ namespace trtlab {
 
 void launch_splitk_reduce(int dtype, const float* scratch, void* C,
                           const float* scale, const float* bias,
-                          const void* residual, int M, int N, int64_t ldc,
-                          int tiles_m, int tiles_n, int splitk, int bm,
-                          int bn, int epi, hipStream_t stream);
+                          const void* residual, float res_scale, int M, int N,
+                          int64_t ldc, int tiles_m, int tiles_n, int splitk,
+                          int bm, int bn, int epi, hipStream_t stream);
 
 template <typename T, Epi E, int BM, int BN, bool SPLIT>
 __global__ __launch_bounds__(256) void gemm_bt_kernel(
     const T* __restrict__ A, const T* __restrict__ B, T* __restrict__ C,
     const float* __restrict__ scale, const float* __restrict__ bias,
-    const T* __restrict__ residual, int M, int N, int K, int64_t lda,
-    int64_t ldb, int64_t ldc, int tiles_n, float* __restrict__ scratch,
-    int splitk, int ktper) {
+    const T* __restrict__ residual, float res_scale, int M, int N, int K,
+    int64_t lda, int64_t ldb, int64_t ldc, int tiles_n,
+    float* __restrict__ scratch, int splitk, int ktper) {
   constexpr int kABytes = BM * 128;
   constexpr int kBuf = (BM + BN) * 128;
 
@@ -45,7 +45,8 @@ __global__ __launch_bounds__(256) void gemm_bt_kernel(
   int m0 = (int)(tile / tiles_n) * BM;
   int n0 = (int)(tile % tiles_n) * BN;
 
-  const int ktiles = K >> 6;  // K % 64 == 0 (host asserts)
+  constexpr int KT = kTileElems<T>;  // 64 (fp16/bf16) or 128 (int8)
+  const int ktiles = K / KT;
   int kt0 = 0, kt1 = ktiles;
   if constexpr (SPLIT) {
     int slice = bid % splitk;
@@ -62,14 +63,14 @@ __global__ __launch_bounds__(256) void gemm_bt_kernel(
   const int wr = wave >> 1;
   const int wc = wave & 1;
 
-  f32x4 acc[BM / 32][BN / 32];
+  typename Mfma16x16x32<T>::accv acc[BM / 32][BN / 32];
 #pragma unroll
   for (int i = 0; i < BM / 32; ++i)
 #pragma unroll
-    for (int j = 0; j < BN / 32; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+    for (int j = 0; j < BN / 32; ++j) acc[i][j] = {0, 0, 0, 0};
 
-  stage_tile<T, BM>(A + (int64_t)m0 * lda + kt0 * 64, lda, m0, M, lds0, tid);
-  stage_tile<T, BN>(B + (int64_t)n0 * ldb + kt0 * 64, ldb, n0, N,
+  stage_tile<T, BM>(A + (int64_t)m0 * lda + kt0 * KT, lda, m0, M, lds0, tid);
+  stage_tile<T, BN>(B + (int64_t)n0 * ldb + kt0 * KT, ldb, n0, N,
                     lds0 + kABytes, tid);
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   __syncthreads();
@@ -78,9 +79,9 @@ __global__ __launch_bounds__(256) void gemm_bt_kernel(
   for (int t = kt0; t < kt1; ++t) {
     if (t + 1 < kt1) {
       uint32_t nb = lds0 + (cur ^ 1) * kBuf;
-      stage_tile<T, BM>(A + (int64_t)m0 * lda + (t + 1) * 64, lda, m0, M, nb,
+      stage_tile<T, BM>(A + (int64_t)m0 * lda + (t + 1) * KT, lda, m0, M, nb,
                         tid);
-      stage_tile<T, BN>(B + (int64_t)n0 * ldb + (t + 1) * 64, ldb, n0, N,
+      stage_tile<T, BN>(B + (int64_t)n0 * ldb + (t + 1) * KT, ldb, n0, N,
                         nb + kABytes, tid);
     }
     const char* As = &smem[cur * kBuf];
@@ -95,14 +96,14 @@ __global__ __launch_bounds__(256) void gemm_bt_kernel(
                             wc);
   } else {
     store_epilogue<T, E, BM, BN>(acc, C, ldc, m0, n0, M, N, scale, bias,
-                                 residual, lane, wr, wc);
+                                 residual, res_scale, lane, wr, wc);
   }
 }
 
 size_t gemm_scratch_bytes(int M, int N, int K) {
   TileCfg cfg = pick_tile(M, N);
   long tiles = cdiv(M, cfg.bm) * cdiv(N, cfg.bn);
-  int splitk = pick_splitk(tiles, K >> 6);
+  int splitk = pick_splitk(tiles, K >> 6);  // fp16 tile count (conservative)
   if (splitk == 1) return 0;
   return (size_t)tiles * splitk * cfg.bm * cfg.bn * 4;
 }
@@ -110,17 +111,19 @@ size_t gemm_scratch_bytes(int M, int N, int K) {
 template <typename T>
 static void launch_gemm_bt_t(const void* A, const void* B, void* C,
                              const float* scale, const float* bias,
-                             const void* residual, int M, int N, int K,
-                             int64_t lda, int64_t ldb, int64_t ldc, int epi,
-                             hipStream_t stream, int tile, float* scratch) {
+                             const void* residual, float res_scale, int M,
+                             int N, int K, int64_t lda, int64_t ldb,
+                             int64_t ldc, int epi, hipStream_t stream,
+                             int tile, float* scratch) {
   TileCfg cfg = tile ? tile_from_code(tile) : pick_tile(M, N);
   int tiles_m = (int)cdiv(M, cfg.bm);
   int tiles_n = (int)cdiv(N, cfg.bn);
   long tiles = (long)tiles_m * tiles_n;
-  int ktiles = K >> 6;
+  int ktiles = K / kTileElems<T>;
   int splitk = (!tile && scratch) ? pick_splitk(tiles, ktiles) : 1;
   dim3 block(256);
-  int dtype = std::is_same<T, _Float16>::value ? 0 : 1;
+  int dtype = std::is_same<T, _Float16>::value ? 0
+              : (std::is_same<T, __bf16>::value ? 1 : 2);
   if (splitk > 1) {
     int ktper = (int)cdiv(ktiles, splitk);
     dim3 grid((unsigned)(tiles * splitk));
@@ -129,12 +132,12 @@ static void launch_gemm_bt_t(const void* A, const void* B, void* C,
       constexpr int BN = decltype(bn)::value;
       hipLaunchKernelGGL((gemm_bt_kernel<T, Epi::kNone, BM, BN, true>), grid,
                          block, 0, stream, (const T*)A, (const T*)B, (T*)C,
-                         scale, bias, (const T*)residual, M, N, K, lda, ldb,
-                         ldc, tiles_n, scratch, splitk, ktper);
+                         scale, bias, (const T*)residual, res_scale, M, N, K,
+                         lda, ldb, ldc, tiles_n, scratch, splitk, ktper);
     });
-    launch_splitk_reduce(dtype, scratch, C, scale, bias, residual, M, N, ldc,
-                         tiles_m, tiles_n, splitk, cfg.bm, cfg.bn, epi,
-                         stream);
+    launch_splitk_reduce(dtype, scratch, C, scale, bias, residual, res_scale,
+                         M, N, ldc, tiles_m, tiles_n, splitk, cfg.bm, cfg.bn,
+                         epi, stream);
     return;
   }
   dim3 grid((unsigned)tiles);
@@ -145,24 +148,34 @@ static void launch_gemm_bt_t(const void* A, const void* B, void* C,
       constexpr int BN = decltype(bn)::value;
       hipLaunchKernelGGL((gemm_bt_kernel<T, EE, BM, BN, false>), grid, block,
                          0, stream, (const T*)A, (const T*)B, (T*)C, scale,
-                         bias, (const T*)residual, M, N, K, lda, ldb, ldc,
-                         tiles_n, (float*)nullptr, 1, ktiles);
+                         bias, (const T*)residual, res_scale, M, N, K, lda,
+                         ldb, ldc, tiles_n, (float*)nullptr, 1, ktiles);
     });
   });
 }
 
-void launch_gemm_bt(int dtype,  // 0 = fp16, 1 = bf16
+void launch_gemm_bt(int dtype,  // 0 = fp16, 1 = bf16, 2 = int8
                     const void* A, const void* B, void* C, const float* scale,
-                    const float* bias, const void* residual, int M, int N,
-                    int K, int64_t lda, int64_t ldb, int64_t ldc, int epi,
-                    hipStream_t stream, int tile, void* scratch) {
+                    const float* bias, const void* residual, float res_scale,
+                    int M, int N, int K, int64_t lda, int64_t ldb, int64_t ldc,
+                    int epi, hipStream_t stream, int tile, void* scratch) {
+  if (dtype == 2) {
+    if (K % 128 != 0)
+      throw std::runtime_error("gemm_bt int8: K must be a multiple of 128");
+    launch_gemm_bt_t<int8_t>(A, B, C, scale, bias, residual, res_scale, M, N,
+                             K, lda, ldb, ldc, epi, stream, tile,
+                             (float*)scratch);
+    return;
+  }
   if (K % 64 != 0) throw std::runtime_error("gemm_bt: K must be a multiple of 64");
   if (dtype == 0)
-    launch_gemm_bt_t<_Float16>(A, B, C, scale, bias, residual, M, N, K, lda,
-                               ldb, ldc, epi, stream, tile, (float*)scratch);
+    launch_gemm_bt_t<_Float16>(A, B, C, scale, bias, residual, res_scale, M,
+                               N, K, lda, ldb, ldc, epi, stream, tile,
+                               (float*)scratch);
   else
-    launch_gemm_bt_t<__bf16>(A, B, C, scale, bias, residual, M, N, K, lda, ldb,
-                             ldc, epi, stream, tile, (float*)scratch);
+    launch_gemm_bt_t<__bf16>(A, B, C, scale, bias, residual, res_scale, M, N,
+                             K, lda, ldb, ldc, epi, stream, tile,
+                             (float*)scratch);
 }
 
 }  // namespace trtlab

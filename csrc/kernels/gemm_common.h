@@ -40,7 +40,13 @@ __device__ __forceinline__ float apply_epi(float acc, float scale, float bias,
 }
 
 // ------------------------------------------------------------- MFMA dispatch
+// All dtypes share the same 128-byte LDS tile row and 16-B fragment reads:
+// fp16/bf16 use mfma_f32_16x16x32 (K=32 elems/instr, 2 per row half),
+// int8 uses mfma_i32_16x16x64 (K=64 elems/instr, 2 per row half) — the
+// byte-level addressing is identical, only the element width differs.
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8v;
+typedef __attribute__((ext_vector_type(16))) char i8x16v;
+typedef __attribute__((ext_vector_type(4))) int i32x4;
 
 template <typename T>
 struct Mfma16x16x32;
@@ -48,6 +54,7 @@ struct Mfma16x16x32;
 template <>
 struct Mfma16x16x32<_Float16> {
   using frag = half8v;
+  using accv = f32x4;
   static __device__ __forceinline__ f32x4 run(frag a, frag b, f32x4 c) {
     return __builtin_amdgcn_mfma_f32_16x16x32_f16(a, b, c, 0, 0, 0);
   }
@@ -56,10 +63,36 @@ struct Mfma16x16x32<_Float16> {
 template <>
 struct Mfma16x16x32<__bf16> {
   using frag = bf16x8v;
+  using accv = f32x4;
   static __device__ __forceinline__ f32x4 run(frag a, frag b, f32x4 c) {
     return __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
   }
 };
+
+template <>
+struct Mfma16x16x32<int8_t> {
+  using frag = i8x16v;
+  using accv = i32x4;
+  static __device__ __forceinline__ i32x4 run(frag a, frag b, i32x4 c) {
+    return __builtin_amdgcn_mfma_i32_16x16x64_i8(a, b, c, 0, 0, 0);
+  }
+};
+
+// Elements per 128-byte K-tile row (the staged BK).
+template <typename T>
+constexpr int kTileElems = 128 / (int)sizeof(T);
+
+// Output store conversion: plain cast for fp16/bf16; round+clamp for int8.
+template <typename T>
+__device__ __forceinline__ T store_cast(float v) {
+  return (T)v;
+}
+template <>
+__device__ __forceinline__ int8_t store_cast<int8_t>(float v) {
+  float r = rintf(v);
+  r = fminf(fmaxf(r, -127.f), 127.f);
+  return (int8_t)r;
+}
 
 // ------------------------------------------------------------------ staging
 __device__ __forceinline__ void glds16(const void* gsrc, uint32_t lds_byte) {
@@ -102,9 +135,9 @@ __device__ __forceinline__ typename Mfma16x16x32<T>::frag read_frag(
 // Shared MFMA core for gemm/conv: 4 waves as 2x2, wave tile (BM/2)x(BN/2),
 // fragments of 16x16x32, BK=64 (two MFMA k-steps per staged tile).
 template <typename T, int BM, int BN>
-__device__ __forceinline__ void mfma_tile(const char* As, const char* Bs,
-                                          int lane, int wr, int wc,
-                                          f32x4 (&acc)[BM / 32][BN / 32]) {
+__device__ __forceinline__ void mfma_tile(
+    const char* As, const char* Bs, int lane, int wr, int wc,
+    typename Mfma16x16x32<T>::accv (&acc)[BM / 32][BN / 32]) {
   using MF = Mfma16x16x32<T>;
   constexpr int MFr = BM / 32, NFr = BN / 32;
 #pragma unroll
@@ -129,10 +162,11 @@ __device__ __forceinline__ void mfma_tile(const char* As, const char* Bs,
 // col = lane&15, row = (lane>>4)*4 + r.
 template <typename T, Epi E, int BM, int BN>
 __device__ __forceinline__ void store_epilogue(
-    f32x4 (&acc)[BM / 32][BN / 32], T* __restrict__ C, int64_t ldc, int m0,
-    int n0, int M, int N, const float* __restrict__ scale,
-    const float* __restrict__ bias, const T* __restrict__ residual, int lane,
-    int wr, int wc) {
+    typename Mfma16x16x32<T>::accv (&acc)[BM / 32][BN / 32],
+    T* __restrict__ C, int64_t ldc, int m0, int n0, int M, int N,
+    const float* __restrict__ scale, const float* __restrict__ bias,
+    const T* __restrict__ residual, float res_scale, int lane, int wr,
+    int wc) {
   constexpr int MFr = BM / 32, NFr = BN / 32;
 #pragma unroll
   for (int i = 0; i < MFr; ++i) {
@@ -151,9 +185,9 @@ __device__ __forceinline__ void store_epilogue(
         if (row >= M) continue;
         float res = 0.0f;
         if constexpr (E == Epi::kScaleBiasAddRelu)
-          res = (float)residual[(int64_t)row * ldc + col];
-        float v = apply_epi<E>(acc[i][j][r], sc, bi, res);
-        C[(int64_t)row * ldc + col] = (T)v;
+          res = (float)residual[(int64_t)row * ldc + col] * res_scale;
+        float v = apply_epi<E>((float)acc[i][j][r], sc, bi, res);
+        C[(int64_t)row * ldc + col] = store_cast<T>(v);
       }
     }
   }
@@ -161,9 +195,9 @@ __device__ __forceinline__ void store_epilogue(
 
 // Split-K slab store: this (tile, slice)'s [BM][BN] fp32 partial sums.
 template <typename T, int BM, int BN>
-__device__ __forceinline__ void store_splitk(f32x4 (&acc)[BM / 32][BN / 32],
-                                             float* __restrict__ slab,
-                                             int lane, int wr, int wc) {
+__device__ __forceinline__ void store_splitk(
+    typename Mfma16x16x32<T>::accv (&acc)[BM / 32][BN / 32],
+    float* __restrict__ slab, int lane, int wr, int wc) {
 #pragma unroll
   for (int i = 0; i < BM / 32; ++i)
 #pragma unroll
@@ -172,7 +206,7 @@ __device__ __forceinline__ void store_splitk(f32x4 (&acc)[BM / 32][BN / 32],
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         int row = wr * (BM / 2) + i * 16 + ((lane >> 4) << 2) + r;
-        slab[row * BN + col] = acc[i][j][r];
+        slab[row * BN + col] = (float)acc[i][j][r];
       }
     }
 }

@@ -5,12 +5,12 @@
 
 namespace trtlab {
 
-// dtype: 0 = fp16, 1 = bf16
+// dtype: 0 = fp16, 1 = bf16, 2 = int8 (symmetric, per-channel weights)
 void launch_gemm_bt(int dtype, const void* A, const void* B, void* C,
                     const float* scale, const float* bias, const void* residual,
-                    int M, int N, int K, int64_t lda, int64_t ldb, int64_t ldc,
-                    int epi, hipStream_t stream, int tile = 0,
-                    void* scratch = nullptr);
+                    float res_scale, int M, int N, int K, int64_t lda,
+                    int64_t ldb, int64_t ldc, int epi, hipStream_t stream,
+                    int tile = 0, void* scratch = nullptr);
 size_t gemm_scratch_bytes(int M, int N, int K);
 
 void launch_conv2d(int dtype, const void* in, const void* Wt, void* out,
@@ -18,7 +18,7 @@ void launch_conv2d(int dtype, const void* in, const void* Wt, void* out,
                    const void* zero_page, int Nb, int H, int W, int C,
                    int Cout, int KH, int KW, int sh, int sw, int ph, int pw,
                    int epi, hipStream_t stream, int tile = 0,
-                   void* scratch = nullptr);
+                   void* scratch = nullptr, float res_scale = 1.0f);
 size_t conv_scratch_bytes(int Nb, int H, int W, int C, int Cout, int KH,
                           int KW, int sh, int sw, int ph, int pw);
 
@@ -47,5 +47,12 @@ void launch_cast(int dtype, bool to_f32, const void* in, void* out, int64_t n,
 
 void launch_attention(int dtype, const void* qkv, void* out, int B, int S,
                       int H, int D, float scale, hipStream_t stream);
+
+// int8 quantization staging: out_i8 = clamp(round(in_f16 / scale)),
+// and the inverse. n % 8 == 0.
+void launch_quantize(const void* in_f16, void* out_i8, int64_t n, float scale,
+                     hipStream_t stream);
+void launch_dequant(const void* in_i8, void* out_f16, int64_t n, float scale,
+                    hipStream_t stream);
 
 }  // namespace trtlab

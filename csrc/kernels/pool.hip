@@ -31,18 +31,21 @@ __global__ void maxpool2d_kernel(const T* __restrict__ in, T* __restrict__ out,
         int iw = ow * sw - pw + kw;
         if ((unsigned)iw >= (unsigned)W) continue;
         const T* p = in + ((((int64_t)n * H + ih) * W + iw) * C) + c8 * 8;
-        short4v v0 = *(const short4v*)p;
-        short4v v1 = *(const short4v*)(p + 4);
-#pragma unroll
-        for (int j = 0; j < 4; ++j) {
-          best[j] = fmaxf(best[j], (float)((const T*)&v0)[j]);
-          best[4 + j] = fmaxf(best[4 + j], (float)((const T*)&v1)[j]);
+        T tmp[8];
+        if constexpr (sizeof(T) == 2) {
+          *(short4v*)&tmp[0] = *(const short4v*)p;
+          *(short4v*)&tmp[4] = *(const short4v*)(p + 4);
+        } else {
+          *(uint2*)&tmp[0] = *(const uint2*)p;
         }
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          best[j] = fmaxf(best[j], (float)tmp[j]);
       }
     }
     T* o = out + ((((int64_t)n * OH + oh) * OW + ow) * C) + c8 * 8;
 #pragma unroll
-    for (int j = 0; j < 8; ++j) o[j] = (T)best[j];
+    for (int j = 0; j < 8; ++j) o[j] = (T)best[j];  // exact for int8 maxima
   }
 }
 
@@ -93,8 +96,10 @@ void launch_maxpool2d(int dtype, const void* in, void* out, int Nb, int H,
   if (C % 8 != 0) throw std::runtime_error("maxpool2d: C % 8 != 0");
   if (dtype == 0)
     launch_maxpool2d_t<_Float16>(in, out, Nb, H, W, C, KH, KW, sh, sw, ph, pw, stream);
-  else
+  else if (dtype == 1)
     launch_maxpool2d_t<__bf16>(in, out, Nb, H, W, C, KH, KW, sh, sw, ph, pw, stream);
+  else
+    launch_maxpool2d_t<int8_t>(in, out, Nb, H, W, C, KH, KW, sh, sw, ph, pw, stream);
 }
 
 void launch_gavgpool(int dtype, const void* in, void* out, int Nb, int HW,

@@ -15,10 +15,10 @@ __global__ void splitk_reduce_kernel(const float* __restrict__ scratch,
                                      T* __restrict__ C,
                                      const float* __restrict__ scale,
                                      const float* __restrict__ bias,
-                                     const T* __restrict__ residual, int M,
-                                     int N, int64_t ldc, int tiles_n,
-                                     int splitk, int log_elems, int bn,
-                                     int64_t total4) {
+                                     const T* __restrict__ residual,
+                                     float res_scale, int M, int N,
+                                     int64_t ldc, int tiles_n, int splitk,
+                                     int log_elems, int bn, int64_t total4) {
   // total4 = tiles * (bm*bn) / 4 vector groups of 4 consecutive elements.
   for (int64_t g4 = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
        g4 < total4; g4 += (int64_t)gridDim.x * blockDim.x) {
@@ -49,18 +49,18 @@ __global__ void splitk_reduce_kernel(const float* __restrict__ scratch,
       if constexpr (E != Epi::kNone) bi = bias[col];
       float res = 0.0f;
       if constexpr (E == Epi::kScaleBiasAddRelu)
-        res = (float)residual[(int64_t)row * ldc + col];
+        res = (float)residual[(int64_t)row * ldc + col] * res_scale;
       float vv = j == 0 ? v.x : (j == 1 ? v.y : (j == 2 ? v.z : v.w));
-      C[(int64_t)row * ldc + col] = (T)apply_epi<E>(vv, sc, bi, res);
+      C[(int64_t)row * ldc + col] = store_cast<T>(apply_epi<E>(vv, sc, bi, res));
     }
   }
 }
 
 void launch_splitk_reduce(int dtype, const float* scratch, void* C,
                           const float* scale, const float* bias,
-                          const void* residual, int M, int N, int64_t ldc,
-                          int tiles_m, int tiles_n, int splitk, int bm,
-                          int bn, int epi, hipStream_t stream) {
+                          const void* residual, float res_scale, int M, int N,
+                          int64_t ldc, int tiles_m, int tiles_n, int splitk,
+                          int bm, int bn, int epi, hipStream_t stream) {
   int elems = bm * bn;  // power of two (64/128 x 64/128)
   int log_elems = 31 - __builtin_clz(elems);
   int64_t total4 = (int64_t)tiles_m * tiles_n * elems / 4;
@@ -72,13 +72,18 @@ void launch_splitk_reduce(int dtype, const float* scratch, void* C,
     if (dtype == 0)
       hipLaunchKernelGGL((splitk_reduce_kernel<_Float16, EE>), grid, block, 0,
                          stream, scratch, (_Float16*)C, scale, bias,
-                         (const _Float16*)residual, M, N, ldc, tiles_n,
-                         splitk, log_elems, bn, total4);
-    else
+                         (const _Float16*)residual, res_scale, M, N, ldc,
+                         tiles_n, splitk, log_elems, bn, total4);
+    else if (dtype == 1)
       hipLaunchKernelGGL((splitk_reduce_kernel<__bf16, EE>), grid, block, 0,
                          stream, scratch, (__bf16*)C, scale, bias,
-                         (const __bf16*)residual, M, N, ldc, tiles_n, splitk,
-                         log_elems, bn, total4);
+                         (const __bf16*)residual, res_scale, M, N, ldc,
+                         tiles_n, splitk, log_elems, bn, total4);
+    else
+      hipLaunchKernelGGL((splitk_reduce_kernel<int8_t, EE>), grid, block, 0,
+                         stream, scratch, (int8_t*)C, scale, bias,
+                         (const int8_t*)residual, res_scale, M, N, ldc,
+                         tiles_n, splitk, log_elems, bn, total4);
   });
 }
 

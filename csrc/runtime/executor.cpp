@@ -91,13 +91,13 @@ void ExecutionContext::enqueue_all(hipStream_t s) {
                       Fp(op.scale_off), Fp(op.bias_off), A(op.in2_off),
                       eng_->zero_page(), op.Nb, op.H, op.W, op.C, op.Cout,
                       op.KH, op.KW, op.sh, op.sw, op.ph, op.pw, op.epi, s, 0,
-                      scratch_);
+                      scratch_, op.res_scale);
         break;
       case kGemmBt:
         launch_gemm_bt(op.dtype, A(op.in_off), Wp(op.w_off), A(op.out_off),
-                       Fp(op.scale_off), Fp(op.bias_off), A(op.in2_off), op.M,
-                       op.N, op.K, op.K /*lda*/, op.K /*ldb*/, op.N /*ldc*/,
-                       op.epi, s, 0, scratch_);
+                       Fp(op.scale_off), Fp(op.bias_off), A(op.in2_off),
+                       op.res_scale, op.M, op.N, op.K, op.K /*lda*/,
+                       op.K /*ldb*/, op.N /*ldc*/, op.epi, s, 0, scratch_);
         break;
       case kMaxPool:
         launch_maxpool2d(op.dtype, A(op.in_off), A(op.out_off), op.Nb, op.H,
@@ -133,6 +133,14 @@ void ExecutionContext::enqueue_all(hipStream_t s) {
       case kAttention:
         launch_attention(op.dtype, A(op.in_off), A(op.out_off), op.B, op.S,
                          op.NH, op.HD, op.att_scale, s);
+        break;
+      case kQuantize:
+        launch_quantize(A(op.in_off), A(op.out_off), op.n_elems, op.q_scale,
+                        s);
+        break;
+      case kDequant:
+        launch_dequant(A(op.in_off), A(op.out_off), op.n_elems, op.q_scale,
+                       s);
         break;
       default:
         throw std::runtime_error("unknown op kind");

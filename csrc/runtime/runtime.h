@@ -49,6 +49,8 @@ enum OpKind : int {
   kElementwise = 7,
   kChannelPad = 8,
   kAttention = 9,
+  kQuantize = 10,   // fp16 -> int8 (symmetric, scale)
+  kDequant = 11,    // int8 -> fp16 (scale)
 };
 
 struct OpDesc {
@@ -70,6 +72,9 @@ struct OpDesc {
   // attention
   int B = 0, S = 0, NH = 0, HD = 0;
   float att_scale = 1.0f;
+  // int8: residual dequant ratio (s_res/s_out); quantize/dequant scale
+  float res_scale = 1.0f;
+  float q_scale = 1.0f;
 };
 
 // A compiled model: weight blob on device + op list + arena layout.

@@ -124,3 +124,27 @@ def test_engine_bert():
     scale = max(np.abs(ref).max(), 1e-6)
     assert np.isfinite(out).all()
     assert err / scale < 0.08, (err, scale)
+
+
+def test_engine_resnet50_int8():
+    """BASELINE config 3 numerics: int8 engine vs the CPU int8 emulation."""
+    from trtlab_amd.engine.planner import DT_I8, Planner
+    from trtlab_amd.engine.reference import run_reference
+    from trtlab_amd.engine.runtime import NativeEngine
+    from trtlab_amd.models import build_resnet
+
+    g = build_resnet(50, batch=2, image=64, seed=0)
+    plan = Planner(dtype=DT_I8).compile(g)
+    eng = NativeEngine(plan)
+    ctx = eng.create_context(capture=True)
+    x = np.random.RandomState(11).randn(*plan.input_shape).astype(np.float32) * 0.5
+    out = ctx.infer(x).astype(np.float32)
+    ref = run_reference(plan, x)
+    err = np.abs(out - ref).max()
+    scale = max(np.abs(ref).max(), 1e-6)
+    assert np.isfinite(out).all()
+    # int8 rounding can differ by ~1 code unit per conv; the fp16 head keeps
+    # the divergence small
+    assert err / scale < 0.15, (err, scale)
+    corr = np.corrcoef(out.ravel(), ref.ravel())[0, 1]
+    assert corr > 0.99, corr

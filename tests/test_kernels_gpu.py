@@ -245,3 +245,64 @@ def test_attention(C):
         / (D ** 0.5), dim=-1)
     ref = (att @ q[:, :, 2].permute(0, 2, 1, 3)).permute(0, 2, 1, 3).reshape(B * S, hid)
     check(out, ref)
+
+
+def test_gemm_bt_int8(C):
+    M, N, K = 256, 512, 384  # K % 128 == 0
+    g = torch.Generator(device="cuda").manual_seed(21)
+    a = torch.randint(-127, 128, (M, K), generator=g, device="cuda",
+                      dtype=torch.int8)
+    b = torch.randint(-127, 128, (N, K), generator=g, device="cuda",
+                      dtype=torch.int8)
+    scale = torch.full((N,), 1e-3, device="cuda").float()
+    bias = torch.randn(N, generator=g, device="cuda").float()
+    out = torch.empty(M, N, dtype=torch.int8, device="cuda")
+    torch.cuda.synchronize()
+    C.ops.gemm_bt(2, a.data_ptr(), b.data_ptr(), out.data_ptr(),
+                  scale=scale.data_ptr(), bias=bias.data_ptr(),
+                  M=M, N=N, K=K, epi=EPI_SB)
+    acc = a.float() @ b.float().t()
+    ref = torch.clamp(torch.round(acc * scale + bias), -127, 127)
+    err = (out.float() - ref).abs()
+    assert (err <= 1).all(), f"max int8 err {err.max()}"
+    assert (err > 0).float().mean() < 1e-3
+
+
+def test_conv2d_int8(C):
+    nb, h, w_, cin, cout = 2, 14, 14, 256, 256
+    g = torch.Generator(device="cuda").manual_seed(22)
+    x = torch.randint(-64, 65, (nb, h, w_, cin), generator=g, device="cuda",
+                      dtype=torch.int8)
+    wt = torch.randint(-64, 65, (cout, cin, 3, 3), generator=g,
+                       device="cuda", dtype=torch.int8)
+    flat = wt.permute(0, 2, 3, 1).reshape(cout, 9 * cin)
+    kp = ((9 * cin + 127) // 128) * 128
+    if kp != flat.shape[1]:
+        flat = torch.nn.functional.pad(flat, (0, kp - flat.shape[1]))
+    wp = flat.contiguous()
+    scale = torch.full((cout,), 2e-5, device="cuda").float()
+    bias = torch.zeros(cout, device="cuda").float()
+    zero = torch.zeros(128, dtype=torch.int8, device="cuda")
+    out = torch.empty(nb, h, w_, cout, dtype=torch.int8, device="cuda")
+    torch.cuda.synchronize()
+    C.ops.conv2d(2, x.data_ptr(), wp.data_ptr(), out.data_ptr(),
+                 scale=scale.data_ptr(), bias=bias.data_ptr(),
+                 zero_page=zero.data_ptr(), Nb=nb, H=h, W=w_, C=cin,
+                 Cout=cout, KH=3, KW=3, sh=1, sw=1, ph=1, pw=1, epi=EPI_SB)
+    acc = torch.nn.functional.conv2d(x.float().permute(0, 3, 1, 2),
+                                     wt.float(), padding=1).permute(0, 2, 3, 1)
+    ref = torch.clamp(torch.round(acc * scale + bias), -127, 127)
+    err = (out.float() - ref).abs()
+    assert (err <= 1).all(), f"max int8 err {err.max()}"
+
+
+def test_quantize_dequant(C):
+    x = t16(64, 128, seed=30, scale=2.0)
+    q = torch.empty(64, 128, dtype=torch.int8, device="cuda")
+    torch.cuda.synchronize()
+    C.ops.quantize(x.data_ptr(), q.data_ptr(), x.numel(), 0.05)
+    ref_q = torch.clamp(torch.round(x.float() / 0.05), -127, 127)
+    assert (q.float() - ref_q).abs().max() <= 1
+    back = torch.empty(64, 128, dtype=torch.half, device="cuda")
+    C.ops.dequant(q.data_ptr(), back.data_ptr(), q.numel(), 0.05)
+    assert (back.float() - q.float() * 0.05).abs().max() < 1e-3

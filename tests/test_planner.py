@@ -130,3 +130,25 @@ def test_fused_reference_matches_unfused_interpreter():
     err = np.abs(out_fused - out_unfused).max()
     scale = np.abs(out_unfused).max()
     assert err / scale < 0.05, (err, scale)
+
+
+def test_int8_plan_structure():
+    from trtlab_amd.engine.planner import (DT_I8, K_DEQUANT, K_QUANTIZE,
+                                           Planner)
+
+    g = build_resnet(50, batch=1, image=64, seed=0)
+    plan = Planner(dtype=DT_I8).compile(g)
+    kinds = [d["kind"] for d in plan.ops]
+    assert kinds.count(K_QUANTIZE) == 1
+    assert kinds.count(K_DEQUANT) == 1
+    convs = [d for d in plan.ops if d["kind"] == K_CONV]
+    assert all(d["dtype"] == 2 for d in convs)
+    assert all(d["C"] % 16 == 0 for d in convs)
+    # residual convs carry a res_scale
+    res = [d for d in convs if d["epi"] == EPI_SCALE_BIAS_ADD_RELU]
+    assert len(res) == 16
+    assert all(d["res_scale"] > 0 for d in res)
+    # weight K padded to 128 for int8 staging
+    from trtlab_amd.engine.planner import K_GEMM
+    gemms = [d for d in plan.ops if d["kind"] == K_GEMM]
+    assert all(d["dtype"] == 0 for d in gemms)  # head stays fp16

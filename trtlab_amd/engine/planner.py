@@ -38,6 +38,10 @@ K_CONV, K_GEMM, K_MAXPOOL, K_GAVGPOOL, K_SOFTMAX, K_LAYERNORM, \
 
 DT_F16 = 0
 DT_BF16 = 1
+DT_I8 = 2
+
+K_QUANTIZE = 10
+K_DEQUANT = 11
 
 
 @dataclass
@@ -74,9 +78,16 @@ class EnginePlan:
 
 
 class Planner:
-    def __init__(self, dtype: int = DT_F16, reuse: bool = True):
+    """dtype=DT_I8 lowers the conv stack to int8 (symmetric, per-channel
+    weights, per-tensor activations calibrated on a synthetic forward);
+    pools stay int8, the classifier head (gavgpool/gemm/softmax) stays fp16
+    with quantize/dequant staging ops — BASELINE config 3."""
+
+    def __init__(self, dtype: int = DT_F16, reuse: bool = True,
+                 calib_sample=None):
         self.dtype = dtype
         self.reuse = reuse  # False: disjoint arena slots (debugging)
+        self.calib_sample = calib_sample  # int8 activation calibration input
 
     # ------------------------------------------------------------- fusion
     def fuse(self, g: Graph) -> List[ExecOp]:
@@ -216,8 +227,9 @@ class Planner:
         assert input_name is not None
         in_shape = shapes[input_name]
         padded_input = input_name
-        if len(in_shape) == 4 and in_shape[3] % 8 != 0:
-            cpad = round_up(in_shape[3], 8)
+        pad_mult = 16 if self.dtype == DT_I8 else 8
+        if len(in_shape) == 4 and in_shape[3] % pad_mult != 0:
+            cpad = round_up(in_shape[3], pad_mult)
             padded_input = input_name + "_padded"
             shapes[padded_input] = (*in_shape[:3], cpad)
             m = in_shape[0] * in_shape[1] * in_shape[2]
@@ -229,9 +241,15 @@ class Planner:
                 op.inputs = [padded_input if t == input_name else t
                              for t in op.inputs]
 
-        # ---- weight prepacking ----
+        # ---- weight prepacking (+ int8 lowering) ----
+        itemsize: Dict[str, int] = {}
+        if self.dtype == DT_I8:
+            from trtlab_amd.engine.quantize import lower_int8
+
+            lower_int8(g, exec_ops, shapes, itemsize, input_name,
+                       padded_input, self.calib_sample)
         for op in exec_ops:
-            if op.kind == K_CONV:
+            if op.kind == K_CONV and op.params.get("int8") is None:
                 w = op.w  # [Cout, Cin, KH, KW] fp32
                 cout, cin, kh, kw = w.shape
                 cpad = round_up(cin, 8)
@@ -292,7 +310,7 @@ class Planner:
             n = 1
             for d in shapes[t]:
                 n *= d
-            return n * 2  # fp16/bf16
+            return n * itemsize.get(t, 2)
 
         arena = ArenaPlanner()
         for t, (s0, e0) in tensors_used.items():
@@ -306,7 +324,9 @@ class Planner:
         op_dicts: List[Dict[str, Any]] = []
         for op in exec_ops:
             w_off, s_off, b_off = w_offs[op.name]
-            d: Dict[str, Any] = dict(dtype=self.dtype, w_off=w_off,
+            op_dtype = op.params.get("dtype", DT_F16 if self.dtype == DT_I8
+                                     else self.dtype)
+            d: Dict[str, Any] = dict(dtype=op_dtype, w_off=w_off,
                                      scale_off=s_off, bias_off=b_off,
                                      in_off=offsets[op.inputs[0]],
                                      out_off=offsets[op.output])
@@ -320,7 +340,8 @@ class Planner:
                          Cout=osh[3], KH=op.params["weight_shape"][2],
                          KW=op.params["weight_shape"][3],
                          sh=op.params["stride"], sw=op.params["stride"],
-                         ph=op.params["padding"], pw=op.params["padding"])
+                         ph=op.params["padding"], pw=op.params["padding"],
+                         res_scale=op.params.get("res_scale", 1.0))
             elif op.kind == K_GEMM:
                 m, k = shapes[op.inputs[0]]
                 nout = shapes[op.output][1]
@@ -354,6 +375,12 @@ class Planner:
             elif op.kind == K_CHANNEL_PAD:
                 d.update(kind=K_CHANNEL_PAD, n_elems=op.params["M"],
                          C=op.params["Cin"], Cout=op.params["Cpad"])
+            elif op.kind in (K_QUANTIZE, K_DEQUANT):
+                n = 1
+                for s_ in shapes[op.output]:
+                    n *= s_
+                d.update(kind=op.kind, n_elems=n,
+                         q_scale=op.params["q_scale"])
             elif op.kind == K_ATTENTION:
                 m, hid = shapes[op.output]
                 heads = op.params["heads"]

@@ -1,0 +1,173 @@
+"""Int8 lowering pass (BASELINE config 3: ResNet-50 int8).
+
+Scheme: symmetric int8; per-channel weight scales, per-tensor activation
+scales calibrated from one fp32 forward on synthetic data (the role of the
+reference's int8 calibrator, examples/ONNX/resnet50/calibrator.py).
+
+The conv stack (convs + maxpool between them) runs entirely in int8:
+  acc_i32 = sum(q_in * q_w)
+  y_real  = acc * s_in * s_w[c] * bn_gamma[c] + bn_beta[c] (+ res * s_res)
+  q_out   = clamp(round(relu(y_real) / s_out))
+Everything folds into the existing epilogue:
+  scale'[c] = s_in * s_w[c] * g[c] / s_out,  bias'[c] = b[c] / s_out,
+  res_scale = s_res / s_out
+The classifier head (gavgpool/gemm/softmax) stays fp16 behind a dequant;
+the padded input is quantized on-device after channel_pad.
+"""
+from __future__ import annotations
+
+from typing import Dict, List
+
+import numpy as np
+
+from trtlab_amd.engine.ir import Graph
+from trtlab_amd.utils import round_up
+
+# local copies to avoid a circular import with planner.py
+_K_CONV, _K_GEMM, _K_MAXPOOL, _K_GAVGPOOL = 0, 1, 2, 3
+_K_CHANNEL_PAD = 8
+_K_QUANTIZE, _K_DEQUANT = 10, 11
+_EPI_NONE, _EPI_BIAS_RELU = 0, 2
+_EPI_SB, _EPI_SB_RELU, _EPI_SB_ADD_RELU = 4, 5, 6
+_DT_F16, _DT_I8 = 0, 2
+
+
+def calibrate_amax(g: Graph, sample: np.ndarray) -> Dict[str, float]:
+    """One node-level fp32 forward; records amax of every tensor."""
+    import torch
+
+    from trtlab_amd.engine import calibrate as C
+
+    # reuse the calibrate interpreter by monkey-capturing outputs: simplest
+    # is to re-run its logic; calibrate_bn already sets BN stats, so here we
+    # interpret without mutating and track amax.
+    t = {g.input_name: torch.from_numpy(np.ascontiguousarray(sample)).float()}
+    amax: Dict[str, float] = {g.input_name: float(np.abs(sample).max())}
+    import torch.nn.functional as F
+
+    for n in g.nodes:
+        if n.kind == "input":
+            continue
+        x = t[n.inputs[0]]
+        if n.kind == "conv2d":
+            w = torch.from_numpy(n.attrs["weight"])
+            y = F.conv2d(x.permute(0, 3, 1, 2), w, stride=n.attrs["stride"],
+                         padding=n.attrs["padding"]).permute(0, 2, 3, 1)
+        elif n.kind == "batchnorm":
+            a = n.attrs
+            y = (x - torch.from_numpy(a["mean"])) / torch.sqrt(
+                torch.from_numpy(a["var"]) + a["eps"])
+            y = y * torch.from_numpy(a["gamma"]) + torch.from_numpy(a["beta"])
+        elif n.kind == "relu":
+            y = F.relu(x)
+        elif n.kind == "add":
+            y = x + t[n.inputs[1]]
+        elif n.kind == "maxpool":
+            y = F.max_pool2d(x.permute(0, 3, 1, 2), n.attrs["kernel"],
+                             stride=n.attrs["stride"],
+                             padding=n.attrs["padding"]).permute(0, 2, 3, 1)
+        elif n.kind == "gavgpool":
+            nb, h, w_, c = x.shape
+            y = x.reshape(nb, h * w_, c).mean(1)
+        elif n.kind == "gemm":
+            y = x @ torch.from_numpy(n.attrs["weight"]).t()
+            if n.attrs.get("bias") is not None:
+                y = y + torch.from_numpy(n.attrs["bias"])
+        elif n.kind == "softmax":
+            y = F.softmax(x, dim=-1)
+        else:
+            raise ValueError(f"int8 calibration: unsupported node {n.kind}")
+        t[n.output] = y
+        amax[n.output] = float(y.abs().max())
+    return amax
+
+
+def lower_int8(g: Graph, exec_ops: List, shapes: Dict, itemsize: Dict,
+               input_name: str, padded_input: str, calib_sample) -> None:
+    """Mutates exec_ops/shapes/itemsize in place for the int8 conv stack."""
+    if calib_sample is None:
+        rng = np.random.RandomState(1234)
+        in_shape = shapes[input_name]
+        calib_sample = (rng.randn(*((min(in_shape[0], 2),) + in_shape[1:]))
+                        .astype(np.float32) * 0.5)
+    amax = calibrate_amax(g, calib_sample)
+
+    scales: Dict[str, float] = {}  # int8 tensor name -> activation scale
+    new_ops: List = []
+    from trtlab_amd.engine.planner import ExecOp
+
+    # which tensors feed fp16-only ops? (gavgpool) — dequant before them
+    for idx, op in enumerate(exec_ops):
+        if op.kind == _K_CHANNEL_PAD:
+            op.params["dtype"] = _DT_F16
+            new_ops.append(op)
+            # quantize the padded input
+            s_in = max(amax[input_name], 1e-6) / 127.0
+            qname = op.output + "_q"
+            shapes[qname] = shapes[op.output]
+            itemsize[qname] = 1
+            qop = ExecOp(_K_QUANTIZE, qname, [op.output], qname,
+                         dict(q_scale=s_in, dtype=_DT_F16))
+            new_ops.append(qop)
+            scales[qname] = s_in
+            # rewrite consumers
+            for o2 in exec_ops[idx + 1:]:
+                o2.inputs = [qname if t == op.output else t for t in o2.inputs]
+        elif op.kind == _K_CONV:
+            s_in = scales[op.inputs[0]]
+            out_name = op.output
+            s_out = max(amax[out_name], 1e-6) / 127.0
+            w = op.w  # [Cout, Cin, KH, KW] fp32 (original)
+            cout, cin, kh, kw = w.shape
+            cpad = round_up(cin, 16)
+            whwc = np.transpose(w, (0, 2, 3, 1))
+            if cpad != cin:
+                whwc = np.pad(whwc, ((0, 0), (0, 0), (0, 0), (0, cpad - cin)))
+            k = kh * kw * cpad
+            kp = round_up(k, 128)
+            flat = whwc.reshape(cout, k).astype(np.float32)
+            if kp != k:
+                flat = np.pad(flat, ((0, 0), (0, kp - k)))
+            sw = np.maximum(np.abs(flat).max(axis=1), 1e-8) / 127.0
+            q = np.clip(np.rint(flat / sw[:, None]), -127, 127).astype(np.int8)
+            op.w = np.ascontiguousarray(q)
+            op.params["C"] = cpad
+            op.params["Kp"] = kp
+            op.params["dtype"] = _DT_I8
+            op.params["int8"] = True
+            # fold scales into the epilogue
+            g_ = op.scale if op.scale is not None else np.ones(cout, np.float32)
+            b_ = op.bias if op.bias is not None else np.zeros(cout, np.float32)
+            op.scale = (s_in * sw * g_ / s_out).astype(np.float32)
+            op.bias = (b_ / s_out).astype(np.float32)
+            epi = op.params["epi"]
+            if epi == _EPI_NONE:
+                op.params["epi"] = _EPI_SB
+            elif epi == _EPI_BIAS_RELU:
+                op.params["epi"] = _EPI_SB_RELU
+            if op.params["epi"] == _EPI_SB_ADD_RELU:
+                s_res = scales[op.inputs[1]]
+                op.params["res_scale"] = float(s_res / s_out)
+            itemsize[out_name] = 1
+            scales[out_name] = s_out
+            new_ops.append(op)
+        elif op.kind == _K_MAXPOOL and op.inputs[0] in scales:
+            op.params["dtype"] = _DT_I8
+            itemsize[op.output] = 1
+            scales[op.output] = scales[op.inputs[0]]  # max() preserves scale
+            new_ops.append(op)
+        elif op.kind == _K_GAVGPOOL and op.inputs[0] in scales:
+            # dequant the final int8 activation back to fp16
+            src = op.inputs[0]
+            dq = src + "_dq"
+            shapes[dq] = shapes[src]
+            itemsize[dq] = 2
+            new_ops.append(ExecOp(_K_DEQUANT, dq, [src], dq,
+                                  dict(q_scale=scales[src], dtype=_DT_F16)))
+            op.inputs = [dq]
+            op.params["dtype"] = _DT_F16
+            new_ops.append(op)
+        else:
+            op.params.setdefault("dtype", _DT_F16)
+            new_ops.append(op)
+    exec_ops[:] = new_ops

@@ -16,7 +16,8 @@ import torch.nn.functional as F
 
 from trtlab_amd.engine.planner import (
     EnginePlan, K_ADD_LAYERNORM, K_ATTENTION, K_CHANNEL_PAD, K_CONV,
-    K_ELEMENTWISE, K_GAVGPOOL, K_GEMM, K_LAYERNORM, K_MAXPOOL, K_SOFTMAX,
+    K_DEQUANT, K_ELEMENTWISE, K_GAVGPOOL, K_GEMM, K_LAYERNORM, K_MAXPOOL,
+    K_QUANTIZE, K_SOFTMAX,
     EPI_BIAS, EPI_BIAS_GELU, EPI_BIAS_RELU, EPI_NONE, EPI_SCALE_BIAS,
     EPI_SCALE_BIAS_ADD_RELU, EPI_SCALE_BIAS_RELU)
 
@@ -68,7 +69,17 @@ def run_reference(plan: EnginePlan, input_nhwc: np.ndarray,
                 bias = torch.from_numpy(op.bias)
             if len(op.inputs) > 1:
                 res = t[op.inputs[1]]
-            t[op.output] = _epi(acc, d["epi"], scale, bias, res)
+                if d.get("res_scale", 1.0) != 1.0:
+                    res = res * d["res_scale"]
+            y = _epi(acc, d["epi"], scale, bias, res)
+            if d["dtype"] == 2:  # int8: emulate the requantized store
+                y = torch.clamp(torch.round(y), -127, 127)
+            t[op.output] = y
+        elif op.kind == K_QUANTIZE:
+            t[op.output] = torch.clamp(torch.round(x / d["q_scale"]),
+                                       -127, 127)
+        elif op.kind == K_DEQUANT:
+            t[op.output] = x * d["q_scale"]
         elif op.kind == K_GEMM:
             wt = torch.from_numpy(op.w.astype(np.float32))  # [N, K]
             acc = x @ wt.t()
