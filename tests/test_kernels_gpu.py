@@ -305,4 +305,5 @@ def test_quantize_dequant(C):
     assert (q.float() - ref_q).abs().max() <= 1
     back = torch.empty(64, 128, dtype=torch.half, device="cuda")
     C.ops.dequant(q.data_ptr(), back.data_ptr(), q.numel(), 0.05)
-    assert (back.float() - q.float() * 0.05).abs().max() < 1e-3
+    # fp16 output rounding: ~2^-11 relative at |x| ~ 6
+    assert (back.float() - q.float() * 0.05).abs().max() < 4e-3
