@@ -35,6 +35,7 @@ def main():
     ap.add_argument("--contexts", type=int, default=3)
     ap.add_argument("--model", default="resnet50",
                     choices=["resnet50", "resnet101", "resnet152", "bert"])
+    ap.add_argument("--dtype", default="fp16", choices=["fp16", "int8"])
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -52,7 +53,7 @@ def main():
         dist.init_process_group("nccl")
 
     import trtlab_amd
-    from trtlab_amd.engine.planner import Planner
+    from trtlab_amd.engine.planner import DT_F16, DT_I8, Planner
     from trtlab_amd.engine.runtime import NativeEngine
     from trtlab_amd.models import build_bert, build_resnet
     from trtlab_amd.parallel import broadcast_weights
@@ -65,7 +66,8 @@ def main():
         depth = int(args.model.replace("resnet", ""))
         g = build_resnet(depth, batch=args.batch, image=224, seed=0)
         cfg_extra = {"image": 224}
-    plan = Planner().compile(g)
+    dtype = DT_I8 if args.dtype == "int8" else DT_F16
+    plan = Planner(dtype=dtype).compile(g)
 
     eng = NativeEngine(plan, device=local_rank)
     if distributed:
@@ -132,7 +134,7 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": round(inf_s / BASELINE_INF_S, 3),
-            "dtype": "fp16",
+            "dtype": args.dtype,
             "data": "synthetic",
             "config": {
                 "model": args.model,
