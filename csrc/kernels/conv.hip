@@ -76,7 +76,7 @@ void launch_splitk_reduce(int dtype, const float* scratch, void* C,
                           int64_t ldc, int tiles_m, int tiles_n, int splitk,
                           int bm, int bn, int epi, hipStream_t stream);
 
-template <typename T, Epi E, int BM, int BN, bool SPLIT>
+template <typename T, Epi E, int BM, int BN, bool SPLIT, int NBUF>
 __global__ __launch_bounds__(256) void conv_igemm_kernel(
     const T* __restrict__ in, const T* __restrict__ Wt, T* __restrict__ out,
     const float* __restrict__ scale, const float* __restrict__ bias,
@@ -91,7 +91,7 @@ __global__ __launch_bounds__(256) void conv_igemm_kernel(
   int m0 = (int)(tile / tiles_n) * BM;
   int n0 = (int)(tile % tiles_n) * BN;
 
-  __shared__ __attribute__((aligned(16))) char smem[2 * kBuf];
+  __shared__ __attribute__((aligned(16))) char smem[NBUF * kBuf];
   uint32_t lds0 = (uint32_t)(uintptr_t)&smem[0];
 
   const int tid = threadIdx.x;
@@ -115,25 +115,38 @@ __global__ __launch_bounds__(256) void conv_igemm_kernel(
     kt1 = min(ktiles, kt0 + ktper);
   }
 
-  stage_conv_a<T, BM>(in, zero_page, p, m0, kt0 * KT, lds0, tid);
-  stage_tile<T, BN>(Wt + (int64_t)n0 * p.K + kt0 * KT, p.K, n0, p.Cout,
-                    lds0 + kABytes, tid);
-  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-  __syncthreads();
+  auto stage = [&](int t, int slot) {
+    uint32_t base = lds0 + slot * kBuf;
+    stage_conv_a<T, BM>(in, zero_page, p, m0, t * KT, base, tid);
+    stage_tile<T, BN>(Wt + (int64_t)n0 * p.K + (int64_t)t * KT, p.K, n0,
+                      p.Cout, base + kABytes, tid);
+  };
 
-  int cur = 0;
-  for (int t = kt0; t < kt1; ++t) {
-    if (t + 1 < kt1) {
-      uint32_t nb = lds0 + (cur ^ 1) * kBuf;
-      stage_conv_a<T, BM>(in, zero_page, p, m0, (t + 1) * KT, nb, tid);
-      stage_tile<T, BN>(Wt + (int64_t)n0 * p.K + (t + 1) * KT, p.K, n0,
-                        p.Cout, nb + kABytes, tid);
+  if constexpr (NBUF == 4) {
+    constexpr int G = BM / 32 + BN / 32;
+    for (int i = 0; i < 3 && kt0 + i < kt1; ++i) stage(kt0 + i, i);
+    for (int t = kt0; t < kt1; ++t) {
+      int ahead = kt1 - 1 - t;
+      if (ahead > 2) ahead = 2;
+      wait_tiles_inflight<G>(ahead);
+      __builtin_amdgcn_s_barrier();
+      if (t + 3 < kt1) stage(t + 3, (t + 3 - kt0) & 3);
+      const char* As = &smem[((t - kt0) & 3) * kBuf];
+      mfma_tile<T, BM, BN>(As, As + kABytes, lane, wr, wc, acc);
     }
-    const char* As = &smem[cur * kBuf];
-    mfma_tile<T, BM, BN>(As, As + kABytes, lane, wr, wc, acc);
+  } else {
+    stage(kt0, 0);
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __syncthreads();
-    cur ^= 1;
+    int cur = 0;
+    for (int t = kt0; t < kt1; ++t) {
+      if (t + 1 < kt1) stage(t + 1, cur ^ 1);
+      const char* As = &smem[cur * kBuf];
+      mfma_tile<T, BM, BN>(As, As + kABytes, lane, wr, wc, acc);
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      __syncthreads();
+      cur ^= 1;
+    }
   }
 
   if constexpr (SPLIT) {
@@ -176,14 +189,22 @@ static void launch_conv2d_t(const void* in, const void* Wt, void* out,
   if (splitk > 1) {
     int ktper = (int)cdiv(ktiles, splitk);
     dim3 grid((unsigned)(tiles * splitk));
+    bool deep = want_deep_pipe(tiles * splitk, ktper);
     tile_dispatch(cfg, [&](auto bm, auto bn) {
       constexpr int BM = decltype(bm)::value;
       constexpr int BN = decltype(bn)::value;
-      hipLaunchKernelGGL((conv_igemm_kernel<T, Epi::kNone, BM, BN, true>),
-                         grid, block, 0, stream, (const T*)in, (const T*)Wt,
-                         (T*)out, scale, bias, (const T*)residual,
-                         (const T*)zero_page, p, tiles_n, scratch, splitk,
-                         ktper);
+      if (deep)
+        hipLaunchKernelGGL((conv_igemm_kernel<T, Epi::kNone, BM, BN, true, 4>),
+                           grid, block, 0, stream, (const T*)in, (const T*)Wt,
+                           (T*)out, scale, bias, (const T*)residual,
+                           (const T*)zero_page, p, tiles_n, scratch, splitk,
+                           ktper);
+      else
+        hipLaunchKernelGGL((conv_igemm_kernel<T, Epi::kNone, BM, BN, true, 2>),
+                           grid, block, 0, stream, (const T*)in, (const T*)Wt,
+                           (T*)out, scale, bias, (const T*)residual,
+                           (const T*)zero_page, p, tiles_n, scratch, splitk,
+                           ktper);
     });
     launch_splitk_reduce(dtype, scratch, out, scale, bias, residual,
                          p.res_scale, p.M, p.Cout, p.Cout, tiles_m, tiles_n,
@@ -191,16 +212,24 @@ static void launch_conv2d_t(const void* in, const void* Wt, void* out,
     return;
   }
   dim3 grid((unsigned)tiles);
+  bool deep = want_deep_pipe(tiles, ktiles);
   epi_dispatch(epi, [&](auto e) {
     constexpr Epi EE = decltype(e)::value;
     tile_dispatch(cfg, [&](auto bm, auto bn) {
       constexpr int BM = decltype(bm)::value;
       constexpr int BN = decltype(bn)::value;
-      hipLaunchKernelGGL((conv_igemm_kernel<T, EE, BM, BN, false>), grid,
-                         block, 0, stream, (const T*)in, (const T*)Wt,
-                         (T*)out, scale, bias, (const T*)residual,
-                         (const T*)zero_page, p, tiles_n, (float*)nullptr, 1,
-                         ktiles);
+      if (deep)
+        hipLaunchKernelGGL((conv_igemm_kernel<T, EE, BM, BN, false, 4>), grid,
+                           block, 0, stream, (const T*)in, (const T*)Wt,
+                           (T*)out, scale, bias, (const T*)residual,
+                           (const T*)zero_page, p, tiles_n, (float*)nullptr,
+                           1, ktiles);
+      else
+        hipLaunchKernelGGL((conv_igemm_kernel<T, EE, BM, BN, false, 2>), grid,
+                           block, 0, stream, (const T*)in, (const T*)Wt,
+                           (T*)out, scale, bias, (const T*)residual,
+                           (const T*)zero_page, p, tiles_n, (float*)nullptr,
+                           1, ktiles);
     });
   });
 }

@@ -30,7 +30,7 @@ void launch_splitk_reduce(int dtype, const float* scratch, void* C,
                           int64_t ldc, int tiles_m, int tiles_n, int splitk,
                           int bm, int bn, int epi, hipStream_t stream);
 
-template <typename T, Epi E, int BM, int BN, bool SPLIT>
+template <typename T, Epi E, int BM, int BN, bool SPLIT, int NBUF>
 __global__ __launch_bounds__(256) void gemm_bt_kernel(
     const T* __restrict__ A, const T* __restrict__ B, T* __restrict__ C,
     const float* __restrict__ scale, const float* __restrict__ bias,
@@ -54,7 +54,7 @@ __global__ __launch_bounds__(256) void gemm_bt_kernel(
     kt1 = min(ktiles, kt0 + ktper);
   }
 
-  __shared__ __attribute__((aligned(16))) char smem[2 * kBuf];
+  __shared__ __attribute__((aligned(16))) char smem[NBUF * kBuf];
   uint32_t lds0 = (uint32_t)(uintptr_t)&smem[0];
 
   const int tid = threadIdx.x;
@@ -69,26 +69,41 @@ __global__ __launch_bounds__(256) void gemm_bt_kernel(
 #pragma unroll
     for (int j = 0; j < BN / 32; ++j) acc[i][j] = {0, 0, 0, 0};
 
-  stage_tile<T, BM>(A + (int64_t)m0 * lda + kt0 * KT, lda, m0, M, lds0, tid);
-  stage_tile<T, BN>(B + (int64_t)n0 * ldb + kt0 * KT, ldb, n0, N,
-                    lds0 + kABytes, tid);
-  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-  __syncthreads();
+  auto stage = [&](int t, int slot) {
+    uint32_t base = lds0 + slot * kBuf;
+    stage_tile<T, BM>(A + (int64_t)m0 * lda + (int64_t)t * KT, lda, m0, M,
+                      base, tid);
+    stage_tile<T, BN>(B + (int64_t)n0 * ldb + (int64_t)t * KT, ldb, n0, N,
+                      base + kABytes, tid);
+  };
 
-  int cur = 0;
-  for (int t = kt0; t < kt1; ++t) {
-    if (t + 1 < kt1) {
-      uint32_t nb = lds0 + (cur ^ 1) * kBuf;
-      stage_tile<T, BM>(A + (int64_t)m0 * lda + (t + 1) * KT, lda, m0, M, nb,
-                        tid);
-      stage_tile<T, BN>(B + (int64_t)n0 * ldb + (t + 1) * KT, ldb, n0, N,
-                        nb + kABytes, tid);
+  if constexpr (NBUF == 4) {
+    // 3-deep staging pipeline: counted vmcnt + raw barrier (loads stay in
+    // flight across barriers; see gemm_common.h wait_tiles_inflight).
+    constexpr int G = BM / 32 + BN / 32;
+    for (int i = 0; i < 3 && kt0 + i < kt1; ++i) stage(kt0 + i, i);
+    for (int t = kt0; t < kt1; ++t) {
+      int ahead = kt1 - 1 - t;
+      if (ahead > 2) ahead = 2;
+      wait_tiles_inflight<G>(ahead);
+      __builtin_amdgcn_s_barrier();
+      if (t + 3 < kt1) stage(t + 3, (t + 3 - kt0) & 3);
+      const char* As = &smem[((t - kt0) & 3) * kBuf];
+      mfma_tile<T, BM, BN>(As, As + kABytes, lane, wr, wc, acc);
     }
-    const char* As = &smem[cur * kBuf];
-    mfma_tile<T, BM, BN>(As, As + kABytes, lane, wr, wc, acc);
+  } else {
+    stage(kt0, 0);
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __syncthreads();
-    cur ^= 1;
+    int cur = 0;
+    for (int t = kt0; t < kt1; ++t) {
+      if (t + 1 < kt1) stage(t + 1, cur ^ 1);
+      const char* As = &smem[cur * kBuf];
+      mfma_tile<T, BM, BN>(As, As + kABytes, lane, wr, wc, acc);
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      __syncthreads();
+      cur ^= 1;
+    }
   }
 
   if constexpr (SPLIT) {
@@ -127,13 +142,22 @@ static void launch_gemm_bt_t(const void* A, const void* B, void* C,
   if (splitk > 1) {
     int ktper = (int)cdiv(ktiles, splitk);
     dim3 grid((unsigned)(tiles * splitk));
+    bool deep = want_deep_pipe(tiles * splitk, ktper);
     tile_dispatch(cfg, [&](auto bm, auto bn) {
       constexpr int BM = decltype(bm)::value;
       constexpr int BN = decltype(bn)::value;
-      hipLaunchKernelGGL((gemm_bt_kernel<T, Epi::kNone, BM, BN, true>), grid,
-                         block, 0, stream, (const T*)A, (const T*)B, (T*)C,
-                         scale, bias, (const T*)residual, res_scale, M, N, K,
-                         lda, ldb, ldc, tiles_n, scratch, splitk, ktper);
+      if (deep)
+        hipLaunchKernelGGL((gemm_bt_kernel<T, Epi::kNone, BM, BN, true, 4>),
+                           grid, block, 0, stream, (const T*)A, (const T*)B,
+                           (T*)C, scale, bias, (const T*)residual, res_scale,
+                           M, N, K, lda, ldb, ldc, tiles_n, scratch, splitk,
+                           ktper);
+      else
+        hipLaunchKernelGGL((gemm_bt_kernel<T, Epi::kNone, BM, BN, true, 2>),
+                           grid, block, 0, stream, (const T*)A, (const T*)B,
+                           (T*)C, scale, bias, (const T*)residual, res_scale,
+                           M, N, K, lda, ldb, ldc, tiles_n, scratch, splitk,
+                           ktper);
     });
     launch_splitk_reduce(dtype, scratch, C, scale, bias, residual, res_scale,
                          M, N, ldc, tiles_m, tiles_n, splitk, cfg.bm, cfg.bn,
@@ -141,15 +165,24 @@ static void launch_gemm_bt_t(const void* A, const void* B, void* C,
     return;
   }
   dim3 grid((unsigned)tiles);
+  bool deep = want_deep_pipe(tiles, ktiles);
   epi_dispatch(epi, [&](auto e) {
     constexpr Epi EE = decltype(e)::value;
     tile_dispatch(cfg, [&](auto bm, auto bn) {
       constexpr int BM = decltype(bm)::value;
       constexpr int BN = decltype(bn)::value;
-      hipLaunchKernelGGL((gemm_bt_kernel<T, EE, BM, BN, false>), grid, block,
-                         0, stream, (const T*)A, (const T*)B, (T*)C, scale,
-                         bias, (const T*)residual, res_scale, M, N, K, lda,
-                         ldb, ldc, tiles_n, (float*)nullptr, 1, ktiles);
+      if (deep)
+        hipLaunchKernelGGL((gemm_bt_kernel<T, EE, BM, BN, false, 4>), grid,
+                           block, 0, stream, (const T*)A, (const T*)B, (T*)C,
+                           scale, bias, (const T*)residual, res_scale, M, N,
+                           K, lda, ldb, ldc, tiles_n, (float*)nullptr, 1,
+                           ktiles);
+      else
+        hipLaunchKernelGGL((gemm_bt_kernel<T, EE, BM, BN, false, 2>), grid,
+                           block, 0, stream, (const T*)A, (const T*)B, (T*)C,
+                           scale, bias, (const T*)residual, res_scale, M, N,
+                           K, lda, ldb, ldc, tiles_n, (float*)nullptr, 1,
+                           ktiles);
     });
   });
 }

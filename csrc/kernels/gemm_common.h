@@ -193,6 +193,41 @@ __device__ __forceinline__ void store_epilogue(
   }
 }
 
+// ------------------------------------------------- deep staging pipeline
+// Counted-vmcnt wait: block until at most `ahead` tiles' glds remain in
+// flight (G = glds instructions per thread per tile). vmcnt retires in
+// issue order, so <= G*ahead outstanding means the current tile landed.
+// (cdna guide §5 "Pipelining across barriers": hipcc's __syncthreads
+// drains glds with vmcnt(0); the counted wait + raw s_barrier keeps
+// later tiles' loads in flight across the barrier.)
+template <int G>
+__device__ __forceinline__ void wait_tiles_inflight(int ahead) {
+  if (ahead >= 2) {
+    if constexpr (G == 4)
+      asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+    else if constexpr (G == 6)
+      asm volatile("s_waitcnt vmcnt(12)" ::: "memory");
+    else
+      asm volatile("s_waitcnt vmcnt(16)" ::: "memory");
+  } else if (ahead == 1) {
+    if constexpr (G == 4)
+      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    else if constexpr (G == 6)
+      asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+    else
+      asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+  } else {
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  }
+}
+
+// Pick the pipeline depth: grid-starved shapes (fewer blocks than ~1.5x
+// the CU count) can't hide the per-K-tile staging latency across blocks,
+// so spend 4x LDS on a 3-deep in-block pipeline instead.
+inline bool want_deep_pipe(long blocks, int ktiles) {
+  return blocks < 384 && ktiles >= 2;
+}
+
 // Split-K slab store: this (tile, slice)'s [BM][BN] fp32 partial sums.
 template <typename T, int BM, int BN>
 __device__ __forceinline__ void store_splitk(
