@@ -275,6 +275,19 @@ PYBIND11_MODULE(_C, m) {
           },
           py::arg("in"), py::arg("out"), py::arg("n"), py::arg("scale"),
           py::arg("stream") = 0, py::arg("sync") = true);
+  ops.def("embedding",
+          [](int dtype, uintptr_t ids, uintptr_t tok, uintptr_t pos,
+             uintptr_t seg, uintptr_t segids, uintptr_t out, int M, int S,
+             int H, uintptr_t stream, bool sync) {
+            launch_embedding(dtype, (void*)ids, (void*)tok, (void*)pos,
+                             (void*)seg, (void*)segids, (void*)out, M, S, H,
+                             as_stream(stream));
+            if (sync) TRT_HIP_CHECK(hipStreamSynchronize(as_stream(stream)));
+          },
+          py::arg("dtype"), py::arg("ids"), py::arg("tok"), py::arg("pos"),
+          py::arg("seg") = 0, py::arg("segids") = 0, py::arg("out") = 0,
+          py::arg("M") = 0, py::arg("S") = 0, py::arg("H") = 0,
+          py::arg("stream") = 0, py::arg("sync") = true);
   ops.def("attention",
           [](int dtype, uintptr_t qkv, uintptr_t out, int B, int S, int H,
              int D, float scale, uintptr_t stream, bool sync) {

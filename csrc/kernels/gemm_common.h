@@ -250,7 +250,7 @@ __device__ __forceinline__ void store_splitk(
 // slab round-trip + reduce launch costs ~8-10 us, so marginal cases lose —
 // measured in tools/tune_tiles.py). Returns 1 = no split.
 inline int pick_splitk(long blocks, int ktiles) {
-  if (blocks > 96 || ktiles < 32) return 1;
+  if (blocks > 144 || ktiles < 24) return 1;
   int splitk = 1;
   while (blocks * splitk < 256 && ktiles / (splitk * 2) >= 4 && splitk < 8)
     splitk *= 2;

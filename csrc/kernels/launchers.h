@@ -48,6 +48,10 @@ void launch_cast(int dtype, bool to_f32, const void* in, void* out, int64_t n,
 void launch_attention(int dtype, const void* qkv, void* out, int B, int S,
                       int H, int D, float scale, hipStream_t stream);
 
+void launch_embedding(int dtype, const void* ids, const void* tok,
+                      const void* pos, const void* seg, const void* segids,
+                      void* out, int M, int S, int H, hipStream_t stream);
+
 // int8 quantization staging: out_i8 = clamp(round(in_f16 / scale)),
 // and the inverse. n % 8 == 0.
 void launch_quantize(const void* in_f16, void* out_i8, int64_t n, float scale,

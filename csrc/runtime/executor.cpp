@@ -142,6 +142,13 @@ void ExecutionContext::enqueue_all(hipStream_t s) {
         launch_dequant(A(op.in_off), A(op.out_off), op.n_elems, op.q_scale,
                        s);
         break;
+      case kEmbedding:
+        // tables live in the weight blob as fp16: tok at w_off, pos at
+        // scale_off, optional seg at bias_off; optional segids at in2_off.
+        launch_embedding(op.dtype, A(op.in_off), Wp(op.w_off),
+                         Wp(op.scale_off), Wp(op.bias_off), A(op.in2_off),
+                         A(op.out_off), op.M, op.S, op.N, s);
+        break;
       default:
         throw std::runtime_error("unknown op kind");
     }

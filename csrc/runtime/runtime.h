@@ -51,6 +51,7 @@ enum OpKind : int {
   kAttention = 9,
   kQuantize = 10,   // fp16 -> int8 (symmetric, scale)
   kDequant = 11,    // int8 -> fp16 (scale)
+  kEmbedding = 12,  // out = tok[ids] + pos[m%S] (+ seg)
 };
 
 struct OpDesc {

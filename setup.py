@@ -23,6 +23,7 @@ SOURCES = [
     "csrc/kernels/normalize.hip",
     "csrc/kernels/elementwise.hip",
     "csrc/kernels/attention.hip",
+    "csrc/kernels/embedding.hip",
     "csrc/runtime/memory.cpp",
     "csrc/runtime/executor.cpp",
     "csrc/ext.cpp",

@@ -151,3 +151,23 @@ def test_engine_resnet50_int8():
     corr_emu = np.corrcoef(ref8.ravel(), ref16.ravel())[0, 1]
     assert corr_gpu > corr_emu - 0.03, (corr_gpu, corr_emu)
     assert corr_gpu > 0.9, corr_gpu
+
+
+def test_engine_bert_with_embeddings():
+    from trtlab_amd.engine.planner import Planner
+    from trtlab_amd.engine.reference import run_reference
+    from trtlab_amd.engine.runtime import NativeEngine
+    from trtlab_amd.models import build_bert
+
+    g = build_bert(batch=2, seq=128, layers=1, seed=0, embeddings=True)
+    plan = Planner().compile(g)
+    assert plan.input_dtype == "i32"
+    eng = NativeEngine(plan)
+    ctx = eng.create_context(capture=True)
+    ids = np.random.RandomState(13).randint(0, 30522, 256).astype(np.int32)
+    out = ctx.infer(ids).astype(np.float32)
+    ref = run_reference(plan, ids)
+    err = np.abs(out - ref).max()
+    scale = max(np.abs(ref).max(), 1e-6)
+    assert np.isfinite(out).all()
+    assert err / scale < 0.08, (err, scale)

@@ -28,7 +28,7 @@ class TensorSpec:
 
     @property
     def nbytes(self) -> int:
-        itemsize = {"f16": 2, "bf16": 2, "f32": 4}[self.dtype]
+        itemsize = {"f16": 2, "bf16": 2, "f32": 4, "i32": 4, "i8": 1}[self.dtype]
         return self.numel * itemsize
 
 
@@ -64,8 +64,9 @@ class Graph:
         self.tensors[name] = TensorSpec(name, tuple(int(s) for s in shape), dtype)
         return name
 
-    def input(self, shape: Tuple[int, ...], name: str = "input") -> str:
-        self.add_tensor(name, shape)
+    def input(self, shape: Tuple[int, ...], name: str = "input",
+              dtype: str = "f16") -> str:
+        self.add_tensor(name, shape, dtype)
         self.nodes.append(Node("input", name, [], name))
         self.input_name = name
         return name
@@ -145,6 +146,20 @@ class Graph:
                           dict(gamma=np.asarray(gamma, np.float32),
                                beta=np.asarray(beta, np.float32), eps=eps),
                           name)
+
+    def embedding(self, ids: str, tok_table: np.ndarray,
+                  pos_table: np.ndarray, seg_table=None,
+                  name: Optional[str] = None) -> str:
+        m = self.tensors[ids].shape[0]
+        v, h = tok_table.shape
+        s_, h2 = pos_table.shape
+        assert h == h2
+        return self._emit("embedding", [ids], (m, h),
+                          dict(tok=np.asarray(tok_table, np.float32),
+                               pos=np.asarray(pos_table, np.float32),
+                               seg=None if seg_table is None
+                               else np.asarray(seg_table, np.float32),
+                               seq=s_), name)
 
     def attention(self, qkv: str, heads: int, seq: int,
                   name: Optional[str] = None) -> str:

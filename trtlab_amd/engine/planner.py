@@ -42,6 +42,7 @@ DT_I8 = 2
 
 K_QUANTIZE = 10
 K_DEQUANT = 11
+K_EMBEDDING = 12
 
 
 @dataclass
@@ -75,6 +76,7 @@ class EnginePlan:
     output_shape: Tuple[int, ...]
     dtype: int = DT_F16
     shapes: Dict[str, Tuple[int, ...]] = field(default_factory=dict)
+    input_dtype: str = "f16"
 
 
 class Planner:
@@ -138,6 +140,14 @@ class Planner:
                             dict(eps=n.attrs["eps"]))
                 op.scale = n.attrs["gamma"].astype(np.float32)
                 op.bias = n.attrs["beta"].astype(np.float32)
+                exec_ops.append(op)
+            elif n.kind == "embedding":
+                op = ExecOp(K_EMBEDDING, n.name, [n.inputs[0]], n.output,
+                            dict(seq=n.attrs["seq"]))
+                op.w = n.attrs["tok"].astype(np.float16)
+                op.scale = n.attrs["pos"].astype(np.float16)
+                op.bias = (None if n.attrs["seg"] is None
+                           else n.attrs["seg"].astype(np.float16))
                 exec_ops.append(op)
             elif n.kind == "attention":
                 exec_ops.append(ExecOp(K_ATTENTION, n.name, [n.inputs[0]],
@@ -242,7 +252,10 @@ class Planner:
                              for t in op.inputs]
 
         # ---- weight prepacking (+ int8 lowering) ----
-        itemsize: Dict[str, int] = {}
+        itemsize: Dict[str, int] = {
+            t: {"f16": 2, "bf16": 2, "f32": 4, "i32": 4, "i8": 1}[spec.dtype]
+            for t, spec in g.tensors.items()
+        }
         if self.dtype == DT_I8:
             from trtlab_amd.engine.quantize import lower_int8
 
@@ -375,6 +388,9 @@ class Planner:
             elif op.kind == K_CHANNEL_PAD:
                 d.update(kind=K_CHANNEL_PAD, n_elems=op.params["M"],
                          C=op.params["Cin"], Cout=op.params["Cpad"])
+            elif op.kind == K_EMBEDDING:
+                m, h = shapes[op.output]
+                d.update(kind=K_EMBEDDING, M=m, S=op.params["seq"], N=h)
             elif op.kind in (K_QUANTIZE, K_DEQUANT):
                 n = 1
                 for s_ in shapes[op.output]:
@@ -409,4 +425,5 @@ class Planner:
             output_shape=shapes[output_name],
             dtype=self.dtype,
             shapes=dict(shapes),
+            input_dtype=g.tensors[input_name].dtype,
         )

@@ -17,7 +17,7 @@ import torch.nn.functional as F
 from trtlab_amd.engine.planner import (
     EnginePlan, K_ADD_LAYERNORM, K_ATTENTION, K_CHANNEL_PAD, K_CONV,
     K_DEQUANT, K_ELEMENTWISE, K_GAVGPOOL, K_GEMM, K_LAYERNORM, K_MAXPOOL,
-    K_QUANTIZE, K_SOFTMAX,
+    K_EMBEDDING, K_QUANTIZE, K_SOFTMAX,
     EPI_BIAS, EPI_BIAS_GELU, EPI_BIAS_RELU, EPI_NONE, EPI_SCALE_BIAS,
     EPI_SCALE_BIAS_ADD_RELU, EPI_SCALE_BIAS_RELU)
 
@@ -74,6 +74,15 @@ def run_reference(plan: EnginePlan, input_nhwc: np.ndarray,
             y = _epi(acc, d["epi"], scale, bias, res)
             if d["dtype"] == 2:  # int8: emulate the requantized store
                 y = torch.clamp(torch.round(y), -127, 127)
+            t[op.output] = y
+        elif op.kind == K_EMBEDDING:
+            ids = x.long()
+            tok = torch.from_numpy(op.w.astype(np.float32))
+            pos = torch.from_numpy(op.scale.astype(np.float32))
+            m = ids.shape[0]
+            y = tok[ids] + pos[torch.arange(m) % d["S"]]
+            if op.bias is not None:
+                y = y + torch.from_numpy(op.bias.astype(np.float32))[0]
             t[op.output] = y
         elif op.kind == K_QUANTIZE:
             t[op.output] = torch.clamp(torch.round(x / d["q_scale"]),

@@ -52,8 +52,10 @@ class NativeContext:
         self.engine = engine
         self.plan = engine.plan
         self.ctx = self._C.ExecutionContext(engine.engine)
+        in_dt = {"f16": np.float16, "bf16": np.float16, "i32": np.int32,
+                 "f32": np.float32}[getattr(self.plan, "input_dtype", "f16")]
         self._in_view = np.frombuffer(
-            self.ctx.input_view(self.plan.input_bytes), dtype=np.float16
+            self.ctx.input_view(self.plan.input_bytes), dtype=in_dt
         ).reshape(self.plan.input_shape)
         self._out_view = np.frombuffer(
             self.ctx.output_view(self.plan.output_bytes), dtype=np.float16
@@ -75,7 +77,8 @@ class NativeContext:
         """Synchronous convenience path; returns a COPY of the output (the
         zero-copy `.output` view is only valid while this context lives)."""
         if batch is not None:
-            np.copyto(self._in_view, batch.astype(np.float16, copy=False))
+            np.copyto(self._in_view,
+                      batch.astype(self._in_view.dtype, copy=False))
         self.ctx.launch()
         self.ctx.synchronize()
         return np.array(self._out_view, copy=True)
@@ -123,6 +126,36 @@ class InferenceManager:
     def infer_runner(self, name: str) -> "InferRunner":
         return InferRunner(self, name)
 
+    # ---- the PyInferenceManager surface (reference pybind/trtlab/infer.cc:
+    # register_tensorrt_engine / update_resources / serve) ----
+    def register_onnx(self, name: str, path: str, batch: int = 8,
+                      dtype: int = 0) -> None:
+        from trtlab_amd.engine.onnx_io import load_onnx
+        from trtlab_amd.engine.planner import Planner
+
+        self.register_model(name, Planner(dtype=dtype).compile(
+            load_onnx(path, batch=batch)))
+
+    def register_plan_file(self, name: str, path: str) -> None:
+        from trtlab_amd.engine.plan_io import load_plan
+
+        self.register_model(name, load_plan(path))
+
+    def serve(self, port: int = 50051, metrics_port: int = 0):
+        """Start a gRPC inference service over the registered models;
+        returns the running Server (reference PyInferenceManager::Serve)."""
+        from trtlab_amd.rpc.server import Server
+        from trtlab_amd.rpc.service import InferenceResources, InferenceService
+        from trtlab_amd.utils.metrics import Metrics
+
+        metrics = Metrics.initialize(metrics_port) if metrics_port else None
+        svc = InferenceService(InferenceResources(self), metrics=metrics)
+        server = Server(f"0.0.0.0:{port}")
+        server.register_service(svc)
+        server.register_service(svc.health_service)
+        server.async_start()
+        return server
+
     def shutdown(self):
         for p in self.thread_pools.values():
             p.shutdown()
@@ -145,7 +178,7 @@ class InferRunner:
         def pre():
             co = self._pool.pop()  # blocks: concurrency limiter
             ctx: NativeContext = co.item
-            np.copyto(ctx.input, batch.astype(np.float16, copy=False))
+            np.copyto(ctx.input, batch.astype(ctx.input.dtype, copy=False))
             tp["hip"].enqueue(hip_stage, co)
 
         def hip_stage(co):

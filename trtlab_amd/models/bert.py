@@ -15,7 +15,11 @@ from trtlab_amd.engine.ir import Graph
 
 def build_bert(batch: int = 8, seq: int = 128, hidden: int = 768,
                layers: int = 12, heads: int = 12, seed: int = 0,
-               intermediate: int | None = None) -> Graph:
+               intermediate: int | None = None, embeddings: bool = False,
+               vocab: int = 30522) -> Graph:
+    """embeddings=True: input is int32 token ids [B*S]; the graph starts
+    with an embedding gather (tok+pos) + LayerNorm. Otherwise the input is
+    the pre-embedded hidden state [B*S, hidden] fp16."""
     inter = intermediate or hidden * 4
     rng = np.random.RandomState(seed)
 
@@ -32,8 +36,15 @@ def build_bert(batch: int = 8, seq: int = 128, hidden: int = 768,
 
     g = Graph(f"bert_base_s{seq}_b{batch}")
     m = batch * seq
-    x = g.input((m, hidden), name="hidden_in")
-    h = x
+    if embeddings:
+        ids = g.input((m,), name="token_ids", dtype="i32")
+        tok = (rng.randn(vocab, hidden) * 0.02).astype(np.float32)
+        pos = (rng.randn(seq, hidden) * 0.02).astype(np.float32)
+        emb = g.embedding(ids, tok, pos, name="embed")
+        ge, be = ln(hidden)
+        h = g.layernorm(emb, ge, be, name="embed_ln")
+    else:
+        h = g.input((m, hidden), name="hidden_in")
     for li in range(layers):
         qkv = g.gemm(h, w(3 * hidden, hidden), b(3 * hidden),
                      name=f"l{li}_qkv")
