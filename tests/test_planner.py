@@ -59,13 +59,6 @@ def test_arena_liveness_safe(rn50_plan):
     s, e = touched[plan.output_name]
     touched[plan.output_name] = (s, len(plan.exec_ops))
 
-    def nbytes(t):
-        # find a shape: inputs/outputs of ops
-        import math
-        for op, d in zip(plan.exec_ops, plan.ops):
-            pass
-        return None
-
     names = list(touched)
     for i, n1 in enumerate(names):
         for n2 in names[i + 1:]:

@@ -87,6 +87,10 @@ class Planner:
 
     def __init__(self, dtype: int = DT_F16, reuse: bool = True,
                  calib_sample=None):
+        if dtype == DT_BF16:
+            raise NotImplementedError(
+                "bf16 engine plans: numpy has no bf16 for the weight blob; "
+                "bf16 kernels are available through the raw ops API")
         self.dtype = dtype
         self.reuse = reuse  # False: disjoint arena slots (debugging)
         self.calib_sample = calib_sample  # int8 activation calibration input
@@ -95,7 +99,6 @@ class Planner:
     def fuse(self, g: Graph) -> List[ExecOp]:
         nodes = g.nodes
         consumed: set = set()
-        by_output: Dict[str, Node] = {n.output: n for n in nodes}
         exec_ops: List[ExecOp] = []
 
         def single_user(t: str) -> Optional[Node]:
