@@ -225,7 +225,11 @@ __device__ __forceinline__ void wait_tiles_inflight(int ahead) {
 // the CU count) can't hide the per-K-tile staging latency across blocks,
 // so spend 4x LDS on a 3-deep in-block pipeline instead.
 inline bool want_deep_pipe(long blocks, int ktiles) {
-  return blocks < 384 && ktiles >= 2;
+  static const long gate = [] {
+    const char* e = getenv("TRTLAB_DEEP_GATE");  // tuning experiments
+    return e ? atol(e) : 384L;
+  }();
+  return blocks < gate && ktiles >= 2;
 }
 
 // Split-K slab store: this (tile, slice)'s [BM][BN] fp32 partial sums.

@@ -111,15 +111,17 @@ __device__ __forceinline__ void layernorm_row(
   float rstd = rsqrtf(q / (float)N + eps);
   nc = 0;
   for (int c = lane; c * 8 < N; c += 64, ++nc) {
-    float g8[8], b8[8];
+    float4v g0 = *(const float4v*)(gamma + c * 8);
+    float4v g1 = *(const float4v*)(gamma + c * 8 + 4);
+    float4v b0 = *(const float4v*)(beta + c * 8);
+    float4v b1 = *(const float4v*)(beta + c * 8 + 4);
 #pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      g8[j] = gamma[c * 8 + j];
-      b8[j] = beta[c * 8 + j];
+    for (int j = 0; j < 4; ++j) {
+      v[nc][j] = (v[nc][j] - mean) * rstd * ((const float*)&g0)[j] +
+                 ((const float*)&b0)[j];
+      v[nc][4 + j] = (v[nc][4 + j] - mean) * rstd * ((const float*)&g1)[j] +
+                     ((const float*)&b1)[j];
     }
-#pragma unroll
-    for (int j = 0; j < 8; ++j)
-      v[nc][j] = (v[nc][j] - mean) * rstd * g8[j] + b8[j];
     store8(dst + c * 8, v[nc]);
   }
 }
