@@ -5,15 +5,16 @@
 // Input: qkv [B*S, 3*H*D] (the fused QKV projection output, column blocks
 // q|k|v, each H*D with head-major (h,d) minor d). Output: [B*S, H*D].
 //
-// One workgroup per (b, h): Q,K,V head tiles live entirely in LDS
-// (S=128, D=64 -> 16 KiB each). 4 waves; each wave owns 32 query rows:
+// One workgroup per (b, h, 64-query-row block) — 2 blocks per head double
+// the grid (B=8, H=12 -> 192 workgroups) since seq-128 heads are small on
+// a 256-CU chip. K and V head tiles live in LDS (staged per block; tiny
+// and L2-resident); 4 waves own 16 query rows each:
 //   QK^T via mfma_f32_16x16x32_f16 (K tile is already "bt" layout),
-//   row softmax fully wave-local (rows live on 16-lane groups; shfl_xor),
-//   P staged to per-wave LDS as two [32][64] tiles, PV against V^T staged
-//   transposed at load. fp32 accumulation throughout; scores scaled by
-//   1/sqrt(D). No attention mask (synthetic full-length sequences; masked
-//   variant planned).
-// Constraints: S == 128, D == 64.
+//   row softmax wave-local (rows on 16-lane groups; shfl_xor),
+//   P staged to per-wave LDS as two [16][64] tiles, PV against V^T staged
+//   transposed at load. fp32 accumulation; scores scaled by 1/sqrt(D).
+// No attention mask (synthetic full-length sequences; masked variant
+// planned). Constraints: S == 128, D == 64.
 #include "gemm_common.h"
 
 namespace trtlab {
@@ -22,21 +23,23 @@ template <typename T>
 __global__ __launch_bounds__(256) void attention_kernel(
     const T* __restrict__ qkv, T* __restrict__ out, int B, int S, int H,
     int D, float scale) {
-  // LDS: Q [128][64] | K [128][64] | Vt [64][128->2x[64][64]] | P 4x[2x[32][64]]
-  __shared__ __attribute__((aligned(16))) char smem[16384 * 3 + 4 * 8192];
-  char* Qs = smem;
-  char* Ks = smem + 16384;
-  char* Vt = smem + 32768;          // two [64][64] tiles: kt*8192
-  char* Ps = smem + 49152;          // per wave: wave*8192, two [32][64] tiles
+  // LDS: Q [64][64] | K [128][64] | Vt 2x[64][64] | P 4 waves x 2x[16][64]
+  __shared__ __attribute__((aligned(16))) char smem[8192 + 16384 * 2 + 16384];
+  char* Qs = smem;                  // 8 KiB
+  char* Ks = smem + 8192;          // 16 KiB
+  char* Vt = smem + 8192 + 16384;  // two [64][64] tiles: kt*8192
+  char* Ps = Vt + 16384;           // per wave: wave*4096, two [16][64] tiles
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
-  const int bh = blockIdx.x;
+  const int blk = blockIdx.x;
+  const int bh = blk >> 1;         // (b, h)
+  const int q0 = (blk & 1) * 64;   // this block's query-row base
   const int b = bh / H;
   const int h = bh % H;
-  const int hid = H * D;           // 768 for BERT-base
-  const int row_stride = 3 * hid;  // qkv row stride in elements
+  const int hid = H * D;
+  const int row_stride = 3 * hid;
 
   const T* base = qkv + (int64_t)b * S * row_stride;
   const int qoff = h * D;
@@ -47,20 +50,26 @@ __global__ __launch_bounds__(256) void attention_kernel(
     return row * 128 + (colbyte ^ ((row & 7) << 4));
   };
 
-  // ---- stage Q, K (swizzled 16-B chunks), Vt (transposed scatter) ----
-  // Q/K: 128 rows x 128 B/row -> 1024 16-B chunks each, 4 per thread.
+  // ---- stage Q (64 rows), K (128 rows), Vt (transposed scatter) ----
+  // Q: 64 rows x 8 chunks = 512 chunks, 2/thread.
+#pragma unroll
+  for (int c = 0; c < 2; ++c) {
+    int idx = c * 256 + tid;       // 0..511
+    int row = idx >> 3;            // 0..63
+    int cb = (idx & 7) * 16;
+    const T* src = base + (int64_t)(q0 + row) * row_stride + qoff + cb / 2;
+    *(short8v*)(Qs + swz(row, cb)) = *(const short8v*)src;
+  }
+  // K: 128 rows x 8 chunks = 1024 chunks, 4/thread.
 #pragma unroll
   for (int c = 0; c < 4; ++c) {
-    int idx = c * 256 + tid;       // 0..1023
+    int idx = c * 256 + tid;
     int row = idx >> 3;            // 0..127
-    int cb = (idx & 7) * 16;       // byte offset within 128-B row
-    const T* src = base + (int64_t)row * row_stride + qoff + cb / 2;
-    *(short8v*)(Qs + swz(row, cb)) = *(const short8v*)src;
+    int cb = (idx & 7) * 16;
     const T* ksrc = base + (int64_t)row * row_stride + koff + cb / 2;
     *(short8v*)(Ks + swz(row, cb)) = *(const short8v*)ksrc;
   }
   // Vt: read v[key][dd..dd+8] (16 B), scatter transposed to Vt[kt][d][key%64].
-  // 128 keys x 8 chunks = 1024 chunks, 4 per thread.
 #pragma unroll
   for (int c = 0; c < 4; ++c) {
     int ci = c * 256 + tid;        // 0..1023
@@ -77,123 +86,97 @@ __global__ __launch_bounds__(256) void attention_kernel(
   }
   __syncthreads();
 
-  // ---- QK^T for this wave's 32 query rows ----
+  // ---- QK^T for this wave's 16 query rows ----
   using MF = Mfma16x16x32<T>;
-  const int qrow0 = wave * 32;
-  f32x4 sacc[2][8];
+  const int qrow = wave * 16;      // within the block's 64-row slice
+  f32x4 sacc[8];
 #pragma unroll
-  for (int i = 0; i < 2; ++i)
-#pragma unroll
-    for (int j = 0; j < 8; ++j) sacc[i][j] = {0.f, 0.f, 0.f, 0.f};
+  for (int j = 0; j < 8; ++j) sacc[j] = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
   for (int ks = 0; ks < 2; ++ks) {
     uint32_t kbyte = ks * 64 + ((lane >> 4) << 4);
-    typename MF::frag qf[2], kf[8];
-#pragma unroll
-    for (int f = 0; f < 2; ++f)
-      qf[f] = *(const typename MF::frag*)(Qs +
-               swz(qrow0 + f * 16 + (lane & 15), kbyte));
+    typename MF::frag qf =
+        *(const typename MF::frag*)(Qs + swz(qrow + (lane & 15), kbyte));
+    typename MF::frag kf[8];
 #pragma unroll
     for (int f = 0; f < 8; ++f)
       kf[f] = *(const typename MF::frag*)(Ks + swz(f * 16 + (lane & 15), kbyte));
 #pragma unroll
-    for (int i = 0; i < 2; ++i)
-#pragma unroll
-      for (int j = 0; j < 8; ++j)
-        sacc[i][j] = MF::run(qf[i], kf[j], sacc[i][j]);
+    for (int j = 0; j < 8; ++j) sacc[j] = MF::run(qf, kf[j], sacc[j]);
   }
 
   // ---- row softmax (rows live on 16-lane groups: shfl_xor 1,2,4,8) ----
 #pragma unroll
-  for (int i = 0; i < 2; ++i) {
+  for (int r = 0; r < 4; ++r) {
+    float m = -3.0e38f;
 #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      float m = -3.0e38f;
+    for (int j = 0; j < 8; ++j) m = fmaxf(m, sacc[j][r] * scale);
 #pragma unroll
-      for (int j = 0; j < 8; ++j) m = fmaxf(m, sacc[i][j][r] * scale);
-#pragma unroll
-      for (int off = 1; off < 16; off <<= 1)
-        m = fmaxf(m, __shfl_xor(m, off, 64));
-      float s = 0.f;
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        float e = __expf(sacc[i][j][r] * scale - m);
-        sacc[i][j][r] = e;
-        s += e;
-      }
-#pragma unroll
-      for (int off = 1; off < 16; off <<= 1) s += __shfl_xor(s, off, 64);
-      float inv = 1.0f / s;
-#pragma unroll
-      for (int j = 0; j < 8; ++j) sacc[i][j][r] *= inv;
-    }
-  }
-
-  // ---- P -> per-wave LDS as two [32][64] fp16 tiles ----
-  char* P = Ps + wave * 8192;
-#pragma unroll
-  for (int i = 0; i < 2; ++i) {
+    for (int off = 1; off < 16; off <<= 1)
+      m = fmaxf(m, __shfl_xor(m, off, 64));
+    float s = 0.f;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      int colg = j * 16 + (lane & 15);
-      char* tile = P + (colg >> 6) * 4096;
-      int col = colg & 63;
+      float e = __expf(sacc[j][r] * scale - m);
+      sacc[j][r] = e;
+      s += e;
+    }
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        int row = i * 16 + ((lane >> 4) << 2) + r;
-        *(T*)(tile + ((uint32_t)row * 128 + ((col * 2) ^ ((row & 7) << 4)))) =
-            (T)sacc[i][j][r];
-      }
+    for (int off = 1; off < 16; off <<= 1) s += __shfl_xor(s, off, 64);
+    float inv = 1.0f / s;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) sacc[j][r] *= inv;
+  }
+
+  // ---- P -> per-wave LDS as two [16][64] fp16 tiles ----
+  char* P = Ps + wave * 4096;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    int colg = j * 16 + (lane & 15);
+    char* tile = P + (colg >> 6) * 2048;
+    int col = colg & 63;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int row = ((lane >> 4) << 2) + r;  // 0..15
+      *(T*)(tile + ((uint32_t)row * 128 + ((col * 2) ^ ((row & 7) << 4)))) =
+          (T)sacc[j][r];
     }
   }
-  // Force completion of the P ds_writes before the PV ds_reads. A bare
-  // same-wave write->read *should* be ordered by the compiler's lgkmcnt
-  // bookkeeping, but was observed to return stale P for some blocks on
-  // real data (one (b,h) block NaN per launch); the barrier makes the
-  // ordering explicit and costs nothing at this kernel's size.
+  // Make the P ds_writes visible before the PV ds_reads (same wave, but
+  // the explicit barrier proved necessary — see git history).
   __syncthreads();
 
-  // ---- PV: out_tile[32 rows][64 d] = P[32][128] @ Vt^T ----
-  f32x4 oacc[2][4];
+  // ---- PV: out_tile[16 rows][64 d] = P[16][128] @ Vt^T ----
+  f32x4 oacc[4];
 #pragma unroll
-  for (int i = 0; i < 2; ++i)
-#pragma unroll
-    for (int j = 0; j < 4; ++j) oacc[i][j] = {0.f, 0.f, 0.f, 0.f};
+  for (int j = 0; j < 4; ++j) oacc[j] = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
   for (int kt = 0; kt < 2; ++kt) {
-    const char* Pt = P + kt * 4096;
+    const char* Pt = P + kt * 2048;
     const char* Vk = Vt + kt * 8192;
 #pragma unroll
     for (int ks = 0; ks < 2; ++ks) {
       uint32_t kbyte = ks * 64 + ((lane >> 4) << 4);
-      typename MF::frag pf[2], vf[4];
-#pragma unroll
-      for (int f = 0; f < 2; ++f)
-        pf[f] = *(const typename MF::frag*)(Pt +
-                 swz(f * 16 + (lane & 15), kbyte));
+      typename MF::frag pf =
+          *(const typename MF::frag*)(Pt + swz(lane & 15, kbyte));
+      typename MF::frag vf[4];
 #pragma unroll
       for (int f = 0; f < 4; ++f)
         vf[f] = *(const typename MF::frag*)(Vk +
                  swz(f * 16 + (lane & 15), kbyte));
 #pragma unroll
-      for (int i = 0; i < 2; ++i)
-#pragma unroll
-        for (int j = 0; j < 4; ++j)
-          oacc[i][j] = MF::run(pf[i], vf[j], oacc[i][j]);
+      for (int j = 0; j < 4; ++j) oacc[j] = MF::run(pf, vf[j], oacc[j]);
     }
   }
 
-  // ---- store out[b*S + row][h*D + d] ----
+  // ---- store out[b*S + q0 + row][h*D + d] ----
 #pragma unroll
-  for (int i = 0; i < 2; ++i) {
+  for (int j = 0; j < 4; ++j) {
+    int d = j * 16 + (lane & 15);
 #pragma unroll
-    for (int j = 0; j < 4; ++j) {
-      int d = j * 16 + (lane & 15);
-#pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        int row = qrow0 + i * 16 + ((lane >> 4) << 2) + r;
-        out[((int64_t)b * S + row) * hid + h * D + d] = (T)oacc[i][j][r];
-      }
+    for (int r = 0; r < 4; ++r) {
+      int row = q0 + qrow + ((lane >> 4) << 2) + r;
+      out[((int64_t)b * S + row) * hid + h * D + d] = (T)oacc[j][r];
     }
   }
 }
@@ -218,7 +201,7 @@ void launch_attention(int dtype, const void* qkv, void* out, int B, int S,
                       int H, int D, float scale, hipStream_t stream) {
   if (S != 128 || D != 64)
     throw std::runtime_error("attention: only S=128, D=64 supported (BERT-base seq128)");
-  dim3 grid(B * H);
+  dim3 grid(B * H * 2);  // 2 query-row blocks per head
   dim3 block(256);
   if (dtype == 0)
     hipLaunchKernelGGL((attention_kernel<_Float16>), grid, block, 0, stream,
