@@ -108,3 +108,34 @@ def test_inference_service_gpu_end_to_end():
     finally:
         server.shutdown()
         mgr.shutdown()
+
+
+def test_metrics_multiple_instances():
+    from trtlab_amd.utils.metrics import Metrics
+
+    m1 = Metrics(port=0)
+    m2 = Metrics(port=0)  # own registries: no duplicate-collector error
+    m1.observe(1.0, 2.0)
+    m2.observe(1.0, 2.0)
+
+
+@pytest.mark.gpu
+def test_multi_model_manager():
+    """Two models behind one InferenceManager (reference RegisterModel xN)."""
+    from trtlab_amd.engine.planner import Planner
+    from trtlab_amd.engine.runtime import InferenceManager
+    from trtlab_amd.models import build_resnet
+
+    g1 = build_resnet(50, batch=1, image=64, seed=0)
+    g2 = build_resnet(50, batch=2, image=64, seed=1)
+    mgr = InferenceManager(max_contexts=1)
+    mgr.register_model("a", Planner().compile(g1))
+    mgr.register_model("b", Planner().compile(g2))
+    mgr.allocate_resources()
+    ra, rb = mgr.infer_runner("a"), mgr.infer_runner("b")
+    xa = np.random.RandomState(0).randn(1, 64, 64, 3).astype(np.float32)
+    xb = np.random.RandomState(1).randn(2, 64, 64, 3).astype(np.float32)
+    oa = ra.infer(xa).result(60)
+    ob = rb.infer(xb).result(60)
+    assert oa.shape == (1, 1000) and ob.shape == (2, 1000)
+    mgr.shutdown()

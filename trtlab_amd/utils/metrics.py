@@ -6,7 +6,8 @@ from __future__ import annotations
 import subprocess
 from typing import Optional
 
-from prometheus_client import Gauge, Histogram, start_http_server
+from prometheus_client import (CollectorRegistry, Gauge, Histogram,
+                               start_http_server)
 
 _LOAD_RATIO_BUCKETS = (1.25, 1.5, 2.0, 10.0, 100.0)  # reference server.cc:96
 
@@ -15,17 +16,23 @@ class Metrics:
     _instance: Optional["Metrics"] = None
 
     def __init__(self, port: int = 50078):
+        # own registry: instantiable more than once per process (tests,
+        # multiple servers)
+        self.registry = CollectorRegistry()
         self.compute_ms = Histogram(
             "trtlab_compute_duration_ms", "GPU compute duration per request",
-            buckets=(1, 2, 3, 5, 8, 13, 21, 34, 55, 89))
+            buckets=(1, 2, 3, 5, 8, 13, 21, 34, 55, 89),
+            registry=self.registry)
         self.request_ms = Histogram(
             "trtlab_request_duration_ms", "total request duration",
-            buckets=(1, 2, 3, 5, 8, 13, 21, 34, 55, 89, 144, 233))
+            buckets=(1, 2, 3, 5, 8, 13, 21, 34, 55, 89, 144, 233),
+            registry=self.registry)
         self.load_ratio = Histogram(
             "trtlab_load_ratio", "request/compute duration ratio "
             "(queueing pressure — the reference's autoscaling signal)",
-            buckets=_LOAD_RATIO_BUCKETS)
-        self.power_w = Gauge("trtlab_gpu_power_watts", "GPU power draw")
+            buckets=_LOAD_RATIO_BUCKETS, registry=self.registry)
+        self.power_w = Gauge("trtlab_gpu_power_watts", "GPU power draw",
+                             registry=self.registry)
         self.port = port
         self._started = False
 
@@ -33,7 +40,7 @@ class Metrics:
     def initialize(cls, port: int = 50078) -> "Metrics":
         if cls._instance is None:
             cls._instance = cls(port)
-            start_http_server(port)
+            start_http_server(port, registry=cls._instance.registry)
             cls._instance._started = True
         return cls._instance
 
