@@ -51,6 +51,8 @@ def main():
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
         torch.cuda.set_device(local_rank)
         dist.init_process_group("nccl")
+        # N ranks calibrate the same plan concurrently on CPU: share cores
+        torch.set_num_threads(max(1, (os.cpu_count() or world) // world))
 
     import trtlab_amd
     from trtlab_amd.engine.planner import DT_F16, DT_I8, Planner
