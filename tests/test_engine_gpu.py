@@ -171,3 +171,24 @@ def test_engine_bert_with_embeddings():
     scale = max(np.abs(ref).max(), 1e-6)
     assert np.isfinite(out).all()
     assert err / scale < 0.08, (err, scale)
+
+
+def test_engine_from_onnx_roundtrip():
+    """ONNX bytes -> import -> plan -> GPU engine matches the source graph's
+    reference output (the reference's ONNX build.py -> engine flow)."""
+    from trtlab_amd.engine.onnx_io import export_onnx, import_onnx
+    from trtlab_amd.engine.planner import Planner
+    from trtlab_amd.engine.reference import run_reference
+    from trtlab_amd.engine.runtime import NativeEngine
+    from trtlab_amd.models import build_resnet
+
+    g = build_resnet(50, batch=1, image=64, seed=0)
+    g2 = import_onnx(export_onnx(g), name="onnx_rt")
+    plan = Planner().compile(g2)
+    eng = NativeEngine(plan)
+    ctx = eng.create_context(capture=True)
+    x = np.random.RandomState(21).randn(*plan.input_shape).astype(np.float32) * 0.5
+    out = ctx.infer(x).astype(np.float32)
+    ref = run_reference(plan, x)
+    err = np.abs(out - ref).max() / max(np.abs(ref).max(), 1e-6)
+    assert err < 0.08, err

@@ -30,3 +30,66 @@ def test_onnx_import_batch_override():
     assert g2.tensors[g2.input_name].shape[0] == 4
     plan = Planner().compile(g2)
     assert plan.input_shape[0] == 4
+
+
+def test_wire_format_against_protobuf_runtime():
+    """Cross-check our hand-rolled protobuf wire codec against the official
+    google.protobuf runtime: parse export_onnx() bytes with a dynamically
+    declared (minimal) onnx schema."""
+    from google.protobuf import descriptor_pb2, descriptor_pool, message_factory
+
+    pool = descriptor_pool.DescriptorPool()  # private pool
+    f = descriptor_pb2.FileDescriptorProto()
+    f.name = "mini_onnx.proto"
+    f.package = "monnx"
+    f.syntax = "proto3"
+    T = descriptor_pb2.FieldDescriptorProto
+
+    def msg(name, fields):
+        m = f.message_type.add()
+        m.name = name
+        for num, fname, ftype, label, tname in fields:
+            fd = m.field.add()
+            fd.name = fname
+            fd.number = num
+            fd.type = ftype
+            fd.label = label
+            if tname:
+                fd.type_name = f".monnx.{tname}"
+
+    R, O = T.LABEL_REPEATED, T.LABEL_OPTIONAL
+    msg("Attr", [(1, "name", T.TYPE_STRING, O, None),
+                 (2, "f", T.TYPE_FLOAT, O, None),
+                 (3, "i", T.TYPE_INT64, O, None),
+                 (8, "ints", T.TYPE_INT64, R, None)])
+    msg("Node", [(1, "input", T.TYPE_STRING, R, None),
+                 (2, "output", T.TYPE_STRING, R, None),
+                 (4, "op_type", T.TYPE_STRING, O, None),
+                 (5, "attribute", T.TYPE_MESSAGE, R, "Attr")])
+    msg("Tensor", [(1, "dims", T.TYPE_INT64, R, None),
+                   (2, "data_type", T.TYPE_INT32, O, None),
+                   (8, "name", T.TYPE_STRING, O, None),
+                   (9, "raw_data", T.TYPE_BYTES, O, None)])
+    msg("Graph", [(1, "node", T.TYPE_MESSAGE, R, "Node"),
+                  (2, "name", T.TYPE_STRING, O, None),
+                  (5, "initializer", T.TYPE_MESSAGE, R, "Tensor")])
+    msg("Model", [(1, "ir_version", T.TYPE_INT64, O, None),
+                  (7, "graph", T.TYPE_MESSAGE, O, "Graph")])
+    pool.Add(f)
+    Model = message_factory.GetMessageClass(
+        pool.FindMessageTypeByName("monnx.Model"))
+
+    g = build_resnet(50, batch=1, image=64, seed=0, calibrate=False)
+    data = export_onnx(g)
+    model = Model.FromString(data)
+    ops = [n.op_type for n in model.graph.node]
+    assert ops.count("Conv") == 53
+    assert ops.count("BatchNormalization") == 53
+    assert ops.count("Gemm") == 1
+    assert len(model.graph.initializer) > 100
+    conv0 = next(n for n in model.graph.node if n.op_type == "Conv")
+    strides = next(a for a in conv0.attribute if a.name == "strides")
+    assert list(strides.ints) == [2, 2]
+    w0 = model.graph.initializer[0]
+    assert list(w0.dims) == [64, 3, 7, 7]
+    assert len(w0.raw_data) == 64 * 3 * 49 * 4
