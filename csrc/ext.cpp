@@ -342,6 +342,12 @@ PYBIND11_MODULE(_C, m) {
              c.synchronize();
            })
       .def("ready", &ExecutionContext::ready)
+      .def("set_timing", &ExecutionContext::set_timing)
+      .def("stage_times_ms",
+           [](ExecutionContext& c) {
+             auto t = c.stage_times_ms();
+             return py::make_tuple(t[0], t[1], t[2]);
+           })
       .def_property_readonly("host_input_ptr", &ExecutionContext::host_input_ptr)
       .def_property_readonly("host_output_ptr", &ExecutionContext::host_output_ptr)
       .def_property_readonly("arena_ptr", &ExecutionContext::arena_ptr)

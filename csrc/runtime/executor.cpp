@@ -61,6 +61,8 @@ ExecutionContext::ExecutionContext(std::shared_ptr<Engine> engine)
 }
 
 ExecutionContext::~ExecutionContext() {
+  for (auto& e : ev_)
+    if (e) hipEventDestroy(e);
   if (graph_exec_) hipGraphExecDestroy(graph_exec_);
   if (graph_) hipGraphDestroy(graph_);
   hipStreamDestroy(stream_);
@@ -81,8 +83,12 @@ void ExecutionContext::enqueue_all(hipStream_t s) {
     return off < 0 ? nullptr : (const float*)(wb + off);
   };
 
+  if (timing_ && !ev_[0])
+    for (auto& e : ev_) TRT_HIP_CHECK(hipEventCreate(&e));
+  if (timing_) TRT_HIP_CHECK(hipEventRecord(ev_[0], s));
   TRT_HIP_CHECK(hipMemcpyAsync(ar + eng_->input_off(), h_in_,
                                eng_->input_bytes(), hipMemcpyHostToDevice, s));
+  if (timing_) TRT_HIP_CHECK(hipEventRecord(ev_[1], s));
 
   for (const OpDesc& op : eng_->ops()) {
     switch (op.kind) {
@@ -154,8 +160,18 @@ void ExecutionContext::enqueue_all(hipStream_t s) {
     }
   }
 
+  if (timing_) TRT_HIP_CHECK(hipEventRecord(ev_[2], s));
   TRT_HIP_CHECK(hipMemcpyAsync(h_out_, ar + eng_->output_off(),
                                eng_->output_bytes(), hipMemcpyDeviceToHost, s));
+  if (timing_) TRT_HIP_CHECK(hipEventRecord(ev_[3], s));
+}
+
+std::array<float, 3> ExecutionContext::stage_times_ms() const {
+  std::array<float, 3> out = {0.f, 0.f, 0.f};
+  if (!ev_[0]) return out;
+  for (int i = 0; i < 3; ++i)
+    TRT_HIP_CHECK(hipEventElapsedTime(&out[i], ev_[i], ev_[i + 1]));
+  return out;
 }
 
 void ExecutionContext::capture() {

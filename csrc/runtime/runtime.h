@@ -2,6 +2,7 @@
 #pragma once
 #include "../common.h"
 
+#include <array>
 #include <atomic>
 #include <memory>
 #include <mutex>
@@ -135,6 +136,12 @@ class ExecutionContext {
   void synchronize();
   bool ready();  // hipStreamQuery == success
 
+  // Per-stage timing (reference TimedBenchmarkWorkspace,
+  // workspace.cc:128-168: 4 events bracket H2D / compute / D2H).
+  // Events are recorded inside the captured graph; call after synchronize.
+  void set_timing(bool enable) { timing_ = enable; }
+  std::array<float, 3> stage_times_ms() const;
+
  private:
   std::shared_ptr<Engine> eng_;
   char* arena_ = nullptr;
@@ -145,6 +152,8 @@ class ExecutionContext {
   hipGraph_t graph_{};
   hipGraphExec_t graph_exec_{};
   bool captured_ = false;
+  bool timing_ = false;
+  hipEvent_t ev_[4] = {nullptr, nullptr, nullptr, nullptr};
 };
 
 }  // namespace trtlab

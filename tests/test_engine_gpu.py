@@ -192,3 +192,18 @@ def test_engine_from_onnx_roundtrip():
     ref = run_reference(plan, x)
     err = np.abs(out - ref).max() / max(np.abs(ref).max(), 1e-6)
     assert err < 0.08, err
+
+
+def test_timed_context_stage_split(rn50_small):
+    """Per-stage H2D/compute/D2H timing inside the captured graph
+    (reference TimedBenchmarkWorkspace, workspace.cc:128-168)."""
+    from trtlab_amd.engine.runtime import NativeEngine
+
+    eng = NativeEngine(rn50_small)
+    ctx = eng.create_context(capture=True, timed=True)
+    x = np.random.RandomState(4).randn(*rn50_small.input_shape).astype(np.float32)
+    ctx.infer(x)
+    h2d, compute, d2h = ctx.stage_times_ms()
+    assert h2d > 0 and compute > 0 and d2h >= 0
+    assert compute > h2d  # forward dominates a 64px batch-2 run
+    assert compute < 50

@@ -40,18 +40,24 @@ class NativeEngine:
     def upload_weights(self, blob: np.ndarray) -> None:
         self.engine.upload_weights(blob)
 
-    def create_context(self, capture: bool = True) -> "NativeContext":
-        return NativeContext(self, capture=capture)
+    def create_context(self, capture: bool = True,
+                       timed: bool = False) -> "NativeContext":
+        return NativeContext(self, capture=capture, timed=timed)
 
 
 class NativeContext:
     """Private stream + activation arena + pinned bindings + hipGraph."""
 
-    def __init__(self, engine: NativeEngine, capture: bool = True):
+    def __init__(self, engine: NativeEngine, capture: bool = True,
+                 timed: bool = False):
         self._C = native()
         self.engine = engine
         self.plan = engine.plan
         self.ctx = self._C.ExecutionContext(engine.engine)
+        if timed:
+            # per-stage H2D/compute/D2H events (reference
+            # TimedBenchmarkWorkspace); recorded inside the captured graph
+            self.ctx.set_timing(True)
         in_dt = {"f16": np.float16, "bf16": np.float16, "i32": np.int32,
                  "f32": np.float32}[getattr(self.plan, "input_dtype", "f16")]
         self._in_view = np.frombuffer(
@@ -91,6 +97,11 @@ class NativeContext:
 
     def ready(self) -> bool:
         return self.ctx.ready()
+
+    def stage_times_ms(self) -> tuple:
+        """(h2d_ms, compute_ms, d2h_ms) of the last completed launch
+        (requires timed=True)."""
+        return self.ctx.stage_times_ms()
 
 
 class InferenceManager:
