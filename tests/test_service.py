@@ -139,3 +139,19 @@ def test_multi_model_manager():
     ob = rb.infer(xb).result(60)
     assert oa.shape == (1, 1000) and ob.shape == (2, 1000)
     mgr.shutdown()
+
+
+def test_shm_zero_copy_input(service_server):
+    """POSIX shared-memory input path (reference SharedMemoryService)."""
+    from trtlab_amd.rpc import ShmInput
+
+    c = SyncClient(f"127.0.0.1:{service_server.port}")
+    x = np.arange(8, dtype=np.float16).reshape(2, 4)
+    with ShmInput(x) as shm:
+        resp = c.call("trtlab.Inference", "Compute",
+                      InferRequest(model="m", shape=[2, 4], dtype="f16",
+                                   shm_name=shm.name, shm_size=shm.size),
+                      InferResponse, timeout=10)
+    out = np.frombuffer(resp.output, dtype=np.float16).reshape(2, 4)
+    assert np.allclose(out, x * 2)
+    c.close()

@@ -9,4 +9,5 @@ from trtlab_amd.rpc.proto import (EchoRequest, EchoResponse, InferRequest,  # no
                                   InferResponse, HealthRequest, HealthResponse)
 from trtlab_amd.rpc.server import (AsyncService, BatchingService, Server,  # noqa: F401
                                    StreamingService, UnaryService)
-from trtlab_amd.rpc.client import AsyncClient, SyncClient, siege  # noqa: F401
+from trtlab_amd.rpc.client import (AsyncClient, ShmInput, SyncClient,  # noqa: F401
+                                   siege)

@@ -13,6 +13,36 @@ import grpc
 import grpc.aio
 
 
+class ShmInput:
+    """Zero-copy local input: place the batch in POSIX shared memory and
+    reference it by name in InferRequest.shm_name (reference
+    SharedMemoryService / sysv_allocator.cc). Use as a context manager."""
+
+    def __init__(self, batch):
+        from multiprocessing import shared_memory
+        import numpy as np
+
+        arr = np.ascontiguousarray(batch)
+        self.shm = shared_memory.SharedMemory(create=True, size=arr.nbytes)
+        self.shm.buf[:arr.nbytes] = arr.tobytes()
+        self.name = self.shm.name
+        self.size = arr.nbytes
+
+    def __enter__(self):
+        return self
+
+    def __exit__(self, *exc):
+        self.close()
+        return False
+
+    def close(self):
+        self.shm.close()
+        try:
+            self.shm.unlink()
+        except FileNotFoundError:
+            pass
+
+
 class SyncClient:
     """Blocking unary client over a shared channel."""
 

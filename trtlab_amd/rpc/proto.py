@@ -10,7 +10,8 @@ surfaces.
     message HealthResponse{ bool ready = 1; string status = 2; }
     message InferRequest  { string model = 1; bytes input = 2;
                             repeated int64 shape = 3; string dtype = 4;
-                            int64 batch_id = 5; }
+                            int64 batch_id = 5;
+                            string shm_name = 6; int64 shm_size = 7; }
     message InferResponse { bytes output = 1; repeated int64 shape = 2;
                             string dtype = 3; int64 batch_id = 4;
                             float compute_ms = 5; float request_ms = 6; }
@@ -51,7 +52,12 @@ _msg("InferRequest", [("model", _T.TYPE_STRING, False),
                       ("input", _T.TYPE_BYTES, False),
                       ("shape", _T.TYPE_INT64, True),
                       ("dtype", _T.TYPE_STRING, False),
-                      ("batch_id", _T.TYPE_INT64, False)])
+                      ("batch_id", _T.TYPE_INT64, False),
+                      # zero-copy local input: POSIX shared memory name
+                      # (reference SysV SharedMemoryService,
+                      # testing.proto:37-48)
+                      ("shm_name", _T.TYPE_STRING, False),
+                      ("shm_size", _T.TYPE_INT64, False)])
 _msg("InferResponse", [("output", _T.TYPE_BYTES, False),
                        ("shape", _T.TYPE_INT64, True),
                        ("dtype", _T.TYPE_STRING, False),

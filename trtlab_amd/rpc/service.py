@@ -56,9 +56,21 @@ class InferenceService(AsyncService):
         runner = resources.runner(request.model)
         plan = resources.manager.get_model(request.model).plan
         dtype = np.float16 if request.dtype in ("", "f16", "float16") else np.dtype(request.dtype)
-        batch = np.frombuffer(request.input, dtype=dtype)
         shape = tuple(request.shape) or plan.input_shape
-        batch = batch.reshape(shape)
+        if request.shm_name:
+            # zero-copy local transport: the tensor lives in POSIX shared
+            # memory (reference's SysV shm input path, 02 server.cc:159)
+            from multiprocessing import shared_memory
+
+            shm = shared_memory.SharedMemory(name=request.shm_name)
+            try:
+                batch = np.frombuffer(
+                    shm.buf[:int(request.shm_size)], dtype=dtype
+                ).reshape(shape).copy()
+            finally:
+                shm.close()
+        else:
+            batch = np.frombuffer(request.input, dtype=dtype).reshape(shape)
 
         t_compute = time.monotonic()
         fut = runner.infer(batch)
