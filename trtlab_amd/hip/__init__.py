@@ -69,3 +69,40 @@ class DeviceInfo:
     @staticmethod
     def gcn_arch(device: int = 0) -> str:
         return device_properties(device)["gcn_arch"]
+
+    @staticmethod
+    def numa_node(device: int = 0) -> Optional[int]:
+        """NUMA node of the GPU via sysfs (reference DeviceInfo::Affinity,
+        device_info.cc:66 — NVML GPU<->CPU affinity -> /sys here)."""
+        import glob
+
+        cards = sorted(glob.glob("/sys/class/drm/card*/device/numa_node"))
+        if device < len(cards):
+            try:
+                with open(cards[device]) as fh:
+                    n = int(fh.read().strip())
+                return n if n >= 0 else None
+            except OSError:
+                return None
+        return None
+
+    @staticmethod
+    def cpu_affinity(device: int = 0) -> List[int]:
+        """CPUs local to the GPU's NUMA node (for pinning the pre/post
+        thread pools next to the device)."""
+        node = DeviceInfo.numa_node(device)
+        if node is None:
+            return list(range(len(__import__("os").sched_getaffinity(0))))
+        try:
+            with open(f"/sys/devices/system/node/node{node}/cpulist") as fh:
+                spec = fh.read().strip()
+            cpus: List[int] = []
+            for part in spec.split(","):
+                if "-" in part:
+                    a, b = part.split("-")
+                    cpus.extend(range(int(a), int(b) + 1))
+                elif part:
+                    cpus.append(int(part))
+            return cpus
+        except OSError:
+            return []
