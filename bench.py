@@ -35,7 +35,8 @@ def main():
     ap.add_argument("--contexts", type=int, default=3)
     ap.add_argument("--model", default="resnet50",
                     choices=["resnet50", "resnet101", "resnet152", "bert"])
-    ap.add_argument("--dtype", default="fp16", choices=["fp16", "int8"])
+    ap.add_argument("--dtype", default="fp16",
+                    choices=["fp16", "int8", "fp8"])
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -68,7 +69,9 @@ def main():
         depth = int(args.model.replace("resnet", ""))
         g = build_resnet(depth, batch=args.batch, image=224, seed=0)
         cfg_extra = {"image": 224}
-    dtype = DT_I8 if args.dtype == "int8" else DT_F16
+    from trtlab_amd.engine.planner import DT_F8
+
+    dtype = {"fp16": DT_F16, "int8": DT_I8, "fp8": DT_F8}[args.dtype]
     plan = Planner(dtype=dtype).compile(g)
 
     eng = NativeEngine(plan, device=local_rank)

@@ -261,20 +261,22 @@ PYBIND11_MODULE(_C, m) {
           py::arg("sync") = true);
   ops.def("quantize",
           [](uintptr_t in, uintptr_t out, int64_t n, float scale,
-             uintptr_t stream, bool sync) {
-            launch_quantize((void*)in, (void*)out, n, scale, as_stream(stream));
+             uintptr_t stream, bool sync, int fmt) {
+            launch_quantize((void*)in, (void*)out, n, scale, as_stream(stream),
+                            fmt);
             if (sync) TRT_HIP_CHECK(hipStreamSynchronize(as_stream(stream)));
           },
           py::arg("in"), py::arg("out"), py::arg("n"), py::arg("scale"),
-          py::arg("stream") = 0, py::arg("sync") = true);
+          py::arg("stream") = 0, py::arg("sync") = true, py::arg("fmt") = 0);
   ops.def("dequant",
           [](uintptr_t in, uintptr_t out, int64_t n, float scale,
-             uintptr_t stream, bool sync) {
-            launch_dequant((void*)in, (void*)out, n, scale, as_stream(stream));
+             uintptr_t stream, bool sync, int fmt) {
+            launch_dequant((void*)in, (void*)out, n, scale, as_stream(stream),
+                           fmt);
             if (sync) TRT_HIP_CHECK(hipStreamSynchronize(as_stream(stream)));
           },
           py::arg("in"), py::arg("out"), py::arg("n"), py::arg("scale"),
-          py::arg("stream") = 0, py::arg("sync") = true);
+          py::arg("stream") = 0, py::arg("sync") = true, py::arg("fmt") = 0);
   ops.def("embedding",
           [](int dtype, uintptr_t ids, uintptr_t tok, uintptr_t pos,
              uintptr_t seg, uintptr_t segids, uintptr_t out, int M, int S,

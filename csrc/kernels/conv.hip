@@ -184,8 +184,11 @@ static void launch_conv2d_t(const void* in, const void* Wt, void* out,
   int ktiles = p.K / kTileElems<T>;
   int splitk = (!tile && scratch) ? pick_splitk(tiles, ktiles) : 1;
   dim3 block(256);
-  int dtype = std::is_same<T, _Float16>::value ? 0
-              : (std::is_same<T, __bf16>::value ? 1 : 2);
+  int dtype = std::is_same<T, _Float16>::value
+                  ? 0
+                  : (std::is_same<T, __bf16>::value
+                         ? 1
+                         : (std::is_same<T, int8_t>::value ? 2 : 3));
   if (splitk > 1) {
     int ktper = (int)cdiv(ktiles, splitk);
     dim3 grid((unsigned)(tiles * splitk));
@@ -249,9 +252,9 @@ void launch_conv2d(int dtype, const void* in, const void* Wt, void* out,
   p.OW = (W + 2 * pw - KW) / sw + 1;
   p.M = Nb * p.OH * p.OW;
   p.Kreal = KH * KW * C;
-  int kt = dtype == 2 ? 128 : 64;
+  int kt = dtype >= 2 ? 128 : 64;  // 1-byte formats: 128 elems per K-tile
   p.K = (int)round_up(p.Kreal, kt);
-  int cmin = dtype == 2 ? 16 : 8;  // one 16-B glds chunk per pixel minimum
+  int cmin = dtype >= 2 ? 16 : 8;  // one 16-B glds chunk per pixel minimum
   if (C % cmin != 0)
     throw std::runtime_error("conv2d: C must be a multiple of 16 B / elem "
                              "size (pad input channels)");
@@ -265,9 +268,13 @@ void launch_conv2d(int dtype, const void* in, const void* Wt, void* out,
   else if (dtype == 1)
     launch_conv2d_t<__bf16>(in, Wt, out, scale, bias, residual, zero_page, p,
                             epi, stream, tile, (float*)scratch);
-  else
+  else if (dtype == 2)
     launch_conv2d_t<int8_t>(in, Wt, out, scale, bias, residual, zero_page, p,
                             epi, stream, tile, (float*)scratch);
+  else
+    launch_conv2d_t<__hip_fp8_e4m3>(in, Wt, out, scale, bias, residual,
+                                    zero_page, p, epi, stream, tile,
+                                    (float*)scratch);
 }
 
 }  // namespace trtlab

@@ -1,6 +1,8 @@
 // trtlab_amd — elementwise kernels (gfx950): vectorized fp16/bf16, 16 B/lane.
 // SURVEY.md §2.8 item 5; also dtype converts and the channel-pad copy used to
 // bring NHWC inputs to C % 8 == 0 for the implicit-GEMM conv.
+#include <hip/hip_fp8.h>
+
 #include "../common.h"
 
 namespace trtlab {
@@ -110,10 +112,14 @@ void launch_channel_pad(int dtype, const void* in, void* out, int64_t M,
     hipLaunchKernelGGL((channel_pad_kernel<__bf16>), dim3(blocks), dim3(256),
                        0, stream, (const __bf16*)in, (__bf16*)out, M, Cin,
                        Cpad);
-  else
+  else if (dtype == 2)
     hipLaunchKernelGGL((channel_pad_kernel<int8_t>), dim3(blocks), dim3(256),
                        0, stream, (const int8_t*)in, (int8_t*)out, M, Cin,
                        Cpad);
+  else
+    hipLaunchKernelGGL((channel_pad_kernel<__hip_fp8_e4m3>), dim3(blocks),
+                       dim3(256), 0, stream, (const __hip_fp8_e4m3*)in,
+                       (__hip_fp8_e4m3*)out, M, Cin, Cpad);
 }
 
 // fp16 -> int8 symmetric quantization (and inverse), vectorized 8-wide.
@@ -152,21 +158,70 @@ __global__ void dequant_kernel(const int8_t* __restrict__ in,
   }
 }
 
-void launch_quantize(const void* in_f16, void* out_i8, int64_t n, float scale,
-                     hipStream_t stream) {
-  if (n % 8 != 0) throw std::runtime_error("quantize: n % 8 != 0");
-  int blocks = ew_blocks(n / 8);
-  hipLaunchKernelGGL(quantize_kernel, dim3(blocks), dim3(256), 0, stream,
-                     (const _Float16*)in_f16, (int8_t*)out_i8, n / 8,
-                     1.0f / scale);
+// fp8 e4m3 variants (fmt = 1): continuous quantization, saturate at 448.
+__global__ void quantize_fp8_kernel(const _Float16* __restrict__ in,
+                                    unsigned char* __restrict__ out,
+                                    int64_t n8, float inv_scale) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n8;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    short4v v0 = *(const short4v*)(in + i * 8);
+    short4v v1 = *(const short4v*)(in + i * 8 + 4);
+    unsigned char r[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float x = (float)((const _Float16*)(j < 4 ? (const void*)&v0
+                                                : (const void*)&v1))[j & 3];
+      float q = fminf(fmaxf(x * inv_scale, -448.f), 448.f);
+      r[j] = __hip_fp8_e4m3(q).__x;
+    }
+    *(uint2*)(out + i * 8) = *(const uint2*)r;
+  }
 }
 
-void launch_dequant(const void* in_i8, void* out_f16, int64_t n, float scale,
-                    hipStream_t stream) {
+__global__ void dequant_fp8_kernel(const unsigned char* __restrict__ in,
+                                   _Float16* __restrict__ out, int64_t n8,
+                                   float scale) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n8;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    unsigned char r[8];
+    *(uint2*)r = *(const uint2*)(in + i * 8);
+    _Float16 o[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      __hip_fp8_e4m3 v;
+      v.__x = r[j];
+      o[j] = (_Float16)((float)v * scale);
+    }
+    *(short4v*)(out + i * 8) = *(const short4v*)&o[0];
+    *(short4v*)(out + i * 8 + 4) = *(const short4v*)&o[4];
+  }
+}
+
+void launch_quantize(const void* in_f16, void* out_q, int64_t n, float scale,
+                     hipStream_t stream, int fmt) {
+  if (n % 8 != 0) throw std::runtime_error("quantize: n % 8 != 0");
+  int blocks = ew_blocks(n / 8);
+  if (fmt == 0)
+    hipLaunchKernelGGL(quantize_kernel, dim3(blocks), dim3(256), 0, stream,
+                       (const _Float16*)in_f16, (int8_t*)out_q, n / 8,
+                       1.0f / scale);
+  else
+    hipLaunchKernelGGL(quantize_fp8_kernel, dim3(blocks), dim3(256), 0,
+                       stream, (const _Float16*)in_f16,
+                       (unsigned char*)out_q, n / 8, 1.0f / scale);
+}
+
+void launch_dequant(const void* in_q, void* out_f16, int64_t n, float scale,
+                    hipStream_t stream, int fmt) {
   if (n % 8 != 0) throw std::runtime_error("dequant: n % 8 != 0");
   int blocks = ew_blocks(n / 8);
-  hipLaunchKernelGGL(dequant_kernel, dim3(blocks), dim3(256), 0, stream,
-                     (const int8_t*)in_i8, (_Float16*)out_f16, n / 8, scale);
+  if (fmt == 0)
+    hipLaunchKernelGGL(dequant_kernel, dim3(blocks), dim3(256), 0, stream,
+                       (const int8_t*)in_q, (_Float16*)out_f16, n / 8, scale);
+  else
+    hipLaunchKernelGGL(dequant_fp8_kernel, dim3(blocks), dim3(256), 0, stream,
+                       (const unsigned char*)in_q, (_Float16*)out_f16, n / 8,
+                       scale);
 }
 
 void launch_cast(int dtype, bool to_f32, const void* in, void* out, int64_t n,

@@ -137,8 +137,11 @@ static void launch_gemm_bt_t(const void* A, const void* B, void* C,
   int ktiles = K / kTileElems<T>;
   int splitk = (!tile && scratch) ? pick_splitk(tiles, ktiles) : 1;
   dim3 block(256);
-  int dtype = std::is_same<T, _Float16>::value ? 0
-              : (std::is_same<T, __bf16>::value ? 1 : 2);
+  int dtype = std::is_same<T, _Float16>::value
+                  ? 0
+                  : (std::is_same<T, __bf16>::value
+                         ? 1
+                         : (std::is_same<T, int8_t>::value ? 2 : 3));
   if (splitk > 1) {
     int ktper = (int)cdiv(ktiles, splitk);
     dim3 grid((unsigned)(tiles * splitk));
@@ -187,17 +190,22 @@ static void launch_gemm_bt_t(const void* A, const void* B, void* C,
   });
 }
 
-void launch_gemm_bt(int dtype,  // 0 = fp16, 1 = bf16, 2 = int8
+void launch_gemm_bt(int dtype,  // 0 = fp16, 1 = bf16, 2 = int8, 3 = fp8 e4m3
                     const void* A, const void* B, void* C, const float* scale,
                     const float* bias, const void* residual, float res_scale,
                     int M, int N, int K, int64_t lda, int64_t ldb, int64_t ldc,
                     int epi, hipStream_t stream, int tile, void* scratch) {
-  if (dtype == 2) {
+  if (dtype == 2 || dtype == 3) {
     if (K % 128 != 0)
-      throw std::runtime_error("gemm_bt int8: K must be a multiple of 128");
-    launch_gemm_bt_t<int8_t>(A, B, C, scale, bias, residual, res_scale, M, N,
-                             K, lda, ldb, ldc, epi, stream, tile,
-                             (float*)scratch);
+      throw std::runtime_error("gemm_bt int8/fp8: K must be a multiple of 128");
+    if (dtype == 2)
+      launch_gemm_bt_t<int8_t>(A, B, C, scale, bias, residual, res_scale, M,
+                               N, K, lda, ldb, ldc, epi, stream, tile,
+                               (float*)scratch);
+    else
+      launch_gemm_bt_t<__hip_fp8_e4m3>(A, B, C, scale, bias, residual,
+                                       res_scale, M, N, K, lda, ldb, ldc,
+                                       epi, stream, tile, (float*)scratch);
     return;
   }
   if (K % 64 != 0) throw std::runtime_error("gemm_bt: K must be a multiple of 64");

@@ -1,6 +1,8 @@
 // trtlab_amd — shared MFMA-GEMM machinery for gfx950 kernels.
 // See gemm.hip for the structure notes.
 #pragma once
+#include <hip/hip_fp8.h>
+
 #include "../common.h"
 
 namespace trtlab {
@@ -51,10 +53,13 @@ typedef __attribute__((ext_vector_type(4))) int i32x4;
 template <typename T>
 struct Mfma16x16x32;
 
+// kFragBytes: bytes of one lane's A/B fragment (ds_read width);
+// kStepBytes: bytes of one MFMA K-step along a 128-B LDS row.
 template <>
 struct Mfma16x16x32<_Float16> {
   using frag = half8v;
   using accv = f32x4;
+  static constexpr int kFragBytes = 16, kStepBytes = 64;
   static __device__ __forceinline__ f32x4 run(frag a, frag b, f32x4 c) {
     return __builtin_amdgcn_mfma_f32_16x16x32_f16(a, b, c, 0, 0, 0);
   }
@@ -64,6 +69,7 @@ template <>
 struct Mfma16x16x32<__bf16> {
   using frag = bf16x8v;
   using accv = f32x4;
+  static constexpr int kFragBytes = 16, kStepBytes = 64;
   static __device__ __forceinline__ f32x4 run(frag a, frag b, f32x4 c) {
     return __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, c, 0, 0, 0);
   }
@@ -73,8 +79,21 @@ template <>
 struct Mfma16x16x32<int8_t> {
   using frag = i8x16v;
   using accv = i32x4;
+  static constexpr int kFragBytes = 16, kStepBytes = 64;
   static __device__ __forceinline__ i32x4 run(frag a, frag b, i32x4 c) {
     return __builtin_amdgcn_mfma_i32_16x16x64_i8(a, b, c, 0, 0, 0);
+  }
+};
+
+// OCP fp8 e4m3 (gfx950-native; NOT the MI300X fnuz variant). K=32 per
+// instruction, 8-byte lane fragments (ds_read_b64); 4 K-steps per 128-B row.
+template <>
+struct Mfma16x16x32<__hip_fp8_e4m3> {
+  using frag = long;
+  using accv = f32x4;
+  static constexpr int kFragBytes = 8, kStepBytes = 32;
+  static __device__ __forceinline__ f32x4 run(frag a, frag b, f32x4 c) {
+    return __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(a, b, c, 0, 0, 0);
   }
 };
 
@@ -92,6 +111,10 @@ __device__ __forceinline__ int8_t store_cast<int8_t>(float v) {
   float r = rintf(v);
   r = fminf(fmaxf(r, -127.f), 127.f);
   return (int8_t)r;
+}
+template <>
+__device__ __forceinline__ __hip_fp8_e4m3 store_cast<__hip_fp8_e4m3>(float v) {
+  return __hip_fp8_e4m3(fminf(fmaxf(v, -448.f), 448.f));
 }
 
 // ------------------------------------------------------------------ staging
@@ -123,7 +146,9 @@ __device__ __forceinline__ void stage_tile(const T* __restrict__ src,
   }
 }
 
-// Swizzled ds_read_b128 of one MFMA fragment (8 consecutive k elements).
+// Swizzled ds_read of one MFMA fragment (8 consecutive k elements, 8 or
+// 16 B per lane). The XOR swizzle flips bits 4-6 of the byte offset, so
+// 8- and 16-byte-aligned fragments stay naturally aligned and contiguous.
 template <typename T>
 __device__ __forceinline__ typename Mfma16x16x32<T>::frag read_frag(
     const char* lds, uint32_t row, uint32_t kbyte) {
@@ -140,10 +165,11 @@ __device__ __forceinline__ void mfma_tile(
     typename Mfma16x16x32<T>::accv (&acc)[BM / 32][BN / 32]) {
   using MF = Mfma16x16x32<T>;
   constexpr int MFr = BM / 32, NFr = BN / 32;
+  constexpr int kSteps = 128 / MF::kStepBytes;
 #pragma unroll
-  for (int ks = 0; ks < 2; ++ks) {
+  for (int ks = 0; ks < kSteps; ++ks) {
     typename MF::frag af[MFr], bf[NFr];
-    uint32_t kbyte = ks * 64 + ((lane >> 4) << 4);
+    uint32_t kbyte = ks * MF::kStepBytes + ((lane >> 4) * MF::kFragBytes);
 #pragma unroll
     for (int f = 0; f < MFr; ++f)
       af[f] = read_frag<T>(As, wr * (BM / 2) + f * 16 + (lane & 15), kbyte);

@@ -54,9 +54,10 @@ void launch_embedding(int dtype, const void* ids, const void* tok,
 
 // int8 quantization staging: out_i8 = clamp(round(in_f16 / scale)),
 // and the inverse. n % 8 == 0.
-void launch_quantize(const void* in_f16, void* out_i8, int64_t n, float scale,
-                     hipStream_t stream);
-void launch_dequant(const void* in_i8, void* out_f16, int64_t n, float scale,
-                    hipStream_t stream);
+// fmt: 0 = int8 (round-to-int codes), 1 = fp8 e4m3 (continuous, sat 448)
+void launch_quantize(const void* in_f16, void* out_q, int64_t n, float scale,
+                     hipStream_t stream, int fmt = 0);
+void launch_dequant(const void* in_q, void* out_f16, int64_t n, float scale,
+                    hipStream_t stream, int fmt = 0);
 
 }  // namespace trtlab

@@ -1,6 +1,8 @@
 // trtlab_amd — pooling kernels (NHWC fp16/bf16), gfx950.
 // Memory-bound: vectorized 8-wide half loads (G13), grid-stride loops.
 // Covers the reference's TensorRT-internal pool ops.
+#include <hip/hip_fp8.h>
+
 #include "../common.h"
 
 namespace trtlab {
@@ -98,8 +100,11 @@ void launch_maxpool2d(int dtype, const void* in, void* out, int Nb, int H,
     launch_maxpool2d_t<_Float16>(in, out, Nb, H, W, C, KH, KW, sh, sw, ph, pw, stream);
   else if (dtype == 1)
     launch_maxpool2d_t<__bf16>(in, out, Nb, H, W, C, KH, KW, sh, sw, ph, pw, stream);
-  else
+  else if (dtype == 2)
     launch_maxpool2d_t<int8_t>(in, out, Nb, H, W, C, KH, KW, sh, sw, ph, pw, stream);
+  else
+    launch_maxpool2d_t<__hip_fp8_e4m3>(in, out, Nb, H, W, C, KH, KW, sh, sw,
+                                       ph, pw, stream);
 }
 
 void launch_gavgpool(int dtype, const void* in, void* out, int Nb, int HW,

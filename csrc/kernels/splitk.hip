@@ -79,11 +79,16 @@ void launch_splitk_reduce(int dtype, const float* scratch, void* C,
                          stream, scratch, (__bf16*)C, scale, bias,
                          (const __bf16*)residual, res_scale, M, N, ldc,
                          tiles_n, splitk, log_elems, bn, total4);
-    else
+    else if (dtype == 2)
       hipLaunchKernelGGL((splitk_reduce_kernel<int8_t, EE>), grid, block, 0,
                          stream, scratch, (int8_t*)C, scale, bias,
                          (const int8_t*)residual, res_scale, M, N, ldc,
                          tiles_n, splitk, log_elems, bn, total4);
+    else
+      hipLaunchKernelGGL((splitk_reduce_kernel<__hip_fp8_e4m3, EE>), grid,
+                         block, 0, stream, scratch, (__hip_fp8_e4m3*)C, scale,
+                         bias, (const __hip_fp8_e4m3*)residual, res_scale, M,
+                         N, ldc, tiles_n, splitk, log_elems, bn, total4);
   });
 }
 

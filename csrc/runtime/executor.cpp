@@ -141,12 +141,13 @@ void ExecutionContext::enqueue_all(hipStream_t s) {
                          op.NH, op.HD, op.att_scale, s);
         break;
       case kQuantize:
+        // op.epi carries the target format (0 = int8, 1 = fp8 e4m3)
         launch_quantize(A(op.in_off), A(op.out_off), op.n_elems, op.q_scale,
-                        s);
+                        s, op.epi);
         break;
       case kDequant:
         launch_dequant(A(op.in_off), A(op.out_off), op.n_elems, op.q_scale,
-                       s);
+                       s, op.epi);
         break;
       case kEmbedding:
         // tables live in the weight blob as fp16: tok at w_off, pos at

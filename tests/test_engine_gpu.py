@@ -207,3 +207,26 @@ def test_timed_context_stage_split(rn50_small):
     assert h2d > 0 and compute > 0 and d2h >= 0
     assert compute > h2d  # forward dominates a 64px batch-2 run
     assert compute < 50
+
+
+def test_engine_resnet50_fp8():
+    """fp8 e4m3 engine: same quality-gate oracle as int8."""
+    from trtlab_amd.engine.planner import DT_F8, Planner
+    from trtlab_amd.engine.reference import run_reference
+    from trtlab_amd.engine.runtime import NativeEngine
+    from trtlab_amd.models import build_resnet
+
+    g = build_resnet(50, batch=2, image=64, seed=0)
+    plan8 = Planner(dtype=DT_F8).compile(g)
+    plan16 = Planner().compile(g)
+    eng = NativeEngine(plan8)
+    ctx = eng.create_context(capture=True)
+    x = np.random.RandomState(11).randn(*plan8.input_shape).astype(np.float32) * 0.5
+    out = ctx.infer(x).astype(np.float32)
+    assert np.isfinite(out).all()
+    ref16 = run_reference(plan16, x)
+    ref8 = run_reference(plan8, x)
+    corr_gpu = np.corrcoef(out.ravel(), ref16.ravel())[0, 1]
+    corr_emu = np.corrcoef(ref8.ravel(), ref16.ravel())[0, 1]
+    assert corr_gpu > corr_emu - 0.03, (corr_gpu, corr_emu)
+    assert corr_gpu > 0.9, corr_gpu

@@ -307,3 +307,38 @@ def test_quantize_dequant(C):
     C.ops.dequant(q.data_ptr(), back.data_ptr(), q.numel(), 0.05)
     # fp16 output rounding: ~2^-11 relative at |x| ~ 6
     assert (back.float() - q.float() * 0.05).abs().max() < 4e-3
+
+
+def test_gemm_bt_fp8(C):
+    """fp8 e4m3 MFMA GEMM vs dequantized torch reference."""
+    M, N, K = 256, 512, 384
+    g = torch.Generator(device="cuda").manual_seed(31)
+    af = torch.randn(M, K, generator=g, device="cuda")
+    bf = torch.randn(N, K, generator=g, device="cuda")
+    a8 = af.to(torch.float8_e4m3fn)
+    b8 = bf.to(torch.float8_e4m3fn)
+    scale = torch.full((N,), 1.0, device="cuda").float()
+    bias = torch.zeros(N, device="cuda").float()
+    out = torch.empty(M, N, dtype=torch.float8_e4m3fn, device="cuda")
+    torch.cuda.synchronize()
+    C.ops.gemm_bt(3, a8.data_ptr(), b8.data_ptr(), out.data_ptr(),
+                  scale=scale.data_ptr(), bias=bias.data_ptr(),
+                  M=M, N=N, K=K, epi=EPI_SB)
+    acc = a8.float() @ b8.float().t()
+    ref = torch.clamp(acc, -448, 448).to(torch.float8_e4m3fn).float()
+    err = (out.float() - ref).abs()
+    tol = 0.07 * ref.abs() + 0.6  # one e4m3 ulp on the requantized output
+    bad = (err > tol).float().mean().item()
+    assert bad < 1e-3, (bad, float(err.max()))
+
+
+def test_quantize_dequant_fp8(C):
+    x = t16(64, 128, seed=32, scale=2.0)
+    q = torch.empty(64, 128, dtype=torch.uint8, device="cuda")
+    torch.cuda.synchronize()
+    C.ops.quantize(x.data_ptr(), q.data_ptr(), x.numel(), 0.05, fmt=1)
+    ref = torch.clamp(x.float() / 0.05, -448, 448).to(torch.float8_e4m3fn)
+    assert (q.view(torch.float8_e4m3fn).float() - ref.float()).abs().max() < 1e-3
+    back = torch.empty(64, 128, dtype=torch.half, device="cuda")
+    C.ops.dequant(q.data_ptr(), back.data_ptr(), q.numel(), 0.05, fmt=1)
+    assert (back.float() - ref.float() * 0.05).abs().max() < 0.05

@@ -39,6 +39,7 @@ K_CONV, K_GEMM, K_MAXPOOL, K_GAVGPOOL, K_SOFTMAX, K_LAYERNORM, \
 DT_F16 = 0
 DT_BF16 = 1
 DT_I8 = 2
+DT_F8 = 3  # OCP fp8 e4m3
 
 K_QUANTIZE = 10
 K_DEQUANT = 11
@@ -240,7 +241,7 @@ class Planner:
         assert input_name is not None
         in_shape = shapes[input_name]
         padded_input = input_name
-        pad_mult = 16 if self.dtype == DT_I8 else 8
+        pad_mult = 16 if self.dtype in (DT_I8, DT_F8) else 8
         if len(in_shape) == 4 and in_shape[3] % pad_mult != 0:
             cpad = round_up(in_shape[3], pad_mult)
             padded_input = input_name + "_padded"
@@ -259,11 +260,12 @@ class Planner:
             t: {"f16": 2, "bf16": 2, "f32": 4, "i32": 4, "i8": 1}[spec.dtype]
             for t, spec in g.tensors.items()
         }
-        if self.dtype == DT_I8:
+        if self.dtype in (DT_I8, DT_F8):
             from trtlab_amd.engine.quantize import lower_int8
 
             lower_int8(g, exec_ops, shapes, itemsize, input_name,
-                       padded_input, self.calib_sample)
+                       padded_input, self.calib_sample,
+                       fmt="f8" if self.dtype == DT_F8 else "i8")
         for op in exec_ops:
             if op.kind == K_CONV and op.params.get("int8") is None:
                 w = op.w  # [Cout, Cin, KH, KW] fp32
@@ -340,8 +342,9 @@ class Planner:
         op_dicts: List[Dict[str, Any]] = []
         for op in exec_ops:
             w_off, s_off, b_off = w_offs[op.name]
-            op_dtype = op.params.get("dtype", DT_F16 if self.dtype == DT_I8
-                                     else self.dtype)
+            op_dtype = op.params.get(
+                "dtype",
+                DT_F16 if self.dtype in (DT_I8, DT_F8) else self.dtype)
             d: Dict[str, Any] = dict(dtype=op_dtype, w_off=w_off,
                                      scale_off=s_off, bias_off=b_off,
                                      in_off=offsets[op.inputs[0]],
@@ -399,7 +402,8 @@ class Planner:
                 for s_ in shapes[op.output]:
                     n *= s_
                 d.update(kind=op.kind, n_elems=n,
-                         q_scale=op.params["q_scale"])
+                         q_scale=op.params["q_scale"],
+                         epi=1 if op.params.get("fmt") == "f8" else 0)
             elif op.kind == K_ATTENTION:
                 m, hid = shapes[op.output]
                 heads = op.params["heads"]

@@ -82,9 +82,28 @@ def calibrate_amax(g: Graph, sample: np.ndarray) -> Dict[str, float]:
     return amax
 
 
+def _quantize_weights(flat: np.ndarray, fmt: str):
+    """Per-channel symmetric quantization of [Cout, K] fp32 weights.
+    Returns (codes uint8-or-int8 array, per-channel scales)."""
+    if fmt == "i8":
+        sw = np.maximum(np.abs(flat).max(axis=1), 1e-8) / 127.0
+        q = np.clip(np.rint(flat / sw[:, None]), -127, 127).astype(np.int8)
+        return q, sw
+    import torch
+
+    sw = np.maximum(np.abs(flat).max(axis=1), 1e-8) / 448.0
+    t = torch.from_numpy(flat / sw[:, None])
+    q = t.to(torch.float8_e4m3fn).view(torch.uint8).numpy()
+    return np.ascontiguousarray(q), sw
+
+
 def lower_int8(g: Graph, exec_ops: List, shapes: Dict, itemsize: Dict,
-               input_name: str, padded_input: str, calib_sample) -> None:
-    """Mutates exec_ops/shapes/itemsize in place for the int8 conv stack."""
+               input_name: str, padded_input: str, calib_sample,
+               fmt: str = "i8") -> None:
+    """Mutates exec_ops/shapes/itemsize in place: lowers the conv stack to
+    int8 (fmt='i8') or OCP fp8 e4m3 (fmt='f8')."""
+    qmax = 127.0 if fmt == "i8" else 448.0
+    dt_q = _DT_I8 if fmt == "i8" else 3
     if calib_sample is None:
         rng = np.random.RandomState(1234)
         in_shape = shapes[input_name]
@@ -102,12 +121,12 @@ def lower_int8(g: Graph, exec_ops: List, shapes: Dict, itemsize: Dict,
             op.params["dtype"] = _DT_F16
             new_ops.append(op)
             # quantize the padded input
-            s_in = max(amax[input_name], 1e-6) / 127.0
+            s_in = max(amax[input_name], 1e-6) / qmax
             qname = op.output + "_q"
             shapes[qname] = shapes[op.output]
             itemsize[qname] = 1
             qop = ExecOp(_K_QUANTIZE, qname, [op.output], qname,
-                         dict(q_scale=s_in, dtype=_DT_F16))
+                         dict(q_scale=s_in, dtype=_DT_F16, fmt=fmt))
             new_ops.append(qop)
             scales[qname] = s_in
             # rewrite consumers
@@ -116,7 +135,7 @@ def lower_int8(g: Graph, exec_ops: List, shapes: Dict, itemsize: Dict,
         elif op.kind == _K_CONV:
             s_in = scales[op.inputs[0]]
             out_name = op.output
-            s_out = max(amax[out_name], 1e-6) / 127.0
+            s_out = max(amax[out_name], 1e-6) / qmax
             w = op.w  # [Cout, Cin, KH, KW] fp32 (original)
             cout, cin, kh, kw = w.shape
             cpad = round_up(cin, 16)
@@ -128,13 +147,13 @@ def lower_int8(g: Graph, exec_ops: List, shapes: Dict, itemsize: Dict,
             flat = whwc.reshape(cout, k).astype(np.float32)
             if kp != k:
                 flat = np.pad(flat, ((0, 0), (0, kp - k)))
-            sw = np.maximum(np.abs(flat).max(axis=1), 1e-8) / 127.0
-            q = np.clip(np.rint(flat / sw[:, None]), -127, 127).astype(np.int8)
-            op.w = np.ascontiguousarray(q)
+            q, sw = _quantize_weights(flat, fmt)
+            op.w = q
             op.params["C"] = cpad
             op.params["Kp"] = kp
-            op.params["dtype"] = _DT_I8
+            op.params["dtype"] = dt_q
             op.params["int8"] = True
+            op.params["fmt"] = fmt
             # fold scales into the epilogue
             g_ = op.scale if op.scale is not None else np.ones(cout, np.float32)
             b_ = op.bias if op.bias is not None else np.zeros(cout, np.float32)
@@ -152,7 +171,7 @@ def lower_int8(g: Graph, exec_ops: List, shapes: Dict, itemsize: Dict,
             scales[out_name] = s_out
             new_ops.append(op)
         elif op.kind == _K_MAXPOOL and op.inputs[0] in scales:
-            op.params["dtype"] = _DT_I8
+            op.params["dtype"] = dt_q
             itemsize[op.output] = 1
             scales[op.output] = scales[op.inputs[0]]  # max() preserves scale
             new_ops.append(op)
@@ -163,7 +182,8 @@ def lower_int8(g: Graph, exec_ops: List, shapes: Dict, itemsize: Dict,
             shapes[dq] = shapes[src]
             itemsize[dq] = 2
             new_ops.append(ExecOp(_K_DEQUANT, dq, [src], dq,
-                                  dict(q_scale=scales[src], dtype=_DT_F16)))
+                                  dict(q_scale=scales[src], dtype=_DT_F16,
+                                       fmt=fmt)))
             op.inputs = [dq]
             op.params["dtype"] = _DT_F16
             new_ops.append(op)

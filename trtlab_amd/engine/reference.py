@@ -22,6 +22,11 @@ from trtlab_amd.engine.planner import (
     EPI_SCALE_BIAS_ADD_RELU, EPI_SCALE_BIAS_RELU)
 
 
+def _fp8_round(x: torch.Tensor) -> torch.Tensor:
+    """Round to the OCP e4m3 grid (saturating), staying in fp32."""
+    return torch.clamp(x, -448, 448).to(torch.float8_e4m3fn).float()
+
+
 def _epi(acc: torch.Tensor, epi: int, scale, bias, res) -> torch.Tensor:
     v = acc
     if epi in (EPI_SCALE_BIAS, EPI_SCALE_BIAS_RELU, EPI_SCALE_BIAS_ADD_RELU):
@@ -56,7 +61,10 @@ def run_reference(plan: EnginePlan, input_nhwc: np.ndarray,
             nb, h, w, c = x.shape
             kh, kw = d["KH"], d["KW"]
             cout = d["Cout"]
-            wt = torch.from_numpy(op.w.astype(np.float32))  # [Cout, Kp]
+            if d["dtype"] == 3:  # fp8 weights stored as uint8 codes
+                wt = torch.from_numpy(op.w).view(torch.float8_e4m3fn).float()
+            else:
+                wt = torch.from_numpy(op.w.astype(np.float32))  # [Cout, Kp]
             k = kh * kw * c
             wt = wt[:, :k].reshape(cout, kh, kw, c).permute(0, 3, 1, 2)
             xc = x.permute(0, 3, 1, 2)
