@@ -196,6 +196,18 @@ PYBIND11_MODULE(_C, m) {
           py::arg("H"), py::arg("W"), py::arg("C"), py::arg("KH"),
           py::arg("KW"), py::arg("sh"), py::arg("sw"), py::arg("ph"),
           py::arg("pw"), py::arg("stream") = 0, py::arg("sync") = true);
+  ops.def("avgpool2d",
+          [](int dtype, uintptr_t in, uintptr_t out, int Nb, int H, int W,
+             int C, int KH, int KW, int sh, int sw, int ph, int pw,
+             uintptr_t stream, bool sync) {
+            launch_avgpool2d(dtype, (void*)in, (void*)out, Nb, H, W, C, KH,
+                             KW, sh, sw, ph, pw, as_stream(stream));
+            if (sync) TRT_HIP_CHECK(hipStreamSynchronize(as_stream(stream)));
+          },
+          py::arg("dtype"), py::arg("in"), py::arg("out"), py::arg("Nb"),
+          py::arg("H"), py::arg("W"), py::arg("C"), py::arg("KH"),
+          py::arg("KW"), py::arg("sh"), py::arg("sw"), py::arg("ph"),
+          py::arg("pw"), py::arg("stream") = 0, py::arg("sync") = true);
   ops.def("gavgpool",
           [](int dtype, uintptr_t in, uintptr_t out, int Nb, int HW, int C,
              uintptr_t stream, bool sync) {

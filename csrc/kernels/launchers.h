@@ -25,6 +25,9 @@ size_t conv_scratch_bytes(int Nb, int H, int W, int C, int Cout, int KH,
 void launch_maxpool2d(int dtype, const void* in, void* out, int Nb, int H,
                       int W, int C, int KH, int KW, int sh, int sw, int ph,
                       int pw, hipStream_t stream);
+void launch_avgpool2d(int dtype, const void* in, void* out, int Nb, int H,
+                      int W, int C, int KH, int KW, int sh, int sw, int ph,
+                      int pw, hipStream_t stream);
 void launch_gavgpool(int dtype, const void* in, void* out, int Nb, int HW,
                      int C, hipStream_t stream);
 

@@ -51,6 +51,70 @@ __global__ void maxpool2d_kernel(const T* __restrict__ in, T* __restrict__ out,
   }
 }
 
+// Windowed average pool (NHWC, count_include_pad=False — torch default).
+template <typename T>
+__global__ void avgpool2d_kernel(const T* __restrict__ in, T* __restrict__ out,
+                                 int Nb, int H, int W, int C, int KH, int KW,
+                                 int sh, int sw, int ph, int pw, int OH,
+                                 int OW) {
+  int cg = C >> 3;
+  int64_t total = (int64_t)Nb * OH * OW * cg;
+  for (int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       idx < total; idx += (int64_t)gridDim.x * blockDim.x) {
+    int c8 = (int)(idx % cg);
+    int64_t q = idx / cg;
+    int ow = (int)(q % OW);
+    q /= OW;
+    int oh = (int)(q % OH);
+    int n = (int)(q / OH);
+    float acc[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+    int cnt = 0;
+    for (int kh = 0; kh < KH; ++kh) {
+      int ih = oh * sh - ph + kh;
+      if ((unsigned)ih >= (unsigned)H) continue;
+      for (int kw = 0; kw < KW; ++kw) {
+        int iw = ow * sw - pw + kw;
+        if ((unsigned)iw >= (unsigned)W) continue;
+        const T* p = in + ((((int64_t)n * H + ih) * W + iw) * C) + c8 * 8;
+        T tmp[8];
+        if constexpr (sizeof(T) == 2) {
+          *(short4v*)&tmp[0] = *(const short4v*)p;
+          *(short4v*)&tmp[4] = *(const short4v*)(p + 4);
+        } else {
+          *(uint2*)&tmp[0] = *(const uint2*)p;
+        }
+#pragma unroll
+        for (int j = 0; j < 8; ++j) acc[j] += (float)tmp[j];
+        ++cnt;
+      }
+    }
+    float inv = cnt ? 1.0f / cnt : 0.0f;
+    T* o = out + ((((int64_t)n * OH + oh) * OW + ow) * C) + c8 * 8;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) o[j] = (T)(acc[j] * inv);
+  }
+}
+
+void launch_avgpool2d(int dtype, const void* in, void* out, int Nb, int H,
+                      int W, int C, int KH, int KW, int sh, int sw, int ph,
+                      int pw, hipStream_t stream) {
+  if (C % 8 != 0) throw std::runtime_error("avgpool2d: C % 8 != 0");
+  int OH = (H + 2 * ph - KH) / sh + 1;
+  int OW = (W + 2 * pw - KW) / sw + 1;
+  int64_t total = (int64_t)Nb * OH * OW * (C / 8);
+  int blocks = (int)std::min<int64_t>(cdiv(total, 256), 2048);
+  if (dtype == 0)
+    hipLaunchKernelGGL((avgpool2d_kernel<_Float16>), dim3(blocks), dim3(256),
+                       0, stream, (const _Float16*)in, (_Float16*)out, Nb, H,
+                       W, C, KH, KW, sh, sw, ph, pw, OH, OW);
+  else if (dtype == 1)
+    hipLaunchKernelGGL((avgpool2d_kernel<__bf16>), dim3(blocks), dim3(256), 0,
+                       stream, (const __bf16*)in, (__bf16*)out, Nb, H, W, C,
+                       KH, KW, sh, sw, ph, pw, OH, OW);
+  else
+    throw std::runtime_error("avgpool2d: fp16/bf16 only");
+}
+
 // Global average pool: in [Nb, HW, C] -> out [Nb, C], fp32 accumulate.
 template <typename T>
 __global__ void gavgpool_kernel(const T* __restrict__ in, T* __restrict__ out,

@@ -110,6 +110,11 @@ void ExecutionContext::enqueue_all(hipStream_t s) {
                          op.W, op.C, op.KH, op.KW, op.sh, op.sw, op.ph, op.pw,
                          s);
         break;
+      case kAvgPool:
+        launch_avgpool2d(op.dtype, A(op.in_off), A(op.out_off), op.Nb, op.H,
+                         op.W, op.C, op.KH, op.KW, op.sh, op.sw, op.ph,
+                         op.pw, s);
+        break;
       case kGAvgPool:
         launch_gavgpool(op.dtype, A(op.in_off), A(op.out_off), op.Nb, op.HW,
                         op.C, s);

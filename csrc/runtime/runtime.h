@@ -53,6 +53,7 @@ enum OpKind : int {
   kQuantize = 10,   // fp16 -> int8 (symmetric, scale)
   kDequant = 11,    // int8 -> fp16 (scale)
   kEmbedding = 12,  // out = tok[ids] + pos[m%S] (+ seg)
+  kAvgPool = 13,
 };
 
 struct OpDesc {

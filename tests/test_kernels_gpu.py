@@ -342,3 +342,15 @@ def test_quantize_dequant_fp8(C):
     back = torch.empty(64, 128, dtype=torch.half, device="cuda")
     C.ops.dequant(q.data_ptr(), back.data_ptr(), q.numel(), 0.05, fmt=1)
     assert (back.float() - ref.float() * 0.05).abs().max() < 0.05
+
+
+def test_avgpool2d(C):
+    x = t16(2, 16, 16, 64, seed=40)
+    out = torch.empty(2, 8, 8, 64, dtype=torch.half, device="cuda")
+    torch.cuda.synchronize()
+    C.ops.avgpool2d(0, x.data_ptr(), out.data_ptr(), 2, 16, 16, 64, 3, 3,
+                    2, 2, 1, 1)
+    ref = torch.nn.functional.avg_pool2d(
+        x.float().permute(0, 3, 1, 2), 3, stride=2, padding=1,
+        count_include_pad=False).permute(0, 2, 3, 1)
+    check(out, ref)

@@ -113,3 +113,40 @@ def test_siege_load_generator(echo_server):
     assert stats["errors"] == 0
     assert stats["completed"] >= 100
     assert stats["p99_ms"] > 0
+
+
+def test_client_deadline_cancel(echo_server):
+    """Client-side deadline on a slow handler: the call fails with
+    DEADLINE_EXCEEDED and the server keeps serving (reference
+    test_pingpong.cc early-cancel/early-finish teardown tests)."""
+    import grpc
+
+    server = echo_server
+    svc = AsyncService("trtlab.Slow")
+
+    async def slow(request, context, resources):
+        import asyncio
+
+        await asyncio.sleep(2.0)
+        return EchoResponse(message=request.message, tag=request.tag)
+
+    svc.register_unary("Echo", slow, EchoRequest, EchoResponse)
+    # register a second service on a NEW server (echo_server already built)
+    s2 = Server("127.0.0.1:0")
+    s2.register_service(svc)
+    s2.async_start()
+    try:
+        c = SyncClient(f"127.0.0.1:{s2.port}")
+        with pytest.raises(grpc.RpcError) as ei:
+            c.call("trtlab.Slow", "Echo", EchoRequest(message="x"),
+                   EchoResponse, timeout=0.2)
+        assert ei.value.code() == grpc.StatusCode.DEADLINE_EXCEEDED
+        c.close()
+        # the original echo server is still healthy
+        c2 = SyncClient(f"127.0.0.1:{server.port}")
+        r = c2.call("trtlab.Echo", "Echo", EchoRequest(message="ok"),
+                    EchoResponse, timeout=5)
+        assert r.message == "ok"
+        c2.close()
+    finally:
+        s2.shutdown()

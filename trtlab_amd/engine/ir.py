@@ -116,6 +116,15 @@ class Graph:
                           dict(kernel=kernel, stride=stride, padding=padding),
                           name)
 
+    def avgpool(self, x: str, kernel: int, stride: int, padding: int = 0,
+                name: Optional[str] = None) -> str:
+        n, h, w, c = self.tensors[x].shape
+        oh = (h + 2 * padding - kernel) // stride + 1
+        ow = (w + 2 * padding - kernel) // stride + 1
+        return self._emit("avgpool", [x], (n, oh, ow, c),
+                          dict(kernel=kernel, stride=stride, padding=padding),
+                          name)
+
     def global_avgpool(self, x: str, name: Optional[str] = None) -> str:
         n, h, w, c = self.tensors[x].shape
         return self._emit("gavgpool", [x], (n, c), dict(hw=h * w), name)

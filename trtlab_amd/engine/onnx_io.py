@@ -155,6 +155,12 @@ def import_onnx(data: bytes, batch: Optional[int] = None,
             out = g.gelu(x)
         elif op == "Add":
             out = g.add(x, remap.get(ins[1], ins[1]))
+        elif op == "AveragePool":
+            ks = attrs.get("kernel_shape", [2, 2])
+            strides = attrs.get("strides", [1, 1])
+            pads = attrs.get("pads", [0, 0, 0, 0])
+            out = g.avgpool(x, kernel=int(ks[0]), stride=int(strides[0]),
+                            padding=int(pads[0]))
         elif op == "MaxPool":
             ks = attrs.get("kernel_shape", [2, 2])
             strides = attrs.get("strides", [1, 1])

@@ -44,6 +44,7 @@ DT_F8 = 3  # OCP fp8 e4m3
 K_QUANTIZE = 10
 K_DEQUANT = 11
 K_EMBEDDING = 12
+K_AVGPOOL = 13
 
 
 @dataclass
@@ -126,6 +127,9 @@ class Planner:
                                        n.output, dict(op=2)))
             elif n.kind == "maxpool":
                 exec_ops.append(ExecOp(K_MAXPOOL, n.name, [n.inputs[0]],
+                                       n.output, dict(n.attrs)))
+            elif n.kind == "avgpool":
+                exec_ops.append(ExecOp(K_AVGPOOL, n.name, [n.inputs[0]],
                                        n.output, dict(n.attrs)))
             elif n.kind == "gavgpool":
                 exec_ops.append(ExecOp(K_GAVGPOOL, n.name, [n.inputs[0]],
@@ -365,9 +369,9 @@ class Planner:
                 m, k = shapes[op.inputs[0]]
                 nout = shapes[op.output][1]
                 d.update(kind=K_GEMM, epi=op.params["epi"], M=m, N=nout, K=k)
-            elif op.kind == K_MAXPOOL:
+            elif op.kind in (K_MAXPOOL, K_AVGPOOL):
                 ish = shapes[op.inputs[0]]
-                d.update(kind=K_MAXPOOL, Nb=ish[0], H=ish[1], W=ish[2],
+                d.update(kind=op.kind, Nb=ish[0], H=ish[1], W=ish[2],
                          C=ish[3], KH=op.params["kernel"],
                          KW=op.params["kernel"], sh=op.params["stride"],
                          sw=op.params["stride"], ph=op.params["padding"],

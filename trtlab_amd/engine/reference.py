@@ -15,7 +15,7 @@ import torch
 import torch.nn.functional as F
 
 from trtlab_amd.engine.planner import (
-    EnginePlan, K_ADD_LAYERNORM, K_ATTENTION, K_CHANNEL_PAD, K_CONV,
+    EnginePlan, K_ADD_LAYERNORM, K_ATTENTION, K_AVGPOOL, K_CHANNEL_PAD, K_CONV,
     K_DEQUANT, K_ELEMENTWISE, K_GAVGPOOL, K_GEMM, K_LAYERNORM, K_MAXPOOL,
     K_EMBEDDING, K_QUANTIZE, K_SOFTMAX,
     EPI_BIAS, EPI_BIAS_GELU, EPI_BIAS_RELU, EPI_NONE, EPI_SCALE_BIAS,
@@ -105,6 +105,11 @@ def run_reference(plan: EnginePlan, input_nhwc: np.ndarray,
         elif op.kind == K_MAXPOOL:
             xc = x.permute(0, 3, 1, 2)
             y = F.max_pool2d(xc, d["KH"], stride=d["sh"], padding=d["ph"])
+            t[op.output] = y.permute(0, 2, 3, 1)
+        elif op.kind == K_AVGPOOL:
+            xc = x.permute(0, 3, 1, 2)
+            y = F.avg_pool2d(xc, d["KH"], stride=d["sh"], padding=d["ph"],
+                             count_include_pad=False)
             t[op.output] = y.permute(0, 2, 3, 1)
         elif op.kind == K_GAVGPOOL:
             nb, h, w, c = x.shape
