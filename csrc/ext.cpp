@@ -47,6 +47,7 @@ OpDesc op_from_dict(const py::dict& d) {
   o.att_scale = gf("att_scale", 1.0f);
   o.res_scale = gf("res_scale", 1.0f);
   o.q_scale = gf("q_scale", 1.0f);
+  o.tile = gi("tile", 0);
   return o;
 }
 

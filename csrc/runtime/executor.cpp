@@ -96,14 +96,15 @@ void ExecutionContext::enqueue_all(hipStream_t s) {
         launch_conv2d(op.dtype, A(op.in_off), Wp(op.w_off), A(op.out_off),
                       Fp(op.scale_off), Fp(op.bias_off), A(op.in2_off),
                       eng_->zero_page(), op.Nb, op.H, op.W, op.C, op.Cout,
-                      op.KH, op.KW, op.sh, op.sw, op.ph, op.pw, op.epi, s, 0,
-                      scratch_, op.res_scale);
+                      op.KH, op.KW, op.sh, op.sw, op.ph, op.pw, op.epi, s,
+                      op.tile, scratch_, op.res_scale);
         break;
       case kGemmBt:
         launch_gemm_bt(op.dtype, A(op.in_off), Wp(op.w_off), A(op.out_off),
                        Fp(op.scale_off), Fp(op.bias_off), A(op.in2_off),
                        op.res_scale, op.M, op.N, op.K, op.K /*lda*/,
-                       op.K /*ldb*/, op.N /*ldc*/, op.epi, s, 0, scratch_);
+                       op.K /*ldb*/, op.N /*ldc*/, op.epi, s, op.tile,
+                       scratch_);
         break;
       case kMaxPool:
         launch_maxpool2d(op.dtype, A(op.in_off), A(op.out_off), op.Nb, op.H,

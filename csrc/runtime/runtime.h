@@ -78,6 +78,8 @@ struct OpDesc {
   // int8: residual dequant ratio (s_res/s_out); quantize/dequant scale
   float res_scale = 1.0f;
   float q_scale = 1.0f;
+  // autotuned tile override (0 = heuristic; 1..4 = fixed BMxBN config)
+  int tile = 0;
 };
 
 // A compiled model: weight blob on device + op list + arena layout.

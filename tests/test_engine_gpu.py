@@ -230,3 +230,25 @@ def test_engine_resnet50_fp8():
     corr_emu = np.corrcoef(ref8.ravel(), ref16.ravel())[0, 1]
     assert corr_gpu > corr_emu - 0.03, (corr_gpu, corr_emu)
     assert corr_gpu > 0.9, corr_gpu
+
+
+def test_autotune_resnet(rn50_small):
+    """Builder-time tactic selection: chosen codes are valid, the tuned
+    engine still matches the reference, and choices persist in the plan."""
+    import copy
+
+    from trtlab_amd.engine.autotune import autotune_plan
+    from trtlab_amd.engine.reference import run_reference
+    from trtlab_amd.engine.runtime import NativeEngine
+
+    plan = copy.deepcopy(rn50_small)
+    chosen = autotune_plan(plan, reps=10, warmup=3)
+    assert chosen and all(0 <= t <= 4 for t in chosen.values())
+    assert any(d.get("tile", 0) in range(5) for d in plan.ops)
+    eng = NativeEngine(plan)
+    ctx = eng.create_context(capture=True)
+    x = np.random.RandomState(30).randn(*plan.input_shape).astype(np.float32) * 0.5
+    out = ctx.infer(x).astype(np.float32)
+    ref = run_reference(plan, x)
+    err = np.abs(out - ref).max() / max(np.abs(ref).max(), 1e-6)
+    assert err < 0.08, err

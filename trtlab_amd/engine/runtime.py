@@ -25,12 +25,20 @@ from trtlab_amd.engine.planner import EnginePlan
 class NativeEngine:
     """Compiled model resident on one GPU (weights + op plan)."""
 
-    def __init__(self, plan: EnginePlan, device: int = 0):
+    def __init__(self, plan: EnginePlan, device: int = 0,
+                 autotune: bool = False):
         self._C = native()
         if self._C.hip.device_count() == 0:
             raise RuntimeError(
                 "NativeEngine requires a GPU; the HIP extension found no "
                 "devices (do NOT fall back to eager torch on a GPU box)")
+        if autotune:
+            # builder-time tactic selection (TensorRT-builder role); the
+            # chosen tile codes are baked into plan.ops and persist through
+            # the plan cache
+            from trtlab_amd.engine.autotune import autotune_plan
+
+            autotune_plan(plan, device=device)
         self.plan = plan
         self.device = device
         self.engine = self._C.Engine(
