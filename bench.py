@@ -37,8 +37,8 @@ def main():
                     choices=["resnet50", "resnet101", "resnet152", "bert"])
     ap.add_argument("--dtype", default="fp16",
                     choices=["fp16", "int8", "fp8"])
-    ap.add_argument("--autotune", action="store_true",
-                    help="builder-time kernel tactic selection")
+    ap.add_argument("--no-autotune", action="store_true",
+                    help="skip builder-time kernel tactic selection")
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -76,7 +76,8 @@ def main():
     dtype = {"fp16": DT_F16, "int8": DT_I8, "fp8": DT_F8}[args.dtype]
     plan = Planner(dtype=dtype).compile(g)
 
-    eng = NativeEngine(plan, device=local_rank, autotune=args.autotune)
+    eng = NativeEngine(plan, device=local_rank,
+                       autotune=not args.no_autotune)
     if distributed:
         # RCCL weight broadcast at load: rank 0's blob is
         # authoritative; replicas receive over xGMI.
