@@ -30,11 +30,12 @@ void launch_splitk_reduce(int dtype, const float* scratch, void* C,
                           int64_t ldc, int tiles_m, int tiles_n, int splitk,
                           int bm, int bn, int epi, hipStream_t stream);
 
-template <typename T, Epi E, int BM, int BN, bool SPLIT, int NBUF>
+template <typename T, typename OT, Epi E, int BM, int BN, bool SPLIT,
+          int NBUF>
 __global__ __launch_bounds__(256) void gemm_bt_kernel(
-    const T* __restrict__ A, const T* __restrict__ B, T* __restrict__ C,
+    const T* __restrict__ A, const T* __restrict__ B, OT* __restrict__ C,
     const float* __restrict__ scale, const float* __restrict__ bias,
-    const T* __restrict__ residual, float res_scale, int M, int N, int K,
+    const OT* __restrict__ residual, float res_scale, int M, int N, int K,
     int64_t lda, int64_t ldb, int64_t ldc, int tiles_n,
     float* __restrict__ scratch, int splitk, int ktper) {
   constexpr int kABytes = BM * 128;
@@ -110,8 +111,8 @@ __global__ __launch_bounds__(256) void gemm_bt_kernel(
     store_splitk<T, BM, BN>(acc, scratch + (int64_t)bid * BM * BN, lane, wr,
                             wc);
   } else {
-    store_epilogue<T, E, BM, BN>(acc, C, ldc, m0, n0, M, N, scale, bias,
-                                 residual, res_scale, lane, wr, wc);
+    store_epilogue<T, E, BM, BN, OT>(acc, C, ldc, m0, n0, M, N, scale, bias,
+                                     residual, res_scale, lane, wr, wc);
   }
 }
 
@@ -123,7 +124,7 @@ size_t gemm_scratch_bytes(int M, int N, int K) {
   return (size_t)tiles * splitk * cfg.bm * cfg.bn * 4;
 }
 
-template <typename T>
+template <typename T, typename OT = T>
 static void launch_gemm_bt_t(const void* A, const void* B, void* C,
                              const float* scale, const float* bias,
                              const void* residual, float res_scale, int M,
@@ -137,11 +138,11 @@ static void launch_gemm_bt_t(const void* A, const void* B, void* C,
   int ktiles = K / kTileElems<T>;
   int splitk = (!tile && scratch) ? pick_splitk(tiles, ktiles) : 1;
   dim3 block(256);
-  int dtype = std::is_same<T, _Float16>::value
-                  ? 0
-                  : (std::is_same<T, __bf16>::value
-                         ? 1
-                         : (std::is_same<T, int8_t>::value ? 2 : 3));
+  int out_dtype = std::is_same<OT, _Float16>::value
+                      ? 0
+                      : (std::is_same<OT, __bf16>::value
+                             ? 1
+                             : (std::is_same<OT, int8_t>::value ? 2 : 3));
   if (splitk > 1) {
     int ktper = (int)cdiv(ktiles, splitk);
     dim3 grid((unsigned)(tiles * splitk));
@@ -150,21 +151,21 @@ static void launch_gemm_bt_t(const void* A, const void* B, void* C,
       constexpr int BM = decltype(bm)::value;
       constexpr int BN = decltype(bn)::value;
       if (deep)
-        hipLaunchKernelGGL((gemm_bt_kernel<T, Epi::kNone, BM, BN, true, 4>),
+        hipLaunchKernelGGL((gemm_bt_kernel<T, OT, Epi::kNone, BM, BN, true, 4>),
                            grid, block, 0, stream, (const T*)A, (const T*)B,
-                           (T*)C, scale, bias, (const T*)residual, res_scale,
-                           M, N, K, lda, ldb, ldc, tiles_n, scratch, splitk,
-                           ktper);
+                           (OT*)C, scale, bias, (const OT*)residual,
+                           res_scale, M, N, K, lda, ldb, ldc, tiles_n,
+                           scratch, splitk, ktper);
       else
-        hipLaunchKernelGGL((gemm_bt_kernel<T, Epi::kNone, BM, BN, true, 2>),
+        hipLaunchKernelGGL((gemm_bt_kernel<T, OT, Epi::kNone, BM, BN, true, 2>),
                            grid, block, 0, stream, (const T*)A, (const T*)B,
-                           (T*)C, scale, bias, (const T*)residual, res_scale,
-                           M, N, K, lda, ldb, ldc, tiles_n, scratch, splitk,
-                           ktper);
+                           (OT*)C, scale, bias, (const OT*)residual,
+                           res_scale, M, N, K, lda, ldb, ldc, tiles_n,
+                           scratch, splitk, ktper);
     });
-    launch_splitk_reduce(dtype, scratch, C, scale, bias, residual, res_scale,
-                         M, N, ldc, tiles_m, tiles_n, splitk, cfg.bm, cfg.bn,
-                         epi, stream);
+    launch_splitk_reduce(out_dtype, scratch, C, scale, bias, residual,
+                         res_scale, M, N, ldc, tiles_m, tiles_n, splitk,
+                         cfg.bm, cfg.bn, epi, stream);
     return;
   }
   dim3 grid((unsigned)tiles);
@@ -175,17 +176,17 @@ static void launch_gemm_bt_t(const void* A, const void* B, void* C,
       constexpr int BM = decltype(bm)::value;
       constexpr int BN = decltype(bn)::value;
       if (deep)
-        hipLaunchKernelGGL((gemm_bt_kernel<T, EE, BM, BN, false, 4>), grid,
-                           block, 0, stream, (const T*)A, (const T*)B, (T*)C,
-                           scale, bias, (const T*)residual, res_scale, M, N,
-                           K, lda, ldb, ldc, tiles_n, (float*)nullptr, 1,
-                           ktiles);
+        hipLaunchKernelGGL((gemm_bt_kernel<T, OT, EE, BM, BN, false, 4>),
+                           grid, block, 0, stream, (const T*)A, (const T*)B,
+                           (OT*)C, scale, bias, (const OT*)residual,
+                           res_scale, M, N, K, lda, ldb, ldc, tiles_n,
+                           (float*)nullptr, 1, ktiles);
       else
-        hipLaunchKernelGGL((gemm_bt_kernel<T, EE, BM, BN, false, 2>), grid,
-                           block, 0, stream, (const T*)A, (const T*)B, (T*)C,
-                           scale, bias, (const T*)residual, res_scale, M, N,
-                           K, lda, ldb, ldc, tiles_n, (float*)nullptr, 1,
-                           ktiles);
+        hipLaunchKernelGGL((gemm_bt_kernel<T, OT, EE, BM, BN, false, 2>),
+                           grid, block, 0, stream, (const T*)A, (const T*)B,
+                           (OT*)C, scale, bias, (const OT*)residual,
+                           res_scale, M, N, K, lda, ldb, ldc, tiles_n,
+                           (float*)nullptr, 1, ktiles);
     });
   });
 }
@@ -195,17 +196,21 @@ void launch_gemm_bt(int dtype,  // 0 = fp16, 1 = bf16, 2 = int8, 3 = fp8 e4m3
                     const float* bias, const void* residual, float res_scale,
                     int M, int N, int K, int64_t lda, int64_t ldb, int64_t ldc,
                     int epi, hipStream_t stream, int tile, void* scratch) {
-  if (dtype == 2 || dtype == 3) {
+  if (dtype == 2 || dtype == 3 || dtype == 4) {
     if (K % 128 != 0)
       throw std::runtime_error("gemm_bt int8/fp8: K must be a multiple of 128");
     if (dtype == 2)
       launch_gemm_bt_t<int8_t>(A, B, C, scale, bias, residual, res_scale, M,
                                N, K, lda, ldb, ldc, epi, stream, tile,
                                (float*)scratch);
-    else
+    else if (dtype == 3)
       launch_gemm_bt_t<__hip_fp8_e4m3>(A, B, C, scale, bias, residual,
                                        res_scale, M, N, K, lda, ldb, ldc,
                                        epi, stream, tile, (float*)scratch);
+    else  // 4: fp8 compute, fp16 output (transformer projections)
+      launch_gemm_bt_t<__hip_fp8_e4m3, _Float16>(
+          A, B, C, scale, bias, residual, res_scale, M, N, K, lda, ldb, ldc,
+          epi, stream, tile, (float*)scratch);
     return;
   }
   if (K % 64 != 0) throw std::runtime_error("gemm_bt: K must be a multiple of 64");

@@ -16,6 +16,7 @@ enum class Epi : int {
   kScaleBias = 4,      // C = acc*scale[n] + bias[n]          (folded BN)
   kScaleBiasRelu = 5,  // C = relu(acc*scale[n] + bias[n])    (conv+BN+ReLU)
   kScaleBiasAddRelu = 6,  // C = relu(acc*scale[n]+bias[n]+res[m][n])
+  kScaleBiasGelu = 7,     // C = gelu(acc*scale[n] + bias[n])  (fp8 gemms)
 };
 
 __device__ __forceinline__ float gelu_tanh(float x) {
@@ -30,14 +31,15 @@ __device__ __forceinline__ float apply_epi(float acc, float scale, float bias,
                                            float res) {
   float v = acc;
   if constexpr (E == Epi::kScaleBias || E == Epi::kScaleBiasRelu ||
-                E == Epi::kScaleBiasAddRelu)
+                E == Epi::kScaleBiasAddRelu || E == Epi::kScaleBiasGelu)
     v = v * scale;
   if constexpr (E != Epi::kNone) v = v + bias;
   if constexpr (E == Epi::kScaleBiasAddRelu) v = v + res;
   if constexpr (E == Epi::kBiasRelu || E == Epi::kScaleBiasRelu ||
                 E == Epi::kScaleBiasAddRelu)
     v = fmaxf(v, 0.0f);
-  if constexpr (E == Epi::kBiasGelu) v = gelu_tanh(v);
+  if constexpr (E == Epi::kBiasGelu || E == Epi::kScaleBiasGelu)
+    v = gelu_tanh(v);
   return v;
 }
 
@@ -185,13 +187,14 @@ __device__ __forceinline__ void mfma_tile(
 }
 
 // Shared predicated epilogue store. D mapping for 16x16x32 MFMA:
-// col = lane&15, row = (lane>>4)*4 + r.
-template <typename T, Epi E, int BM, int BN>
+// col = lane&15, row = (lane>>4)*4 + r. OT = output/residual element type
+// (defaults to the compute type; fp8-in/fp16-out gemms set OT=_Float16).
+template <typename T, Epi E, int BM, int BN, typename OT = T>
 __device__ __forceinline__ void store_epilogue(
     typename Mfma16x16x32<T>::accv (&acc)[BM / 32][BN / 32],
-    T* __restrict__ C, int64_t ldc, int m0, int n0, int M, int N,
+    OT* __restrict__ C, int64_t ldc, int m0, int n0, int M, int N,
     const float* __restrict__ scale, const float* __restrict__ bias,
-    const T* __restrict__ residual, float res_scale, int lane, int wr,
+    const OT* __restrict__ residual, float res_scale, int lane, int wr,
     int wc) {
   constexpr int MFr = BM / 32, NFr = BN / 32;
 #pragma unroll
@@ -202,7 +205,7 @@ __device__ __forceinline__ void store_epilogue(
       if (col >= N) continue;
       float sc = 1.0f, bi = 0.0f;
       if constexpr (E == Epi::kScaleBias || E == Epi::kScaleBiasRelu ||
-                    E == Epi::kScaleBiasAddRelu)
+                    E == Epi::kScaleBiasAddRelu || E == Epi::kScaleBiasGelu)
         sc = scale[col];
       if constexpr (E != Epi::kNone) bi = bias[col];
 #pragma unroll
@@ -213,7 +216,7 @@ __device__ __forceinline__ void store_epilogue(
         if constexpr (E == Epi::kScaleBiasAddRelu)
           res = (float)residual[(int64_t)row * ldc + col] * res_scale;
         float v = apply_epi<E>((float)acc[i][j][r], sc, bi, res);
-        C[(int64_t)row * ldc + col] = store_cast<T>(v);
+        C[(int64_t)row * ldc + col] = store_cast<OT>(v);
       }
     }
   }
@@ -346,6 +349,7 @@ inline void epi_dispatch(int epi, F&& f) {
     case Epi::kScaleBias: f(std::integral_constant<Epi, Epi::kScaleBias>{}); break;
     case Epi::kScaleBiasRelu: f(std::integral_constant<Epi, Epi::kScaleBiasRelu>{}); break;
     case Epi::kScaleBiasAddRelu: f(std::integral_constant<Epi, Epi::kScaleBiasAddRelu>{}); break;
+    case Epi::kScaleBiasGelu: f(std::integral_constant<Epi, Epi::kScaleBiasGelu>{}); break;
   }
 }
 

@@ -44,7 +44,7 @@ __global__ void splitk_reduce_kernel(const float* __restrict__ scratch,
       if (col >= N) continue;
       float sc = 1.0f, bi = 0.0f;
       if constexpr (E == Epi::kScaleBias || E == Epi::kScaleBiasRelu ||
-                    E == Epi::kScaleBiasAddRelu)
+                    E == Epi::kScaleBiasAddRelu || E == Epi::kScaleBiasGelu)
         sc = scale[col];
       if constexpr (E != Epi::kNone) bi = bias[col];
       float res = 0.0f;

@@ -252,3 +252,28 @@ def test_autotune_resnet(rn50_small):
     ref = run_reference(plan, x)
     err = np.abs(out - ref).max() / max(np.abs(ref).max(), 1e-6)
     assert err < 0.08, err
+
+
+def test_engine_bert_fp8():
+    """fp8 transformer projections (fp8 compute, fp16 out): quality gate
+    against the fp16 reference at the emulation's own level."""
+    from trtlab_amd.engine.planner import DT_F8, Planner
+    from trtlab_amd.engine.reference import run_reference
+    from trtlab_amd.engine.runtime import NativeEngine
+    from trtlab_amd.models import build_bert
+
+    g = build_bert(batch=2, seq=128, layers=2, seed=0)
+    plan8 = Planner(dtype=DT_F8).compile(g)
+    plan16 = Planner().compile(g)
+    assert any(d["kind"] == 1 and d["dtype"] == 4 for d in plan8.ops)
+    eng = NativeEngine(plan8)
+    ctx = eng.create_context(capture=True)
+    x = np.random.RandomState(15).randn(*plan8.input_shape).astype(np.float32)
+    out = ctx.infer(x).astype(np.float32)
+    assert np.isfinite(out).all()
+    ref16 = run_reference(plan16, x)
+    ref8 = run_reference(plan8, x)
+    corr_gpu = np.corrcoef(out.ravel(), ref16.ravel())[0, 1]
+    corr_emu = np.corrcoef(ref8.ravel(), ref16.ravel())[0, 1]
+    assert corr_gpu > corr_emu - 0.05, (corr_gpu, corr_emu)
+    assert corr_gpu > 0.85, corr_gpu

@@ -31,6 +31,7 @@ EPI_BIAS_GELU = 3
 EPI_SCALE_BIAS = 4
 EPI_SCALE_BIAS_RELU = 5
 EPI_SCALE_BIAS_ADD_RELU = 6
+EPI_SCALE_BIAS_GELU = 7
 
 # op kinds — keep in sync with csrc/runtime/runtime.h OpKind
 K_CONV, K_GEMM, K_MAXPOOL, K_GAVGPOOL, K_SOFTMAX, K_LAYERNORM, \
@@ -286,7 +287,7 @@ class Planner:
                 op.w = np.ascontiguousarray(flat, np.float16)
                 op.params["C"] = cpad
                 op.params["Kp"] = kp
-            elif op.kind == K_GEMM:
+            elif op.kind == K_GEMM and op.params.get("int8") is None:
                 w = op.w  # [Nout, K] fp32
                 nout, k = w.shape
                 if k % 64 != 0:

@@ -27,9 +27,11 @@ from trtlab_amd.utils import round_up
 _K_CONV, _K_GEMM, _K_MAXPOOL, _K_GAVGPOOL = 0, 1, 2, 3
 _K_CHANNEL_PAD = 8
 _K_QUANTIZE, _K_DEQUANT = 10, 11
-_EPI_NONE, _EPI_BIAS_RELU = 0, 2
-_EPI_SB, _EPI_SB_RELU, _EPI_SB_ADD_RELU = 4, 5, 6
+_EPI_NONE, _EPI_BIAS, _EPI_BIAS_RELU, _EPI_BIAS_GELU = 0, 1, 2, 3
+_EPI_SB, _EPI_SB_RELU, _EPI_SB_ADD_RELU, _EPI_SB_GELU = 4, 5, 6, 7
 _DT_F16, _DT_I8 = 0, 2
+_DT_F8_F16OUT = 4  # fp8 compute, fp16 output (transformer projections)
+_MIN_GEMM_FLOPS = 256e6  # don't quantize tiny classifier heads
 
 
 def calibrate_amax(g: Graph, sample: np.ndarray) -> Dict[str, float]:
@@ -75,6 +77,30 @@ def calibrate_amax(g: Graph, sample: np.ndarray) -> Dict[str, float]:
                 y = y + torch.from_numpy(n.attrs["bias"])
         elif n.kind == "softmax":
             y = F.softmax(x, dim=-1)
+        elif n.kind == "gelu":
+            y = F.gelu(x, approximate="tanh")
+        elif n.kind == "layernorm":
+            y = F.layer_norm(x, (x.shape[-1],),
+                             torch.from_numpy(n.attrs["gamma"]),
+                             torch.from_numpy(n.attrs["beta"]), n.attrs["eps"])
+        elif n.kind == "add_layernorm":
+            ssum = x + t[n.inputs[1]]
+            y = F.layer_norm(ssum, (ssum.shape[-1],),
+                             torch.from_numpy(n.attrs["gamma"]),
+                             torch.from_numpy(n.attrs["beta"]), n.attrs["eps"])
+        elif n.kind == "attention":
+            a = n.attrs
+            b = x.shape[0] // a["seq"]
+            qkv = x.reshape(b, a["seq"], 3, a["heads"], a["head_dim"])
+            q, k, v = (qkv[:, :, i].permute(0, 2, 1, 3) for i in range(3))
+            att = torch.softmax(
+                q @ k.transpose(-1, -2) / np.sqrt(a["head_dim"]), dim=-1)
+            y = (att @ v).permute(0, 2, 1, 3).reshape(x.shape[0], -1)
+        elif n.kind == "embedding":
+            ids = x.long()
+            tok = torch.from_numpy(n.attrs["tok"])
+            pos = torch.from_numpy(n.attrs["pos"])
+            y = tok[ids] + pos[torch.arange(ids.shape[0]) % n.attrs["seq"]]
         else:
             raise ValueError(f"int8 calibration: unsupported node {n.kind}")
         t[n.output] = y
@@ -107,8 +133,11 @@ def lower_int8(g: Graph, exec_ops: List, shapes: Dict, itemsize: Dict,
     if calib_sample is None:
         rng = np.random.RandomState(1234)
         in_shape = shapes[input_name]
-        calib_sample = (rng.randn(*((min(in_shape[0], 2),) + in_shape[1:]))
-                        .astype(np.float32) * 0.5)
+        if len(in_shape) == 4:  # NHWC images: a small batch suffices
+            cs = (min(in_shape[0], 2),) + tuple(in_shape[1:])
+        else:  # token/row inputs: keep the full [B*S, ...] layout
+            cs = tuple(in_shape)
+        calib_sample = rng.randn(*cs).astype(np.float32) * 0.5
     amax = calibrate_amax(g, calib_sample)
 
     scales: Dict[str, float] = {}  # int8 tensor name -> activation scale
@@ -186,6 +215,39 @@ def lower_int8(g: Graph, exec_ops: List, shapes: Dict, itemsize: Dict,
                                        fmt=fmt)))
             op.inputs = [dq]
             op.params["dtype"] = _DT_F16
+            new_ops.append(op)
+        elif (op.kind == _K_GEMM and fmt == "f8"
+              and op.params["weight_shape"][1] % 128 == 0
+              and op.params["epi"] in (_EPI_NONE, _EPI_BIAS, _EPI_BIAS_GELU)):
+            # fp8 projection GEMM: quantize the fp16 input per-tensor,
+            # fp8 weights per-channel, fp16 output with the dequant folded
+            # into a scale-bias epilogue.
+            src = op.inputs[0]
+            nout, kin = op.params["weight_shape"]
+            m = shapes[src][0]
+            if 2.0 * m * nout * kin < _MIN_GEMM_FLOPS:
+                op.params.setdefault("dtype", _DT_F16)
+                new_ops.append(op)
+                continue
+            s_in = max(amax[src], 1e-6) / qmax
+            qname = src + "_q8"
+            if qname not in shapes:
+                shapes[qname] = shapes[src]
+                itemsize[qname] = 1
+                new_ops.append(ExecOp(_K_QUANTIZE, qname, [src], qname,
+                                      dict(q_scale=s_in, dtype=_DT_F16,
+                                           fmt=fmt)))
+            flat = op.w.astype(np.float32)  # [Nout, K] original fp32
+            q, sw = _quantize_weights(flat, fmt)
+            op.w = q
+            op.inputs[0] = qname
+            op.params["dtype"] = _DT_F8_F16OUT
+            op.params["int8"] = True  # skip the fp16 prepack pass
+            b_ = op.bias if op.bias is not None else np.zeros(nout, np.float32)
+            op.scale = (s_in * sw).astype(np.float32)
+            op.bias = b_.astype(np.float32)
+            epi = op.params["epi"]
+            op.params["epi"] = _EPI_SB_GELU if epi == _EPI_BIAS_GELU else _EPI_SB
             new_ops.append(op)
         else:
             op.params.setdefault("dtype", _DT_F16)

@@ -19,7 +19,7 @@ from trtlab_amd.engine.planner import (
     K_DEQUANT, K_ELEMENTWISE, K_GAVGPOOL, K_GEMM, K_LAYERNORM, K_MAXPOOL,
     K_EMBEDDING, K_QUANTIZE, K_SOFTMAX,
     EPI_BIAS, EPI_BIAS_GELU, EPI_BIAS_RELU, EPI_NONE, EPI_SCALE_BIAS,
-    EPI_SCALE_BIAS_ADD_RELU, EPI_SCALE_BIAS_RELU)
+    EPI_SCALE_BIAS_ADD_RELU, EPI_SCALE_BIAS_GELU, EPI_SCALE_BIAS_RELU)
 
 
 def _fp8_round(x: torch.Tensor) -> torch.Tensor:
@@ -29,7 +29,8 @@ def _fp8_round(x: torch.Tensor) -> torch.Tensor:
 
 def _epi(acc: torch.Tensor, epi: int, scale, bias, res) -> torch.Tensor:
     v = acc
-    if epi in (EPI_SCALE_BIAS, EPI_SCALE_BIAS_RELU, EPI_SCALE_BIAS_ADD_RELU):
+    if epi in (EPI_SCALE_BIAS, EPI_SCALE_BIAS_RELU, EPI_SCALE_BIAS_ADD_RELU,
+               EPI_SCALE_BIAS_GELU):
         v = v * scale
     if epi != EPI_NONE:
         v = v + bias
@@ -37,7 +38,7 @@ def _epi(acc: torch.Tensor, epi: int, scale, bias, res) -> torch.Tensor:
         v = v + res
     if epi in (EPI_BIAS_RELU, EPI_SCALE_BIAS_RELU, EPI_SCALE_BIAS_ADD_RELU):
         v = F.relu(v)
-    if epi == EPI_BIAS_GELU:
+    if epi in (EPI_BIAS_GELU, EPI_SCALE_BIAS_GELU):
         v = F.gelu(v, approximate="tanh")
     return v
 
@@ -98,10 +99,18 @@ def run_reference(plan: EnginePlan, input_nhwc: np.ndarray,
         elif op.kind == K_DEQUANT:
             t[op.output] = x * d["q_scale"]
         elif op.kind == K_GEMM:
-            wt = torch.from_numpy(op.w.astype(np.float32))  # [N, K]
-            acc = x @ wt.t()
-            bias = torch.from_numpy(op.bias) if op.bias is not None else None
-            t[op.output] = _epi(acc, d["epi"], None, bias, None)
+            if d["dtype"] == 4:  # fp8 compute, fp16 out: x is fp8-grid values
+                wt = torch.from_numpy(op.w).view(torch.float8_e4m3fn).float()
+                acc = x @ wt.t()
+                scale = torch.from_numpy(op.scale)
+                bias = torch.from_numpy(op.bias)
+                t[op.output] = _epi(acc, d["epi"], scale, bias, None)
+            else:
+                wt = torch.from_numpy(op.w.astype(np.float32))  # [N, K]
+                acc = x @ wt.t()
+                bias = (torch.from_numpy(op.bias)
+                        if op.bias is not None else None)
+                t[op.output] = _epi(acc, d["epi"], None, bias, None)
         elif op.kind == K_MAXPOOL:
             xc = x.permute(0, 3, 1, 2)
             y = F.max_pool2d(xc, d["KH"], stride=d["sh"], padding=d["ph"])
