@@ -60,18 +60,21 @@ def autotune_plan(plan: EnginePlan, device: int = 0, reps: int = 30,
             if key in chosen:
                 d["tile"] = chosen[key]
                 continue
-            esize = 1 if d["dtype"] >= 2 else 2
+            # element sizes: dtype 2/3 are 1-byte in AND out; dtype 4 is
+            # fp8 in / fp16 OUT (transformer projections)
+            in_esize = 1 if d["dtype"] in (2, 3, 4) else 2
+            out_esize = 1 if d["dtype"] in (2, 3) else 2
             if d["kind"] == K_CONV:
                 oh = (d["H"] + 2 * d["ph"] - d["KH"]) // d["sh"] + 1
                 ow = (d["W"] + 2 * d["pw"] - d["KW"]) // d["sw"] + 1
                 kk = ((d["KH"] * d["KW"] * d["C"] + 127) // 128) * 128
-                a = buf("in", d["Nb"] * d["H"] * d["W"] * d["C"] * esize)
-                w = buf("w", d["Cout"] * kk * esize)
-                o = buf("out", d["Nb"] * oh * ow * d["Cout"] * esize)
+                a = buf("in", d["Nb"] * d["H"] * d["W"] * d["C"] * in_esize)
+                w = buf("w", d["Cout"] * kk * in_esize)
+                o = buf("out", d["Nb"] * oh * ow * d["Cout"] * out_esize)
             else:
-                a = buf("in", d["M"] * d["K"] * esize)
-                w = buf("w", d["N"] * d["K"] * esize)
-                o = buf("out", d["M"] * d["N"] * esize)
+                a = buf("in", d["M"] * d["K"] * in_esize)
+                w = buf("w", d["N"] * d["K"] * in_esize)
+                o = buf("out", d["M"] * d["N"] * out_esize)
             sc = buf("scale", max(d.get("Cout", 0), d.get("N", 0)) * 4)
             bi = buf("bias", max(d.get("Cout", 0), d.get("N", 0)) * 4)
             zp = buf("zero", 256)
