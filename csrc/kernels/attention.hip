@@ -19,10 +19,12 @@
 
 namespace trtlab {
 
-template <typename T>
+// OT: output element type (fp8 e4m3 with out_scale = 1/s_q fuses the
+// producer-side quantization for the following projection GEMM).
+template <typename T, typename OT = T>
 __global__ __launch_bounds__(256) void attention_kernel(
-    const T* __restrict__ qkv, T* __restrict__ out, int B, int S, int H,
-    int D, float scale) {
+    const T* __restrict__ qkv, OT* __restrict__ out, int B, int S, int H,
+    int D, float scale, float out_scale) {
   // LDS: Q [64][64] | K [128][64] | Vt 2x[64][64] | P 4 waves x 2x[16][64]
   __shared__ __attribute__((aligned(16))) char smem[8192 + 16384 * 2 + 16384];
   char* Qs = smem;                  // 8 KiB
@@ -176,7 +178,8 @@ __global__ __launch_bounds__(256) void attention_kernel(
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       int row = q0 + qrow + ((lane >> 4) << 2) + r;
-      out[((int64_t)b * S + row) * hid + h * D + d] = (T)oacc[j][r];
+      out[((int64_t)b * S + row) * hid + h * D + d] =
+          store_cast<OT>(oacc[j][r] * out_scale);
     }
   }
 }
@@ -198,18 +201,26 @@ void launch_attention_probe(int* dbg, int B, int S, int H, int D,
 }
 
 void launch_attention(int dtype, const void* qkv, void* out, int B, int S,
-                      int H, int D, float scale, hipStream_t stream) {
+                      int H, int D, float scale, hipStream_t stream,
+                      int out_dtype, float out_scale) {
   if (S != 128 || D != 64)
     throw std::runtime_error("attention: only S=128, D=64 supported (BERT-base seq128)");
   dim3 grid(B * H * 2);  // 2 query-row blocks per head
   dim3 block(256);
-  if (dtype == 0)
-    hipLaunchKernelGGL((attention_kernel<_Float16>), grid, block, 0, stream,
-                       (const _Float16*)qkv, (_Float16*)out, B, S, H, D,
-                       scale);
-  else
-    hipLaunchKernelGGL((attention_kernel<__bf16>), grid, block, 0, stream,
-                       (const __bf16*)qkv, (__bf16*)out, B, S, H, D, scale);
+  if (dtype == 0) {
+    if (out_dtype == 3)  // fused fp8 output for the projection GEMM
+      hipLaunchKernelGGL((attention_kernel<_Float16, __hip_fp8_e4m3>), grid,
+                         block, 0, stream, (const _Float16*)qkv,
+                         (__hip_fp8_e4m3*)out, B, S, H, D, scale, out_scale);
+    else
+      hipLaunchKernelGGL((attention_kernel<_Float16, _Float16>), grid, block,
+                         0, stream, (const _Float16*)qkv, (_Float16*)out, B,
+                         S, H, D, scale, out_scale);
+  } else {
+    hipLaunchKernelGGL((attention_kernel<__bf16, __bf16>), grid, block, 0,
+                       stream, (const __bf16*)qkv, (__bf16*)out, B, S, H, D,
+                       scale, out_scale);
+  }
 }
 
 }  // namespace trtlab

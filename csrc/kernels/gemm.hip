@@ -35,8 +35,8 @@ template <typename T, typename OT, Epi E, int BM, int BN, bool SPLIT,
 __global__ __launch_bounds__(256) void gemm_bt_kernel(
     const T* __restrict__ A, const T* __restrict__ B, OT* __restrict__ C,
     const float* __restrict__ scale, const float* __restrict__ bias,
-    const OT* __restrict__ residual, float res_scale, int M, int N, int K,
-    int64_t lda, int64_t ldb, int64_t ldc, int tiles_n,
+    const OT* __restrict__ residual, float res_scale, float out_scale,
+    int M, int N, int K, int64_t lda, int64_t ldb, int64_t ldc, int tiles_n,
     float* __restrict__ scratch, int splitk, int ktper) {
   constexpr int kABytes = BM * 128;
   constexpr int kBuf = (BM + BN) * 128;
@@ -112,7 +112,8 @@ __global__ __launch_bounds__(256) void gemm_bt_kernel(
                             wc);
   } else {
     store_epilogue<T, E, BM, BN, OT>(acc, C, ldc, m0, n0, M, N, scale, bias,
-                                     residual, res_scale, lane, wr, wc);
+                                     residual, res_scale, lane, wr, wc,
+                                     out_scale);
   }
 }
 
@@ -127,10 +128,10 @@ size_t gemm_scratch_bytes(int M, int N, int K) {
 template <typename T, typename OT = T>
 static void launch_gemm_bt_t(const void* A, const void* B, void* C,
                              const float* scale, const float* bias,
-                             const void* residual, float res_scale, int M,
-                             int N, int K, int64_t lda, int64_t ldb,
-                             int64_t ldc, int epi, hipStream_t stream,
-                             int tile, float* scratch) {
+                             const void* residual, float res_scale,
+                             float out_scale, int M, int N, int K,
+                             int64_t lda, int64_t ldb, int64_t ldc, int epi,
+                             hipStream_t stream, int tile, float* scratch) {
   TileCfg cfg = tile ? tile_from_code(tile) : pick_tile(M, N);
   int tiles_m = (int)cdiv(M, cfg.bm);
   int tiles_n = (int)cdiv(N, cfg.bn);
@@ -154,14 +155,14 @@ static void launch_gemm_bt_t(const void* A, const void* B, void* C,
         hipLaunchKernelGGL((gemm_bt_kernel<T, OT, Epi::kNone, BM, BN, true, 4>),
                            grid, block, 0, stream, (const T*)A, (const T*)B,
                            (OT*)C, scale, bias, (const OT*)residual,
-                           res_scale, M, N, K, lda, ldb, ldc, tiles_n,
-                           scratch, splitk, ktper);
+                           res_scale, out_scale, M, N, K, lda, ldb, ldc,
+                           tiles_n, scratch, splitk, ktper);
       else
         hipLaunchKernelGGL((gemm_bt_kernel<T, OT, Epi::kNone, BM, BN, true, 2>),
                            grid, block, 0, stream, (const T*)A, (const T*)B,
                            (OT*)C, scale, bias, (const OT*)residual,
-                           res_scale, M, N, K, lda, ldb, ldc, tiles_n,
-                           scratch, splitk, ktper);
+                           res_scale, out_scale, M, N, K, lda, ldb, ldc,
+                           tiles_n, scratch, splitk, ktper);
     });
     launch_splitk_reduce(out_dtype, scratch, C, scale, bias, residual,
                          res_scale, M, N, ldc, tiles_m, tiles_n, splitk,
@@ -179,49 +180,54 @@ static void launch_gemm_bt_t(const void* A, const void* B, void* C,
         hipLaunchKernelGGL((gemm_bt_kernel<T, OT, EE, BM, BN, false, 4>),
                            grid, block, 0, stream, (const T*)A, (const T*)B,
                            (OT*)C, scale, bias, (const OT*)residual,
-                           res_scale, M, N, K, lda, ldb, ldc, tiles_n,
-                           (float*)nullptr, 1, ktiles);
+                           res_scale, out_scale, M, N, K, lda, ldb, ldc,
+                           tiles_n, (float*)nullptr, 1, ktiles);
       else
         hipLaunchKernelGGL((gemm_bt_kernel<T, OT, EE, BM, BN, false, 2>),
                            grid, block, 0, stream, (const T*)A, (const T*)B,
                            (OT*)C, scale, bias, (const OT*)residual,
-                           res_scale, M, N, K, lda, ldb, ldc, tiles_n,
-                           (float*)nullptr, 1, ktiles);
+                           res_scale, out_scale, M, N, K, lda, ldb, ldc,
+                           tiles_n, (float*)nullptr, 1, ktiles);
     });
   });
 }
 
-void launch_gemm_bt(int dtype,  // 0 = fp16, 1 = bf16, 2 = int8, 3 = fp8 e4m3
+void launch_gemm_bt(int dtype,  // 0 fp16, 1 bf16, 2 int8, 3 fp8, 4 fp8->fp16
                     const void* A, const void* B, void* C, const float* scale,
                     const float* bias, const void* residual, float res_scale,
                     int M, int N, int K, int64_t lda, int64_t ldb, int64_t ldc,
-                    int epi, hipStream_t stream, int tile, void* scratch) {
+                    int epi, hipStream_t stream, int tile, void* scratch,
+                    float out_scale) {
   if (dtype == 2 || dtype == 3 || dtype == 4) {
     if (K % 128 != 0)
       throw std::runtime_error("gemm_bt int8/fp8: K must be a multiple of 128");
+    // the split-K reduce path does not apply out_scale; disable split
+    // when a post-activation quant scale is present
+    void* sk = out_scale == 1.0f ? scratch : nullptr;
     if (dtype == 2)
-      launch_gemm_bt_t<int8_t>(A, B, C, scale, bias, residual, res_scale, M,
-                               N, K, lda, ldb, ldc, epi, stream, tile,
-                               (float*)scratch);
+      launch_gemm_bt_t<int8_t>(A, B, C, scale, bias, residual, res_scale,
+                               out_scale, M, N, K, lda, ldb, ldc, epi,
+                               stream, tile, (float*)sk);
     else if (dtype == 3)
       launch_gemm_bt_t<__hip_fp8_e4m3>(A, B, C, scale, bias, residual,
-                                       res_scale, M, N, K, lda, ldb, ldc,
-                                       epi, stream, tile, (float*)scratch);
+                                       res_scale, out_scale, M, N, K, lda,
+                                       ldb, ldc, epi, stream, tile,
+                                       (float*)sk);
     else  // 4: fp8 compute, fp16 output (transformer projections)
       launch_gemm_bt_t<__hip_fp8_e4m3, _Float16>(
-          A, B, C, scale, bias, residual, res_scale, M, N, K, lda, ldb, ldc,
-          epi, stream, tile, (float*)scratch);
+          A, B, C, scale, bias, residual, res_scale, out_scale, M, N, K, lda,
+          ldb, ldc, epi, stream, tile, (float*)sk);
     return;
   }
   if (K % 64 != 0) throw std::runtime_error("gemm_bt: K must be a multiple of 64");
   if (dtype == 0)
-    launch_gemm_bt_t<_Float16>(A, B, C, scale, bias, residual, res_scale, M,
-                               N, K, lda, ldb, ldc, epi, stream, tile,
-                               (float*)scratch);
+    launch_gemm_bt_t<_Float16>(A, B, C, scale, bias, residual, res_scale,
+                               out_scale, M, N, K, lda, ldb, ldc, epi,
+                               stream, tile, (float*)scratch);
   else
-    launch_gemm_bt_t<__bf16>(A, B, C, scale, bias, residual, res_scale, M, N,
-                             K, lda, ldb, ldc, epi, stream, tile,
-                             (float*)scratch);
+    launch_gemm_bt_t<__bf16>(A, B, C, scale, bias, residual, res_scale,
+                             out_scale, M, N, K, lda, ldb, ldc, epi, stream,
+                             tile, (float*)scratch);
 }
 
 }  // namespace trtlab

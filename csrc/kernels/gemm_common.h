@@ -195,7 +195,7 @@ __device__ __forceinline__ void store_epilogue(
     OT* __restrict__ C, int64_t ldc, int m0, int n0, int M, int N,
     const float* __restrict__ scale, const float* __restrict__ bias,
     const OT* __restrict__ residual, float res_scale, int lane, int wr,
-    int wc) {
+    int wc, float out_scale = 1.0f) {
   constexpr int MFr = BM / 32, NFr = BN / 32;
 #pragma unroll
   for (int i = 0; i < MFr; ++i) {
@@ -216,7 +216,7 @@ __device__ __forceinline__ void store_epilogue(
         if constexpr (E == Epi::kScaleBiasAddRelu)
           res = (float)residual[(int64_t)row * ldc + col] * res_scale;
         float v = apply_epi<E>((float)acc[i][j][r], sc, bi, res);
-        C[(int64_t)row * ldc + col] = store_cast<OT>(v);
+        C[(int64_t)row * ldc + col] = store_cast<OT>(v * out_scale);
       }
     }
   }

@@ -10,7 +10,8 @@ void launch_gemm_bt(int dtype, const void* A, const void* B, void* C,
                     const float* scale, const float* bias, const void* residual,
                     float res_scale, int M, int N, int K, int64_t lda,
                     int64_t ldb, int64_t ldc, int epi, hipStream_t stream,
-                    int tile = 0, void* scratch = nullptr);
+                    int tile = 0, void* scratch = nullptr,
+                    float out_scale = 1.0f);
 size_t gemm_scratch_bytes(int M, int N, int K);
 
 void launch_conv2d(int dtype, const void* in, const void* Wt, void* out,
@@ -33,13 +34,17 @@ void launch_gavgpool(int dtype, const void* in, void* out, int Nb, int HW,
 
 void launch_softmax_rows(int dtype, const void* in, void* out, int M, int N,
                          int64_t ld, hipStream_t stream);
+// q_out (optional): fused fp8-e4m3 quantized copy of the output row at
+// scale q_scale (producer-side quantization for fp8 projections).
 void launch_layernorm(int dtype, const void* in, const float* gamma,
                       const float* beta, void* out, int M, int N, int64_t ld,
-                      float eps, hipStream_t stream);
+                      float eps, hipStream_t stream, void* q_out = nullptr,
+                      float q_scale = 0.f);
 void launch_add_layernorm(int dtype, const void* x, const void* res,
                           const float* gamma, const float* beta, void* out,
                           void* sum_out, int M, int N, int64_t ld, float eps,
-                          hipStream_t stream);
+                          hipStream_t stream, void* q_out = nullptr,
+                          float q_scale = 0.f);
 
 void launch_elementwise(int dtype, int op, const void* a, const void* b,
                         void* out, int64_t n, hipStream_t stream);
@@ -48,8 +53,11 @@ void launch_channel_pad(int dtype, const void* in, void* out, int64_t M,
 void launch_cast(int dtype, bool to_f32, const void* in, void* out, int64_t n,
                  hipStream_t stream);
 
+// out_dtype 3 = fp8-e4m3 output at out_scale (fused quantization for the
+// following projection); otherwise the compute dtype.
 void launch_attention(int dtype, const void* qkv, void* out, int B, int S,
-                      int H, int D, float scale, hipStream_t stream);
+                      int H, int D, float scale, hipStream_t stream,
+                      int out_dtype = -1, float out_scale = 1.0f);
 
 void launch_embedding(int dtype, const void* ids, const void* tok,
                       const void* pos, const void* seg, const void* segids,

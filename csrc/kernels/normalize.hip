@@ -5,6 +5,8 @@
 //
 // Layout: one wave per row, 4 waves per block. Each lane owns 8-element
 // chunks (chunk c = lane + 64*p); rows up to 2048 elements (N % 8 == 0).
+#include <hip/hip_fp8.h>
+
 #include "../common.h"
 
 namespace trtlab {
@@ -73,13 +75,26 @@ __global__ void softmax_rows_kernel(const T* __restrict__ in,
   }
 }
 
-// layernorm core shared by the plain and residual-add variants.
+__device__ __forceinline__ void store8_fp8(unsigned char* p,
+                                           const float* src,
+                                           float inv_scale) {
+  unsigned char out[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j)
+    out[j] = __hip_fp8_e4m3(
+                 fminf(fmaxf(src[j] * inv_scale, -448.f), 448.f)).__x;
+  *(uint2*)p = *(const uint2*)out;
+}
+
+// layernorm core shared by the plain and residual-add variants. q_out, if
+// set, receives the fp8-e4m3 quantized row (value / q_scale) — the fused
+// producer-side quantization for fp8 transformer projections.
 template <typename T, bool ADD>
 __device__ __forceinline__ void layernorm_row(
     const T* __restrict__ src, const T* __restrict__ res,
     const float* __restrict__ gamma, const float* __restrict__ beta,
-    T* __restrict__ dst, T* __restrict__ sum_out, int N, float eps,
-    int lane) {
+    T* __restrict__ dst, T* __restrict__ sum_out, int N, float eps, int lane,
+    unsigned char* __restrict__ q_out = nullptr, float q_inv_scale = 1.0f) {
   float v[kMaxChunks][8];
   int nc = 0;
   float s = 0.f;
@@ -123,6 +138,7 @@ __device__ __forceinline__ void layernorm_row(
                      ((const float*)&b1)[j];
     }
     store8(dst + c * 8, v[nc]);
+    if (q_out) store8_fp8(q_out + c * 8, v[nc], q_inv_scale);
   }
 }
 
@@ -130,13 +146,17 @@ template <typename T>
 __global__ void layernorm_kernel(const T* __restrict__ in,
                                  const float* __restrict__ gamma,
                                  const float* __restrict__ beta,
-                                 T* __restrict__ out, int M, int N, int64_t ld,
-                                 float eps) {
+                                 T* __restrict__ out,
+                                 unsigned char* __restrict__ q_out, int M,
+                                 int N, int64_t ld, float eps,
+                                 float q_inv_scale) {
   int lane = threadIdx.x & 63;
   int row = (blockIdx.x * (blockDim.x >> 6)) + (threadIdx.x >> 6);
   if (row >= M) return;
   layernorm_row<T, false>(in + (int64_t)row * ld, nullptr, gamma, beta,
-                          out + (int64_t)row * ld, nullptr, N, eps, lane);
+                          out + (int64_t)row * ld, nullptr, N, eps, lane,
+                          q_out ? q_out + (int64_t)row * ld : nullptr,
+                          q_inv_scale);
 }
 
 template <typename T>
@@ -145,15 +165,19 @@ __global__ void add_layernorm_kernel(const T* __restrict__ x,
                                      const float* __restrict__ gamma,
                                      const float* __restrict__ beta,
                                      T* __restrict__ out,
-                                     T* __restrict__ sum_out, int M, int N,
-                                     int64_t ld, float eps) {
+                                     T* __restrict__ sum_out,
+                                     unsigned char* __restrict__ q_out, int M,
+                                     int N, int64_t ld, float eps,
+                                     float q_inv_scale) {
   int lane = threadIdx.x & 63;
   int row = (blockIdx.x * (blockDim.x >> 6)) + (threadIdx.x >> 6);
   if (row >= M) return;
   layernorm_row<T, true>(x + (int64_t)row * ld, res + (int64_t)row * ld,
                          gamma, beta, out + (int64_t)row * ld,
                          sum_out ? sum_out + (int64_t)row * ld : nullptr, N,
-                         eps, lane);
+                         eps, lane,
+                         q_out ? q_out + (int64_t)row * ld : nullptr,
+                         q_inv_scale);
 }
 
 static inline dim3 rows_grid(int M) { return dim3((unsigned)cdiv(M, 4)); }
@@ -172,32 +196,37 @@ void launch_softmax_rows(int dtype, const void* in, void* out, int M, int N,
 
 void launch_layernorm(int dtype, const void* in, const float* gamma,
                       const float* beta, void* out, int M, int N, int64_t ld,
-                      float eps, hipStream_t stream) {
+                      float eps, hipStream_t stream, void* q_out,
+                      float q_scale) {
   if (N > 2048 || N % 8 != 0) throw std::runtime_error("layernorm: bad N");
+  float inv = q_scale != 0.f ? 1.0f / q_scale : 1.0f;
   if (dtype == 0)
     hipLaunchKernelGGL((layernorm_kernel<_Float16>), rows_grid(M), dim3(256), 0,
                        stream, (const _Float16*)in, gamma, beta, (_Float16*)out,
-                       M, N, ld, eps);
+                       (unsigned char*)q_out, M, N, ld, eps, inv);
   else
     hipLaunchKernelGGL((layernorm_kernel<__bf16>), rows_grid(M), dim3(256), 0,
-                       stream, (const __bf16*)in, gamma, beta, (__bf16*)out, M,
-                       N, ld, eps);
+                       stream, (const __bf16*)in, gamma, beta, (__bf16*)out,
+                       (unsigned char*)q_out, M, N, ld, eps, inv);
 }
 
 void launch_add_layernorm(int dtype, const void* x, const void* res,
                           const float* gamma, const float* beta, void* out,
                           void* sum_out, int M, int N, int64_t ld, float eps,
-                          hipStream_t stream) {
+                          hipStream_t stream, void* q_out, float q_scale) {
   if (N > 2048 || N % 8 != 0) throw std::runtime_error("add_layernorm: bad N");
+  float inv = q_scale != 0.f ? 1.0f / q_scale : 1.0f;
   if (dtype == 0)
     hipLaunchKernelGGL((add_layernorm_kernel<_Float16>), rows_grid(M),
                        dim3(256), 0, stream, (const _Float16*)x,
                        (const _Float16*)res, gamma, beta, (_Float16*)out,
-                       (_Float16*)sum_out, M, N, ld, eps);
+                       (_Float16*)sum_out, (unsigned char*)q_out, M, N, ld,
+                       eps, inv);
   else
     hipLaunchKernelGGL((add_layernorm_kernel<__bf16>), rows_grid(M), dim3(256),
                        0, stream, (const __bf16*)x, (const __bf16*)res, gamma,
-                       beta, (__bf16*)out, (__bf16*)sum_out, M, N, ld, eps);
+                       beta, (__bf16*)out, (__bf16*)sum_out,
+                       (unsigned char*)q_out, M, N, ld, eps, inv);
 }
 
 }  // namespace trtlab

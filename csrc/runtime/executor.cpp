@@ -104,7 +104,7 @@ void ExecutionContext::enqueue_all(hipStream_t s) {
                        Fp(op.scale_off), Fp(op.bias_off), A(op.in2_off),
                        op.res_scale, op.M, op.N, op.K, op.K /*lda*/,
                        op.K /*ldb*/, op.N /*ldc*/, op.epi, s, op.tile,
-                       scratch_);
+                       scratch_, op.q_scale /*post-epilogue out scale*/);
         break;
       case kMaxPool:
         launch_maxpool2d(op.dtype, A(op.in_off), A(op.out_off), op.Nb, op.H,
@@ -125,14 +125,16 @@ void ExecutionContext::enqueue_all(hipStream_t s) {
                             op.N, s);
         break;
       case kLayerNorm:
+        // out2 (optional) = fused fp8 copy at q_scale
         launch_layernorm(op.dtype, A(op.in_off), Fp(op.scale_off),
                          Fp(op.bias_off), A(op.out_off), op.M, op.N, op.N,
-                         op.eps, s);
+                         op.eps, s, A(op.out2_off), op.q_scale);
         break;
       case kAddLayerNorm:
         launch_add_layernorm(op.dtype, A(op.in_off), A(op.in2_off),
                              Fp(op.scale_off), Fp(op.bias_off), A(op.out_off),
-                             A(op.out2_off), op.M, op.N, op.N, op.eps, s);
+                             nullptr /*sum_out*/, op.M, op.N, op.N, op.eps, s,
+                             A(op.out2_off) /*fused fp8 copy*/, op.q_scale);
         break;
       case kElementwise:
         launch_elementwise(op.dtype, op.epi, A(op.in_off), A(op.in2_off),
@@ -143,8 +145,12 @@ void ExecutionContext::enqueue_all(hipStream_t s) {
                            op.C, op.Cout, s);
         break;
       case kAttention:
+        // epi = output dtype flag (3 -> fused fp8 out at 1/q_scale)
         launch_attention(op.dtype, A(op.in_off), A(op.out_off), op.B, op.S,
-                         op.NH, op.HD, op.att_scale, s);
+                         op.NH, op.HD, op.att_scale, s, op.epi,
+                         op.q_scale != 0.f && op.epi == 3
+                             ? 1.0f / op.q_scale
+                             : 1.0f);
         break;
       case kQuantize:
         // op.epi carries the target format (0 = int8, 1 = fp8 e4m3)

@@ -324,6 +324,8 @@ class Planner:
             for t in op.inputs:
                 touch(t, i)
             touch(op.output, i)
+            if "q_out" in op.params:  # fused fp8 second output
+                touch(op.params["q_out"], i)
         output_name = exec_ops[-1].output
         # input live from the start; output live to the end
         s, e = tensors_used[output_name]
@@ -369,7 +371,8 @@ class Planner:
             elif op.kind == K_GEMM:
                 m, k = shapes[op.inputs[0]]
                 nout = shapes[op.output][1]
-                d.update(kind=K_GEMM, epi=op.params["epi"], M=m, N=nout, K=k)
+                d.update(kind=K_GEMM, epi=op.params["epi"], M=m, N=nout, K=k,
+                         q_scale=op.params.get("out_scale", 1.0))
             elif op.kind in (K_MAXPOOL, K_AVGPOOL):
                 ish = shapes[op.inputs[0]]
                 d.update(kind=op.kind, Nb=ish[0], H=ish[1], W=ish[2],
@@ -386,11 +389,15 @@ class Planner:
                 d.update(kind=K_SOFTMAX, M=m, N=ncol)
             elif op.kind == K_LAYERNORM:
                 m, ncol = shapes[op.inputs[0]]
-                d.update(kind=K_LAYERNORM, M=m, N=ncol, eps=op.params["eps"])
+                d.update(kind=K_LAYERNORM, M=m, N=ncol, eps=op.params["eps"],
+                         out2_off=offsets.get(op.params.get("q_out"), -1),
+                         q_scale=op.params.get("q_scale", 0.0))
             elif op.kind == K_ADD_LAYERNORM:
                 m, ncol = shapes[op.inputs[0]]
                 d.update(kind=K_ADD_LAYERNORM, M=m, N=ncol,
-                         eps=op.params["eps"], out2_off=-1)
+                         eps=op.params["eps"],
+                         out2_off=offsets.get(op.params.get("q_out"), -1),
+                         q_scale=op.params.get("q_scale", 0.0))
             elif op.kind == K_ELEMENTWISE:
                 n = 1
                 for s_ in shapes[op.output]:
@@ -415,7 +422,9 @@ class Planner:
                 seq = op.params["seq"]
                 hd = op.params["head_dim"]
                 d.update(kind=K_ATTENTION, B=m // seq, S=seq, NH=heads, HD=hd,
-                         att_scale=1.0 / float(np.sqrt(hd)))
+                         att_scale=1.0 / float(np.sqrt(hd)),
+                         epi=op.params.get("out_dtype", 0),
+                         q_scale=op.params.get("q_scale", 0.0))
             else:
                 raise ValueError(f"bad exec op kind {op.kind}")
             op_dicts.append(d)

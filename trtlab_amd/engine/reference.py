@@ -99,12 +99,15 @@ def run_reference(plan: EnginePlan, input_nhwc: np.ndarray,
         elif op.kind == K_DEQUANT:
             t[op.output] = x * d["q_scale"]
         elif op.kind == K_GEMM:
-            if d["dtype"] == 4:  # fp8 compute, fp16 out: x is fp8-grid values
+            if d["dtype"] in (3, 4):  # fp8 compute (3: fp8 out, 4: fp16 out)
                 wt = torch.from_numpy(op.w).view(torch.float8_e4m3fn).float()
                 acc = x @ wt.t()
                 scale = torch.from_numpy(op.scale)
                 bias = torch.from_numpy(op.bias)
-                t[op.output] = _epi(acc, d["epi"], scale, bias, None)
+                y = _epi(acc, d["epi"], scale, bias, None)
+                if d["dtype"] == 3:
+                    y = _fp8_round(y * d.get("q_scale", 1.0))
+                t[op.output] = y
             else:
                 wt = torch.from_numpy(op.w.astype(np.float32))  # [N, K]
                 acc = x @ wt.t()
@@ -128,12 +131,18 @@ def run_reference(plan: EnginePlan, input_nhwc: np.ndarray,
         elif op.kind == K_LAYERNORM:
             g_ = torch.from_numpy(op.scale)
             b_ = torch.from_numpy(op.bias)
-            t[op.output] = F.layer_norm(x, (x.shape[-1],), g_, b_, d["eps"])
+            y = F.layer_norm(x, (x.shape[-1],), g_, b_, d["eps"])
+            t[op.output] = y
+            if op.params.get("q_out"):
+                t[op.params["q_out"]] = _fp8_round(y / d["q_scale"])
         elif op.kind == K_ADD_LAYERNORM:
             s = x + t[op.inputs[1]]
             g_ = torch.from_numpy(op.scale)
             b_ = torch.from_numpy(op.bias)
-            t[op.output] = F.layer_norm(s, (s.shape[-1],), g_, b_, d["eps"])
+            y = F.layer_norm(s, (s.shape[-1],), g_, b_, d["eps"])
+            t[op.output] = y
+            if op.params.get("q_out"):
+                t[op.params["q_out"]] = _fp8_round(y / d["q_scale"])
         elif op.kind == K_ELEMENTWISE:
             code = d["epi"]
             if code == 0:
@@ -153,6 +162,8 @@ def run_reference(plan: EnginePlan, input_nhwc: np.ndarray,
             v = qkv[:, :, 2].permute(0, 2, 1, 3)
             att = torch.softmax(q @ k.transpose(-1, -2) * d["att_scale"], dim=-1)
             y = (att @ v).permute(0, 2, 1, 3).reshape(b * s, hid)
+            if d.get("epi") == 3:  # fused fp8 output
+                y = _fp8_round(y / d["q_scale"])
             t[op.output] = y
         else:
             raise ValueError(f"bad op kind {op.kind}")
