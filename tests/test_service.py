@@ -155,3 +155,20 @@ def test_shm_zero_copy_input(service_server):
     out = np.frombuffer(resp.output, dtype=np.float16).reshape(2, 4)
     assert np.allclose(out, x * 2)
     c.close()
+
+
+def test_remote_inference_manager(service_server):
+    """Client-side manager mirroring the local surface (reference
+    PyRemoteInferenceManager)."""
+    from trtlab_amd.rpc.remote import RemoteInferenceManager
+
+    m = RemoteInferenceManager(f"127.0.0.1:{service_server.port}")
+    assert m.ready()
+    runner = m.infer_runner("m")
+    x = np.arange(8, dtype=np.float16).reshape(2, 4)
+    out = runner.infer(x).result(timeout=10)
+    assert np.allclose(out, x * 2)
+    # shm transport variant
+    out2 = m.infer_runner("m", use_shm=True).infer(x).result(timeout=10)
+    assert np.allclose(out2, x * 2)
+    m.close()

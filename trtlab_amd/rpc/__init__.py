@@ -11,3 +11,5 @@ from trtlab_amd.rpc.server import (AsyncService, BatchingService, Server,  # noq
                                    StreamingService, UnaryService)
 from trtlab_amd.rpc.client import (AsyncClient, ShmInput, SyncClient,  # noqa: F401
                                    siege)
+from trtlab_amd.rpc.remote import (RemoteInferenceManager,  # noqa: F401
+                                   RemoteInferRunner)
