@@ -51,9 +51,10 @@ def _parse_tensor(buf: bytes) -> tuple[str, np.ndarray]:
         np_dt = {_F32: np.float32, _F16: np.float16, _I64: np.int64}[dtype]
         arr = np.frombuffer(raw, dtype=np_dt)
     elif _T_FLOAT_DATA in d:
-        arr = np.array([struct.unpack("<f", struct.pack("<I", v))[0]
-                        if isinstance(v, int) else v
-                        for v in d[_T_FLOAT_DATA]], dtype=np.float32)
+        arr = np.array(
+            [struct.unpack("<f", struct.pack("<I", v & 0xFFFFFFFF))[0]
+             if isinstance(v, int) else v
+             for v in d[_T_FLOAT_DATA]], dtype=np.float32)
     elif _T_INT64_DATA in d:
         arr = np.array([w.varint_to_sint64(v) for v in d[_T_INT64_DATA]],
                        dtype=np.int64)
