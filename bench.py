@@ -36,7 +36,7 @@ def main():
     ap.add_argument("--model", default="resnet50",
                     choices=["resnet50", "resnet101", "resnet152", "bert"])
     ap.add_argument("--dtype", default="fp16",
-                    choices=["fp16", "int8", "fp8"])
+                    choices=["fp16", "bf16", "int8", "fp8"])
     ap.add_argument("--no-autotune", action="store_true",
                     help="skip builder-time kernel tactic selection")
     args = ap.parse_args()
@@ -71,9 +71,10 @@ def main():
         depth = int(args.model.replace("resnet", ""))
         g = build_resnet(depth, batch=args.batch, image=224, seed=0)
         cfg_extra = {"image": 224}
-    from trtlab_amd.engine.planner import DT_F8
+    from trtlab_amd.engine.planner import DT_BF16, DT_F8
 
-    dtype = {"fp16": DT_F16, "int8": DT_I8, "fp8": DT_F8}[args.dtype]
+    dtype = {"fp16": DT_F16, "bf16": DT_BF16, "int8": DT_I8,
+             "fp8": DT_F8}[args.dtype]
     plan = Planner(dtype=dtype).compile(g)
 
     eng = NativeEngine(plan, device=local_rank,

@@ -147,6 +147,35 @@ def test_int8_plan_structure():
     assert all(d["dtype"] == 0 for d in gemms)  # head stays fp16
 
 
+def test_bf16_plan_structure():
+    """bf16 plans: same graph/offsets as fp16, weight blob re-encoded to
+    bf16 bit patterns (decoded here via torch) while exec_ops stay fp16
+    numeric for the CPU reference."""
+    import torch
+
+    from trtlab_amd.engine.planner import DT_BF16, K_GEMM, Planner
+
+    g = build_resnet(50, batch=1, image=64, seed=0)
+    plan = Planner(dtype=DT_BF16).compile(g)
+    assert plan.input_dtype == "bf16"
+    assert all(d["dtype"] == 1 for d in plan.ops)
+    # decode the head gemm's weights from the blob and compare to exec_ops
+    gd = next(d for d in plan.ops if d["kind"] == K_GEMM)
+    op = next(o for o in plan.exec_ops if o.kind == K_GEMM)
+    n = op.w.size
+    raw = plan.weights[gd["w_off"]:gd["w_off"] + 2 * n].view(np.int16)
+    dec = torch.from_numpy(raw.copy()).view(torch.bfloat16).to(
+        torch.float32).numpy().reshape(op.w.shape)
+    ref = op.w.astype(np.float32)
+    scale = max(np.abs(ref).max(), 1e-6)
+    assert np.abs(dec - ref).max() / scale < 0.01  # bf16 rounding only
+    # fp32 reference executor still runs on the numeric exec_ops
+    from trtlab_amd.engine.reference import run_reference
+
+    x = np.random.RandomState(7).randn(*plan.input_shape).astype(np.float32) * 0.5
+    assert np.isfinite(run_reference(plan, x)).all()
+
+
 def test_fp8_plan_structure():
     from trtlab_amd.engine.planner import DT_F8, K_QUANTIZE, Planner
 

@@ -48,6 +48,15 @@ K_EMBEDDING = 12
 K_AVGPOOL = 13
 
 
+def _bf16_bits(arr: np.ndarray) -> np.ndarray:
+    """fp16/fp32 -> bf16 bit patterns as int16 (numpy has no bf16 dtype;
+    torch supplies the round-to-nearest-even conversion)."""
+    import torch
+
+    t = torch.from_numpy(np.ascontiguousarray(arr, np.float32))
+    return t.to(torch.bfloat16).view(torch.int16).numpy()
+
+
 @dataclass
 class ExecOp:
     kind: int
@@ -90,10 +99,6 @@ class Planner:
 
     def __init__(self, dtype: int = DT_F16, reuse: bool = True,
                  calib_sample=None):
-        if dtype == DT_BF16:
-            raise NotImplementedError(
-                "bf16 engine plans: numpy has no bf16 for the weight blob; "
-                "bf16 kernels are available through the raw ops API")
         self.dtype = dtype
         self.reuse = reuse  # False: disjoint arena slots (debugging)
         self.calib_sample = calib_sample  # int8 activation calibration input
@@ -305,9 +310,23 @@ class Planner:
             blob.extend(np.ascontiguousarray(arr).tobytes())
             return off
 
+        def pack_half(arr: Optional[np.ndarray]) -> int:
+            # compute-dtype tensor slot: fp16 numeric in exec_ops (so the
+            # CPU reference executor and autotune stay numeric), re-encoded
+            # to bf16 bit patterns in the blob for DT_BF16 plans
+            if arr is not None and self.dtype == DT_BF16 and \
+                    arr.dtype in (np.float16, np.float32):
+                return pack(_bf16_bits(arr))
+            return pack(arr)
+
         w_offs: Dict[str, Tuple[int, int, int]] = {}
         for op in exec_ops:
-            w_offs[op.name] = (pack(op.w), pack(op.scale), pack(op.bias))
+            if op.kind == K_EMBEDDING:  # pos/seg tables are compute-dtype
+                w_offs[op.name] = (pack_half(op.w), pack_half(op.scale),
+                                   pack_half(op.bias))
+            else:  # scale/bias are fp32 epilogue params, w is compute-dtype
+                w_offs[op.name] = (pack_half(op.w), pack(op.scale),
+                                   pack(op.bias))
 
         # ---- liveness + arena offsets ----
         tensors_used: Dict[str, Tuple[int, int]] = {}
@@ -446,5 +465,9 @@ class Planner:
             output_shape=shapes[output_name],
             dtype=self.dtype,
             shapes=dict(shapes),
-            input_dtype=g.tensors[input_name].dtype,
+            # bf16 plans read the input buffer as bf16 bits (the host side
+            # converts); integer inputs (BERT token ids) stay as declared
+            input_dtype=("bf16" if self.dtype == DT_BF16 and
+                         g.tensors[input_name].dtype == "f16"
+                         else g.tensors[input_name].dtype),
         )

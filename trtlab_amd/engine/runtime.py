@@ -66,13 +66,19 @@ class NativeContext:
             # per-stage H2D/compute/D2H events (reference
             # TimedBenchmarkWorkspace); recorded inside the captured graph
             self.ctx.set_timing(True)
-        in_dt = {"f16": np.float16, "bf16": np.float16, "i32": np.int32,
-                 "f32": np.float32}[getattr(self.plan, "input_dtype", "f16")]
+        # bf16 plans carry bf16 bit patterns in the bindings (numpy has no
+        # bf16 dtype); infer() converts via torch at the edges
+        in_name = getattr(self.plan, "input_dtype", "f16")
+        self._in_bf16 = in_name == "bf16"
+        self._out_bf16 = getattr(self.plan, "dtype", 0) == 1  # DT_BF16
+        in_dt = {"f16": np.float16, "bf16": np.int16, "i32": np.int32,
+                 "f32": np.float32}[in_name]
         self._in_view = np.frombuffer(
             self.ctx.input_view(self.plan.input_bytes), dtype=in_dt
         ).reshape(self.plan.input_shape)
         self._out_view = np.frombuffer(
-            self.ctx.output_view(self.plan.output_bytes), dtype=np.float16
+            self.ctx.output_view(self.plan.output_bytes),
+            dtype=np.int16 if self._out_bf16 else np.float16
         ).reshape(self.plan.output_shape)
         if capture:
             self.ctx.capture()
@@ -91,11 +97,23 @@ class NativeContext:
         """Synchronous convenience path; returns a COPY of the output (the
         zero-copy `.output` view is only valid while this context lives)."""
         if batch is not None:
-            np.copyto(self._in_view,
-                      batch.astype(self._in_view.dtype, copy=False))
+            if self._in_bf16 and batch.dtype != np.int16:
+                from trtlab_amd.engine.planner import _bf16_bits
+
+                np.copyto(self._in_view,
+                          _bf16_bits(batch).reshape(self._in_view.shape))
+            else:
+                np.copyto(self._in_view,
+                          batch.astype(self._in_view.dtype, copy=False))
         self.ctx.launch()
         self.ctx.synchronize()
-        return np.array(self._out_view, copy=True)
+        out = np.array(self._out_view, copy=True)
+        if self._out_bf16:
+            import torch
+
+            out = torch.from_numpy(out).view(torch.bfloat16).to(
+                torch.float32).numpy()
+        return out
 
     def launch(self):
         self.ctx.launch()
