@@ -88,6 +88,11 @@ def main():
 
     rng = np.random.RandomState(123 + rank)
     batch = (rng.randn(*plan.input_shape) * 0.5).astype(np.float16)
+    if args.dtype == "bf16":
+        # bf16 bindings carry bit patterns (numpy has no bf16 dtype)
+        from trtlab_amd.engine.planner import _bf16_bits
+
+        batch = _bf16_bits(batch).reshape(plan.input_shape)
     for c in ctxs:
         np.copyto(c.input, batch)
 

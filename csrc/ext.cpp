@@ -79,6 +79,17 @@ PYBIND11_MODULE(_C, m) {
   mem.def("pinned_free", [](uintptr_t p, size_t b) { pinned_free((void*)p, b); });
   mem.def("device_bytes_in_use", &device_bytes_in_use);
   mem.def("pinned_bytes_in_use", &pinned_bytes_in_use);
+  mem.def("huge_malloc", [](size_t b, bool pin) {
+    bool hugetlb = false;
+    void* p = huge_malloc(b, pin, &hugetlb);
+    return py::make_tuple((uintptr_t)p, hugetlb);
+  });
+  mem.def("huge_free",
+          [](uintptr_t p, size_t b, bool pin) { huge_free((void*)p, b, pin); });
+  mem.def("huge_bytes_in_use", &huge_bytes_in_use);
+  mem.def("host_view", [](uintptr_t p, size_t bytes) {
+    return py::memoryview::from_memory((void*)p, bytes);
+  });
   mem.def("memcpy_h2d", [](uintptr_t dst, py::buffer src, size_t bytes) {
     py::buffer_info info = src.request();
     TRT_HIP_CHECK(hipMemcpy((void*)dst, info.ptr, bytes, hipMemcpyHostToDevice));

@@ -6,6 +6,10 @@
 //   - BlockPool: fixed-size block pool with O(1) acquire/release free list
 //     (reference memory_pool.h:65)
 //   - byte accounting per memory type (reference tracking.h:183)
+#include <sys/mman.h>
+
+#include <cstring>
+
 #include "runtime.h"
 
 namespace trtlab {
@@ -38,6 +42,45 @@ void pinned_free(void* p, size_t bytes) {
 }
 int64_t device_bytes_in_use() { return g_device_bytes.load(); }
 int64_t pinned_bytes_in_use() { return g_pinned_bytes.load(); }
+
+// ---- huge-page host allocator ----
+// (reference trtlab/memory raw allocator set included a huge-page
+// allocator; here: explicit 2 MiB hugetlb pages when the system pool has
+// them, THP madvise fallback otherwise. The memset commits pages on the
+// calling thread -> first-touch NUMA placement; allocate from an
+// affinity-pinned thread for locality. pin=true hipHostRegisters the range
+// so the GPU can DMA it like hipHostMalloc memory.)
+static std::atomic<int64_t> g_huge_bytes{0};
+static constexpr size_t kHugePageBytes = size_t(2) << 20;
+
+void* huge_malloc(size_t bytes, bool pin, bool* hugetlb_out) {
+  size_t sz = round_up(bytes, kHugePageBytes);
+  void* p = mmap(nullptr, sz, PROT_READ | PROT_WRITE,
+                 MAP_PRIVATE | MAP_ANONYMOUS | MAP_HUGETLB, -1, 0);
+  bool hugetlb = (p != MAP_FAILED);
+  if (!hugetlb) {
+    p = mmap(nullptr, sz, PROT_READ | PROT_WRITE,
+             MAP_PRIVATE | MAP_ANONYMOUS, -1, 0);
+    if (p == MAP_FAILED)
+      throw std::runtime_error("huge_malloc: mmap failed");
+    madvise(p, sz, MADV_HUGEPAGE);
+  }
+  memset(p, 0, sz);  // first-touch commit
+  if (pin) TRT_HIP_CHECK(hipHostRegister(p, sz, hipHostRegisterDefault));
+  g_huge_bytes += (int64_t)sz;
+  if (hugetlb_out) *hugetlb_out = hugetlb;
+  return p;
+}
+
+void huge_free(void* p, size_t bytes, bool pin) {
+  if (!p) return;
+  size_t sz = round_up(bytes, kHugePageBytes);
+  if (pin) (void)hipHostUnregister(p);
+  munmap(p, sz);
+  g_huge_bytes -= (int64_t)sz;
+}
+
+int64_t huge_bytes_in_use() { return g_huge_bytes.load(); }
 
 BlockPool::BlockPool(size_t block_bytes, int count, int device)
     : block_bytes_(round_up(block_bytes, 256)), device_(device) {

@@ -17,6 +17,9 @@ void* pinned_malloc(size_t bytes);
 void pinned_free(void* p, size_t bytes);
 int64_t device_bytes_in_use();
 int64_t pinned_bytes_in_use();
+void* huge_malloc(size_t bytes, bool pin, bool* hugetlb_out);
+void huge_free(void* p, size_t bytes, bool pin);
+int64_t huge_bytes_in_use();
 
 class BlockPool {
  public:

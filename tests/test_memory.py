@@ -54,3 +54,38 @@ def test_arena_planner_serial_chain_reuses():
     planner.add("c", 1000, 2, 3)
     offsets, total = planner.plan()
     assert total <= 2 * 1024
+
+
+def test_huge_page_buffer_host():
+    """Huge-page host allocator: 2 MiB rounding, zero-copy views, byte
+    accounting (reference trtlab/memory huge-page raw allocator)."""
+    from trtlab_amd import native
+    from trtlab_amd.memory import HugePageBuffer
+
+    C = native()
+    before = C.memory.huge_bytes_in_use()
+    b = HugePageBuffer(3 << 20)  # rounds to 2 x 2 MiB
+    assert C.memory.huge_bytes_in_use() - before == 4 << 20
+    assert isinstance(b.hugetlb, bool)  # explicit hugetlb or THP fallback
+    a = b.array(np.float32)
+    a[:] = 1.5
+    assert float(b.array(np.float32)[-1]) == 1.5
+    b.close()
+    assert C.memory.huge_bytes_in_use() == before
+
+
+@pytest.mark.gpu
+def test_huge_page_buffer_pinned_gpu():
+    """pin=True hipHostRegisters the range: usable for GPU DMA."""
+    from trtlab_amd.memory import DeviceBuffer, HugePageBuffer
+
+    b = HugePageBuffer(1 << 20, pin=True)
+    a = b.array(np.float32)
+    a[:] = 2.25
+    d = DeviceBuffer(1 << 20)
+    d.upload(a)
+    out = np.zeros(1 << 18, np.float32)
+    d.download(out)
+    assert out[0] == 2.25 and out[-1] == 2.25
+    d.close()
+    b.close()

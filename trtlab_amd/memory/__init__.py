@@ -14,6 +14,8 @@ trtlab/tensorrt/src/workspace.cc:40-41).
 from __future__ import annotations
 
 import threading
+
+import numpy as np
 from dataclasses import dataclass, field
 from typing import List, Optional, Tuple
 
@@ -78,6 +80,42 @@ class PinnedBuffer:
     def close(self):
         if getattr(self, "ptr", 0):
             self._C.memory.pinned_free(self.ptr, self.nbytes)
+            self.ptr = 0
+
+    def __del__(self):
+        try:
+            self.close()
+        except Exception:
+            pass
+
+
+class HugePageBuffer:
+    """Huge-page-backed host buffer: explicit 2 MiB hugetlb pages when the
+    system pool has them (`hugetlb` attr says which), transparent-huge-page
+    madvise fallback otherwise. First-touch happens on the allocating
+    thread — allocate from an affinity-pinned `ThreadPool` worker for NUMA
+    locality. `pin=True` hipHostRegisters the range so the GPU can DMA it
+    (requires a GPU). Reference: the huge-page raw allocator in
+    trtlab/memory (memory/include/trtlab/memory/allocator/malloc.h family).
+    """
+
+    def __init__(self, nbytes: int, pin: bool = False):
+        from trtlab_amd import native
+
+        self._C = native()
+        self.nbytes = int(nbytes)
+        self.pin = bool(pin)
+        self.ptr, self.hugetlb = self._C.memory.huge_malloc(self.nbytes,
+                                                            self.pin)
+
+    def array(self, dtype=np.uint8) -> np.ndarray:
+        """Zero-copy numpy view of the buffer (valid while it lives)."""
+        return np.frombuffer(
+            self._C.memory.host_view(self.ptr, self.nbytes), dtype=dtype)
+
+    def close(self):
+        if getattr(self, "ptr", 0):
+            self._C.memory.huge_free(self.ptr, self.nbytes, self.pin)
             self.ptr = 0
 
     def __del__(self):
