@@ -93,3 +93,56 @@ def test_wire_format_against_protobuf_runtime():
     w0 = model.graph.initializer[0]
     assert list(w0.dims) == [64, 3, 7, 7]
     assert len(w0.raw_data) == 64 * 3 * 49 * 4
+
+
+def test_onnx_import_passthrough_and_clip_ops():
+    """Importer breadth: Identity/Dropout/Reshape pass-throughs and Clip
+    (attr-form and initializer-form) lowering to relu."""
+    import trtlab_amd.engine.onnx_wire as w
+    from trtlab_amd.engine.onnx_io import (_GRAPH_INIT, _GRAPH_INPUT,
+                                           _GRAPH_NAME, _GRAPH_OUTPUT,
+                                           _MODEL_GRAPH, _attr_f, _attr_i,
+                                           _node, _tensor_bytes, _value_info)
+    from trtlab_amd.engine.reference import run_reference
+    from trtlab_amd.engine.planner import Planner
+
+    rng = np.random.RandomState(0)
+    cw = (rng.randn(16, 8, 3, 3) * 0.2).astype(np.float32)
+    gw = (rng.randn(10, 64) * 0.1).astype(np.float32)
+    fake_min = np.zeros((), np.float32)  # Clip initializer bounds
+
+    nodes = b""
+    nodes += _node("Conv", ["x", "cw"], ["c1"],
+                   # kernel 3, stride 1, pad 1
+                   *[])
+    nodes += _node("Clip", ["c1"], ["r1"], _attr_f("min", 0.0))
+    nodes += _node("Identity", ["r1"], ["i1"])
+    nodes += _node("Conv", ["i1", "cw2"], ["c2"])
+    nodes += _node("Clip", ["c2", "mn"], ["r2"])  # opset-11 style bounds
+    nodes += _node("Dropout", ["r2"], ["d1", "d1_mask"])
+    nodes += _node("GlobalAveragePool", ["d1"], ["gap"])
+    nodes += _node("Reshape", ["gap", "shape0"], ["flat"])
+    nodes += _node("Gemm", ["flat", "gw"], ["y"], _attr_i("transB", 1))
+
+    # second conv 1x1: 8 -> 64 channels so K % 64 holds for the gemm
+    cw2 = (rng.randn(64, 16, 1, 1) * 0.2).astype(np.float32)
+    inits = b""
+    for nm, arr in [("cw", cw), ("cw2", cw2), ("gw", gw),
+                    ("mn", fake_min),
+                    ("shape0", np.array([0, -1], np.int64))]:
+        inits += w.f_bytes(_GRAPH_INIT, _tensor_bytes(nm, arr))
+    in_vi = w.f_bytes(_GRAPH_INPUT, _value_info("x", [2, 8, 8, 8]))
+    out_vi = w.f_bytes(_GRAPH_OUTPUT, _value_info("y", []))
+    graph = nodes + w.f_string(_GRAPH_NAME, "breadth") + inits + in_vi + out_vi
+    data = w.f_varint(1, 8) + w.f_bytes(_MODEL_GRAPH, graph)
+
+    from trtlab_amd.engine.onnx_io import import_onnx
+
+    g = import_onnx(data)
+    kinds = [n.kind for n in g.nodes]
+    assert kinds.count("relu") == 2           # both Clips lowered
+    assert "identity" not in kinds            # pass-throughs erased
+    plan = Planner().compile(g)
+    x = rng.randn(2, 8, 8, 8).astype(np.float32) * 0.5
+    out = run_reference(plan, x)
+    assert out.shape == (2, 10) and np.isfinite(out).all()

@@ -170,8 +170,26 @@ def import_onnx(data: bytes, batch: Optional[int] = None,
                             padding=int(pads[0]))
         elif op == "GlobalAveragePool":
             out = g.global_avgpool(x)
-        elif op == "Flatten":
-            out = x  # GAP output is already [N, C] in the IR
+        elif op in ("Flatten", "Reshape", "Squeeze", "Unsqueeze"):
+            # shape plumbing between GAP and Gemm (torchvision-style heads);
+            # the IR's gavgpool output is already [N, C], so these are views
+            out = x
+        elif op in ("Identity", "Dropout"):
+            out = x  # inference mode: both are pass-through
+        elif op == "Clip":
+            # opset<11: attrs; opset>=11: optional min/max initializer inputs
+            mn = attrs.get("min")
+            mx = attrs.get("max")
+            if mn is None and len(ins) > 1 and ins[1] and ins[1] in inits:
+                mn = float(inits[ins[1]])
+            if mx is None and len(ins) > 2 and ins[2] and ins[2] in inits:
+                mx = float(inits[ins[2]])
+            if (mn is None or mn == 0.0) and (mx is None or mx >= 3e38):
+                out = g.relu(x)
+            else:
+                raise ValueError(
+                    f"ONNX Clip with bounds ({mn}, {mx}) not supported "
+                    "(only ReLU-equivalent clips)")
         elif op in ("Gemm", "MatMul"):
             wt = inits[ins[1]].astype(np.float32)
             if op == "MatMul" or not attrs.get("transB", 0):
