@@ -181,9 +181,9 @@ def import_onnx(data: bytes, batch: Optional[int] = None,
             mn = attrs.get("min")
             mx = attrs.get("max")
             if mn is None and len(ins) > 1 and ins[1] and ins[1] in inits:
-                mn = float(inits[ins[1]])
+                mn = float(np.asarray(inits[ins[1]]).reshape(-1)[0])
             if mx is None and len(ins) > 2 and ins[2] and ins[2] in inits:
-                mx = float(inits[ins[2]])
+                mx = float(np.asarray(inits[ins[2]]).reshape(-1)[0])
             if (mn is None or mn == 0.0) and (mx is None or mx >= 3e38):
                 out = g.relu(x)
             else:
