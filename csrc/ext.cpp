@@ -330,7 +330,7 @@ PYBIND11_MODULE(_C, m) {
       .def(py::init([](int device, py::buffer weights, size_t arena_bytes,
                        std::vector<py::dict> ops_in, int64_t input_off,
                        size_t input_bytes, int64_t output_off,
-                       size_t output_bytes) {
+                       size_t output_bytes, bool managed_weights) {
              py::buffer_info wi = weights.request();
              std::vector<OpDesc> ops;
              ops.reserve(ops_in.size());
@@ -338,11 +338,12 @@ PYBIND11_MODULE(_C, m) {
              return std::make_shared<Engine>(
                  device, wi.ptr, (size_t)(wi.size * wi.itemsize), arena_bytes,
                  std::move(ops), input_off, input_bytes, output_off,
-                 output_bytes);
+                 output_bytes, managed_weights);
            }),
            py::arg("device"), py::arg("weights"), py::arg("arena_bytes"),
            py::arg("ops"), py::arg("input_off"), py::arg("input_bytes"),
-           py::arg("output_off"), py::arg("output_bytes"))
+           py::arg("output_off"), py::arg("output_bytes"),
+           py::arg("managed_weights") = false)
       .def_property_readonly("device", &Engine::device)
       .def_property_readonly("arena_bytes", &Engine::arena_bytes)
       .def_property_readonly("input_bytes", &Engine::input_bytes)

@@ -2,6 +2,8 @@
 // reference's TensorRT workspace/enqueueV2 path, trtlab/tensorrt/src/
 // workspace.cc:21-75: bindings + activation scratch + warm-up + cudaGraph
 // capture + replay — here with our own op plan over CDNA4 kernels).
+#include <cstring>
+
 #include "runtime.h"
 #include "../kernels/launchers.h"
 
@@ -9,9 +11,11 @@ namespace trtlab {
 
 Engine::Engine(int device, const void* weights, size_t weight_bytes,
                size_t arena_bytes, std::vector<OpDesc> ops, int64_t input_off,
-               size_t input_bytes, int64_t output_off, size_t output_bytes)
+               size_t input_bytes, int64_t output_off, size_t output_bytes,
+               bool managed_weights)
     : device_(device),
       weight_bytes_(weight_bytes),
+      managed_weights_(managed_weights),
       arena_bytes_(arena_bytes),
       ops_(std::move(ops)),
       input_off_(input_off),
@@ -19,9 +23,22 @@ Engine::Engine(int device, const void* weights, size_t weight_bytes,
       input_bytes_(input_bytes),
       output_bytes_(output_bytes) {
   TRT_HIP_CHECK(hipSetDevice(device_));
-  weights_ = device_malloc(weight_bytes_ ? weight_bytes_ : 256, device_);
-  if (weight_bytes_)
-    TRT_HIP_CHECK(hipMemcpy(weights_, weights, weight_bytes_, hipMemcpyHostToDevice));
+  size_t wb = weight_bytes_ ? weight_bytes_ : 256;
+  if (managed_weights_) {
+    // reference ManagedRuntime: weights in managed memory, advised
+    // read-mostly and prefetched so steady-state reads hit HBM
+    TRT_HIP_CHECK(hipMallocManaged(&weights_, wb));
+    if (weight_bytes_) {
+      std::memcpy(weights_, weights, weight_bytes_);
+      (void)hipMemAdvise(weights_, wb, hipMemAdviseSetReadMostly, device_);
+      (void)hipMemPrefetchAsync(weights_, wb, device_, 0);
+    }
+  } else {
+    weights_ = device_malloc(wb, device_);
+    if (weight_bytes_)
+      TRT_HIP_CHECK(
+          hipMemcpy(weights_, weights, weight_bytes_, hipMemcpyHostToDevice));
+  }
   zero_page_ = device_malloc(256, device_);
   TRT_HIP_CHECK(hipMemset(zero_page_, 0, 256));
   // Split-K slab workspace: sized to the worst op in the plan.
@@ -39,14 +56,21 @@ Engine::Engine(int device, const void* weights, size_t weight_bytes,
 }
 
 Engine::~Engine() {
-  device_free(weights_, weight_bytes_ ? weight_bytes_ : 256);
+  if (managed_weights_)
+    (void)hipFree(weights_);
+  else
+    device_free(weights_, weight_bytes_ ? weight_bytes_ : 256);
   device_free(zero_page_, 256);
 }
 
 void Engine::upload_weights(const void* src, size_t bytes) {
   if (bytes > weight_bytes_) throw std::runtime_error("upload_weights: too large");
   TRT_HIP_CHECK(hipSetDevice(device_));
-  TRT_HIP_CHECK(hipMemcpy(weights_, src, bytes, hipMemcpyHostToDevice));
+  TRT_HIP_CHECK(hipMemcpy(weights_, src, bytes,
+                          managed_weights_ ? hipMemcpyDefault
+                                           : hipMemcpyHostToDevice));
+  if (managed_weights_)
+    (void)hipMemPrefetchAsync(weights_, bytes, device_, 0);
 }
 
 ExecutionContext::ExecutionContext(std::shared_ptr<Engine> engine)

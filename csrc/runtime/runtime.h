@@ -90,9 +90,15 @@ struct OpDesc {
 // here the "engine" is an explicit op plan over hand-written CDNA4 kernels.
 class Engine {
  public:
+  // managed_weights: place the weight blob in hipMallocManaged memory with
+  // hipMemAdvise(ReadMostly) + device prefetch — the reference's
+  // ManagedRuntime / NvAllocator::use_weights_allocator path
+  // (trtlab/tensorrt/src/allocator.cc:12-56). Default off: explicit device
+  // memory is faster and the MI355X has 288 GB of HBM3E per GPU.
   Engine(int device, const void* weights, size_t weight_bytes,
          size_t arena_bytes, std::vector<OpDesc> ops, int64_t input_off,
-         size_t input_bytes, int64_t output_off, size_t output_bytes);
+         size_t input_bytes, int64_t output_off, size_t output_bytes,
+         bool managed_weights = false);
   ~Engine();
 
   int device() const { return device_; }
@@ -114,6 +120,7 @@ class Engine {
   int device_;
   void* weights_ = nullptr;
   size_t weight_bytes_ = 0;
+  bool managed_weights_ = false;
   void* zero_page_ = nullptr;
   size_t arena_bytes_;
   size_t scratch_bytes_ = 0;  // split-K slab workspace (max over ops)

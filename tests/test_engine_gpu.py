@@ -303,3 +303,15 @@ def test_engine_bert_fp8():
     corr_emu = np.corrcoef(ref8.ravel(), ref16.ravel())[0, 1]
     assert corr_gpu > corr_emu - 0.05, (corr_gpu, corr_emu)
     assert corr_gpu > 0.85, corr_gpu
+
+
+def test_engine_managed_weights(rn50_small):
+    """Weights in hipMallocManaged memory (advised read-mostly + prefetched)
+    must produce bit-identical results to explicit device weights."""
+    from trtlab_amd.engine.runtime import NativeEngine
+
+    plan = rn50_small
+    x = np.random.RandomState(9).randn(*plan.input_shape).astype(np.float32) * 0.5
+    dev = NativeEngine(plan).create_context().infer(x).copy()
+    man = NativeEngine(plan, managed_weights=True).create_context().infer(x)
+    assert np.array_equal(dev, man)

@@ -26,7 +26,11 @@ class NativeEngine:
     """Compiled model resident on one GPU (weights + op plan)."""
 
     def __init__(self, plan: EnginePlan, device: int = 0,
-                 autotune: bool = False):
+                 autotune: bool = False, managed_weights: bool = False):
+        """managed_weights: weight blob in hipMallocManaged memory advised
+        read-mostly (reference ManagedRuntime / NvAllocator weights path,
+        trtlab/tensorrt/src/allocator.cc:12-56). Default off — explicit
+        HBM residency is faster and MI355X has 288 GB per GPU."""
         self._C = native()
         if self._C.hip.device_count() == 0:
             raise RuntimeError(
@@ -43,7 +47,8 @@ class NativeEngine:
         self.device = device
         self.engine = self._C.Engine(
             device, plan.weights, plan.arena_bytes, plan.ops, plan.input_off,
-            plan.input_bytes, plan.output_off, plan.output_bytes)
+            plan.input_bytes, plan.output_off, plan.output_bytes,
+            managed_weights=managed_weights)
 
     def upload_weights(self, blob: np.ndarray) -> None:
         self.engine.upload_weights(blob)
