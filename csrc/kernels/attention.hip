@@ -21,10 +21,15 @@ namespace trtlab {
 
 // OT: output element type (fp8 e4m3 with out_scale = 1/s_q fuses the
 // producer-side quantization for the following projection GEMM).
+// seqlens (optional, [B] device ints): valid token count per sequence for
+// right-padded variable-length batches — keys >= seqlens[b] are masked to
+// -inf before softmax (zero attention weight). Padded query rows still
+// attend to the valid keys; their outputs are ignored downstream.
 template <typename T, typename OT = T>
 __global__ __launch_bounds__(256) void attention_kernel(
     const T* __restrict__ qkv, OT* __restrict__ out, int B, int S, int H,
-    int D, float scale, float out_scale) {
+    int D, float scale, float out_scale,
+    const int* __restrict__ seqlens) {
   // LDS: Q [64][64] | K [128][64] | Vt 2x[64][64] | P 4 waves x 2x[16][64]
   __shared__ __attribute__((aligned(16))) char smem[8192 + 16384 * 2 + 16384];
   char* Qs = smem;                  // 8 KiB
@@ -105,6 +110,24 @@ __global__ __launch_bounds__(256) void attention_kernel(
       kf[f] = *(const typename MF::frag*)(Ks + swz(f * 16 + (lane & 15), kbyte));
 #pragma unroll
     for (int j = 0; j < 8; ++j) sacc[j] = MF::run(qf, kf[j], sacc[j]);
+  }
+
+  // ---- key masking for variable-length sequences ----
+  // score fragment layout: key = j*16 + (lane & 15) (MFMA C col), so the
+  // mask is uniform over r. limit clamped to [1, S] so every query row
+  // keeps at least one finite score (padded queries attend to key 0..L).
+  if (seqlens) {
+    int limit = seqlens[b];
+    limit = limit < 1 ? 1 : (limit > S ? S : limit);
+    if (limit < S) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        if (j * 16 + (lane & 15) >= limit) {
+#pragma unroll
+          for (int r = 0; r < 4; ++r) sacc[j][r] = -3.0e38f;
+        }
+      }
+    }
   }
 
   // ---- row softmax (rows live on 16-lane groups: shfl_xor 1,2,4,8) ----
@@ -202,25 +225,48 @@ void launch_attention_probe(int* dbg, int B, int S, int H, int D,
 
 void launch_attention(int dtype, const void* qkv, void* out, int B, int S,
                       int H, int D, float scale, hipStream_t stream,
-                      int out_dtype, float out_scale) {
+                      int out_dtype, float out_scale, const void* seqlens) {
   if (S != 128 || D != 64)
     throw std::runtime_error("attention: only S=128, D=64 supported (BERT-base seq128)");
   dim3 grid(B * H * 2);  // 2 query-row blocks per head
   dim3 block(256);
+  const int* lens = (const int*)seqlens;
   if (dtype == 0) {
     if (out_dtype == 3)  // fused fp8 output for the projection GEMM
       hipLaunchKernelGGL((attention_kernel<_Float16, __hip_fp8_e4m3>), grid,
                          block, 0, stream, (const _Float16*)qkv,
-                         (__hip_fp8_e4m3*)out, B, S, H, D, scale, out_scale);
+                         (__hip_fp8_e4m3*)out, B, S, H, D, scale, out_scale,
+                         lens);
     else
       hipLaunchKernelGGL((attention_kernel<_Float16, _Float16>), grid, block,
                          0, stream, (const _Float16*)qkv, (_Float16*)out, B,
-                         S, H, D, scale, out_scale);
+                         S, H, D, scale, out_scale, lens);
   } else {
     hipLaunchKernelGGL((attention_kernel<__bf16, __bf16>), grid, block, 0,
                        stream, (const __bf16*)qkv, (__bf16*)out, B, S, H, D,
-                       scale, out_scale);
+                       scale, out_scale, lens);
   }
+}
+
+// ---- seqlens from right-padded token ids ----
+// lens[b] = count of ids[b*S + i] != pad_id (clamped to >= 1). One wave
+// per sequence; the planner schedules this once per forward when any
+// attention op runs in variable-length mode.
+__global__ __launch_bounds__(64) void seqlens_kernel(
+    const int* __restrict__ ids, int* __restrict__ lens, int S, int pad_id) {
+  int b = blockIdx.x;
+  int lane = threadIdx.x;
+  int cnt = 0;
+  for (int i = lane; i < S; i += 64) cnt += (ids[b * S + i] != pad_id);
+#pragma unroll
+  for (int off = 32; off; off >>= 1) cnt += __shfl_down(cnt, off, 64);
+  if (lane == 0) lens[b] = cnt > 0 ? cnt : 1;
+}
+
+void launch_seqlens(const void* ids, void* lens, int B, int S, int pad_id,
+                    hipStream_t stream) {
+  hipLaunchKernelGGL(seqlens_kernel, dim3(B), dim3(64), 0, stream,
+                     (const int*)ids, (int*)lens, S, pad_id);
 }
 
 }  // namespace trtlab

@@ -57,7 +57,13 @@ void launch_cast(int dtype, bool to_f32, const void* in, void* out, int64_t n,
 // following projection); otherwise the compute dtype.
 void launch_attention(int dtype, const void* qkv, void* out, int B, int S,
                       int H, int D, float scale, hipStream_t stream,
-                      int out_dtype = -1, float out_scale = 1.0f);
+                      int out_dtype = -1, float out_scale = 1.0f,
+                      const void* seqlens = nullptr);
+
+// lens[b] = count of non-pad tokens in right-padded ids (>=1); feeds the
+// variable-length attention mask.
+void launch_seqlens(const void* ids, void* lens, int B, int S, int pad_id,
+                    hipStream_t stream);
 
 void launch_embedding(int dtype, const void* ids, const void* tok,
                       const void* pos, const void* seg, const void* segids,

@@ -174,7 +174,12 @@ void ExecutionContext::enqueue_all(hipStream_t s) {
                          op.NH, op.HD, op.att_scale, s, op.epi,
                          op.q_scale != 0.f && op.epi == 3
                              ? 1.0f / op.q_scale
-                             : 1.0f);
+                             : 1.0f,
+                         A(op.in2_off) /*seqlens or null*/);
+        break;
+      case kSeqLens:
+        // op.epi carries pad_id
+        launch_seqlens(A(op.in_off), A(op.out_off), op.B, op.S, op.epi, s);
         break;
       case kQuantize:
         // op.epi carries the target format (0 = int8, 1 = fp8 e4m3)

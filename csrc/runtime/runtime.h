@@ -57,6 +57,7 @@ enum OpKind : int {
   kDequant = 11,    // int8 -> fp16 (scale)
   kEmbedding = 12,  // out = tok[ids] + pos[m%S] (+ seg)
   kAvgPool = 13,
+  kSeqLens = 14,  // ids -> per-sequence valid length (varlen attention)
 };
 
 struct OpDesc {

@@ -29,3 +29,47 @@ def test_bert_reference_runs():
     assert out.shape == plan.output_shape
     assert np.isfinite(out).all()
     assert np.abs(out).max() < 50  # layernorm keeps things bounded
+
+
+def test_varlen_plan_structure_and_noop_equivalence():
+    """varlen BERT: a seqlens op is scheduled once, every attention op reads
+    it; with full-length (pad-free) ids the masked plan must match the
+    unmasked plan exactly."""
+    import numpy as np
+
+    from trtlab_amd.engine.planner import K_ATTENTION, K_SEQLENS, Planner
+    from trtlab_amd.engine.reference import run_reference
+    from trtlab_amd.models import build_bert
+
+    g = build_bert(batch=2, seq=128, layers=2, seed=0, embeddings=True,
+                   varlen=True)
+    plan = Planner().compile(g)
+    kinds = [d["kind"] for d in plan.ops]
+    assert kinds.count(K_SEQLENS) == 1
+    atts = [d for d in plan.ops if d["kind"] == K_ATTENTION]
+    assert len(atts) == 2 and all("in2_off" in d for d in atts)
+
+    rng = np.random.RandomState(3)
+    ids_full = rng.randint(1, 30522, 256).astype(np.int32)  # no pad tokens
+    out_v = run_reference(plan, ids_full)
+
+    g2 = build_bert(batch=2, seq=128, layers=2, seed=0, embeddings=True)
+    plan2 = Planner().compile(g2)
+    out_p = run_reference(plan2, ids_full)
+    assert np.allclose(out_v, out_p), "full-length varlen must be a no-op"
+
+    # padded batch: lens derived correctly, masking changes the result
+    ids_pad = ids_full.copy().reshape(2, 128)
+    ids_pad[1, 57:] = 0  # right-pad sequence 1 to length 57
+    ids_pad = ids_pad.reshape(-1)
+    all_t = run_reference(plan, ids_pad, return_all=True)
+    assert list(all_t["_seqlens"].astype(int)) == [128, 57]
+    out_masked = all_t[plan.output_name]
+    out_unmasked = run_reference(plan2, ids_pad)
+    assert np.isfinite(out_masked).all()
+    # sequence 0 rows see a fully-valid batch -> unaffected by masking of
+    # sequence 1 (attention never crosses sequences)
+    assert np.allclose(out_masked[:128], out_unmasked[:128], atol=1e-5)
+    # sequence 1 valid rows must change (padded keys no longer attended)
+    assert not np.allclose(out_masked[128:128 + 57],
+                           out_unmasked[128:128 + 57], atol=1e-3)

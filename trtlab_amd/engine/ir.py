@@ -171,11 +171,15 @@ class Graph:
                                seq=s_), name)
 
     def attention(self, qkv: str, heads: int, seq: int,
+                  varlen: bool = False, pad_id: int = 0,
                   name: Optional[str] = None) -> str:
+        """varlen: mask keys beyond each sequence's valid length (derived
+        from right-padded token ids; requires an i32 ids graph input)."""
         m, k3 = self.tensors[qkv].shape
         hid = k3 // 3
         return self._emit("attention", [qkv], (m, hid),
-                          dict(heads=heads, seq=seq, head_dim=hid // heads),
+                          dict(heads=heads, seq=seq, head_dim=hid // heads,
+                               varlen=varlen, pad_id=pad_id),
                           name)
 
     # ------------------------------------------------------------ helpers

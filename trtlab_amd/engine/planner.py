@@ -46,6 +46,7 @@ K_QUANTIZE = 10
 K_DEQUANT = 11
 K_EMBEDDING = 12
 K_AVGPOOL = 13
+K_SEQLENS = 14  # token ids -> per-sequence valid length (varlen attention)
 
 
 def _bf16_bits(arr: np.ndarray) -> np.ndarray:
@@ -270,6 +271,28 @@ class Planner:
             t: {"f16": 2, "bf16": 2, "f32": 4, "i32": 4, "i8": 1}[spec.dtype]
             for t, spec in g.tensors.items()
         }
+
+        # ---- variable-length attention: derive per-sequence lengths ----
+        # one seqlens op per forward (ids -> [B] i32 in the arena); every
+        # varlen attention op reads it as a second input (in2_off).
+        att_varlen = [op for op in exec_ops
+                      if op.kind == K_ATTENTION and op.params.get("varlen")]
+        if att_varlen:
+            if g.tensors[input_name].dtype != "i32":
+                raise ValueError(
+                    "varlen attention requires an i32 token-id graph input "
+                    "(build the model with embeddings=True)")
+            seq = att_varlen[0].params["seq"]
+            bsz = shapes[input_name][0] // seq
+            lens_name = "_seqlens"
+            shapes[lens_name] = (bsz,)
+            itemsize[lens_name] = 4
+            exec_ops.insert(0, ExecOp(
+                K_SEQLENS, lens_name, [input_name], lens_name,
+                dict(B=bsz, S=seq,
+                     pad_id=att_varlen[0].params.get("pad_id", 0))))
+            for op in att_varlen:
+                op.inputs.append(lens_name)
         if self.dtype in (DT_I8, DT_F8):
             from trtlab_amd.engine.quantize import lower_int8
 
@@ -428,6 +451,9 @@ class Planner:
             elif op.kind == K_EMBEDDING:
                 m, h = shapes[op.output]
                 d.update(kind=K_EMBEDDING, M=m, S=op.params["seq"], N=h)
+            elif op.kind == K_SEQLENS:
+                d.update(kind=K_SEQLENS, B=op.params["B"], S=op.params["S"],
+                         epi=op.params["pad_id"])
             elif op.kind in (K_QUANTIZE, K_DEQUANT):
                 n = 1
                 for s_ in shapes[op.output]:

@@ -17,7 +17,7 @@ import torch.nn.functional as F
 from trtlab_amd.engine.planner import (
     EnginePlan, K_ADD_LAYERNORM, K_ATTENTION, K_AVGPOOL, K_CHANNEL_PAD, K_CONV,
     K_DEQUANT, K_ELEMENTWISE, K_GAVGPOOL, K_GEMM, K_LAYERNORM, K_MAXPOOL,
-    K_EMBEDDING, K_QUANTIZE, K_SOFTMAX,
+    K_EMBEDDING, K_QUANTIZE, K_SEQLENS, K_SOFTMAX,
     EPI_BIAS, EPI_BIAS_GELU, EPI_BIAS_RELU, EPI_NONE, EPI_SCALE_BIAS,
     EPI_SCALE_BIAS_ADD_RELU, EPI_SCALE_BIAS_GELU, EPI_SCALE_BIAS_RELU)
 
@@ -153,6 +153,10 @@ def run_reference(plan: EnginePlan, input_nhwc: np.ndarray,
                 t[op.output] = x + t[op.inputs[1]]
             elif code == 3:
                 t[op.output] = F.relu(x + t[op.inputs[1]])
+        elif op.kind == K_SEQLENS:
+            bsz, seq = d["B"], d["S"]
+            lens = (x.reshape(bsz, seq).long() != d["epi"]).sum(1)
+            t[op.output] = torch.clamp(lens, min=1)
         elif op.kind == K_ATTENTION:
             b, s, nh, hd = d["B"], d["S"], d["NH"], d["HD"]
             hid = nh * hd
@@ -160,7 +164,13 @@ def run_reference(plan: EnginePlan, input_nhwc: np.ndarray,
             q = qkv[:, :, 0].permute(0, 2, 1, 3)  # [B, NH, S, HD]
             k = qkv[:, :, 1].permute(0, 2, 1, 3)
             v = qkv[:, :, 2].permute(0, 2, 1, 3)
-            att = torch.softmax(q @ k.transpose(-1, -2) * d["att_scale"], dim=-1)
+            scores = q @ k.transpose(-1, -2) * d["att_scale"]
+            if len(op.inputs) > 1:  # varlen: mask right-padded keys
+                lens = t[op.inputs[1]].long()
+                keymask = torch.arange(s)[None, :] >= lens[:, None]  # [B,S]
+                scores = scores.masked_fill(
+                    keymask[:, None, None, :], float("-inf"))
+            att = torch.softmax(scores, dim=-1)
             y = (att @ v).permute(0, 2, 1, 3).reshape(b * s, hid)
             if d.get("epi") == 3:  # fused fp8 output
                 y = _fp8_round(y / d["q_scale"])

@@ -16,10 +16,16 @@ from trtlab_amd.engine.ir import Graph
 def build_bert(batch: int = 8, seq: int = 128, hidden: int = 768,
                layers: int = 12, heads: int = 12, seed: int = 0,
                intermediate: int | None = None, embeddings: bool = False,
+               varlen: bool = False, pad_id: int = 0,
                vocab: int = 30522) -> Graph:
     """embeddings=True: input is int32 token ids [B*S]; the graph starts
     with an embedding gather (tok+pos) + LayerNorm. Otherwise the input is
-    the pre-embedded hidden state [B*S, hidden] fp16."""
+    the pre-embedded hidden state [B*S, hidden] fp16.
+    varlen=True (needs embeddings): right-padded variable-length batches —
+    per-sequence valid lengths are derived on-device from ids != pad_id and
+    padded keys are masked out of every attention softmax."""
+    if varlen and not embeddings:
+        raise ValueError("varlen requires embeddings=True (ids input)")
     inter = intermediate or hidden * 4
     rng = np.random.RandomState(seed)
 
@@ -48,7 +54,8 @@ def build_bert(batch: int = 8, seq: int = 128, hidden: int = 768,
     for li in range(layers):
         qkv = g.gemm(h, w(3 * hidden, hidden), b(3 * hidden),
                      name=f"l{li}_qkv")
-        att = g.attention(qkv, heads=heads, seq=seq, name=f"l{li}_att")
+        att = g.attention(qkv, heads=heads, seq=seq, varlen=varlen,
+                          pad_id=pad_id, name=f"l{li}_att")
         proj = g.gemm(att, w(hidden, hidden), b(hidden), name=f"l{li}_proj")
         ga, ba = ln(hidden)
         h1 = g.add_layernorm(proj, h, ga, ba, name=f"l{li}_ln1")
