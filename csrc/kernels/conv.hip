@@ -171,12 +171,95 @@ size_t conv_scratch_bytes(int Nb, int H, int W, int C, int Cout, int KH,
   return (size_t)tiles * splitk * cfg.bm * cfg.bn * 4;
 }
 
+// ---- direct-to-VGPR small-K fast path ----
+// 1x1/s1/p0 convs with exactly one 128-byte K-tile (K == 64 fp16 elems,
+// e.g. the 16 C=64 1x1 convs in ResNet-50) are pure GEMMs with a single
+// MFMA K-pass. For those the staged kernel's LDS round-trip + barrier IS
+// the ~5-8 us latency floor (profiles/README "known next levers"), so this
+// variant reads A and B fragments straight from global memory — each A
+// fragment row is one 128-B cacheline, B (the 64xK weight panel) stays
+// L2-resident across all M-tiles — and needs no LDS and no barrier.
+// MFMA A/B lane layout: lane holds 8 contiguous k at row lane&15,
+// k = (lane>>4)*8; D: col = lane&15, row = (lane>>4)*4 + r.
+template <typename T, Epi E>
+__global__ __launch_bounds__(256) void conv_smallk_kernel(
+    const T* __restrict__ A, const T* __restrict__ Bw, T* __restrict__ C,
+    const float* __restrict__ scale, const float* __restrict__ bias,
+    const T* __restrict__ residual, int M, int N, int K, float res_scale,
+    int tiles_n) {
+  using MF = Mfma16x16x32<T>;
+  uint32_t blk = xcd_swizzle(blockIdx.x, gridDim.x);
+  int m0 = (int)(blk / tiles_n) * 64, n0 = (int)(blk % tiles_n) * 64;
+  int lane = threadIdx.x & 63, wave = threadIdx.x >> 6;
+  int arow = m0 + wave * 16 + (lane & 15);
+  if (arow >= M) arow = M - 1;  // clamp loads; stores are predicated
+  constexpr int kEps = MF::kStepBytes / (int)sizeof(T);    // 32 (fp16)
+  constexpr int kFe = MF::kFragBytes / (int)sizeof(T);     // 8
+  constexpr int kSteps = 128 / MF::kStepBytes;             // 2
+  f32x4 acc[4];
+#pragma unroll
+  for (int f = 0; f < 4; ++f) acc[f] = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+  for (int s = 0; s < kSteps; ++s) {
+    int kb = s * kEps + (lane >> 4) * kFe;
+    typename MF::frag af =
+        *(const typename MF::frag*)(A + (int64_t)arow * K + kb);
+#pragma unroll
+    for (int f = 0; f < 4; ++f) {
+      int brow = n0 + f * 16 + (lane & 15);
+      if (brow >= N) brow = N - 1;
+      typename MF::frag bf =
+          *(const typename MF::frag*)(Bw + (int64_t)brow * K + kb);
+      acc[f] = MF::run(af, bf, acc[f]);
+    }
+  }
+#pragma unroll
+  for (int f = 0; f < 4; ++f) {
+    int col = n0 + f * 16 + (lane & 15);
+    if (col >= N) continue;
+    float sc = 1.0f, bi = 0.0f;
+    if constexpr (E == Epi::kScaleBias || E == Epi::kScaleBiasRelu ||
+                  E == Epi::kScaleBiasAddRelu || E == Epi::kScaleBiasGelu)
+      sc = scale[col];
+    if constexpr (E != Epi::kNone) bi = bias[col];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int row = m0 + wave * 16 + ((lane >> 4) << 2) + r;
+      if (row >= M) continue;
+      float res = 0.0f;
+      if constexpr (E == Epi::kScaleBiasAddRelu)
+        res = (float)residual[(int64_t)row * N + col] * res_scale;
+      C[(int64_t)row * N + col] =
+          store_cast<T>(apply_epi<E>(acc[f][r], sc, bi, res));
+    }
+  }
+}
+
 template <typename T>
 static void launch_conv2d_t(const void* in, const void* Wt, void* out,
                             const float* scale, const float* bias,
                             const void* residual, const void* zero_page,
                             const ConvParams& p, int epi, hipStream_t stream,
                             int tile, float* scratch) {
+  // small-K fast path (2-byte dtypes; K fits one tile with no zero pad).
+  // Taken regardless of the autotuned tile code: it has no tiling choice.
+  // (constexpr guard keeps the kernel uninstantiated for 1-byte formats,
+  // whose single K-tile would still need zero-fill past C.)
+  if constexpr (sizeof(T) == 2) {
+  if (p.KH == 1 && p.KW == 1 && p.sh == 1 && p.sw == 1 &&
+      p.ph == 0 && p.pw == 0 && p.Kreal == p.K && p.K == kTileElems<T>) {
+    dim3 grid((unsigned)(cdiv(p.M, 64) * cdiv(p.Cout, 64)));
+    int tn = (int)cdiv(p.Cout, 64);
+    epi_dispatch(epi, [&](auto e) {
+      constexpr Epi EE = decltype(e)::value;
+      hipLaunchKernelGGL((conv_smallk_kernel<T, EE>), grid, dim3(256), 0,
+                         stream, (const T*)in, (const T*)Wt, (T*)out, scale,
+                         bias, (const T*)residual, p.M, p.Cout, p.K,
+                         p.res_scale, tn);
+    });
+    return;
+  }
+  }
   TileCfg cfg = tile ? tile_from_code(tile) : pick_tile(p.M, p.Cout);
   int tiles_m = (int)cdiv(p.M, cfg.bm);
   int tiles_n = (int)cdiv(p.Cout, cfg.bn);
