@@ -34,7 +34,8 @@ def main():
     ap.add_argument("--batch", type=int, default=8)
     ap.add_argument("--contexts", type=int, default=3)
     ap.add_argument("--model", default="resnet50",
-                    choices=["resnet50", "resnet101", "resnet152", "bert"])
+                    choices=["resnet50", "resnet101", "resnet152", "bert",
+                             "bert-large"])
     ap.add_argument("--dtype", default="fp16",
                     choices=["fp16", "bf16", "int8", "fp8"])
     ap.add_argument("--no-autotune", action="store_true",
@@ -67,6 +68,10 @@ def main():
     if args.model == "bert":
         g = build_bert(batch=args.batch, seq=128, layers=12, seed=0)
         cfg_extra = {"seq_len": 128, "hidden": 768, "layers": 12}
+    elif args.model == "bert-large":
+        g = build_bert(batch=args.batch, seq=128, hidden=1024, heads=16,
+                       layers=24, seed=0)
+        cfg_extra = {"seq_len": 128, "hidden": 1024, "layers": 24}
     else:
         depth = int(args.model.replace("resnet", ""))
         g = build_resnet(depth, batch=args.batch, image=224, seed=0)

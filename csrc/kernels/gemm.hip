@@ -120,7 +120,7 @@ __global__ __launch_bounds__(256) void gemm_bt_kernel(
 size_t gemm_scratch_bytes(int M, int N, int K) {
   TileCfg cfg = pick_tile(M, N);
   long tiles = cdiv(M, cfg.bm) * cdiv(N, cfg.bn);
-  int splitk = pick_splitk(tiles, K >> 6);  // fp16 tile count (conservative)
+  int splitk = pick_splitk_gemm(tiles, K >> 6);  // fp16 tiles (conservative)
   if (splitk == 1) return 0;
   return (size_t)tiles * splitk * cfg.bm * cfg.bn * 4;
 }
@@ -137,7 +137,7 @@ static void launch_gemm_bt_t(const void* A, const void* B, void* C,
   int tiles_n = (int)cdiv(N, cfg.bn);
   long tiles = (long)tiles_m * tiles_n;
   int ktiles = K / kTileElems<T>;
-  int splitk = (!tile && scratch) ? pick_splitk(tiles, ktiles) : 1;
+  int splitk = (!tile && scratch) ? pick_splitk_gemm(tiles, ktiles) : 1;
   dim3 block(256);
   int out_dtype = std::is_same<OT, _Float16>::value
                       ? 0

@@ -290,6 +290,17 @@ inline int pick_splitk(long blocks, int ktiles) {
   return splitk;
 }
 
+// Plain GEMMs amortize the reduce pass better than convs (contiguous A,
+// no im2col address work in the partials), so the gate is wider: BERT's
+// K=768 projections (12 K-tiles, 48-144 blocks at M=1024) qualify.
+inline int pick_splitk_gemm(long blocks, int ktiles) {
+  if (blocks > 144 || ktiles < 8) return 1;
+  int splitk = 1;
+  while (blocks * splitk < 256 && ktiles / (splitk * 2) >= 3 && splitk < 8)
+    splitk *= 2;
+  return splitk;
+}
+
 // Host-side tile-config choice: prefer the config that fills the chip
 // (>=512 workgroups) at the highest tile utilization; otherwise maximize
 // parallelism x utilization. Small deep-layer shapes (ResNet stage 4/5 at
