@@ -314,6 +314,16 @@ PYBIND11_MODULE(_C, m) {
           py::arg("seg") = 0, py::arg("segids") = 0, py::arg("out") = 0,
           py::arg("M") = 0, py::arg("S") = 0, py::arg("H") = 0,
           py::arg("stream") = 0, py::arg("sync") = true);
+  ops.def("gemm_mxfp8",
+          [](uintptr_t A, uintptr_t B, uintptr_t Sa, uintptr_t Sb,
+             uintptr_t C, int M, int N, int K, uintptr_t stream, bool sync) {
+            launch_gemm_mxfp8((void*)A, (void*)B, (void*)Sa, (void*)Sb,
+                              (void*)C, M, N, K, as_stream(stream));
+            if (sync) TRT_HIP_CHECK(hipStreamSynchronize(as_stream(stream)));
+          },
+          py::arg("A"), py::arg("B"), py::arg("Sa"), py::arg("Sb"),
+          py::arg("C"), py::arg("M"), py::arg("N"), py::arg("K"),
+          py::arg("stream") = 0, py::arg("sync") = true);
   ops.def("attention",
           [](int dtype, uintptr_t qkv, uintptr_t out, int B, int S, int H,
              int D, float scale, uintptr_t stream, bool sync) {

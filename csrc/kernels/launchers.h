@@ -65,6 +65,13 @@ void launch_attention(int dtype, const void* qkv, void* out, int B, int S,
 void launch_seqlens(const void* ids, void* lens, int B, int S, int pad_id,
                     hipStream_t stream);
 
+// MXFP8 (OCP MX: fp8-e4m3 elements + e8m0 per-32-block scales) GEMM on
+// the CDNA4 scaled MFMA 16x16x128. B transposed [N][K]; scales u8
+// [M][K/32] / [N][K/32]; C fp32.
+void launch_gemm_mxfp8(const void* A, const void* B, const void* Sa,
+                       const void* Sb, void* C, int M, int N, int K,
+                       hipStream_t stream);
+
 void launch_embedding(int dtype, const void* ids, const void* tok,
                       const void* pos, const void* seg, const void* segids,
                       void* out, int M, int S, int H, hipStream_t stream);

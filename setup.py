@@ -17,6 +17,7 @@ ARCH = os.environ.get("TRTLAB_GPU_ARCH", "gfx950")
 
 SOURCES = [
     "csrc/kernels/gemm.hip",
+    "csrc/kernels/gemm_mx.hip",
     "csrc/kernels/conv.hip",
     "csrc/kernels/splitk.hip",
     "csrc/kernels/pool.hip",
