@@ -314,6 +314,17 @@ PYBIND11_MODULE(_C, m) {
           py::arg("seg") = 0, py::arg("segids") = 0, py::arg("out") = 0,
           py::arg("M") = 0, py::arg("S") = 0, py::arg("H") = 0,
           py::arg("stream") = 0, py::arg("sync") = true);
+  ops.def("mx_probe",
+          [](uintptr_t A, uintptr_t B, uintptr_t Sa, uintptr_t Sb,
+             uintptr_t D) {
+            launch_mx_probe((void*)A, (void*)B, (void*)Sa, (void*)Sb,
+                            (void*)D, 0);
+            TRT_HIP_CHECK(hipStreamSynchronize(0));
+          });
+  ops.def("mx_frag_dump", [](uintptr_t A, uintptr_t out, int K) {
+    launch_mx_frag_dump((void*)A, (void*)out, K, 0);
+    TRT_HIP_CHECK(hipStreamSynchronize(0));
+  });
   ops.def("gemm_mxfp8",
           [](uintptr_t A, uintptr_t B, uintptr_t Sa, uintptr_t Sb,
              uintptr_t C, int M, int N, int K, uintptr_t stream, bool sync) {

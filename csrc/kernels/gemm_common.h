@@ -236,6 +236,8 @@ __device__ __forceinline__ void wait_tiles_inflight(int ahead) {
       asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
     else if constexpr (G == 6)
       asm volatile("s_waitcnt vmcnt(12)" ::: "memory");
+    else if constexpr (G == 9)
+      asm volatile("s_waitcnt vmcnt(18)" ::: "memory");
     else
       asm volatile("s_waitcnt vmcnt(16)" ::: "memory");
   } else if (ahead == 1) {
@@ -243,6 +245,8 @@ __device__ __forceinline__ void wait_tiles_inflight(int ahead) {
       asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
     else if constexpr (G == 6)
       asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+    else if constexpr (G == 9)
+      asm volatile("s_waitcnt vmcnt(9)" ::: "memory");
     else
       asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
   } else {

@@ -7,115 +7,223 @@
 // C[M,N] = sum_k (a[m,k] * 2^(sa[m,k/32]-127)) * (b[n,k] * 2^(sb[n,k/32]-127))
 //
 // B is prepacked transposed [N][K] like every other "bt" GEMM here; scales
-// are row-major u8 [M][K/32] / [N][K/32]. A/B fragments: 32 bytes per lane
-// (row = lane&15, k = (lane>>4)*32 + j — the 16x16x32 layout scaled 4x in
-// K), staged through the shared 128-byte-row XOR-swizzled LDS tiles
-// (one LDS row == one instruction's K). Lane's 32 elements are exactly one
-// MX block, so the per-lane scale operand is the block scale, selected
-// via opsel byte 0. cbsz/blgp = 0 = fp8 e4m3 for both operands (fp6/fp4
-// use the same instruction with different format codes — planned).
+// are row-major u8 [M][K/32] / [N][K/32]. Operand layout (pinned down
+// empirically — tools/probe_mx.py): lane-group g's 32-byte register holds
+// logical k [g*16, g*16+16) (low half) and [64+g*16, 64+g*16+16) (high
+// half); lane-group g's scale byte (opsel 0) applies to logical MX block
+// g, i.e. k [g*32, g*32+32). Rows stage through the shared 128-byte-row
+// XOR-swizzled LDS tiles (one LDS row == one instruction's K). cbsz/blgp
+// select the element format (0 = fp8 e4m3; fp6/fp4 reuse the same
+// instruction — planned).
 #include "gemm_common.h"
 
 namespace trtlab {
 
 typedef __attribute__((ext_vector_type(8))) int i32x8v;
 
+__device__ __forceinline__ void glds4(const void* gsrc, uint32_t lds_byte) {
+  __builtin_amdgcn_global_load_lds(
+      (const __attribute__((address_space(1))) uint32_t*)gsrc,
+      (__attribute__((address_space(3))) uint32_t*)(uintptr_t)lds_byte, 4, 0,
+      0);
+}
+
 // Read a 32-byte fragment (two swizzled 16-B LDS chunks) for tile row
-// `row` at K-chunk `(lane>>4)`.
+// `row`. Lane-group g's register covers logical k [g*16, g*16+16) in its
+// low 16 bytes and [64+g*16, ...) in its high 16 bytes (pinned down by
+// tools/probe_mx.py: experiment E1 = 48, the split-half signature); the
+// scale byte of lane-group g applies to logical block g (k [g*32,+32)).
 __device__ __forceinline__ i32x8v read_frag32(const char* lds, int row,
                                               int lane) {
-  uint32_t cb0 = (uint32_t)(lane >> 4) * 32;
+  uint32_t g16 = (uint32_t)(lane >> 4) * 16;
   auto swz = [&](uint32_t cb) {
     return (uint32_t)row * 128 + (cb ^ (((uint32_t)row & 7) << 4));
   };
-  i32x4 lo = *(const i32x4*)(lds + swz(cb0));
-  i32x4 hi = *(const i32x4*)(lds + swz(cb0 + 16));
+  i32x4 lo = *(const i32x4*)(lds + swz(g16));
+  i32x4 hi = *(const i32x4*)(lds + swz(64 + g16));
   i32x8v f;
   f[0] = lo[0]; f[1] = lo[1]; f[2] = lo[2]; f[3] = lo[3];
   f[4] = hi[0]; f[5] = hi[1]; f[6] = hi[2]; f[7] = hi[3];
   return f;
 }
 
-// BM = BN = 64, one workgroup = 4 waves; wave w computes rows
-// [w*16, w*16+16) x all 64 cols. One K-tile (128 elems) per MFMA issue.
+// BM = BN = 128, one workgroup = 4 waves; wave (wr, wc) = (w>>1, w&1)
+// computes the [wr*64, +64) x [wc*64, +64) quadrant (4x4 fragment pairs,
+// 16 scaled MFMAs per K-tile from 8 32-byte fragment reads). Double
+// buffered: 2 x (128+128) rows x 128 B = 64 KiB LDS -> 2 blocks/CU.
 __global__ __launch_bounds__(256) void gemm_mxfp8_kernel(
     const uint8_t* __restrict__ A, const uint8_t* __restrict__ B,
     const uint8_t* __restrict__ Sa, const uint8_t* __restrict__ Sb,
     float* __restrict__ C, int M, int N, int K, int tiles_n) {
-  constexpr int kABytes = 64 * 128;
-  __shared__ __attribute__((aligned(16))) char smem[2 * 2 * kABytes];
+  // 3 slots x (A tile + B tile + scale slabs): counted-vmcnt pipeline
+  // keeps 2 tiles' staging in flight behind the MFMAs (9 glds per thread
+  // per tile: 8 data + 1 scale u32).
+  constexpr int kABytes = 128 * 128;
+  constexpr int kSlot = 2 * kABytes + 1024;  // + sa[128] u32 + sb[128] u32
+  __shared__ __attribute__((aligned(16))) char smem[3 * kSlot];
 
   uint32_t bid = xcd_swizzle(blockIdx.x, gridDim.x);
-  int m0 = (int)(bid / tiles_n) * 64;
-  int n0 = (int)(bid % tiles_n) * 64;
+  int m0 = (int)(bid / tiles_n) * 128;
+  int n0 = (int)(bid % tiles_n) * 128;
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
+  const int wr = wave >> 1, wc = wave & 1;
   const int ktiles = K >> 7;       // 128 elems per tile
   const int kblocks = K >> 5;      // 32-elem MX blocks per row
 
-  f32x4 acc[4];
+  f32x4 acc[4][4];
 #pragma unroll
-  for (int f = 0; f < 4; ++f) acc[f] = {0.f, 0.f, 0.f, 0.f};
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int f = 0; f < 4; ++f) acc[i][f] = {0.f, 0.f, 0.f, 0.f};
 
   auto stage = [&](int t, int slot) {
-    char* base = smem + slot * 2 * kABytes;
-    // A rows [m0, m0+64), B rows [n0, n0+64): 64 rows x 8 chunks each.
+    char* base = smem + slot * kSlot;
+    // A rows [m0, m0+128), B rows [n0, n0+128): 128 rows x 8 chunks each.
 #pragma unroll
-    for (int c = 0; c < 2; ++c) {
-      int idx = c * 256 + tid;  // 0..511
+    for (int c = 0; c < 4; ++c) {
+      int idx = c * 256 + tid;  // 0..1023
       uint32_t row = (uint32_t)idx >> 3;
       uint32_t cb = ((uint32_t)idx & 7) * 16;
+      uint32_t kb = cb ^ ((row & 7) << 4);  // source-side swizzle (rule 21)
       int ar = m0 + (int)row;
       if (ar >= M) ar = M - 1;
-      glds16((const char*)(A + (int64_t)ar * K + t * 128 + cb),
-             (uint32_t)(uintptr_t)base + row * 128 + (cb ^ ((row & 7) << 4)));
+      glds16((const char*)(A + (int64_t)ar * K + t * 128 + kb),
+             (uint32_t)(uintptr_t)base + row * 128 + cb);
       int br = n0 + (int)row;
       if (br >= N) br = N - 1;
-      glds16((const char*)(B + (int64_t)br * K + t * 128 + cb),
-             (uint32_t)(uintptr_t)(base + kABytes) + row * 128 +
-                 (cb ^ ((row & 7) << 4)));
+      glds16((const char*)(B + (int64_t)br * K + t * 128 + kb),
+             (uint32_t)(uintptr_t)(base + kABytes) + row * 128 + cb);
+    }
+    // scales: one u32 (4 block-scales) per row per tile, via glds so the
+    // per-thread glds count stays uniform (scattered 1-byte global loads
+    // per-MFMA were the v1 bottleneck: 402 -> 747 TF when staged)
+    if (tid < 128) {
+      int ar = m0 + tid;
+      if (ar >= M) ar = M - 1;
+      glds4(Sa + (int64_t)ar * kblocks + t * 4,
+            (uint32_t)(uintptr_t)(base + 2 * kABytes) + (uint32_t)tid * 4);
+    } else {
+      int br = n0 + tid - 128;
+      if (br >= N) br = N - 1;
+      glds4(Sb + (int64_t)br * kblocks + t * 4,
+            (uint32_t)(uintptr_t)(base + 2 * kABytes) +
+                (uint32_t)tid * 4);  // tid-128 -> second 512-B half
     }
   };
 
   stage(0, 0);
-  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-  __syncthreads();
-  int cur = 0;
-  int arow = m0 + wave * 16 + (lane & 15);
-  if (arow >= M) arow = M - 1;
+  if (ktiles > 1) stage(1, 1);
   for (int t = 0; t < ktiles; ++t) {
-    if (t + 1 < ktiles) stage(t + 1, cur ^ 1);
-    const char* As = smem + cur * 2 * kABytes;
+    int ahead = ktiles - 1 - t;
+    if (ahead > 1) ahead = 1;
+    wait_tiles_inflight<9>(ahead);
+    __builtin_amdgcn_s_barrier();
+    if (t + 2 < ktiles) stage(t + 2, (t + 2) % 3);
+    const char* As = smem + (t % 3) * kSlot;
     const char* Bs = As + kABytes;
-    int kb = t * 4 + (lane >> 4);  // this lane's MX block index
-    i32x8v af = read_frag32(As, wave * 16 + (lane & 15), lane);
-    int sa = Sa[(int64_t)arow * kblocks + kb];
+    const uint8_t* sas = (const uint8_t*)(As + 2 * kABytes);
+    const uint8_t* sbs = sas + 512;
+    int g = lane >> 4;  // lane's MX block within the tile (scale byte idx)
+    i32x8v af[4];
+    int sa[4];
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      int arow = wr * 64 + i * 16 + (lane & 15);
+      af[i] = read_frag32(As, arow, lane);
+      sa[i] = sas[arow * 4 + g];
+    }
 #pragma unroll
     for (int f = 0; f < 4; ++f) {
-      int brow = n0 + f * 16 + (lane & 15);
-      if (brow >= N) brow = N - 1;
-      i32x8v bf = read_frag32(Bs, f * 16 + (lane & 15), lane);
-      int sb = Sb[(int64_t)brow * kblocks + kb];
-      acc[f] = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
-          af, bf, acc[f], 0 /*cbsz: A fp8*/, 0 /*blgp: B fp8*/, 0, sa, 0, sb);
+      int brow = wc * 64 + f * 16 + (lane & 15);
+      i32x8v bf = read_frag32(Bs, brow, lane);
+      int sb = sbs[brow * 4 + g];
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+        acc[i][f] = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
+            af[i], bf, acc[i][f], 0 /*cbsz*/, 0 /*blgp*/, 0, sa[i], 0, sb);
     }
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    __syncthreads();
-    cur ^= 1;
   }
 
 #pragma unroll
-  for (int f = 0; f < 4; ++f) {
-    int col = n0 + f * 16 + (lane & 15);
-    if (col >= N) continue;
+  for (int i = 0; i < 4; ++i)
 #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      int row = m0 + wave * 16 + ((lane >> 4) << 2) + r;
-      if (row >= M) continue;
-      C[(int64_t)row * N + col] = acc[f][r];
+    for (int f = 0; f < 4; ++f) {
+      int col = n0 + wc * 64 + f * 16 + (lane & 15);
+      if (col >= N) continue;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = m0 + wr * 64 + i * 16 + ((lane >> 4) << 2) + r;
+        if (row >= M) continue;
+        C[(int64_t)row * N + col] = acc[i][f][r];
+      }
     }
+}
+
+// Stage 64 rows through the swizzled LDS tile exactly like the GEMM
+// kernel, then dump wave-0's assembled fragments so the host can verify
+// the LDS round trip byte-for-byte.
+__global__ __launch_bounds__(256) void mx_frag_dump_kernel(
+    const uint8_t* __restrict__ A, uint8_t* __restrict__ out, int K) {
+  __shared__ __attribute__((aligned(16))) char smem[64 * 128];
+  const int tid = threadIdx.x;
+#pragma unroll
+  for (int c = 0; c < 2; ++c) {
+    int idx = c * 256 + tid;
+    uint32_t row = (uint32_t)idx >> 3;
+    uint32_t cb = ((uint32_t)idx & 7) * 16;
+    glds16((const char*)(A + (int64_t)row * K + (cb ^ ((row & 7) << 4))),
+           (uint32_t)(uintptr_t)smem + row * 128 + cb);
   }
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+  int lane = tid & 63, wave = tid >> 6;
+  if (wave >= 4) return;
+  i32x8v f = read_frag32(smem, wave * 16 + (lane & 15), lane);
+  *(i32x8v*)(out + (wave * 64 + lane) * 32) = f;
+}
+
+void launch_mx_frag_dump(const void* A, void* out, int K, hipStream_t s) {
+  hipLaunchKernelGGL(mx_frag_dump_kernel, dim3(1), dim3(256), 0, s,
+                     (const uint8_t*)A, (uint8_t*)out, K);
+}
+
+// Layout probe: one wave, direct global loads, no LDS. M=N=16, K=128.
+// Lane l loads its register halves from logical k [g*16,+16) and
+// [64+g*16,+16); scales Sa[row][g]. Dumps acc for host-side hypothesis
+// falsification (tools/probe_mx.py).
+__global__ __launch_bounds__(64) void mx_probe_kernel(
+    const uint8_t* __restrict__ A, const uint8_t* __restrict__ B,
+    const uint8_t* __restrict__ Sa, const uint8_t* __restrict__ Sb,
+    float* __restrict__ D) {
+  int lane = threadIdx.x;
+  int row = lane & 15, g = lane >> 4;
+  i32x8v af, bf;
+  {
+    i32x4 lo = *(const i32x4*)(A + row * 128 + g * 16);
+    i32x4 hi = *(const i32x4*)(A + row * 128 + 64 + g * 16);
+    af[0] = lo[0]; af[1] = lo[1]; af[2] = lo[2]; af[3] = lo[3];
+    af[4] = hi[0]; af[5] = hi[1]; af[6] = hi[2]; af[7] = hi[3];
+    lo = *(const i32x4*)(B + row * 128 + g * 16);
+    hi = *(const i32x4*)(B + row * 128 + 64 + g * 16);
+    bf[0] = lo[0]; bf[1] = lo[1]; bf[2] = lo[2]; bf[3] = lo[3];
+    bf[4] = hi[0]; bf[5] = hi[1]; bf[6] = hi[2]; bf[7] = hi[3];
+  }
+  int sa = Sa[row * 4 + g];
+  int sb = Sb[row * 4 + g];
+  f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+  acc = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(af, bf, acc, 0, 0,
+                                                         0, sa, 0, sb);
+#pragma unroll
+  for (int r = 0; r < 4; ++r) D[((g << 2) + r) * 16 + row] = acc[r];
+}
+
+void launch_mx_probe(const void* A, const void* B, const void* Sa,
+                     const void* Sb, void* D, hipStream_t stream) {
+  hipLaunchKernelGGL(mx_probe_kernel, dim3(1), dim3(64), 0, stream,
+                     (const uint8_t*)A, (const uint8_t*)B, (const uint8_t*)Sa,
+                     (const uint8_t*)Sb, (float*)D);
 }
 
 void launch_gemm_mxfp8(const void* A, const void* B, const void* Sa,
@@ -123,8 +231,8 @@ void launch_gemm_mxfp8(const void* A, const void* B, const void* Sa,
                        hipStream_t stream) {
   if (K % 128 != 0)
     throw std::runtime_error("gemm_mxfp8: K must be a multiple of 128");
-  int tiles_n = (int)cdiv(N, 64);
-  dim3 grid((unsigned)(cdiv(M, 64) * tiles_n));
+  int tiles_n = (int)cdiv(N, 128);
+  dim3 grid((unsigned)(cdiv(M, 128) * tiles_n));
   hipLaunchKernelGGL(gemm_mxfp8_kernel, grid, dim3(256), 0, stream,
                      (const uint8_t*)A, (const uint8_t*)B,
                      (const uint8_t*)Sa, (const uint8_t*)Sb, (float*)C, M, N,

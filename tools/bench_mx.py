@@ -5,7 +5,11 @@ Times square GEMMs and reports effective TFLOP/s next to the fp16
 16x16x32 path for context. Run on an MI355X:
     python tools/bench_mx.py
 """
+import os
+import sys
 import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import numpy as np
 import torch
