@@ -325,6 +325,23 @@ PYBIND11_MODULE(_C, m) {
     launch_mx_frag_dump((void*)A, (void*)out, K, 0);
     TRT_HIP_CHECK(hipStreamSynchronize(0));
   });
+  ops.def("gemm_mxfp4",
+          [](uintptr_t A, uintptr_t B, uintptr_t Sa, uintptr_t Sb,
+             uintptr_t C, int M, int N, int K, uintptr_t stream, bool sync) {
+            launch_gemm_mxfp4((void*)A, (void*)B, (void*)Sa, (void*)Sb,
+                              (void*)C, M, N, K, as_stream(stream));
+            if (sync) TRT_HIP_CHECK(hipStreamSynchronize(as_stream(stream)));
+          },
+          py::arg("A"), py::arg("B"), py::arg("Sa"), py::arg("Sb"),
+          py::arg("C"), py::arg("M"), py::arg("N"), py::arg("K"),
+          py::arg("stream") = 0, py::arg("sync") = true);
+  ops.def("mx4_probe",
+          [](uintptr_t A, uintptr_t B, uintptr_t Sa, uintptr_t Sb,
+             uintptr_t D) {
+            launch_mx4_probe((void*)A, (void*)B, (void*)Sa, (void*)Sb,
+                             (void*)D, 0);
+            TRT_HIP_CHECK(hipStreamSynchronize(0));
+          });
   ops.def("gemm_mxfp8",
           [](uintptr_t A, uintptr_t B, uintptr_t Sa, uintptr_t Sb,
              uintptr_t C, int M, int N, int K, uintptr_t stream, bool sync) {

@@ -51,6 +51,7 @@ __device__ __forceinline__ float apply_epi(float acc, float scale, float bias,
 typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8v;
 typedef __attribute__((ext_vector_type(16))) char i8x16v;
 typedef __attribute__((ext_vector_type(4))) int i32x4;
+typedef __attribute__((ext_vector_type(2))) int i32x2;
 
 template <typename T>
 struct Mfma16x16x32;
@@ -238,6 +239,8 @@ __device__ __forceinline__ void wait_tiles_inflight(int ahead) {
       asm volatile("s_waitcnt vmcnt(12)" ::: "memory");
     else if constexpr (G == 9)
       asm volatile("s_waitcnt vmcnt(18)" ::: "memory");
+    else if constexpr (G == 10)
+      asm volatile("s_waitcnt vmcnt(20)" ::: "memory");
     else
       asm volatile("s_waitcnt vmcnt(16)" ::: "memory");
   } else if (ahead == 1) {
@@ -247,6 +250,8 @@ __device__ __forceinline__ void wait_tiles_inflight(int ahead) {
       asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
     else if constexpr (G == 9)
       asm volatile("s_waitcnt vmcnt(9)" ::: "memory");
+    else if constexpr (G == 10)
+      asm volatile("s_waitcnt vmcnt(10)" ::: "memory");
     else
       asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
   } else {

@@ -219,6 +219,182 @@ __global__ __launch_bounds__(64) void mx_probe_kernel(
   for (int r = 0; r < 4; ++r) D[((g << 2) + r) * 16 + row] = acc[r];
 }
 
+// ---- MXFP4 (fp4 e2m1, e8m0 scales per 32 elems) ----
+// Same scaled MFMA with cbsz/blgp = 4. fp4 operand layout is SIMPLER than
+// fp8 (tools/probe_mx4.py): lane-group g's 16 packed bytes (low 4 dwords,
+// upper 4 zero) cover logical k [g*32, g*32+32) contiguously, and its
+// scale byte covers exactly that block. One 128-byte LDS row holds TWO
+// 128-k instruction windows (256 logical k); K must be % 256.
+__global__ __launch_bounds__(256) void gemm_mxfp4_kernel(
+    const uint8_t* __restrict__ A, const uint8_t* __restrict__ B,
+    const uint8_t* __restrict__ Sa, const uint8_t* __restrict__ Sb,
+    float* __restrict__ C, int M, int N, int K, int tiles_n) {
+  constexpr int kABytes = 128 * 128;
+  constexpr int kSlot = 2 * kABytes + 2048;  // + 8 scale bytes/row (A, B)
+  __shared__ __attribute__((aligned(16))) char smem[3 * kSlot];
+
+  uint32_t bid = xcd_swizzle(blockIdx.x, gridDim.x);
+  int m0 = (int)(bid / tiles_n) * 128;
+  int n0 = (int)(bid % tiles_n) * 128;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wr = wave >> 1, wc = wave & 1;
+  const int kb2 = K >> 1;          // packed bytes per row
+  const int ktiles = K >> 8;       // 256 logical elems per LDS tile
+  const int kblocks = K >> 5;
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int f = 0; f < 4; ++f) acc[i][f] = {0.f, 0.f, 0.f, 0.f};
+
+  auto stage = [&](int t, int slot) {
+    char* base = smem + slot * kSlot;
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+      int idx = c * 256 + tid;  // 0..1023
+      uint32_t row = (uint32_t)idx >> 3;
+      uint32_t cb = ((uint32_t)idx & 7) * 16;
+      uint32_t kb = cb ^ ((row & 7) << 4);  // source-side swizzle
+      int ar = m0 + (int)row;
+      if (ar >= M) ar = M - 1;
+      glds16((const char*)(A + (int64_t)ar * kb2 + t * 128 + kb),
+             (uint32_t)(uintptr_t)base + row * 128 + cb);
+      int br = n0 + (int)row;
+      if (br >= N) br = N - 1;
+      glds16((const char*)(B + (int64_t)br * kb2 + t * 128 + kb),
+             (uint32_t)(uintptr_t)(base + kABytes) + row * 128 + cb);
+    }
+    // 8 e8m0 bytes per row per 256-k tile, as four 512-B slabs
+    // (A-window0 | A-window1 | B-window0 | B-window1). global_load_lds
+    // packs lanes at consecutive 4-B slots from a uniform base, so each
+    // slab gets its own call with a lane-natural destination.
+    uint32_t sbase = (uint32_t)(uintptr_t)(base + 2 * kABytes);
+    if (tid < 128) {
+      int ar = m0 + tid;
+      if (ar >= M) ar = M - 1;
+      glds4(Sa + (int64_t)ar * kblocks + t * 8, sbase + (uint32_t)tid * 4);
+      glds4(Sa + (int64_t)ar * kblocks + t * 8 + 4,
+            sbase + 512 + (uint32_t)tid * 4);
+    } else {
+      int br = n0 + tid - 128;
+      if (br >= N) br = N - 1;
+      glds4(Sb + (int64_t)br * kblocks + t * 8,
+            sbase + 1024 + (uint32_t)(tid - 128) * 4);
+      glds4(Sb + (int64_t)br * kblocks + t * 8 + 4,
+            sbase + 1536 + (uint32_t)(tid - 128) * 4);
+    }
+  };
+
+  auto frag16 = [&](const char* lds, int row, int w) {
+    uint32_t cb = (uint32_t)w * 64 + (uint32_t)(lane >> 4) * 16;
+    i32x4 v = *(const i32x4*)(
+        lds + (uint32_t)row * 128 + (cb ^ (((uint32_t)row & 7) << 4)));
+    i32x8v f = {v[0], v[1], v[2], v[3], 0, 0, 0, 0};
+    return f;
+  };
+
+  stage(0, 0);
+  if (ktiles > 1) stage(1, 1);
+  for (int t = 0; t < ktiles; ++t) {
+    int ahead = ktiles - 1 - t;
+    if (ahead > 1) ahead = 1;
+    wait_tiles_inflight<10>(ahead);
+    __builtin_amdgcn_s_barrier();
+    if (t + 2 < ktiles) stage(t + 2, (t + 2) % 3);
+    const char* As = smem + (t % 3) * kSlot;
+    const char* Bs = As + kABytes;
+    const uint8_t* sas = (const uint8_t*)(As + 2 * kABytes);
+    const uint8_t* sbs = sas + 1024;
+    int g = lane >> 4;
+#pragma unroll
+    for (int w = 0; w < 2; ++w) {
+      i32x8v af[4];
+      int sa[4];
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        int arow = wr * 64 + i * 16 + (lane & 15);
+        af[i] = frag16(As, arow, w);
+        sa[i] = sas[w * 512 + arow * 4 + g];
+      }
+#pragma unroll
+      for (int f = 0; f < 4; ++f) {
+        int brow = wc * 64 + f * 16 + (lane & 15);
+        i32x8v bf = frag16(Bs, brow, w);
+        int sb = sbs[w * 512 + brow * 4 + g];
+#pragma unroll
+        for (int i = 0; i < 4; ++i)
+          acc[i][f] = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
+              af[i], bf, acc[i][f], 4 /*cbsz*/, 4 /*blgp*/, 0, sa[i], 0, sb);
+      }
+    }
+  }
+
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int f = 0; f < 4; ++f) {
+      int col = n0 + wc * 64 + f * 16 + (lane & 15);
+      if (col >= N) continue;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = m0 + wr * 64 + i * 16 + ((lane >> 4) << 2) + r;
+        if (row >= M) continue;
+        C[(int64_t)row * N + col] = acc[i][f][r];
+      }
+    }
+}
+
+void launch_gemm_mxfp4(const void* A, const void* B, const void* Sa,
+                       const void* Sb, void* C, int M, int N, int K,
+                       hipStream_t stream) {
+  if (K % 256 != 0)
+    throw std::runtime_error("gemm_mxfp4: K must be a multiple of 256");
+  int tiles_n = (int)cdiv(N, 128);
+  dim3 grid((unsigned)(cdiv(M, 128) * tiles_n));
+  hipLaunchKernelGGL(gemm_mxfp4_kernel, grid, dim3(256), 0, stream,
+                     (const uint8_t*)A, (const uint8_t*)B,
+                     (const uint8_t*)Sa, (const uint8_t*)Sb, (float*)C, M, N,
+                     K, tiles_n);
+}
+
+// fp4 variant of the layout probe: A/B rows are 64 packed bytes (2 elems
+// per byte); lane-group g is assumed to cover logical k [g*16,+16) and
+// [64+g*16,+16) like fp8 -> bytes [g*8,+8) and [32+g*8,+8), passed in the
+// low 4 dwords (CK: f4 operands occupy v4i32 zero-extended to v8i32).
+__global__ __launch_bounds__(64) void mx4_probe_kernel(
+    const uint8_t* __restrict__ A, const uint8_t* __restrict__ B,
+    const uint8_t* __restrict__ Sa, const uint8_t* __restrict__ Sb,
+    float* __restrict__ D) {
+  int lane = threadIdx.x;
+  int row = lane & 15, g = lane >> 4;
+  i32x8v af = {0, 0, 0, 0, 0, 0, 0, 0}, bf = {0, 0, 0, 0, 0, 0, 0, 0};
+  {
+    // contiguous mapping: lane-group g holds logical elems [g*32,+32)
+    // (= 16 packed bytes) and its scale byte covers exactly that block
+    i32x4 v = *(const i32x4*)(A + row * 64 + g * 16);
+    af[0] = v[0]; af[1] = v[1]; af[2] = v[2]; af[3] = v[3];
+    v = *(const i32x4*)(B + row * 64 + g * 16);
+    bf[0] = v[0]; bf[1] = v[1]; bf[2] = v[2]; bf[3] = v[3];
+  }
+  int sa = Sa[row * 4 + g];
+  int sb = Sb[row * 4 + g];
+  f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+  acc = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(af, bf, acc, 4, 4,
+                                                         0, sa, 0, sb);
+#pragma unroll
+  for (int r = 0; r < 4; ++r) D[((g << 2) + r) * 16 + row] = acc[r];
+}
+
+void launch_mx4_probe(const void* A, const void* B, const void* Sa,
+                      const void* Sb, void* D, hipStream_t stream) {
+  hipLaunchKernelGGL(mx4_probe_kernel, dim3(1), dim3(64), 0, stream,
+                     (const uint8_t*)A, (const uint8_t*)B, (const uint8_t*)Sa,
+                     (const uint8_t*)Sb, (float*)D);
+}
+
 void launch_mx_probe(const void* A, const void* B, const void* Sa,
                      const void* Sb, void* D, hipStream_t stream) {
   hipLaunchKernelGGL(mx_probe_kernel, dim3(1), dim3(64), 0, stream,

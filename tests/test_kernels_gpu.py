@@ -386,3 +386,29 @@ def test_gemm_mxfp8(C, M, N, K):
     full = torch.from_numpy(a32) @ torch.from_numpy(b32).t()
     corr = np.corrcoef(out.cpu().numpy().ravel(), full.numpy().ravel())[0, 1]
     assert corr > 0.99, corr
+
+
+@pytest.mark.parametrize("M,N,K", [(128, 128, 256), (256, 384, 512),
+                                   (1000, 768, 768)])
+def test_gemm_mxfp4(C, M, N, K):
+    """MXFP4 GEMM (scaled MFMA, cbsz=4) vs the dequantized-fp32 oracle."""
+    from trtlab_amd.engine.mx import dequantize_mxfp4, quantize_mxfp4
+
+    rng = np.random.RandomState(M + N + K)
+    a32 = (rng.randn(M, K) * np.exp(rng.randn(M, 1))).astype(np.float32)
+    b32 = (rng.randn(N, K) * np.exp(rng.randn(N, 1))).astype(np.float32)
+    aq, asc = quantize_mxfp4(a32)
+    bq, bsc = quantize_mxfp4(b32)
+    a = torch.from_numpy(aq).cuda()
+    b = torch.from_numpy(bq).cuda()
+    sa = torch.from_numpy(asc).cuda()
+    sb = torch.from_numpy(bsc).cuda()
+    out = torch.empty(M, N, dtype=torch.float32, device="cuda")
+    C.ops.gemm_mxfp4(a.data_ptr(), b.data_ptr(), sa.data_ptr(), sb.data_ptr(),
+                     out.data_ptr(), M, N, K)
+    ref = torch.from_numpy(dequantize_mxfp4(aq, asc)) @ \
+        torch.from_numpy(dequantize_mxfp4(bq, bsc)).t()
+    check(out, ref.cuda(), rtol=1e-2, atol=1e-2)
+    full = torch.from_numpy(a32) @ torch.from_numpy(b32).t()
+    corr = np.corrcoef(out.cpu().numpy().ravel(), full.numpy().ravel())[0, 1]
+    assert corr > 0.95, corr  # fp4: coarser grid than fp8

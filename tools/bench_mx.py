@@ -15,7 +15,7 @@ import numpy as np
 import torch
 
 import trtlab_amd
-from trtlab_amd.engine.mx import quantize_mxfp8
+from trtlab_amd.engine.mx import quantize_mxfp4, quantize_mxfp8
 
 C = trtlab_amd.native()
 
@@ -49,6 +49,16 @@ def main():
             a.data_ptr(), b.data_ptr(), sa.data_ptr(), sb.data_ptr(),
             out.data_ptr(), M, N, K, 0, False))
 
+        a4, as4 = quantize_mxfp4(a32)
+        b4, bs4 = quantize_mxfp4(b32)
+        a4t = torch.from_numpy(a4).cuda()
+        b4t = torch.from_numpy(b4).cuda()
+        sa4 = torch.from_numpy(as4).cuda()
+        sb4 = torch.from_numpy(bs4).cuda()
+        tmx4 = time_op(lambda: C.ops.gemm_mxfp4(
+            a4t.data_ptr(), b4t.data_ptr(), sa4.data_ptr(), sb4.data_ptr(),
+            out.data_ptr(), M, N, K, 0, False))
+
         ah = torch.from_numpy(a32).half().cuda()
         bh = torch.from_numpy(b32).half().cuda()
         oh = torch.empty(M, N, dtype=torch.half, device="cuda")
@@ -58,6 +68,7 @@ def main():
 
         flops = 2.0 * M * N * K
         print(f"{size}^3: mxfp8 {tmx*1e3:7.3f} ms = {flops/tmx/1e12:7.1f} TF"
+              f" | mxfp4 {tmx4*1e3:7.3f} ms = {flops/tmx4/1e12:7.1f} TF"
               f" | fp16 {tfp16*1e3:7.3f} ms = {flops/tfp16/1e12:7.1f} TF",
               flush=True)
 

@@ -44,3 +44,44 @@ def dequantize_mxfp8(codes: np.ndarray, scales: np.ndarray) -> np.ndarray:
     scale = torch.pow(2.0, torch.from_numpy(scales).float() - 127)
     out = vals.reshape(r, k // BLOCK, BLOCK) * scale[:, :, None]
     return out.reshape(r, k).numpy()
+
+
+# ---- MXFP4 (fp4 e2m1 elements, e8m0 block scales) ----
+_E2M1_VALS = [0.0, 0.5, 1.0, 1.5, 2.0, 3.0, 4.0, 6.0]
+_E2M1_EMAX = 2  # 6 = 1.5 * 2^2
+
+
+def quantize_mxfp4(x: np.ndarray) -> tuple[np.ndarray, np.ndarray]:
+    """[R, K] fp32 -> (packed u8 [R, K/2] (low nibble = even element),
+    scales u8 [R, K/32])."""
+    r, k = x.shape
+    assert k % BLOCK == 0
+    t = torch.from_numpy(np.ascontiguousarray(x, np.float32))
+    blocks = t.reshape(r, k // BLOCK, BLOCK)
+    amax = blocks.abs().amax(dim=2)
+    e = torch.where(amax > 0, torch.floor(torch.log2(amax)),
+                    torch.zeros_like(amax)) - _E2M1_EMAX
+    e = torch.clamp(e, -127, 127)
+    scale = torch.pow(2.0, e)
+    q = blocks / scale[:, :, None]
+    vals = torch.tensor(_E2M1_VALS)
+    mids = (vals[1:] + vals[:-1]) / 2  # round-to-nearest boundaries
+    mag = torch.bucketize(q.abs().reshape(-1), mids).to(torch.uint8)
+    code = torch.where(q.reshape(-1) < 0, mag | 8, mag).reshape(r, k)
+    packed = (code[:, 0::2] | (code[:, 1::2] << 4)).to(torch.uint8)
+    return packed.numpy().copy(), (e + 127).to(torch.uint8).numpy().copy()
+
+
+def dequantize_mxfp4(packed: np.ndarray, scales: np.ndarray) -> np.ndarray:
+    r, kh = packed.shape
+    k = kh * 2
+    p = torch.from_numpy(packed)
+    code = torch.empty(r, k, dtype=torch.uint8)
+    code[:, 0::2] = p & 15
+    code[:, 1::2] = p >> 4
+    vals = torch.tensor(_E2M1_VALS)
+    mag = vals[(code & 7).long()]
+    v = torch.where(code >= 8, -mag, mag)
+    scale = torch.pow(2.0, torch.from_numpy(scales).float() - 127)
+    return (v.reshape(r, k // BLOCK, BLOCK) * scale[:, :, None]).reshape(
+        r, k).numpy()
