@@ -34,6 +34,7 @@ OpDesc op_from_dict(const py::dict& d) {
   o.out_off = gl("out_off", -1);
   o.out2_off = gl("out2_off", -1);
   o.w_off = gl("w_off", -1);
+  o.w2_off = gl("w2_off", -1);
   o.scale_off = gl("scale_off", -1);
   o.bias_off = gl("bias_off", -1);
   o.M = gi("M", 0); o.N = gi("N", 0); o.K = gi("K", 0);
@@ -325,6 +326,15 @@ PYBIND11_MODULE(_C, m) {
     launch_mx_frag_dump((void*)A, (void*)out, K, 0);
     TRT_HIP_CHECK(hipStreamSynchronize(0));
   });
+  ops.def("quantize_mxfp4",
+          [](uintptr_t x, uintptr_t codes, uintptr_t scales, int64_t m,
+             int64_t k, uintptr_t stream, bool sync) {
+            launch_quantize_mxfp4((void*)x, (void*)codes, (void*)scales, m, k,
+                                  as_stream(stream));
+            if (sync) TRT_HIP_CHECK(hipStreamSynchronize(as_stream(stream)));
+          },
+          py::arg("x"), py::arg("codes"), py::arg("scales"), py::arg("m"),
+          py::arg("k"), py::arg("stream") = 0, py::arg("sync") = true);
   ops.def("gemm_mxfp4",
           [](uintptr_t A, uintptr_t B, uintptr_t Sa, uintptr_t Sb,
              uintptr_t C, int M, int N, int K, uintptr_t stream, bool sync) {

@@ -219,16 +219,82 @@ __global__ __launch_bounds__(64) void mx_probe_kernel(
   for (int r = 0; r < 4; ++r) D[((g << 2) + r) * 16 + row] = acc[r];
 }
 
+// Row-wise dynamic MXFP4 quantization of fp16 activations:
+// x [M, K] fp16 -> codes [M, K/2] (low nibble = even elem) + e8m0 scales
+// [M, K/32]. One thread per 32-element block (grid-strided): per-block
+// amax -> shared exponent 2^(floor(log2(amax)) - 2) -> round-to-nearest
+// onto the e2m1 grid {0, .5, 1, 1.5, 2, 3, 4, 6}.
+__global__ __launch_bounds__(256) void quantize_mxfp4_kernel(
+    const _Float16* __restrict__ x, uint8_t* __restrict__ codes,
+    uint8_t* __restrict__ scales, int64_t nblocks) {
+  int64_t b = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (b >= nblocks) return;
+  const _Float16* src = x + b * 32;
+  float v[32];
+  float amax = 0.f;
+#pragma unroll
+  for (int c = 0; c < 4; ++c) {
+    half8v h = *(const half8v*)(src + c * 8);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float f = (float)((const _Float16*)&h)[j];
+      v[c * 8 + j] = f;
+      amax = fmaxf(amax, fabsf(f));
+    }
+  }
+  int e = (amax > 0.f) ? (int)floorf(log2f(amax)) - 2 : 0;
+  e = e < -127 ? -127 : (e > 127 ? 127 : e);
+  scales[b] = (uint8_t)(e + 127);
+  float inv = exp2f((float)-e);
+  uint8_t out[16];
+#pragma unroll
+  for (int j = 0; j < 16; ++j) {
+    uint8_t byte = 0;
+#pragma unroll
+    for (int h = 0; h < 2; ++h) {
+      float q = v[2 * j + h] * inv;
+      float a = fabsf(q);
+      // round-to-nearest onto {0,.5,1,1.5,2,3,4,6}
+      int m;
+      if (a < 0.25f) m = 0;
+      else if (a < 0.75f) m = 1;
+      else if (a < 1.25f) m = 2;
+      else if (a < 1.75f) m = 3;
+      else if (a < 2.5f) m = 4;
+      else if (a < 3.5f) m = 5;
+      else if (a < 5.0f) m = 6;
+      else m = 7;
+      uint8_t code = (uint8_t)(q < 0.f ? (m | 8) : m);
+      byte |= (uint8_t)(code << (4 * h));
+    }
+    out[j] = byte;
+  }
+  *(f32x4*)(codes + b * 16) = *(const f32x4*)out;
+}
+
+void launch_quantize_mxfp4(const void* x, void* codes, void* scales,
+                           int64_t m, int64_t k, hipStream_t stream) {
+  if (k % 32 != 0)
+    throw std::runtime_error("quantize_mxfp4: K must be a multiple of 32");
+  int64_t nblocks = m * (k / 32);
+  int64_t blocks = (nblocks + 255) / 256;
+  hipLaunchKernelGGL(quantize_mxfp4_kernel, dim3((unsigned)blocks), dim3(256),
+                     0, stream, (const _Float16*)x, (uint8_t*)codes,
+                     (uint8_t*)scales, nblocks);
+}
+
 // ---- MXFP4 (fp4 e2m1, e8m0 scales per 32 elems) ----
 // Same scaled MFMA with cbsz/blgp = 4. fp4 operand layout is SIMPLER than
 // fp8 (tools/probe_mx4.py): lane-group g's 16 packed bytes (low 4 dwords,
 // upper 4 zero) cover logical k [g*32, g*32+32) contiguously, and its
 // scale byte covers exactly that block. One 128-byte LDS row holds TWO
 // 128-k instruction windows (256 logical k); K must be % 256.
+template <typename OT, Epi E>
 __global__ __launch_bounds__(256) void gemm_mxfp4_kernel(
     const uint8_t* __restrict__ A, const uint8_t* __restrict__ B,
     const uint8_t* __restrict__ Sa, const uint8_t* __restrict__ Sb,
-    float* __restrict__ C, int M, int N, int K, int tiles_n) {
+    OT* __restrict__ C, const float* __restrict__ scale,
+    const float* __restrict__ bias, int M, int N, int K, int tiles_n) {
   constexpr int kABytes = 128 * 128;
   constexpr int kSlot = 2 * kABytes + 2048;  // + 8 scale bytes/row (A, B)
   __shared__ __attribute__((aligned(16))) char smem[3 * kSlot];
@@ -338,26 +404,45 @@ __global__ __launch_bounds__(256) void gemm_mxfp4_kernel(
     for (int f = 0; f < 4; ++f) {
       int col = n0 + wc * 64 + f * 16 + (lane & 15);
       if (col >= N) continue;
+      float sc = 1.0f, bi = 0.0f;
+      if constexpr (E == Epi::kScaleBias || E == Epi::kScaleBiasRelu ||
+                    E == Epi::kScaleBiasAddRelu || E == Epi::kScaleBiasGelu)
+        sc = scale[col];
+      if constexpr (E != Epi::kNone) bi = bias[col];
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         int row = m0 + wr * 64 + i * 16 + ((lane >> 4) << 2) + r;
         if (row >= M) continue;
-        C[(int64_t)row * N + col] = acc[i][f][r];
+        C[(int64_t)row * N + col] =
+            store_cast<OT>(apply_epi<E>(acc[i][f][r], sc, bi, 0.0f));
       }
     }
 }
 
+// out_dtype: 0 = fp16 (engine path, with epilogue), 2 = fp32 raw C.
 void launch_gemm_mxfp4(const void* A, const void* B, const void* Sa,
                        const void* Sb, void* C, int M, int N, int K,
-                       hipStream_t stream) {
+                       hipStream_t stream, int out_dtype, int epi,
+                       const float* scale, const float* bias) {
   if (K % 256 != 0)
     throw std::runtime_error("gemm_mxfp4: K must be a multiple of 256");
   int tiles_n = (int)cdiv(N, 128);
   dim3 grid((unsigned)(cdiv(M, 128) * tiles_n));
-  hipLaunchKernelGGL(gemm_mxfp4_kernel, grid, dim3(256), 0, stream,
-                     (const uint8_t*)A, (const uint8_t*)B,
-                     (const uint8_t*)Sa, (const uint8_t*)Sb, (float*)C, M, N,
-                     K, tiles_n);
+  if (out_dtype == 2) {
+    hipLaunchKernelGGL((gemm_mxfp4_kernel<float, Epi::kNone>), grid,
+                       dim3(256), 0, stream, (const uint8_t*)A,
+                       (const uint8_t*)B, (const uint8_t*)Sa,
+                       (const uint8_t*)Sb, (float*)C, nullptr, nullptr, M, N,
+                       K, tiles_n);
+    return;
+  }
+  epi_dispatch(epi, [&](auto e) {
+    constexpr Epi EE = decltype(e)::value;
+    hipLaunchKernelGGL((gemm_mxfp4_kernel<_Float16, EE>), grid, dim3(256), 0,
+                       stream, (const uint8_t*)A, (const uint8_t*)B,
+                       (const uint8_t*)Sa, (const uint8_t*)Sb, (_Float16*)C,
+                       scale, bias, M, N, K, tiles_n);
+  });
 }
 
 // fp4 variant of the layout probe: A/B rows are 64 packed bytes (2 elems

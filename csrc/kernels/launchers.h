@@ -69,9 +69,13 @@ void launch_seqlens(const void* ids, void* lens, int B, int S, int pad_id,
 // the CDNA4 scaled MFMA 16x16x128. B transposed [N][K]; scales u8
 // [M][K/32] / [N][K/32]; C fp32.
 void launch_mx_frag_dump(const void* A, void* out, int K, hipStream_t s);
+void launch_quantize_mxfp4(const void* x, void* codes, void* scales,
+                           int64_t m, int64_t k, hipStream_t stream);
 void launch_gemm_mxfp4(const void* A, const void* B, const void* Sa,
                        const void* Sb, void* C, int M, int N, int K,
-                       hipStream_t stream);
+                       hipStream_t stream, int out_dtype = 2,
+                       int epi = 0, const float* scale = nullptr,
+                       const float* bias = nullptr);
 void launch_mx4_probe(const void* A, const void* B, const void* Sa,
                       const void* Sb, void* D, hipStream_t stream);
 void launch_mx_probe(const void* A, const void* B, const void* Sa,

@@ -58,6 +58,8 @@ enum OpKind : int {
   kEmbedding = 12,  // out = tok[ids] + pos[m%S] (+ seg)
   kAvgPool = 13,
   kSeqLens = 14,  // ids -> per-sequence valid length (varlen attention)
+  kQuantMx4 = 15,  // fp16 rows -> MXFP4 codes + e8m0 block scales
+  kGemmMx4 = 16,   // MXFP4 x MXFP4 scaled-MFMA GEMM, fp16 out + epilogue
 };
 
 struct OpDesc {
@@ -68,6 +70,7 @@ struct OpDesc {
   int64_t in_off = -1, in2_off = -1, out_off = -1, out2_off = -1;
   // weight-blob offsets in bytes (-1 = absent)
   int64_t w_off = -1, scale_off = -1, bias_off = -1;
+  int64_t w2_off = -1;  // second weight slab (MX weight scales)
   // gemm
   int M = 0, N = 0, K = 0;
   // conv / pool

@@ -73,3 +73,32 @@ def test_varlen_plan_structure_and_noop_equivalence():
     # sequence 1 valid rows must change (padded keys no longer attended)
     assert not np.allclose(out_masked[128:128 + 57],
                            out_unmasked[128:128 + 57], atol=1e-3)
+
+
+def test_mxfp4_plan_structure():
+    """DT_MX4 lowers every BERT GEMM to quantize + MXFP4 scaled-MFMA GEMM;
+    the fp32 reference emulation stays close to the fp16 plan."""
+    import numpy as np
+
+    from trtlab_amd.engine.planner import (DT_MX4, K_GEMM, K_GEMM_MX4,
+                                           K_QUANT_MX4, Planner)
+    from trtlab_amd.engine.reference import run_reference
+    from trtlab_amd.models import build_bert
+
+    g = build_bert(batch=2, seq=128, layers=2, seed=0)
+    plan = Planner(dtype=DT_MX4).compile(g)
+    kinds = [d["kind"] for d in plan.ops]
+    assert kinds.count(K_GEMM_MX4) == 8  # 4 gemms x 2 layers
+    assert kinds.count(K_QUANT_MX4) == 8
+    assert kinds.count(K_GEMM) == 0
+    mx = [d for d in plan.ops if d["kind"] == K_GEMM_MX4]
+    assert all(d["w2_off"] >= 0 and "in2_off" in d for d in mx)
+
+    x = np.random.RandomState(3).randn(*plan.input_shape).astype(
+        np.float32) * 0.5
+    out = run_reference(plan, x)
+    assert np.isfinite(out).all()
+    g2 = build_bert(batch=2, seq=128, layers=2, seed=0)
+    ref16 = run_reference(Planner().compile(g2), x)
+    corr = np.corrcoef(out.ravel(), ref16.ravel())[0, 1]
+    assert corr > 0.95, corr  # fp4 weights+activations: coarse but sane
