@@ -37,7 +37,7 @@ def main():
                     choices=["resnet50", "resnet101", "resnet152", "bert",
                              "bert-large"])
     ap.add_argument("--dtype", default="fp16",
-                    choices=["fp16", "bf16", "int8", "fp8", "mxfp4"])
+                    choices=["fp16", "bf16", "int8", "fp8", "mxfp4", "mxfp8"])
     ap.add_argument("--no-autotune", action="store_true",
                     help="skip builder-time kernel tactic selection")
     args = ap.parse_args()
@@ -78,10 +78,10 @@ def main():
         cfg_extra = {"image": 224}
     from trtlab_amd.engine.planner import DT_BF16, DT_F8
 
-    from trtlab_amd.engine.planner import DT_MX4
+    from trtlab_amd.engine.planner import DT_MX4, DT_MX8
 
     dtype = {"fp16": DT_F16, "bf16": DT_BF16, "int8": DT_I8,
-             "fp8": DT_F8, "mxfp4": DT_MX4}[args.dtype]
+             "fp8": DT_F8, "mxfp4": DT_MX4, "mxfp8": DT_MX8}[args.dtype]
     plan = Planner(dtype=dtype).compile(g)
 
     eng = NativeEngine(plan, device=local_rank,

@@ -51,10 +51,12 @@ __device__ __forceinline__ i32x8v read_frag32(const char* lds, int row,
 // computes the [wr*64, +64) x [wc*64, +64) quadrant (4x4 fragment pairs,
 // 16 scaled MFMAs per K-tile from 8 32-byte fragment reads). Double
 // buffered: 2 x (128+128) rows x 128 B = 64 KiB LDS -> 2 blocks/CU.
+template <typename OT, Epi E>
 __global__ __launch_bounds__(256) void gemm_mxfp8_kernel(
     const uint8_t* __restrict__ A, const uint8_t* __restrict__ B,
     const uint8_t* __restrict__ Sa, const uint8_t* __restrict__ Sb,
-    float* __restrict__ C, int M, int N, int K, int tiles_n) {
+    OT* __restrict__ C, const float* __restrict__ scale,
+    const float* __restrict__ bias, int M, int N, int K, int tiles_n) {
   // 3 slots x (A tile + B tile + scale slabs): counted-vmcnt pipeline
   // keeps 2 tiles' staging in flight behind the MFMAs (9 glds per thread
   // per tile: 8 data + 1 scale u32).
@@ -152,11 +154,17 @@ __global__ __launch_bounds__(256) void gemm_mxfp8_kernel(
     for (int f = 0; f < 4; ++f) {
       int col = n0 + wc * 64 + f * 16 + (lane & 15);
       if (col >= N) continue;
+      float sc = 1.0f, bi = 0.0f;
+      if constexpr (E == Epi::kScaleBias || E == Epi::kScaleBiasRelu ||
+                    E == Epi::kScaleBiasAddRelu || E == Epi::kScaleBiasGelu)
+        sc = scale[col];
+      if constexpr (E != Epi::kNone) bi = bias[col];
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         int row = m0 + wr * 64 + i * 16 + ((lane >> 4) << 2) + r;
         if (row >= M) continue;
-        C[(int64_t)row * N + col] = acc[i][f][r];
+        C[(int64_t)row * N + col] =
+            store_cast<OT>(apply_epi<E>(acc[i][f][r], sc, bi, 0.0f));
       }
     }
 }
@@ -279,6 +287,51 @@ void launch_quantize_mxfp4(const void* x, void* codes, void* scales,
   int64_t nblocks = m * (k / 32);
   int64_t blocks = (nblocks + 255) / 256;
   hipLaunchKernelGGL(quantize_mxfp4_kernel, dim3((unsigned)blocks), dim3(256),
+                     0, stream, (const _Float16*)x, (uint8_t*)codes,
+                     (uint8_t*)scales, nblocks);
+}
+
+// Row-wise dynamic MXFP8 quantization: fp16 [M, K] -> e4m3 codes [M, K]
+// + e8m0 scales [M, K/32]. One thread per 32-element block.
+__global__ __launch_bounds__(256) void quantize_mxfp8_kernel(
+    const _Float16* __restrict__ x, uint8_t* __restrict__ codes,
+    uint8_t* __restrict__ scales, int64_t nblocks) {
+  int64_t b = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (b >= nblocks) return;
+  const _Float16* src = x + b * 32;
+  float v[32];
+  float amax = 0.f;
+#pragma unroll
+  for (int c = 0; c < 4; ++c) {
+    half8v h = *(const half8v*)(src + c * 8);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float f = (float)((const _Float16*)&h)[j];
+      v[c * 8 + j] = f;
+      amax = fmaxf(amax, fabsf(f));
+    }
+  }
+  int e = (amax > 0.f) ? (int)floorf(log2f(amax)) - 8 : 0;  // e4m3 emax 8
+  e = e < -127 ? -127 : (e > 127 ? 127 : e);
+  scales[b] = (uint8_t)(e + 127);
+  float inv = exp2f((float)-e);
+  uint8_t out[32];
+#pragma unroll
+  for (int j = 0; j < 32; ++j) {
+    __hip_fp8_e4m3 q = store_cast<__hip_fp8_e4m3>(v[j] * inv);
+    out[j] = *(const uint8_t*)&q;
+  }
+  *(f32x4*)(codes + b * 32) = *(const f32x4*)out;
+  *(f32x4*)(codes + b * 32 + 16) = *(const f32x4*)(out + 16);
+}
+
+void launch_quantize_mxfp8(const void* x, void* codes, void* scales,
+                           int64_t m, int64_t k, hipStream_t stream) {
+  if (k % 32 != 0)
+    throw std::runtime_error("quantize_mxfp8: K must be a multiple of 32");
+  int64_t nblocks = m * (k / 32);
+  int64_t blocks = (nblocks + 255) / 256;
+  hipLaunchKernelGGL(quantize_mxfp8_kernel, dim3((unsigned)blocks), dim3(256),
                      0, stream, (const _Float16*)x, (uint8_t*)codes,
                      (uint8_t*)scales, nblocks);
 }
@@ -487,17 +540,30 @@ void launch_mx_probe(const void* A, const void* B, const void* Sa,
                      (const uint8_t*)Sb, (float*)D);
 }
 
+// out_dtype: 0 = fp16 (engine path, with epilogue), 2 = fp32 raw C.
 void launch_gemm_mxfp8(const void* A, const void* B, const void* Sa,
                        const void* Sb, void* C, int M, int N, int K,
-                       hipStream_t stream) {
+                       hipStream_t stream, int out_dtype, int epi,
+                       const float* scale, const float* bias) {
   if (K % 128 != 0)
     throw std::runtime_error("gemm_mxfp8: K must be a multiple of 128");
   int tiles_n = (int)cdiv(N, 128);
   dim3 grid((unsigned)(cdiv(M, 128) * tiles_n));
-  hipLaunchKernelGGL(gemm_mxfp8_kernel, grid, dim3(256), 0, stream,
-                     (const uint8_t*)A, (const uint8_t*)B,
-                     (const uint8_t*)Sa, (const uint8_t*)Sb, (float*)C, M, N,
-                     K, tiles_n);
+  if (out_dtype == 2) {
+    hipLaunchKernelGGL((gemm_mxfp8_kernel<float, Epi::kNone>), grid,
+                       dim3(256), 0, stream, (const uint8_t*)A,
+                       (const uint8_t*)B, (const uint8_t*)Sa,
+                       (const uint8_t*)Sb, (float*)C, nullptr, nullptr, M, N,
+                       K, tiles_n);
+    return;
+  }
+  epi_dispatch(epi, [&](auto e) {
+    constexpr Epi EE = decltype(e)::value;
+    hipLaunchKernelGGL((gemm_mxfp8_kernel<_Float16, EE>), grid, dim3(256), 0,
+                       stream, (const uint8_t*)A, (const uint8_t*)B,
+                       (const uint8_t*)Sa, (const uint8_t*)Sb, (_Float16*)C,
+                       scale, bias, M, N, K, tiles_n);
+  });
 }
 
 }  // namespace trtlab

@@ -82,7 +82,11 @@ void launch_mx_probe(const void* A, const void* B, const void* Sa,
                      const void* Sb, void* D, hipStream_t stream);
 void launch_gemm_mxfp8(const void* A, const void* B, const void* Sa,
                        const void* Sb, void* C, int M, int N, int K,
-                       hipStream_t stream);
+                       hipStream_t stream, int out_dtype = 2, int epi = 0,
+                       const float* scale = nullptr,
+                       const float* bias = nullptr);
+void launch_quantize_mxfp8(const void* x, void* codes, void* scales,
+                           int64_t m, int64_t k, hipStream_t stream);
 
 void launch_embedding(int dtype, const void* ids, const void* tok,
                       const void* pos, const void* seg, const void* segids,

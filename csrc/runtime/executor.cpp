@@ -193,6 +193,16 @@ void ExecutionContext::enqueue_all(hipStream_t s) {
                           0 /*fp16 out*/, op.epi, Fp(op.scale_off),
                           Fp(op.bias_off));
         break;
+      case kQuantMx8:
+        launch_quantize_mxfp8(A(op.in_off), A(op.out_off), A(op.out2_off),
+                              op.M, op.K, s);
+        break;
+      case kGemmMx8:
+        launch_gemm_mxfp8(A(op.in_off), Wp(op.w_off), A(op.in2_off),
+                          Wp(op.w2_off), A(op.out_off), op.M, op.N, op.K, s,
+                          0 /*fp16 out*/, op.epi, Fp(op.scale_off),
+                          Fp(op.bias_off));
+        break;
       case kQuantize:
         // op.epi carries the target format (0 = int8, 1 = fp8 e4m3)
         launch_quantize(A(op.in_off), A(op.out_off), op.n_elems, op.q_scale,

@@ -60,6 +60,8 @@ enum OpKind : int {
   kSeqLens = 14,  // ids -> per-sequence valid length (varlen attention)
   kQuantMx4 = 15,  // fp16 rows -> MXFP4 codes + e8m0 block scales
   kGemmMx4 = 16,   // MXFP4 x MXFP4 scaled-MFMA GEMM, fp16 out + epilogue
+  kQuantMx8 = 17,  // fp16 rows -> MXFP8 (e4m3) codes + e8m0 block scales
+  kGemmMx8 = 18,   // MXFP8 x MXFP8 scaled-MFMA GEMM, fp16 out + epilogue
 };
 
 struct OpDesc {

@@ -102,3 +102,26 @@ def test_mxfp4_plan_structure():
     ref16 = run_reference(Planner().compile(g2), x)
     corr = np.corrcoef(out.ravel(), ref16.ravel())[0, 1]
     assert corr > 0.95, corr  # fp4 weights+activations: coarse but sane
+
+
+def test_mxfp8_plan_structure():
+    """DT_MX8: same lowering with e4m3 elements (1 byte/elem codes)."""
+    import numpy as np
+
+    from trtlab_amd.engine.planner import (DT_MX8, K_GEMM, K_GEMM_MX8,
+                                           K_QUANT_MX8, Planner)
+    from trtlab_amd.engine.reference import run_reference
+    from trtlab_amd.models import build_bert
+
+    g = build_bert(batch=2, seq=128, layers=1, seed=0)
+    plan = Planner(dtype=DT_MX8).compile(g)
+    kinds = [d["kind"] for d in plan.ops]
+    assert kinds.count(K_GEMM_MX8) == 4 and kinds.count(K_QUANT_MX8) == 4
+    assert kinds.count(K_GEMM) == 0
+    x = np.random.RandomState(3).randn(*plan.input_shape).astype(
+        np.float32) * 0.5
+    out = run_reference(plan, x)
+    g2 = build_bert(batch=2, seq=128, layers=1, seed=0)
+    ref16 = run_reference(Planner().compile(g2), x)
+    corr = np.corrcoef(out.ravel(), ref16.ravel())[0, 1]
+    assert corr > 0.99, corr  # fp8 elements: much tighter than fp4

@@ -373,3 +373,26 @@ def test_engine_bert_mxfp4():
     corr_emu = np.corrcoef(refmx.ravel(), ref16.ravel())[0, 1]
     assert corr_gpu > corr_emu - 0.03, (corr_gpu, corr_emu)
     assert corr_gpu > 0.9, corr_gpu
+
+
+def test_engine_bert_mxfp8():
+    """BERT with MXFP8 GEMMs (device-quantized e4m3 + e8m0 block scales)."""
+    from trtlab_amd.engine.planner import DT_MX8, Planner
+    from trtlab_amd.engine.reference import run_reference
+    from trtlab_amd.engine.runtime import NativeEngine
+    from trtlab_amd.models import build_bert
+
+    g = build_bert(batch=2, seq=128, layers=2, seed=0)
+    plan = Planner(dtype=DT_MX8).compile(g)
+    ctx = NativeEngine(plan).create_context(capture=True)
+    x = np.random.RandomState(33).randn(*plan.input_shape).astype(
+        np.float32) * 0.5
+    out = ctx.infer(x).astype(np.float32)
+    assert np.isfinite(out).all()
+    ref16 = run_reference(Planner().compile(
+        build_bert(batch=2, seq=128, layers=2, seed=0)), x)
+    refmx = run_reference(plan, x)
+    corr_gpu = np.corrcoef(out.ravel(), ref16.ravel())[0, 1]
+    corr_emu = np.corrcoef(refmx.ravel(), ref16.ravel())[0, 1]
+    assert corr_gpu > corr_emu - 0.03, (corr_gpu, corr_emu)
+    assert corr_gpu > 0.95, corr_gpu

@@ -42,6 +42,7 @@ DT_BF16 = 1
 DT_I8 = 2
 DT_F8 = 3  # OCP fp8 e4m3
 DT_MX4 = 4  # OCP MXFP4 (e2m1 + e8m0 block scales) for GEMMs; rest fp16
+DT_MX8 = 5  # OCP MXFP8 (e4m3 + e8m0 block scales) for GEMMs; rest fp16
 
 K_QUANTIZE = 10
 K_DEQUANT = 11
@@ -50,6 +51,8 @@ K_AVGPOOL = 13
 K_SEQLENS = 14  # token ids -> per-sequence valid length (varlen attention)
 K_QUANT_MX4 = 15  # fp16 rows -> MXFP4 codes + e8m0 block scales
 K_GEMM_MX4 = 16  # MXFP4 x MXFP4 scaled-MFMA GEMM (fp16 out + epilogue)
+K_QUANT_MX8 = 17  # fp16 rows -> MXFP8 (e4m3) codes + e8m0 block scales
+K_GEMM_MX8 = 18  # MXFP8 x MXFP8 scaled-MFMA GEMM (fp16 out + epilogue)
 
 
 def _bf16_bits(arr: np.ndarray) -> np.ndarray:
@@ -244,37 +247,42 @@ class Planner:
         return op
 
     # ---------------------------------------------------------- MX lowering
-    def _lower_mx4(self, exec_ops, shapes, itemsize):
-        """Lower eligible GEMMs to MXFP4 (OCP Microscaling fp4-e2m1 +
-        e8m0 per-32-block scales) on the CDNA4 scaled MFMA: weights are
-        block-quantized at build time; a K_QUANT_MX4 op quantizes the
-        activation rows on device before each lowered GEMM. Epilogues
+    def _lower_mx(self, exec_ops, shapes, itemsize, fp4: bool):
+        """Lower eligible GEMMs to OCP Microscaling (fp4-e2m1 or fp8-e4m3
+        elements + e8m0 per-32-block scales) on the CDNA4 scaled MFMA:
+        weights are block-quantized at build time; a quantize op converts
+        the activation rows on device before each lowered GEMM. Epilogues
         (bias/GeLU) stay fused. Everything else stays fp16."""
-        from trtlab_amd.engine.mx import quantize_mxfp4
+        from trtlab_amd.engine.mx import quantize_mxfp4, quantize_mxfp8
 
+        kmult = 256 if fp4 else 128  # one LDS tile of logical K
+        quantize = quantize_mxfp4 if fp4 else quantize_mxfp8
+        kq = K_QUANT_MX4 if fp4 else K_QUANT_MX8
+        kg = K_GEMM_MX4 if fp4 else K_GEMM_MX8
         new_ops = []
         for op in exec_ops:
             m, k = (None, None)
             if op.kind == K_GEMM:
                 m, k = shapes[op.inputs[0]]
-            if op.kind != K_GEMM or k % 256 != 0 or                     op.params["epi"] not in (EPI_NONE, EPI_BIAS,
+            if op.kind != K_GEMM or k % kmult != 0 or \
+                    op.params["epi"] not in (EPI_NONE, EPI_BIAS,
                                              EPI_BIAS_RELU, EPI_BIAS_GELU):
                 new_ops.append(op)
                 continue
             nout = op.params["weight_shape"][0]
-            wq, wsc = quantize_mxfp4(op.w.astype(np.float32))
+            wq, wsc = quantize(op.w.astype(np.float32))
             codes_t = op.name + "_mxq"
             scales_t = op.name + "_mxs"
-            shapes[codes_t] = (m, k // 2)
+            shapes[codes_t] = (m, k // 2 if fp4 else k)
             shapes[scales_t] = (m, k // 32)
             itemsize[codes_t] = 1
             itemsize[scales_t] = 1
-            q = ExecOp(K_QUANT_MX4, op.name + "_quant", [op.inputs[0]],
+            q = ExecOp(kq, op.name + "_quant", [op.inputs[0]],
                        codes_t, dict(M=m, K=k, q_out=scales_t))
-            g = ExecOp(K_GEMM_MX4, op.name, [codes_t, scales_t], op.output,
+            g = ExecOp(kg, op.name, [codes_t, scales_t], op.output,
                        dict(epi=op.params["epi"], M=m, N=nout, K=k,
                             mx=True))
-            g.w = wq                     # fp4 codes [N, K/2]
+            g.w = wq                     # packed element codes
             g.bias = op.bias             # fp32 epilogue bias
             g.params["mx_wscales"] = wsc  # e8m0 [N, K/32] -> w2_off
             new_ops.append(q)
@@ -334,8 +342,9 @@ class Planner:
                      pad_id=att_varlen[0].params.get("pad_id", 0))))
             for op in att_varlen:
                 op.inputs.append(lens_name)
-        if self.dtype == DT_MX4:
-            self._lower_mx4(exec_ops, shapes, itemsize)
+        if self.dtype in (DT_MX4, DT_MX8):
+            self._lower_mx(exec_ops, shapes, itemsize,
+                           fp4=self.dtype == DT_MX4)
         if self.dtype in (DT_I8, DT_F8):
             from trtlab_amd.engine.quantize import lower_int8
 
@@ -439,7 +448,7 @@ class Planner:
             w_off, s_off, b_off = w_offs[op.name]
             op_dtype = op.params.get(
                 "dtype",
-                DT_F16 if self.dtype in (DT_I8, DT_F8, DT_MX4)
+                DT_F16 if self.dtype in (DT_I8, DT_F8, DT_MX4, DT_MX8)
                 else self.dtype)
             d: Dict[str, Any] = dict(dtype=op_dtype, w_off=w_off,
                                      scale_off=s_off, bias_off=b_off,
@@ -501,12 +510,12 @@ class Planner:
             elif op.kind == K_SEQLENS:
                 d.update(kind=K_SEQLENS, B=op.params["B"], S=op.params["S"],
                          epi=op.params["pad_id"])
-            elif op.kind == K_QUANT_MX4:
-                d.update(kind=K_QUANT_MX4, M=op.params["M"],
+            elif op.kind in (K_QUANT_MX4, K_QUANT_MX8):
+                d.update(kind=op.kind, M=op.params["M"],
                          K=op.params["K"],
                          out2_off=offsets[op.params["q_out"]])
-            elif op.kind == K_GEMM_MX4:
-                d.update(kind=K_GEMM_MX4, epi=op.params["epi"],
+            elif op.kind in (K_GEMM_MX4, K_GEMM_MX8):
+                d.update(kind=op.kind, epi=op.params["epi"],
                          M=op.params["M"], N=op.params["N"],
                          K=op.params["K"], w2_off=w2_offs[op.name])
             elif op.kind in (K_QUANTIZE, K_DEQUANT):

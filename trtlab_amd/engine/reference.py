@@ -17,7 +17,8 @@ import torch.nn.functional as F
 from trtlab_amd.engine.planner import (
     EnginePlan, K_ADD_LAYERNORM, K_ATTENTION, K_AVGPOOL, K_CHANNEL_PAD, K_CONV,
     K_DEQUANT, K_ELEMENTWISE, K_GAVGPOOL, K_GEMM, K_LAYERNORM, K_MAXPOOL,
-    K_EMBEDDING, K_GEMM_MX4, K_QUANT_MX4, K_QUANTIZE, K_SEQLENS, K_SOFTMAX,
+    K_EMBEDDING, K_GEMM_MX4, K_GEMM_MX8, K_QUANT_MX4, K_QUANT_MX8,
+    K_QUANTIZE, K_SEQLENS, K_SOFTMAX,
     EPI_BIAS, EPI_BIAS_GELU, EPI_BIAS_RELU, EPI_NONE, EPI_SCALE_BIAS,
     EPI_SCALE_BIAS_ADD_RELU, EPI_SCALE_BIAS_GELU, EPI_SCALE_BIAS_RELU)
 
@@ -153,18 +154,21 @@ def run_reference(plan: EnginePlan, input_nhwc: np.ndarray,
                 t[op.output] = x + t[op.inputs[1]]
             elif code == 3:
                 t[op.output] = F.relu(x + t[op.inputs[1]])
-        elif op.kind == K_QUANT_MX4:
-            from trtlab_amd.engine.mx import quantize_mxfp4
+        elif op.kind in (K_QUANT_MX4, K_QUANT_MX8):
+            from trtlab_amd.engine.mx import quantize_mxfp4, quantize_mxfp8
 
-            codes, scales = quantize_mxfp4(x.numpy().astype(np.float32))
+            qf = quantize_mxfp4 if op.kind == K_QUANT_MX4 else quantize_mxfp8
+            codes, scales = qf(x.numpy().astype(np.float32))
             t[op.output] = torch.from_numpy(codes)
             t[op.params["q_out"]] = torch.from_numpy(scales)
-        elif op.kind == K_GEMM_MX4:
-            from trtlab_amd.engine.mx import dequantize_mxfp4
+        elif op.kind in (K_GEMM_MX4, K_GEMM_MX8):
+            from trtlab_amd.engine.mx import (dequantize_mxfp4,
+                                              dequantize_mxfp8)
 
-            a = dequantize_mxfp4(x.numpy(),
-                                 t[op.inputs[1]].numpy())
-            w = dequantize_mxfp4(op.w, op.params["mx_wscales"])
+            df = (dequantize_mxfp4 if op.kind == K_GEMM_MX4
+                  else dequantize_mxfp8)
+            a = df(x.numpy(), t[op.inputs[1]].numpy())
+            w = df(op.w, op.params["mx_wscales"])
             acc = torch.from_numpy(a) @ torch.from_numpy(w).T
             bias = (torch.from_numpy(op.bias) if op.bias is not None
                     else None)
