@@ -46,6 +46,7 @@ OpDesc op_from_dict(const py::dict& d) {
   o.n_elems = gl("n_elems", 0);
   o.B = gi("B", 0); o.S = gi("S", 0); o.NH = gi("NH", 0); o.HD = gi("HD", 0);
   o.att_scale = gf("att_scale", 1.0f);
+  o.causal = gi("causal", 0);
   o.res_scale = gf("res_scale", 1.0f);
   o.q_scale = gf("q_scale", 1.0f);
   o.tile = gi("tile", 0);
@@ -364,14 +365,17 @@ PYBIND11_MODULE(_C, m) {
           py::arg("stream") = 0, py::arg("sync") = true);
   ops.def("attention",
           [](int dtype, uintptr_t qkv, uintptr_t out, int B, int S, int H,
-             int D, float scale, uintptr_t stream, bool sync) {
+             int D, float scale, uintptr_t stream, bool sync, int causal,
+             uintptr_t seqlens) {
             launch_attention(dtype, (void*)qkv, (void*)out, B, S, H, D, scale,
-                             as_stream(stream));
+                             as_stream(stream), -1, 1.0f, (void*)seqlens,
+                             causal);
             if (sync) TRT_HIP_CHECK(hipStreamSynchronize(as_stream(stream)));
           },
           py::arg("dtype"), py::arg("qkv"), py::arg("out"), py::arg("B"),
           py::arg("S"), py::arg("H"), py::arg("D"), py::arg("scale"),
-          py::arg("stream") = 0, py::arg("sync") = true);
+          py::arg("stream") = 0, py::arg("sync") = true,
+          py::arg("causal") = 0, py::arg("seqlens") = 0);
 
   // ------------------------------------------------------------- engine
   py::class_<Engine, std::shared_ptr<Engine>>(m, "Engine")

@@ -1,35 +1,36 @@
-// trtlab_amd — fused multi-head self-attention for the BERT encoder path
-// (gfx950). SURVEY.md §2.8 item 3: batched GEMM + softmax fused in one
-// kernel for seq=128.
+// trtlab_amd — fused multi-head self-attention (gfx950). SURVEY.md §2.8
+// item 3: batched GEMM + softmax fused in one kernel.
 //
 // Input: qkv [B*S, 3*H*D] (the fused QKV projection output, column blocks
 // q|k|v, each H*D with head-major (h,d) minor d). Output: [B*S, H*D].
 //
-// One workgroup per (b, h, 64-query-row block) — 2 blocks per head double
-// the grid (B=8, H=12 -> 192 workgroups) since seq-128 heads are small on
-// a 256-CU chip. K and V head tiles live in LDS (staged per block; tiny
-// and L2-resident); 4 waves own 16 query rows each:
-//   QK^T via mfma_f32_16x16x32_f16 (K tile is already "bt" layout),
-//   row softmax wave-local (rows on 16-lane groups; shfl_xor),
-//   P staged to per-wave LDS as two [16][64] tiles, PV against V^T staged
-//   transposed at load. fp32 accumulation; scores scaled by 1/sqrt(D).
-// No attention mask (synthetic full-length sequences; masked variant
-// planned). Constraints: S == 128, D == 64.
+// One workgroup per (b, h, 64-query-row block); 4 waves own 16 query rows
+// each. Keys/values stream through LDS in 128-key tiles with an ONLINE
+// softmax (flash-attention style running max/sum with accumulator
+// rescaling), so S is any multiple of 128 (BERT 128 ... GPT-2 1024) at
+// fixed LDS footprint:
+//   per key tile: QK^T via mfma_f32_16x16x32 (K tile is "bt" layout),
+//   wave-local running softmax (rows on 16-lane groups; shfl_xor),
+//   P staged to per-wave LDS, PV accumulated against V^T (staged
+//   transposed at load), output rescaled by exp(m_old - m_new).
+// fp32 accumulation; scores scaled by 1/sqrt(D). Constraints: D == 64,
+// S % 128 == 0.
+//
+// Masks: seqlens (optional, [B] device ints) masks right-padded keys
+// >= seqlens[b] (variable-length batches; key tiles past the valid length
+// are skipped entirely). causal != 0 masks keys > query (decoder-style),
+// and skips key tiles entirely above the block's query range.
 #include "gemm_common.h"
 
 namespace trtlab {
 
 // OT: output element type (fp8 e4m3 with out_scale = 1/s_q fuses the
 // producer-side quantization for the following projection GEMM).
-// seqlens (optional, [B] device ints): valid token count per sequence for
-// right-padded variable-length batches — keys >= seqlens[b] are masked to
-// -inf before softmax (zero attention weight). Padded query rows still
-// attend to the valid keys; their outputs are ignored downstream.
 template <typename T, typename OT = T>
 __global__ __launch_bounds__(256) void attention_kernel(
     const T* __restrict__ qkv, OT* __restrict__ out, int B, int S, int H,
-    int D, float scale, float out_scale,
-    const int* __restrict__ seqlens) {
+    int D, float scale, float out_scale, const int* __restrict__ seqlens,
+    int causal) {
   // LDS: Q [64][64] | K [128][64] | Vt 2x[64][64] | P 4 waves x 2x[16][64]
   __shared__ __attribute__((aligned(16))) char smem[8192 + 16384 * 2 + 16384];
   char* Qs = smem;                  // 8 KiB
@@ -40,9 +41,10 @@ __global__ __launch_bounds__(256) void attention_kernel(
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
+  const int qblocks = S >> 6;      // 64-query-row blocks per (b, h)
   const int blk = blockIdx.x;
-  const int bh = blk >> 1;         // (b, h)
-  const int q0 = (blk & 1) * 64;   // this block's query-row base
+  const int bh = blk / qblocks;
+  const int q0 = (blk % qblocks) * 64;  // this block's query-row base
   const int b = bh / H;
   const int h = bh % H;
   const int hid = H * D;
@@ -57,8 +59,7 @@ __global__ __launch_bounds__(256) void attention_kernel(
     return row * 128 + (colbyte ^ ((row & 7) << 4));
   };
 
-  // ---- stage Q (64 rows), K (128 rows), Vt (transposed scatter) ----
-  // Q: 64 rows x 8 chunks = 512 chunks, 2/thread.
+  // ---- stage Q once (64 rows x 8 chunks = 512 chunks, 2/thread) ----
 #pragma unroll
   for (int c = 0; c < 2; ++c) {
     int idx = c * 256 + tid;       // 0..511
@@ -67,142 +68,176 @@ __global__ __launch_bounds__(256) void attention_kernel(
     const T* src = base + (int64_t)(q0 + row) * row_stride + qoff + cb / 2;
     *(short8v*)(Qs + swz(row, cb)) = *(const short8v*)src;
   }
-  // K: 128 rows x 8 chunks = 1024 chunks, 4/thread.
-#pragma unroll
-  for (int c = 0; c < 4; ++c) {
-    int idx = c * 256 + tid;
-    int row = idx >> 3;            // 0..127
-    int cb = (idx & 7) * 16;
-    const T* ksrc = base + (int64_t)row * row_stride + koff + cb / 2;
-    *(short8v*)(Ks + swz(row, cb)) = *(const short8v*)ksrc;
-  }
-  // Vt: read v[key][dd..dd+8] (16 B), scatter transposed to Vt[kt][d][key%64].
-#pragma unroll
-  for (int c = 0; c < 4; ++c) {
-    int ci = c * 256 + tid;        // 0..1023
-    int key = ci >> 3;
-    int dd = (ci & 7) * 8;
-    const T* vsrc = base + (int64_t)key * row_stride + voff + dd;
-    short8v v = *(const short8v*)vsrc;
-    char* tile = Vt + (key >> 6) * 8192;
-    int kcol = key & 63;
-#pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      *(T*)(tile + swz(dd + j, kcol * 2)) = ((const T*)&v)[j];
-    }
-  }
-  __syncthreads();
 
-  // ---- QK^T for this wave's 16 query rows ----
+  int limit = S;
+  if (seqlens) {
+    limit = seqlens[b];
+    limit = limit < 1 ? 1 : (limit > S ? S : limit);
+  }
+  int ntiles = (limit + 127) >> 7;          // key tiles with any valid key
+  if (causal) {
+    int tmax = (q0 + 64 + 127) >> 7;        // keys above the block's queries
+    ntiles = ntiles < tmax ? ntiles : tmax; // contribute nothing
+  }
+
   using MF = Mfma16x16x32<T>;
   const int qrow = wave * 16;      // within the block's 64-row slice
-  f32x4 sacc[8];
+  float m_run[4], l_run[4];
+  f32x4 oacc[4];
 #pragma unroll
-  for (int j = 0; j < 8; ++j) sacc[j] = {0.f, 0.f, 0.f, 0.f};
-#pragma unroll
-  for (int ks = 0; ks < 2; ++ks) {
-    uint32_t kbyte = ks * 64 + ((lane >> 4) << 4);
-    typename MF::frag qf =
-        *(const typename MF::frag*)(Qs + swz(qrow + (lane & 15), kbyte));
-    typename MF::frag kf[8];
-#pragma unroll
-    for (int f = 0; f < 8; ++f)
-      kf[f] = *(const typename MF::frag*)(Ks + swz(f * 16 + (lane & 15), kbyte));
-#pragma unroll
-    for (int j = 0; j < 8; ++j) sacc[j] = MF::run(qf, kf[j], sacc[j]);
+  for (int r = 0; r < 4; ++r) {
+    m_run[r] = -3.0e38f;
+    l_run[r] = 0.f;
   }
+#pragma unroll
+  for (int j = 0; j < 4; ++j) oacc[j] = {0.f, 0.f, 0.f, 0.f};
 
-  // ---- key masking for variable-length sequences ----
-  // score fragment layout: key = j*16 + (lane & 15) (MFMA C col), so the
-  // mask is uniform over r. limit clamped to [1, S] so every query row
-  // keeps at least one finite score (padded queries attend to key 0..L).
-  if (seqlens) {
-    int limit = seqlens[b];
-    limit = limit < 1 ? 1 : (limit > S ? S : limit);
-    if (limit < S) {
+  for (int t = 0; t < ntiles; ++t) {
+    // previous tile's PV reads must finish before K/Vt are overwritten
+    __syncthreads();
+    // ---- stage K tile (128 rows) + Vt (transposed scatter) ----
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+      int idx = c * 256 + tid;
+      int row = idx >> 3;            // 0..127
+      int cb = (idx & 7) * 16;
+      const T* ksrc =
+          base + (int64_t)(t * 128 + row) * row_stride + koff + cb / 2;
+      *(short8v*)(Ks + swz(row, cb)) = *(const short8v*)ksrc;
+    }
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+      int ci = c * 256 + tid;        // 0..1023
+      int key = ci >> 3;
+      int dd = (ci & 7) * 8;
+      const T* vsrc =
+          base + (int64_t)(t * 128 + key) * row_stride + voff + dd;
+      short8v v = *(const short8v*)vsrc;
+      char* tile = Vt + (key >> 6) * 8192;
+      int kcol = key & 63;
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        if (j * 16 + (lane & 15) >= limit) {
+        *(T*)(tile + swz(dd + j, kcol * 2)) = ((const T*)&v)[j];
+      }
+    }
+    __syncthreads();
+
+    // ---- QK^T for this wave's 16 query rows ----
+    f32x4 sacc[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) sacc[j] = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      uint32_t kbyte = ks * 64 + ((lane >> 4) << 4);
+      typename MF::frag qf =
+          *(const typename MF::frag*)(Qs + swz(qrow + (lane & 15), kbyte));
+      typename MF::frag kf[8];
+#pragma unroll
+      for (int f = 0; f < 8; ++f)
+        kf[f] =
+            *(const typename MF::frag*)(Ks + swz(f * 16 + (lane & 15), kbyte));
+#pragma unroll
+      for (int j = 0; j < 8; ++j) sacc[j] = MF::run(qf, kf[j], sacc[j]);
+    }
+
+    // ---- masking ----
+    // score fragment layout: key = t*128 + j*16 + (lane & 15) (MFMA C col),
+    // query row = q0 + qrow + (lane>>4)*4 + r.
+    if (limit - t * 128 < 128) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        if (t * 128 + j * 16 + (lane & 15) >= limit) {
 #pragma unroll
           for (int r = 0; r < 4; ++r) sacc[j][r] = -3.0e38f;
         }
       }
     }
-  }
-
-  // ---- row softmax (rows live on 16-lane groups: shfl_xor 1,2,4,8) ----
+    if (causal && t * 128 + 127 > q0 + qrow) {
 #pragma unroll
-  for (int r = 0; r < 4; ++r) {
-    float m = -3.0e38f;
+      for (int j = 0; j < 8; ++j) {
+        int key = t * 128 + j * 16 + (lane & 15);
 #pragma unroll
-    for (int j = 0; j < 8; ++j) m = fmaxf(m, sacc[j][r] * scale);
-#pragma unroll
-    for (int off = 1; off < 16; off <<= 1)
-      m = fmaxf(m, __shfl_xor(m, off, 64));
-    float s = 0.f;
-#pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      float e = __expf(sacc[j][r] * scale - m);
-      sacc[j][r] = e;
-      s += e;
+        for (int r = 0; r < 4; ++r) {
+          int q = q0 + qrow + ((lane >> 4) << 2) + r;
+          if (key > q) sacc[j][r] = -3.0e38f;
+        }
+      }
     }
-#pragma unroll
-    for (int off = 1; off < 16; off <<= 1) s += __shfl_xor(s, off, 64);
-    float inv = 1.0f / s;
-#pragma unroll
-    for (int j = 0; j < 8; ++j) sacc[j][r] *= inv;
-  }
 
-  // ---- P -> per-wave LDS as two [16][64] fp16 tiles ----
-  char* P = Ps + wave * 4096;
-#pragma unroll
-  for (int j = 0; j < 8; ++j) {
-    int colg = j * 16 + (lane & 15);
-    char* tile = P + (colg >> 6) * 2048;
-    int col = colg & 63;
+    // ---- online softmax update (rows on 16-lane groups; shfl_xor) ----
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
-      int row = ((lane >> 4) << 2) + r;  // 0..15
-      *(T*)(tile + ((uint32_t)row * 128 + ((col * 2) ^ ((row & 7) << 4)))) =
-          (T)sacc[j][r];
+      float m = m_run[r];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) m = fmaxf(m, sacc[j][r] * scale);
+#pragma unroll
+      for (int off = 1; off < 16; off <<= 1)
+        m = fmaxf(m, __shfl_xor(m, off, 64));
+      float c = __expf(m_run[r] - m);  // 0 on the first tile (m_run = -inf)
+      float s = 0.f;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float e = __expf(sacc[j][r] * scale - m);
+        sacc[j][r] = e;
+        s += e;
+      }
+#pragma unroll
+      for (int off = 1; off < 16; off <<= 1) s += __shfl_xor(s, off, 64);
+      l_run[r] = l_run[r] * c + s;
+      m_run[r] = m;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) oacc[j][r] *= c;
+    }
+
+    // ---- P (unnormalized exp weights) -> per-wave LDS, two [16][64] ----
+    char* P = Ps + wave * 4096;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      int colg = j * 16 + (lane & 15);
+      char* tile = P + (colg >> 6) * 2048;
+      int col = colg & 63;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = ((lane >> 4) << 2) + r;  // 0..15
+        *(T*)(tile + ((uint32_t)row * 128 + ((col * 2) ^ ((row & 7) << 4)))) =
+            (T)sacc[j][r];
+      }
+    }
+    // Make the P ds_writes visible before the PV ds_reads (same wave, but
+    // the explicit barrier proved necessary — see git history).
+    __syncthreads();
+
+    // ---- PV: oacc[16 rows][64 d] += P[16][128] @ Vt^T ----
+#pragma unroll
+    for (int kt = 0; kt < 2; ++kt) {
+      const char* Pt = P + kt * 2048;
+      const char* Vk = Vt + kt * 8192;
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        uint32_t kbyte = ks * 64 + ((lane >> 4) << 4);
+        typename MF::frag pf =
+            *(const typename MF::frag*)(Pt + swz(lane & 15, kbyte));
+        typename MF::frag vf[4];
+#pragma unroll
+        for (int f = 0; f < 4; ++f)
+          vf[f] = *(const typename MF::frag*)(Vk +
+                   swz(f * 16 + (lane & 15), kbyte));
+#pragma unroll
+        for (int j = 0; j < 4; ++j) oacc[j] = MF::run(pf, vf[j], oacc[j]);
+      }
     }
   }
-  // Make the P ds_writes visible before the PV ds_reads (same wave, but
-  // the explicit barrier proved necessary — see git history).
-  __syncthreads();
 
-  // ---- PV: out_tile[16 rows][64 d] = P[16][128] @ Vt^T ----
-  f32x4 oacc[4];
-#pragma unroll
-  for (int j = 0; j < 4; ++j) oacc[j] = {0.f, 0.f, 0.f, 0.f};
-#pragma unroll
-  for (int kt = 0; kt < 2; ++kt) {
-    const char* Pt = P + kt * 2048;
-    const char* Vk = Vt + kt * 8192;
-#pragma unroll
-    for (int ks = 0; ks < 2; ++ks) {
-      uint32_t kbyte = ks * 64 + ((lane >> 4) << 4);
-      typename MF::frag pf =
-          *(const typename MF::frag*)(Pt + swz(lane & 15, kbyte));
-      typename MF::frag vf[4];
-#pragma unroll
-      for (int f = 0; f < 4; ++f)
-        vf[f] = *(const typename MF::frag*)(Vk +
-                 swz(f * 16 + (lane & 15), kbyte));
-#pragma unroll
-      for (int j = 0; j < 4; ++j) oacc[j] = MF::run(pf, vf[j], oacc[j]);
-    }
-  }
-
-  // ---- store out[b*S + q0 + row][h*D + d] ----
+  // ---- store out[b*S + q0 + row][h*D + d] = oacc / l ----
 #pragma unroll
   for (int j = 0; j < 4; ++j) {
     int d = j * 16 + (lane & 15);
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       int row = q0 + qrow + ((lane >> 4) << 2) + r;
+      float inv = l_run[r] > 0.f ? 1.0f / l_run[r] : 0.f;
       out[((int64_t)b * S + row) * hid + h * D + d] =
-          store_cast<OT>(oacc[j][r] * out_scale);
+          store_cast<OT>(oacc[j][r] * inv * out_scale);
     }
   }
 }
@@ -225,10 +260,12 @@ void launch_attention_probe(int* dbg, int B, int S, int H, int D,
 
 void launch_attention(int dtype, const void* qkv, void* out, int B, int S,
                       int H, int D, float scale, hipStream_t stream,
-                      int out_dtype, float out_scale, const void* seqlens) {
-  if (S != 128 || D != 64)
-    throw std::runtime_error("attention: only S=128, D=64 supported (BERT-base seq128)");
-  dim3 grid(B * H * 2);  // 2 query-row blocks per head
+                      int out_dtype, float out_scale, const void* seqlens,
+                      int causal) {
+  if (S % 128 != 0 || D != 64)
+    throw std::runtime_error(
+        "attention: S must be a multiple of 128 and D == 64");
+  dim3 grid(B * H * (S / 64));  // 64-query-row blocks
   dim3 block(256);
   const int* lens = (const int*)seqlens;
   if (dtype == 0) {
@@ -236,15 +273,15 @@ void launch_attention(int dtype, const void* qkv, void* out, int B, int S,
       hipLaunchKernelGGL((attention_kernel<_Float16, __hip_fp8_e4m3>), grid,
                          block, 0, stream, (const _Float16*)qkv,
                          (__hip_fp8_e4m3*)out, B, S, H, D, scale, out_scale,
-                         lens);
+                         lens, causal);
     else
       hipLaunchKernelGGL((attention_kernel<_Float16, _Float16>), grid, block,
                          0, stream, (const _Float16*)qkv, (_Float16*)out, B,
-                         S, H, D, scale, out_scale, lens);
+                         S, H, D, scale, out_scale, lens, causal);
   } else {
     hipLaunchKernelGGL((attention_kernel<__bf16, __bf16>), grid, block, 0,
                        stream, (const __bf16*)qkv, (__bf16*)out, B, S, H, D,
-                       scale, out_scale, lens);
+                       scale, out_scale, lens, causal);
   }
 }
 

@@ -58,7 +58,7 @@ void launch_cast(int dtype, bool to_f32, const void* in, void* out, int64_t n,
 void launch_attention(int dtype, const void* qkv, void* out, int B, int S,
                       int H, int D, float scale, hipStream_t stream,
                       int out_dtype = -1, float out_scale = 1.0f,
-                      const void* seqlens = nullptr);
+                      const void* seqlens = nullptr, int causal = 0);
 
 // lens[b] = count of non-pad tokens in right-padded ids (>=1); feeds the
 // variable-length attention mask.

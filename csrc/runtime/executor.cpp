@@ -175,7 +175,7 @@ void ExecutionContext::enqueue_all(hipStream_t s) {
                          op.q_scale != 0.f && op.epi == 3
                              ? 1.0f / op.q_scale
                              : 1.0f,
-                         A(op.in2_off) /*seqlens or null*/);
+                         A(op.in2_off) /*seqlens or null*/, op.causal);
         break;
       case kSeqLens:
         // op.epi carries pad_id

@@ -83,6 +83,7 @@ struct OpDesc {
   int64_t n_elems = 0;
   // attention
   int B = 0, S = 0, NH = 0, HD = 0;
+  int causal = 0;  // decoder-style key > query masking
   float att_scale = 1.0f;
   // int8: residual dequant ratio (s_res/s_out); quantize/dequant scale
   float res_scale = 1.0f;

@@ -412,3 +412,28 @@ def test_gemm_mxfp4(C, M, N, K):
     full = torch.from_numpy(a32) @ torch.from_numpy(b32).t()
     corr = np.corrcoef(out.cpu().numpy().ravel(), full.numpy().ravel())[0, 1]
     assert corr > 0.95, corr  # fp4: coarser grid than fp8
+
+
+@pytest.mark.parametrize("S,causal", [(128, 0), (256, 0), (512, 0),
+                                      (128, 1), (256, 1)])
+def test_attention_long_and_causal(C, S, causal):
+    """Online-softmax attention at S > 128 (streamed key tiles) and with
+    the decoder-style causal mask, vs plain torch attention."""
+    B, H, D = 2, 4, 64
+    hid = H * D
+    qkv = t16(B * S, 3 * hid, seed=100 + S + causal)
+    out = torch.empty(B * S, hid, dtype=torch.half, device="cuda")
+    torch.cuda.synchronize()
+    C.ops.attention(0, qkv.data_ptr(), out.data_ptr(), B, S, H, D,
+                    1.0 / (D ** 0.5), causal=causal)
+    q = qkv.float().reshape(B, S, 3, H, D)
+    scores = (q[:, :, 0].permute(0, 2, 1, 3) @
+              q[:, :, 1].permute(0, 2, 1, 3).transpose(-1, -2)) / (D ** 0.5)
+    if causal:
+        cm = torch.arange(S, device="cuda")[None, :] > \
+            torch.arange(S, device="cuda")[:, None]
+        scores = scores.masked_fill(cm[None, None], float("-inf"))
+    att = torch.softmax(scores, dim=-1)
+    ref = (att @ q[:, :, 2].permute(0, 2, 1, 3)).permute(
+        0, 2, 1, 3).reshape(B * S, hid)
+    check(out, ref)

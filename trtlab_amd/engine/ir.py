@@ -172,14 +172,17 @@ class Graph:
 
     def attention(self, qkv: str, heads: int, seq: int,
                   varlen: bool = False, pad_id: int = 0,
+                  causal: bool = False,
                   name: Optional[str] = None) -> str:
         """varlen: mask keys beyond each sequence's valid length (derived
-        from right-padded token ids; requires an i32 ids graph input)."""
+        from right-padded token ids; requires an i32 ids graph input).
+        causal: decoder-style key > query masking (GPT-family)."""
         m, k3 = self.tensors[qkv].shape
         hid = k3 // 3
         return self._emit("attention", [qkv], (m, hid),
                           dict(heads=heads, seq=seq, head_dim=hid // heads,
-                               varlen=varlen, pad_id=pad_id),
+                               varlen=varlen, pad_id=pad_id,
+                               causal=causal),
                           name)
 
     # ------------------------------------------------------------ helpers

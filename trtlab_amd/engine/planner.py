@@ -533,7 +533,8 @@ class Planner:
                 d.update(kind=K_ATTENTION, B=m // seq, S=seq, NH=heads, HD=hd,
                          att_scale=1.0 / float(np.sqrt(hd)),
                          epi=op.params.get("out_dtype", 0),
-                         q_scale=op.params.get("q_scale", 0.0))
+                         q_scale=op.params.get("q_scale", 0.0),
+                         causal=1 if op.params.get("causal") else 0)
             else:
                 raise ValueError(f"bad exec op kind {op.kind}")
             op_dicts.append(d)

@@ -191,6 +191,9 @@ def run_reference(plan: EnginePlan, input_nhwc: np.ndarray,
                 keymask = torch.arange(s)[None, :] >= lens[:, None]  # [B,S]
                 scores = scores.masked_fill(
                     keymask[:, None, None, :], float("-inf"))
+            if d.get("causal"):
+                cm = torch.arange(s)[None, :] > torch.arange(s)[:, None]
+                scores = scores.masked_fill(cm[None, None], float("-inf"))
             att = torch.softmax(scores, dim=-1)
             y = (att @ v).permute(0, 2, 1, 3).reshape(b * s, hid)
             if d.get("epi") == 3:  # fused fp8 output
