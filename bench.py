@@ -35,7 +35,7 @@ def main():
     ap.add_argument("--contexts", type=int, default=3)
     ap.add_argument("--model", default="resnet50",
                     choices=["resnet50", "resnet101", "resnet152", "bert",
-                             "bert-large"])
+                             "bert-large", "gpt2"])
     ap.add_argument("--dtype", default="fp16",
                     choices=["fp16", "bf16", "int8", "fp8", "mxfp4", "mxfp8"])
     ap.add_argument("--no-autotune", action="store_true",
@@ -72,6 +72,13 @@ def main():
         g = build_bert(batch=args.batch, seq=128, hidden=1024, heads=16,
                        layers=24, seed=0)
         cfg_extra = {"seq_len": 128, "hidden": 1024, "layers": 24}
+    elif args.model == "gpt2":
+        # full-sequence forward = prefill (causal online-softmax attention)
+        from trtlab_amd.models import build_gpt2
+
+        g = build_gpt2(batch=args.batch, seq=1024, layers=12, seed=0)
+        cfg_extra = {"seq_len": 1024, "hidden": 768, "layers": 12,
+                     "phase": "prefill"}
     else:
         depth = int(args.model.replace("resnet", ""))
         g = build_resnet(depth, batch=args.batch, image=224, seed=0)
