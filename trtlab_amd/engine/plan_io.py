@@ -46,8 +46,14 @@ def save_plan(plan: EnginePlan, path: str) -> None:
             arrays[f"s{i}"] = o.scale
         if o.bias is not None:
             arrays[f"b{i}"] = o.bias
-    np.savez_compressed(path, meta=np.frombuffer(
-        json.dumps(meta).encode(), dtype=np.uint8), **arrays)
+        for k, v in o.params.items():  # ndarray params (e.g. MX wt scales)
+            if isinstance(v, np.ndarray):
+                arrays[f"p{i}_{k}"] = v
+    # write through a file object so the exact path is honored
+    # (np.savez appends .npz to bare string paths)
+    with open(path, "wb") as f:
+        np.savez_compressed(f, meta=np.frombuffer(
+            json.dumps(meta).encode(), dtype=np.uint8), **arrays)
 
 
 def load_plan(path: str) -> EnginePlan:
@@ -65,6 +71,9 @@ def load_plan(path: str) -> EnginePlan:
             op.scale = z[f"s{i}"]
         if f"b{i}" in z:
             op.bias = z[f"b{i}"]
+        for key in z.files:
+            if key.startswith(f"p{i}_"):
+                op.params[key[len(f"p{i}_"):]] = z[key]
         exec_ops.append(op)
     return EnginePlan(
         name=meta["name"],
