@@ -142,6 +142,26 @@ PYBIND11_MODULE(_C, m) {
     py::gil_scoped_release rel;
     TRT_HIP_CHECK(hipStreamSynchronize(as_stream(s)));
   });
+  // generic stream capture (DecodeSession and power users build their own
+  // replayable graphs from raw ops)
+  hip.def("stream_begin_capture", [](uintptr_t s) {
+    TRT_HIP_CHECK(hipStreamBeginCapture(as_stream(s),
+                                        hipStreamCaptureModeRelaxed));
+  });
+  hip.def("stream_end_capture", [](uintptr_t s) {
+    hipGraph_t g{};
+    TRT_HIP_CHECK(hipStreamEndCapture(as_stream(s), &g));
+    hipGraphExec_t e{};
+    TRT_HIP_CHECK(hipGraphInstantiate(&e, g, nullptr, nullptr, 0));
+    TRT_HIP_CHECK(hipGraphDestroy(g));
+    return (uintptr_t)e;
+  });
+  hip.def("graph_launch", [](uintptr_t e, uintptr_t s) {
+    TRT_HIP_CHECK(hipGraphLaunch((hipGraphExec_t)e, as_stream(s)));
+  });
+  hip.def("graph_destroy", [](uintptr_t e) {
+    (void)hipGraphExecDestroy((hipGraphExec_t)e);
+  });
   hip.def("device_properties", [](int dev) {
     hipDeviceProp_t p;
     TRT_HIP_CHECK(hipGetDeviceProperties(&p, dev));
@@ -363,6 +383,45 @@ PYBIND11_MODULE(_C, m) {
           py::arg("A"), py::arg("B"), py::arg("Sa"), py::arg("Sb"),
           py::arg("C"), py::arg("M"), py::arg("N"), py::arg("K"),
           py::arg("stream") = 0, py::arg("sync") = true);
+  ops.def("kv_append",
+          [](uintptr_t qkv, uintptr_t kc, uintptr_t vc, uintptr_t pos, int B,
+             int H, int smax, uintptr_t stream, bool sync) {
+            launch_kv_append((void*)qkv, (void*)kc, (void*)vc, (void*)pos, B,
+                             H, smax, as_stream(stream));
+            if (sync) TRT_HIP_CHECK(hipStreamSynchronize(as_stream(stream)));
+          },
+          py::arg("qkv"), py::arg("kcache"), py::arg("vcache"),
+          py::arg("pos"), py::arg("B"), py::arg("H"), py::arg("smax"),
+          py::arg("stream") = 0, py::arg("sync") = true);
+  ops.def("decode_attention",
+          [](uintptr_t qkv, uintptr_t kc, uintptr_t vc, uintptr_t out,
+             uintptr_t pos, int B, int H, int smax, float scale,
+             uintptr_t stream, bool sync) {
+            launch_decode_attention((void*)qkv, (void*)kc, (void*)vc,
+                                    (void*)out, (void*)pos, B, H, smax,
+                                    scale, as_stream(stream));
+            if (sync) TRT_HIP_CHECK(hipStreamSynchronize(as_stream(stream)));
+          },
+          py::arg("qkv"), py::arg("kcache"), py::arg("vcache"),
+          py::arg("out"), py::arg("pos"), py::arg("B"), py::arg("H"),
+          py::arg("smax"), py::arg("scale"), py::arg("stream") = 0,
+          py::arg("sync") = true);
+  ops.def("decode_embed",
+          [](uintptr_t ids, uintptr_t tok, uintptr_t pe, uintptr_t out,
+             uintptr_t pos, int B, int hidden, uintptr_t stream, bool sync) {
+            launch_decode_embed((void*)ids, (void*)tok, (void*)pe, (void*)out,
+                                (void*)pos, B, hidden, as_stream(stream));
+            if (sync) TRT_HIP_CHECK(hipStreamSynchronize(as_stream(stream)));
+          },
+          py::arg("ids"), py::arg("tok"), py::arg("posemb"), py::arg("out"),
+          py::arg("pos"), py::arg("B"), py::arg("hidden"),
+          py::arg("stream") = 0, py::arg("sync") = true);
+  ops.def("advance_pos", [](uintptr_t pos, int smax, uintptr_t stream,
+                            bool sync) {
+    launch_advance_pos((void*)pos, smax, as_stream(stream));
+    if (sync) TRT_HIP_CHECK(hipStreamSynchronize(as_stream(stream)));
+  }, py::arg("pos"), py::arg("smax"), py::arg("stream") = 0,
+     py::arg("sync") = true);
   ops.def("attention",
           [](int dtype, uintptr_t qkv, uintptr_t out, int B, int S, int H,
              int D, float scale, uintptr_t stream, bool sync, int causal,

@@ -88,6 +88,19 @@ void launch_gemm_mxfp8(const void* A, const void* B, const void* Sa,
 void launch_quantize_mxfp8(const void* x, void* codes, void* scales,
                            int64_t m, int64_t k, hipStream_t stream);
 
+// ---- incremental decode (KV cache; see csrc/kernels/decode.hip) ----
+void launch_kv_append(const void* qkv, void* kcache, void* vcache,
+                      const void* pos, int B, int H, int smax,
+                      hipStream_t stream);
+void launch_decode_attention(const void* qkv, const void* kcache,
+                             const void* vcache, void* out, const void* pos,
+                             int B, int H, int smax, float scale,
+                             hipStream_t stream);
+void launch_decode_embed(const void* ids, const void* tok, const void* posemb,
+                         void* out, const void* pos, int B, int hidden,
+                         hipStream_t stream);
+void launch_advance_pos(void* pos, int smax, hipStream_t stream);
+
 void launch_embedding(int dtype, const void* ids, const void* tok,
                       const void* pos, const void* seg, const void* segids,
                       void* out, int M, int S, int H, hipStream_t stream);

@@ -18,6 +18,7 @@ ARCH = os.environ.get("TRTLAB_GPU_ARCH", "gfx950")
 SOURCES = [
     "csrc/kernels/gemm.hip",
     "csrc/kernels/gemm_mx.hip",
+    "csrc/kernels/decode.hip",
     "csrc/kernels/conv.hip",
     "csrc/kernels/splitk.hip",
     "csrc/kernels/pool.hip",

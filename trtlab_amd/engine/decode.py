@@ -1,0 +1,202 @@
+"""Incremental (KV-cache) decoding for GPT-2-family models.
+
+Beyond-reference capability (the CUDA reference served static TensorRT
+engines only): a `DecodeSession` holds per-layer K/V caches resident in
+HBM and replays ONE hipGraph per generated token. All position dependence
+flows through a device-side counter (csrc/kernels/decode.hip), so the
+captured graph needs no re-instantiation between steps:
+
+    embed(ids, pos) -> [per layer: ln1 -> qkv gemm -> kv_append ->
+    decode_attention -> proj gemm -> +res -> ln2 -> ff1+gelu -> ff2 ->
+    +res] -> ln_f -> advance_pos
+
+Weights come from a `build_gpt2(embeddings=True)` IR graph (the same
+random-init builder the full-sequence engine uses, so prefill/decode can
+be cross-checked). Prefill here is sequential priming (step the prompt
+token by token); fused prefill-into-cache is round-2 work.
+"""
+from __future__ import annotations
+
+from typing import Dict, List, Optional
+
+import numpy as np
+
+
+class DecodeSession:
+    """One generation session: fixed batch, growing position."""
+
+    def __init__(self, graph, batch: int, smax: int = 1024, device: int = 0,
+                 capture: bool = True, lm_head: bool = False):
+        import torch
+
+        from trtlab_amd import native
+        from trtlab_amd.engine.planner import (EPI_BIAS, EPI_BIAS_GELU,
+                                               EPI_NONE)
+
+        self._C = native()
+        self._torch = torch
+        torch.cuda.set_device(device)
+        self.batch = batch
+        self.smax = smax
+        self.capture = capture
+        self._epi_bias = EPI_BIAS
+        self._epi_gelu = EPI_BIAS_GELU
+        self._epi_none = EPI_NONE
+
+        # ---- pull weights out of the IR graph by node kind/name ----
+        nodes = {n.name: n for n in graph.nodes}
+        emb = next(n for n in graph.nodes if n.kind == "embedding")
+        self.hidden = emb.attrs["tok"].shape[1]
+        self.vocab = emb.attrs["tok"].shape[0]
+        assert emb.attrs["pos"].shape[0] >= 1
+
+        def dev16(a):
+            return torch.from_numpy(np.ascontiguousarray(a, np.float32)) \
+                .half().cuda()
+
+        def dev32(a):
+            return torch.from_numpy(np.ascontiguousarray(a, np.float32)).cuda()
+
+        self.tok = dev16(emb.attrs["tok"])
+        self.posemb = dev16(emb.attrs["pos"])
+        if self.posemb.shape[0] < smax:
+            self.smax = smax = int(self.posemb.shape[0])
+
+        self.layers: List[Dict] = []
+        li = 0
+        while f"l{li}_qkv" in nodes:
+            att = nodes[f"l{li}_att"]
+            self.heads = att.attrs["heads"]
+            lay = {}
+            for key, nm in (("ln1", f"l{li}_ln1"), ("ln2", f"l{li}_ln2")):
+                lay[key + "_g"] = dev32(nodes[nm].attrs["gamma"])
+                lay[key + "_b"] = dev32(nodes[nm].attrs["beta"])
+            for key in ("qkv", "proj", "ff1", "ff2"):
+                n = nodes[f"l{li}_{key}"]
+                lay[key + "_w"] = dev16(n.attrs["weight"])
+                lay[key + "_b"] = dev32(n.attrs["bias"])
+            lay["kcache"] = torch.zeros(batch, self.heads, smax, 64,
+                                        dtype=torch.half, device="cuda")
+            lay["vcache"] = torch.zeros_like(lay["kcache"])
+            self.layers.append(lay)
+            li += 1
+        self.n_layers = li
+        self.inter = self.layers[0]["ff1_w"].shape[0]
+        gf = nodes["ln_f"]
+        self.lnf_g = dev32(gf.attrs["gamma"])
+        self.lnf_b = dev32(gf.attrs["beta"])
+        self.lm_head = lm_head  # logits = h @ tok^T (weight tying)
+
+        B, Hd = batch, self.hidden
+        self.ids = torch.zeros(B, dtype=torch.int32, device="cuda")
+        self.pos = torch.zeros(1, dtype=torch.int32, device="cuda")
+        self.h = torch.zeros(B, Hd, dtype=torch.half, device="cuda")
+        self.x = torch.zeros(B, Hd, dtype=torch.half, device="cuda")
+        self.qkv = torch.zeros(B, 3 * Hd, dtype=torch.half, device="cuda")
+        self.att = torch.zeros(B, Hd, dtype=torch.half, device="cuda")
+        self.ff = torch.zeros(B, self.inter, dtype=torch.half, device="cuda")
+        self.out = torch.zeros(B, Hd, dtype=torch.half, device="cuda")
+        self.logits = (torch.zeros(B, self.vocab, dtype=torch.half,
+                                   device="cuda") if lm_head else None)
+        self.stream = self._C.hip.stream_create()
+        self._graph = 0
+        self._steps = 0
+
+    # ------------------------------------------------------------ plumbing
+    def _enqueue(self):
+        """Record one decode step's kernels on self.stream (pos-relative:
+        kv_append/decode_attention/embed all read the device counter)."""
+        C, s = self._C, self.stream
+        B, Hd = self.batch, self.hidden
+        ops = C.ops
+        ops.decode_embed(self.ids.data_ptr(), self.tok.data_ptr(),
+                         self.posemb.data_ptr(), self.h.data_ptr(),
+                         self.pos.data_ptr(), B, Hd, stream=s, sync=False)
+        for lay in self.layers:
+            ops.layernorm(0, self.h.data_ptr(), lay["ln1_g"].data_ptr(),
+                          lay["ln1_b"].data_ptr(), self.x.data_ptr(), B, Hd,
+                          stream=s, sync=False)
+            ops.gemm_bt(0, self.x.data_ptr(), lay["qkv_w"].data_ptr(),
+                        self.qkv.data_ptr(), bias=lay["qkv_b"].data_ptr(),
+                        M=B, N=3 * Hd, K=Hd, epi=self._epi_bias, stream=s,
+                        sync=False)
+            ops.kv_append(self.qkv.data_ptr(), lay["kcache"].data_ptr(),
+                          lay["vcache"].data_ptr(), self.pos.data_ptr(), B,
+                          self.heads, self.smax, stream=s, sync=False)
+            ops.decode_attention(self.qkv.data_ptr(),
+                                 lay["kcache"].data_ptr(),
+                                 lay["vcache"].data_ptr(),
+                                 self.att.data_ptr(), self.pos.data_ptr(), B,
+                                 self.heads, self.smax,
+                                 1.0 / float(np.sqrt(64.0)), stream=s,
+                                 sync=False)
+            ops.gemm_bt(0, self.att.data_ptr(), lay["proj_w"].data_ptr(),
+                        self.x.data_ptr(), bias=lay["proj_b"].data_ptr(),
+                        M=B, N=Hd, K=Hd, epi=self._epi_bias, stream=s,
+                        sync=False)
+            ops.elementwise(0, 2, self.h.data_ptr(), self.x.data_ptr(),
+                            self.h.data_ptr(), B * Hd, stream=s, sync=False)
+            ops.layernorm(0, self.h.data_ptr(), lay["ln2_g"].data_ptr(),
+                          lay["ln2_b"].data_ptr(), self.x.data_ptr(), B, Hd,
+                          stream=s, sync=False)
+            ops.gemm_bt(0, self.x.data_ptr(), lay["ff1_w"].data_ptr(),
+                        self.ff.data_ptr(), bias=lay["ff1_b"].data_ptr(),
+                        M=B, N=self.inter, K=Hd, epi=self._epi_gelu,
+                        stream=s, sync=False)
+            ops.gemm_bt(0, self.ff.data_ptr(), lay["ff2_w"].data_ptr(),
+                        self.x.data_ptr(), bias=lay["ff2_b"].data_ptr(),
+                        M=B, N=Hd, K=self.inter, epi=self._epi_bias,
+                        stream=s, sync=False)
+            ops.elementwise(0, 2, self.h.data_ptr(), self.x.data_ptr(),
+                            self.h.data_ptr(), B * Hd, stream=s, sync=False)
+        ops.layernorm(0, self.h.data_ptr(), self.lnf_g.data_ptr(),
+                      self.lnf_b.data_ptr(), self.out.data_ptr(), B, Hd,
+                      stream=s, sync=False)
+        if self.logits is not None:
+            ops.gemm_bt(0, self.out.data_ptr(), self.tok.data_ptr(),
+                        self.logits.data_ptr(), M=B, N=self.vocab, K=Hd,
+                        epi=self._epi_none, stream=s, sync=False)
+        ops.advance_pos(self.pos.data_ptr(), self.smax, stream=s, sync=False)
+
+    def step(self, ids: np.ndarray) -> np.ndarray:
+        """Feed one token per sequence; returns the final hidden state
+        [B, hidden] fp32 (or logits [B, vocab] with lm_head=True). The
+        device-side position counter starts at 0 and the captured graph
+        advances it, so replays need no host-side position plumbing."""
+        if self._steps >= self.smax:
+            raise RuntimeError("DecodeSession: sequence limit reached")
+        arr = np.ascontiguousarray(ids, np.int32)
+        # synchronous H2D keeps the token copy ordered before the replay
+        self._C.memory.memcpy_h2d(self.ids.data_ptr(), arr, arr.nbytes)
+        if self.capture:
+            if not self._graph:
+                # step 0 runs eagerly as the warm-up, then a fresh step is
+                # RECORDED (capture does not execute) for replay from step 1
+                self._enqueue()
+                self._C.hip.stream_synchronize(self.stream)
+                self._C.hip.stream_begin_capture(self.stream)
+                self._enqueue()
+                self._graph = self._C.hip.stream_end_capture(self.stream)
+                self._steps += 1
+                out = self.out.float().cpu().numpy()
+                return (self.logits.float().cpu().numpy()
+                        if self.logits is not None else out)
+            self._C.hip.graph_launch(self._graph, self.stream)
+        else:
+            self._enqueue()
+        self._C.hip.stream_synchronize(self.stream)
+        self._steps += 1
+        out = self.out.float().cpu().numpy()
+        return (self.logits.float().cpu().numpy()
+                if self.logits is not None else out)
+
+    def close(self):
+        if self._graph:
+            self._C.hip.graph_destroy(self._graph)
+            self._graph = 0
+
+    def __del__(self):
+        try:
+            self.close()
+        except Exception:
+            pass
