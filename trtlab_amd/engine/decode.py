@@ -94,6 +94,7 @@ class DecodeSession:
         self.x = torch.zeros(B, Hd, dtype=torch.half, device="cuda")
         self.qkv = torch.zeros(B, 3 * Hd, dtype=torch.half, device="cuda")
         self.att = torch.zeros(B, Hd, dtype=torch.half, device="cuda")
+        self.x2 = torch.zeros(B, Hd, dtype=torch.half, device="cuda")
         self.ff = torch.zeros(B, self.inter, dtype=torch.half, device="cuda")
         self.out = torch.zeros(B, Hd, dtype=torch.half, device="cuda")
         self.logits = (torch.zeros(B, self.vocab, dtype=torch.half,
@@ -105,21 +106,31 @@ class DecodeSession:
     # ------------------------------------------------------------ plumbing
     def _enqueue(self):
         """Record one decode step's kernels on self.stream (pos-relative:
-        kv_append/decode_attention/embed all read the device counter)."""
+        kv_append/decode_attention/embed all read the device counter).
+
+        The step is kernel-COUNT bound (~4.6 us graph-replay floor per
+        kernel regardless of size - profiles/dec_kernel_stats), so every
+        residual add is fused into the following layernorm (sum_out
+        updates the residual stream in the same kernel) and split-K is
+        disabled on the tiny M=B gemms via the tile hint (the fp32-slab
+        reduce kernel doubled the gemm count for no win at this floor).
+        """
         C, s = self._C, self.stream
         B, Hd = self.batch, self.hidden
+        T4 = 4  # 64x64 tile hint: no split-K
         ops = C.ops
         ops.decode_embed(self.ids.data_ptr(), self.tok.data_ptr(),
                          self.posemb.data_ptr(), self.h.data_ptr(),
                          self.pos.data_ptr(), B, Hd, stream=s, sync=False)
-        for lay in self.layers:
-            ops.layernorm(0, self.h.data_ptr(), lay["ln1_g"].data_ptr(),
-                          lay["ln1_b"].data_ptr(), self.x.data_ptr(), B, Hd,
-                          stream=s, sync=False)
+        ops.layernorm(0, self.h.data_ptr(),
+                      self.layers[0]["ln1_g"].data_ptr(),
+                      self.layers[0]["ln1_b"].data_ptr(), self.x.data_ptr(),
+                      B, Hd, stream=s, sync=False)
+        for li, lay in enumerate(self.layers):
             ops.gemm_bt(0, self.x.data_ptr(), lay["qkv_w"].data_ptr(),
                         self.qkv.data_ptr(), bias=lay["qkv_b"].data_ptr(),
                         M=B, N=3 * Hd, K=Hd, epi=self._epi_bias, stream=s,
-                        sync=False)
+                        sync=False, tile=T4)
             ops.kv_append(self.qkv.data_ptr(), lay["kcache"].data_ptr(),
                           lay["vcache"].data_ptr(), self.pos.data_ptr(), B,
                           self.heads, self.smax, stream=s, sync=False)
@@ -131,27 +142,31 @@ class DecodeSession:
                                  1.0 / float(np.sqrt(64.0)), stream=s,
                                  sync=False)
             ops.gemm_bt(0, self.att.data_ptr(), lay["proj_w"].data_ptr(),
-                        self.x.data_ptr(), bias=lay["proj_b"].data_ptr(),
+                        self.x2.data_ptr(), bias=lay["proj_b"].data_ptr(),
                         M=B, N=Hd, K=Hd, epi=self._epi_bias, stream=s,
-                        sync=False)
-            ops.elementwise(0, 2, self.h.data_ptr(), self.x.data_ptr(),
-                            self.h.data_ptr(), B * Hd, stream=s, sync=False)
-            ops.layernorm(0, self.h.data_ptr(), lay["ln2_g"].data_ptr(),
-                          lay["ln2_b"].data_ptr(), self.x.data_ptr(), B, Hd,
-                          stream=s, sync=False)
+                        sync=False, tile=T4)
+            # h += proj; x = ln2(h)   (one kernel: sum_out = new residual)
+            ops.add_layernorm(0, self.x2.data_ptr(), self.h.data_ptr(),
+                              lay["ln2_g"].data_ptr(),
+                              lay["ln2_b"].data_ptr(), self.x.data_ptr(),
+                              sum_out=self.h.data_ptr(), M=B, N=Hd,
+                              stream=s, sync=False)
             ops.gemm_bt(0, self.x.data_ptr(), lay["ff1_w"].data_ptr(),
                         self.ff.data_ptr(), bias=lay["ff1_b"].data_ptr(),
                         M=B, N=self.inter, K=Hd, epi=self._epi_gelu,
-                        stream=s, sync=False)
+                        stream=s, sync=False, tile=T4)
             ops.gemm_bt(0, self.ff.data_ptr(), lay["ff2_w"].data_ptr(),
-                        self.x.data_ptr(), bias=lay["ff2_b"].data_ptr(),
+                        self.x2.data_ptr(), bias=lay["ff2_b"].data_ptr(),
                         M=B, N=Hd, K=self.inter, epi=self._epi_bias,
-                        stream=s, sync=False)
-            ops.elementwise(0, 2, self.h.data_ptr(), self.x.data_ptr(),
-                            self.h.data_ptr(), B * Hd, stream=s, sync=False)
-        ops.layernorm(0, self.h.data_ptr(), self.lnf_g.data_ptr(),
-                      self.lnf_b.data_ptr(), self.out.data_ptr(), B, Hd,
-                      stream=s, sync=False)
+                        stream=s, sync=False, tile=T4)
+            # h += ff2; x = next ln1(h) (or ln_f at the end)
+            nxt = (self.layers[li + 1] if li + 1 < self.n_layers else None)
+            gptr = (nxt["ln1_g"] if nxt else self.lnf_g).data_ptr()
+            bptr = (nxt["ln1_b"] if nxt else self.lnf_b).data_ptr()
+            dst = (self.x if nxt else self.out).data_ptr()
+            ops.add_layernorm(0, self.x2.data_ptr(), self.h.data_ptr(),
+                              gptr, bptr, dst, sum_out=self.h.data_ptr(),
+                              M=B, N=Hd, stream=s, sync=False)
         if self.logits is not None:
             ops.gemm_bt(0, self.out.data_ptr(), self.tok.data_ptr(),
                         self.logits.data_ptr(), M=B, N=self.vocab, K=Hd,
