@@ -117,7 +117,7 @@ class DecodeSession:
         """
         C, s = self._C, self.stream
         B, Hd = self.batch, self.hidden
-        T4 = 4  # 64x64 tile hint: no split-K
+        T4 = 0  # heuristic tiles (split-K allowed)
         ops = C.ops
         ops.decode_embed(self.ids.data_ptr(), self.tok.data_ptr(),
                          self.posemb.data_ptr(), self.h.data_ptr(),
