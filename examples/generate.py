@@ -1,0 +1,61 @@
+#!/usr/bin/env python3
+"""GPT-2-style greedy generation on the incremental decode path (GPU).
+
+Random-init weights (no network for checkpoints), so the "text" is noise —
+this demonstrates the serving mechanics: one hipGraph replay per token
+against resident KV caches, with logits from the weight-tied lm head.
+
+    python examples/generate.py --batch 4 --prompt-len 16 --new-tokens 32
+"""
+import argparse
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+import numpy as np
+
+from trtlab_amd.engine.decode import DecodeSession
+from trtlab_amd.models import build_gpt2
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--batch", type=int, default=4)
+    ap.add_argument("--layers", type=int, default=12)
+    ap.add_argument("--prompt-len", type=int, default=16)
+    ap.add_argument("--new-tokens", type=int, default=32)
+    args = ap.parse_args()
+
+    g = build_gpt2(batch=args.batch, seq=1024, layers=args.layers, seed=0,
+                   embeddings=True)
+    sess = DecodeSession(g, batch=args.batch, smax=1024, lm_head=True)
+
+    rng = np.random.RandomState(0)
+    prompt = rng.randint(1, 50257, (args.batch, args.prompt_len))
+    # prefill = sequential priming (fused prefill-into-cache: round 2)
+    for t in range(args.prompt_len):
+        logits = sess.step(prompt[:, t].astype(np.int32))
+
+    toks = np.argmax(logits, axis=1).astype(np.int32)
+    out = [toks]
+    t0 = time.perf_counter()
+    for _ in range(args.new_tokens - 1):
+        logits = sess.step(toks)
+        toks = np.argmax(logits, axis=1).astype(np.int32)
+        out.append(toks)
+    dt = time.perf_counter() - t0
+    seqs = np.stack(out, axis=1)
+    print("generated token ids (greedy):")
+    for b in range(args.batch):
+        print(f"  seq{b}: {seqs[b][:16].tolist()} ...")
+    rate = args.batch * (args.new_tokens - 1) / dt
+    print(f"decode rate: {rate:,.0f} tok/s "
+          f"({dt / (args.new_tokens - 1) * 1e3:.2f} ms/step, "
+          f"batch {args.batch}, lm head incl.)")
+    sess.close()
+
+
+if __name__ == "__main__":
+    main()

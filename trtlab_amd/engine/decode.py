@@ -17,7 +17,7 @@ token by token); fused prefill-into-cache is round-2 work.
 """
 from __future__ import annotations
 
-from typing import Dict, List, Optional
+from typing import Dict, List
 
 import numpy as np
 
