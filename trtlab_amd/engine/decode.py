@@ -10,6 +10,11 @@ captured graph needs no re-instantiation between steps:
     decode_attention -> proj gemm -> +res -> ln2 -> ff1+gelu -> ff2 ->
     +res] -> ln_f -> advance_pos
 
+Caveat: the raw-op GEMMs share the lazily-grown module-level scratch
+buffer (csrc/ext.cpp test_scratch); the captured graph bakes its address,
+so other raw-op users must not force a regrow while a session is live
+(engine-managed contexts are unaffected).
+
 Weights come from a `build_gpt2(embeddings=True)` IR graph (the same
 random-init builder the full-sequence engine uses, so prefill/decode can
 be cross-checked). Prefill here is sequential priming (step the prompt
