@@ -38,7 +38,7 @@ try:
         time.sleep(1)
 
     mgr = RemoteInferenceManager("127.0.0.1:50953")
-    runner = mgr.infer_runner("resnet50", use_shm=True)
+    runner = mgr.infer_runner("resnet50", use_shm=True)  # pooled segments
     batch = np.random.randn(8, 224, 224, 3).astype(np.float16)
 
     # pipelined: keep `depth` requests in flight
@@ -65,6 +65,7 @@ try:
         if depth == 1:
             extra = f"  (sync loop)"
         print(f"shm depth={depth:>2}: {infs:8.0f} inf/s{extra}", flush=True)
+    runner.close()
     mgr.close()
 finally:
     server.terminate()

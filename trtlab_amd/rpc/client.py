@@ -43,6 +43,51 @@ class ShmInput:
             pass
 
 
+
+
+class ShmPool:
+    """Reusable POSIX shared-memory segments for the zero-copy transport:
+    per-request segment creation (shm_open + mmap + unlink) measured as
+    the dominant client-side cost of ShmInput (profiles/README), so a
+    runner checks segments out of this pool and returns them when the
+    response lands. Sized for `depth` concurrent requests."""
+
+    def __init__(self, nbytes: int, depth: int = 8):
+        import queue
+        from multiprocessing import shared_memory
+
+        self.nbytes = nbytes
+        self._q = queue.Queue()
+        self._all = []
+        for _ in range(depth):
+            shm = shared_memory.SharedMemory(create=True, size=nbytes)
+            self._all.append(shm)
+            self._q.put(shm)
+
+    def checkout(self, arr) -> "tuple":
+        import numpy as np
+
+        a = np.ascontiguousarray(arr)
+        assert a.nbytes <= self.nbytes
+        shm = self._q.get()  # blocks when `depth` requests are in flight
+        shm.buf[:a.nbytes] = a.tobytes()
+        return shm, shm.name, a.nbytes
+
+    def release(self, shm) -> None:
+        self._q.put(shm)
+
+    def close(self):
+        while not self._q.empty():
+            self._q.get_nowait()
+        for shm in self._all:
+            shm.close()
+            try:
+                shm.unlink()
+            except FileNotFoundError:
+                pass
+        self._all.clear()
+
+
 class SyncClient:
     """Blocking unary client over a shared channel."""
 

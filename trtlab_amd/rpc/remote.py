@@ -8,7 +8,7 @@ from typing import Optional, Tuple
 
 import numpy as np
 
-from trtlab_amd.rpc.client import AsyncClient, ShmInput
+from trtlab_amd.rpc.client import AsyncClient, ShmInput, ShmPool
 from trtlab_amd.rpc.proto import (HealthRequest, HealthResponse, InferRequest,
                                   InferResponse)
 
@@ -17,26 +17,32 @@ class RemoteInferRunner:
     """infer(batch) -> Future[np.ndarray], like the local InferRunner."""
 
     def __init__(self, manager: "RemoteInferenceManager", model: str,
-                 use_shm: bool = False):
+                 use_shm: bool = False, shm_depth: int = 8):
         self._m = manager
         self.model = model
-        self.use_shm = use_shm  # zero-copy local transport
+        self.use_shm = use_shm  # zero-copy local transport (pooled)
+        self._shm_depth = shm_depth
+        self._pool: ShmPool | None = None
 
     def infer(self, batch: np.ndarray) -> Future:
         batch = np.ascontiguousarray(batch, np.float16)
-        shm = ShmInput(batch) if self.use_shm else None
+        seg = name = None
+        size = 0
+        if self.use_shm:
+            if self._pool is None:
+                self._pool = ShmPool(batch.nbytes, depth=self._shm_depth)
+            seg, name, size = self._pool.checkout(batch)
         req = InferRequest(
             model=self.model, shape=list(batch.shape), dtype="f16",
-            input=b"" if shm else batch.tobytes(),
-            shm_name=shm.name if shm else "",
-            shm_size=shm.size if shm else 0)
+            input=b"" if seg is not None else batch.tobytes(),
+            shm_name=name or "", shm_size=size)
         inner = self._m._client.call("trtlab.Inference", "Compute", req,
                                      InferResponse, timeout=self._m.timeout)
         out: Future = Future()
 
         def done(f):
-            if shm:
-                shm.close()
+            if seg is not None:
+                self._pool.release(seg)
             exc = f.exception()
             if exc is not None:
                 out.set_exception(exc)
@@ -48,6 +54,11 @@ class RemoteInferRunner:
 
         inner.add_done_callback(done)
         return out
+
+    def close(self):
+        if self._pool is not None:
+            self._pool.close()
+            self._pool = None
 
 
 class RemoteInferenceManager:

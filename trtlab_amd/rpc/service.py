@@ -41,6 +41,7 @@ class InferenceService(AsyncService):
     def __init__(self, resources: InferenceResources, metrics=None):
         super().__init__("trtlab.Inference", resources)
         self.metrics = metrics
+        self._shm_cache: dict = {}  # name -> SharedMemory (pooled clients)
         self.register_unary("Compute", self._compute, InferRequest,
                             InferResponse)
 
@@ -59,16 +60,21 @@ class InferenceService(AsyncService):
         shape = tuple(request.shape) or plan.input_shape
         if request.shm_name:
             # zero-copy local transport: the tensor lives in POSIX shared
-            # memory (reference's SysV shm input path, 02 server.cc:159)
+            # memory (reference's SysV shm input path, 02 server.cc:159).
+            # Mappings are cached by name: clients pool and reuse segments
+            # (rpc.client.ShmPool), so re-mmapping per request is waste.
             from multiprocessing import shared_memory
 
-            shm = shared_memory.SharedMemory(name=request.shm_name)
-            try:
-                batch = np.frombuffer(
-                    shm.buf[:int(request.shm_size)], dtype=dtype
-                ).reshape(shape).copy()
-            finally:
-                shm.close()
+            shm = self._shm_cache.get(request.shm_name)
+            if shm is None:
+                shm = shared_memory.SharedMemory(name=request.shm_name)
+                if len(self._shm_cache) >= 64:  # bounded
+                    old = self._shm_cache.popitem()[1]
+                    old.close()
+                self._shm_cache[request.shm_name] = shm
+            batch = np.frombuffer(
+                shm.buf[:int(request.shm_size)], dtype=dtype
+            ).reshape(shape).copy()
         else:
             batch = np.frombuffer(request.input, dtype=dtype).reshape(shape)
 
