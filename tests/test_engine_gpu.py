@@ -465,3 +465,22 @@ def test_decode_session_matches_full_model():
     sess2 = DecodeSession(g, batch=2, smax=128, capture=False)
     outs2 = [sess2.step(ids[:, t]) for t in range(4)]
     assert np.allclose(outs[3], outs2[3], atol=1e-3)
+
+
+def test_engine_resnet18():
+    """Basic-block ResNet (18) end-to-end on the fused conv path."""
+    from trtlab_amd.engine.planner import Planner
+    from trtlab_amd.engine.reference import run_reference
+    from trtlab_amd.engine.runtime import NativeEngine
+    from trtlab_amd.models import build_resnet
+
+    g = build_resnet(18, batch=2, image=64, seed=0)
+    plan = Planner().compile(g)
+    ctx = NativeEngine(plan).create_context(capture=True)
+    x = np.random.RandomState(51).randn(*plan.input_shape).astype(
+        np.float32) * 0.5
+    out = ctx.infer(x).astype(np.float32)
+    ref = run_reference(plan, x)
+    err = np.abs(out - ref).max()
+    scale = max(np.abs(ref).max(), 1e-6)
+    assert err / scale < 0.08, (err, scale)

@@ -11,6 +11,8 @@ import numpy as np
 from trtlab_amd.engine.ir import Graph
 
 _LAYERS = {50: (3, 4, 6, 3), 101: (3, 4, 23, 3), 152: (3, 8, 36, 3)}
+# basic-block depths (two 3x3 convs per block, expansion 1)
+_LAYERS_BASIC = {18: (2, 2, 2, 2), 34: (3, 4, 6, 3)}
 
 
 class _Init:
@@ -58,12 +60,33 @@ def _bottleneck(g: Graph, init: _Init, x: str, cin: int, mid: int,
     return g.relu(h)
 
 
+def _basic_block(g: Graph, init: _Init, x: str, cin: int, mid: int,
+                 stride: int) -> str:
+    """ResNet-18/34 basic block: 3x3 -> 3x3 with identity/1x1 shortcut
+    (torchvision BasicBlock). Downsample emitted first so the planner
+    fuses the residual add + relu into the second conv's epilogue."""
+    if stride != 1 or cin != mid:
+        ds = g.conv2d(x, init.conv(mid, cin, 1, 1), stride=stride)
+        ds = g.batchnorm(ds, **init.bn(mid))
+        residual = ds
+    else:
+        residual = x
+    h = g.conv2d(x, init.conv(mid, cin, 3, 3), stride=stride, padding=1)
+    h = g.batchnorm(h, **init.bn(mid))
+    h = g.relu(h)
+    h = g.conv2d(h, init.conv(mid, mid, 3, 3), padding=1)
+    h = g.batchnorm(h, **init.bn(mid))
+    h = g.add(h, residual)
+    return g.relu(h)
+
+
 def build_resnet(depth: int = 50, batch: int = 8, image: int = 224,
                  classes: int = 1000, seed: int = 0, softmax: bool = False,
                  calibrate: bool = True) -> Graph:
-    if depth not in _LAYERS:
+    if depth not in _LAYERS and depth not in _LAYERS_BASIC:
         raise ValueError(f"unsupported resnet depth {depth}")
-    blocks = _LAYERS[depth]
+    basic = depth in _LAYERS_BASIC
+    blocks = _LAYERS_BASIC[depth] if basic else _LAYERS[depth]
     init = _Init(seed)
     g = Graph(f"resnet{depth}_b{batch}")
     x = g.input((batch, image, image, 3))
@@ -76,8 +99,12 @@ def build_resnet(depth: int = 50, batch: int = 8, image: int = 224,
         mid = 64 * (2 ** stage)
         for b in range(nblocks):
             stride = 2 if (stage > 0 and b == 0) else 1
-            h = _bottleneck(g, init, h, cin, mid, stride)
-            cin = mid * 4
+            if basic:
+                h = _basic_block(g, init, h, cin, mid, stride)
+                cin = mid
+            else:
+                h = _bottleneck(g, init, h, cin, mid, stride)
+                cin = mid * 4
     h = g.global_avgpool(h)
     w, bias = init.fc(classes, cin)
     h = g.gemm(h, w, bias)
