@@ -65,16 +65,25 @@ class InferenceService(AsyncService):
             # (rpc.client.ShmPool), so re-mmapping per request is waste.
             from multiprocessing import shared_memory
 
-            shm = self._shm_cache.get(request.shm_name)
-            if shm is None:
-                shm = shared_memory.SharedMemory(name=request.shm_name)
-                if len(self._shm_cache) >= 64:  # bounded
-                    old = self._shm_cache.popitem()[1]
+            def read_shm():
+                shm = self._shm_cache.get(request.shm_name)
+                if shm is None:
+                    shm = shared_memory.SharedMemory(name=request.shm_name)
+                    if len(self._shm_cache) >= 64:  # bounded
+                        self._shm_cache.popitem()[1].close()
+                    self._shm_cache[request.shm_name] = shm
+                return np.frombuffer(
+                    shm.buf[:int(request.shm_size)], dtype=dtype
+                ).reshape(shape).copy()
+
+            try:
+                batch = read_shm()
+            except Exception:
+                # stale cached mapping (client restarted): drop + reopen
+                old = self._shm_cache.pop(request.shm_name, None)
+                if old is not None:
                     old.close()
-                self._shm_cache[request.shm_name] = shm
-            batch = np.frombuffer(
-                shm.buf[:int(request.shm_size)], dtype=dtype
-            ).reshape(shape).copy()
+                batch = read_shm()
         else:
             batch = np.frombuffer(request.input, dtype=dtype).reshape(shape)
 
