@@ -210,3 +210,29 @@ def test_resnet18_basic_block_plan():
         np.float32) * 0.5
     out = run_reference(plan, x)
     assert out.shape == (1, 1000) and np.isfinite(out).all()
+
+
+def test_planner_error_paths():
+    """Malformed graphs fail loudly at compile time, not on the GPU."""
+    from trtlab_amd.engine.planner import Planner
+    from trtlab_amd.models import build_bert
+
+    # gemm K % 64
+    g = Graph("bad")
+    x = g.input((4, 100))
+    with pytest.raises(AssertionError):
+        g.gemm(x, np.zeros((10, 96), np.float32))  # K mismatch vs input
+
+    g2 = Graph("bad2")
+    x2 = g2.input((4, 100))
+    g2.gemm(x2, np.zeros((10, 100), np.float32))
+    with pytest.raises(ValueError, match="must be %64"):
+        Planner().compile(g2)
+
+    # varlen without an ids input
+    g3 = build_bert(batch=1, seq=128, layers=1, seed=0)
+    for n in g3.nodes:
+        if n.kind == "attention":
+            n.attrs["varlen"] = True
+    with pytest.raises(ValueError, match="token-id"):
+        Planner().compile(g3)

@@ -172,3 +172,40 @@ def test_remote_inference_manager(service_server):
     out2 = m.infer_runner("m", use_shm=True).infer(x).result(timeout=10)
     assert np.allclose(out2, x * 2)
     m.close()
+
+
+def test_unknown_model_is_an_rpc_error():
+    """Requests for unregistered models surface as client-side errors, not
+    server crashes; the service keeps serving afterwards."""
+    class _StrictManager(_FakeManager):
+        def infer_runner(self, name):
+            if name != "m":
+                raise KeyError(name)
+            return _FakeRunner()
+
+        def get_model(self, name):
+            if name != "m":
+                raise KeyError(name)
+            return _FakeEngine()
+
+    server = Server("127.0.0.1:0")
+    svc = InferenceService(InferenceResources(_StrictManager()))
+    server.register_service(svc)
+    server.register_service(svc.health_service)
+    server.async_start()
+    try:
+        c = SyncClient(f"127.0.0.1:{server.port}")
+        bad = InferRequest(model="nope", shape=[2, 4], dtype="f16",
+                           input=np.zeros(8, np.float16).tobytes())
+        with pytest.raises(Exception):
+            c.call("trtlab.Inference", "Compute", bad, InferResponse,
+                   timeout=5)
+        # still alive after the error
+        x = np.arange(8, dtype=np.float16).reshape(2, 4)
+        good = InferRequest(model="m", input=x.tobytes(), shape=[2, 4],
+                            dtype="f16")
+        r = c.call("trtlab.Inference", "Compute", good, InferResponse,
+                   timeout=10)
+        assert list(r.shape) == [2, 4]
+    finally:
+        server.shutdown()
