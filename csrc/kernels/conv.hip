@@ -264,7 +264,10 @@ static void launch_conv2d_t(const void* in, const void* Wt, void* out,
     return;
   }
   }
-  TileCfg cfg = tile ? tile_from_code(tile) : pick_tile(p.M, p.Cout);
+  // code 5 (256-wide) is a GEMM-only tactic: LDS would overflow the deep
+  // conv pipeline, so fall back to the heuristic
+  TileCfg cfg = (tile && tile != 5) ? tile_from_code(tile)
+                                    : pick_tile(p.M, p.Cout);
   int tiles_m = (int)cdiv(p.M, cfg.bm);
   int tiles_n = (int)cdiv(p.Cout, cfg.bn);
   long tiles = (long)tiles_m * tiles_n;

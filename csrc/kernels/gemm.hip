@@ -137,6 +137,20 @@ static void launch_gemm_bt_t(const void* A, const void* B, void* C,
   int tiles_n = (int)cdiv(N, cfg.bn);
   long tiles = (long)tiles_m * tiles_n;
   int ktiles = K / kTileElems<T>;
+  if (cfg.bm == 256) {
+    // 256x128 tactic (code 5): 2-phase only (4-buffer LDS would overflow),
+    // no split-K; picked by autotune on staging-bound large-M shapes.
+    dim3 grid5((unsigned)tiles);
+    epi_dispatch(epi, [&](auto e) {
+      constexpr Epi EE = decltype(e)::value;
+      hipLaunchKernelGGL((gemm_bt_kernel<T, OT, EE, 256, 128, false, 2>),
+                         grid5, dim3(256), 0, stream, (const T*)A,
+                         (const T*)B, (OT*)C, scale, bias,
+                         (const OT*)residual, res_scale, out_scale, M, N, K,
+                         lda, ldb, ldc, tiles_n, (float*)nullptr, 1, ktiles);
+    });
+    return;
+  }
   int splitk = (!tile && scratch) ? pick_splitk_gemm(tiles, ktiles) : 1;
   dim3 block(256);
   int out_dtype = std::is_same<OT, _Float16>::value

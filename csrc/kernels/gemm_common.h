@@ -332,12 +332,15 @@ inline TileCfg pick_tile(int M, int N) {
   return best;
 }
 
-// Explicit tile override codes: 1=(128,128) 2=(128,64) 3=(64,128) 4=(64,64).
+// Explicit tile override codes: 1=(128,128) 2=(128,64) 3=(64,128)
+// 4=(64,64) 5=(256,128) (GEMM-only: halves staged bytes per FLOP for the
+// staging-bandwidth-bound large-M transformer shapes; conv clamps it).
 inline TileCfg tile_from_code(int code) {
   switch (code) {
     case 1: return {128, 128};
     case 2: return {128, 64};
     case 3: return {64, 128};
+    case 5: return {256, 128};
     default: return {64, 64};
   }
 }

@@ -437,3 +437,15 @@ def test_attention_long_and_causal(C, S, causal):
     ref = (att @ q[:, :, 2].permute(0, 2, 1, 3)).permute(
         0, 2, 1, 3).reshape(B * S, hid)
     check(out, ref)
+
+
+def test_gemm_bt_tile256(C):
+    """The 256x128 GEMM tactic (code 5) vs torch on a transformer shape."""
+    M, N, K = 2048, 768, 768
+    a = t16(M, K, seed=61)
+    b = t16(N, K, seed=62)
+    out = torch.empty(M, N, dtype=torch.half, device="cuda")
+    torch.cuda.synchronize()
+    C.ops.gemm_bt(0, a.data_ptr(), b.data_ptr(), out.data_ptr(),
+                  M=M, N=N, K=K, tile=5)
+    check(out, (a.float() @ b.float().t()).half())

@@ -14,7 +14,8 @@ from typing import Dict, Tuple
 
 from trtlab_amd.engine.planner import EnginePlan, K_CONV, K_GEMM
 
-_CANDIDATES = (0, 1, 2, 3, 4)
+_CANDIDATES = (0, 1, 2, 3, 4)          # conv: heuristic + 4 fixed tiles
+_CANDIDATES_GEMM = (0, 1, 2, 3, 4, 5)  # + 256x128 (staging-bound shapes)
 
 
 def _conv_key(d: dict) -> Tuple:
@@ -93,7 +94,9 @@ def autotune_plan(plan: EnginePlan, device: int = 0, reps: int = 30,
                                   sync=sync, tile=tile)
 
             best, best_us = 0, float("inf")
-            for tile in _CANDIDATES:
+            cands = (_CANDIDATES if d["kind"] == K_CONV
+                     else _CANDIDATES_GEMM)
+            for tile in cands:
                 for _ in range(warmup):
                     run(tile)
                 C.hip.device_synchronize()
