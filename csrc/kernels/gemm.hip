@@ -92,6 +92,21 @@ __global__ __launch_bounds__(256) void gemm_bt_kernel(
       const char* As = &smem[((t - kt0) & 3) * kBuf];
       mfma_tile<T, BM, BN>(As, As + kABytes, lane, wr, wc, acc);
     }
+  } else if constexpr (NBUF == 3) {
+    // 2-deep counted pipeline (the 256-wide tile: 4 slots would overflow
+    // LDS; occupancy is 1, so in-block overlap does the latency hiding)
+    constexpr int G = BM / 32 + BN / 32;
+    stage(kt0, 0);
+    if (kt0 + 1 < kt1) stage(kt0 + 1, 1);
+    for (int t = kt0; t < kt1; ++t) {
+      int ahead = kt1 - 1 - t;
+      if (ahead > 1) ahead = 1;
+      wait_tiles_inflight<G>(ahead);
+      __builtin_amdgcn_s_barrier();
+      if (t + 2 < kt1) stage(t + 2, (t + 2 - kt0) % 3);
+      const char* As = &smem[((t - kt0) % 3) * kBuf];
+      mfma_tile<T, BM, BN>(As, As + kABytes, lane, wr, wc, acc);
+    }
   } else {
     stage(kt0, 0);
     asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
@@ -138,12 +153,12 @@ static void launch_gemm_bt_t(const void* A, const void* B, void* C,
   long tiles = (long)tiles_m * tiles_n;
   int ktiles = K / kTileElems<T>;
   if (cfg.bm == 256) {
-    // 256x128 tactic (code 5): 2-phase only (4-buffer LDS would overflow),
+    // 256x128 tactic (code 5): 3-slot counted pipeline (144 KiB LDS),
     // no split-K; picked by autotune on staging-bound large-M shapes.
     dim3 grid5((unsigned)tiles);
     epi_dispatch(epi, [&](auto e) {
       constexpr Epi EE = decltype(e)::value;
-      hipLaunchKernelGGL((gemm_bt_kernel<T, OT, EE, 256, 128, false, 2>),
+      hipLaunchKernelGGL((gemm_bt_kernel<T, OT, EE, 256, 128, false, 3>),
                          grid5, dim3(256), 0, stream, (const T*)A,
                          (const T*)B, (OT*)C, scale, bias,
                          (const OT*)residual, res_scale, out_scale, M, N, K,

@@ -241,6 +241,8 @@ __device__ __forceinline__ void wait_tiles_inflight(int ahead) {
       asm volatile("s_waitcnt vmcnt(18)" ::: "memory");
     else if constexpr (G == 10)
       asm volatile("s_waitcnt vmcnt(20)" ::: "memory");
+    else if constexpr (G == 12)
+      asm volatile("s_waitcnt vmcnt(24)" ::: "memory");
     else
       asm volatile("s_waitcnt vmcnt(16)" ::: "memory");
   } else if (ahead == 1) {
@@ -252,6 +254,8 @@ __device__ __forceinline__ void wait_tiles_inflight(int ahead) {
       asm volatile("s_waitcnt vmcnt(9)" ::: "memory");
     else if constexpr (G == 10)
       asm volatile("s_waitcnt vmcnt(10)" ::: "memory");
+    else if constexpr (G == 12)
+      asm volatile("s_waitcnt vmcnt(12)" ::: "memory");
     else
       asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
   } else {
