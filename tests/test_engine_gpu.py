@@ -484,3 +484,27 @@ def test_engine_resnet18():
     err = np.abs(out - ref).max()
     scale = max(np.abs(ref).max(), 1e-6)
     assert err / scale < 0.08, (err, scale)
+
+
+def test_engine_gpt2_fp8():
+    """GPT-2 with fp8 gemms (producer-fused quantization) + causal
+    attention: quality gate vs the fp16 reference."""
+    from trtlab_amd.engine.planner import DT_F8, Planner
+    from trtlab_amd.engine.reference import run_reference
+    from trtlab_amd.engine.runtime import NativeEngine
+    from trtlab_amd.models import build_gpt2
+
+    g = build_gpt2(batch=2, seq=128, layers=2, seed=0)
+    plan = Planner(dtype=DT_F8).compile(g)
+    ctx = NativeEngine(plan).create_context(capture=True)
+    x = np.random.RandomState(61).randn(*plan.input_shape).astype(
+        np.float32) * 0.5
+    out = ctx.infer(x).astype(np.float32)
+    assert np.isfinite(out).all()
+    ref16 = run_reference(Planner().compile(
+        build_gpt2(batch=2, seq=128, layers=2, seed=0)), x)
+    ref8 = run_reference(plan, x)
+    corr_gpu = np.corrcoef(out.ravel(), ref16.ravel())[0, 1]
+    corr_emu = np.corrcoef(ref8.ravel(), ref16.ravel())[0, 1]
+    assert corr_gpu > corr_emu - 0.03, (corr_gpu, corr_emu)
+    assert corr_gpu > 0.95, corr_gpu
