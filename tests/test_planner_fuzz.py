@@ -146,3 +146,123 @@ def test_fuzz_random_graph_gpu(seed):
     scale = max(np.abs(ref).max(), 1e-3)
     err = np.abs(out - ref).max() / scale
     assert err < 0.05, (seed, err)
+
+
+def _random_transformer(rng: np.random.RandomState) -> Graph:
+    """Random BERT/GPT-style stacks: pre- or post-LN, optional gelu,
+    random layer counts — exercises gemm/attention/LN fusion paths."""
+    g = Graph("tfuzz")
+    b = int(rng.choice([1, 2]))
+    seq = 128
+    hidden = int(rng.choice([128, 256]))
+    heads = hidden // 64
+    m = b * seq
+    h = g.input((m, hidden))
+    pre_ln = bool(rng.rand() < 0.5)
+    causal = bool(rng.rand() < 0.5)
+
+    def w(no, ni):
+        return (rng.randn(no, ni) * np.sqrt(1.0 / ni)).astype(np.float32)
+
+    def ln_p():
+        return (rng.uniform(0.9, 1.1, hidden).astype(np.float32),
+                (rng.randn(hidden) * 0.05).astype(np.float32))
+
+    for li in range(int(rng.randint(1, 3))):
+        if pre_ln:
+            ga, be = ln_p()
+            x = g.layernorm(h, ga, be)
+        else:
+            x = h
+        qkv = g.gemm(x, w(3 * hidden, hidden),
+                     (rng.randn(3 * hidden) * 0.02).astype(np.float32))
+        att = g.attention(qkv, heads=heads, seq=seq, causal=causal)
+        proj = g.gemm(att, w(hidden, hidden))
+        if pre_ln:
+            h = g.add(h, proj)
+        else:
+            ga, be = ln_p()
+            h = g.add_layernorm(proj, h, ga, be)
+        inter = hidden * int(rng.choice([2, 4]))
+        ff = g.gemm(h, w(inter, hidden),
+                    (rng.randn(inter) * 0.02).astype(np.float32))
+        if rng.rand() < 0.7:
+            ff = g.gelu(ff)
+        h = g.gemm(ff, w(hidden, inter))
+    return g
+
+
+@pytest.mark.parametrize("seed", range(8))
+def test_fuzz_random_transformer(seed):
+    """Random transformer stacks: fused plan vs torch node interpretation."""
+    rng = np.random.RandomState(2000 + seed)
+    g = _random_transformer(rng)
+    plan = Planner().compile(g)
+    x = (rng.randn(*plan.input_shape) * 0.5).astype(np.float32)
+    fused = run_reference(plan, x)
+
+    t = {g.input_name: torch.from_numpy(x).float()}
+    for n in g.nodes:
+        if n.kind == "input":
+            continue
+        xx = t[n.inputs[0]]
+        if n.kind == "gemm":
+            y = xx @ torch.from_numpy(n.attrs["weight"]).t()
+            if n.attrs.get("bias") is not None:
+                y = y + torch.from_numpy(n.attrs["bias"])
+        elif n.kind == "layernorm":
+            a = n.attrs
+            mu = xx.mean(-1, keepdim=True)
+            var = xx.var(-1, unbiased=False, keepdim=True)
+            y = (xx - mu) / torch.sqrt(var + a["eps"])
+            y = y * torch.from_numpy(a["gamma"]) + torch.from_numpy(a["beta"])
+        elif n.kind == "add_layernorm":
+            a = n.attrs
+            s = xx + t[n.inputs[1]]
+            mu = s.mean(-1, keepdim=True)
+            var = s.var(unbiased=False, dim=-1, keepdim=True)
+            y = (s - mu) / torch.sqrt(var + a["eps"])
+            y = y * torch.from_numpy(a["gamma"]) + torch.from_numpy(a["beta"])
+        elif n.kind == "add":
+            y = xx + t[n.inputs[1]]
+        elif n.kind == "gelu":
+            y = F.gelu(xx, approximate="tanh")
+        elif n.kind == "attention":
+            a = n.attrs
+            bsz = xx.shape[0] // a["seq"]
+            s_, nh, hd = a["seq"], a["heads"], a["head_dim"]
+            qkv = xx.reshape(bsz, s_, 3, nh, hd)
+            q = qkv[:, :, 0].permute(0, 2, 1, 3)
+            k = qkv[:, :, 1].permute(0, 2, 1, 3)
+            v = qkv[:, :, 2].permute(0, 2, 1, 3)
+            sc = q @ k.transpose(-1, -2) / np.sqrt(hd)
+            if a.get("causal"):
+                cm = torch.arange(s_)[None, :] > torch.arange(s_)[:, None]
+                sc = sc.masked_fill(cm[None, None], float("-inf"))
+            y = (torch.softmax(sc, -1) @ v).permute(0, 2, 1, 3).reshape(
+                bsz * s_, nh * hd)
+        else:
+            raise AssertionError(n.kind)
+        t[n.output] = y
+    unfused = t[g.output_name].numpy()
+    scale = max(np.abs(unfused).max(), 1e-3)
+    err = np.abs(fused - unfused).max() / scale
+    assert err < 0.03, (seed, err)
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("seed", range(4))
+def test_fuzz_random_transformer_gpu(seed):
+    """Same fuzz transformers through the native engine (graph-captured)."""
+    from trtlab_amd.engine.runtime import NativeEngine
+
+    rng = np.random.RandomState(2000 + seed)
+    g = _random_transformer(rng)
+    plan = Planner().compile(g)
+    ctx = NativeEngine(plan).create_context(capture=True)
+    x = (rng.randn(*plan.input_shape) * 0.5).astype(np.float32)
+    out = ctx.infer(x).astype(np.float32)
+    ref = run_reference(plan, x)
+    scale = max(np.abs(ref).max(), 1e-3)
+    err = np.abs(out - ref).max() / scale
+    assert err < 0.05, (seed, err)
