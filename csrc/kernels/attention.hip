@@ -92,34 +92,48 @@ __global__ __launch_bounds__(256) void attention_kernel(
 #pragma unroll
   for (int j = 0; j < 4; ++j) oacc[j] = {0.f, 0.f, 0.f, 0.f};
 
-  for (int t = 0; t < ntiles; ++t) {
-    // previous tile's PV reads must finish before K/Vt are overwritten
-    __syncthreads();
-    // ---- stage K tile (128 rows) + Vt (transposed scatter) ----
+  // K/V staging is register-double-buffered: tile t+1's global loads are
+  // issued while tile t computes (their ~1-2 us latency was serialized
+  // with compute before), and only the cheap ds_writes sit between the
+  // barriers. LDS footprint unchanged (occupancy stays 2/SIMD).
+  short8v kv_k[4], kv_v[4];
+  auto load_kv = [&](int t) {
 #pragma unroll
     for (int c = 0; c < 4; ++c) {
       int idx = c * 256 + tid;
       int row = idx >> 3;            // 0..127
       int cb = (idx & 7) * 16;
-      const T* ksrc =
-          base + (int64_t)(t * 128 + row) * row_stride + koff + cb / 2;
-      *(short8v*)(Ks + swz(row, cb)) = *(const short8v*)ksrc;
-    }
-#pragma unroll
-    for (int c = 0; c < 4; ++c) {
+      kv_k[c] = *(const short8v*)(
+          base + (int64_t)(t * 128 + row) * row_stride + koff + cb / 2);
       int ci = c * 256 + tid;        // 0..1023
       int key = ci >> 3;
       int dd = (ci & 7) * 8;
-      const T* vsrc =
-          base + (int64_t)(t * 128 + key) * row_stride + voff + dd;
-      short8v v = *(const short8v*)vsrc;
+      kv_v[c] = *(const short8v*)(
+          base + (int64_t)(t * 128 + key) * row_stride + voff + dd);
+    }
+  };
+  if (ntiles > 0) load_kv(0);
+  for (int t = 0; t < ntiles; ++t) {
+    // previous tile's PV reads must finish before K/Vt are overwritten
+    __syncthreads();
+    // ---- write the prefetched K tile (128 rows) + Vt (transposed) ----
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+      int idx = c * 256 + tid;
+      int row = idx >> 3;
+      int cb = (idx & 7) * 16;
+      *(short8v*)(Ks + swz(row, cb)) = kv_k[c];
+      int ci = c * 256 + tid;
+      int key = ci >> 3;
+      int dd = (ci & 7) * 8;
       char* tile = Vt + (key >> 6) * 8192;
       int kcol = key & 63;
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        *(T*)(tile + swz(dd + j, kcol * 2)) = ((const T*)&v)[j];
+        *(T*)(tile + swz(dd + j, kcol * 2)) = ((const T*)&kv_v[c])[j];
       }
     }
+    if (t + 1 < ntiles) load_kv(t + 1);  // overlaps this tile's compute
     __syncthreads();
 
     // ---- QK^T for this wave's 16 query rows ----
