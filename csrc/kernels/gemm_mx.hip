@@ -472,6 +472,121 @@ __global__ __launch_bounds__(256) void gemm_mxfp4_kernel(
     }
 }
 
+
+// 64x64 small-tile MXFP4 variant (grid-starved shapes; cf. the fp8 one).
+template <typename OT, Epi E>
+__global__ __launch_bounds__(256) void gemm_mxfp4_small_kernel(
+    const uint8_t* __restrict__ A, const uint8_t* __restrict__ B,
+    const uint8_t* __restrict__ Sa, const uint8_t* __restrict__ Sb,
+    OT* __restrict__ C, const float* __restrict__ scale,
+    const float* __restrict__ bias, int M, int N, int K, int tiles_n) {
+  constexpr int kABytes = 64 * 128;  // 64 rows x 256 logical k (packed)
+  __shared__ __attribute__((aligned(16))) char smem[2 * 2 * kABytes];
+  __shared__ __attribute__((aligned(4))) uint32_t sa_s[2][128], sb_s[2][128];
+
+  uint32_t bid = xcd_swizzle(blockIdx.x, gridDim.x);
+  int m0 = (int)(bid / tiles_n) * 64;
+  int n0 = (int)(bid % tiles_n) * 64;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int kb2 = K >> 1;
+  const int ktiles = K >> 8;  // 256 logical elems per LDS tile
+  const int kblocks = K >> 5;
+
+  f32x4 acc[4];
+#pragma unroll
+  for (int f = 0; f < 4; ++f) acc[f] = {0.f, 0.f, 0.f, 0.f};
+
+  auto stage = [&](int t, int slot) {
+    char* base = smem + slot * 2 * kABytes;
+#pragma unroll
+    for (int c = 0; c < 2; ++c) {
+      int idx = c * 256 + tid;
+      uint32_t row = (uint32_t)idx >> 3;
+      uint32_t cb = ((uint32_t)idx & 7) * 16;
+      uint32_t kb = cb ^ ((row & 7) << 4);
+      int ar = m0 + (int)row;
+      if (ar >= M) ar = M - 1;
+      glds16((const char*)(A + (int64_t)ar * kb2 + t * 128 + kb),
+             (uint32_t)(uintptr_t)base + row * 128 + cb);
+      int br = n0 + (int)row;
+      if (br >= N) br = N - 1;
+      glds16((const char*)(B + (int64_t)br * kb2 + t * 128 + kb),
+             (uint32_t)(uintptr_t)(base + kABytes) + row * 128 + cb);
+    }
+    // 8 scale bytes per row per 256-k tile: two 64-u32 slabs per side
+    if (tid < 64) {
+      int ar = m0 + tid;
+      if (ar >= M) ar = M - 1;
+      sa_s[slot][tid] = *(const uint32_t*)(Sa + (int64_t)ar * kblocks + t * 8);
+      sa_s[slot][64 + tid] =
+          *(const uint32_t*)(Sa + (int64_t)ar * kblocks + t * 8 + 4);
+    } else if (tid < 128) {
+      int br = n0 + tid - 64;
+      if (br >= N) br = N - 1;
+      sb_s[slot][tid - 64] =
+          *(const uint32_t*)(Sb + (int64_t)br * kblocks + t * 8);
+      sb_s[slot][tid] =
+          *(const uint32_t*)(Sb + (int64_t)br * kblocks + t * 8 + 4);
+    }
+  };
+
+  auto frag16 = [&](const char* lds, int row, int w) {
+    uint32_t cb = (uint32_t)w * 64 + (uint32_t)(lane >> 4) * 16;
+    i32x4 v = *(const i32x4*)(
+        lds + (uint32_t)row * 128 + (cb ^ (((uint32_t)row & 7) << 4)));
+    i32x8v f = {v[0], v[1], v[2], v[3], 0, 0, 0, 0};
+    return f;
+  };
+
+  stage(0, 0);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+  int cur = 0;
+  for (int t = 0; t < ktiles; ++t) {
+    if (t + 1 < ktiles) stage(t + 1, cur ^ 1);
+    const char* As = smem + cur * 2 * kABytes;
+    const char* Bs = As + kABytes;
+    int g = lane >> 4;
+#pragma unroll
+    for (int w = 0; w < 2; ++w) {
+      int arow = wave * 16 + (lane & 15);
+      i32x8v af = frag16(As, arow, w);
+      int sa = ((const uint8_t*)&sa_s[cur][w * 64 + arow])[g];
+#pragma unroll
+      for (int f = 0; f < 4; ++f) {
+        int brow = f * 16 + (lane & 15);
+        i32x8v bf = frag16(Bs, brow, w);
+        int sb = ((const uint8_t*)&sb_s[cur][w * 64 + brow])[g];
+        acc[f] = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
+            af, bf, acc[f], 4, 4, 0, sa, 0, sb);
+      }
+    }
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+    cur ^= 1;
+  }
+
+#pragma unroll
+  for (int f = 0; f < 4; ++f) {
+    int col = n0 + f * 16 + (lane & 15);
+    if (col >= N) continue;
+    float sc = 1.0f, bi = 0.0f;
+    if constexpr (E == Epi::kScaleBias || E == Epi::kScaleBiasRelu ||
+                  E == Epi::kScaleBiasAddRelu || E == Epi::kScaleBiasGelu)
+      sc = scale[col];
+    if constexpr (E != Epi::kNone) bi = bias[col];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int row = m0 + wave * 16 + ((lane >> 4) << 2) + r;
+      if (row >= M) continue;
+      C[(int64_t)row * N + col] =
+          store_cast<OT>(apply_epi<E>(acc[f][r], sc, bi, 0.0f));
+    }
+  }
+}
+
 // out_dtype: 0 = fp16 (engine path, with epilogue), 2 = fp32 raw C.
 void launch_gemm_mxfp4(const void* A, const void* B, const void* Sa,
                        const void* Sb, void* C, int M, int N, int K,
@@ -481,6 +596,28 @@ void launch_gemm_mxfp4(const void* A, const void* B, const void* Sa,
     throw std::runtime_error("gemm_mxfp4: K must be a multiple of 256");
   int tiles_n = (int)cdiv(N, 128);
   dim3 grid((unsigned)(cdiv(M, 128) * tiles_n));
+  if ((long)cdiv(M, 128) * tiles_n < 64) {
+    // grid-starved: 4x the workgroups at 64x64
+    int tn64 = (int)cdiv(N, 64);
+    dim3 g64((unsigned)(cdiv(M, 64) * tn64));
+    if (out_dtype == 2) {
+      hipLaunchKernelGGL((gemm_mxfp4_small_kernel<float, Epi::kNone>), g64,
+                         dim3(256), 0, stream, (const uint8_t*)A,
+                         (const uint8_t*)B, (const uint8_t*)Sa,
+                         (const uint8_t*)Sb, (float*)C, nullptr, nullptr, M,
+                         N, K, tn64);
+      return;
+    }
+    epi_dispatch(epi, [&](auto e) {
+      constexpr Epi EE = decltype(e)::value;
+      hipLaunchKernelGGL((gemm_mxfp4_small_kernel<_Float16, EE>), g64,
+                         dim3(256), 0, stream, (const uint8_t*)A,
+                         (const uint8_t*)B, (const uint8_t*)Sa,
+                         (const uint8_t*)Sb, (_Float16*)C, scale, bias, M, N,
+                         K, tn64);
+    });
+    return;
+  }
   if (out_dtype == 2) {
     hipLaunchKernelGGL((gemm_mxfp4_kernel<float, Epi::kNone>), grid,
                        dim3(256), 0, stream, (const uint8_t*)A,
@@ -540,6 +677,105 @@ void launch_mx_probe(const void* A, const void* B, const void* Sa,
                      (const uint8_t*)Sb, (float*)D);
 }
 
+
+// 64x64 small-tile MXFP8 variant for grid-starved serving shapes (e.g.
+// BERT b8: M=1024 gives the 128-tile kernel only 48 workgroups on 256
+// CUs). 4 waves x [16 rows x 64 cols]; 2-phase staging (latency-bound
+// shapes; the deep pipeline is the 128-tile's job).
+template <typename OT, Epi E>
+__global__ __launch_bounds__(256) void gemm_mxfp8_small_kernel(
+    const uint8_t* __restrict__ A, const uint8_t* __restrict__ B,
+    const uint8_t* __restrict__ Sa, const uint8_t* __restrict__ Sb,
+    OT* __restrict__ C, const float* __restrict__ scale,
+    const float* __restrict__ bias, int M, int N, int K, int tiles_n) {
+  constexpr int kABytes = 64 * 128;
+  __shared__ __attribute__((aligned(16))) char smem[2 * 2 * kABytes];
+  __shared__ __attribute__((aligned(4))) uint32_t sa_s[2][64], sb_s[2][64];
+
+  uint32_t bid = xcd_swizzle(blockIdx.x, gridDim.x);
+  int m0 = (int)(bid / tiles_n) * 64;
+  int n0 = (int)(bid % tiles_n) * 64;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int ktiles = K >> 7;
+  const int kblocks = K >> 5;
+
+  f32x4 acc[4];
+#pragma unroll
+  for (int f = 0; f < 4; ++f) acc[f] = {0.f, 0.f, 0.f, 0.f};
+
+  auto stage = [&](int t, int slot) {
+    char* base = smem + slot * 2 * kABytes;
+#pragma unroll
+    for (int c = 0; c < 2; ++c) {
+      int idx = c * 256 + tid;  // 0..511
+      uint32_t row = (uint32_t)idx >> 3;
+      uint32_t cb = ((uint32_t)idx & 7) * 16;
+      uint32_t kb = cb ^ ((row & 7) << 4);
+      int ar = m0 + (int)row;
+      if (ar >= M) ar = M - 1;
+      glds16((const char*)(A + (int64_t)ar * K + t * 128 + kb),
+             (uint32_t)(uintptr_t)base + row * 128 + cb);
+      int br = n0 + (int)row;
+      if (br >= N) br = N - 1;
+      glds16((const char*)(B + (int64_t)br * K + t * 128 + kb),
+             (uint32_t)(uintptr_t)(base + kABytes) + row * 128 + cb);
+    }
+    if (tid < 64) {
+      int ar = m0 + tid;
+      if (ar >= M) ar = M - 1;
+      sa_s[slot][tid] = *(const uint32_t*)(Sa + (int64_t)ar * kblocks + t * 4);
+    } else if (tid < 128) {
+      int br = n0 + tid - 64;
+      if (br >= N) br = N - 1;
+      sb_s[slot][tid - 64] =
+          *(const uint32_t*)(Sb + (int64_t)br * kblocks + t * 4);
+    }
+  };
+
+  stage(0, 0);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+  int cur = 0;
+  for (int t = 0; t < ktiles; ++t) {
+    if (t + 1 < ktiles) stage(t + 1, cur ^ 1);
+    const char* As = smem + cur * 2 * kABytes;
+    const char* Bs = As + kABytes;
+    int g = lane >> 4;
+    i32x8v af = read_frag32(As, wave * 16 + (lane & 15), lane);
+    int sa = ((const uint8_t*)&sa_s[cur][wave * 16 + (lane & 15)])[g];
+#pragma unroll
+    for (int f = 0; f < 4; ++f) {
+      i32x8v bf = read_frag32(Bs, f * 16 + (lane & 15), lane);
+      int sb = ((const uint8_t*)&sb_s[cur][f * 16 + (lane & 15)])[g];
+      acc[f] = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
+          af, bf, acc[f], 0, 0, 0, sa, 0, sb);
+    }
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+    cur ^= 1;
+  }
+
+#pragma unroll
+  for (int f = 0; f < 4; ++f) {
+    int col = n0 + f * 16 + (lane & 15);
+    if (col >= N) continue;
+    float sc = 1.0f, bi = 0.0f;
+    if constexpr (E == Epi::kScaleBias || E == Epi::kScaleBiasRelu ||
+                  E == Epi::kScaleBiasAddRelu || E == Epi::kScaleBiasGelu)
+      sc = scale[col];
+    if constexpr (E != Epi::kNone) bi = bias[col];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int row = m0 + wave * 16 + ((lane >> 4) << 2) + r;
+      if (row >= M) continue;
+      C[(int64_t)row * N + col] =
+          store_cast<OT>(apply_epi<E>(acc[f][r], sc, bi, 0.0f));
+    }
+  }
+}
+
 // out_dtype: 0 = fp16 (engine path, with epilogue), 2 = fp32 raw C.
 void launch_gemm_mxfp8(const void* A, const void* B, const void* Sa,
                        const void* Sb, void* C, int M, int N, int K,
@@ -549,6 +785,28 @@ void launch_gemm_mxfp8(const void* A, const void* B, const void* Sa,
     throw std::runtime_error("gemm_mxfp8: K must be a multiple of 128");
   int tiles_n = (int)cdiv(N, 128);
   dim3 grid((unsigned)(cdiv(M, 128) * tiles_n));
+  if ((long)cdiv(M, 128) * tiles_n < 64) {
+    // grid-starved: 4x the workgroups at 64x64
+    int tn64 = (int)cdiv(N, 64);
+    dim3 g64((unsigned)(cdiv(M, 64) * tn64));
+    if (out_dtype == 2) {
+      hipLaunchKernelGGL((gemm_mxfp8_small_kernel<float, Epi::kNone>), g64,
+                         dim3(256), 0, stream, (const uint8_t*)A,
+                         (const uint8_t*)B, (const uint8_t*)Sa,
+                         (const uint8_t*)Sb, (float*)C, nullptr, nullptr, M,
+                         N, K, tn64);
+      return;
+    }
+    epi_dispatch(epi, [&](auto e) {
+      constexpr Epi EE = decltype(e)::value;
+      hipLaunchKernelGGL((gemm_mxfp8_small_kernel<_Float16, EE>), g64,
+                         dim3(256), 0, stream, (const uint8_t*)A,
+                         (const uint8_t*)B, (const uint8_t*)Sa,
+                         (const uint8_t*)Sb, (_Float16*)C, scale, bias, M, N,
+                         K, tn64);
+    });
+    return;
+  }
   if (out_dtype == 2) {
     hipLaunchKernelGGL((gemm_mxfp8_kernel<float, Epi::kNone>), grid,
                        dim3(256), 0, stream, (const uint8_t*)A,

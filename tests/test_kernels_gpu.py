@@ -357,7 +357,8 @@ def test_avgpool2d(C):
 
 
 @pytest.mark.parametrize("M,N,K", [(128, 128, 128), (256, 384, 256),
-                                   (1000, 768, 768), (100, 64, 1024)])
+                                   (1000, 768, 768), (100, 64, 1024),
+                                   (2048, 2048, 512)])  # last: 128-tile path
 def test_gemm_mxfp8(C, M, N, K):
     """MXFP8 GEMM (scaled MFMA 16x16x128, e8m0 block scales) vs the fp32
     matmul of the dequantized operands (exact oracle: the kernel's math is
@@ -389,7 +390,8 @@ def test_gemm_mxfp8(C, M, N, K):
 
 
 @pytest.mark.parametrize("M,N,K", [(128, 128, 256), (256, 384, 512),
-                                   (1000, 768, 768)])
+                                   (1000, 768, 768),
+                                   (2048, 2048, 512)])  # last: 128-tile path
 def test_gemm_mxfp4(C, M, N, K):
     """MXFP4 GEMM (scaled MFMA, cbsz=4) vs the dequantized-fp32 oracle."""
     from trtlab_amd.engine.mx import dequantize_mxfp4, quantize_mxfp4
