@@ -393,6 +393,16 @@ PYBIND11_MODULE(_C, m) {
           py::arg("qkv"), py::arg("kcache"), py::arg("vcache"),
           py::arg("pos"), py::arg("B"), py::arg("H"), py::arg("smax"),
           py::arg("stream") = 0, py::arg("sync") = true);
+  ops.def("kv_append_range",
+          [](uintptr_t qkv, uintptr_t kc, uintptr_t vc, int B, int H, int P,
+             int smax, uintptr_t stream, bool sync) {
+            launch_kv_append_range((void*)qkv, (void*)kc, (void*)vc, B, H, P,
+                                   smax, as_stream(stream));
+            if (sync) TRT_HIP_CHECK(hipStreamSynchronize(as_stream(stream)));
+          },
+          py::arg("qkv"), py::arg("kcache"), py::arg("vcache"), py::arg("B"),
+          py::arg("H"), py::arg("P"), py::arg("smax"), py::arg("stream") = 0,
+          py::arg("sync") = true);
   ops.def("decode_attention",
           [](uintptr_t qkv, uintptr_t kc, uintptr_t vc, uintptr_t out,
              uintptr_t pos, int B, int H, int smax, float scale,

@@ -37,6 +37,32 @@ void launch_kv_append(const void* qkv, void* kcache, void* vcache,
                      (_Float16*)vcache, (const int*)pos, B, H, smax);
 }
 
+// Batch-scatter a full prompt's K/V head rows (qkv [B*P, 3*H*64], rows
+// ordered b-major then position) into the caches at positions [0, P).
+// Grid B*H*P; fused prefill runs the prompt through the full-sequence
+// kernels once, then decode continues from position P.
+__global__ __launch_bounds__(64) void kv_append_range_kernel(
+    const _Float16* __restrict__ qkv, _Float16* __restrict__ kcache,
+    _Float16* __restrict__ vcache, int B, int H, int P, int smax) {
+  int p = blockIdx.x % P;
+  int bh = blockIdx.x / P;
+  int b = bh / H, h = bh % H;
+  int d = threadIdx.x;
+  int hid = H * 64;
+  int64_t src = ((int64_t)b * P + p) * 3 * hid + h * 64 + d;
+  int64_t dst = (((int64_t)b * H + h) * smax + p) * 64 + d;
+  kcache[dst] = qkv[src + hid];
+  vcache[dst] = qkv[src + 2 * hid];
+}
+
+void launch_kv_append_range(const void* qkv, void* kcache, void* vcache,
+                            int B, int H, int P, int smax,
+                            hipStream_t stream) {
+  hipLaunchKernelGGL(kv_append_range_kernel, dim3(B * H * P), dim3(64), 0,
+                     stream, (const _Float16*)qkv, (_Float16*)kcache,
+                     (_Float16*)vcache, B, H, P, smax);
+}
+
 // Single-query attention against the cache: out[b, h*64+d] =
 // softmax(q . K[0..pos]) @ V[0..pos]. One wave per (b, h); each lane owns
 // keys lane, lane+64, ... for the score pass (its K rows are whole

@@ -92,6 +92,9 @@ void launch_quantize_mxfp8(const void* x, void* codes, void* scales,
 void launch_kv_append(const void* qkv, void* kcache, void* vcache,
                       const void* pos, int B, int H, int smax,
                       hipStream_t stream);
+void launch_kv_append_range(const void* qkv, void* kcache, void* vcache,
+                            int B, int H, int P, int smax,
+                            hipStream_t stream);
 void launch_decode_attention(const void* qkv, const void* kcache,
                              const void* vcache, void* out, const void* pos,
                              int B, int H, int smax, float scale,

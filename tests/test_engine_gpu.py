@@ -508,3 +508,28 @@ def test_engine_gpt2_fp8():
     corr_emu = np.corrcoef(ref8.ravel(), ref16.ravel())[0, 1]
     assert corr_gpu > corr_emu - 0.03, (corr_gpu, corr_emu)
     assert corr_gpu > 0.95, corr_gpu
+
+
+def test_decode_prefill_matches_sequential():
+    """Fused prefill-into-cache must leave the session in the same state
+    as token-by-token priming: the next decoded steps agree."""
+    from trtlab_amd.engine.decode import DecodeSession
+    from trtlab_amd.models import build_gpt2
+
+    g = build_gpt2(batch=2, seq=256, layers=2, seed=0, embeddings=True)
+    rng = np.random.RandomState(71)
+    prompt = rng.randint(1, 50257, (2, 100)).astype(np.int32)  # P % 128 != 0
+    nxt = rng.randint(1, 50257, (2, 3)).astype(np.int32)
+
+    a = DecodeSession(g, batch=2, smax=256, capture=False)
+    for t in range(100):
+        out_seq = a.step(prompt[:, t])
+    b = DecodeSession(g, batch=2, smax=256, capture=False)
+    out_pre = b.prefill(prompt)
+    scale = max(np.abs(out_seq).max(), 1e-6)
+    assert np.abs(out_pre - out_seq).max() / scale < 0.05
+
+    for t in range(3):
+        sa = a.step(nxt[:, t])
+        sb = b.step(nxt[:, t])
+        assert np.abs(sa - sb).max() / max(np.abs(sa).max(), 1e-6) < 0.05, t
