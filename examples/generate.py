@@ -33,10 +33,9 @@ def main():
     sess = DecodeSession(g, batch=args.batch, smax=1024, lm_head=True)
 
     rng = np.random.RandomState(0)
-    prompt = rng.randint(1, 50257, (args.batch, args.prompt_len))
-    # prefill = sequential priming (fused prefill-into-cache: round 2)
-    for t in range(args.prompt_len):
-        logits = sess.step(prompt[:, t].astype(np.int32))
+    prompt = rng.randint(1, 50257,
+                         (args.batch, args.prompt_len)).astype(np.int32)
+    logits = sess.prefill(prompt)  # one fused pass fills the KV caches
 
     toks = np.argmax(logits, axis=1).astype(np.int32)
     out = [toks]
