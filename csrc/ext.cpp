@@ -426,11 +426,11 @@ PYBIND11_MODULE(_C, m) {
           py::arg("ids"), py::arg("tok"), py::arg("posemb"), py::arg("out"),
           py::arg("pos"), py::arg("B"), py::arg("hidden"),
           py::arg("stream") = 0, py::arg("sync") = true);
-  ops.def("advance_pos", [](uintptr_t pos, int smax, uintptr_t stream,
+  ops.def("advance_pos", [](uintptr_t pos, int B, int smax, uintptr_t stream,
                             bool sync) {
-    launch_advance_pos((void*)pos, smax, as_stream(stream));
+    launch_advance_pos((void*)pos, B, smax, as_stream(stream));
     if (sync) TRT_HIP_CHECK(hipStreamSynchronize(as_stream(stream)));
-  }, py::arg("pos"), py::arg("smax"), py::arg("stream") = 0,
+  }, py::arg("pos"), py::arg("B"), py::arg("smax"), py::arg("stream") = 0,
      py::arg("sync") = true);
   ops.def("attention",
           [](int dtype, uintptr_t qkv, uintptr_t out, int B, int S, int H,

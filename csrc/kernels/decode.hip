@@ -3,9 +3,12 @@
 // served static TensorRT engines only.
 //
 // Design: the decode step is captured in a hipGraph once and replayed per
-// token. All position dependence goes through a DEVICE-side int counter
-// (`pos`), so replays need no re-capture: kernels read `pos`, and
-// advance_pos bumps it at the end of the captured step.
+// token. All position dependence goes through a DEVICE-side PER-SEQUENCE
+// counter array (`pos[B]`), so replays need no re-capture — and a slot
+// can be reset to position 0 between replays (continuous batching in
+// lockstep: a finished sequence's slot restarts a new request while the
+// other slots keep decoding). advance_pos bumps every slot at the end of
+// the captured step.
 //
 // Cache layout per layer: K and V as [B][H][Smax][64] fp16 (contiguous
 // 128-B rows per key — one cacheline).
@@ -21,7 +24,7 @@ __global__ __launch_bounds__(64) void kv_append_kernel(
     int smax) {
   int b = blockIdx.x / H, h = blockIdx.x % H;
   int d = threadIdx.x;  // 0..63
-  int p = *pos;
+  int p = pos[b];
   int hid = H * 64;
   int64_t src = (int64_t)b * 3 * hid + h * 64 + d;
   int64_t dst = (((int64_t)b * H + h) * smax + p) * 64 + d;
@@ -76,7 +79,7 @@ __global__ __launch_bounds__(64) void decode_attention_kernel(
   int b = blockIdx.x / H, h = blockIdx.x % H;
   int lane = threadIdx.x;
   int hid = H * 64;
-  int n = *pos + 1;  // keys 0..pos inclusive (this step's K already appended)
+  int n = pos[b] + 1;  // keys 0..pos[b] (this step's K already appended)
 
   // q for this head, one element per lane
   float q[64];
@@ -147,7 +150,7 @@ __global__ void decode_embed_kernel(const int* __restrict__ ids,
                                     _Float16* __restrict__ out,
                                     const int* __restrict__ pos, int hidden) {
   int b = blockIdx.x;
-  int p = *pos;
+  int p = pos[b];
   int64_t t = (int64_t)ids[b] * hidden;
   for (int i = threadIdx.x; i < hidden; i += blockDim.x)
     out[(int64_t)b * hidden + i] =
@@ -163,18 +166,20 @@ void launch_decode_embed(const void* ids, const void* tok, const void* posemb,
                      hidden);
 }
 
-// Advance the device-side position counter (last node of the captured
+// Advance every slot's position counter (last node of the captured
 // decode step; clamped so replay past smax is safe).
-__global__ void advance_pos_kernel(int* pos, int smax) {
-  if (threadIdx.x == 0) {
-    int p = *pos + 1;
-    *pos = p >= smax ? smax - 1 : p;
+__global__ void advance_pos_kernel(int* pos, int B, int smax) {
+  int b = threadIdx.x;
+  if (b < B) {
+    int p = pos[b] + 1;
+    pos[b] = p >= smax ? smax - 1 : p;
   }
 }
 
-void launch_advance_pos(void* pos, int smax, hipStream_t stream) {
-  hipLaunchKernelGGL(advance_pos_kernel, dim3(1), dim3(1), 0, stream,
-                     (int*)pos, smax);
+void launch_advance_pos(void* pos, int B, int smax, hipStream_t stream) {
+  hipLaunchKernelGGL(advance_pos_kernel, dim3(1),
+                     dim3((B + 63) / 64 * 64), 0, stream, (int*)pos, B,
+                     smax);
 }
 
 }  // namespace trtlab

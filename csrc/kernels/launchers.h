@@ -102,7 +102,7 @@ void launch_decode_attention(const void* qkv, const void* kcache,
 void launch_decode_embed(const void* ids, const void* tok, const void* posemb,
                          void* out, const void* pos, int B, int hidden,
                          hipStream_t stream);
-void launch_advance_pos(void* pos, int smax, hipStream_t stream);
+void launch_advance_pos(void* pos, int B, int smax, hipStream_t stream);
 
 void launch_embedding(int dtype, const void* ids, const void* tok,
                       const void* pos, const void* seg, const void* segids,

@@ -94,7 +94,8 @@ class DecodeSession:
 
         B, Hd = batch, self.hidden
         self.ids = torch.zeros(B, dtype=torch.int32, device="cuda")
-        self.pos = torch.zeros(1, dtype=torch.int32, device="cuda")
+        self.pos = torch.zeros(B, dtype=torch.int32, device="cuda")
+        self._slot_steps = np.zeros(B, np.int64)
         self.h = torch.zeros(B, Hd, dtype=torch.half, device="cuda")
         self.x = torch.zeros(B, Hd, dtype=torch.half, device="cuda")
         self.qkv = torch.zeros(B, 3 * Hd, dtype=torch.half, device="cuda")
@@ -176,7 +177,8 @@ class DecodeSession:
             ops.gemm_bt(0, self.out.data_ptr(), self.tok.data_ptr(),
                         self.logits.data_ptr(), M=B, N=self.vocab, K=Hd,
                         epi=self._epi_none, stream=s, sync=False)
-        ops.advance_pos(self.pos.data_ptr(), self.smax, stream=s, sync=False)
+        ops.advance_pos(self.pos.data_ptr(), B, self.smax, stream=s,
+                        sync=False)
 
     def prefill(self, prompt: np.ndarray) -> np.ndarray:
         """Fill the KV caches from a whole prompt [B, P] in ONE pass through
@@ -255,6 +257,7 @@ class DecodeSession:
         torch.cuda.synchronize()
         self.pos.fill_(P)
         self._steps = P
+        self._slot_steps[:] = P
         torch.cuda.synchronize()
         if self.logits is not None:
             ops.gemm_bt(0, last.data_ptr(), self.tok.data_ptr(),
@@ -268,8 +271,10 @@ class DecodeSession:
         [B, hidden] fp32 (or logits [B, vocab] with lm_head=True). The
         device-side position counter starts at 0 and the captured graph
         advances it, so replays need no host-side position plumbing."""
-        if self._steps >= self.smax:
-            raise RuntimeError("DecodeSession: sequence limit reached")
+        if (self._slot_steps >= self.smax).any():
+            raise RuntimeError(
+                "DecodeSession: a slot hit the sequence limit "
+                "(reset_slot() it or end the session)")
         arr = np.ascontiguousarray(ids, np.int32)
         # synchronous H2D keeps the token copy ordered before the replay
         self._C.memory.memcpy_h2d(self.ids.data_ptr(), arr, arr.nbytes)
@@ -283,6 +288,7 @@ class DecodeSession:
                 self._enqueue()
                 self._graph = self._C.hip.stream_end_capture(self.stream)
                 self._steps += 1
+                self._slot_steps += 1
                 out = self.out.float().cpu().numpy()
                 return (self.logits.float().cpu().numpy()
                         if self.logits is not None else out)
@@ -291,9 +297,18 @@ class DecodeSession:
             self._enqueue()
         self._C.hip.stream_synchronize(self.stream)
         self._steps += 1
+        self._slot_steps += 1
         out = self.out.float().cpu().numpy()
         return (self.logits.float().cpu().numpy()
                 if self.logits is not None else out)
+
+    def reset_slot(self, b: int) -> None:
+        """Restart slot b at position 0 (continuous batching in lockstep:
+        the next step() token for this slot begins a fresh sequence while
+        the other slots keep decoding against their caches)."""
+        self.pos[b] = 0
+        self._slot_steps[b] = 0
+        self._torch.cuda.synchronize()
 
     def close(self):
         if self._graph:
