@@ -560,14 +560,9 @@ def test_decode_slot_reset_continuous_batching():
     mix[1] = toks[0]  # slot 1 now follows slot 0's fresh sequence
     outs = [tst.step(mix[:, t - 8]) for t in range(8, 16)]
 
-    # slot 0 (never reset) must match the control exactly at every step
-    for i, o in enumerate(outs):
-        pass
     scale = max(np.abs(ctl_out[-1]).max(), 1e-6)
-    # slot 0 of tst at step 15 saw tokens toks[0, 0:8] then toks[0, 0:8]
-    # again -- NOT the control stream; instead verify slot independence:
-    # slot 1 of tst (reset, fed toks[0,0..7]) must equal slot 0 of a FRESH
-    # session fed the same tokens.
+    # slot independence: slot 1 of tst (reset, then fed toks[0, 0:8]) must
+    # equal slot 0 of a FRESH session fed the same tokens.
     fresh = DecodeSession(g, batch=2, smax=128, capture=False)
     f_out = None
     for t in range(8):
