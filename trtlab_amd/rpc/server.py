@@ -138,7 +138,10 @@ class Server:
     def _run_loop(self):
         self._loop = asyncio.new_event_loop()
         asyncio.set_event_loop(self._loop)
-        self._loop.run_until_complete(self._serve())
+        try:
+            self._loop.run_until_complete(self._serve())
+        finally:
+            self._loop.close()
 
     async def _serve(self):
         self._server = grpc.aio.server()
