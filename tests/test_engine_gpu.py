@@ -569,3 +569,29 @@ def test_decode_slot_reset_continuous_batching():
         f_out = fresh.step(np.stack([toks[0, t], toks[0, t]]))
     err = np.abs(outs[-1][1] - f_out[0]).max()
     assert err / scale < 0.05, err
+
+
+def test_decode_slot_reset_under_capture():
+    """reset_slot between hipGraph replays (the serving configuration):
+    host writes to the device position array are honored by subsequent
+    replays, and match the eager path."""
+    from trtlab_amd.engine.decode import DecodeSession
+    from trtlab_amd.models import build_gpt2
+
+    g = build_gpt2(batch=2, seq=128, layers=2, seed=0, embeddings=True)
+    rng = np.random.RandomState(91)
+    toks = rng.randint(1, 50257, (2, 12)).astype(np.int32)
+
+    cap = DecodeSession(g, batch=2, smax=128, capture=True)
+    eag = DecodeSession(g, batch=2, smax=128, capture=False)
+    for t in range(6):
+        oc = cap.step(toks[:, t])
+        oe = eag.step(toks[:, t])
+    cap.reset_slot(1)
+    eag.reset_slot(1)
+    for t in range(6, 12):
+        oc = cap.step(toks[:, t])
+        oe = eag.step(toks[:, t])
+        scale = max(np.abs(oe).max(), 1e-6)
+        assert np.abs(oc - oe).max() / scale < 0.05, t
+    cap.close()
