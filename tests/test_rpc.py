@@ -150,3 +150,46 @@ def test_client_deadline_cancel(echo_server):
         c2.close()
     finally:
         s2.shutdown()
+
+
+def test_shm_pool_checkout_release():
+    """ShmPool: depth-bounded checkout, contents visible across handles,
+    release unblocks waiters, close unlinks."""
+    import threading
+
+    import numpy as np
+
+    from trtlab_amd.rpc.client import ShmPool
+
+    pool = ShmPool(1024, depth=2)
+    a1 = np.arange(256, dtype=np.float32)
+    seg1, name1, size1 = pool.checkout(a1)
+    seg2, name2, size2 = pool.checkout(a1 * 2)
+    assert name1 != name2 and size1 == a1.nbytes
+
+    # reader sees the bytes through an independent mapping
+    from multiprocessing import shared_memory
+
+    rd = shared_memory.SharedMemory(name=name1)
+    got = np.frombuffer(rd.buf[:size1], dtype=np.float32).copy()
+    rd.close()
+    assert np.array_equal(got, a1)
+
+    # third checkout blocks until a release
+    acquired = threading.Event()
+
+    def taker():
+        s, _, _ = pool.checkout(a1)
+        acquired.set()
+        pool.release(s)
+
+    t = threading.Thread(target=taker, daemon=True)
+    t.start()
+    assert not acquired.wait(timeout=0.2)
+    pool.release(seg1)
+    assert acquired.wait(timeout=5)
+    pool.release(seg2)
+    t.join(timeout=5)
+    pool.close()
+    with pytest.raises(FileNotFoundError):
+        shared_memory.SharedMemory(name=name2)
