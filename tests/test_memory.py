@@ -89,3 +89,29 @@ def test_huge_page_buffer_pinned_gpu():
     assert out[0] == 2.25 and out[-1] == 2.25
     d.close()
     b.close()
+
+
+def test_arena_planner_reuses_disjoint_lifetimes():
+    """Best-fit arena: tensors with disjoint lifetimes share bytes, total
+    stays below the no-reuse sum (the property the engine's activation
+    memory depends on)."""
+    ap = ArenaPlanner()
+    ap.add("a", 1000, 0, 2)
+    ap.add("b", 1000, 3, 5)   # disjoint from a -> can share
+    ap.add("c", 1000, 1, 4)   # overlaps both
+    offsets, total = ap.plan()
+    assert offsets["a"] == offsets["b"]
+    assert offsets["c"] != offsets["a"]
+    assert total < 3 * 1024  # reuse happened (256-B aligned blocks)
+
+
+def test_transactional_stack_nesting():
+    st = TransactionalStack(4096)
+    st.begin()
+    st.allocate(100)
+    st.begin()
+    st.allocate(100)
+    st.commit()   # inner rollback point
+    inner_top = st.high_water
+    st.commit()
+    assert st.high_water < inner_top
