@@ -146,3 +146,20 @@ def test_onnx_import_passthrough_and_clip_ops():
     x = rng.randn(2, 8, 8, 8).astype(np.float32) * 0.5
     out = run_reference(plan, x)
     assert out.shape == (2, 10) and np.isfinite(out).all()
+
+
+def test_onnx_roundtrip_resnet18_basic_blocks():
+    """Basic-block graphs survive export -> import -> compile -> reference."""
+    from trtlab_amd.engine.planner import Planner
+    from trtlab_amd.engine.reference import run_reference
+    from trtlab_amd.models import build_resnet
+
+    g = build_resnet(18, batch=1, image=64, seed=0)
+    g2 = import_onnx(export_onnx(g), name="rt18")
+    assert len(g2.nodes) == len(g.nodes)
+    plan = Planner().compile(g2)
+    x = np.random.RandomState(9).randn(*plan.input_shape).astype(
+        np.float32) * 0.5
+    out = run_reference(plan, x)
+    ref = run_reference(Planner().compile(g), x)
+    assert np.allclose(out, ref, atol=1e-4)
