@@ -84,3 +84,25 @@ def test_dispatcher_batches_by_size_and_timeout():
     assert calls[0] == [1, 2]
     assert calls[1] == [3]
     d.shutdown()
+
+
+def test_batcher_timeout_closes_partial_batch():
+    """StandardBatcher: a partial batch closes on the timeout path, not
+    only on max size (reference batcher.h close_batch :134)."""
+    import time
+
+    from trtlab_amd.core import Dispatcher
+
+    seen = []
+
+    def compute(batch):
+        seen.append(len(batch))
+        return [x * 2 for x in batch]
+
+    d = Dispatcher(max_batch_size=8, timeout_s=0.02,
+                   compute_batch_fn=compute,
+                   workers=1)
+    futs = [d.enqueue(i) for i in (1, 2, 3)]  # < max_batch_size
+    assert [f.result(5) for f in futs] == [2, 4, 6]
+    assert seen and seen[0] == 3  # closed by timeout as one partial batch
+    d.shutdown()
