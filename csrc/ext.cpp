@@ -54,12 +54,16 @@ OpDesc op_from_dict(const py::dict& d) {
 }
 
 // Lazily-grown scratch for the raw ops.* test path (serial use only; the
-// engine allocates per-context scratch for concurrent replay).
+// engine allocates per-context scratch for concurrent replay). Grow-only:
+// a captured hipGraph (DecodeSession) bakes the scratch address into its
+// nodes, so an old allocation must stay live for the process lifetime —
+// regrow allocates a NEW slab and retires the old one without freeing it.
 void* test_scratch(size_t need) {
+  static std::vector<void*> retired;  // kept live: graphs may reference them
   static void* p = nullptr;
   static size_t cap = 0;
   if (need > cap) {
-    if (p) hipFree(p);
+    if (p) retired.push_back(p);
     TRT_HIP_CHECK(hipMalloc(&p, need));
     cap = need;
   }

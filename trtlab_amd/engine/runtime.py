@@ -98,18 +98,29 @@ class NativeContext:
         """Pinned host output binding (valid after synchronize)."""
         return self._out_view
 
+    def write_input(self, batch: np.ndarray) -> None:
+        """Copy a host batch into the pinned input binding, handling the
+        bf16 bit-pattern representation (numpy has no bf16 dtype — float
+        inputs are converted to bf16 BITS, never numerically cast to int16)
+        and validating shape up front so malformed requests fail loudly."""
+        if tuple(batch.shape) != tuple(self._in_view.shape):
+            raise ValueError(
+                f"input shape {tuple(batch.shape)} != plan shape "
+                f"{tuple(self._in_view.shape)}")
+        if self._in_bf16 and batch.dtype != np.int16:
+            from trtlab_amd.engine.planner import _bf16_bits
+
+            np.copyto(self._in_view,
+                      _bf16_bits(batch).reshape(self._in_view.shape))
+        else:
+            np.copyto(self._in_view,
+                      batch.astype(self._in_view.dtype, copy=False))
+
     def infer(self, batch: Optional[np.ndarray] = None) -> np.ndarray:
         """Synchronous convenience path; returns a COPY of the output (the
         zero-copy `.output` view is only valid while this context lives)."""
         if batch is not None:
-            if self._in_bf16 and batch.dtype != np.int16:
-                from trtlab_amd.engine.planner import _bf16_bits
-
-                np.copyto(self._in_view,
-                          _bf16_bits(batch).reshape(self._in_view.shape))
-            else:
-                np.copyto(self._in_view,
-                          batch.astype(self._in_view.dtype, copy=False))
+            self.write_input(batch)
         self.ctx.launch()
         self.ctx.synchronize()
         out = np.array(self._out_view, copy=True)
@@ -217,25 +228,38 @@ class InferRunner:
         fut: Future = Future()
         tp = self.manager.thread_pools
 
+        # Every stage propagates failures into `fut` and releases the
+        # checkout — a malformed request (bad shape/dtype) must reject its
+        # own Future, never hang the awaiting RPC handler.
         def pre():
             co = self._pool.pop()  # blocks: concurrency limiter
-            ctx: NativeContext = co.item
-            np.copyto(ctx.input, batch.astype(ctx.input.dtype, copy=False))
-            tp["hip"].enqueue(hip_stage, co)
+            try:
+                ctx: NativeContext = co.item
+                ctx.write_input(batch)  # bf16-bit aware + shape-validated
+                tp["hip"].enqueue(hip_stage, co)
+            except BaseException as e:  # noqa: BLE001
+                fut.set_exception(e)
+                co.release()
 
         def hip_stage(co):
-            co.item.launch()
-            tp["post"].enqueue(post, co)
+            try:
+                co.item.launch()
+                tp["post"].enqueue(post, co)
+            except BaseException as e:  # noqa: BLE001
+                fut.set_exception(e)
+                co.release()
 
         def post(co):
             try:
-                co.item.synchronize()
-                out = np.array(co.item.output, copy=True)
+                out = None
+                try:
+                    co.item.synchronize()
+                    out = np.array(co.item.output, copy=True)
+                finally:
+                    co.release()
                 fut.set_result(out)
             except BaseException as e:  # noqa: BLE001
                 fut.set_exception(e)
-            finally:
-                co.release()
 
         tp["pre"].enqueue(pre)
         return fut

@@ -143,13 +143,22 @@ class BlockingBlockPool:
         self._cv = threading.Condition()
 
     def pop(self, timeout: Optional[float] = None) -> int:
+        import time as _time
+
         with self._cv:
-            deadline = None
+            deadline = None if timeout is None else _time.monotonic() + timeout
             while True:
                 p = self._native.acquire()
                 if p:
                     return p
-                if not self._cv.wait(timeout):
+                # monotonic deadline: a notify that loses the race to another
+                # waiter must not restart the full timeout window
+                remaining = None
+                if deadline is not None:
+                    remaining = deadline - _time.monotonic()
+                    if remaining <= 0:
+                        raise TimeoutError("BlockingBlockPool.pop timed out")
+                if not self._cv.wait(remaining):
                     raise TimeoutError("BlockingBlockPool.pop timed out")
 
     def push(self, ptr: int) -> None:

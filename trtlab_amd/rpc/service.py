@@ -41,7 +41,12 @@ class InferenceService(AsyncService):
     def __init__(self, resources: InferenceResources, metrics=None):
         super().__init__("trtlab.Inference", resources)
         self.metrics = metrics
-        self._shm_cache: dict = {}  # name -> SharedMemory (pooled clients)
+        from collections import OrderedDict
+
+        # name -> SharedMemory (pooled clients); true LRU: hits move to the
+        # back, eviction pops the FRONT (oldest) so stale segments from dead
+        # clients age out instead of thrashing the newest mapping
+        self._shm_cache: "OrderedDict[str, object]" = OrderedDict()
         self.register_unary("Compute", self._compute, InferRequest,
                             InferResponse)
 
@@ -69,9 +74,11 @@ class InferenceService(AsyncService):
                 shm = self._shm_cache.get(request.shm_name)
                 if shm is None:
                     shm = shared_memory.SharedMemory(name=request.shm_name)
-                    if len(self._shm_cache) >= 64:  # bounded
-                        self._shm_cache.popitem()[1].close()
+                    if len(self._shm_cache) >= 64:  # bounded, evict oldest
+                        self._shm_cache.popitem(last=False)[1].close()
                     self._shm_cache[request.shm_name] = shm
+                else:
+                    self._shm_cache.move_to_end(request.shm_name)
                 return np.frombuffer(
                     shm.buf[:int(request.shm_size)], dtype=dtype
                 ).reshape(shape).copy()
