@@ -47,22 +47,25 @@ def main():
     local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
     n_gpus = max(world, 1)
 
-    import torch
-    import torch.distributed as dist
-
-    distributed = world > 1
-    if distributed:
-        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
-        torch.cuda.set_device(local_rank)
-        dist.init_process_group("nccl")
-        # N ranks calibrate the same plan concurrently on CPU: share cores
-        torch.set_num_threads(max(1, (os.cpu_count() or world) // world))
-
     import trtlab_amd
     from trtlab_amd.engine.planner import DT_F16, DT_I8, Planner
     from trtlab_amd.engine.runtime import NativeEngine
     from trtlab_amd.models import build_bert, build_resnet
-    from trtlab_amd.parallel import broadcast_weights
+    from trtlab_amd.parallel import Communicator, broadcast_weights
+
+    # Device = local rank, wrapped by the visible device count so the
+    # 2-rank single-GPU proof run works within a 1-GPU lease (identity
+    # mapping on the driver's 8-GPU node).
+    ndev = max(trtlab_amd.native().hip.device_count(), 1)
+    local_rank = local_rank % ndev
+
+    distributed = world > 1
+    comm = None
+    if distributed:
+        # Owned RCCL communicator (csrc/runtime/comm.cpp): torchrun is only
+        # the launcher — comm bring-up is our file-based uid rendezvous,
+        # collectives run on our own streams (no torch.distributed).
+        comm = Communicator(rank=rank, world=world, device=local_rank)
 
     # Build the plan (identical on every rank: same seed).
     if args.model == "bert":
@@ -95,8 +98,9 @@ def main():
                        autotune=not args.no_autotune)
     if distributed:
         # RCCL weight broadcast at load: rank 0's blob is
-        # authoritative; replicas receive over xGMI.
-        broadcast_weights(eng, src_rank=0, device=local_rank)
+        # authoritative; replicas receive over xGMI (one fused in-place
+        # broadcast of the whole blob on the engine's weight memory).
+        broadcast_weights(eng, comm, src_rank=0)
 
     ctxs = [eng.create_context(capture=True) for _ in range(args.contexts)]
 
@@ -133,18 +137,20 @@ def main():
     run_steps(args.warmup, record=False)
 
     # ---- timed region ----
+    C = trtlab_amd.native()
     if distributed:
-        dist.barrier()
-    torch.cuda.synchronize()
+        comm.barrier()
+    C.hip.device_synchronize()
     t0 = time.perf_counter()
     run_steps(args.steps, record=True)
-    torch.cuda.synchronize()
+    C.hip.device_synchronize()
     elapsed = time.perf_counter() - t0
     if distributed:
-        t = torch.tensor([elapsed], device="cuda", dtype=torch.float64)
-        dist.all_reduce(t, op=dist.ReduceOp.MAX)
-        elapsed = float(t.item())
-        dist.barrier()
+        # whole-job time = MAX over ranks (RCCL fp64 all-reduce)
+        from trtlab_amd.parallel import OP_MAX
+
+        elapsed = comm.all_reduce_scalar(elapsed, op=OP_MAX)
+        comm.barrier()
 
     # ---- single-inflight latency phase (well-defined request latency:
     # H2D + graph replay + D2H, one at a time; not part of the throughput
@@ -156,7 +162,7 @@ def main():
         ctxs[0].synchronize()
         lat1.append((time.perf_counter() - t1) * 1e3)
     if distributed:
-        dist.barrier()
+        comm.barrier()
 
     if rank == 0:
         inf_s = n_gpus * args.steps * args.batch / elapsed
@@ -193,7 +199,8 @@ def main():
         }), flush=True)
 
     if distributed:
-        dist.destroy_process_group()
+        comm.barrier()
+        comm.close()
 
 
 if __name__ == "__main__":

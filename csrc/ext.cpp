@@ -6,6 +6,7 @@
 #include <pybind11/stl.h>
 
 #include "kernels/launchers.h"
+#include "runtime/comm.h"
 #include "runtime/runtime.h"
 
 namespace py = pybind11;
@@ -449,6 +450,89 @@ PYBIND11_MODULE(_C, m) {
           py::arg("S"), py::arg("H"), py::arg("D"), py::arg("scale"),
           py::arg("stream") = 0, py::arg("sync") = true,
           py::arg("causal") = 0, py::arg("seqlens") = 0);
+
+  // --------------------------------------------------------------- comm
+  // Owned RCCL collective layer (no torch.distributed in the data path).
+  auto comm = m.def_submodule("comm");
+  comm.def("unique_id", [] { return py::bytes(Communicator::unique_id()); });
+  py::class_<Communicator>(comm, "Communicator")
+      .def(py::init([](int rank, int world, py::bytes uid, int device) {
+             // clique connect can block for seconds — release the GIL so
+             // multi-rank-per-process tests don't deadlock
+             std::string u = uid;
+             py::gil_scoped_release rel;
+             return new Communicator(rank, world, u, device);
+           }),
+           py::arg("rank"), py::arg("world"), py::arg("uid"),
+           py::arg("device") = 0)
+      .def_property_readonly("rank", &Communicator::rank)
+      .def_property_readonly("world", &Communicator::world)
+      .def_property_readonly("device", &Communicator::device)
+      .def("broadcast",
+           [](Communicator& c, uintptr_t ptr, size_t bytes, int root,
+              uintptr_t stream) {
+             py::gil_scoped_release rel;
+             c.broadcast((void*)ptr, bytes, root, as_stream(stream));
+           },
+           py::arg("ptr"), py::arg("bytes"), py::arg("root") = 0,
+           py::arg("stream") = 0)
+      .def("all_reduce",
+           [](Communicator& c, uintptr_t ptr, size_t count, int dtype, int op,
+              uintptr_t stream) {
+             py::gil_scoped_release rel;
+             c.all_reduce((void*)ptr, count, dtype, op, as_stream(stream));
+           },
+           py::arg("ptr"), py::arg("count"), py::arg("dtype") = 3,
+           py::arg("op") = 0, py::arg("stream") = 0)
+      .def("all_gather",
+           [](Communicator& c, uintptr_t send, uintptr_t recv,
+              size_t bytes_per_rank, uintptr_t stream) {
+             py::gil_scoped_release rel;
+             c.all_gather((const void*)send, (void*)recv, bytes_per_rank,
+                          as_stream(stream));
+           },
+           py::arg("send"), py::arg("recv"), py::arg("bytes_per_rank"),
+           py::arg("stream") = 0)
+      .def("reduce_scatter",
+           [](Communicator& c, uintptr_t send, uintptr_t recv,
+              size_t count_per_rank, int dtype, int op, uintptr_t stream) {
+             py::gil_scoped_release rel;
+             c.reduce_scatter((const void*)send, (void*)recv, count_per_rank,
+                              dtype, op, as_stream(stream));
+           },
+           py::arg("send"), py::arg("recv"), py::arg("count_per_rank"),
+           py::arg("dtype") = 3, py::arg("op") = 0, py::arg("stream") = 0)
+      .def("send",
+           [](Communicator& c, uintptr_t ptr, size_t bytes, int peer,
+              uintptr_t stream) {
+             py::gil_scoped_release rel;
+             c.send((const void*)ptr, bytes, peer, as_stream(stream));
+           },
+           py::arg("ptr"), py::arg("bytes"), py::arg("peer"),
+           py::arg("stream") = 0)
+      .def("recv",
+           [](Communicator& c, uintptr_t ptr, size_t bytes, int peer,
+              uintptr_t stream) {
+             py::gil_scoped_release rel;
+             c.recv((void*)ptr, bytes, peer, as_stream(stream));
+           },
+           py::arg("ptr"), py::arg("bytes"), py::arg("peer"),
+           py::arg("stream") = 0)
+      .def("barrier",
+           [](Communicator& c) {
+             py::gil_scoped_release rel;
+             c.barrier();
+           })
+      .def("all_reduce_scalar",
+           [](Communicator& c, double v, int op) {
+             py::gil_scoped_release rel;
+             return c.all_reduce_scalar(v, op);
+           },
+           py::arg("v"), py::arg("op") = 0)
+      .def("stream_synchronize", [](Communicator& c) {
+        py::gil_scoped_release rel;
+        c.stream_synchronize();
+      });
 
   // ------------------------------------------------------------- engine
   py::class_<Engine, std::shared_ptr<Engine>>(m, "Engine")

@@ -28,6 +28,7 @@ SOURCES = [
     "csrc/kernels/embedding.hip",
     "csrc/runtime/memory.cpp",
     "csrc/runtime/executor.cpp",
+    "csrc/runtime/comm.cpp",
     "csrc/ext.cpp",
 ]
 
@@ -36,6 +37,7 @@ HEADERS = [
     "csrc/kernels/gemm_common.h",
     "csrc/kernels/launchers.h",
     "csrc/runtime/runtime.h",
+    "csrc/runtime/comm.h",
 ]
 
 
@@ -91,7 +93,8 @@ def build(verbose: bool = True) -> Path:
         list(ex.map(run, jobs))
 
     if jobs or not out.exists():
-        link = [HIPCC, "-shared", "-fPIC", *[str(o) for o in objs], "-o", str(out)]
+        link = [HIPCC, "-shared", "-fPIC", *[str(o) for o in objs],
+                "-L/opt/rocm/lib", "-lrccl", "-o", str(out)]
         if verbose:
             print(f"[link] {out.relative_to(ROOT)}", flush=True)
         r = subprocess.run(link, capture_output=True, text=True)

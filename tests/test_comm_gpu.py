@@ -1,0 +1,129 @@
+"""GPU tests for the owned RCCL collective layer (csrc/runtime/comm.cpp).
+
+Runs a REAL 2-rank RCCL clique on however many GPUs are present — both
+ranks share device 0 on a 1-GPU box (RCCL multi-rank-per-GPU), which is
+how the multi-GPU path is proven within a single-GPU lease (VERDICT.md
+next-round item 1). No torch.distributed anywhere: comm bring-up is the
+file-based unique-id rendezvous in trtlab_amd.parallel.
+"""
+import os
+import subprocess
+import sys
+from pathlib import Path
+
+import pytest
+
+ROOT = Path(__file__).resolve().parent.parent
+
+pytestmark = pytest.mark.gpu
+
+_WORKER = r"""
+import os, sys
+import numpy as np
+
+sys.path.insert(0, os.environ["TRTLAB_ROOT"])
+rank = int(os.environ["RANK"])
+world = int(os.environ["WORLD_SIZE"])
+
+from trtlab_amd import native
+from trtlab_amd.parallel import (Communicator, OP_MAX, OP_SUM, DT_F32,
+                                 broadcast_weights)
+
+C = native()
+ndev = C.hip.device_count()
+device = rank % max(ndev, 1)
+comm = Communicator(rank=rank, world=world, device=device,
+                    rendezvous_path=os.environ["RDV_PATH"])
+
+# 1) fused byte-blob broadcast, in place on device memory
+n = 1 << 20
+host = (np.arange(n, dtype=np.uint8) * 7 + 13).astype(np.uint8)
+ptr = C.memory.device_malloc(n, device)
+if rank == 0:
+    C.memory.memcpy_h2d(ptr, host, n)
+else:
+    C.memory.memset_d(ptr, 0, n)
+comm.broadcast(ptr, n, root=0)
+comm.synchronize()
+back = np.zeros(n, dtype=np.uint8)
+C.memory.memcpy_d2h(back, ptr, n)
+assert np.array_equal(back, host), "broadcast blob mismatch"
+
+# 2) all-reduce on fp32 device data
+m = 4096
+vals = np.full(m, float(rank + 1), dtype=np.float32)
+p2 = C.memory.device_malloc(m * 4, device)
+C.memory.memcpy_h2d(p2, vals, m * 4)
+comm.all_reduce(p2, m, dtype=DT_F32, op=OP_SUM)
+comm.synchronize()
+out = np.zeros(m, dtype=np.float32)
+C.memory.memcpy_d2h(out, p2, m * 4)
+expect = world * (world + 1) / 2.0
+assert np.allclose(out, expect), f"allreduce got {out[:4]} want {expect}"
+
+# 3) the bench choreography: barrier + MAX-of-scalar
+comm.barrier()
+mx = comm.all_reduce_scalar(0.1 * (rank + 1), op=OP_MAX)
+assert abs(mx - 0.1 * world) < 1e-9, mx
+
+# 4) broadcast_weights on a real engine: rank!=0 zeroes its blob, receives
+# rank 0's, and the forward then matches the fp32 reference
+from trtlab_amd.models import build_resnet
+from trtlab_amd.engine.planner import Planner
+from trtlab_amd.engine.runtime import NativeEngine
+from trtlab_amd.engine.reference import run_reference
+
+g = build_resnet(18, batch=1, image=64, seed=0)
+plan = Planner().compile(g)
+eng = NativeEngine(plan, device=device)
+if rank != 0:
+    eng.upload_weights(np.zeros_like(plan.weights))
+broadcast_weights(eng, comm, src_rank=0)
+ctx = eng.create_context(capture=True)
+x = np.random.RandomState(7).randn(*plan.input_shape).astype(np.float32) * 0.5
+out = ctx.infer(x).astype(np.float32)
+ref = run_reference(plan, x)
+err = float(np.abs(out - ref).max() / max(np.abs(ref).max(), 1e-6))
+assert err < 0.1, f"post-broadcast forward mismatch: {err}"
+
+comm.barrier()
+comm.close()
+print(f"rank {rank} COMM_OK", flush=True)
+"""
+
+
+def test_rccl_two_rank_clique(tmp_path):
+    """2 ranks, possibly sharing one GPU: broadcast / all-reduce / barrier /
+    scalar-max / engine weight broadcast, all through the owned RCCL layer."""
+    script = tmp_path / "worker.py"
+    script.write_text(_WORKER)
+    procs = []
+    for rank in range(2):
+        env = dict(os.environ)
+        env.update(
+            TRTLAB_ROOT=str(ROOT), RANK=str(rank), WORLD_SIZE="2",
+            RDV_PATH=str(tmp_path / "uid.bin"),
+            HSA_ENABLE_IPC_MODE_LEGACY="0",
+        )
+        procs.append(subprocess.Popen(
+            [sys.executable, str(script)], env=env, cwd=str(ROOT),
+            stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True))
+    outs = []
+    for p in procs:
+        try:
+            out, _ = p.communicate(timeout=600)
+        except subprocess.TimeoutExpired:
+            for q in procs:
+                q.kill()
+            raise
+        outs.append(out)
+    for rank, (p, out) in enumerate(zip(procs, outs)):
+        assert p.returncode == 0, f"rank {rank} failed:\n{out}"
+        assert "COMM_OK" in out, f"rank {rank}:\n{out}"
+
+
+def test_unique_id_native():
+    from trtlab_amd import native
+
+    uid = native().comm.unique_id()
+    assert isinstance(uid, bytes) and len(uid) == 128

@@ -61,3 +61,82 @@ def test_gloo_world2_bench_flow(tmp_path):
         capture_output=True, text=True, timeout=240, env=env, cwd=str(ROOT))
     assert r.returncode == 0, f"stdout:\n{r.stdout}\nstderr:\n{r.stderr}"
     assert r.stdout.count("OK") == 2
+
+
+# ---------------------------------------------------------------------------
+# Owned-comm support logic that runs without a GPU: the unique-id rendezvous
+# (trtlab_amd.parallel.exchange_unique_id) and the replica scheduler.
+
+def test_rendezvous_publish_and_read(tmp_path):
+    import threading
+
+    from trtlab_amd.parallel import exchange_unique_id
+
+    path = str(tmp_path / "uid.bin")
+    uid = bytes(range(128))
+    got = {}
+
+    def reader():
+        got["uid"] = exchange_unique_id(1, 2, path=path, timeout=10)
+
+    t = threading.Thread(target=reader)
+    t.start()
+    exchange_unique_id(0, 2, uid=uid, path=path)
+    t.join(timeout=10)
+    assert got["uid"] == uid
+
+
+def test_rendezvous_rejects_stale_file(tmp_path):
+    import os
+    import time
+
+    from trtlab_amd.parallel import exchange_unique_id
+
+    path = str(tmp_path / "uid.bin")
+    with open(path, "wb") as f:
+        f.write(b"\x01" * 128)
+    old = time.time() - 3600
+    os.utime(path, (old, old))  # a file from a previous launch
+    with pytest.raises(TimeoutError):
+        exchange_unique_id(1, 2, path=path, timeout=0.5)
+
+
+def test_rendezvous_rank0_requires_uid(tmp_path):
+    from trtlab_amd.parallel import exchange_unique_id
+
+    with pytest.raises(ValueError):
+        exchange_unique_id(0, 2, path=str(tmp_path / "x.bin"))
+
+
+def test_communicator_world1_is_noop():
+    from trtlab_amd.parallel import Communicator
+
+    c = Communicator(rank=0, world=1, device=0)
+    c.broadcast(0, 0)
+    c.barrier()
+    assert c.all_reduce_scalar(3.5) == 3.5
+    c.close()
+
+
+def test_replica_group_least_outstanding_and_failover():
+    from trtlab_amd.parallel import ReplicaGroup
+
+    g = ReplicaGroup(engines=["e0", "e1", "e2"])
+    a = g.acquire()
+    b = g.acquire()
+    c = g.acquire()
+    assert sorted([a, b, c]) == [0, 1, 2]  # spreads across all replicas
+    g.release(b)
+    assert g.acquire() == b  # least-outstanding wins
+    # failover: unhealthy replicas are never picked
+    g.mark_unhealthy(0)
+    picks = {g.acquire() for _ in range(4)}
+    assert 0 not in picks
+    g.mark_healthy(0)
+    g.outstanding = [0, 5, 5]
+    assert g.next_index() == 0
+    # all down -> loud failure
+    g2 = ReplicaGroup(engines=["x"])
+    g2.mark_unhealthy(0)
+    with pytest.raises(RuntimeError):
+        g2.next_index()
