@@ -537,26 +537,36 @@ PYBIND11_MODULE(_C, m) {
   // ------------------------------------------------------------- engine
   py::class_<Engine, std::shared_ptr<Engine>>(m, "Engine")
       .def(py::init([](int device, py::buffer weights, size_t arena_bytes,
-                       std::vector<py::dict> ops_in, int64_t input_off,
-                       size_t input_bytes, int64_t output_off,
-                       size_t output_bytes, bool managed_weights) {
+                       std::vector<py::dict> ops_in,
+                       std::vector<std::pair<int64_t, size_t>> inputs,
+                       std::vector<std::pair<int64_t, size_t>> outputs,
+                       bool managed_weights) {
              py::buffer_info wi = weights.request();
              std::vector<OpDesc> ops;
              ops.reserve(ops_in.size());
              for (auto& d : ops_in) ops.push_back(op_from_dict(d));
+             auto to_bindings =
+                 [](const std::vector<std::pair<int64_t, size_t>>& v) {
+                   std::vector<BindingDesc> out;
+                   for (auto& p : v) out.push_back({p.first, p.second});
+                   return out;
+                 };
              return std::make_shared<Engine>(
                  device, wi.ptr, (size_t)(wi.size * wi.itemsize), arena_bytes,
-                 std::move(ops), input_off, input_bytes, output_off,
-                 output_bytes, managed_weights);
+                 std::move(ops), to_bindings(inputs), to_bindings(outputs),
+                 managed_weights);
            }),
            py::arg("device"), py::arg("weights"), py::arg("arena_bytes"),
-           py::arg("ops"), py::arg("input_off"), py::arg("input_bytes"),
-           py::arg("output_off"), py::arg("output_bytes"),
+           py::arg("ops"), py::arg("inputs"), py::arg("outputs"),
            py::arg("managed_weights") = false)
       .def_property_readonly("device", &Engine::device)
       .def_property_readonly("arena_bytes", &Engine::arena_bytes)
       .def_property_readonly("input_bytes", &Engine::input_bytes)
       .def_property_readonly("output_bytes", &Engine::output_bytes)
+      .def_property_readonly("n_inputs",
+                             [](Engine& e) { return e.inputs().size(); })
+      .def_property_readonly("n_outputs",
+                             [](Engine& e) { return e.outputs().size(); })
       .def_property_readonly("weights_ptr", &Engine::weights_ptr)
       .def_property_readonly("weight_bytes", &Engine::weight_bytes)
       .def("upload_weights", [](Engine& e, py::buffer b) {
@@ -584,15 +594,25 @@ PYBIND11_MODULE(_C, m) {
              auto t = c.stage_times_ms();
              return py::make_tuple(t[0], t[1], t[2]);
            })
-      .def_property_readonly("host_input_ptr", &ExecutionContext::host_input_ptr)
-      .def_property_readonly("host_output_ptr", &ExecutionContext::host_output_ptr)
+      .def_property_readonly("host_input_ptr",
+                             [](ExecutionContext& c) {
+                               return c.host_input_ptr(0);
+                             })
+      .def_property_readonly("host_output_ptr",
+                             [](ExecutionContext& c) {
+                               return c.host_output_ptr(0);
+                             })
       .def_property_readonly("arena_ptr", &ExecutionContext::arena_ptr)
       .def("input_view",
-           [](ExecutionContext& c, size_t bytes) {
-             return py::memoryview::from_memory((void*)c.host_input_ptr(),
+           [](ExecutionContext& c, size_t bytes, int i) {
+             return py::memoryview::from_memory((void*)c.host_input_ptr(i),
                                                 bytes);
-           })
-      .def("output_view", [](ExecutionContext& c, size_t bytes) {
-        return py::memoryview::from_memory((void*)c.host_output_ptr(), bytes);
-      });
+           },
+           py::arg("bytes"), py::arg("i") = 0)
+      .def("output_view",
+           [](ExecutionContext& c, size_t bytes, int i) {
+             return py::memoryview::from_memory((void*)c.host_output_ptr(i),
+                                                bytes);
+           },
+           py::arg("bytes"), py::arg("i") = 0);
 }

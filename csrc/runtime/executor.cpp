@@ -10,18 +10,18 @@
 namespace trtlab {
 
 Engine::Engine(int device, const void* weights, size_t weight_bytes,
-               size_t arena_bytes, std::vector<OpDesc> ops, int64_t input_off,
-               size_t input_bytes, int64_t output_off, size_t output_bytes,
-               bool managed_weights)
+               size_t arena_bytes, std::vector<OpDesc> ops,
+               std::vector<BindingDesc> inputs,
+               std::vector<BindingDesc> outputs, bool managed_weights)
     : device_(device),
       weight_bytes_(weight_bytes),
       managed_weights_(managed_weights),
       arena_bytes_(arena_bytes),
       ops_(std::move(ops)),
-      input_off_(input_off),
-      output_off_(output_off),
-      input_bytes_(input_bytes),
-      output_bytes_(output_bytes) {
+      inputs_(std::move(inputs)),
+      outputs_(std::move(outputs)) {
+  if (inputs_.empty() || outputs_.empty())
+    throw std::runtime_error("Engine needs >=1 input and >=1 output binding");
   TRT_HIP_CHECK(hipSetDevice(device_));
   size_t wb = weight_bytes_ ? weight_bytes_ : 256;
   if (managed_weights_) {
@@ -79,8 +79,18 @@ ExecutionContext::ExecutionContext(std::shared_ptr<Engine> engine)
   arena_ = (char*)device_malloc(eng_->arena_bytes(), eng_->device());
   if (eng_->scratch_bytes())
     scratch_ = (char*)device_malloc(eng_->scratch_bytes(), eng_->device());
-  h_in_ = (char*)pinned_malloc(eng_->input_bytes());
-  h_out_ = (char*)pinned_malloc(eng_->output_bytes());
+  // Carve one pinned slab per direction into per-binding regions, each
+  // 256-B aligned (reference Buffers::CreateBindings carving pattern).
+  for (const BindingDesc& b : eng_->inputs()) {
+    in_hoff_.push_back(h_in_bytes_);
+    h_in_bytes_ += round_up(b.bytes, 256);
+  }
+  for (const BindingDesc& b : eng_->outputs()) {
+    out_hoff_.push_back(h_out_bytes_);
+    h_out_bytes_ += round_up(b.bytes, 256);
+  }
+  h_in_ = (char*)pinned_malloc(h_in_bytes_);
+  h_out_ = (char*)pinned_malloc(h_out_bytes_);
   TRT_HIP_CHECK(hipStreamCreateWithFlags(&stream_, hipStreamNonBlocking));
 }
 
@@ -92,8 +102,8 @@ ExecutionContext::~ExecutionContext() {
   hipStreamDestroy(stream_);
   device_free(arena_, eng_->arena_bytes());
   if (scratch_) device_free(scratch_, eng_->scratch_bytes());
-  pinned_free(h_in_, eng_->input_bytes());
-  pinned_free(h_out_, eng_->output_bytes());
+  pinned_free(h_in_, h_in_bytes_);
+  pinned_free(h_out_, h_out_bytes_);
 }
 
 void ExecutionContext::enqueue_all(hipStream_t s) {
@@ -110,8 +120,10 @@ void ExecutionContext::enqueue_all(hipStream_t s) {
   if (timing_ && !ev_[0])
     for (auto& e : ev_) TRT_HIP_CHECK(hipEventCreate(&e));
   if (timing_) TRT_HIP_CHECK(hipEventRecord(ev_[0], s));
-  TRT_HIP_CHECK(hipMemcpyAsync(ar + eng_->input_off(), h_in_,
-                               eng_->input_bytes(), hipMemcpyHostToDevice, s));
+  const auto& ins = eng_->inputs();
+  for (size_t i = 0; i < ins.size(); ++i)
+    TRT_HIP_CHECK(hipMemcpyAsync(ar + ins[i].off, h_in_ + in_hoff_[i],
+                                 ins[i].bytes, hipMemcpyHostToDevice, s));
   if (timing_) TRT_HIP_CHECK(hipEventRecord(ev_[1], s));
 
   for (const OpDesc& op : eng_->ops()) {
@@ -225,8 +237,10 @@ void ExecutionContext::enqueue_all(hipStream_t s) {
   }
 
   if (timing_) TRT_HIP_CHECK(hipEventRecord(ev_[2], s));
-  TRT_HIP_CHECK(hipMemcpyAsync(h_out_, ar + eng_->output_off(),
-                               eng_->output_bytes(), hipMemcpyDeviceToHost, s));
+  const auto& outs = eng_->outputs();
+  for (size_t i = 0; i < outs.size(); ++i)
+    TRT_HIP_CHECK(hipMemcpyAsync(h_out_ + out_hoff_[i], ar + outs[i].off,
+                                 outs[i].bytes, hipMemcpyDeviceToHost, s));
   if (timing_) TRT_HIP_CHECK(hipEventRecord(ev_[3], s));
 }
 

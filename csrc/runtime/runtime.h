@@ -92,6 +92,14 @@ struct OpDesc {
   int tile = 0;
 };
 
+// One engine I/O binding: a named arena region mirrored by pinned host
+// staging in each ExecutionContext. The reference carves N host+device
+// addresses per model the same way (trtlab/tensorrt/bindings.h:60-120).
+struct BindingDesc {
+  int64_t off = -1;   // arena offset in bytes
+  size_t bytes = 0;   // binding size
+};
+
 // A compiled model: weight blob on device + op list + arena layout.
 // Replaces the reference's Model/ICudaEngine (trtlab/tensorrt/model.h:17):
 // here the "engine" is an explicit op plan over hand-written CDNA4 kernels.
@@ -103,18 +111,21 @@ class Engine {
   // (trtlab/tensorrt/src/allocator.cc:12-56). Default off: explicit device
   // memory is faster and the MI355X has 288 GB of HBM3E per GPU.
   Engine(int device, const void* weights, size_t weight_bytes,
-         size_t arena_bytes, std::vector<OpDesc> ops, int64_t input_off,
-         size_t input_bytes, int64_t output_off, size_t output_bytes,
+         size_t arena_bytes, std::vector<OpDesc> ops,
+         std::vector<BindingDesc> inputs, std::vector<BindingDesc> outputs,
          bool managed_weights = false);
   ~Engine();
 
   int device() const { return device_; }
   size_t arena_bytes() const { return arena_bytes_; }
   size_t scratch_bytes() const { return scratch_bytes_; }
-  size_t input_bytes() const { return input_bytes_; }
-  size_t output_bytes() const { return output_bytes_; }
-  int64_t input_off() const { return input_off_; }
-  int64_t output_off() const { return output_off_; }
+  const std::vector<BindingDesc>& inputs() const { return inputs_; }
+  const std::vector<BindingDesc>& outputs() const { return outputs_; }
+  // Primary (first-input / first-output) conveniences.
+  size_t input_bytes() const { return inputs_[0].bytes; }
+  size_t output_bytes() const { return outputs_[0].bytes; }
+  int64_t input_off() const { return inputs_[0].off; }
+  int64_t output_off() const { return outputs_[0].off; }
   const std::vector<OpDesc>& ops() const { return ops_; }
   const char* weights() const { return (const char*)weights_; }
   const char* zero_page() const { return (const char*)zero_page_; }
@@ -132,8 +143,7 @@ class Engine {
   size_t arena_bytes_;
   size_t scratch_bytes_ = 0;  // split-K slab workspace (max over ops)
   std::vector<OpDesc> ops_;
-  int64_t input_off_, output_off_;
-  size_t input_bytes_, output_bytes_;
+  std::vector<BindingDesc> inputs_, outputs_;
 };
 
 // Per-request execution context: private stream + activation arena + pinned
@@ -145,9 +155,14 @@ class ExecutionContext {
   ExecutionContext(std::shared_ptr<Engine> engine);
   ~ExecutionContext();
 
-  // Raw host staging buffers (pinned) for input/output bindings.
-  uintptr_t host_input_ptr() const { return (uintptr_t)h_in_; }
-  uintptr_t host_output_ptr() const { return (uintptr_t)h_out_; }
+  // Raw host staging buffers (pinned), one per binding. Index 0 keeps the
+  // single-binding API shape.
+  uintptr_t host_input_ptr(int i = 0) const {
+    return (uintptr_t)(h_in_ + in_hoff_[i]);
+  }
+  uintptr_t host_output_ptr(int i = 0) const {
+    return (uintptr_t)(h_out_ + out_hoff_[i]);
+  }
   uintptr_t arena_ptr() const { return (uintptr_t)arena_; }
 
   void enqueue_all(hipStream_t s);  // H2D + ops + D2H on stream s
@@ -166,8 +181,13 @@ class ExecutionContext {
   std::shared_ptr<Engine> eng_;
   char* arena_ = nullptr;
   char* scratch_ = nullptr;
+  // One pinned slab per direction, carved per binding (in_hoff_[i] = host
+  // offset of binding i) — the reference's Buffers/Bindings carving
+  // (trtlab/tensorrt/src/buffers.cc, bindings.h:60-120).
   char* h_in_ = nullptr;
   char* h_out_ = nullptr;
+  std::vector<size_t> in_hoff_, out_hoff_;
+  size_t h_in_bytes_ = 0, h_out_bytes_ = 0;
   hipStream_t stream_{};
   hipGraph_t graph_{};
   hipGraphExec_t graph_exec_{};

@@ -51,6 +51,12 @@ class Graph:
         self.name = name
         self.nodes: List[Node] = []
         self.tensors: Dict[str, TensorSpec] = {}
+        # N-binding I/O (reference bindings.h:60-120 carves N host+device
+        # addresses per model): input_names in declaration order;
+        # output_names pinned via mark_output() (default: the last op's
+        # output). input_name/output_name remain the primary bindings.
+        self.input_names: List[str] = []
+        self.output_names: List[str] = []
         self.input_name: Optional[str] = None
         self.output_name: Optional[str] = None
         self._ctr = 0
@@ -68,7 +74,19 @@ class Graph:
               dtype: str = "f16") -> str:
         self.add_tensor(name, shape, dtype)
         self.nodes.append(Node("input", name, [], name))
-        self.input_name = name
+        if self.input_name is None:
+            self.input_name = name  # primary binding = first declared
+        self.input_names.append(name)
+        return name
+
+    def mark_output(self, name: str) -> str:
+        """Pin a tensor as an engine output binding (in addition to the
+        default last-op output). Order of mark_output calls = binding
+        order; the primary output stays output_name unless marked first."""
+        if name not in self.tensors:
+            raise KeyError(f"mark_output: unknown tensor {name}")
+        if name not in self.output_names:
+            self.output_names.append(name)
         return name
 
     def _emit(self, kind: str, inputs: List[str], out_shape: Tuple[int, ...],
@@ -158,12 +176,16 @@ class Graph:
 
     def embedding(self, ids: str, tok_table: np.ndarray,
                   pos_table: np.ndarray, seg_table=None,
+                  segids: Optional[str] = None,
                   name: Optional[str] = None) -> str:
+        """out[m] = tok[ids[m]] + pos[m % S] (+ seg[segids[m]]). segids:
+        optional i32 input-tensor name (BERT token_type_ids binding)."""
         m = self.tensors[ids].shape[0]
         v, h = tok_table.shape
         s_, h2 = pos_table.shape
         assert h == h2
-        return self._emit("embedding", [ids], (m, h),
+        ins = [ids] if segids is None else [ids, segids]
+        return self._emit("embedding", ins, (m, h),
                           dict(tok=np.asarray(tok_table, np.float32),
                                pos=np.asarray(pos_table, np.float32),
                                seg=None if seg_table is None

@@ -32,6 +32,8 @@ def save_plan(plan: EnginePlan, path: str) -> None:
         dtype=plan.dtype,
         shapes={k: list(v) for k, v in plan.shapes.items()},
         input_dtype=plan.input_dtype,
+        inputs=[{**b, "shape": list(b["shape"])} for b in plan.inputs],
+        outputs=[{**b, "shape": list(b["shape"])} for b in plan.outputs],
         exec_meta=[dict(kind=o.kind, name=o.name, inputs=o.inputs,
                         output=o.output,
                         params={k: v for k, v in o.params.items()
@@ -93,4 +95,8 @@ def load_plan(path: str) -> EnginePlan:
         dtype=meta["dtype"],
         shapes={k: tuple(v) for k, v in meta["shapes"].items()},
         input_dtype=meta.get("input_dtype", "f16"),
+        inputs=[{**b, "shape": tuple(b["shape"])}
+                for b in meta.get("inputs", [])],
+        outputs=[{**b, "shape": tuple(b["shape"])}
+                 for b in meta.get("outputs", [])],
     )

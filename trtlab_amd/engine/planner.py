@@ -96,6 +96,24 @@ class EnginePlan:
     dtype: int = DT_F16
     shapes: Dict[str, Tuple[int, ...]] = field(default_factory=dict)
     input_dtype: str = "f16"
+    # N-binding I/O (reference bindings.h:60-120): ordered binding dicts
+    # {name, off, bytes, shape, dtype}; [0] is the primary binding the
+    # legacy single-binding fields mirror.
+    inputs: List[Dict[str, Any]] = field(default_factory=list)
+    outputs: List[Dict[str, Any]] = field(default_factory=list)
+
+    def __post_init__(self):
+        if not self.inputs:
+            self.inputs = [dict(name=self.input_name, off=self.input_off,
+                                bytes=self.input_bytes,
+                                shape=tuple(self.input_shape),
+                                dtype=self.input_dtype)]
+        if not self.outputs:
+            self.outputs = [dict(name=self.output_name, off=self.output_off,
+                                 bytes=self.output_bytes,
+                                 shape=tuple(self.output_shape),
+                                 dtype="bf16" if self.dtype == DT_BF16
+                                 else "f16")]
 
 
 class Planner:
@@ -163,7 +181,8 @@ class Planner:
                 op.bias = n.attrs["beta"].astype(np.float32)
                 exec_ops.append(op)
             elif n.kind == "embedding":
-                op = ExecOp(K_EMBEDDING, n.name, [n.inputs[0]], n.output,
+                # inputs = [ids] or [ids, segids] (segids -> in2_off)
+                op = ExecOp(K_EMBEDDING, n.name, list(n.inputs), n.output,
                             dict(seq=n.attrs["seq"]))
                 op.w = n.attrs["tok"].astype(np.float16)
                 op.scale = n.attrs["pos"].astype(np.float16)
@@ -327,19 +346,27 @@ class Planner:
         att_varlen = [op for op in exec_ops
                       if op.kind == K_ATTENTION and op.params.get("varlen")]
         if att_varlen:
-            if g.tensors[input_name].dtype != "i32":
+            # lengths come from a dedicated attention_mask binding when the
+            # graph has one (count of non-zero entries == pad_id 0 scan),
+            # else from the token-id input vs pad_id
+            if "attention_mask" in g.input_names:
+                lens_src = "attention_mask"
+                pad_id = 0
+            else:
+                lens_src = input_name
+                pad_id = att_varlen[0].params.get("pad_id", 0)
+            if g.tensors[lens_src].dtype != "i32":
                 raise ValueError(
-                    "varlen attention requires an i32 token-id graph input "
-                    "(build the model with embeddings=True)")
+                    "varlen attention requires an i32 token-id or "
+                    "attention_mask graph input")
             seq = att_varlen[0].params["seq"]
-            bsz = shapes[input_name][0] // seq
+            bsz = shapes[lens_src][0] // seq
             lens_name = "_seqlens"
             shapes[lens_name] = (bsz,)
             itemsize[lens_name] = 4
             exec_ops.insert(0, ExecOp(
-                K_SEQLENS, lens_name, [input_name], lens_name,
-                dict(B=bsz, S=seq,
-                     pad_id=att_varlen[0].params.get("pad_id", 0))))
+                K_SEQLENS, lens_name, [lens_src], lens_name,
+                dict(B=bsz, S=seq, pad_id=pad_id)))
             for op in att_varlen:
                 op.inputs.append(lens_name)
         if self.dtype in (DT_MX4, DT_MX8):
@@ -416,7 +443,9 @@ class Planner:
             else:
                 tensors_used[t] = (i, i)
 
-        touch(input_name, 0)
+        input_names = list(g.input_names) or [input_name]
+        for t in input_names:
+            touch(t, 0)  # all input bindings live from the start (H2D)
         for i, op in enumerate(exec_ops):
             for t in op.inputs:
                 touch(t, i)
@@ -424,9 +453,14 @@ class Planner:
             if "q_out" in op.params:  # fused fp8 second output
                 touch(op.params["q_out"], i)
         output_name = exec_ops[-1].output
-        # input live from the start; output live to the end
-        s, e = tensors_used[output_name]
-        tensors_used[output_name] = (s, len(exec_ops))
+        # N output bindings: the final op's output is the primary binding
+        # [0]; tensors pinned by g.mark_output follow. All live to the end
+        # (each gets a D2H copy).
+        output_names = [output_name] + [t for t in g.output_names
+                                        if t != output_name]
+        for t in output_names:
+            s, e = tensors_used[t]
+            tensors_used[t] = (s, len(exec_ops))
 
         def nbytes_of(t: str) -> int:
             n = 1
@@ -539,6 +573,16 @@ class Planner:
                 raise ValueError(f"bad exec op kind {op.kind}")
             op_dicts.append(d)
 
+        # bf16 plans read float bindings as bf16 bits (the host side
+        # converts); integer inputs (BERT token ids) stay as declared
+        def bind_dtype(t: str) -> str:
+            dt = g.tensors[t].dtype if t in g.tensors else "f16"
+            return "bf16" if self.dtype == DT_BF16 and dt == "f16" else dt
+
+        def binding(t: str) -> Dict[str, Any]:
+            return dict(name=t, off=offsets[t], bytes=nbytes_of(t),
+                        shape=tuple(shapes[t]), dtype=bind_dtype(t))
+
         return EnginePlan(
             name=g.name,
             ops=op_dicts,
@@ -556,9 +600,7 @@ class Planner:
             output_shape=shapes[output_name],
             dtype=self.dtype,
             shapes=dict(shapes),
-            # bf16 plans read the input buffer as bf16 bits (the host side
-            # converts); integer inputs (BERT token ids) stay as declared
-            input_dtype=("bf16" if self.dtype == DT_BF16 and
-                         g.tensors[input_name].dtype == "f16"
-                         else g.tensors[input_name].dtype),
+            input_dtype=bind_dtype(input_name),
+            inputs=[binding(t) for t in input_names],
+            outputs=[binding(t) for t in output_names],
         )

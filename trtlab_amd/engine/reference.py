@@ -44,13 +44,18 @@ def _epi(acc: torch.Tensor, epi: int, scale, bias, res) -> torch.Tensor:
     return v
 
 
-def run_reference(plan: EnginePlan, input_nhwc: np.ndarray,
-                  return_all: bool = False):
-    """Run the plan on CPU (fp32). input is the RAW (unpadded) input.
-    With return_all=True, returns the dict of ALL tensors (debugging)."""
-    t: Dict[str, torch.Tensor] = {
-        plan.input_name: torch.from_numpy(np.ascontiguousarray(input_nhwc)).float()
-    }
+def run_reference(plan: EnginePlan, input_nhwc, return_all: bool = False):
+    """Run the plan on CPU (fp32). input is the RAW (unpadded) primary
+    input, or a {name: array} dict for multi-binding models. With
+    return_all=True, returns the dict of ALL tensors (debugging)."""
+    if isinstance(input_nhwc, dict):
+        t: Dict[str, torch.Tensor] = {
+            k: torch.from_numpy(np.ascontiguousarray(v)).float()
+            for k, v in input_nhwc.items()
+        }
+    else:
+        t = {plan.input_name:
+             torch.from_numpy(np.ascontiguousarray(input_nhwc)).float()}
     # shapes registry from the planner's op dicts + exec op metadata
     for op, d in zip(plan.exec_ops, plan.ops):
         x = t[op.inputs[0]]
@@ -92,7 +97,11 @@ def run_reference(plan: EnginePlan, input_nhwc: np.ndarray,
             m = ids.shape[0]
             y = tok[ids] + pos[torch.arange(m) % d["S"]]
             if op.bias is not None:
-                y = y + torch.from_numpy(op.bias.astype(np.float32))[0]
+                seg = torch.from_numpy(op.bias.astype(np.float32))
+                if len(op.inputs) > 1:  # segids binding (token_type_ids)
+                    y = y + seg[t[op.inputs[1]].long()]
+                else:
+                    y = y + seg[0]
             t[op.output] = y
         elif op.kind == K_QUANTIZE:
             t[op.output] = torch.clamp(torch.round(x / d["q_scale"]),

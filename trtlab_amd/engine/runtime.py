@@ -46,8 +46,9 @@ class NativeEngine:
         self.plan = plan
         self.device = device
         self.engine = self._C.Engine(
-            device, plan.weights, plan.arena_bytes, plan.ops, plan.input_off,
-            plan.input_bytes, plan.output_off, plan.output_bytes,
+            device, plan.weights, plan.arena_bytes, plan.ops,
+            [(b["off"], b["bytes"]) for b in plan.inputs],
+            [(b["off"], b["bytes"]) for b in plan.outputs],
             managed_weights=managed_weights)
 
     def upload_weights(self, blob: np.ndarray) -> None:
@@ -73,18 +74,29 @@ class NativeContext:
             self.ctx.set_timing(True)
         # bf16 plans carry bf16 bit patterns in the bindings (numpy has no
         # bf16 dtype); infer() converts via torch at the edges
-        in_name = getattr(self.plan, "input_dtype", "f16")
-        self._in_bf16 = in_name == "bf16"
-        self._out_bf16 = getattr(self.plan, "dtype", 0) == 1  # DT_BF16
-        in_dt = {"f16": np.float16, "bf16": np.int16, "i32": np.int32,
-                 "f32": np.float32}[in_name]
-        self._in_view = np.frombuffer(
-            self.ctx.input_view(self.plan.input_bytes), dtype=in_dt
-        ).reshape(self.plan.input_shape)
-        self._out_view = np.frombuffer(
-            self.ctx.output_view(self.plan.output_bytes),
-            dtype=np.int16 if self._out_bf16 else np.float16
-        ).reshape(self.plan.output_shape)
+        np_dt = {"f16": np.float16, "bf16": np.int16, "i32": np.int32,
+                 "f32": np.float32, "i8": np.int8}
+
+        # N named bindings, pinned zero-copy views (reference Bindings
+        # carving, bindings.h:60-120); index 0 = primary binding.
+        self.inputs: Dict[str, np.ndarray] = {}
+        self._in_bf16_map: Dict[str, bool] = {}
+        for i, b in enumerate(self.plan.inputs):
+            v = np.frombuffer(self.ctx.input_view(b["bytes"], i),
+                              dtype=np_dt[b["dtype"]]).reshape(b["shape"])
+            self.inputs[b["name"]] = v
+            self._in_bf16_map[b["name"]] = b["dtype"] == "bf16"
+        self.outputs: Dict[str, np.ndarray] = {}
+        for i, b in enumerate(self.plan.outputs):
+            v = np.frombuffer(self.ctx.output_view(b["bytes"], i),
+                              dtype=np_dt[b["dtype"]]).reshape(b["shape"])
+            self.outputs[b["name"]] = v
+
+        in0 = self.plan.inputs[0]["name"]
+        self._in_bf16 = self._in_bf16_map[in0]
+        self._out_bf16 = self.plan.outputs[0]["dtype"] == "bf16"
+        self._in_view = self.inputs[in0]
+        self._out_view = self.outputs[self.plan.outputs[0]["name"]]
         if capture:
             self.ctx.capture()
 
@@ -98,27 +110,35 @@ class NativeContext:
         """Pinned host output binding (valid after synchronize)."""
         return self._out_view
 
-    def write_input(self, batch: np.ndarray) -> None:
-        """Copy a host batch into the pinned input binding, handling the
-        bf16 bit-pattern representation (numpy has no bf16 dtype — float
-        inputs are converted to bf16 BITS, never numerically cast to int16)
-        and validating shape up front so malformed requests fail loudly."""
-        if tuple(batch.shape) != tuple(self._in_view.shape):
+    def write_input(self, batch, name: Optional[str] = None) -> None:
+        """Copy a host batch into a pinned input binding (primary binding
+        by default; pass `name` or a {name: array} dict for multi-input
+        models). Handles the bf16 bit-pattern representation (numpy has no
+        bf16 dtype — float inputs are converted to bf16 BITS, never
+        numerically cast to int16) and validates shape up front so
+        malformed requests fail loudly."""
+        if isinstance(batch, dict):
+            for k, v in batch.items():
+                self.write_input(v, name=k)
+            return
+        view = self._in_view if name is None else self.inputs[name]
+        bf16 = self._in_bf16 if name is None else self._in_bf16_map[name]
+        if tuple(batch.shape) != tuple(view.shape):
             raise ValueError(
                 f"input shape {tuple(batch.shape)} != plan shape "
-                f"{tuple(self._in_view.shape)}")
-        if self._in_bf16 and batch.dtype != np.int16:
+                f"{tuple(view.shape)} for binding "
+                f"{name or self.plan.inputs[0]['name']}")
+        if bf16 and batch.dtype != np.int16:
             from trtlab_amd.engine.planner import _bf16_bits
 
-            np.copyto(self._in_view,
-                      _bf16_bits(batch).reshape(self._in_view.shape))
+            np.copyto(view, _bf16_bits(batch).reshape(view.shape))
         else:
-            np.copyto(self._in_view,
-                      batch.astype(self._in_view.dtype, copy=False))
+            np.copyto(view, batch.astype(view.dtype, copy=False))
 
-    def infer(self, batch: Optional[np.ndarray] = None) -> np.ndarray:
-        """Synchronous convenience path; returns a COPY of the output (the
-        zero-copy `.output` view is only valid while this context lives)."""
+    def infer(self, batch=None):
+        """Synchronous convenience path; returns a COPY of the primary
+        output (the zero-copy `.output` view is only valid while this
+        context lives). Multi-output models: use infer_all()."""
         if batch is not None:
             self.write_input(batch)
         self.ctx.launch()
@@ -130,6 +150,15 @@ class NativeContext:
             out = torch.from_numpy(out).view(torch.bfloat16).to(
                 torch.float32).numpy()
         return out
+
+    def infer_all(self, batch=None) -> Dict[str, np.ndarray]:
+        """Run one forward and return copies of ALL output bindings by
+        name (3-input/2-output models etc.)."""
+        if batch is not None:
+            self.write_input(batch)
+        self.ctx.launch()
+        self.ctx.synchronize()
+        return {k: np.array(v, copy=True) for k, v in self.outputs.items()}
 
     def launch(self):
         self.ctx.launch()

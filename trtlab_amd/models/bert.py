@@ -17,15 +17,26 @@ def build_bert(batch: int = 8, seq: int = 128, hidden: int = 768,
                layers: int = 12, heads: int = 12, seed: int = 0,
                intermediate: int | None = None, embeddings: bool = False,
                varlen: bool = False, pad_id: int = 0,
-               vocab: int = 30522) -> Graph:
+               vocab: int = 30522, segments: bool = False,
+               mask_input: bool = False) -> Graph:
     """embeddings=True: input is int32 token ids [B*S]; the graph starts
     with an embedding gather (tok+pos) + LayerNorm. Otherwise the input is
     the pre-embedded hidden state [B*S, hidden] fp16.
     varlen=True (needs embeddings): right-padded variable-length batches —
     per-sequence valid lengths are derived on-device from ids != pad_id and
-    padded keys are masked out of every attention softmax."""
+    padded keys are masked out of every attention softmax.
+    segments=True: adds a segment_ids i32 input binding + learned
+    token-type table (BERT token_type_ids). mask_input=True: adds an
+    attention_mask i32 input binding that drives the varlen key masking
+    instead of the pad-id scan — together these express BERT with its REAL
+    (ids, mask, segments) bindings (reference bindings.h:60-120 carves one
+    device address per binding the same way)."""
     if varlen and not embeddings:
         raise ValueError("varlen requires embeddings=True (ids input)")
+    if (segments or mask_input) and not embeddings:
+        raise ValueError("segments/mask_input require embeddings=True")
+    if mask_input and not varlen:
+        raise ValueError("mask_input requires varlen=True")
     inter = intermediate or hidden * 4
     rng = np.random.RandomState(seed)
 
@@ -44,9 +55,16 @@ def build_bert(batch: int = 8, seq: int = 128, hidden: int = 768,
     m = batch * seq
     if embeddings:
         ids = g.input((m,), name="token_ids", dtype="i32")
+        segids = (g.input((m,), name="segment_ids", dtype="i32")
+                  if segments else None)
+        if mask_input:
+            g.input((m,), name="attention_mask", dtype="i32")
         tok = (rng.randn(vocab, hidden) * 0.02).astype(np.float32)
         pos = (rng.randn(seq, hidden) * 0.02).astype(np.float32)
-        emb = g.embedding(ids, tok, pos, name="embed")
+        seg = ((rng.randn(2, hidden) * 0.02).astype(np.float32)
+               if segments else None)
+        emb = g.embedding(ids, tok, pos, seg_table=seg, segids=segids,
+                          name="embed")
         ge, be = ln(hidden)
         h = g.layernorm(emb, ge, be, name="embed_ln")
     else:

@@ -87,7 +87,16 @@ class Communicator:
     def __init__(self, rank: Optional[int] = None,
                  world: Optional[int] = None, device: Optional[int] = None,
                  uid: Optional[bytes] = None,
-                 rendezvous_path: Optional[str] = None):
+                 rendezvous_path: Optional[str] = None,
+                 force: bool = False):
+        """force=True builds the native RCCL communicator even at world
+        size 1 (a real self-clique: ncclCommInitRank + collectives execute
+        on the GPU). Default world-1 behavior is no-op passthrough.
+
+        NOTE: this RCCL build rejects two ranks on ONE device
+        ("Duplicate GPU detected", ncclInvalidUsage) — a genuine N>1
+        clique needs N distinct GPUs; the world-1 forced clique is the
+        deepest single-GPU proof available."""
         from trtlab_amd import native
 
         C = native()
@@ -100,13 +109,14 @@ class Communicator:
             if ndev > 0:
                 device = device % ndev  # multi-rank single-GPU proof mode
         self.device = device
-        if self.world <= 1:
+        if self.world <= 1 and not force:
             self._comm = None
             return
         if uid is None:
             uid = C.comm.unique_id() if self.rank == 0 else None
-            uid = exchange_unique_id(self.rank, self.world, uid,
-                                     path=rendezvous_path)
+            if self.world > 1:
+                uid = exchange_unique_id(self.rank, self.world, uid,
+                                         path=rendezvous_path)
         self._comm = C.comm.Communicator(self.rank, self.world, uid,
                                          self.device)
 
@@ -163,8 +173,8 @@ def broadcast_weights(engine, comm: Communicator, src_rank: int = 0) -> None:
     weight memory — no staging tensor, no torch (bigger transfers amortize
     per-link xGMI latency better than per-tensor messages).
     """
-    if comm is None or comm.world <= 1:
-        return
+    if comm is None or comm._comm is None:
+        return  # world-1 no-op comm (a forced self-clique still broadcasts)
     comm.broadcast(engine.engine.weights_ptr, engine.engine.weight_bytes,
                    root=src_rank)
     comm.synchronize()
