@@ -283,7 +283,11 @@ class InferRunner:
                 out = None
                 try:
                     co.item.synchronize()
-                    out = np.array(co.item.output, copy=True)
+                    if len(co.item.plan.outputs) > 1:
+                        out = {k: np.array(v, copy=True)
+                               for k, v in co.item.outputs.items()}
+                    else:
+                        out = np.array(co.item.output, copy=True)
                 finally:
                     co.release()
                 fut.set_result(out)

@@ -6,7 +6,8 @@ built on grpc.aio completion-queue machinery. Messages are defined with
 dynamic protobuf descriptors (no protoc needed offline) in rpc.proto.
 """
 from trtlab_amd.rpc.proto import (EchoRequest, EchoResponse, InferRequest,  # noqa: F401
-                                  InferResponse, HealthRequest, HealthResponse)
+                                  InferResponse, HealthRequest,
+                                  HealthResponse, NamedTensor)
 from trtlab_amd.rpc.server import (AsyncService, BatchingService, Server,  # noqa: F401
                                    StreamingService, UnaryService)
 from trtlab_amd.rpc.client import (AsyncClient, ShmInput, SyncClient,  # noqa: F401

@@ -48,6 +48,25 @@ _msg("EchoResponse", [("message", _T.TYPE_STRING, False),
 _msg("HealthRequest", [])
 _msg("HealthResponse", [("ready", _T.TYPE_BOOL, False),
                         ("status", _T.TYPE_STRING, False)])
+# Named tensor for multi-binding models (reference Bindings carve N
+# addresses per model, bindings.h:60-120; TRTIS InferRequestHeader names
+# its inputs the same way).
+_msg("NamedTensor", [("name", _T.TYPE_STRING, False),
+                     ("data", _T.TYPE_BYTES, False),
+                     ("shape", _T.TYPE_INT64, True),
+                     ("dtype", _T.TYPE_STRING, False)])
+
+
+def _msg_field(m_name, fname, num, type_name):
+    m = next(m for m in _f.message_type if m.name == m_name)
+    fd = m.field.add()
+    fd.name = fname
+    fd.number = num
+    fd.type = _T.TYPE_MESSAGE
+    fd.type_name = f".trtlab.{type_name}"
+    fd.label = _T.LABEL_REPEATED
+
+
 _msg("InferRequest", [("model", _T.TYPE_STRING, False),
                       ("input", _T.TYPE_BYTES, False),
                       ("shape", _T.TYPE_INT64, True),
@@ -64,6 +83,9 @@ _msg("InferResponse", [("output", _T.TYPE_BYTES, False),
                        ("batch_id", _T.TYPE_INT64, False),
                        ("compute_ms", _T.TYPE_FLOAT, False),
                        ("request_ms", _T.TYPE_FLOAT, False)])
+# field 8: repeated NamedTensor inputs; field 7: repeated NamedTensor outputs
+_msg_field("InferRequest", "inputs", 8, "NamedTensor")
+_msg_field("InferResponse", "outputs", 7, "NamedTensor")
 
 _file_desc = _pool.Add(_f)
 
@@ -73,6 +95,7 @@ def _cls(name):
 
 
 EchoRequest = _cls("EchoRequest")
+NamedTensor = _cls("NamedTensor")
 EchoResponse = _cls("EchoResponse")
 HealthRequest = _cls("HealthRequest")
 HealthResponse = _cls("HealthResponse")

@@ -63,7 +63,15 @@ class InferenceService(AsyncService):
         plan = resources.manager.get_model(request.model).plan
         dtype = np.float16 if request.dtype in ("", "f16", "float16") else np.dtype(request.dtype)
         shape = tuple(request.shape) or plan.input_shape
-        if request.shm_name:
+        if request.inputs:
+            # named multi-binding request (reference carves N addresses per
+            # model; TRTIS names request inputs the same way)
+            np_dt = {"f16": np.float16, "i32": np.int32, "f32": np.float32,
+                     "bf16": np.int16, "i8": np.int8, "": np.float16}
+            batch = {
+                t.name: np.frombuffer(t.data, dtype=np_dt[t.dtype]).reshape(
+                    tuple(t.shape)) for t in request.inputs}
+        elif request.shm_name:
             # zero-copy local transport: the tensor lives in POSIX shared
             # memory (reference's SysV shm input path, 02 server.cc:159).
             # Mappings are cached by name: clients pool and reuse segments
@@ -101,6 +109,18 @@ class InferenceService(AsyncService):
         request_ms = (time.monotonic() - t_start) * 1e3
         if self.metrics:
             self.metrics.observe(compute_ms, request_ms)
+        if isinstance(out, dict):  # multi-output model -> named tensors
+            from trtlab_amd.rpc.proto import NamedTensor
+
+            prim = out[plan.outputs[0]["name"]]
+            return InferResponse(
+                output=prim.tobytes(), shape=list(prim.shape), dtype="f16",
+                batch_id=request.batch_id, compute_ms=compute_ms,
+                request_ms=request_ms,
+                outputs=[NamedTensor(name=k, data=v.tobytes(),
+                                     shape=list(v.shape),
+                                     dtype=str(v.dtype).replace("float", "f"))
+                         for k, v in out.items()])
         return InferResponse(
             output=out.tobytes(), shape=list(out.shape), dtype="f16",
             batch_id=request.batch_id, compute_ms=compute_ms,
