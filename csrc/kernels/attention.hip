@@ -13,8 +13,13 @@
 //   wave-local running softmax (rows on 16-lane groups; shfl_xor),
 //   P staged to per-wave LDS, PV accumulated against V^T (staged
 //   transposed at load), output rescaled by exp(m_old - m_new).
-// fp32 accumulation; scores scaled by 1/sqrt(D). Constraints: D == 64,
-// S % 128 == 0.
+// fp32 accumulation; scores scaled by 1/sqrt(D).
+//
+// Generality: S is ARBITRARY (tail query rows clamp
+// their loads and skip their stores; tail key tiles clamp loads and mask
+// scores >= S), and D in {64, 128} — the D=128 variant below retiles to
+// 64-key tiles so Q[64][128] + K[64][128] + Vt[128][64] + P fit in 56 KiB
+// of LDS.
 //
 // Masks: seqlens (optional, [B] device ints) masks right-padded keys
 // >= seqlens[b] (variable-length batches; key tiles past the valid length
@@ -41,7 +46,7 @@ __global__ __launch_bounds__(256) void attention_kernel(
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
-  const int qblocks = S >> 6;      // 64-query-row blocks per (b, h)
+  const int qblocks = (S + 63) >> 6;  // 64-query-row blocks per (b, h)
   const int blk = blockIdx.x;
   const int bh = blk / qblocks;
   const int q0 = (blk % qblocks) * 64;  // this block's query-row base
@@ -60,12 +65,15 @@ __global__ __launch_bounds__(256) void attention_kernel(
   };
 
   // ---- stage Q once (64 rows x 8 chunks = 512 chunks, 2/thread) ----
+  // tail blocks clamp out-of-range query rows (their scores are computed
+  // but never stored)
 #pragma unroll
   for (int c = 0; c < 2; ++c) {
     int idx = c * 256 + tid;       // 0..511
     int row = idx >> 3;            // 0..63
     int cb = (idx & 7) * 16;
-    const T* src = base + (int64_t)(q0 + row) * row_stride + qoff + cb / 2;
+    int qr = q0 + row < S ? q0 + row : S - 1;
+    const T* src = base + (int64_t)qr * row_stride + qoff + cb / 2;
     *(short8v*)(Qs + swz(row, cb)) = *(const short8v*)src;
   }
 
@@ -103,13 +111,15 @@ __global__ __launch_bounds__(256) void attention_kernel(
       int idx = c * 256 + tid;
       int row = idx >> 3;            // 0..127
       int cb = (idx & 7) * 16;
+      int kr = t * 128 + row < S ? t * 128 + row : S - 1;  // tail clamp
       kv_k[c] = *(const short8v*)(
-          base + (int64_t)(t * 128 + row) * row_stride + koff + cb / 2);
+          base + (int64_t)kr * row_stride + koff + cb / 2);
       int ci = c * 256 + tid;        // 0..1023
       int key = ci >> 3;
       int dd = (ci & 7) * 8;
+      int vr = t * 128 + key < S ? t * 128 + key : S - 1;
       kv_v[c] = *(const short8v*)(
-          base + (int64_t)(t * 128 + key) * row_stride + voff + dd);
+          base + (int64_t)vr * row_stride + voff + dd);
     }
   };
   if (ntiles > 0) load_kv(0);
@@ -249,6 +259,228 @@ __global__ __launch_bounds__(256) void attention_kernel(
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       int row = q0 + qrow + ((lane >> 4) << 2) + r;
+      if (row >= S) continue;  // tail block: clamped rows are not stored
+      float inv = l_run[r] > 0.f ? 1.0f / l_run[r] : 0.f;
+      out[((int64_t)b * S + row) * hid + h * D + d] =
+          store_cast<OT>(oacc[j][r] * inv * out_scale);
+    }
+  }
+}
+
+// ---- D = 128 variant (head_dim-128 decoders / LLaMA-family shapes) ----
+// Retiled for the bigger head: 64-key tiles keep the whole working set in
+// 56 KiB LDS (Q[64][128] 16K + K[64][128] 16K + Vt[128][64] 16K + P 8K).
+// Same online softmax, masking, clamped-tail generality as the D=64 path.
+template <typename T, typename OT = T>
+__global__ __launch_bounds__(256) void attention_kernel_d128(
+    const T* __restrict__ qkv, OT* __restrict__ out, int B, int S, int H,
+    float scale, float out_scale, const int* __restrict__ seqlens,
+    int causal) {
+  constexpr int D = 128;
+  __shared__ __attribute__((aligned(16))) char smem[16384 * 3 + 8192];
+  char* Qs = smem;                   // [64 q][128 d], 256-B rows
+  char* Ks = smem + 16384;           // [64 k][128 d], 256-B rows
+  char* Vt = smem + 32768;           // [128 d][64 k], 128-B rows
+  char* Ps = smem + 49152;           // per wave [16 q][64 k], 128-B rows
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int qblocks = (S + 63) >> 6;
+  const int blk = blockIdx.x;
+  const int bh = blk / qblocks;
+  const int q0 = (blk % qblocks) * 64;
+  const int b = bh / H;
+  const int h = bh % H;
+  const int hid = H * D;
+  const int row_stride = 3 * hid;
+
+  const T* base = qkv + (int64_t)b * S * row_stride;
+  const int qoff = h * D;
+  const int koff = hid + h * D;
+  const int voff = 2 * hid + h * D;
+
+  // 256-B-row swizzle: 16 distinct 16-B offsets per 16 rows (bank period
+  // = 256 B on 64x4-B LDS banks) -> conflict-free frag reads and writes
+  auto swz128 = [](uint32_t row, uint32_t colbyte) {
+    return row * 256 + (colbyte ^ ((row & 15) << 4));
+  };
+  auto swz = [](uint32_t row, uint32_t colbyte) {  // 128-B rows (Vt, P)
+    return row * 128 + (colbyte ^ ((row & 7) << 4));
+  };
+
+  // ---- stage Q once: 64 rows x 16 chunks = 1024 chunks, 4/thread ----
+#pragma unroll
+  for (int c = 0; c < 4; ++c) {
+    int idx = c * 256 + tid;
+    int row = idx >> 4;              // 0..63
+    int cb = (idx & 15) * 16;
+    int qr = q0 + row < S ? q0 + row : S - 1;
+    *(short8v*)(Qs + swz128(row, cb)) =
+        *(const short8v*)(base + (int64_t)qr * row_stride + qoff + cb / 2);
+  }
+
+  int limit = S;
+  if (seqlens) {
+    limit = seqlens[b];
+    limit = limit < 1 ? 1 : (limit > S ? S : limit);
+  }
+  int ntiles = (limit + 63) >> 6;           // 64-key tiles
+  if (causal) {
+    int tmax = (q0 + 64 + 63) >> 6;
+    ntiles = ntiles < tmax ? ntiles : tmax;
+  }
+
+  using MF = Mfma16x16x32<T>;
+  const int qrow = wave * 16;
+  float m_run[4], l_run[4];
+  f32x4 oacc[8];
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    m_run[r] = -3.0e38f;
+    l_run[r] = 0.f;
+  }
+#pragma unroll
+  for (int j = 0; j < 8; ++j) oacc[j] = {0.f, 0.f, 0.f, 0.f};
+
+  short8v kv_k[4], kv_v[4];
+  auto load_kv = [&](int t) {
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+      int idx = c * 256 + tid;
+      int row = idx >> 4;            // 0..63 (key)
+      int cb = (idx & 15) * 16;
+      int kr = t * 64 + row < S ? t * 64 + row : S - 1;
+      kv_k[c] = *(const short8v*)(
+          base + (int64_t)kr * row_stride + koff + cb / 2);
+      int key = idx >> 4;
+      int dd = (idx & 15) * 8;
+      int vr = t * 64 + key < S ? t * 64 + key : S - 1;
+      kv_v[c] = *(const short8v*)(
+          base + (int64_t)vr * row_stride + voff + dd);
+    }
+  };
+  if (ntiles > 0) load_kv(0);
+  for (int t = 0; t < ntiles; ++t) {
+    __syncthreads();
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+      int idx = c * 256 + tid;
+      int row = idx >> 4;
+      int cb = (idx & 15) * 16;
+      *(short8v*)(Ks + swz128(row, cb)) = kv_k[c];
+      int key = idx >> 4;
+      int dd = (idx & 15) * 8;
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        *(T*)(Vt + swz(dd + j, key * 2)) = ((const T*)&kv_v[c])[j];
+    }
+    if (t + 1 < ntiles) load_kv(t + 1);
+    __syncthreads();
+
+    // ---- QK^T: 16 q rows x 64 keys, k-dim 128 (4 MFMA k-steps) ----
+    f32x4 sacc[4];
+#pragma unroll
+    for (int j = 0; j < 4; ++j) sacc[j] = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+    for (int ks = 0; ks < 4; ++ks) {
+      uint32_t kbyte = ks * 64 + ((lane >> 4) << 4);
+      typename MF::frag qf = *(const typename MF::frag*)(
+          Qs + swz128(qrow + (lane & 15), kbyte));
+      typename MF::frag kf[4];
+#pragma unroll
+      for (int f = 0; f < 4; ++f)
+        kf[f] = *(const typename MF::frag*)(
+            Ks + swz128(f * 16 + (lane & 15), kbyte));
+#pragma unroll
+      for (int j = 0; j < 4; ++j) sacc[j] = MF::run(qf, kf[j], sacc[j]);
+    }
+
+    // ---- masking (key = t*64 + j*16 + (lane & 15)) ----
+    if (limit - t * 64 < 64) {
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        if (t * 64 + j * 16 + (lane & 15) >= limit) {
+#pragma unroll
+          for (int r = 0; r < 4; ++r) sacc[j][r] = -3.0e38f;
+        }
+      }
+    }
+    if (causal && t * 64 + 63 > q0 + qrow) {
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        int key = t * 64 + j * 16 + (lane & 15);
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          int q = q0 + qrow + ((lane >> 4) << 2) + r;
+          if (key > q) sacc[j][r] = -3.0e38f;
+        }
+      }
+    }
+
+    // ---- online softmax ----
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      float m = m_run[r];
+#pragma unroll
+      for (int j = 0; j < 4; ++j) m = fmaxf(m, sacc[j][r] * scale);
+#pragma unroll
+      for (int off = 1; off < 16; off <<= 1)
+        m = fmaxf(m, __shfl_xor(m, off, 64));
+      float c = __expf(m_run[r] - m);
+      float s = 0.f;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        float e = __expf(sacc[j][r] * scale - m);
+        sacc[j][r] = e;
+        s += e;
+      }
+#pragma unroll
+      for (int off = 1; off < 16; off <<= 1) s += __shfl_xor(s, off, 64);
+      l_run[r] = l_run[r] * c + s;
+      m_run[r] = m;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) oacc[j][r] *= c;
+    }
+
+    // ---- P -> per-wave LDS [16 q][64 k] ----
+    char* P = Ps + wave * 2048;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      int col = j * 16 + (lane & 15);
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = ((lane >> 4) << 2) + r;
+        *(T*)(P + ((uint32_t)row * 128 + ((col * 2) ^ ((row & 7) << 4)))) =
+            (T)sacc[j][r];
+      }
+    }
+    __syncthreads();
+
+    // ---- PV: oacc[16 q][128 d] += P[16][64] @ Vt^T ----
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      uint32_t kbyte = ks * 64 + ((lane >> 4) << 4);
+      typename MF::frag pf =
+          *(const typename MF::frag*)(P + swz(lane & 15, kbyte));
+      typename MF::frag vf[8];
+#pragma unroll
+      for (int f = 0; f < 8; ++f)
+        vf[f] = *(const typename MF::frag*)(
+            Vt + swz(f * 16 + (lane & 15), kbyte));
+#pragma unroll
+      for (int j = 0; j < 8; ++j) oacc[j] = MF::run(pf, vf[j], oacc[j]);
+    }
+  }
+
+  // ---- store ----
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    int d = j * 16 + (lane & 15);
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int row = q0 + qrow + ((lane >> 4) << 2) + r;
+      if (row >= S) continue;
       float inv = l_run[r] > 0.f ? 1.0f / l_run[r] : 0.f;
       out[((int64_t)b * S + row) * hid + h * D + d] =
           store_cast<OT>(oacc[j][r] * inv * out_scale);
