@@ -188,10 +188,14 @@ PYBIND11_MODULE(_C, m) {
   ops.def("gemm_bt",
           [](int dtype, uintptr_t A, uintptr_t B, uintptr_t C, uintptr_t scale,
              uintptr_t bias, uintptr_t residual, int M, int N, int K, int epi,
-             uintptr_t stream, bool sync, int tile, float res_scale) {
+             uintptr_t stream, bool sync, int tile, float res_scale,
+             int64_t lda, int64_t ldb, int64_t ldc) {
+            // strided operands: lda/ldb/ldc in ELEMENTS (0 = contiguous
+            // [M,K]/[N,K]/[M,N]) — sub-matrix views / transposed outputs
             launch_gemm_bt(dtype, (void*)A, (void*)B, (void*)C, (float*)scale,
                            (float*)bias, (void*)residual, res_scale, M, N, K,
-                           K, K, N, epi, as_stream(stream), tile,
+                           lda ? lda : K, ldb ? ldb : K, ldc ? ldc : N, epi,
+                           as_stream(stream), tile,
                            test_scratch(gemm_scratch_bytes(M, N, K)));
             if (sync) TRT_HIP_CHECK(hipStreamSynchronize(as_stream(stream)));
           },
@@ -199,7 +203,8 @@ PYBIND11_MODULE(_C, m) {
           py::arg("scale") = 0, py::arg("bias") = 0, py::arg("residual") = 0,
           py::arg("M") = 0, py::arg("N") = 0, py::arg("K") = 0,
           py::arg("epi") = 0, py::arg("stream") = 0, py::arg("sync") = true,
-          py::arg("tile") = 0, py::arg("res_scale") = 1.0f);
+          py::arg("tile") = 0, py::arg("res_scale") = 1.0f,
+          py::arg("lda") = 0, py::arg("ldb") = 0, py::arg("ldc") = 0);
   ops.def("conv2d",
           [](int dtype, uintptr_t in, uintptr_t Wt, uintptr_t out,
              uintptr_t scale, uintptr_t bias, uintptr_t residual,

@@ -508,12 +508,31 @@ void launch_attention(int dtype, const void* qkv, void* out, int B, int S,
                       int H, int D, float scale, hipStream_t stream,
                       int out_dtype, float out_scale, const void* seqlens,
                       int causal) {
-  if (S % 128 != 0 || D != 64)
-    throw std::runtime_error(
-        "attention: S must be a multiple of 128 and D == 64");
-  dim3 grid(B * H * (S / 64));  // 64-query-row blocks
+  if (D != 64 && D != 128)
+    throw std::runtime_error("attention: head_dim must be 64 or 128");
+  if (S < 1) throw std::runtime_error("attention: S must be >= 1");
+  dim3 grid(B * H * ((S + 63) / 64));  // 64-query-row blocks (+ tail)
   dim3 block(256);
   const int* lens = (const int*)seqlens;
+  if (D == 128) {
+    if (dtype == 0) {
+      if (out_dtype == 3)
+        hipLaunchKernelGGL((attention_kernel_d128<_Float16, __hip_fp8_e4m3>),
+                           grid, block, 0, stream, (const _Float16*)qkv,
+                           (__hip_fp8_e4m3*)out, B, S, H, scale, out_scale,
+                           lens, causal);
+      else
+        hipLaunchKernelGGL((attention_kernel_d128<_Float16, _Float16>), grid,
+                           block, 0, stream, (const _Float16*)qkv,
+                           (_Float16*)out, B, S, H, scale, out_scale, lens,
+                           causal);
+    } else {
+      hipLaunchKernelGGL((attention_kernel_d128<__bf16, __bf16>), grid, block,
+                         0, stream, (const __bf16*)qkv, (__bf16*)out, B, S, H,
+                         scale, out_scale, lens, causal);
+    }
+    return;
+  }
   if (dtype == 0) {
     if (out_dtype == 3)  // fused fp8 output for the projection GEMM
       hipLaunchKernelGGL((attention_kernel<_Float16, __hip_fp8_e4m3>), grid,

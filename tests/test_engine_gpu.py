@@ -673,3 +673,65 @@ def test_bert_real_bindings_gpu():
     # sequence 0 is unaffected by sequence 1's padding
     assert np.allclose(out2.reshape(b, seq, -1)[0],
                        out.reshape(b, seq, -1)[0], atol=1e-2)
+
+
+def test_bert_s384_matches_reference():
+    """BERT at seq 384 (3 key tiles/query block) through the captured
+    engine vs the fp32 reference — VERDICT item 5 generality gate."""
+    from trtlab_amd.engine.planner import Planner
+    from trtlab_amd.engine.reference import run_reference
+    from trtlab_amd.engine.runtime import NativeEngine
+    from trtlab_amd.models import build_bert
+
+    g = build_bert(batch=1, seq=384, layers=2, seed=3)
+    plan = Planner().compile(g)
+    eng = NativeEngine(plan)
+    ctx = eng.create_context(capture=True)
+    x = np.random.RandomState(4).randn(*plan.input_shape).astype(np.float32)
+    out = ctx.infer(x).astype(np.float32)
+    ref = run_reference(plan, x)
+    err = np.abs(out - ref).max() / max(np.abs(ref).max(), 1e-6)
+    assert err < 0.08, err
+
+
+def test_bert_odd_seq_matches_reference():
+    """Arbitrary (non-tile-multiple) sequence length s=200 end-to-end."""
+    from trtlab_amd.engine.planner import Planner
+    from trtlab_amd.engine.reference import run_reference
+    from trtlab_amd.engine.runtime import NativeEngine
+    from trtlab_amd.models import build_bert
+
+    g = build_bert(batch=2, seq=200, layers=1, seed=5)
+    plan = Planner().compile(g)
+    eng = NativeEngine(plan)
+    ctx = eng.create_context(capture=True)
+    x = np.random.RandomState(6).randn(*plan.input_shape).astype(np.float32)
+    out = ctx.infer(x).astype(np.float32)
+    ref = run_reference(plan, x)
+    err = np.abs(out - ref).max() / max(np.abs(ref).max(), 1e-6)
+    assert err < 0.08, err
+
+
+def test_head_dim_128_decoder_matches_reference():
+    """A head_dim-128 causal decoder layer stack (LLaMA-ish head shape:
+    hidden 768, 6 heads x 128) through the captured engine."""
+    from trtlab_amd.engine.planner import Planner
+    from trtlab_amd.engine.reference import run_reference
+    from trtlab_amd.engine.runtime import NativeEngine
+    from trtlab_amd.models import build_bert
+
+    g = build_bert(batch=2, seq=128, layers=2, heads=6, hidden=768, seed=7)
+    # bert builder with heads=6 gives head_dim 128; flip attention causal
+    for n in g.nodes:
+        if n.kind == "attention":
+            n.attrs["causal"] = True
+    plan = Planner().compile(g)
+    att = next(d for d in plan.ops if d["kind"] == 9)
+    assert att["HD"] == 128 and att["causal"] == 1
+    eng = NativeEngine(plan)
+    ctx = eng.create_context(capture=True)
+    x = np.random.RandomState(8).randn(*plan.input_shape).astype(np.float32)
+    out = ctx.infer(x).astype(np.float32)
+    ref = run_reference(plan, x)
+    err = np.abs(out - ref).max() / max(np.abs(ref).max(), 1e-6)
+    assert err < 0.08, err

@@ -416,14 +416,20 @@ def test_gemm_mxfp4(C, M, N, K):
     assert corr > 0.95, corr  # fp4: coarser grid than fp8
 
 
-@pytest.mark.parametrize("S,causal", [(128, 0), (256, 0), (512, 0),
-                                      (128, 1), (256, 1)])
-def test_attention_long_and_causal(C, S, causal):
-    """Online-softmax attention at S > 128 (streamed key tiles) and with
-    the decoder-style causal mask, vs plain torch attention."""
-    B, H, D = 2, 4, 64
+@pytest.mark.parametrize("S,causal,D", [
+    (128, 0, 64), (256, 0, 64), (512, 0, 64), (128, 1, 64), (256, 1, 64),
+    # arbitrary S: tail query blocks + masked tail key tiles
+    (200, 0, 64), (100, 1, 64), (77, 0, 64), (384, 0, 64),
+    # head_dim 128 variant (64-key LDS tiles)
+    (128, 0, 128), (256, 1, 128), (200, 0, 128), (77, 0, 128),
+])
+def test_attention_long_and_causal(C, S, causal, D):
+    """Online-softmax attention: streamed key tiles at any S (incl. non-
+    multiples of the tile sizes), decoder-style causal mask, and both head
+    dims {64, 128}, vs plain torch attention."""
+    B, H = 2, 4
     hid = H * D
-    qkv = t16(B * S, 3 * hid, seed=100 + S + causal)
+    qkv = t16(B * S, 3 * hid, seed=100 + S + causal + D)
     out = torch.empty(B * S, hid, dtype=torch.half, device="cuda")
     torch.cuda.synchronize()
     C.ops.attention(0, qkv.data_ptr(), out.data_ptr(), B, S, H, D,
@@ -441,6 +447,33 @@ def test_attention_long_and_causal(C, S, causal):
     check(out, ref)
 
 
+@pytest.mark.parametrize("S,D", [(200, 64), (200, 128), (384, 128)])
+def test_attention_varlen_odd_s(C, S, D):
+    """Variable-length key masking (seqlens) combined with arbitrary S and
+    both head dims: rows of each sequence attend only its valid keys."""
+    B, H = 2, 4
+    hid = H * D
+    qkv = t16(B * S, 3 * hid, seed=300 + S + D)
+    lens = torch.tensor([S, S * 2 // 3], dtype=torch.int32, device="cuda")
+    out = torch.empty(B * S, hid, dtype=torch.half, device="cuda")
+    torch.cuda.synchronize()
+    C.ops.attention(0, qkv.data_ptr(), out.data_ptr(), B, S, H, D,
+                    1.0 / (D ** 0.5), seqlens=lens.data_ptr())
+    q = qkv.float().reshape(B, S, 3, H, D)
+    scores = (q[:, :, 0].permute(0, 2, 1, 3) @
+              q[:, :, 1].permute(0, 2, 1, 3).transpose(-1, -2)) / (D ** 0.5)
+    for b in range(B):
+        scores[b, :, :, lens[b]:] = float("-inf")
+    att = torch.softmax(scores, dim=-1)
+    ref = (att @ q[:, :, 2].permute(0, 2, 1, 3)).permute(
+        0, 2, 1, 3).reshape(B * S, hid)
+    # compare only valid query rows (padded rows are don't-care)
+    valid = torch.zeros(B * S, dtype=torch.bool)
+    for b in range(B):
+        valid[b * S:b * S + int(lens[b])] = True
+    check(out[valid.to(out.device)], ref[valid.to(ref.device)])
+
+
 def test_gemm_bt_tile256(C):
     """The 256x128 GEMM tactic (code 5) vs torch on a transformer shape."""
     M, N, K = 2048, 768, 768
@@ -451,3 +484,21 @@ def test_gemm_bt_tile256(C):
     C.ops.gemm_bt(0, a.data_ptr(), b.data_ptr(), out.data_ptr(),
                   M=M, N=N, K=K, tile=5)
     check(out, (a.float() @ b.float().t()).half())
+
+
+def test_gemm_bt_strided_operands(C):
+    """Strided lda/ldb/ldc (sub-matrix views): C view[M,N] inside a larger
+    buffer = A view @ B view^T, vs torch on the same views."""
+    LDA, LDB, LDC = 1024, 896, 1536
+    M, N, K = 192, 256, 512
+    a_full = t16(M, LDA, seed=71)
+    b_full = t16(N, LDB, seed=72)
+    c_full = torch.zeros(M, LDC, dtype=torch.half, device="cuda")
+    torch.cuda.synchronize()
+    C.ops.gemm_bt(0, a_full.data_ptr(), b_full.data_ptr(),
+                  c_full.data_ptr(), M=M, N=N, K=K,
+                  lda=LDA, ldb=LDB, ldc=LDC)
+    ref = a_full[:, :K].float() @ b_full[:, :K].float().t()
+    check(c_full[:, :N], ref)
+    # untouched tail of each C row stays zero
+    assert (c_full[:, N:] == 0).all()
