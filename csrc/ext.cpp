@@ -112,6 +112,34 @@ PYBIND11_MODULE(_C, m) {
     TRT_HIP_CHECK(hipMemset((void*)p, v, bytes));
   });
 
+  // Growing best-fit device allocator (reference bfit_allocator +
+  // growing block_arena) for multi-model serving / dynamic shapes.
+  py::class_<DeviceArena>(mem, "DeviceArena")
+      .def(py::init<int, size_t, size_t, size_t>(), py::arg("device") = 0,
+           py::arg("initial_bytes") = 0, py::arg("max_bytes") = 0,
+           py::arg("growth_bytes") = 0)
+      .def("allocate",
+           [](DeviceArena& a, size_t bytes, size_t align) {
+             return (uintptr_t)a.allocate(bytes, align);
+           },
+           py::arg("bytes"), py::arg("align") = 256)
+      .def("deallocate",
+           [](DeviceArena& a, uintptr_t p) { a.deallocate((void*)p); })
+      .def("stats", [](DeviceArena& a) {
+        auto s = a.stats();
+        py::dict d;
+        d["capacity"] = s.capacity;
+        d["in_use"] = s.in_use;
+        d["high_water"] = s.high_water;
+        d["largest_free"] = s.largest_free;
+        d["free_nodes"] = s.free_nodes;
+        d["live_allocs"] = s.live_allocs;
+        py::list h;
+        for (int i = 0; i < 48; ++i) h.append(s.histogram[i]);
+        d["histogram"] = h;
+        return d;
+      });
+
   py::class_<BlockPool>(mem, "BlockPool")
       .def(py::init<size_t, int, int>(), py::arg("block_bytes"),
            py::arg("count"), py::arg("device") = 0)
@@ -580,7 +608,8 @@ PYBIND11_MODULE(_C, m) {
       });
 
   py::class_<ExecutionContext>(m, "ExecutionContext")
-      .def(py::init<std::shared_ptr<Engine>>())
+      .def(py::init<std::shared_ptr<Engine>, uintptr_t>(), py::arg("engine"),
+           py::arg("external_arena") = 0)
       .def("capture",
            [](ExecutionContext& c) {
              py::gil_scoped_release rel;

@@ -73,10 +73,16 @@ void Engine::upload_weights(const void* src, size_t bytes) {
     (void)hipMemPrefetchAsync(weights_, bytes, device_, 0);
 }
 
-ExecutionContext::ExecutionContext(std::shared_ptr<Engine> engine)
+ExecutionContext::ExecutionContext(std::shared_ptr<Engine> engine,
+                                   uintptr_t external_arena)
     : eng_(std::move(engine)) {
   TRT_HIP_CHECK(hipSetDevice(eng_->device()));
-  arena_ = (char*)device_malloc(eng_->arena_bytes(), eng_->device());
+  if (external_arena) {
+    arena_ = (char*)external_arena;  // caller-owned (shared DeviceArena)
+    owns_arena_ = false;
+  } else {
+    arena_ = (char*)device_malloc(eng_->arena_bytes(), eng_->device());
+  }
   if (eng_->scratch_bytes())
     scratch_ = (char*)device_malloc(eng_->scratch_bytes(), eng_->device());
   // Carve one pinned slab per direction into per-binding regions, each
@@ -100,7 +106,7 @@ ExecutionContext::~ExecutionContext() {
   if (graph_exec_) hipGraphExecDestroy(graph_exec_);
   if (graph_) hipGraphDestroy(graph_);
   hipStreamDestroy(stream_);
-  device_free(arena_, eng_->arena_bytes());
+  if (owns_arena_) device_free(arena_, eng_->arena_bytes());
   if (scratch_) device_free(scratch_, eng_->scratch_bytes());
   pinned_free(h_in_, h_in_bytes_);
   pinned_free(h_out_, h_out_bytes_);

@@ -4,8 +4,10 @@
 
 #include <array>
 #include <atomic>
+#include <map>
 #include <memory>
 #include <mutex>
+#include <set>
 #include <vector>
 
 namespace trtlab {
@@ -38,6 +40,52 @@ class BlockPool {
   int total_;
   std::vector<char*> free_;
   std::mutex mu_;
+};
+
+// Growing best-fit device allocator for dynamic workloads (multi-model
+// arena sharing, decode KV growth) — reference bfit_allocator.h:121 +
+// block_arena.h:177. All bookkeeping host-side (device memory cannot hold
+// list nodes: the reference's block_list_oob insight).
+class DeviceArena {
+ public:
+  struct Stats {
+    size_t capacity = 0, in_use = 0, high_water = 0, largest_free = 0;
+    size_t free_nodes = 0, live_allocs = 0;
+    uint64_t histogram[48] = {0};  // log2 allocation-size buckets
+  };
+
+  DeviceArena(int device, size_t initial_bytes, size_t max_bytes = 0,
+              size_t growth_bytes = 0);
+  ~DeviceArena();
+  DeviceArena(const DeviceArena&) = delete;
+  DeviceArena& operator=(const DeviceArena&) = delete;
+
+  void* allocate(size_t bytes, size_t align = 256);
+  void deallocate(void* p);
+  Stats stats() const;
+  int device() const { return device_; }
+
+ private:
+  struct Slab {
+    char* base;
+    size_t bytes;
+  };
+  static inline size_t round_up(int64_t a, int64_t b) {
+    return (size_t)(((a + b - 1) / b) * b);
+  }
+  void grow(size_t need);
+  void insert_free(char* p, size_t bytes);
+  void erase_size(char* p, size_t bytes);
+
+  int device_;
+  size_t max_bytes_, growth_bytes_;
+  size_t capacity_ = 0, in_use_ = 0, high_water_ = 0;
+  std::vector<Slab> slabs_;
+  std::map<char*, size_t> by_addr_;              // free nodes by address
+  std::set<std::pair<size_t, char*>> by_size_;   // free nodes by size
+  std::map<char*, size_t> live_;                 // outstanding allocations
+  uint64_t hist_[48] = {0};
+  mutable std::mutex mu_;
 };
 
 // ---------------------------------------------------------------- executor
@@ -152,7 +200,11 @@ class Engine {
 // allocate bindings/scratch, warm up, capture enqueue into a graph, replay.
 class ExecutionContext {
  public:
-  ExecutionContext(std::shared_ptr<Engine> engine);
+  // external_arena: optional caller-owned device pointer for the
+  // activation arena (e.g. carved from a shared DeviceArena so multiple
+  // models serve from one pool); 0 = allocate privately.
+  ExecutionContext(std::shared_ptr<Engine> engine,
+                   uintptr_t external_arena = 0);
   ~ExecutionContext();
 
   // Raw host staging buffers (pinned), one per binding. Index 0 keeps the
@@ -180,6 +232,7 @@ class ExecutionContext {
  private:
   std::shared_ptr<Engine> eng_;
   char* arena_ = nullptr;
+  bool owns_arena_ = true;
   char* scratch_ = nullptr;
   // One pinned slab per direction, carved per binding (in_hoff_[i] = host
   // offset of binding i) — the reference's Buffers/Bindings carving

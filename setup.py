@@ -29,6 +29,7 @@ SOURCES = [
     "csrc/runtime/memory.cpp",
     "csrc/runtime/executor.cpp",
     "csrc/runtime/comm.cpp",
+    "csrc/runtime/bfit.cpp",
     "csrc/ext.cpp",
 ]
 

@@ -54,20 +54,25 @@ class NativeEngine:
     def upload_weights(self, blob: np.ndarray) -> None:
         self.engine.upload_weights(blob)
 
-    def create_context(self, capture: bool = True,
-                       timed: bool = False) -> "NativeContext":
-        return NativeContext(self, capture=capture, timed=timed)
+    def create_context(self, capture: bool = True, timed: bool = False,
+                       arena_ptr: int = 0) -> "NativeContext":
+        """arena_ptr: optional caller-owned device memory for the
+        activation arena (carve from a shared memory.DeviceArena so
+        multiple models serve from one pool — reference growing
+        block_arena / bfit role)."""
+        return NativeContext(self, capture=capture, timed=timed,
+                             arena_ptr=arena_ptr)
 
 
 class NativeContext:
     """Private stream + activation arena + pinned bindings + hipGraph."""
 
     def __init__(self, engine: NativeEngine, capture: bool = True,
-                 timed: bool = False):
+                 timed: bool = False, arena_ptr: int = 0):
         self._C = native()
         self.engine = engine
         self.plan = engine.plan
-        self.ctx = self._C.ExecutionContext(engine.engine)
+        self.ctx = self._C.ExecutionContext(engine.engine, arena_ptr)
         if timed:
             # per-stage H2D/compute/D2H events (reference
             # TimedBenchmarkWorkspace); recorded inside the captured graph
@@ -182,11 +187,23 @@ class InferenceManager:
 
     def __init__(self, max_contexts: int = 2, device: int = 0,
                  pre_threads: int = 1, hip_threads: int = 1,
-                 post_threads: int = 2):
+                 post_threads: int = 2, shared_arena: bool = False,
+                 arena_max_bytes: int = 0):
+        """shared_arena=True: all models' activation arenas are carved from
+        ONE growing best-fit device pool (native DeviceArena — reference
+        bfit_allocator/block_arena role) instead of per-context hipMallocs;
+        pool stats (high-water, histogram) export via arena_stats()."""
         self.device = device
         self.max_contexts = max_contexts
         self._models: Dict[str, NativeEngine] = {}
         self._ctx_pools: Dict[str, Pool] = {}
+        self.arena = None
+        self._arena_allocs: List[int] = []
+        if shared_arena:
+            from trtlab_amd import native
+
+            self.arena = native().memory.DeviceArena(
+                device, 0, arena_max_bytes)
         self.thread_pools: Dict[str, ThreadPool] = {
             "pre": ThreadPool(pre_threads, "pre"),
             "hip": ThreadPool(hip_threads, "hip"),
@@ -199,8 +216,16 @@ class InferenceManager:
     def allocate_resources(self) -> None:
         for name, eng in self._models.items():
             if name not in self._ctx_pools:
+                ptrs = [0] * self.max_contexts
+                if self.arena is not None:
+                    ptrs = [self.arena.allocate(eng.plan.arena_bytes)
+                            for _ in range(self.max_contexts)]
+                    self._arena_allocs.extend(ptrs)
                 self._ctx_pools[name] = Pool(
-                    [eng.create_context() for _ in range(self.max_contexts)])
+                    [eng.create_context(arena_ptr=p) for p in ptrs])
+
+    def arena_stats(self) -> Optional[dict]:
+        return self.arena.stats() if self.arena is not None else None
 
     def get_model(self, name: str) -> NativeEngine:
         return self._models[name]
