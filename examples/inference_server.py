@@ -27,6 +27,9 @@ def main():
     ap.add_argument("--contexts", type=int, default=4)
     ap.add_argument("--devices", default="0",
                     help="comma-separated GPU ids for replica-per-GPU")
+    ap.add_argument("--shared-arena", action="store_true",
+                    help="serve all models from one growing best-fit device "
+                         "pool (native DeviceArena); exports pool gauges")
     args = ap.parse_args()
 
     from trtlab_amd.engine.planner import Planner
@@ -46,7 +49,8 @@ def main():
     devices = [int(d) for d in args.devices.split(",")]
     managers = []
     for dev in devices:
-        mgr = InferenceManager(max_contexts=args.contexts, device=dev)
+        mgr = InferenceManager(max_contexts=args.contexts, device=dev,
+                               shared_arena=args.shared_arena)
         mgr.register_model(args.model, plan)
         mgr.allocate_resources()
         managers.append(mgr)
@@ -62,8 +66,11 @@ def main():
     print(f"serving {args.model} b{args.batch} on :{args.port} "
           f"(devices {devices}, {args.contexts} contexts each; "
           f"metrics :{args.metrics_port})")
-    server.run(control_interval_s=2.0,
-               control_fn=lambda: metrics.update_power(devices[0]))
+    def control():
+        metrics.update_power(devices[0])
+        metrics.update_arena(managers[0])
+
+    server.run(control_interval_s=2.0, control_fn=control)
 
 
 class _ReplicaResources:

@@ -735,3 +735,69 @@ def test_head_dim_128_decoder_matches_reference():
     ref = run_reference(plan, x)
     err = np.abs(out - ref).max() / max(np.abs(ref).max(), 1e-6)
     assert err < 0.08, err
+
+
+def test_device_arena_best_fit_and_coalescing():
+    """Growing best-fit allocator: growth on demand, address-ordered
+    coalescing on free, high-water + log2 histogram tracking."""
+    C = trtlab_amd.native()
+    a = C.memory.DeviceArena(0, 1 << 20, growth_bytes=1 << 20)
+    p1 = a.allocate(100 << 10)
+    p2 = a.allocate(200 << 10)
+    p3 = a.allocate(300 << 10)
+    s = a.stats()
+    assert s["live_allocs"] == 3
+    assert s["in_use"] >= (600 << 10)
+    # free the middle block, then ask for something that fits only there
+    a.deallocate(p2)
+    p4 = a.allocate(150 << 10)
+    assert p4 == p2  # best-fit reuses the freed hole
+    a.deallocate(p1)
+    a.deallocate(p3)
+    a.deallocate(p4)
+    s = a.stats()
+    assert s["live_allocs"] == 0 and s["in_use"] == 0
+    # all frees coalesced back into one node per slab
+    assert s["free_nodes"] == 1
+    assert s["high_water"] >= (600 << 10)
+    assert sum(s["histogram"]) == 4
+    # growth: allocation larger than current capacity triggers a new slab
+    big = a.allocate(8 << 20)
+    assert a.stats()["capacity"] >= (9 << 20)
+    a.deallocate(big)
+    # double free / unknown free fails loudly
+    with pytest.raises(RuntimeError):
+        a.deallocate(big)
+
+
+def test_two_models_share_one_arena():
+    """Two models of different sizes served from ONE shared device pool
+    (VERDICT item 8 'done' gate), with high-water stats exported."""
+    from trtlab_amd.engine.planner import Planner
+    from trtlab_amd.engine.reference import run_reference
+    from trtlab_amd.engine.runtime import InferenceManager
+    from trtlab_amd.models import build_resnet
+
+    mgr = InferenceManager(max_contexts=2, shared_arena=True)
+    g18 = build_resnet(18, batch=2, image=64, seed=0)
+    g50 = build_resnet(50, batch=2, image=64, seed=1)
+    p18 = Planner().compile(g18)
+    p50 = Planner().compile(g50)
+    mgr.register_model("rn18", p18)
+    mgr.register_model("rn50", p50)
+    mgr.allocate_resources()
+    s = mgr.arena_stats()
+    assert s is not None
+    assert s["live_allocs"] == 4  # 2 contexts x 2 models
+    assert s["in_use"] >= 2 * (p18.arena_bytes + p50.arena_bytes)
+    assert s["high_water"] == s["in_use"]
+    # both models compute correctly from the shared pool
+    for name, plan in (("rn18", p18), ("rn50", p50)):
+        r = mgr.infer_runner(name)
+        x = np.random.RandomState(3).randn(
+            *plan.input_shape).astype(np.float32) * 0.5
+        out = r.infer(x).result(timeout=120).astype(np.float32)
+        ref = run_reference(plan, x)
+        err = np.abs(out - ref).max() / max(np.abs(ref).max(), 1e-6)
+        assert err < 0.08, (name, err)
+    mgr.shutdown()

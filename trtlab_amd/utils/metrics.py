@@ -33,6 +33,17 @@ class Metrics:
             buckets=_LOAD_RATIO_BUCKETS, registry=self.registry)
         self.power_w = Gauge("trtlab_gpu_power_watts", "GPU power draw",
                              registry=self.registry)
+        # shared device-arena pool gauges (native DeviceArena stats —
+        # reference histogram_tracker / memory tracking role)
+        self.arena_in_use = Gauge("trtlab_arena_in_use_bytes",
+                                  "shared device arena bytes in use",
+                                  registry=self.registry)
+        self.arena_high_water = Gauge("trtlab_arena_high_water_bytes",
+                                      "shared device arena high-water mark",
+                                      registry=self.registry)
+        self.arena_capacity = Gauge("trtlab_arena_capacity_bytes",
+                                    "shared device arena capacity",
+                                    registry=self.registry)
         self.port = port
         self._started = False
 
@@ -49,6 +60,14 @@ class Metrics:
         self.request_ms.observe(request_ms)
         if compute_ms > 0:
             self.load_ratio.observe(request_ms / compute_ms)
+
+    def update_arena(self, manager) -> None:
+        """Refresh pool gauges from an InferenceManager's shared arena."""
+        s = manager.arena_stats() if manager is not None else None
+        if s:
+            self.arena_in_use.set(s["in_use"])
+            self.arena_high_water.set(s["high_water"])
+            self.arena_capacity.set(s["capacity"])
 
     def update_power(self, device: int = 0) -> Optional[float]:
         """Refresh the power gauge from rocm-smi/amd-smi (reference:
