@@ -51,6 +51,8 @@ OpDesc op_from_dict(const py::dict& d) {
   o.res_scale = gf("res_scale", 1.0f);
   o.q_scale = gf("q_scale", 1.0f);
   o.tile = gi("tile", 0);
+  o.fork = gi("fork", 0);
+  o.join = gi("join", 0);
   return o;
 }
 

@@ -98,13 +98,28 @@ ExecutionContext::ExecutionContext(std::shared_ptr<Engine> engine,
   h_in_ = (char*)pinned_malloc(h_in_bytes_);
   h_out_ = (char*)pinned_malloc(h_out_bytes_);
   TRT_HIP_CHECK(hipStreamCreateWithFlags(&stream_, hipStreamNonBlocking));
+  // fork/join resources: one side stream + 2 events per forked op + a
+  // private split-K slab so forked convs never race the main chain's
+  int nfork = 0;
+  for (const OpDesc& op : eng_->ops()) nfork += op.fork ? 1 : 0;
+  if (nfork) {
+    TRT_HIP_CHECK(hipStreamCreateWithFlags(&side_, hipStreamNonBlocking));
+    fork_ev_.resize(2 * nfork);
+    for (auto& e : fork_ev_)
+      TRT_HIP_CHECK(hipEventCreateWithFlags(&e, hipEventDisableTiming));
+    if (eng_->scratch_bytes())
+      scratch2_ = (char*)device_malloc(eng_->scratch_bytes(), eng_->device());
+  }
 }
 
 ExecutionContext::~ExecutionContext() {
   for (auto& e : ev_)
     if (e) hipEventDestroy(e);
+  for (auto& e : fork_ev_) hipEventDestroy(e);
   if (graph_exec_) hipGraphExecDestroy(graph_exec_);
   if (graph_) hipGraphDestroy(graph_);
+  if (side_) hipStreamDestroy(side_);
+  if (scratch2_) device_free(scratch2_, eng_->scratch_bytes());
   hipStreamDestroy(stream_);
   if (owns_arena_) device_free(arena_, eng_->arena_bytes());
   if (scratch_) device_free(scratch_, eng_->scratch_bytes());
@@ -132,64 +147,81 @@ void ExecutionContext::enqueue_all(hipStream_t s) {
                                  ins[i].bytes, hipMemcpyHostToDevice, s));
   if (timing_) TRT_HIP_CHECK(hipEventRecord(ev_[1], s));
 
+  // fork/join dual-stream schedule: forked ops (downsample convs) run on
+  // side_ overlapping the main chain; the consuming op waits on the
+  // branch's event. Event record/wait inside stream capture become graph
+  // dependencies, so the replayed hipGraph keeps the fork/join shape.
+  int evi = 0;
+  hipEvent_t pending_join = nullptr;
   for (const OpDesc& op : eng_->ops()) {
+    hipStream_t os = s;
+    if (op.join && pending_join) {
+      TRT_HIP_CHECK(hipStreamWaitEvent(s, pending_join, 0));
+      pending_join = nullptr;
+    }
+    if (op.fork && side_) {
+      hipEvent_t ef = fork_ev_[evi++];
+      TRT_HIP_CHECK(hipEventRecord(ef, s));
+      TRT_HIP_CHECK(hipStreamWaitEvent(side_, ef, 0));
+      os = side_;
+    }
     switch (op.kind) {
       case kConv2d:
         launch_conv2d(op.dtype, A(op.in_off), Wp(op.w_off), A(op.out_off),
                       Fp(op.scale_off), Fp(op.bias_off), A(op.in2_off),
                       eng_->zero_page(), op.Nb, op.H, op.W, op.C, op.Cout,
-                      op.KH, op.KW, op.sh, op.sw, op.ph, op.pw, op.epi, s,
-                      op.tile, scratch_, op.res_scale);
+                      op.KH, op.KW, op.sh, op.sw, op.ph, op.pw, op.epi, os,
+                      op.tile, op.fork ? scratch2_ : scratch_, op.res_scale);
         break;
       case kGemmBt:
         launch_gemm_bt(op.dtype, A(op.in_off), Wp(op.w_off), A(op.out_off),
                        Fp(op.scale_off), Fp(op.bias_off), A(op.in2_off),
                        op.res_scale, op.M, op.N, op.K, op.K /*lda*/,
-                       op.K /*ldb*/, op.N /*ldc*/, op.epi, s, op.tile,
+                       op.K /*ldb*/, op.N /*ldc*/, op.epi, os, op.tile,
                        scratch_, op.q_scale /*post-epilogue out scale*/);
         break;
       case kMaxPool:
         launch_maxpool2d(op.dtype, A(op.in_off), A(op.out_off), op.Nb, op.H,
                          op.W, op.C, op.KH, op.KW, op.sh, op.sw, op.ph, op.pw,
-                         s);
+                         os);
         break;
       case kAvgPool:
         launch_avgpool2d(op.dtype, A(op.in_off), A(op.out_off), op.Nb, op.H,
                          op.W, op.C, op.KH, op.KW, op.sh, op.sw, op.ph,
-                         op.pw, s);
+                         op.pw, os);
         break;
       case kGAvgPool:
         launch_gavgpool(op.dtype, A(op.in_off), A(op.out_off), op.Nb, op.HW,
-                        op.C, s);
+                        op.C, os);
         break;
       case kSoftmax:
         launch_softmax_rows(op.dtype, A(op.in_off), A(op.out_off), op.M, op.N,
-                            op.N, s);
+                            op.N, os);
         break;
       case kLayerNorm:
         // out2 (optional) = fused fp8 copy at q_scale
         launch_layernorm(op.dtype, A(op.in_off), Fp(op.scale_off),
                          Fp(op.bias_off), A(op.out_off), op.M, op.N, op.N,
-                         op.eps, s, A(op.out2_off), op.q_scale);
+                         op.eps, os, A(op.out2_off), op.q_scale);
         break;
       case kAddLayerNorm:
         launch_add_layernorm(op.dtype, A(op.in_off), A(op.in2_off),
                              Fp(op.scale_off), Fp(op.bias_off), A(op.out_off),
-                             nullptr /*sum_out*/, op.M, op.N, op.N, op.eps, s,
+                             nullptr /*sum_out*/, op.M, op.N, op.N, op.eps, os,
                              A(op.out2_off) /*fused fp8 copy*/, op.q_scale);
         break;
       case kElementwise:
         launch_elementwise(op.dtype, op.epi, A(op.in_off), A(op.in2_off),
-                           A(op.out_off), op.n_elems, s);
+                           A(op.out_off), op.n_elems, os);
         break;
       case kChannelPad:
         launch_channel_pad(op.dtype, A(op.in_off), A(op.out_off), op.n_elems,
-                           op.C, op.Cout, s);
+                           op.C, op.Cout, os);
         break;
       case kAttention:
         // epi = output dtype flag (3 -> fused fp8 out at 1/q_scale)
         launch_attention(op.dtype, A(op.in_off), A(op.out_off), op.B, op.S,
-                         op.NH, op.HD, op.att_scale, s, op.epi,
+                         op.NH, op.HD, op.att_scale, os, op.epi,
                          op.q_scale != 0.f && op.epi == 3
                              ? 1.0f / op.q_scale
                              : 1.0f,
@@ -197,50 +229,57 @@ void ExecutionContext::enqueue_all(hipStream_t s) {
         break;
       case kSeqLens:
         // op.epi carries pad_id
-        launch_seqlens(A(op.in_off), A(op.out_off), op.B, op.S, op.epi, s);
+        launch_seqlens(A(op.in_off), A(op.out_off), op.B, op.S, op.epi, os);
         break;
       case kQuantMx4:
         // out = fp4 codes, out2 = e8m0 block scales
         launch_quantize_mxfp4(A(op.in_off), A(op.out_off), A(op.out2_off),
-                              op.M, op.K, s);
+                              op.M, op.K, os);
         break;
       case kGemmMx4:
         // A/Sa = quantized activations (arena), B/Sb = weights (blob)
         launch_gemm_mxfp4(A(op.in_off), Wp(op.w_off), A(op.in2_off),
-                          Wp(op.w2_off), A(op.out_off), op.M, op.N, op.K, s,
+                          Wp(op.w2_off), A(op.out_off), op.M, op.N, op.K, os,
                           0 /*fp16 out*/, op.epi, Fp(op.scale_off),
                           Fp(op.bias_off));
         break;
       case kQuantMx8:
         launch_quantize_mxfp8(A(op.in_off), A(op.out_off), A(op.out2_off),
-                              op.M, op.K, s);
+                              op.M, op.K, os);
         break;
       case kGemmMx8:
         launch_gemm_mxfp8(A(op.in_off), Wp(op.w_off), A(op.in2_off),
-                          Wp(op.w2_off), A(op.out_off), op.M, op.N, op.K, s,
+                          Wp(op.w2_off), A(op.out_off), op.M, op.N, op.K, os,
                           0 /*fp16 out*/, op.epi, Fp(op.scale_off),
                           Fp(op.bias_off));
         break;
       case kQuantize:
         // op.epi carries the target format (0 = int8, 1 = fp8 e4m3)
         launch_quantize(A(op.in_off), A(op.out_off), op.n_elems, op.q_scale,
-                        s, op.epi);
+                        os, op.epi);
         break;
       case kDequant:
         launch_dequant(A(op.in_off), A(op.out_off), op.n_elems, op.q_scale,
-                       s, op.epi);
+                       os, op.epi);
         break;
       case kEmbedding:
         // tables live in the weight blob as fp16: tok at w_off, pos at
         // scale_off, optional seg at bias_off; optional segids at in2_off.
         launch_embedding(op.dtype, A(op.in_off), Wp(op.w_off),
                          Wp(op.scale_off), Wp(op.bias_off), A(op.in2_off),
-                         A(op.out_off), op.M, op.S, op.N, s);
+                         A(op.out_off), op.M, op.S, op.N, os);
         break;
       default:
         throw std::runtime_error("unknown op kind");
     }
+    if (op.fork && side_) {  // branch completion event for the join
+      hipEvent_t ej = fork_ev_[evi++];
+      TRT_HIP_CHECK(hipEventRecord(ej, side_));
+      pending_join = ej;
+    }
   }
+  if (pending_join)  // safety: never leave a branch dangling past the D2H
+    TRT_HIP_CHECK(hipStreamWaitEvent(s, pending_join, 0));
 
   if (timing_) TRT_HIP_CHECK(hipEventRecord(ev_[2], s));
   const auto& outs = eng_->outputs();

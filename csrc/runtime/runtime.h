@@ -138,6 +138,12 @@ struct OpDesc {
   float q_scale = 1.0f;
   // autotuned tile override (0 = heuristic; 1..4 = fixed BMxBN config)
   int tile = 0;
+  // fork/join dual-stream schedule (ResNet downsample pattern): fork=1
+  // launches this op on the context's side stream (its inputs are ready
+  // before the main chain it overlaps); join=1 makes the op wait for the
+  // pending forked op's event first. Pairs are strictly sequential.
+  int fork = 0;
+  int join = 0;
 };
 
 // One engine I/O binding: a named arena region mirrored by pinned host
@@ -242,6 +248,11 @@ class ExecutionContext {
   std::vector<size_t> in_hoff_, out_hoff_;
   size_t h_in_bytes_ = 0, h_out_bytes_ = 0;
   hipStream_t stream_{};
+  // fork/join side stream (captured as a forked branch of the main
+  // stream's graph): downsample convs overlap the main bottleneck chain
+  hipStream_t side_{};
+  char* scratch2_ = nullptr;  // private split-K slab for forked ops
+  std::vector<hipEvent_t> fork_ev_;  // 2 events per fork op (fork + join)
   hipGraph_t graph_{};
   hipGraphExec_t graph_exec_{};
   bool captured_ = false;

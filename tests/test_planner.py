@@ -236,3 +236,26 @@ def test_planner_error_paths():
             n.attrs["varlen"] = True
     with pytest.raises(ValueError, match="token-id"):
         Planner().compile(g3)
+
+
+def test_fork_join_downsample_marking():
+    """ResNet downsample convs are marked fork (side-stream) with their
+    consuming residual conv marked join; pairs are strictly sequential and
+    the forked inputs stay live until the join (no arena aliasing while
+    the side stream reads them)."""
+    from trtlab_amd.engine.planner import K_CONV, Planner
+    from trtlab_amd.models import build_resnet
+
+    plan = Planner().compile(build_resnet(50, batch=2, image=64, seed=0))
+    forks = [i for i, d in enumerate(plan.ops) if d.get("fork")]
+    joins = [i for i, d in enumerate(plan.ops) if d.get("join")]
+    assert len(forks) == 4 and len(joins) == 4  # one per stage transition
+    seq = sorted([(i, "f") for i in forks] + [(j, "j") for j in joins])
+    kinds = "".join(k for _, k in seq)
+    assert kinds == "fjfjfjfj"  # strictly alternating pairs
+    for f, j in zip(forks, joins):
+        assert j > f + 1  # there is work to overlap
+        assert plan.ops[f]["kind"] == K_CONV
+        assert plan.ops[j]["kind"] == K_CONV
+        # the join consumes the forked output as its residual
+        assert plan.ops[j]["in2_off"] == plan.ops[f]["out_off"]

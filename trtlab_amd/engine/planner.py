@@ -433,6 +433,35 @@ class Planner:
             if op.params.get("mx_wscales") is not None:
                 w2_offs[op.name] = pack(op.params["mx_wscales"])
 
+        # ---- fork/join dual-stream schedule (downsample overlap) ----
+        # A conv whose output's SOLE use is the residual (in2) input of a
+        # later conv runs on the context's side stream, overlapping the
+        # main bottleneck chain (the ResNet downsample pattern). Pairs are
+        # strictly sequential; the executor turns them into hipEvent
+        # dependencies inside the captured graph (executor.cpp fork/join).
+        consumers_of: Dict[str, List[int]] = {}
+        for i, op in enumerate(exec_ops):
+            for t in op.inputs:
+                consumers_of.setdefault(t, []).append(i)
+        fork_pairs: List[Tuple[int, int]] = []
+        last_join = -1
+        for i, op in enumerate(exec_ops):
+            if op.kind != K_CONV or i <= last_join:
+                continue
+            cons = consumers_of.get(op.output, [])
+            if len(cons) != 1:
+                continue
+            j = cons[0]
+            cj = exec_ops[j]
+            if j <= i + 1 or cj.kind != K_CONV:
+                continue
+            if len(cj.inputs) < 2 or cj.inputs[1] != op.output:
+                continue
+            op.params["fork"] = 1
+            cj.params["join"] = 1
+            fork_pairs.append((i, j))
+            last_join = j
+
         # ---- liveness + arena offsets ----
         tensors_used: Dict[str, Tuple[int, int]] = {}
 
@@ -461,6 +490,14 @@ class Planner:
         for t in output_names:
             s, e = tensors_used[t]
             tensors_used[t] = (s, len(exec_ops))
+        # forked ops run concurrently with ops (i, j): their INPUT regions
+        # must stay live until the join so no intermediate output aliases
+        # memory the side stream is still reading
+        for fi, fj in fork_pairs:
+            for t in exec_ops[fi].inputs:
+                if t in tensors_used:
+                    s, e = tensors_used[t]
+                    tensors_used[t] = (s, max(e, fj))
 
         def nbytes_of(t: str) -> int:
             n = 1
@@ -487,7 +524,9 @@ class Planner:
             d: Dict[str, Any] = dict(dtype=op_dtype, w_off=w_off,
                                      scale_off=s_off, bias_off=b_off,
                                      in_off=offsets[op.inputs[0]],
-                                     out_off=offsets[op.output])
+                                     out_off=offsets[op.output],
+                                     fork=op.params.get("fork", 0),
+                                     join=op.params.get("join", 0))
             if len(op.inputs) > 1:
                 d["in2_off"] = offsets[op.inputs[1]]
             if op.kind == K_CONV:
