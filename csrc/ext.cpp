@@ -68,6 +68,9 @@ void* test_scratch(size_t need) {
   if (need > cap) {
     if (p) retired.push_back(p);
     TRT_HIP_CHECK(hipMalloc(&p, need));
+    // zero the split-K slice-counter prefix (fused reduce contract)
+    TRT_HIP_CHECK(
+        hipMemset(p, 0, std::min<size_t>(need, kSplitkCtrPrefixBytes)));
     cap = need;
   }
   return need ? p : nullptr;

@@ -5,6 +5,11 @@
 
 namespace trtlab {
 
+// Split-K scratch slabs lead with a counter area that MUST be zeroed once
+// at allocation (the fused in-kernel reduce counts finished slices there;
+// counters self-reset after each use). Keep in sync with gemm_common.h.
+constexpr int kSplitkCtrPrefixBytes = 16384;
+
 // dtype: 0 = fp16, 1 = bf16, 2 = int8 (symmetric, per-channel weights)
 void launch_gemm_bt(int dtype, const void* A, const void* B, void* C,
                     const float* scale, const float* bias, const void* residual,

@@ -83,8 +83,14 @@ ExecutionContext::ExecutionContext(std::shared_ptr<Engine> engine,
   } else {
     arena_ = (char*)device_malloc(eng_->arena_bytes(), eng_->device());
   }
-  if (eng_->scratch_bytes())
+  if (eng_->scratch_bytes()) {
     scratch_ = (char*)device_malloc(eng_->scratch_bytes(), eng_->device());
+    // the fused split-K reduce's slice counters must start at zero (they
+    // self-reset after every use, so once is enough)
+    TRT_HIP_CHECK(hipMemset(scratch_, 0,
+                            std::min<size_t>(eng_->scratch_bytes(),
+                                             kSplitkCtrPrefixBytes)));
+  }
   // Carve one pinned slab per direction into per-binding regions, each
   // 256-B aligned (reference Buffers::CreateBindings carving pattern).
   for (const BindingDesc& b : eng_->inputs()) {
@@ -107,8 +113,12 @@ ExecutionContext::ExecutionContext(std::shared_ptr<Engine> engine,
     fork_ev_.resize(2 * nfork);
     for (auto& e : fork_ev_)
       TRT_HIP_CHECK(hipEventCreateWithFlags(&e, hipEventDisableTiming));
-    if (eng_->scratch_bytes())
+    if (eng_->scratch_bytes()) {
       scratch2_ = (char*)device_malloc(eng_->scratch_bytes(), eng_->device());
+      TRT_HIP_CHECK(hipMemset(scratch2_, 0,
+                              std::min<size_t>(eng_->scratch_bytes(),
+                                               kSplitkCtrPrefixBytes)));
+    }
   }
 }
 

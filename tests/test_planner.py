@@ -246,7 +246,8 @@ def test_fork_join_downsample_marking():
     from trtlab_amd.engine.planner import K_CONV, Planner
     from trtlab_amd.models import build_resnet
 
-    plan = Planner().compile(build_resnet(50, batch=2, image=64, seed=0))
+    plan = Planner(fork_join=True).compile(
+        build_resnet(50, batch=2, image=64, seed=0))
     forks = [i for i, d in enumerate(plan.ops) if d.get("fork")]
     joins = [i for i, d in enumerate(plan.ops) if d.get("join")]
     assert len(forks) == 4 and len(joins) == 4  # one per stage transition

@@ -123,10 +123,17 @@ class Planner:
     with quantize/dequant staging ops — BASELINE config 3."""
 
     def __init__(self, dtype: int = DT_F16, reuse: bool = True,
-                 calib_sample=None):
+                 calib_sample=None, fork_join: bool = False):
         self.dtype = dtype
         self.reuse = reuse  # False: disjoint arena slots (debugging)
         self.calib_sample = calib_sample  # int8 activation calibration input
+        # fork_join: dual-stream downsample overlap inside the captured
+        # graph. MEASURED ON MI355X (gpurun_out/check4): it REGRESSES
+        # pipelined throughput 20.4k -> 13.1k inf/s rn50 b8 — with 3
+        # contexts in flight the GPU is already saturated and the event
+        # nodes + CU contention outweigh the overlap. Keep OFF for serving;
+        # useful only for single-inflight latency experiments.
+        self.fork_join = fork_join
 
     # ------------------------------------------------------------- fusion
     def fuse(self, g: Graph) -> List[ExecOp]:
@@ -446,6 +453,8 @@ class Planner:
         fork_pairs: List[Tuple[int, int]] = []
         last_join = -1
         for i, op in enumerate(exec_ops):
+            if not self.fork_join:
+                break
             if op.kind != K_CONV or i <= last_join:
                 continue
             cons = consumers_of.get(op.output, [])
