@@ -68,9 +68,6 @@ void* test_scratch(size_t need) {
   if (need > cap) {
     if (p) retired.push_back(p);
     TRT_HIP_CHECK(hipMalloc(&p, need));
-    // zero the split-K slice-counter prefix (fused reduce contract)
-    TRT_HIP_CHECK(
-        hipMemset(p, 0, std::min<size_t>(need, kSplitkCtrPrefixBytes)));
     cap = need;
   }
   return need ? p : nullptr;
@@ -338,6 +335,35 @@ PYBIND11_MODULE(_C, m) {
           py::arg("dtype"), py::arg("op"), py::arg("a"), py::arg("b") = 0,
           py::arg("out") = 0, py::arg("n") = 0, py::arg("stream") = 0,
           py::arg("sync") = true);
+  ops.def("clip",
+          [](int dtype, uintptr_t in, uintptr_t out, int64_t n, float mn,
+             float mx, uintptr_t stream, bool sync) {
+            launch_clip(dtype, (void*)in, (void*)out, n, mn, mx,
+                        as_stream(stream));
+            if (sync) TRT_HIP_CHECK(hipStreamSynchronize(as_stream(stream)));
+          },
+          py::arg("dtype"), py::arg("in"), py::arg("out"), py::arg("n"),
+          py::arg("mn"), py::arg("mx"), py::arg("stream") = 0,
+          py::arg("sync") = true);
+  ops.def("transpose2d",
+          [](int dtype, uintptr_t in, uintptr_t out, int M, int N,
+             uintptr_t stream, bool sync) {
+            launch_transpose2d(dtype, (void*)in, (void*)out, M, N,
+                               as_stream(stream));
+            if (sync) TRT_HIP_CHECK(hipStreamSynchronize(as_stream(stream)));
+          },
+          py::arg("dtype"), py::arg("in"), py::arg("out"), py::arg("M"),
+          py::arg("N"), py::arg("stream") = 0, py::arg("sync") = true);
+  ops.def("copy2d",
+          [](int dtype, uintptr_t src, uintptr_t dst, int64_t M, int C,
+             int ldd, int coff, uintptr_t stream, bool sync) {
+            launch_copy2d(dtype, (void*)src, (void*)dst, M, C, ldd, coff,
+                          as_stream(stream));
+            if (sync) TRT_HIP_CHECK(hipStreamSynchronize(as_stream(stream)));
+          },
+          py::arg("dtype"), py::arg("src"), py::arg("dst"), py::arg("M"),
+          py::arg("C"), py::arg("ldd"), py::arg("coff") = 0,
+          py::arg("stream") = 0, py::arg("sync") = true);
   ops.def("channel_pad",
           [](int dtype, uintptr_t in, uintptr_t out, int64_t M, int Cin,
              int Cpad, uintptr_t stream, bool sync) {

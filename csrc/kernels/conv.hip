@@ -70,6 +70,12 @@ __device__ __forceinline__ void stage_conv_a(
   }
 }
 
+void launch_splitk_reduce(int dtype, const float* scratch, void* C,
+                          const float* scale, const float* bias,
+                          const void* residual, float res_scale, int M, int N,
+                          int64_t ldc, int tiles_m, int tiles_n, int splitk,
+                          int bm, int bn, int epi, hipStream_t stream);
+
 template <typename T, Epi E, int BM, int BN, bool SPLIT, int NBUF>
 __global__ __launch_bounds__(256) void conv_igemm_kernel(
     const T* __restrict__ in, const T* __restrict__ Wt, T* __restrict__ out,
@@ -144,11 +150,8 @@ __global__ __launch_bounds__(256) void conv_igemm_kernel(
   }
 
   if constexpr (SPLIT) {
-    // fused reduce: the last slice's block sums the partials and applies
-    // the epilogue — no separate splitk_reduce launch
-    splitk_store_and_reduce<T, T, E, BM, BN>(
-        acc, scratch, bid, tile, splitk, out, scale, bias, residual,
-        p.res_scale, 1.0f, m0, n0, p.M, p.Cout, p.Cout, lane, wr, wc, tid);
+    store_splitk<T, BM, BN>(acc, scratch + (int64_t)bid * BM * BN, lane, wr,
+                            wc);
   } else {
     store_epilogue<T, E, BM, BN>(acc, out, p.Cout, m0, n0, p.M, p.Cout, scale,
                                  bias, residual, p.res_scale, lane, wr, wc);
@@ -165,8 +168,7 @@ size_t conv_scratch_bytes(int Nb, int H, int W, int C, int Cout, int KH,
   long tiles = cdiv(M, cfg.bm) * cdiv(Cout, cfg.bn);
   int splitk = pick_splitk(tiles, K >> 6);
   if (splitk == 1) return 0;
-  // leading counter area + fp32 partials (see splitk_store_and_reduce)
-  return kSplitkCtrBytes + (size_t)tiles * splitk * cfg.bm * cfg.bn * 4;
+  return (size_t)tiles * splitk * cfg.bm * cfg.bn * 4;
 }
 
 // ---- direct-to-VGPR small-K fast path ----
@@ -281,25 +283,25 @@ static void launch_conv2d_t(const void* in, const void* Wt, void* out,
     int ktper = (int)cdiv(ktiles, splitk);
     dim3 grid((unsigned)(tiles * splitk));
     bool deep = want_deep_pipe(tiles * splitk, ktper);
-    epi_dispatch(epi, [&](auto e) {
-      constexpr Epi EE = decltype(e)::value;
-      tile_dispatch(cfg, [&](auto bm, auto bn) {
-        constexpr int BM = decltype(bm)::value;
-        constexpr int BN = decltype(bn)::value;
-        if (deep)
-          hipLaunchKernelGGL((conv_igemm_kernel<T, EE, BM, BN, true, 4>),
-                             grid, block, 0, stream, (const T*)in,
-                             (const T*)Wt, (T*)out, scale, bias,
-                             (const T*)residual, (const T*)zero_page, p,
-                             tiles_n, scratch, splitk, ktper);
-        else
-          hipLaunchKernelGGL((conv_igemm_kernel<T, EE, BM, BN, true, 2>),
-                             grid, block, 0, stream, (const T*)in,
-                             (const T*)Wt, (T*)out, scale, bias,
-                             (const T*)residual, (const T*)zero_page, p,
-                             tiles_n, scratch, splitk, ktper);
-      });
+    tile_dispatch(cfg, [&](auto bm, auto bn) {
+      constexpr int BM = decltype(bm)::value;
+      constexpr int BN = decltype(bn)::value;
+      if (deep)
+        hipLaunchKernelGGL((conv_igemm_kernel<T, Epi::kNone, BM, BN, true, 4>),
+                           grid, block, 0, stream, (const T*)in, (const T*)Wt,
+                           (T*)out, scale, bias, (const T*)residual,
+                           (const T*)zero_page, p, tiles_n, scratch, splitk,
+                           ktper);
+      else
+        hipLaunchKernelGGL((conv_igemm_kernel<T, Epi::kNone, BM, BN, true, 2>),
+                           grid, block, 0, stream, (const T*)in, (const T*)Wt,
+                           (T*)out, scale, bias, (const T*)residual,
+                           (const T*)zero_page, p, tiles_n, scratch, splitk,
+                           ktper);
     });
+    launch_splitk_reduce(dtype, scratch, out, scale, bias, residual,
+                         p.res_scale, p.M, p.Cout, p.Cout, tiles_m, tiles_n,
+                         splitk, cfg.bm, cfg.bn, epi, stream);
     return;
   }
   dim3 grid((unsigned)tiles);

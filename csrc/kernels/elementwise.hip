@@ -247,4 +247,112 @@ void launch_cast(int dtype, bool to_f32, const void* in, void* out, int64_t n,
   }
 }
 
+
+// ---- general clip: out = min(max(x, mn), mx) (ONNX Clip with bounds) ----
+template <typename T>
+__global__ void clip_kernel(const T* __restrict__ in, T* __restrict__ out,
+                            int64_t n8, float mn, float mx) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n8;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    short4v a0 = *(const short4v*)((const T*)in + i * 8);
+    short4v a1 = *(const short4v*)((const T*)in + i * 8 + 4);
+    T o[8];
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      o[j] = (T)fminf(fmaxf((float)((const T*)&a0)[j], mn), mx);
+      o[4 + j] = (T)fminf(fmaxf((float)((const T*)&a1)[j], mn), mx);
+    }
+    *(short4v*)(out + i * 8) = *(const short4v*)&o[0];
+    *(short4v*)(out + i * 8 + 4) = *(const short4v*)&o[4];
+  }
+}
+
+void launch_clip(int dtype, const void* in, void* out, int64_t n, float mn,
+                 float mx, hipStream_t stream) {
+  if (n % 8 != 0) throw std::runtime_error("clip: n % 8 != 0");
+  int blocks = ew_blocks(n / 8);
+  if (dtype == 0)
+    hipLaunchKernelGGL((clip_kernel<_Float16>), dim3(blocks), dim3(256), 0,
+                       stream, (const _Float16*)in, (_Float16*)out, n / 8,
+                       mn, mx);
+  else
+    hipLaunchKernelGGL((clip_kernel<__bf16>), dim3(blocks), dim3(256), 0,
+                       stream, (const __bf16*)in, (__bf16*)out, n / 8, mn,
+                       mx);
+}
+
+// ---- tiled 2-D transpose: out[N][M] = in[M][N]^T ----
+// 64x64 tiles through LDS with a +1-element row skew (65*2 B rows) so both
+// the coalesced row reads and the transposed column reads are conflict-
+// free. SURVEY.md §2.8 item 9 (general layout/transpose kernel).
+template <typename T>
+__global__ __launch_bounds__(256) void transpose2d_kernel(
+    const T* __restrict__ in, T* __restrict__ out, int M, int N,
+    int tiles_n) {
+  __shared__ T tile[64][65];
+  int tm = blockIdx.x / tiles_n, tn = blockIdx.x % tiles_n;
+  int r0 = tm * 64, c0 = tn * 64;
+  // 256 threads: each loads 16 rows x 4-elem chunks? simpler: 64x4 layout
+  int lr = threadIdx.x & 63;        // column within the tile row group
+  int lw = threadIdx.x >> 6;        // 0..3
+#pragma unroll
+  for (int rr = 0; rr < 16; ++rr) {
+    int row = r0 + lw * 16 + rr;
+    int col = c0 + lr;
+    if (row < M && col < N) tile[lw * 16 + rr][lr] = in[(int64_t)row * N + col];
+  }
+  __syncthreads();
+#pragma unroll
+  for (int rr = 0; rr < 16; ++rr) {
+    int orow = c0 + lw * 16 + rr;   // output row = input col
+    int ocol = r0 + lr;             // output col = input row
+    if (orow < N && ocol < M)
+      out[(int64_t)orow * M + ocol] = tile[lr][lw * 16 + rr];
+  }
+}
+
+void launch_transpose2d(int dtype, const void* in, void* out, int M, int N,
+                        hipStream_t stream) {
+  int tiles_m = (int)cdiv(M, 64), tiles_n = (int)cdiv(N, 64);
+  dim3 grid((unsigned)(tiles_m * tiles_n));
+  if (dtype == 0)
+    hipLaunchKernelGGL((transpose2d_kernel<_Float16>), grid, dim3(256), 0,
+                       stream, (const _Float16*)in, (_Float16*)out, M, N,
+                       tiles_n);
+  else
+    hipLaunchKernelGGL((transpose2d_kernel<__bf16>), grid, dim3(256), 0,
+                       stream, (const __bf16*)in, (__bf16*)out, M, N,
+                       tiles_n);
+}
+
+// ---- strided 2-D copy: dst[m][coff + c] = src[m][c] (concat lowering) ----
+// Each ONNX Concat input becomes one copy of its [M, C] block into the
+// output's column range at offset coff with destination row stride ldd.
+template <typename T>
+__global__ void copy2d_kernel(const T* __restrict__ src, T* __restrict__ dst,
+                              int64_t M, int C, int ldd) {
+  int64_t total = M * (int64_t)C;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int64_t m = i / C;
+    int c = (int)(i - m * C);
+    dst[m * ldd + c] = src[i];
+  }
+}
+
+void launch_copy2d(int dtype, const void* src, void* dst, int64_t M, int C,
+                   int ldd, int coff, hipStream_t stream) {
+  int64_t total = M * (int64_t)C;
+  int blocks = ew_blocks(total);
+  if (dtype == 0)
+    hipLaunchKernelGGL((copy2d_kernel<_Float16>), dim3(blocks), dim3(256), 0,
+                       stream, (const _Float16*)src, (_Float16*)dst + coff,
+                       M, C, ldd);
+  else
+    hipLaunchKernelGGL((copy2d_kernel<__bf16>), dim3(blocks), dim3(256), 0,
+                       stream, (const __bf16*)src, (__bf16*)dst + coff, M, C,
+                       ldd);
+}
+
 }  // namespace trtlab
+

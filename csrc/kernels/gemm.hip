@@ -24,6 +24,12 @@
 
 namespace trtlab {
 
+void launch_splitk_reduce(int dtype, const float* scratch, void* C,
+                          const float* scale, const float* bias,
+                          const void* residual, float res_scale, int M, int N,
+                          int64_t ldc, int tiles_m, int tiles_n, int splitk,
+                          int bm, int bn, int epi, hipStream_t stream);
+
 template <typename T, typename OT, Epi E, int BM, int BN, bool SPLIT,
           int NBUF>
 __global__ __launch_bounds__(256) void gemm_bt_kernel(
@@ -117,11 +123,8 @@ __global__ __launch_bounds__(256) void gemm_bt_kernel(
   }
 
   if constexpr (SPLIT) {
-    // fused reduce: the last slice's block sums the partials and applies
-    // the epilogue — no separate splitk_reduce launch
-    splitk_store_and_reduce<T, OT, E, BM, BN>(
-        acc, scratch, bid, tile, splitk, C, scale, bias, residual, res_scale,
-        out_scale, m0, n0, M, N, ldc, lane, wr, wc, (int)threadIdx.x);
+    store_splitk<T, BM, BN>(acc, scratch + (int64_t)bid * BM * BN, lane, wr,
+                            wc);
   } else {
     store_epilogue<T, E, BM, BN, OT>(acc, C, ldc, m0, n0, M, N, scale, bias,
                                      residual, res_scale, lane, wr, wc,
@@ -134,8 +137,7 @@ size_t gemm_scratch_bytes(int M, int N, int K) {
   long tiles = cdiv(M, cfg.bm) * cdiv(N, cfg.bn);
   int splitk = pick_splitk_gemm(tiles, K >> 6);  // fp16 tiles (conservative)
   if (splitk == 1) return 0;
-  // leading counter area + fp32 partials (see splitk_store_and_reduce)
-  return kSplitkCtrBytes + (size_t)tiles * splitk * cfg.bm * cfg.bn * 4;
+  return (size_t)tiles * splitk * cfg.bm * cfg.bn * 4;
 }
 
 template <typename T, typename OT = T>
@@ -175,25 +177,25 @@ static void launch_gemm_bt_t(const void* A, const void* B, void* C,
     int ktper = (int)cdiv(ktiles, splitk);
     dim3 grid((unsigned)(tiles * splitk));
     bool deep = want_deep_pipe(tiles * splitk, ktper);
-    epi_dispatch(epi, [&](auto e) {
-      constexpr Epi EE = decltype(e)::value;
-      tile_dispatch(cfg, [&](auto bm, auto bn) {
-        constexpr int BM = decltype(bm)::value;
-        constexpr int BN = decltype(bn)::value;
-        if (deep)
-          hipLaunchKernelGGL((gemm_bt_kernel<T, OT, EE, BM, BN, true, 4>),
-                             grid, block, 0, stream, (const T*)A, (const T*)B,
-                             (OT*)C, scale, bias, (const OT*)residual,
-                             res_scale, out_scale, M, N, K, lda, ldb, ldc,
-                             tiles_n, scratch, splitk, ktper);
-        else
-          hipLaunchKernelGGL((gemm_bt_kernel<T, OT, EE, BM, BN, true, 2>),
-                             grid, block, 0, stream, (const T*)A, (const T*)B,
-                             (OT*)C, scale, bias, (const OT*)residual,
-                             res_scale, out_scale, M, N, K, lda, ldb, ldc,
-                             tiles_n, scratch, splitk, ktper);
-      });
+    tile_dispatch(cfg, [&](auto bm, auto bn) {
+      constexpr int BM = decltype(bm)::value;
+      constexpr int BN = decltype(bn)::value;
+      if (deep)
+        hipLaunchKernelGGL((gemm_bt_kernel<T, OT, Epi::kNone, BM, BN, true, 4>),
+                           grid, block, 0, stream, (const T*)A, (const T*)B,
+                           (OT*)C, scale, bias, (const OT*)residual,
+                           res_scale, out_scale, M, N, K, lda, ldb, ldc,
+                           tiles_n, scratch, splitk, ktper);
+      else
+        hipLaunchKernelGGL((gemm_bt_kernel<T, OT, Epi::kNone, BM, BN, true, 2>),
+                           grid, block, 0, stream, (const T*)A, (const T*)B,
+                           (OT*)C, scale, bias, (const OT*)residual,
+                           res_scale, out_scale, M, N, K, lda, ldb, ldc,
+                           tiles_n, scratch, splitk, ktper);
     });
+    launch_splitk_reduce(out_dtype, scratch, C, scale, bias, residual,
+                         res_scale, M, N, ldc, tiles_m, tiles_n, splitk,
+                         cfg.bm, cfg.bn, epi, stream);
     return;
   }
   dim3 grid((unsigned)tiles);

@@ -292,81 +292,6 @@ __device__ __forceinline__ void store_splitk(
     }
 }
 
-// ---- fused split-K reduce (no separate launch) ----
-// Split-K slabs lead with a fixed counter area: ctr[tile] counts finished
-// slices; the LAST slice's block reduces the tile's partials and applies
-// the epilogue in-kernel (deterministic: slices are summed in fixed
-// order). This removes the separate reduce launch that was 8-11% of conv
-// time on its shapes (profiles/README PMC section). Counters self-reset
-// to 0 so graph replays need no re-initialization; the slab's counter
-// prefix is zeroed ONCE at allocation.
-constexpr int kSplitkCtrBytes = 16384;  // up to 4096 tiles (gate: <=144)
-
-template <typename OT, Epi E, int BM, int BN>
-__device__ __forceinline__ void reduce_splitk_tile(
-    const float* __restrict__ tile_base,  // splitk slices of [BM][BN] fp32
-    int splitk, OT* __restrict__ C, const float* __restrict__ scale,
-    const float* __restrict__ bias, const OT* __restrict__ residual,
-    float res_scale, float out_scale, int m0, int n0, int M, int N,
-    int64_t ldc, int tid) {
-  constexpr int elems = BM * BN;
-  for (int e4 = tid * 4; e4 < elems; e4 += 256 * 4) {
-    float4v v = *(const float4v*)(tile_base + e4);
-    for (int s = 1; s < splitk; ++s) {
-      const float4v u = *(const float4v*)(tile_base + s * elems + e4);
-      v.x += u.x; v.y += u.y; v.z += u.z; v.w += u.w;
-    }
-    int row = m0 + e4 / BN;
-    int col0 = n0 + (e4 % BN);  // 4 consecutive cols (BN % 4 == 0)
-    if (row >= M) continue;
-    float vv[4] = {v.x, v.y, v.z, v.w};
-#pragma unroll
-    for (int j = 0; j < 4; ++j) {
-      int col = col0 + j;
-      if (col >= N) continue;
-      float sc = 1.0f, bi = 0.0f;
-      if constexpr (E == Epi::kScaleBias || E == Epi::kScaleBiasRelu ||
-                    E == Epi::kScaleBiasAddRelu || E == Epi::kScaleBiasGelu)
-        sc = scale[col];
-      if constexpr (E != Epi::kNone) bi = bias[col];
-      float res = 0.0f;
-      if constexpr (E == Epi::kScaleBiasAddRelu)
-        res = (float)residual[(int64_t)row * ldc + col] * res_scale;
-      C[(int64_t)row * ldc + col] =
-          store_cast<OT>(apply_epi<E>(vv[j], sc, bi, res) * out_scale);
-    }
-  }
-}
-
-// Tail sequence for split-K kernels: store partials, count completion,
-// last block reduces. Returns only after this block's role is done.
-template <typename T, typename OT, Epi E, int BM, int BN>
-__device__ __forceinline__ void splitk_store_and_reduce(
-    typename Mfma16x16x32<T>::accv (&acc)[BM / 32][BN / 32],
-    float* __restrict__ scratch, uint32_t bid, uint32_t tile, int splitk,
-    OT* __restrict__ C, const float* __restrict__ scale,
-    const float* __restrict__ bias, const OT* __restrict__ residual,
-    float res_scale, float out_scale, int m0, int n0, int M, int N,
-    int64_t ldc, int lane, int wr, int wc, int tid) {
-  float* partials = scratch + kSplitkCtrBytes / 4;
-  store_splitk<T, BM, BN>(acc, partials + (int64_t)bid * BM * BN, lane, wr,
-                          wc);
-  __threadfence();  // release: partials visible device-wide (cross-XCD L2)
-  __shared__ int last;
-  if (tid == 0) {
-    int* ctr = (int*)scratch;
-    last = (atomicAdd(&ctr[tile], 1) == splitk - 1);
-    if (last) atomicExch(&ctr[tile], 0);  // self-reset for the next replay
-  }
-  __syncthreads();
-  if (last) {
-    __threadfence();  // acquire: observe every slice's partials
-    reduce_splitk_tile<OT, E, BM, BN>(
-        partials + (int64_t)tile * splitk * BM * BN, splitk, C, scale, bias,
-        residual, res_scale, out_scale, m0, n0, M, N, ldc, tid);
-  }
-}
-
 // Split-K decision: only for severely grid-starved, K-heavy shapes (the
 // slab round-trip + reduce launch costs ~8-10 us, so marginal cases lose —
 // measured in tools/tune_tiles.py). Returns 1 = no split.

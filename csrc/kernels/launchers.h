@@ -5,11 +5,6 @@
 
 namespace trtlab {
 
-// Split-K scratch slabs lead with a counter area that MUST be zeroed once
-// at allocation (the fused in-kernel reduce counts finished slices there;
-// counters self-reset after each use). Keep in sync with gemm_common.h.
-constexpr int kSplitkCtrPrefixBytes = 16384;
-
 // dtype: 0 = fp16, 1 = bf16, 2 = int8 (symmetric, per-channel weights)
 void launch_gemm_bt(int dtype, const void* A, const void* B, void* C,
                     const float* scale, const float* bias, const void* residual,
@@ -53,6 +48,12 @@ void launch_add_layernorm(int dtype, const void* x, const void* res,
 
 void launch_elementwise(int dtype, int op, const void* a, const void* b,
                         void* out, int64_t n, hipStream_t stream);
+void launch_clip(int dtype, const void* in, void* out, int64_t n, float mn,
+                 float mx, hipStream_t stream);
+void launch_transpose2d(int dtype, const void* in, void* out, int M, int N,
+                        hipStream_t stream);
+void launch_copy2d(int dtype, const void* src, void* dst, int64_t M, int C,
+                   int ldd, int coff, hipStream_t stream);
 void launch_channel_pad(int dtype, const void* in, void* out, int64_t M,
                         int Cin, int Cpad, hipStream_t stream);
 void launch_cast(int dtype, bool to_f32, const void* in, void* out, int64_t n,

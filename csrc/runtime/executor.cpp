@@ -83,14 +83,8 @@ ExecutionContext::ExecutionContext(std::shared_ptr<Engine> engine,
   } else {
     arena_ = (char*)device_malloc(eng_->arena_bytes(), eng_->device());
   }
-  if (eng_->scratch_bytes()) {
+  if (eng_->scratch_bytes())
     scratch_ = (char*)device_malloc(eng_->scratch_bytes(), eng_->device());
-    // the fused split-K reduce's slice counters must start at zero (they
-    // self-reset after every use, so once is enough)
-    TRT_HIP_CHECK(hipMemset(scratch_, 0,
-                            std::min<size_t>(eng_->scratch_bytes(),
-                                             kSplitkCtrPrefixBytes)));
-  }
   // Carve one pinned slab per direction into per-binding regions, each
   // 256-B aligned (reference Buffers::CreateBindings carving pattern).
   for (const BindingDesc& b : eng_->inputs()) {
@@ -113,12 +107,8 @@ ExecutionContext::ExecutionContext(std::shared_ptr<Engine> engine,
     fork_ev_.resize(2 * nfork);
     for (auto& e : fork_ev_)
       TRT_HIP_CHECK(hipEventCreateWithFlags(&e, hipEventDisableTiming));
-    if (eng_->scratch_bytes()) {
+    if (eng_->scratch_bytes())
       scratch2_ = (char*)device_malloc(eng_->scratch_bytes(), eng_->device());
-      TRT_HIP_CHECK(hipMemset(scratch2_, 0,
-                              std::min<size_t>(eng_->scratch_bytes(),
-                                               kSplitkCtrPrefixBytes)));
-    }
   }
 }
 
@@ -278,6 +268,20 @@ void ExecutionContext::enqueue_all(hipStream_t s) {
         launch_embedding(op.dtype, A(op.in_off), Wp(op.w_off),
                          Wp(op.scale_off), Wp(op.bias_off), A(op.in2_off),
                          A(op.out_off), op.M, op.S, op.N, os);
+        break;
+      case kClip:
+        // res_scale = min bound, q_scale = max bound
+        launch_clip(op.dtype, A(op.in_off), A(op.out_off), op.n_elems,
+                    op.res_scale, op.q_scale, os);
+        break;
+      case kTranspose2D:
+        launch_transpose2d(op.dtype, A(op.in_off), A(op.out_off), op.M, op.N,
+                           os);
+        break;
+      case kCopy2D:
+        // epi = destination column offset; Cout = destination row stride
+        launch_copy2d(op.dtype, A(op.in_off), A(op.out_off), op.M, op.C,
+                      op.Cout, op.epi, os);
         break;
       default:
         throw std::runtime_error("unknown op kind");

@@ -110,6 +110,9 @@ enum OpKind : int {
   kGemmMx4 = 16,   // MXFP4 x MXFP4 scaled-MFMA GEMM, fp16 out + epilogue
   kQuantMx8 = 17,  // fp16 rows -> MXFP8 (e4m3) codes + e8m0 block scales
   kGemmMx8 = 18,   // MXFP8 x MXFP8 scaled-MFMA GEMM, fp16 out + epilogue
+  kClip = 19,        // out = min(max(x, res_scale), q_scale) — ONNX Clip
+  kTranspose2D = 20, // out[N][M] = in[M][N]^T (tiled LDS transpose)
+  kCopy2D = 21,      // dst[m][epi + c] = src[m][c], dst row stride Cout
 };
 
 struct OpDesc {

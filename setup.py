@@ -20,6 +20,7 @@ SOURCES = [
     "csrc/kernels/gemm_mx.hip",
     "csrc/kernels/decode.hip",
     "csrc/kernels/conv.hip",
+    "csrc/kernels/splitk.hip",
     "csrc/kernels/pool.hip",
     "csrc/kernels/normalize.hip",
     "csrc/kernels/elementwise.hip",
