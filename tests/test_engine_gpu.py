@@ -801,3 +801,57 @@ def test_two_models_share_one_arena():
         err = np.abs(out - ref).max() / max(np.abs(ref).max(), 1e-6)
         assert err < 0.08, (name, err)
     mgr.shutdown()
+
+
+def test_stock_style_resnet50_onnx_on_gpu():
+    """The torch-export-style resnet50.onnx (independent plain-torch
+    oracle, tools/torch_resnet.py) imports unmodified and the captured
+    engine matches the torch module's forward."""
+    import sys
+    from pathlib import Path
+
+    sys.path.insert(0, str(Path(__file__).resolve().parent.parent / "tools"))
+    import torch
+    from torch_resnet import TorchResNet, export_resnet_onnx
+
+    from trtlab_amd.engine.onnx_io import import_onnx
+    from trtlab_amd.engine.planner import Planner
+    from trtlab_amd.engine.runtime import NativeEngine
+
+    m = TorchResNet(layers=(3, 4, 6, 3), seed=3)
+    g = import_onnx(export_resnet_onnx(m, batch=2, image=64))
+    plan = Planner().compile(g)
+    eng = NativeEngine(plan)
+    ctx = eng.create_context(capture=True)
+    x = (np.random.RandomState(1).randn(2, 64, 64, 3) * 0.5).astype(
+        np.float32)
+    out = ctx.infer(x).astype(np.float32)
+    with torch.no_grad():
+        ref = m(torch.from_numpy(x).permute(0, 3, 1, 2)).numpy()
+    err = np.abs(out - ref).max() / max(np.abs(ref).max(), 1e-6)
+    assert err < 0.08, err
+
+
+def test_concat_clip_engine_path():
+    """Concat + general Clip through the captured engine (ONNX breadth
+    kernels: copy2d + clip) vs the fp32 reference."""
+    from trtlab_amd.engine.ir import Graph
+    from trtlab_amd.engine.planner import Planner
+    from trtlab_amd.engine.reference import run_reference
+    from trtlab_amd.engine.runtime import NativeEngine
+
+    rng = np.random.RandomState(0)
+    g = Graph("cc")
+    x = g.input((64, 128), name="x")
+    a = g.clip(x, -0.5, 1.5, name="a")
+    b = g.gemm(x, (rng.randn(64, 128) * 0.1).astype(np.float32),
+               (rng.randn(64) * 0.1).astype(np.float32), name="b")
+    g.concat([a, b], name="cat")
+    plan = Planner().compile(g)
+    eng = NativeEngine(plan)
+    ctx = eng.create_context(capture=True)
+    xv = rng.randn(64, 128).astype(np.float32)
+    out = ctx.infer(xv).astype(np.float32)
+    ref = run_reference(plan, xv)
+    err = np.abs(out - ref).max() / max(np.abs(ref).max(), 1e-6)
+    assert err < 0.05, err

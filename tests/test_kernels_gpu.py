@@ -502,3 +502,32 @@ def test_gemm_bt_strided_operands(C):
     check(c_full[:, :N], ref)
     # untouched tail of each C row stays zero
     assert (c_full[:, N:] == 0).all()
+
+
+def test_clip_kernel(C):
+    x = t16(64, 256, seed=91, scale=3.0)
+    out = torch.empty_like(x)
+    torch.cuda.synchronize()
+    C.ops.clip(0, x.data_ptr(), out.data_ptr(), x.numel(), -1.5, 2.0)
+    ref = torch.clamp(x.float(), -1.5, 2.0)
+    check(out, ref, rtol=1e-3, atol=1e-3)
+
+
+@pytest.mark.parametrize("M,N", [(64, 64), (100, 200), (1000, 768), (65, 1)])
+def test_transpose2d_kernel(C, M, N):
+    x = t16(M, N, seed=92)
+    out = torch.empty(N, M, dtype=torch.half, device="cuda")
+    torch.cuda.synchronize()
+    C.ops.transpose2d(0, x.data_ptr(), out.data_ptr(), M, N)
+    assert torch.equal(out, x.t().contiguous())
+
+
+def test_copy2d_concat(C):
+    a = t16(128, 96, seed=93)
+    b = t16(128, 160, seed=94)
+    out = torch.zeros(128, 256, dtype=torch.half, device="cuda")
+    torch.cuda.synchronize()
+    C.ops.copy2d(0, a.data_ptr(), out.data_ptr(), 128, 96, 256, 0)
+    C.ops.copy2d(0, b.data_ptr(), out.data_ptr(), 128, 160, 256, 96)
+    ref = torch.cat([a, b], dim=1)
+    assert torch.equal(out, ref)

@@ -163,3 +163,109 @@ def test_onnx_roundtrip_resnet18_basic_blocks():
     out = run_reference(plan, x)
     ref = run_reference(Planner().compile(g), x)
     assert np.allclose(out, ref, atol=1e-4)
+
+
+# ---------------------------------------------------------------------------
+# Stock-export parity: a torch.onnx.export-style
+# resnet50.onnx (unfused BatchNormalization + Flatten + Gemm transB=1,
+# generated by the INDEPENDENT plain-torch model in tools/torch_resnet.py)
+# imports unmodified and matches the torch module's forward.
+
+def _tools():
+    import sys
+    from pathlib import Path
+
+    sys.path.insert(0, str(Path(__file__).resolve().parent.parent / "tools"))
+
+
+def test_stock_style_resnet50_import_matches_torch():
+    import torch
+
+    _tools()
+    from torch_resnet import TorchResNet, export_resnet_onnx
+
+    from trtlab_amd.engine.onnx_io import import_onnx
+    from trtlab_amd.engine.planner import Planner
+    from trtlab_amd.engine.reference import run_reference
+
+    m = TorchResNet(layers=(3, 4, 6, 3), seed=3)  # real resnet50 depth
+    data = export_resnet_onnx(m, batch=1, image=64)
+    g = import_onnx(data)
+    plan = Planner().compile(g)
+    # the full conv inventory made it through the importer + fusion
+    assert sum(1 for d in plan.ops if d["kind"] == 0) == 53  # convs
+    x = (np.random.RandomState(0).randn(1, 64, 64, 3) * 0.5).astype(
+        np.float32)
+    out = run_reference(plan, x)
+    with torch.no_grad():
+        ref = m(torch.from_numpy(x).permute(0, 3, 1, 2)).numpy()
+    err = np.abs(out - ref).max() / max(np.abs(ref).max(), 1e-6)
+    assert err < 0.03, err
+
+
+def _mini_onnx(nodes, inits, in_name, in_dims, out_name, out_dims):
+    from trtlab_amd.engine import onnx_wire as w
+    from trtlab_amd.engine.onnx_io import (_GRAPH_INIT, _GRAPH_INPUT,
+                                           _GRAPH_NAME, _GRAPH_OUTPUT,
+                                           _MODEL_GRAPH, _tensor_bytes,
+                                           _value_info)
+
+    gparts = [w.f_string(_GRAPH_NAME, "mini")]
+    gparts += nodes
+    gparts += [w.f_bytes(_GRAPH_INIT, _tensor_bytes(n, a))
+               for n, a in inits]
+    gparts.append(w.f_bytes(_GRAPH_INPUT, _value_info(in_name, in_dims)))
+    gparts.append(w.f_bytes(_GRAPH_OUTPUT, _value_info(out_name, out_dims)))
+    return w.f_bytes(_MODEL_GRAPH, b"".join(gparts))
+
+
+def test_import_nary_sum_concat_transpose_clip():
+    """The round-2 importer breadth ops: Sum>2 inputs, Concat, Transpose
+    chain (NCHW<->NHWC cancel + 2-D transpose), general Clip bounds."""
+    import torch
+
+    from trtlab_amd.engine.onnx_io import _attr_f, _attr_ints, _node, import_onnx
+    from trtlab_amd.engine.planner import Planner
+    from trtlab_amd.engine.reference import run_reference
+
+    rng = np.random.RandomState(1)
+    w1 = (rng.randn(32, 16) * 0.2).astype(np.float32)  # Gemm B [N,K] via transB? use MatMul style [K,N]
+    nodes = [
+        # three branches of the input, then 3-ary Sum
+        _node("Relu", ["x"], ["a"]),
+        _node("Clip", ["x"], ["b"], _attr_f("min", -0.5), _attr_f("max", 0.5)),
+        _node("Identity", ["x"], ["c"]),
+        _node("Sum", ["a", "b", "c"], ["s"]),
+        # Concat the sum with the clipped branch along the feature axis
+        _node("Concat", ["s", "b"], ["cat"], _attr_ints("axis", [1])),
+        # transpose chain: T then T back (cancels shape-wise via kernel)
+        _node("Transpose", ["cat"], ["t1"], _attr_ints("perm", [1, 0])),
+        _node("Transpose", ["t1"], ["out"], _attr_ints("perm", [1, 0])),
+    ]
+    data = _mini_onnx(nodes, [], "x", [8, 16], "out", [8, 32])
+    g = import_onnx(data)
+    plan = Planner().compile(g)
+    x = rng.randn(8, 16).astype(np.float32)
+    out = run_reference(plan, x)
+    xt = torch.from_numpy(x)
+    s = torch.relu(xt) + torch.clamp(xt, -0.5, 0.5) + xt
+    ref = torch.cat([s, torch.clamp(xt, -0.5, 0.5)], dim=1).numpy()
+    assert np.allclose(out, ref, atol=1e-4), np.abs(out - ref).max()
+
+
+def test_import_nchw_nhwc_transpose_cancels():
+    from trtlab_amd.engine.onnx_io import _attr_ints, _node, import_onnx
+    from trtlab_amd.engine.planner import Planner
+    from trtlab_amd.engine.reference import run_reference
+
+    nodes = [
+        _node("Transpose", ["x"], ["t1"], _attr_ints("perm", [0, 2, 3, 1])),
+        _node("Transpose", ["t1"], ["t2"], _attr_ints("perm", [0, 3, 1, 2])),
+        _node("Relu", ["t2"], ["out"]),
+    ]
+    data = _mini_onnx(nodes, [], "x", [2, 8, 4, 4], "out", [2, 8, 4, 4])
+    g = import_onnx(data)
+    plan = Planner().compile(g)
+    x = np.random.RandomState(2).randn(2, 4, 4, 8).astype(np.float32)
+    out = run_reference(plan, x)
+    assert np.allclose(out, np.maximum(x, 0), atol=1e-6)

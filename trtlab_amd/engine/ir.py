@@ -192,6 +192,28 @@ class Graph:
                                else np.asarray(seg_table, np.float32),
                                seq=s_), name)
 
+    def clip(self, x: str, mn: float, mx: float,
+             name: Optional[str] = None) -> str:
+        """out = min(max(x, mn), mx) (ONNX Clip with arbitrary bounds)."""
+        return self._emit("clip", [x], self.tensors[x].shape,
+                          dict(mn=float(mn), mx=float(mx)), name)
+
+    def transpose2d(self, x: str, name: Optional[str] = None) -> str:
+        """out[N][M] = x[M][N]^T (2-D only; the general layout kernel)."""
+        m, n = self.tensors[x].shape
+        return self._emit("transpose2d", [x], (n, m), {}, name)
+
+    def concat(self, xs: List[str], name: Optional[str] = None) -> str:
+        """Concatenate along the LAST axis (channels in NHWC / features in
+        2-D) — the common inception/densenet pattern. Lowers to one
+        strided copy per input (kCopy2D)."""
+        shapes = [self.tensors[x].shape for x in xs]
+        lead = shapes[0][:-1]
+        assert all(s[:-1] == lead for s in shapes), \
+            f"concat: leading dims differ: {shapes}"
+        ctot = sum(s[-1] for s in shapes)
+        return self._emit("concat", list(xs), (*lead, ctot), {}, name)
+
     def attention(self, qkv: str, heads: int, seq: int,
                   varlen: bool = False, pad_id: int = 0,
                   causal: bool = False,

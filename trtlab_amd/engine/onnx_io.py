@@ -116,10 +116,15 @@ def import_onnx(data: bytes, batch: Optional[int] = None,
         nm, dims = _parse_vi_shape(vi)
         if nm in inits:
             continue
-        n, c, h, ww = dims  # NCHW
-        if batch:
-            n = batch
-        input_name = g.input((n, h, ww, c), name=nm)  # IR is NHWC
+        if len(dims) == 4:
+            n, c, h, ww = dims  # NCHW -> IR NHWC
+            if batch:
+                n = batch
+            input_name = g.input((n, h, ww, c), name=nm)
+        else:  # 2-D (features) or other ranks: layout-free
+            if batch and dims:
+                dims = [batch] + list(dims[1:])
+            input_name = g.input(tuple(dims), name=nm)
     assert input_name is not None, "no graph input found"
 
     # name remapping: ONNX tensor name -> IR tensor name
@@ -185,11 +190,41 @@ def import_onnx(data: bytes, batch: Optional[int] = None,
             if mx is None and len(ins) > 2 and ins[2] and ins[2] in inits:
                 mx = float(np.asarray(inits[ins[2]]).reshape(-1)[0])
             if (mn is None or mn == 0.0) and (mx is None or mx >= 3e38):
-                out = g.relu(x)
+                out = g.relu(x)  # ReLU-equivalent: fusable epilogue
+            else:  # general bounds (ReLU6 etc.) -> dedicated clip kernel
+                out = g.clip(x, -3e38 if mn is None else mn,
+                             3e38 if mx is None else mx)
+        elif op == "Sum":
+            # N-ary Sum -> chain of adds (2 inputs = plain Add)
+            out = remap.get(ins[0], ins[0])
+            for extra in ins[1:]:
+                out = g.add(out, remap.get(extra, extra))
+        elif op == "Concat":
+            a = attrs.get("axis", 1)
+            axis = int(a[0] if isinstance(a, (list, tuple)) else a)
+            srcs = [remap.get(i, i) for i in ins]
+            rank = len(g.tensors[srcs[0]].shape)
+            # ONNX graphs are NCHW; the IR is NHWC — channel concat
+            # (axis=1, rank 4) is last-axis concat here; last-axis concat
+            # (features, rank 2) maps directly
+            if not (axis in (1, -1, rank - 1) or
+                    (rank == 4 and axis == 1)):
+                raise ValueError(f"ONNX Concat axis {axis} not supported")
+            out = g.concat(srcs)
+        elif op == "Transpose":
+            perm = [int(v) for v in attrs.get("perm", [])]
+            rank = len(g.tensors[x].shape)
+            if rank == 2 and (perm == [1, 0] or not perm):
+                out = g.transpose2d(x)
+            elif perm == list(range(rank)):
+                out = x  # identity permutation
+            elif rank == 4 and perm in ([0, 2, 3, 1], [0, 3, 1, 2]):
+                # NCHW<->NHWC annotations: our activations are ALREADY
+                # NHWC, so a single layout flip is a no-op view; chains
+                # (flip + flip back) cancel
+                out = x
             else:
-                raise ValueError(
-                    f"ONNX Clip with bounds ({mn}, {mx}) not supported "
-                    "(only ReLU-equivalent clips)")
+                raise ValueError(f"ONNX Transpose perm {perm} not supported")
         elif op in ("Gemm", "MatMul"):
             wt = inits[ins[1]].astype(np.float32)
             if op == "MatMul" or not attrs.get("transB", 0):

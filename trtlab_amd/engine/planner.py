@@ -53,6 +53,9 @@ K_QUANT_MX4 = 15  # fp16 rows -> MXFP4 codes + e8m0 block scales
 K_GEMM_MX4 = 16  # MXFP4 x MXFP4 scaled-MFMA GEMM (fp16 out + epilogue)
 K_QUANT_MX8 = 17  # fp16 rows -> MXFP8 (e4m3) codes + e8m0 block scales
 K_GEMM_MX8 = 18  # MXFP8 x MXFP8 scaled-MFMA GEMM (fp16 out + epilogue)
+K_CLIP = 19  # out = min(max(x, mn), mx) — ONNX Clip with arbitrary bounds
+K_TRANSPOSE2D = 20  # out[N][M] = in[M][N]^T (tiled LDS transpose kernel)
+K_COPY2D = 21  # dst[m][coff + c] = src[m][c] — Concat lowering
 
 
 def _bf16_bits(arr: np.ndarray) -> np.ndarray:
@@ -199,6 +202,24 @@ class Planner:
             elif n.kind == "attention":
                 exec_ops.append(ExecOp(K_ATTENTION, n.name, [n.inputs[0]],
                                        n.output, dict(n.attrs)))
+            elif n.kind == "clip":
+                exec_ops.append(ExecOp(K_CLIP, n.name, [n.inputs[0]],
+                                       n.output,
+                                       dict(mn=n.attrs["mn"],
+                                            mx=n.attrs["mx"])))
+            elif n.kind == "transpose2d":
+                exec_ops.append(ExecOp(K_TRANSPOSE2D, n.name, [n.inputs[0]],
+                                       n.output, {}))
+            elif n.kind == "concat":
+                # one strided copy per input into its column range
+                coff = 0
+                ctot = g.tensors[n.output].shape[-1]
+                for ci, t in enumerate(n.inputs):
+                    c = g.tensors[t].shape[-1]
+                    exec_ops.append(ExecOp(
+                        K_COPY2D, f"{n.name}_part{ci}", [t], n.output,
+                        dict(coff=coff, C=c, ldd=ctot)))
+                    coff += c
             elif n.kind == "flatten":
                 raise ValueError("flatten should be a view, not a node")
             else:
@@ -617,6 +638,22 @@ class Planner:
                          epi=op.params.get("out_dtype", 0),
                          q_scale=op.params.get("q_scale", 0.0),
                          causal=1 if op.params.get("causal") else 0)
+            elif op.kind == K_CLIP:
+                n = 1
+                for s_ in shapes[op.output]:
+                    n *= s_
+                # res_scale/q_scale carry the clip bounds (see executor.cpp)
+                d.update(kind=K_CLIP, n_elems=n, res_scale=op.params["mn"],
+                         q_scale=op.params["mx"])
+            elif op.kind == K_TRANSPOSE2D:
+                m, ncol = shapes[op.inputs[0]]
+                d.update(kind=K_TRANSPOSE2D, M=m, N=ncol)
+            elif op.kind == K_COPY2D:
+                rows = 1
+                for s_ in shapes[op.inputs[0]][:-1]:
+                    rows *= s_
+                d.update(kind=K_COPY2D, M=rows, C=op.params["C"],
+                         Cout=op.params["ldd"], epi=op.params["coff"])
             else:
                 raise ValueError(f"bad exec op kind {op.kind}")
             op_dicts.append(d)

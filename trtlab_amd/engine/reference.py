@@ -15,6 +15,7 @@ import torch
 import torch.nn.functional as F
 
 from trtlab_amd.engine.planner import (
+    K_CLIP, K_COPY2D, K_TRANSPOSE2D,
     EnginePlan, K_ADD_LAYERNORM, K_ATTENTION, K_AVGPOOL, K_CHANNEL_PAD, K_CONV,
     K_DEQUANT, K_ELEMENTWISE, K_GAVGPOOL, K_GEMM, K_LAYERNORM, K_MAXPOOL,
     K_EMBEDDING, K_GEMM_MX4, K_GEMM_MX8, K_QUANT_MX4, K_QUANT_MX8,
@@ -208,6 +209,16 @@ def run_reference(plan: EnginePlan, input_nhwc, return_all: bool = False):
             if d.get("epi") == 3:  # fused fp8 output
                 y = _fp8_round(y / d["q_scale"])
             t[op.output] = y
+        elif op.kind == K_CLIP:
+            t[op.output] = torch.clamp(x, d["res_scale"], d["q_scale"])
+        elif op.kind == K_TRANSPOSE2D:
+            t[op.output] = x.reshape(d["M"], d["N"]).t().contiguous()
+        elif op.kind == K_COPY2D:
+            rows, c, ldd, coff = d["M"], d["C"], d["Cout"], d["epi"]
+            if op.output not in t:
+                t[op.output] = torch.zeros(*plan.shapes[op.output])
+            t[op.output].reshape(rows, ldd)[:, coff:coff + c] = \
+                x.reshape(rows, c)
         else:
             raise ValueError(f"bad op kind {op.kind}")
     if return_all:
