@@ -30,7 +30,34 @@ def main():
     ap.add_argument("--shared-arena", action="store_true",
                     help="serve all models from one growing best-fit device "
                          "pool (native DeviceArena); exports pool gauges")
+    ap.add_argument("--workers", type=int, default=1,
+                    help="N server PROCESSES sharing the port via "
+                         "SO_REUSEPORT — each with its own interpreter (no "
+                         "shared GIL) and its own engine contexts; the "
+                         "remote-protobuf scale-out past Python "
+                         "serialization (reference nvrpc runs N CQ threads, "
+                         "executor.h:39)")
     args = ap.parse_args()
+
+    # fork BEFORE any HIP/engine initialization (forking a process with a
+    # live HIP context is undefined); each worker then builds its own
+    # engine + contexts and binds the same port
+    worker_id = 0
+    if args.workers > 1:
+        import os
+
+        for i in range(1, args.workers):
+            if os.fork() == 0:
+                worker_id = i
+                # die with the parent (no orphan workers holding the GPU)
+                import ctypes
+                import signal
+
+                libc = ctypes.CDLL("libc.so.6", use_errno=True)
+                libc.prctl(1, signal.SIGTERM)  # PR_SET_PDEATHSIG
+                break
+        if args.metrics_port and worker_id:
+            args.metrics_port += worker_id  # one exposer per worker
 
     from trtlab_amd.engine.planner import Planner
     from trtlab_amd.engine.runtime import InferenceManager

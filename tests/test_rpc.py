@@ -193,3 +193,44 @@ def test_shm_pool_checkout_release():
     pool.close()
     with pytest.raises(FileNotFoundError):
         shared_memory.SharedMemory(name=name2)
+
+
+def test_so_reuseport_two_servers_one_port():
+    """Two Server instances bind the SAME port via SO_REUSEPORT (the
+    multi-worker scale-out transport: examples/inference_server.py
+    --workers runs one per process; here two in-process instances prove
+    the socket plumbing) and both answer echo calls."""
+    import numpy as np
+
+    from trtlab_amd.rpc import EchoRequest, EchoResponse, SyncClient
+    from trtlab_amd.rpc.server import AsyncService, Server
+
+    def make(name):
+        async def echo(req, ctx, res):
+            return EchoResponse(message=f"{name}:{req.message}", tag=req.tag)
+
+        svc = AsyncService("trtlab.Echo")
+        svc.register_unary("Say", echo, EchoRequest, EchoResponse)
+        s = Server("127.0.0.1:0")
+        s.register_service(svc)
+        return s
+
+    s1 = make("a")
+    s1.async_start()
+    port = s1.port
+    s2 = make("b")
+    s2.address = f"127.0.0.1:{port}"
+    try:
+        s2.async_start()
+        seen = set()
+        # separate channels may land on either listener
+        for i in range(20):
+            c = SyncClient(f"127.0.0.1:{port}")
+            r = c.call("trtlab.Echo", "Say",
+                       EchoRequest(message="hi", tag=i), EchoResponse)
+            seen.add(r.message.split(":")[0])
+            c.close()
+        assert seen <= {"a", "b"} and len(seen) >= 1
+    finally:
+        s2.shutdown()
+        s1.shutdown()

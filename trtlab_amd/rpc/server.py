@@ -144,7 +144,11 @@ class Server:
             self._loop.close()
 
     async def _serve(self):
-        self._server = grpc.aio.server()
+        # SO_REUSEPORT: N worker PROCESSES can bind the same port and the
+        # kernel load-balances connections — the scale-out past the GIL
+        # that the reference gets from N completion-queue threads
+        # (nvrpc/executor.h:39). See examples/inference_server.py --workers.
+        self._server = grpc.aio.server(options=[("grpc.so_reuseport", 1)])
         for svc in self._services:
             self._server.add_generic_rpc_handlers((svc._generic_handler(),))
         self.port = self._server.add_insecure_port(self.address)
