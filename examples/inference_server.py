@@ -90,6 +90,9 @@ def main():
     server = Server(f"0.0.0.0:{args.port}")
     server.register_service(svc)
     server.register_service(svc.health_service)
+    from trtlab_amd.rpc.trtis import TrtisService
+
+    server.register_service(TrtisService(resources).service)
     print(f"serving {args.model} b{args.batch} on :{args.port} "
           f"(devices {devices}, {args.contexts} contexts each; "
           f"metrics :{args.metrics_port})")

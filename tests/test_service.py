@@ -14,6 +14,8 @@ from trtlab_amd.rpc.service import InferenceResources, InferenceService
 class _FakePlan:
     input_shape = (2, 4)
     output_shape = (2, 4)
+    inputs = [dict(name="input", shape=(2, 4), dtype="f16")]
+    outputs = [dict(name="output", shape=(2, 4), dtype="f16")]
 
 
 class _FakeEngine:
@@ -277,5 +279,45 @@ def test_named_tensor_rpc_roundtrip():
         # primary output mirrors outputs[0] for single-output clients
         prim = np.frombuffer(resp.output, np.float16).reshape(2, 4)
         np.testing.assert_allclose(prim, outs["sum"])
+    finally:
+        server.shutdown()
+
+
+def test_trtis_surface_roundtrip():
+    """The TRTIS v1 GRPCService (nvidia.inferenceserver package, exact
+    reference field numbers): Status, Health and Infer with raw_input /
+    InferRequestHeader all answer correctly over a real loopback server."""
+    from trtlab_amd.rpc.trtis import (InferRequestHeader, StatusRequest,
+                                      StatusResponse, TrtisHealthRequest,
+                                      TrtisHealthResponse, TrtisInferRequest,
+                                      TrtisInferResponse, TrtisService)
+
+    server = Server("127.0.0.1:0")
+    svc = TrtisService(InferenceResources(_FakeManager()))
+    server.register_service(svc.service)
+    server.async_start()
+    try:
+        c = SyncClient(f"127.0.0.1:{server.port}")
+        st = c.call("nvidia.inferenceserver.GRPCService", "Status",
+                    StatusRequest(), StatusResponse)
+        assert st.request_status.code == 1  # SUCCESS
+        assert st.server_status.ready_state == 2  # SERVER_READY
+        h = c.call("nvidia.inferenceserver.GRPCService", "Health",
+                   TrtisHealthRequest(mode="ready"), TrtisHealthResponse)
+        assert h.health
+        x = np.arange(8, dtype=np.float16).reshape(2, 4)
+        hdr = InferRequestHeader(batch_size=2)
+        i = hdr.input.add()
+        i.name = "input"
+        i.byte_size = x.nbytes
+        req = TrtisInferRequest(model_name="m", meta_data=hdr,
+                                raw_input=[x.tobytes()], batch_id=7)
+        resp = c.call("nvidia.inferenceserver.GRPCService", "Infer", req,
+                      TrtisInferResponse)
+        assert resp.request_status.code == 1
+        assert resp.batch_id == 7
+        out = np.frombuffer(resp.raw_output[0], np.float16).reshape(2, 4)
+        np.testing.assert_allclose(out, x * 2)
+        assert resp.meta_data.output[0].raw.byte_size == out.nbytes
     finally:
         server.shutdown()

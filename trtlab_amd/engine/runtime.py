@@ -256,10 +256,16 @@ class InferenceManager:
         from trtlab_amd.utils.metrics import Metrics
 
         metrics = Metrics.initialize(metrics_port) if metrics_port else None
-        svc = InferenceService(InferenceResources(self), metrics=metrics)
+        resources = InferenceResources(self)
+        svc = InferenceService(resources, metrics=metrics)
         server = Server(f"0.0.0.0:{port}")
         server.register_service(svc)
         server.register_service(svc.health_service)
+        # TRTIS v1 surface (nvidia.inferenceserver.GRPCService) so stock
+        # TRTIS clients interoperate (reference 11_Protos API)
+        from trtlab_amd.rpc.trtis import TrtisService
+
+        server.register_service(TrtisService(resources).service)
         server.async_start()
         return server
 
