@@ -73,6 +73,65 @@ void* test_scratch(size_t need) {
   return need ? p : nullptr;
 }
 
+// ---- DLPack interop (reference core/types.h:83 DLPack dtypes) ----
+// Minimal DLPack ABI structs (the stable v0.x layout torch consumes);
+// capsules BORROW our device memory (the owning DeviceBuffer/arena must
+// outlive any tensor created from the capsule).
+struct DLDevice_ {
+  int32_t device_type;
+  int32_t device_id;
+};
+struct DLDataType_ {
+  uint8_t code;
+  uint8_t bits;
+  uint16_t lanes;
+};
+struct DLTensor_ {
+  void* data;
+  DLDevice_ device;
+  int32_t ndim;
+  DLDataType_ dtype;
+  int64_t* shape;
+  int64_t* strides;
+  uint64_t byte_offset;
+};
+struct DLManagedTensor_ {
+  DLTensor_ dl_tensor;
+  void* manager_ctx;
+  void (*deleter)(DLManagedTensor_*);
+};
+constexpr int kDLROCM = 10;
+
+struct DLHolder {
+  DLManagedTensor_ mt;
+  std::vector<int64_t> shape;
+};
+
+py::capsule make_dlpack(uintptr_t ptr, std::vector<int64_t> shape,
+                        int dtype_code, int bits, int device) {
+  auto* h = new DLHolder();
+  h->shape = std::move(shape);
+  h->mt.dl_tensor.data = (void*)ptr;
+  h->mt.dl_tensor.device = {kDLROCM, device};
+  h->mt.dl_tensor.ndim = (int32_t)h->shape.size();
+  h->mt.dl_tensor.dtype = {(uint8_t)dtype_code, (uint8_t)bits, 1};
+  h->mt.dl_tensor.shape = h->shape.data();
+  h->mt.dl_tensor.strides = nullptr;  // compact row-major
+  h->mt.dl_tensor.byte_offset = 0;
+  h->mt.manager_ctx = h;
+  h->mt.deleter = [](DLManagedTensor_* mt) {
+    delete (DLHolder*)mt->manager_ctx;  // memory itself is borrowed
+  };
+  return py::capsule(&h->mt, "dltensor", [](PyObject* cap) {
+    // unconsumed capsule: torch renames it to "used_dltensor" when it
+    // takes ownership, so only delete if still named "dltensor"
+    if (PyCapsule_IsValid(cap, "dltensor")) {
+      auto* mt = (DLManagedTensor_*)PyCapsule_GetPointer(cap, "dltensor");
+      if (mt && mt->deleter) mt->deleter(mt);
+    }
+  });
+}
+
 }  // namespace
 
 PYBIND11_MODULE(_C, m) {
@@ -113,6 +172,18 @@ PYBIND11_MODULE(_C, m) {
   mem.def("memset_d", [](uintptr_t p, int v, size_t bytes) {
     TRT_HIP_CHECK(hipMemset((void*)p, v, bytes));
   });
+
+  // DLPack export: dtype_code 2 = float (fp16 bits=16, fp32 bits=32),
+  // 0 = int, 1 = uint, 4 = bfloat. torch.from_dlpack(capsule) gives a
+  // ZERO-COPY tensor view of our device memory.
+  mem.def("to_dlpack",
+          [](uintptr_t ptr, std::vector<int64_t> shape, int dtype_code,
+             int bits, int device) {
+            return make_dlpack(ptr, std::move(shape), dtype_code, bits,
+                               device);
+          },
+          py::arg("ptr"), py::arg("shape"), py::arg("dtype_code") = 2,
+          py::arg("bits") = 16, py::arg("device") = 0);
 
   // Growing best-fit device allocator (reference bfit_allocator +
   // growing block_arena) for multi-model serving / dynamic shapes.

@@ -855,3 +855,41 @@ def test_concat_clip_engine_path():
     ref = run_reference(plan, xv)
     err = np.abs(out - ref).max() / max(np.abs(ref).max(), 1e-6)
     assert err < 0.05, err
+
+
+def test_dlpack_zero_copy_into_torch():
+    """DLPack interop: a torch tensor created from our DeviceBuffer's
+    capsule shares the SAME device memory (writes through torch are
+    visible to our runtime and vice versa)."""
+    import torch
+
+    from trtlab_amd.memory import DeviceBuffer
+
+    C = trtlab_amd.native()
+    buf = DeviceBuffer(256 * 2)
+    src = np.arange(256, dtype=np.float16)
+    buf.upload(src)
+    t = torch.from_dlpack(buf.dlpack((256,), "f16"))
+    assert t.device.type == "cuda" and t.dtype == torch.float16
+    assert np.array_equal(t.cpu().numpy(), src)  # zero-copy view sees data
+    t += 1  # write through torch...
+    back = np.zeros_like(src)
+    buf.download(back)
+    torch.cuda.synchronize()
+    assert np.array_equal(back, src + 1)  # ...lands in our buffer
+    del t
+    buf.close()
+
+
+def test_fenced_device_buffer():
+    from trtlab_amd.memory import FENCE_BYTES, FencedDeviceBuffer
+
+    C = trtlab_amd.native()
+    b = FencedDeviceBuffer(1024)
+    C.memory.memset_d(b.payload, 0x33, 1024)
+    b.check()  # payload writes don't touch the fences
+    # simulate an overrun into the back fence
+    C.memory.memset_d(b.payload + 1024, 0, 4)
+    with pytest.raises(MemoryError):
+        b.check()
+    b.close()

@@ -66,6 +66,15 @@ class DeviceBuffer:
     def download(self, arr) -> None:
         self._C.memory.memcpy_d2h(arr, self.ptr, min(arr.nbytes, self.nbytes))
 
+    def dlpack(self, shape, dtype: str = "f16"):
+        """Zero-copy DLPack capsule over this buffer (consume with
+        torch.from_dlpack; reference core/types.h:83 DLPack interop).
+        The buffer must outlive tensors created from the capsule."""
+        code, bits = {"f16": (2, 16), "f32": (2, 32), "i32": (0, 32),
+                      "u8": (1, 8), "bf16": (4, 16)}[dtype]
+        return self._C.memory.to_dlpack(self.ptr, list(shape), code, bits,
+                                        self.device)
+
 
 class PinnedBuffer:
     """Page-locked host staging buffer (reference host_pinned_memory)."""
@@ -255,3 +264,115 @@ class ArenaPlanner:
             placed.append(it)
             total = max(total, it.offset + it.size)
         return {it.name: it.offset for it in items}, round_up(total, self.alignment)
+
+
+# --------------------------------------------------------------------------
+# Allocation-size histogram tracker (reference trackers.h:37
+# histogram_tracker): wraps any allocate/deallocate pair with log2-bucket
+# counts + byte counters; report() renders the reference-style table.
+class HistogramTracker:
+    def __init__(self, name: str = "alloc"):
+        self.name = name
+        self.buckets = [0] * 48      # count of allocations by ceil(log2)
+        self.total_allocs = 0
+        self.total_bytes = 0
+        self.in_use = 0
+        self.high_water = 0
+
+    def on_allocate(self, nbytes: int) -> None:
+        b = max(int(nbytes - 1).bit_length(), 0) if nbytes > 1 else 0
+        self.buckets[min(b, 47)] += 1
+        self.total_allocs += 1
+        self.total_bytes += nbytes
+        self.in_use += nbytes
+        self.high_water = max(self.high_water, self.in_use)
+
+    def on_deallocate(self, nbytes: int) -> None:
+        self.in_use -= nbytes
+
+    def report(self) -> str:
+        lines = [f"[{self.name}] allocs={self.total_allocs} "
+                 f"bytes={self.total_bytes} in_use={self.in_use} "
+                 f"high_water={self.high_water}"]
+        for i, c in enumerate(self.buckets):
+            if c:
+                lines.append(f"  2^{i:<2d} ({1 << i:>12d} B): {c}")
+        return "\n".join(lines)
+
+
+class TrackedDeviceAllocator:
+    """Device allocator with a histogram tracker (reference
+    make_tracked_allocator, tracking.h:370)."""
+
+    def __init__(self, device: int = 0, name: str = "device"):
+        from trtlab_amd import native
+
+        self._C = native()
+        self.device = device
+        self.tracker = HistogramTracker(name)
+        self._sizes: dict = {}
+
+    def allocate(self, nbytes: int) -> int:
+        p = self._C.memory.device_malloc(nbytes, self.device)
+        self._sizes[p] = nbytes
+        self.tracker.on_allocate(nbytes)
+        return p
+
+    def deallocate(self, ptr: int) -> None:
+        nbytes = self._sizes.pop(ptr)
+        self._C.memory.device_free(ptr, nbytes)
+        self.tracker.on_deallocate(nbytes)
+
+
+# --------------------------------------------------------------------------
+# Debug fill/fence (reference debugging.h:1-114 + detail/debug_helpers.h:
+# fill fresh memory with a pattern, fence both ends, verify on free).
+FENCE_BYTES = 64
+_FENCE = 0xFD
+_FILL = 0xCD
+
+
+class FencedHostBuffer:
+    """Host buffer with guard fences: allocates nbytes + 2 fences, fills
+    the payload with the debug pattern, and check()/close() raise on any
+    fence corruption (buffer overrun detector for host staging code)."""
+
+    def __init__(self, nbytes: int):
+        self.nbytes = int(nbytes)
+        self._raw = np.empty(self.nbytes + 2 * FENCE_BYTES, np.uint8)
+        self._raw[:FENCE_BYTES] = _FENCE
+        self._raw[-FENCE_BYTES:] = _FENCE
+        self._raw[FENCE_BYTES:-FENCE_BYTES] = _FILL
+        self.array = self._raw[FENCE_BYTES:FENCE_BYTES + self.nbytes]
+
+    def check(self) -> None:
+        if (self._raw[:FENCE_BYTES] != _FENCE).any():
+            raise MemoryError("front fence corrupted (underrun)")
+        if (self._raw[-FENCE_BYTES:] != _FENCE).any():
+            raise MemoryError("back fence corrupted (overrun)")
+
+    def close(self) -> None:
+        self.check()
+
+
+class FencedDeviceBuffer(DeviceBuffer):
+    """DeviceBuffer with device-side guard fences (GPU debug builds)."""
+
+    def __init__(self, nbytes: int, device: int = 0):
+        super().__init__(int(nbytes) + 2 * FENCE_BYTES, device)
+        self.payload = self.ptr + FENCE_BYTES
+        self.payload_bytes = int(nbytes)
+        pat = np.full(FENCE_BYTES, _FENCE, np.uint8)
+        self._C.memory.memcpy_h2d(self.ptr, pat, FENCE_BYTES)
+        self._C.memory.memcpy_h2d(self.payload + self.payload_bytes, pat,
+                                  FENCE_BYTES)
+
+    def check(self) -> None:
+        got = np.zeros(FENCE_BYTES, np.uint8)
+        self._C.memory.memcpy_d2h(got, self.ptr, FENCE_BYTES)
+        if (got != _FENCE).any():
+            raise MemoryError("front fence corrupted (underrun)")
+        self._C.memory.memcpy_d2h(got, self.payload + self.payload_bytes,
+                                  FENCE_BYTES)
+        if (got != _FENCE).any():
+            raise MemoryError("back fence corrupted (overrun)")
