@@ -24,10 +24,16 @@ def main():
 
     shape = tuple(int(s) for s in args.shape.split(","))
     batch = (np.random.RandomState(0).randn(*shape) * 0.5).astype(np.float16)
+    # build the payload ONCE: re-serializing 1.2 MB of protobuf per request
+    # makes the CLIENT the bottleneck at high rates (the server is 3
+    # SO_REUSEPORT workers) — the load generator should generate load,
+    # not burn its core on repeated identical serialization
+    payload = batch.tobytes()
+    req = InferRequest(model=args.model, input=payload, shape=list(shape),
+                      dtype="f16")
 
     def make_request(i):
-        return InferRequest(model=args.model, input=batch.tobytes(),
-                            shape=list(shape), dtype="f16", batch_id=i)
+        return req
 
     stats = siege(args.target, "trtlab.Inference", "Compute", make_request,
                   InferResponse, rate_hz=args.rate, duration_s=args.seconds,
