@@ -37,7 +37,7 @@ _TT_ELEM, _TT_SHAPE = 1, 2
 _TS_DIM = 1
 _TD_VALUE = 1
 
-_F32, _F16, _I64 = 1, 10, 7
+_F32, _F16, _I64, _I32 = 1, 10, 7, 6
 
 
 # -------------------------------------------------------------------- load
@@ -83,19 +83,21 @@ def _parse_attrs(bufs: List[bytes]) -> Dict[str, object]:
     return out
 
 
-def _parse_vi_shape(buf: bytes) -> tuple[str, List[int]]:
+def _parse_vi_shape(buf: bytes) -> tuple[str, List[int], int]:
     d = w.fields_dict(buf)
     name = d[_VI_NAME][0].decode()
     dims: List[int] = []
+    elem = _F32
     if _VI_TYPE in d:
         tp = w.fields_dict(d[_VI_TYPE][0])
         if _TP_TENSOR in tp:
             tt = w.fields_dict(tp[_TP_TENSOR][0])
+            elem = tt.get(_TT_ELEM, [_F32])[0]
             if _TT_SHAPE in tt:
                 for dim_buf in w.fields_dict(tt[_TT_SHAPE][0]).get(_TS_DIM, []):
                     dd = w.fields_dict(dim_buf)
                     dims.append(w.varint_to_sint64(dd.get(_TD_VALUE, [0])[0]))
-    return name, dims
+    return name, dims, elem
 
 
 def import_onnx(data: bytes, batch: Optional[int] = None,
@@ -113,18 +115,22 @@ def import_onnx(data: bytes, batch: Optional[int] = None,
     # graph input (the one without an initializer)
     input_name = None
     for vi in gd.get(_GRAPH_INPUT, []):
-        nm, dims = _parse_vi_shape(vi)
+        nm, dims, elem = _parse_vi_shape(vi)
         if nm in inits:
             continue
+        dt = {_F32: "f32", _F16: "f16", _I32: "i32", _I64: "i32"}.get(
+            elem, "f16")
+        if dt == "f32":
+            dt = "f16"  # fp32 graph inputs feed the fp16 compute path
         if len(dims) == 4:
             n, c, h, ww = dims  # NCHW -> IR NHWC
             if batch:
                 n = batch
             input_name = g.input((n, h, ww, c), name=nm)
-        else:  # 2-D (features) or other ranks: layout-free
+        else:  # 2-D (features) / 1-D (token ids): layout-free
             if batch and dims:
                 dims = [batch] + list(dims[1:])
-            input_name = g.input(tuple(dims), name=nm)
+            input_name = g.input(tuple(dims), name=nm, dtype=dt)
     assert input_name is not None, "no graph input found"
 
     # name remapping: ONNX tensor name -> IR tensor name
@@ -237,6 +243,22 @@ def import_onnx(data: bytes, batch: Optional[int] = None,
             out = g.layernorm(x, inits[ins[1]].astype(np.float32),
                               inits[ins[2]].astype(np.float32),
                               eps=float(attrs.get("epsilon", 1e-5)))
+        elif op == "TrtlabEmbedding":
+            # custom-domain round-trip op: [ids, tok, pos, (seg), (segids)]
+            tables = [i for i in ins[1:] if i in inits]
+            tensors = [i for i in ins[1:] if i not in inits]
+            tok = inits[tables[0]].astype(np.float32)
+            pos = inits[tables[1]].astype(np.float32)
+            seg = (inits[tables[2]].astype(np.float32)
+                   if len(tables) > 2 else None)
+            segids = remap.get(tensors[0], tensors[0]) if tensors else None
+            out = g.embedding(x, tok, pos, seg_table=seg, segids=segids)
+        elif op == "TrtlabAttention":
+            out = g.attention(x, heads=int(attrs["heads"]),
+                              seq=int(attrs["seq"]),
+                              causal=bool(attrs.get("causal", 0)),
+                              varlen=bool(attrs.get("varlen", 0)),
+                              pad_id=int(attrs.get("pad_id", 0)))
         else:
             raise ValueError(f"ONNX op {op} not supported by the importer")
         remap[outs[0]] = out
@@ -285,11 +307,11 @@ def _node(op: str, ins: List[str], outs: List[str], *attrs: bytes) -> bytes:
     return w.f_bytes(_GRAPH_NODE, body)
 
 
-def _value_info(name: str, dims: List[int]) -> bytes:
+def _value_info(name: str, dims: List[int], elem: int = _F32) -> bytes:
     dim_bufs = b"".join(
         w.f_bytes(_TS_DIM, w.f_varint(_TD_VALUE, d)) for d in dims)
     shape = w.f_bytes(_TT_SHAPE, dim_bufs)
-    tt = w.f_varint(_TT_ELEM, _F32) + shape
+    tt = w.f_varint(_TT_ELEM, elem) + shape
     tp = w.f_bytes(_TP_TENSOR, tt)
     return w.f_string(_VI_NAME, name) + w.f_bytes(_VI_TYPE, tp)
 
@@ -351,13 +373,65 @@ def export_onnx(g: Graph) -> bytes:
                            [n.inputs[0], add_init(a["gamma"]),
                             add_init(a["beta"])], [n.output],
                            _attr_f("epsilon", a["eps"]))
+        elif n.kind == "avgpool":
+            a = n.attrs
+            nodes += _node("AveragePool", [n.inputs[0]], [n.output],
+                           _attr_ints("kernel_shape", [a["kernel"]] * 2),
+                           _attr_ints("strides", [a["stride"]] * 2),
+                           _attr_ints("pads", [a["padding"]] * 4))
+        elif n.kind == "clip":
+            nodes += _node("Clip", [n.inputs[0]], [n.output],
+                           _attr_f("min", n.attrs["mn"]),
+                           _attr_f("max", n.attrs["mx"]))
+        elif n.kind == "transpose2d":
+            nodes += _node("Transpose", [n.inputs[0]], [n.output],
+                           _attr_ints("perm", [1, 0]))
+        elif n.kind == "concat":
+            rank = len(g.tensors[n.inputs[0]].shape)
+            nodes += _node("Concat", list(n.inputs), [n.output],
+                           _attr_i("axis", rank - 1))
+        elif n.kind == "add_layernorm":
+            # decompose to standard ops: Add + LayerNormalization
+            a = n.attrs
+            mid = n.output + "_sum"
+            nodes += _node("Add", list(n.inputs), [mid])
+            nodes += _node("LayerNormalization",
+                           [mid, add_init(a["gamma"]), add_init(a["beta"])],
+                           [n.output], _attr_f("epsilon", a["eps"]))
+        elif n.kind == "embedding":
+            # custom-domain op with the tables as initializers (transformer
+            # round-trip; standard ONNX would need Gather+Add chains the IR
+            # has no generic ops for)
+            a = n.attrs
+            ins = [n.inputs[0], add_init(a["tok"]), add_init(a["pos"])]
+            if a.get("seg") is not None:
+                ins.append(add_init(a["seg"]))
+            nodes += _node("TrtlabEmbedding", ins + list(n.inputs[1:]),
+                           [n.output], _attr_i("seq", a["seq"]))
+        elif n.kind == "attention":
+            a = n.attrs
+            nodes += _node("TrtlabAttention", [n.inputs[0]], [n.output],
+                           _attr_i("heads", a["heads"]),
+                           _attr_i("seq", a["seq"]),
+                           _attr_i("causal", 1 if a.get("causal") else 0),
+                           _attr_i("varlen", 1 if a.get("varlen") else 0),
+                           _attr_i("pad_id", a.get("pad_id", 0)))
         else:
             raise ValueError(f"export: unsupported node kind {n.kind}")
 
-    nb, h, ww, c = (list(g.tensors[g.input_name].shape) + [0, 0, 0, 0])[:4]
-    in_vi = w.f_bytes(_GRAPH_INPUT, _value_info(g.input_name, [nb, c, h, ww]))
+    shape = list(g.tensors[g.input_name].shape)
+    if len(shape) == 4:
+        nb, h, ww, c = shape
+        shape = [nb, c, h, ww]  # IR NHWC -> ONNX NCHW
+    in_vis = b""
+    for nm in (g.input_names or [g.input_name]):
+        s_ = list(g.tensors[nm].shape)
+        if len(s_) == 4:
+            s_ = [s_[0], s_[3], s_[1], s_[2]]
+        elem = _I32 if g.tensors[nm].dtype == "i32" else _F32
+        in_vis += w.f_bytes(_GRAPH_INPUT, _value_info(nm, s_, elem))
     out_vi = w.f_bytes(_GRAPH_OUTPUT, _value_info(g.output_name, []))
-    graph = nodes + w.f_string(_GRAPH_NAME, g.name) + inits + in_vi + out_vi
+    graph = nodes + w.f_string(_GRAPH_NAME, g.name) + inits + in_vis + out_vi
     model = w.f_varint(1, 8)  # ir_version
     model += w.f_bytes(_MODEL_GRAPH, graph)
     return model
