@@ -893,3 +893,31 @@ def test_fenced_device_buffer():
     with pytest.raises(MemoryError):
         b.check()
     b.close()
+
+
+@pytest.mark.parametrize("dtype_name", ["int8", "fp8"])
+def test_quantized_per_layer_gate(dtype_name):
+    """Per-layer int8/fp8 correctness gate: every
+    readable intermediate tensor must correlate with the quantization-
+    emulating CPU reference — a correlation oracle applied PER LAYER so a
+    systematic bias in any single kernel cannot hide inside an acceptable
+    end-to-end score."""
+    import sys
+    from pathlib import Path
+
+    sys.path.insert(0, str(Path(__file__).resolve().parent.parent / "tools"))
+    from debug_engine import per_layer_errors
+
+    from trtlab_amd.engine.planner import DT_F8, DT_I8, Planner
+    from trtlab_amd.models import build_resnet
+
+    dt = DT_I8 if dtype_name == "int8" else DT_F8
+    g = build_resnet(18, batch=2, image=64, seed=0)
+    plan = Planner(dtype=dt, reuse=False).compile(g)
+    x = np.random.RandomState(9).randn(*plan.input_shape).astype(
+        np.float32) * 0.5
+    rows = per_layer_errors(plan, x)
+    assert len(rows) >= 10  # reuse=False: most tensors readable
+    for i, t, rel, corr, nans in rows:
+        assert nans == 0, f"op {i} {t}: {nans} NaNs"
+        assert corr > 0.98, f"op {i} {t}: per-layer corr {corr}"
