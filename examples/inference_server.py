@@ -93,6 +93,16 @@ def main():
     from trtlab_amd.rpc.trtis import TrtisService
 
     server.register_service(TrtisService(resources).service)
+    # GC tuning for the serving loop: the engine/plan object graph is
+    # permanent — freeze it out of collection, and raise gen0 threshold so
+    # request-object churn (1.2 MB protobufs) doesn't trigger frequent
+    # collections whose pauses land in the latency tail
+    import gc
+
+    gc.collect()
+    gc.freeze()
+    gc.set_threshold(50000, 100, 100)
+
     print(f"serving {args.model} b{args.batch} on :{args.port} "
           f"(devices {devices}, {args.contexts} contexts each; "
           f"metrics :{args.metrics_port})")
