@@ -115,3 +115,54 @@ def test_transactional_stack_nesting():
     inner_top = st.high_water
     st.commit()
     assert st.high_water < inner_top
+
+
+class _FakeInner:
+    """CPU stand-in for a device allocator (composability tests)."""
+
+    def __init__(self):
+        self.n = 0
+        self.live = set()
+
+    def allocate(self, nbytes):
+        self.n += 1
+        self.live.add(self.n)
+        return self.n
+
+    def deallocate(self, ptr):
+        self.live.discard(ptr)
+
+
+def test_count_limited_allocator():
+    from trtlab_amd.memory import CountLimitedAllocator
+
+    a = CountLimitedAllocator(_FakeInner(), max_count=2)
+    p1, p2 = a.allocate(10), a.allocate(10)
+    with pytest.raises(MemoryError):
+        a.allocate(10)
+    a.deallocate(p1)
+    a.allocate(10)  # slot freed
+
+
+def test_size_limited_allocator():
+    from trtlab_amd.memory import SizeLimitedAllocator
+
+    a = SizeLimitedAllocator(_FakeInner(), max_bytes=100)
+    p = a.allocate(80)
+    with pytest.raises(MemoryError):
+        a.allocate(30)
+    a.deallocate(p)
+    a.allocate(95)
+    assert a.in_use == 95
+
+
+def test_tracked_allocator_composes():
+    from trtlab_amd.memory import (CountLimitedAllocator, TrackedAllocator)
+
+    a = TrackedAllocator(CountLimitedAllocator(_FakeInner(), 8), "t")
+    ps = [a.allocate(1 << i) for i in range(4)]
+    for p in ps:
+        a.deallocate(p)
+    assert a.tracker.total_allocs == 4
+    assert a.tracker.in_use == 0
+    assert a.tracker.high_water == (1 + 2 + 4 + 8)

@@ -376,3 +376,104 @@ class FencedDeviceBuffer(DeviceBuffer):
                                   FENCE_BYTES)
         if (got != _FENCE).any():
             raise MemoryError("back fence corrupted (overrun)")
+
+
+# --------------------------------------------------------------------------
+# Composable allocator algebra (reference block_allocators.h:186-461:
+# growing / count-limited / size-limited block allocators + factory).
+# Wrappers compose over any inner allocator exposing allocate/deallocate —
+# the native DeviceArena, TrackedDeviceAllocator, or a test fake.
+class CountLimitedAllocator:
+    """Cap the number of LIVE allocations (reference
+    count_limited_block_allocator, block_allocators.h:258)."""
+
+    def __init__(self, inner, max_count: int):
+        self.inner = inner
+        self.max_count = max_count
+        self._live = 0
+
+    def allocate(self, nbytes: int):
+        if self._live >= self.max_count:
+            raise MemoryError(
+                f"allocation count limit reached ({self.max_count})")
+        p = self.inner.allocate(nbytes)
+        self._live += 1
+        return p
+
+    def deallocate(self, ptr) -> None:
+        self.inner.deallocate(ptr)
+        self._live = max(0, self._live - 1)
+
+
+class SizeLimitedAllocator:
+    """Cap the total LIVE bytes (reference size_limited_block_allocator,
+    block_allocators.h:342). Sized for 288 GB of HBM3E by default."""
+
+    def __init__(self, inner, max_bytes: int = 288 << 30):
+        self.inner = inner
+        self.max_bytes = max_bytes
+        self._sizes: dict = {}
+        self.in_use = 0
+
+    def allocate(self, nbytes: int):
+        if self.in_use + nbytes > self.max_bytes:
+            raise MemoryError(
+                f"size limit: {self.in_use} + {nbytes} > {self.max_bytes}")
+        p = self.inner.allocate(nbytes)
+        self._sizes[p] = nbytes
+        self.in_use += nbytes
+        return p
+
+    def deallocate(self, ptr) -> None:
+        self.inner.deallocate(ptr)
+        self.in_use -= self._sizes.pop(ptr, 0)
+
+
+class TrackedAllocator:
+    """Attach a HistogramTracker to any allocator (reference
+    make_tracked_allocator, tracking.h:370)."""
+
+    def __init__(self, inner, name: str = "tracked"):
+        self.inner = inner
+        self.tracker = HistogramTracker(name)
+        self._sizes: dict = {}
+
+    def allocate(self, nbytes: int):
+        p = self.inner.allocate(nbytes)
+        self._sizes[p] = nbytes
+        self.tracker.on_allocate(nbytes)
+        return p
+
+    def deallocate(self, ptr) -> None:
+        self.inner.deallocate(ptr)
+        self.tracker.on_deallocate(self._sizes.pop(ptr, 0))
+
+
+def make_device_allocator(device: int = 0, max_bytes: int = 0,
+                          max_count: int = 0, tracked: bool = False,
+                          initial_bytes: int = 0):
+    """Factory composing the device allocator stack (reference
+    make_block_allocator, block_allocators.h:432): growing best-fit
+    DeviceArena at the bottom, then optional size / count limits and
+    tracking."""
+    from trtlab_amd import native
+
+    class _ArenaAdapter:
+        def __init__(self):
+            self.arena = native().memory.DeviceArena(device, initial_bytes,
+                                                     max_bytes)
+
+        def allocate(self, nbytes):
+            return self.arena.allocate(nbytes)
+
+        def deallocate(self, ptr):
+            self.arena.deallocate(ptr)
+
+    alloc = _ArenaAdapter()
+    if max_bytes:
+        alloc = SizeLimitedAllocator(alloc, max_bytes)
+    if max_count:
+        alloc = CountLimitedAllocator(alloc, max_count)
+    if tracked:
+        alloc = TrackedAllocator(alloc, f"device{device}")
+    return alloc
