@@ -234,3 +234,25 @@ def test_so_reuseport_two_servers_one_port():
     finally:
         s2.shutdown()
         s1.shutdown()
+
+
+def test_streaming_call_write_queue(echo_server):
+    """Incremental streaming client (reference v2/v3 richness): write(),
+    interleaved responses via callback, close_writes() half-close, and a
+    status future that resolves on server finish."""
+    from trtlab_amd.rpc.client import AsyncClient
+
+    got = []
+    c = AsyncClient(f"127.0.0.1:{echo_server.port}")
+    try:
+        call = c.open_stream("trtlab.Echo", "EchoStream", EchoResponse,
+                             on_response=lambda r: got.append(r.tag))
+        for i in range(5):
+            call.write(EchoRequest(message=f"m{i}", tag=i))
+        call.close_writes()
+        resps = list(call.responses())
+        assert [r.tag for r in resps] == list(range(5))
+        assert got == list(range(5))  # callback saw them too
+        assert call.status.result(timeout=5) is True
+    finally:
+        c.close()

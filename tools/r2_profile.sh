@@ -1,36 +1,44 @@
 #!/bin/bash
-# Round-2 conv-perf profiling: kernel stats + PMC (separate runs; gpurun
-# refuses pmc+trace combinations).
-cd /tmp && export TMPDIR=/tmp
+# Round-2 conv-perf profiling. Traces go to /tmp (NOT gpurun_out — the
+# copy-back limit is 64 MiB); only CSV summaries are returned.
+export TMPDIR=/tmp
 cd /root/repo
-mkdir -p gpurun_out/prof_r2
+mkdir -p gpurun_out
+rm -rf /tmp/prof_r2 && mkdir -p /tmp/prof_r2
 echo "=== kernel stats (rocprofv3 --kernel-trace --stats) ==="
-timeout 500 rocprofv3 --kernel-trace --stats -d gpurun_out/prof_r2 -o r2stats \
-  -- python bench.py --steps 60 --warmup 20 > gpurun_out/prof_bench.log 2>&1
-find gpurun_out/prof_r2 -name "*kernel_stats*" -exec cp {} gpurun_out/r2_kernel_stats.csv \;
-head -14 gpurun_out/r2_kernel_stats.csv 2>/dev/null | cut -c1-150
+(cd /tmp && timeout 500 rocprofv3 --kernel-trace --stats -d /tmp/prof_r2 \
+  -o r2stats -- python /root/repo/bench.py --steps 40 --warmup 15) \
+  > gpurun_out/prof_bench.log 2>&1
+echo "rocprof rc=$?"
+tail -3 gpurun_out/prof_bench.log
+find /tmp/prof_r2 -name "*kernel_stats*" -exec cp {} gpurun_out/r2_kernel_stats.csv \;
+echo "--- top kernels ---"
+head -14 gpurun_out/r2_kernel_stats.csv 2>/dev/null | cut -c1-140
 echo "=== PMC counters ==="
-timeout 500 rocprofv3 --pmc SQ_BUSY_CYCLES SQ_VALU_MFMA_BUSY_CYCLES \
-  SQ_LDS_BANK_CONFLICT SQ_WAIT_ANY SQ_INSTS_VALU SQ_INSTS_MFMA \
-  -d gpurun_out/prof_r2 -o r2pmc -- python bench.py --steps 20 --warmup 10 \
-  --no-autotune > gpurun_out/prof_pmc.log 2>&1 || \
-timeout 500 rocprofv3 --pmc SQ_BUSY_CYCLES SQ_VALU_MFMA_BUSY_CYCLES \
-  SQ_LDS_BANK_CONFLICT -d gpurun_out/prof_r2 -o r2pmc -- python bench.py \
-  --steps 20 --warmup 10 --no-autotune > gpurun_out/prof_pmc.log 2>&1
-find gpurun_out/prof_r2 -name "*counter*" -exec cp {} gpurun_out/r2_pmc_counters.csv \;
+rm -rf /tmp/prof_pmc && mkdir -p /tmp/prof_pmc
+(cd /tmp && timeout 500 rocprofv3 --pmc SQ_BUSY_CYCLES SQ_VALU_MFMA_BUSY_CYCLES \
+  SQ_LDS_BANK_CONFLICT SQ_INSTS_VALU SQ_INSTS_MFMA -d /tmp/prof_pmc -o r2pmc \
+  -- python /root/repo/bench.py --steps 15 --warmup 8 --no-autotune) \
+  > gpurun_out/prof_pmc.log 2>&1
+echo "pmc rc=$?"
+tail -3 gpurun_out/prof_pmc.log
+find /tmp/prof_pmc -name "*counter_collection*" -o -name "*counter*csv" | head -3
+find /tmp/prof_pmc -name "*counter*" -exec cp {} gpurun_out/r2_pmc_counters.csv \;
 python - << 'PYEOF'
 import csv, collections
 try:
     agg = collections.defaultdict(lambda: collections.defaultdict(float))
-    with open("gpurun_out/r2_pmc_counters.csv") as f:
+    with open("/root/repo/gpurun_out/r2_pmc_counters.csv") as f:
         for row in csv.DictReader(f):
-            agg[row["Kernel_Name"][:80]][row["Counter_Name"]] += float(row["Counter_Value"])
-    print(f"{'kernel':80} {'mfma/busy':>10} {'conflicts':>10}")
+            agg[row["Kernel_Name"][:70]][row["Counter_Name"]] += float(row["Counter_Value"])
+    print(f"{'kernel':70} {'mfma/busy':>10} {'conflicts':>10} {'valu':>12}")
     for k, c in sorted(agg.items(), key=lambda kv: -kv[1].get("SQ_BUSY_CYCLES", 0))[:12]:
         busy = c.get("SQ_BUSY_CYCLES", 1)
-        print(f"{k:80} {c.get('SQ_VALU_MFMA_BUSY_CYCLES',0)/busy:10.3f} "
-              f"{c.get('SQ_LDS_BANK_CONFLICT',0):10.0f}")
+        print(f"{k:70} {c.get('SQ_VALU_MFMA_BUSY_CYCLES',0)/busy:10.3f} "
+              f"{c.get('SQ_LDS_BANK_CONFLICT',0):10.0f} "
+              f"{c.get('SQ_INSTS_VALU',0):12.0f}")
 except Exception as e:
     print("pmc summary failed:", e)
 PYEOF
+du -sh gpurun_out 2>/dev/null
 echo "=== done ==="
