@@ -10,6 +10,11 @@
 // other slots keep decoding). advance_pos bumps every slot at the end of
 // the captured step.
 //
+// Idle-slot masking: pos[b] < 0 marks slot b IDLE — every per-slot kernel
+// early-exits for it (no cache writes, no attention scan, position stays
+// parked), so empty slots in a continuous batch cost ~nothing while the
+// captured graph keeps its fixed shape.
+//
 // Cache layout per layer: K and V as [B][H][Smax][64] fp16 (contiguous
 // 128-B rows per key — one cacheline).
 #include "gemm_common.h"
@@ -25,6 +30,7 @@ __global__ __launch_bounds__(64) void kv_append_kernel(
   int b = blockIdx.x / H, h = blockIdx.x % H;
   int d = threadIdx.x;  // 0..63
   int p = pos[b];
+  if (p < 0) return;  // idle slot
   int hid = H * 64;
   int64_t src = (int64_t)b * 3 * hid + h * 64 + d;
   int64_t dst = (((int64_t)b * H + h) * smax + p) * 64 + d;
@@ -79,6 +85,7 @@ __global__ __launch_bounds__(64) void decode_attention_kernel(
   int b = blockIdx.x / H, h = blockIdx.x % H;
   int lane = threadIdx.x;
   int hid = H * 64;
+  if (pos[b] < 0) return;  // idle slot: no scan, stale output row unused
   int n = pos[b] + 1;  // keys 0..pos[b] (this step's K already appended)
 
   // q for this head, one element per lane
@@ -151,6 +158,7 @@ __global__ void decode_embed_kernel(const int* __restrict__ ids,
                                     const int* __restrict__ pos, int hidden) {
   int b = blockIdx.x;
   int p = pos[b];
+  if (p < 0) return;  // idle slot
   int64_t t = (int64_t)ids[b] * hidden;
   for (int i = threadIdx.x; i < hidden; i += blockDim.x)
     out[(int64_t)b * hidden + i] =
@@ -170,7 +178,7 @@ void launch_decode_embed(const void* ids, const void* tok, const void* posemb,
 // decode step; clamped so replay past smax is safe).
 __global__ void advance_pos_kernel(int* pos, int B, int smax) {
   int b = threadIdx.x;
-  if (b < B) {
+  if (b < B && pos[b] >= 0) {  // idle slots stay parked at -1
     int p = pos[b] + 1;
     pos[b] = p >= smax ? smax - 1 : p;
   }

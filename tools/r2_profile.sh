@@ -2,6 +2,7 @@
 # Round-2 conv-perf profiling. Traces go to /tmp (NOT gpurun_out — the
 # copy-back limit is 64 MiB); only CSV summaries are returned.
 export TMPDIR=/tmp
+export PYTHONPATH=/root/repo
 cd /root/repo
 mkdir -p gpurun_out
 rm -rf /tmp/prof_r2 && mkdir -p /tmp/prof_r2

@@ -96,6 +96,7 @@ class DecodeSession:
         self.ids = torch.zeros(B, dtype=torch.int32, device="cuda")
         self.pos = torch.zeros(B, dtype=torch.int32, device="cuda")
         self._slot_steps = np.zeros(B, np.int64)
+        self._active = np.ones(B, bool)  # host mirror of pos[b] >= 0
         self.h = torch.zeros(B, Hd, dtype=torch.half, device="cuda")
         self.x = torch.zeros(B, Hd, dtype=torch.half, device="cuda")
         self.qkv = torch.zeros(B, 3 * Hd, dtype=torch.half, device="cuda")
@@ -256,6 +257,7 @@ class DecodeSession:
         last = x.reshape(B, Pp, Hd)[:, P - 1].contiguous()
         torch.cuda.synchronize()
         self.pos.fill_(P)
+        self._active[:] = True
         self._steps = P
         self._slot_steps[:] = P
         torch.cuda.synchronize()
@@ -271,7 +273,7 @@ class DecodeSession:
         [B, hidden] fp32 (or logits [B, vocab] with lm_head=True). The
         device-side position counter starts at 0 and the captured graph
         advances it, so replays need no host-side position plumbing."""
-        if (self._slot_steps >= self.smax).any():
+        if (self._slot_steps[self._active] >= self.smax).any():
             raise RuntimeError(
                 "DecodeSession: a slot hit the sequence limit "
                 "(reset_slot() it or end the session)")
@@ -308,6 +310,17 @@ class DecodeSession:
         the other slots keep decoding against their caches)."""
         self.pos[b] = 0
         self._slot_steps[b] = 0
+        self._active[b] = True
+        self._torch.cuda.synchronize()
+
+    def idle_slot(self, b: int) -> None:
+        """Park slot b (no active request): pos[b] = -1 makes every
+        per-slot decode kernel early-exit for it, so an empty slot in the
+        continuous batch costs ~nothing per replayed step. reset_slot(b)
+        re-activates it for a fresh sequence."""
+        self.pos[b] = -1
+        self._slot_steps[b] = 0
+        self._active[b] = False
         self._torch.cuda.synchronize()
 
     def close(self):
