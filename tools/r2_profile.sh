@@ -7,7 +7,7 @@ cd /root/repo
 mkdir -p gpurun_out
 rm -rf /tmp/prof_r2 && mkdir -p /tmp/prof_r2
 echo "=== kernel stats (rocprofv3 --kernel-trace --stats) ==="
-(cd /tmp && timeout 500 rocprofv3 --kernel-trace --stats -d /tmp/prof_r2 \
+(cd /tmp && timeout 500 rocprofv3 --kernel-trace --stats --output-format csv -d /tmp/prof_r2 \
   -o r2stats -- python /root/repo/bench.py --steps 40 --warmup 15) \
   > gpurun_out/prof_bench.log 2>&1
 echo "rocprof rc=$?"
@@ -17,7 +17,7 @@ echo "--- top kernels ---"
 head -14 gpurun_out/r2_kernel_stats.csv 2>/dev/null | cut -c1-140
 echo "=== PMC counters ==="
 rm -rf /tmp/prof_pmc && mkdir -p /tmp/prof_pmc
-(cd /tmp && timeout 500 rocprofv3 --pmc SQ_BUSY_CYCLES SQ_VALU_MFMA_BUSY_CYCLES \
+(cd /tmp && timeout 500 rocprofv3 --output-format csv --pmc SQ_BUSY_CYCLES SQ_VALU_MFMA_BUSY_CYCLES \
   SQ_LDS_BANK_CONFLICT SQ_INSTS_VALU SQ_INSTS_MFMA -d /tmp/prof_pmc -o r2pmc \
   -- python /root/repo/bench.py --steps 15 --warmup 8 --no-autotune) \
   > gpurun_out/prof_pmc.log 2>&1
