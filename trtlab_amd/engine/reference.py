@@ -105,8 +105,11 @@ def run_reference(plan: EnginePlan, input_nhwc, return_all: bool = False):
                     y = y + seg[0]
             t[op.output] = y
         elif op.kind == K_QUANTIZE:
-            t[op.output] = torch.clamp(torch.round(x / d["q_scale"]),
-                                       -127, 127)
+            if d.get("epi") == 1:  # fp8 e4m3 (continuous grid, sat 448)
+                t[op.output] = _fp8_round(x / d["q_scale"])
+            else:  # int8 codes
+                t[op.output] = torch.clamp(torch.round(x / d["q_scale"]),
+                                           -127, 127)
         elif op.kind == K_DEQUANT:
             t[op.output] = x * d["q_scale"]
         elif op.kind == K_GEMM:

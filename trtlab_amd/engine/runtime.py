@@ -188,13 +188,23 @@ class InferenceManager:
     def __init__(self, max_contexts: int = 2, device: int = 0,
                  pre_threads: int = 1, hip_threads: int = 1,
                  post_threads: int = 2, shared_arena: bool = False,
-                 arena_max_bytes: int = 0):
+                 arena_max_bytes: int = 0, max_executions: int = 0):
         """shared_arena=True: all models' activation arenas are carved from
         ONE growing best-fit device pool (native DeviceArena — reference
         bfit_allocator/block_arena role) instead of per-context hipMallocs;
-        pool stats (high-water, histogram) export via arena_stats()."""
+        pool stats (high-water, histogram) export via arena_stats().
+
+        max_executions: GLOBAL cap on concurrently-launched forwards across
+        ALL models — the reference's two-level concurrency limiter
+        (inference_manager.cc:254-282: the global ExecutionContext pool
+        gates on top of the per-model context pools). 0 = per-model pools
+        only."""
+        import threading as _threading
+
         self.device = device
         self.max_contexts = max_contexts
+        self._exec_sem = (_threading.Semaphore(max_executions)
+                          if max_executions > 0 else None)
         self._models: Dict[str, NativeEngine] = {}
         self._ctx_pools: Dict[str, Pool] = {}
         self.arena = None
@@ -301,11 +311,17 @@ class InferRunner:
                 fut.set_exception(e)
                 co.release()
 
+        sem = self.manager._exec_sem
+
         def hip_stage(co):
             try:
+                if sem is not None:
+                    sem.acquire()  # level 2: global cross-model cap
                 co.item.launch()
                 tp["post"].enqueue(post, co)
             except BaseException as e:  # noqa: BLE001
+                if sem is not None:
+                    sem.release()
                 fut.set_exception(e)
                 co.release()
 
@@ -320,6 +336,8 @@ class InferRunner:
                     else:
                         out = np.array(co.item.output, copy=True)
                 finally:
+                    if sem is not None:
+                        sem.release()
                     co.release()
                 fut.set_result(out)
             except BaseException as e:  # noqa: BLE001

@@ -23,7 +23,7 @@ def build_gpt2(batch: int = 8, seq: int = 1024, hidden: int = 768,
                embeddings: bool = False, vocab: int = 50257) -> Graph:
     """embeddings=True: int32 token ids input [B*S] -> tok+pos gather.
     Otherwise pre-embedded hidden states [B*S, hidden] fp16."""
-    assert seq % 128 == 0, "attention kernel: seq must be a multiple of 128"
+    assert seq >= 1  # attention streams key tiles at ANY sequence length
     assert hidden % heads == 0 and hidden // heads == 64, \
         "attention kernel: head_dim must be 64"
     inter = hidden * 4
