@@ -172,3 +172,50 @@ def test_fenced_host_buffer_detects_overrun():
     b._raw[-1] = 0  # simulate an overrun into the back fence
     with _pytest.raises(MemoryError):
         b.check()
+
+
+def test_two_level_execution_limiter():
+    """reference inference_manager.cc:254-282: per-model context pools
+    gate level 1; a GLOBAL max_executions semaphore gates level 2 across
+    all models. With max_executions=1 two models' forwards serialize."""
+    import threading
+    import time
+
+    import numpy as np
+
+    from trtlab_amd.core import Pool
+    from trtlab_amd.engine.runtime import InferenceManager, InferRunner
+
+    mgr = InferenceManager(max_contexts=2, max_executions=1)
+    inflight = [0]
+    peak = [0]
+    lock = threading.Lock()
+
+    class _FakePlan:
+        outputs = [dict(name="y")]
+
+    class _FakeCtx:
+        plan = _FakePlan()
+        output = np.zeros(4)
+
+        def write_input(self, batch, name=None):
+            pass
+
+        def launch(self):
+            with lock:
+                inflight[0] += 1
+                peak[0] = max(peak[0], inflight[0])
+            time.sleep(0.05)  # "GPU" time while holding the global slot
+
+        def synchronize(self):
+            with lock:
+                inflight[0] -= 1
+
+    for name in ("m1", "m2"):
+        mgr._ctx_pools[name] = Pool([_FakeCtx(), _FakeCtx()])
+    futs = [InferRunner(mgr, name).infer(np.zeros(4))
+            for name in ("m1", "m2") for _ in range(3)]
+    for f in futs:
+        f.result(timeout=30)
+    assert peak[0] == 1  # global limiter held launches to one at a time
+    mgr.shutdown()
