@@ -64,3 +64,36 @@ def test_device_windowed_stack_roundtrip():
         C.memory.memcpy_d2h(back, ptr, back.nbytes)
         assert np.array_equal(back, w)
     stack.close()
+
+
+def test_reserved_windowed_stack():
+    from trtlab_amd.core.windowed import ReservedWindowedStack
+
+    seen = []
+    st = ReservedWindowedStack(8, 3, on_window=lambda w, i: seen.append(
+        (i, w.copy())))
+    v = st.reserve_window()
+    assert v.shape == (5,)  # stride = window - overlap
+    v[:] = np.arange(5)
+    st.commit_window()
+    v2 = st.reserve_window()
+    v2[:] = np.arange(5) + 10
+    st.commit_window()
+    assert len(seen) == 2
+    # window 1 carries window 0's trailing overlap samples
+    np.testing.assert_array_equal(seen[1][1][:3], seen[0][1][-3:])
+    np.testing.assert_array_equal(seen[1][1][3:], np.arange(5) + 10)
+    with pytest.raises(RuntimeError):
+        st.commit_window()  # nothing reserved
+
+
+def test_windowed_task_executor():
+    from trtlab_amd.core.windowed import WindowedTaskExecutor
+
+    ex = WindowedTaskExecutor(4, 1, lambda w, i: (i, float(w.sum())),
+                              workers=2)
+    ex.push(np.arange(16, dtype=np.float32))
+    res = ex.results()
+    assert [i for i, _ in res] == list(range(len(res)))
+    assert len(res) >= 4
+    ex.shutdown()

@@ -7,6 +7,7 @@ window is contiguous).
 """
 from __future__ import annotations
 
+import threading
 from typing import Callable, Iterator, List, Optional
 
 import numpy as np
@@ -107,3 +108,89 @@ class DeviceCyclicWindowedStack:
             self.close()
         except Exception:
             pass
+
+
+class ReservedWindowedStack(CyclicWindowedBuffer):
+    """Zero-copy producer variant (reference
+    cyclic_windowed_reserved_stack::reserve_window,
+    cyclic_windowed_buffer.h:287-365): reserve_window() hands the producer
+    a writable view of the NEXT window's fresh region — the overlap with
+    the previous window is pre-filled — and commit_window() emits it.
+    Avoids the push() concatenate/copy for producers that can write in
+    place (e.g. a decoder writing feature frames)."""
+
+    def __init__(self, window_size: int, overlap: int, sample_shape=(),
+                 dtype=np.float32, on_window=None):
+        super().__init__(window_size, overlap, sample_shape=sample_shape,
+                         dtype=dtype, on_window=on_window)
+        self._reserved: Optional[np.ndarray] = None
+
+    def reserve_window(self) -> np.ndarray:
+        """Writable [stride, *sample_shape] view for the window's NEW
+        samples (the first `overlap` samples are carried over)."""
+        if self._reserved is not None:
+            raise RuntimeError("a window is already reserved (commit first)")
+        self._win = np.zeros((self.window_size, *self.sample_shape),
+                             self.dtype)
+        if self.overlap and len(self._pending) >= self.overlap:
+            self._win[:self.overlap] = self._pending[-self.overlap:]
+        self._reserved = self._win[self.overlap:]
+        return self._reserved
+
+    def commit_window(self) -> int:
+        """Emit the reserved window; returns its window id."""
+        if self._reserved is None:
+            raise RuntimeError("no reserved window")
+        wid = self._window_id
+        if self._on_window:
+            self._on_window(self._win, wid)
+        else:
+            self._out.append(self._win)
+        self._window_id += 1
+        # trailing overlap becomes the next window's carried samples
+        self._pending = self._win[-self.overlap:].copy() if self.overlap \
+            else np.zeros((0, *self.sample_shape), self.dtype)
+        self._reserved = None
+        return wid
+
+
+class WindowedTaskExecutor:
+    """Per-window compute executor (reference
+    cyclic_windowed_task_executor::on_compute_window,
+    cyclic_windowed_buffer.h:369-440): push samples in, `compute_fn`
+    runs on a worker thread for every completed window, results are
+    collected in order."""
+
+    def __init__(self, window_size: int, overlap: int, compute_fn,
+                 sample_shape=(), dtype=np.float32, workers: int = 1):
+        from trtlab_amd.core import ThreadPool
+
+        self._pool = ThreadPool(workers, "windowed")
+        self._results: dict = {}
+        self._lock = threading.Lock()
+        self._futs: List = []
+
+        def on_window(win: np.ndarray, wid: int):
+            def run(w=win.copy(), i=wid):
+                r = compute_fn(w, i)
+                with self._lock:
+                    self._results[i] = r
+
+            self._futs.append(self._pool.enqueue(run))
+
+        self.buffer = CyclicWindowedBuffer(window_size, overlap,
+                                           sample_shape=sample_shape,
+                                           dtype=dtype, on_window=on_window)
+
+    def push(self, data: np.ndarray) -> int:
+        return self.buffer.push(data)
+
+    def results(self) -> List:
+        """Wait for all scheduled windows; results in window order."""
+        for f in self._futs:
+            f.result()
+        with self._lock:
+            return [self._results[i] for i in sorted(self._results)]
+
+    def shutdown(self):
+        self._pool.shutdown()
