@@ -566,6 +566,42 @@ PYBIND11_MODULE(_C, m) {
           py::arg("ids"), py::arg("tok"), py::arg("posemb"), py::arg("out"),
           py::arg("pos"), py::arg("B"), py::arg("hidden"),
           py::arg("stream") = 0, py::arg("sync") = true);
+  ops.def("kv_append_chunk",
+          [](uintptr_t qkv, uintptr_t kc, uintptr_t vc, uintptr_t pos, int B,
+             int H, int K, int smax, uintptr_t stream, bool sync) {
+            launch_kv_append_chunk((void*)qkv, (void*)kc, (void*)vc,
+                                   (void*)pos, B, H, K, smax,
+                                   as_stream(stream));
+            if (sync) TRT_HIP_CHECK(hipStreamSynchronize(as_stream(stream)));
+          },
+          py::arg("qkv"), py::arg("kcache"), py::arg("vcache"),
+          py::arg("pos"), py::arg("B"), py::arg("H"), py::arg("K"),
+          py::arg("smax"), py::arg("stream") = 0, py::arg("sync") = true);
+  ops.def("chunk_attention",
+          [](uintptr_t qkv, uintptr_t kc, uintptr_t vc, uintptr_t out,
+             uintptr_t pos, int B, int H, int K, int smax, float scale,
+             uintptr_t stream, bool sync) {
+            launch_chunk_attention((void*)qkv, (void*)kc, (void*)vc,
+                                   (void*)out, (void*)pos, B, H, K, smax,
+                                   scale, as_stream(stream));
+            if (sync) TRT_HIP_CHECK(hipStreamSynchronize(as_stream(stream)));
+          },
+          py::arg("qkv"), py::arg("kcache"), py::arg("vcache"),
+          py::arg("out"), py::arg("pos"), py::arg("B"), py::arg("H"),
+          py::arg("K"), py::arg("smax"), py::arg("scale"),
+          py::arg("stream") = 0, py::arg("sync") = true);
+  ops.def("chunk_embed",
+          [](uintptr_t ids, uintptr_t tok, uintptr_t pe, uintptr_t out,
+             uintptr_t pos, int B, int K, int smax, int hidden,
+             uintptr_t stream, bool sync) {
+            launch_chunk_embed((void*)ids, (void*)tok, (void*)pe, (void*)out,
+                               (void*)pos, B, K, smax, hidden,
+                               as_stream(stream));
+            if (sync) TRT_HIP_CHECK(hipStreamSynchronize(as_stream(stream)));
+          },
+          py::arg("ids"), py::arg("tok"), py::arg("posemb"), py::arg("out"),
+          py::arg("pos"), py::arg("B"), py::arg("K"), py::arg("smax"),
+          py::arg("hidden"), py::arg("stream") = 0, py::arg("sync") = true);
   ops.def("advance_pos", [](uintptr_t pos, int B, int smax, uintptr_t stream,
                             bool sync) {
     launch_advance_pos((void*)pos, B, smax, as_stream(stream));

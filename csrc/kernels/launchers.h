@@ -108,6 +108,17 @@ void launch_decode_attention(const void* qkv, const void* kcache,
 void launch_decode_embed(const void* ids, const void* tok, const void* posemb,
                          void* out, const void* pos, int B, int hidden,
                          hipStream_t stream);
+// speculative-decoding verification chunk (see decode.hip)
+void launch_kv_append_chunk(const void* qkv, void* kcache, void* vcache,
+                            const void* pos, int B, int H, int K, int smax,
+                            hipStream_t stream);
+void launch_chunk_attention(const void* qkv, const void* kcache,
+                            const void* vcache, void* out, const void* pos,
+                            int B, int H, int K, int smax, float scale,
+                            hipStream_t stream);
+void launch_chunk_embed(const void* ids, const void* tok, const void* posemb,
+                        void* out, const void* pos, int B, int K, int smax,
+                        int hidden, hipStream_t stream);
 void launch_advance_pos(void* pos, int B, int smax, hipStream_t stream);
 
 void launch_embedding(int dtype, const void* ids, const void* tok,

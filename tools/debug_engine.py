@@ -78,7 +78,7 @@ def per_layer_errors(plan, x, device: int = 0):
         shape = plan.shapes[t]
         dt = out_dtype.get(t, 0)
         np_dt = {0: np.float16, 1: np.int16, 2: np.int8,
-                 3: np.uint8}[dt]
+                 3: np.uint8, 4: np.float16}[dt]
         buf = np.empty(int(np.prod(shape)), dtype=np_dt)
         C.memory.memcpy_d2h(buf, arena + plan.offsets[t], buf.nbytes)
         if dt == 3:  # fp8 e4m3 codes -> float via torch

@@ -90,6 +90,8 @@ def run_reference(plan: EnginePlan, input_nhwc, return_all: bool = False):
             y = _epi(acc, d["epi"], scale, bias, res)
             if d["dtype"] == 2:  # int8: emulate the requantized store
                 y = torch.clamp(torch.round(y), -127, 127)
+            elif d["dtype"] == 3:  # fp8: emulate the e4m3 output store
+                y = _fp8_round(y)
             t[op.output] = y
         elif op.kind == K_EMBEDDING:
             ids = x.long()
