@@ -959,3 +959,71 @@ def test_decode_idle_slot_masking():
         np.testing.assert_allclose(replay[1], ref_outs[i][0],
                                    rtol=2e-2, atol=2e-2)
     s.close()
+
+
+def test_chunk_attention_matches_sequential():
+    """Spec-decode chunk kernels vs the sequential decode path: feeding K
+    tokens through verify_chunk must produce the same logits as K
+    sequential step() calls (same caches, same math)."""
+    from trtlab_amd.engine.decode import DecodeSession
+    from trtlab_amd.models import build_gpt2
+
+    g = build_gpt2(batch=2, seq=32, layers=2, seed=0, embeddings=True)
+    rng = np.random.RandomState(1)
+    toks = rng.randint(1, 4000, size=(5, 2)).astype(np.int32)
+
+    seq = DecodeSession(g, batch=2, smax=64, capture=False, lm_head=True)
+    seq_logits = [seq.step(toks[i]) for i in range(5)]
+    seq.close()
+
+    ch = DecodeSession(g, batch=2, smax=64, capture=False, lm_head=True)
+    first = ch.verify_chunk(toks[0][:, None])[:, 0]
+    np.testing.assert_allclose(first, seq_logits[0], rtol=2e-2, atol=2e-2)
+    ch.add_pos(np.ones(2, np.int64))
+    chunk = toks[1:5].T.copy()  # [B, 4]
+    cl = ch.verify_chunk(chunk)
+    for i in range(4):
+        np.testing.assert_allclose(cl[:, i], seq_logits[i + 1], rtol=3e-2,
+                                   atol=3e-2)
+    ch.close()
+
+
+def test_speculative_decoding_invariant():
+    """THE spec-decode correctness property: the generated stream equals
+    target-only greedy decoding REGARDLESS of the draft — identical-draft
+    (acceptance ~1.0) and mismatched-draft (low acceptance) both
+    reproduce the baseline sequence exactly."""
+    from trtlab_amd.engine.decode import DecodeSession, SpeculativeDecoder
+    from trtlab_amd.models import build_gpt2
+
+    g = build_gpt2(batch=2, seq=32, layers=2, seed=0, embeddings=True)
+    g_draft_same = build_gpt2(batch=2, seq=32, layers=2, seed=0,
+                              embeddings=True)
+    g_draft_diff = build_gpt2(batch=2, seq=32, layers=1, seed=9,
+                              embeddings=True)
+    seed_tok = np.array([17, 23], np.int32)
+    STEPS = 12
+
+    # baseline: target-only greedy
+    base = DecodeSession(g, batch=2, smax=96, capture=False, lm_head=True)
+    cur = seed_tok
+    ref = []
+    for _ in range(STEPS):
+        lg = base.step(cur)
+        cur = lg.argmax(-1).astype(np.int32)
+        ref.append(cur)
+    base.close()
+    ref = np.stack(ref, axis=1)  # [B, STEPS]
+
+    for gd, expect_high in ((g_draft_same, True), (g_draft_diff, False)):
+        t = DecodeSession(g, batch=2, smax=96, capture=False, lm_head=True)
+        d = DecodeSession(gd, batch=2, smax=96, capture=False, lm_head=True)
+        sd = SpeculativeDecoder(t, d, k=3)
+        toks, rate = sd.generate(seed_tok, STEPS)
+        np.testing.assert_array_equal(toks, ref)
+        if expect_high:
+            assert rate > 0.9, f"identical draft should accept ~all: {rate}"
+        else:
+            assert rate < 0.9, f"mismatched draft acceptance: {rate}"
+        t.close()
+        d.close()

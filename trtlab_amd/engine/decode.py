@@ -304,6 +304,120 @@ class DecodeSession:
         return (self.logits.float().cpu().numpy()
                 if self.logits is not None else out)
 
+    # ------------------------------------------- speculative verification
+    def verify_chunk(self, tokens: np.ndarray) -> np.ndarray:
+        """Score K proposed tokens per slot in ONE chunked forward
+        (speculative decoding's verification step): tokens [B, K] are
+        consumed at positions pos[b]..pos[b]+K-1 (their K/V overwrite any
+        stale entries), and the TARGET logits for each position come back
+        as [B, K, vocab] fp32. pos is NOT advanced — call add_pos() with
+        the per-slot accepted counts. Requires lm_head=True."""
+        import torch
+
+        if self.logits is None:
+            raise RuntimeError("verify_chunk requires lm_head=True")
+        tokens = np.ascontiguousarray(tokens, np.int32)
+        B, K = tokens.shape
+        assert B == self.batch and K >= 1
+        M = B * K
+        Hd, inter = self.hidden, self.inter
+        ops, s = self._C.ops, self.stream
+        torch = self._torch
+
+        cb = getattr(self, "_chunk_bufs", None)
+        if cb is None or cb["K"] != K:
+            cb = dict(
+                K=K,
+                ids=torch.zeros(M, dtype=torch.int32, device="cuda"),
+                h=torch.zeros(M, Hd, dtype=torch.half, device="cuda"),
+                x=torch.zeros(M, Hd, dtype=torch.half, device="cuda"),
+                x2=torch.zeros(M, Hd, dtype=torch.half, device="cuda"),
+                qkv=torch.zeros(M, 3 * Hd, dtype=torch.half, device="cuda"),
+                att=torch.zeros(M, Hd, dtype=torch.half, device="cuda"),
+                ff=torch.zeros(M, inter, dtype=torch.half, device="cuda"),
+                logits=torch.zeros(M, self.vocab, dtype=torch.half,
+                                   device="cuda"),
+            )
+            self._chunk_bufs = cb
+        self._C.memory.memcpy_h2d(cb["ids"].data_ptr(), tokens.reshape(-1),
+                                  tokens.nbytes)
+        ops.chunk_embed(cb["ids"].data_ptr(), self.tok.data_ptr(),
+                        self.posemb.data_ptr(), cb["h"].data_ptr(),
+                        self.pos.data_ptr(), B, K, self.smax, Hd, stream=s,
+                        sync=False)
+        ops.layernorm(0, cb["h"].data_ptr(),
+                      self.layers[0]["ln1_g"].data_ptr(),
+                      self.layers[0]["ln1_b"].data_ptr(),
+                      cb["x"].data_ptr(), M, Hd, stream=s, sync=False)
+        for li, lay in enumerate(self.layers):
+            ops.gemm_bt(0, cb["x"].data_ptr(), lay["qkv_w"].data_ptr(),
+                        cb["qkv"].data_ptr(), bias=lay["qkv_b"].data_ptr(),
+                        M=M, N=3 * Hd, K=Hd, epi=self._epi_bias, stream=s,
+                        sync=False)
+            ops.kv_append_chunk(cb["qkv"].data_ptr(),
+                                lay["kcache"].data_ptr(),
+                                lay["vcache"].data_ptr(),
+                                self.pos.data_ptr(), B, self.heads, K,
+                                self.smax, stream=s, sync=False)
+            ops.chunk_attention(cb["qkv"].data_ptr(),
+                                lay["kcache"].data_ptr(),
+                                lay["vcache"].data_ptr(),
+                                cb["att"].data_ptr(), self.pos.data_ptr(),
+                                B, self.heads, K, self.smax,
+                                1.0 / float(np.sqrt(64.0)), stream=s,
+                                sync=False)
+            ops.gemm_bt(0, cb["att"].data_ptr(), lay["proj_w"].data_ptr(),
+                        cb["x2"].data_ptr(), bias=lay["proj_b"].data_ptr(),
+                        M=M, N=Hd, K=Hd, epi=self._epi_bias, stream=s,
+                        sync=False)
+            ops.add_layernorm(0, cb["x2"].data_ptr(), cb["h"].data_ptr(),
+                              lay["ln2_g"].data_ptr(),
+                              lay["ln2_b"].data_ptr(), cb["x"].data_ptr(),
+                              sum_out=cb["h"].data_ptr(), M=M, N=Hd,
+                              stream=s, sync=False)
+            ops.gemm_bt(0, cb["x"].data_ptr(), lay["ff1_w"].data_ptr(),
+                        cb["ff"].data_ptr(), bias=lay["ff1_b"].data_ptr(),
+                        M=M, N=inter, K=Hd, epi=self._epi_gelu, stream=s,
+                        sync=False)
+            ops.gemm_bt(0, cb["ff"].data_ptr(), lay["ff2_w"].data_ptr(),
+                        cb["x2"].data_ptr(), bias=lay["ff2_b"].data_ptr(),
+                        M=M, N=Hd, K=inter, epi=self._epi_bias, stream=s,
+                        sync=False)
+            nxt = (self.layers[li + 1] if li + 1 < self.n_layers else None)
+            gptr = (nxt["ln1_g"] if nxt else self.lnf_g).data_ptr()
+            bptr = (nxt["ln1_b"] if nxt else self.lnf_b).data_ptr()
+            ops.add_layernorm(0, cb["x2"].data_ptr(), cb["h"].data_ptr(),
+                              gptr, bptr, cb["x"].data_ptr(),
+                              sum_out=cb["h"].data_ptr(), M=M, N=Hd,
+                              stream=s, sync=False)
+        ops.gemm_bt(0, cb["x"].data_ptr(), self.tok.data_ptr(),
+                    cb["logits"].data_ptr(), M=M, N=self.vocab, K=Hd,
+                    epi=self._epi_none, stream=s, sync=False)
+        self._C.hip.stream_synchronize(s)
+        return cb["logits"].float().cpu().numpy().reshape(B, K, self.vocab)
+
+    def add_pos(self, counts: np.ndarray) -> None:
+        """Advance each slot's position by counts[b] (speculative
+        acceptance: the chunk's first counts[b] tokens are now consumed)."""
+        counts = np.asarray(counts, np.int64)
+        cur = self.pos.cpu().numpy().astype(np.int64)
+        new = np.where(cur >= 0, np.minimum(cur + counts, self.smax - 1),
+                       cur)
+        self.pos.copy_(self._torch.from_numpy(new.astype(np.int32)).cuda())
+        self._slot_steps = np.where(cur >= 0,
+                                    self._slot_steps + counts,
+                                    self._slot_steps)
+        self._torch.cuda.synchronize()
+
+    def set_pos(self, positions: np.ndarray) -> None:
+        """Set per-slot absolute positions (speculative rollback of a
+        draft session to the target's accepted frontier)."""
+        arr = np.ascontiguousarray(positions, np.int32)
+        self.pos.copy_(self._torch.from_numpy(arr).cuda())
+        self._slot_steps = arr.astype(np.int64).clip(min=0)
+        self._active = arr >= 0
+        self._torch.cuda.synchronize()
+
     def reset_slot(self, b: int) -> None:
         """Restart slot b at position 0 (continuous batching in lockstep:
         the next step() token for this slot begins a fresh sequence while
@@ -333,3 +447,80 @@ class DecodeSession:
             self.close()
         except Exception:
             pass
+
+
+class SpeculativeDecoder:
+    """Greedy speculative decoding over two DecodeSessions (beyond-
+    reference serving capability): a cheap DRAFT model proposes k tokens
+    per round, the TARGET verifies all k in ONE chunked forward
+    (verify_chunk), and the longest matching prefix is accepted plus the
+    target's own correction token. The emitted stream is IDENTICAL to
+    greedy decoding with the target alone — the draft only changes speed,
+    never output (the invariant the GPU test checks with a mismatched
+    draft). Per-slot positions let different slots accept different
+    lengths each round.
+
+    Both sessions need lm_head=True and the same vocab; prime both on the
+    same context before generate().
+    """
+
+    def __init__(self, target: "DecodeSession", draft: "DecodeSession",
+                 k: int = 4):
+        if target.logits is None or draft.logits is None:
+            raise ValueError("both sessions need lm_head=True")
+        if target.vocab != draft.vocab or target.batch != draft.batch:
+            raise ValueError("vocab/batch mismatch between target and draft")
+        self.target = target
+        self.draft = draft
+        self.k = k
+        self.proposed = 0
+        self.accepted = 0
+
+    def generate(self, seed: np.ndarray, steps: int):
+        """Greedy-generate `steps` tokens per slot after consuming `seed`
+        [B]. Returns (tokens [B, steps] int32, acceptance_rate)."""
+        B, k = self.target.batch, self.k
+        seed = np.ascontiguousarray(seed, np.int32)
+        lt = self.target.verify_chunk(seed[:, None])[:, 0]  # consume seed
+        self.target.add_pos(np.ones(B, np.int64))
+        ld = self.draft.verify_chunk(seed[:, None])[:, 0]
+        self.draft.add_pos(np.ones(B, np.int64))
+
+        out = [[] for _ in range(B)]
+        while min(len(o) for o in out) < steps:
+            # ---- draft chain: k greedy proposals ----
+            props = np.zeros((B, k), np.int32)
+            cur = ld.argmax(-1).astype(np.int32)
+            for i in range(k):
+                props[:, i] = cur
+                ld = self.draft.verify_chunk(cur[:, None])[:, 0]
+                self.draft.add_pos(np.ones(B, np.int64))
+                cur = ld.argmax(-1).astype(np.int32)
+            # ---- target verifies the whole chunk at once ----
+            tl = self.target.verify_chunk(props)  # [B, k, vocab]
+            accept = np.zeros(B, np.int64)
+            corr = np.zeros(B, np.int32)
+            for b in range(B):
+                g = int(lt[b].argmax())
+                a = 0
+                while a < k and props[b, a] == g:
+                    out[b].append(g)
+                    g = int(tl[b, a].argmax())
+                    a += 1
+                accept[b] = a
+                corr[b] = g          # target's own next token
+                out[b].append(g)
+            self.proposed += B * k
+            self.accepted += int(accept.sum())
+            # ---- advance target past the accepted prefix, consume the
+            # correction token (its K/V overwrites the rejected slot) ----
+            self.target.add_pos(accept)
+            lt = self.target.verify_chunk(corr[:, None])[:, 0]
+            self.target.add_pos(np.ones(B, np.int64))
+            # ---- roll the draft back to the target's frontier ----
+            tpos = self.target.pos.cpu().numpy().astype(np.int32)
+            self.draft.set_pos(tpos - 1)
+            ld = self.draft.verify_chunk(corr[:, None])[:, 0]
+            self.draft.add_pos(np.ones(B, np.int64))
+        rate = self.accepted / max(self.proposed, 1)
+        return (np.array([o[:steps] for o in out], np.int32), rate)
