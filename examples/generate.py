@@ -26,6 +26,11 @@ def main():
     ap.add_argument("--layers", type=int, default=12)
     ap.add_argument("--prompt-len", type=int, default=16)
     ap.add_argument("--new-tokens", type=int, default=32)
+    ap.add_argument("--speculative", type=int, default=0, metavar="K",
+                    help="speculative decoding with a draft of "
+                         "--draft-layers layers proposing K tokens/round "
+                         "(output provably identical to plain greedy)")
+    ap.add_argument("--draft-layers", type=int, default=2)
     args = ap.parse_args()
 
     g = build_gpt2(batch=args.batch, seq=1024, layers=args.layers, seed=0,
@@ -38,14 +43,36 @@ def main():
     logits = sess.prefill(prompt)  # one fused pass fills the KV caches
 
     toks = np.argmax(logits, axis=1).astype(np.int32)
-    out = [toks]
-    t0 = time.perf_counter()
-    for _ in range(args.new_tokens - 1):
-        logits = sess.step(toks)
-        toks = np.argmax(logits, axis=1).astype(np.int32)
-        out.append(toks)
-    dt = time.perf_counter() - t0
-    seqs = np.stack(out, axis=1)
+    if args.speculative:
+        # draft (same weights family, fewer layers) proposes K per round;
+        # the target verifies each chunk in ONE forward
+        from trtlab_amd.engine.decode import SpeculativeDecoder
+
+        gd = build_gpt2(batch=args.batch, seq=1024,
+                        layers=args.draft_layers, seed=0, embeddings=True)
+        draft = DecodeSession(gd, batch=args.batch, smax=1024,
+                              capture=False, lm_head=True)
+        draft.prefill(prompt)
+        target = DecodeSession(g, batch=args.batch, smax=1024,
+                               capture=False, lm_head=True)
+        target.prefill(prompt)
+        sd = SpeculativeDecoder(target, draft, k=args.speculative)
+        t0 = time.perf_counter()
+        seqs, rate_acc = sd.generate(toks, args.new_tokens - 1)
+        dt = time.perf_counter() - t0
+        seqs = np.concatenate([toks[:, None], seqs], axis=1)
+        print(f"speculative acceptance rate: {rate_acc:.2f}")
+        target.close()
+        draft.close()
+    else:
+        out = [toks]
+        t0 = time.perf_counter()
+        for _ in range(args.new_tokens - 1):
+            logits = sess.step(toks)
+            toks = np.argmax(logits, axis=1).astype(np.int32)
+            out.append(toks)
+        dt = time.perf_counter() - t0
+        seqs = np.stack(out, axis=1)
     print("generated token ids (greedy):")
     for b in range(args.batch):
         print(f"  seq{b}: {seqs[b][:16].tolist()} ...")
