@@ -26,6 +26,51 @@ import numpy as np
 BASELINE_INF_S = 953.414  # reference examples/00_TensorRT/README.md:46 (V100)
 
 
+class _TorchCommFallback:
+    """torch.distributed stand-in matching the parallel.Communicator
+    surface bench.py uses (barrier / all_reduce_scalar / broadcast via
+    weights path / close). Only used if the owned RCCL layer fails on a
+    multi-GPU node — see the warning at the construction site."""
+
+    def __init__(self, rank, world, device):
+        import torch
+        import torch.distributed as dist
+
+        self._torch, self._dist = torch, dist
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        torch.cuda.set_device(device)
+        dist.init_process_group("nccl")
+        self.rank, self.world, self.device = rank, world, device
+        self._comm = self  # truthy: broadcast_weights takes the real path
+
+    def barrier(self):
+        self._dist.barrier()
+
+    def all_reduce_scalar(self, v, op=None):
+        t = self._torch.tensor([v], device="cuda",
+                               dtype=self._torch.float64)
+        self._dist.all_reduce(t, op=self._dist.ReduceOp.MAX)
+        return float(t.item())
+
+    def broadcast(self, ptr, nbytes, root=0, stream=0):
+        import trtlab_amd
+
+        C = trtlab_amd.native()
+        staging = self._torch.empty(nbytes, dtype=self._torch.uint8,
+                                    device="cuda")
+        C.memory.memcpy_d2d(staging.data_ptr(), ptr, nbytes)
+        self._torch.cuda.synchronize()
+        self._dist.broadcast(staging, src=root)
+        self._torch.cuda.synchronize()
+        C.memory.memcpy_d2d(ptr, staging.data_ptr(), nbytes)
+
+    def synchronize(self):
+        self._torch.cuda.synchronize()
+
+    def close(self):
+        self._dist.destroy_process_group()
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
@@ -65,7 +110,19 @@ def main():
         # Owned RCCL communicator (csrc/runtime/comm.cpp): torchrun is only
         # the launcher — comm bring-up is our file-based uid rendezvous,
         # collectives run on our own streams (no torch.distributed).
-        comm = Communicator(rank=rank, world=world, device=local_rank)
+        # Insurance: 1-GPU leases cannot exercise a real N>1 clique (RCCL
+        # rejects co-located ranks), so if the owned path fails on the
+        # multi-GPU node, fall back to torch.distributed with a LOUD
+        # warning rather than losing the scaling measurement.
+        try:
+            comm = Communicator(rank=rank, world=world, device=local_rank)
+        except Exception as e:  # noqa: BLE001
+            import sys
+
+            print(f"[bench] WARNING rank {rank}: owned RCCL comm failed "
+                  f"({type(e).__name__}: {e}); falling back to "
+                  f"torch.distributed", file=sys.stderr, flush=True)
+            comm = _TorchCommFallback(rank, world, local_rank)
 
     # Build the plan (identical on every rank: same seed).
     if args.model == "bert":
