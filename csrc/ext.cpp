@@ -566,6 +566,30 @@ PYBIND11_MODULE(_C, m) {
           py::arg("ids"), py::arg("tok"), py::arg("posemb"), py::arg("out"),
           py::arg("pos"), py::arg("B"), py::arg("hidden"),
           py::arg("stream") = 0, py::arg("sync") = true);
+  ops.def("decode_gemm_fused",
+          [](int pro, int epi, uintptr_t x, uintptr_t r, uintptr_t h_out,
+             uintptr_t gamma, uintptr_t beta, uintptr_t Bw, uintptr_t bias,
+             uintptr_t C, uintptr_t ids, uintptr_t tok, uintptr_t posemb,
+             uintptr_t pos, uintptr_t kcache, uintptr_t vcache, int M,
+             int N, int K, int heads, int smax, float eps, uintptr_t stream,
+             bool sync) {
+            launch_decode_gemm_fused(
+                pro, epi, (const void*)x, (const void*)r, (void*)h_out,
+                (const float*)gamma, (const float*)beta, (const void*)Bw,
+                (const float*)bias, (void*)C, (const void*)ids,
+                (const void*)tok, (const void*)posemb, (const void*)pos,
+                (void*)kcache, (void*)vcache, M, N, K, heads, smax, eps,
+                as_stream(stream));
+            if (sync) TRT_HIP_CHECK(hipStreamSynchronize(as_stream(stream)));
+          },
+          py::arg("pro"), py::arg("epi"), py::arg("x") = 0, py::arg("r") = 0,
+          py::arg("h_out") = 0, py::arg("gamma") = 0, py::arg("beta") = 0,
+          py::arg("B") = 0, py::arg("bias") = 0, py::arg("C") = 0,
+          py::arg("ids") = 0, py::arg("tok") = 0, py::arg("posemb") = 0,
+          py::arg("pos") = 0, py::arg("kcache") = 0, py::arg("vcache") = 0,
+          py::arg("M") = 0, py::arg("N") = 0, py::arg("K") = 0,
+          py::arg("heads") = 0, py::arg("smax") = 0, py::arg("eps") = 1e-5f,
+          py::arg("stream") = 0, py::arg("sync") = true);
   ops.def("kv_append_chunk",
           [](uintptr_t qkv, uintptr_t kc, uintptr_t vc, uintptr_t pos, int B,
              int H, int K, int smax, uintptr_t stream, bool sync) {

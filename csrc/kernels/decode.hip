@@ -334,4 +334,228 @@ void launch_advance_pos(void* pos, int B, int smax, hipStream_t stream) {
                      smax);
 }
 
+
+// ---- fused decode GEMM (horizontal fusion for the latency-bound step) ----
+// The decode step is kernel-COUNT bound (~4.6 us replay floor/kernel), so
+// this kernel folds the surrounding elementwise work into the small-M
+// (M = batch <= 64) GEMMs:
+//   prologue PRO: 1 = LN(x), 2 = ADD_LN(x + r, also storing the new
+//   residual stream h_out — blocks write identical values, benign),
+//   3 = EMBED_LN(tok[ids[b]] + posemb[pos[b]])
+//   epilogue EPI: 0 = +bias, 1 = gelu(+bias), 2 = +bias AND scatter the
+//   K/V column ranges into the caches at pos[b] (replaces kv_append)
+// One block per 64 output columns; rows are the whole batch. LN stats are
+// computed in-block (4 lanes per row, shfl-combined) and the normalized
+// A-tile is ds_written per K-step with the same XOR swizzle the MFMA
+// fragment reads expect. Cuts the GPT-2 step from ~8 to 5 kernels/layer.
+template <int PRO, int EPI>
+__global__ __launch_bounds__(256) void decode_gemm_fused_kernel(
+    const _Float16* __restrict__ x, const _Float16* __restrict__ r,
+    _Float16* __restrict__ h_out, const float* __restrict__ gamma,
+    const float* __restrict__ beta, const _Float16* __restrict__ Bw,
+    const float* __restrict__ bias, _Float16* __restrict__ C,
+    const int* __restrict__ ids, const _Float16* __restrict__ tok,
+    const _Float16* __restrict__ posemb, const int* __restrict__ pos,
+    _Float16* __restrict__ kcache, _Float16* __restrict__ vcache, int M,
+    int N, int K, int heads, int smax, float eps) {
+  constexpr int BM = 64, BN = 64;
+  __shared__ __attribute__((aligned(16))) char smem[2 * (BM + BN) * 128];
+  __shared__ float sm_mean[BM], sm_inv[BM];
+  uint32_t lds0 = (uint32_t)(uintptr_t)&smem[0];
+  constexpr int kABytes = BM * 128;
+  constexpr int kBuf = (BM + BN) * 128;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wr = wave >> 1;
+  const int wc = wave & 1;
+  const int n0 = (int)blockIdx.x * BN;
+
+  // source value of the pre-norm row (x + r / embed), 8 elems at kk
+  auto load8 = [&](int row, int kk, _Float16* v8) {
+    if (PRO == 3) {
+      int p = pos[row];
+      if (p < 0) p = 0;
+      const _Float16* trow = tok + (int64_t)ids[row] * K + kk;
+      const _Float16* prow = posemb + (int64_t)p * K + kk;
+      half8v a = *(const half8v*)trow;
+      half8v b = *(const half8v*)prow;
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        v8[j] = (_Float16)((float)((const _Float16*)&a)[j] +
+                           (float)((const _Float16*)&b)[j]);
+    } else if (PRO == 2) {
+      half8v a = *(const half8v*)(x + (int64_t)row * K + kk);
+      half8v b = *(const half8v*)(r + (int64_t)row * K + kk);
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        v8[j] = (_Float16)((float)((const _Float16*)&a)[j] +
+                           (float)((const _Float16*)&b)[j]);
+    } else {
+      *(half8v*)v8 = *(const half8v*)(x + (int64_t)row * K + kk);
+    }
+  };
+
+  // ---- LN stats: 4 lanes per row, vectorized strided reads ----
+  {
+    int row = tid >> 2;
+    int part = tid & 3;
+    float s = 0.f, ss = 0.f;
+    if (row < M) {
+      for (int kk = part * 8; kk < K; kk += 32) {
+        _Float16 v8[8];
+        load8(row, kk, v8);
+        // block 0 persists the new residual stream h_out = x + r. h_out
+        // MUST be a different buffer than r (ping-pong in the session):
+        // other blocks re-read r during their own staging, so an in-place
+        // update would race. Every block computes identical values; the
+        // n0 gate just avoids duplicate HBM traffic.
+        if ((PRO == 2 || PRO == 3) && n0 == 0 && h_out)
+          *(half8v*)(h_out + (int64_t)row * K + kk) = *(const half8v*)v8;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          float f = (float)v8[j];
+          s += f;
+          ss += f * f;
+        }
+      }
+    }
+    // combine the 4 partial lanes of this row (consecutive lanes)
+    s += __shfl_xor(s, 1, 64);
+    s += __shfl_xor(s, 2, 64);
+    ss += __shfl_xor(ss, 1, 64);
+    ss += __shfl_xor(ss, 2, 64);
+    if (part == 0 && row < M) {
+      float mean = s / K;
+      float var = ss / K - mean * mean;
+      sm_mean[row] = mean;
+      sm_inv[row] = rsqrtf(var + eps);
+    }
+  }
+  __syncthreads();
+
+  typename Mfma16x16x32<_Float16>::accv acc[2][2];
+#pragma unroll
+  for (int i = 0; i < 2; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j) acc[i][j] = {0, 0, 0, 0};
+
+  const int ktiles = K / 64;
+  // stage A' (normalized) + B for tile t into slot
+  auto stage = [&](int t, int slot) {
+    uint32_t base = lds0 + slot * kBuf;
+    // B via async glds (weights)
+    stage_tile<_Float16, BN>(Bw + (int64_t)n0 * K + (int64_t)t * 64, K, n0,
+                             N, base + kABytes, tid);
+    // A' via VALU normalize + ds_write (2 chunks of 8 elems per thread)
+#pragma unroll
+    for (int c = 0; c < 2; ++c) {
+      int idx = c * 256 + tid;        // 0..511
+      int row = idx >> 3;             // 0..63
+      int cb = (idx & 7) * 16;        // byte offset in the 128-B row
+      int kk = t * 64 + cb / 2;
+      _Float16 v8[8];
+      _Float16 o8[8];
+      int rrow = row < M ? row : M - 1;
+      load8(rrow, kk, v8);
+      float mean = sm_mean[rrow], inv = sm_inv[rrow];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float g = gamma ? gamma[kk + j] : 1.0f;
+        float b = beta ? beta[kk + j] : 0.0f;
+        o8[j] = (_Float16)(((float)v8[j] - mean) * inv * g + b);
+      }
+      *(short8v*)((char*)(uintptr_t)(base + row * 128 +
+                                     (cb ^ ((row & 7) << 4)))) =
+          *(const short8v*)o8;
+    }
+  };
+
+  int cur = 0;
+  stage(0, 0);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+  for (int t = 0; t < ktiles; ++t) {
+    if (t + 1 < ktiles) stage(t + 1, cur ^ 1);
+    const char* As = &smem[cur * kBuf];
+    mfma_tile<_Float16, BM, BN>(As, As + kABytes, lane, wr, wc, acc);
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+    cur ^= 1;
+  }
+
+  // ---- epilogue: bias (+gelu) (+K/V scatter) ----
+  const int hid = (EPI == 2) ? N / 3 : 0;  // qkv gemm: N = 3*H*64
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+      int col = n0 + wc * 32 + j * 16 + (lane & 15);
+      if (col >= N) continue;
+      float bi = bias ? bias[col] : 0.0f;
+#pragma unroll
+      for (int rr = 0; rr < 4; ++rr) {
+        int row = wr * 32 + i * 16 + ((lane >> 4) << 2) + rr;
+        if (row >= M) continue;
+        float v = (float)acc[i][j][rr] + bi;
+        if (EPI == 1) v = gelu_tanh(v);
+        _Float16 hv = (_Float16)v;
+        C[(int64_t)row * N + col] = hv;
+        if (EPI == 2) {
+          int p = pos[row];
+          if (p >= 0 && col >= hid) {  // K or V column range
+            if (p >= smax) p = smax - 1;
+            int rel = col - hid;
+            bool is_v = rel >= hid;
+            if (is_v) rel -= hid;
+            int h_idx = rel >> 6, d = rel & 63;
+            _Float16* cache = is_v ? vcache : kcache;
+            cache[(((int64_t)row * heads + h_idx) * smax + p) * 64 + d] = hv;
+          }
+        }
+      }
+    }
+  }
+}
+
+void launch_decode_gemm_fused(int pro, int epi, const void* x, const void* r,
+                              void* h_out, const float* gamma,
+                              const float* beta, const void* Bw,
+                              const float* bias, void* C, const void* ids,
+                              const void* tok, const void* posemb,
+                              const void* pos, void* kcache, void* vcache,
+                              int M, int N, int K, int heads, int smax,
+                              float eps, hipStream_t stream) {
+  if (M > 64) throw std::runtime_error("decode_gemm_fused: M > 64");
+  if (K % 64 != 0) throw std::runtime_error("decode_gemm_fused: K % 64");
+  dim3 grid((unsigned)cdiv(N, 64));
+  dim3 block(256);
+  auto L = [&](auto pro_c, auto epi_c) {
+    hipLaunchKernelGGL(
+        (decode_gemm_fused_kernel<decltype(pro_c)::value,
+                                  decltype(epi_c)::value>),
+        grid, block, 0, stream, (const _Float16*)x, (const _Float16*)r,
+        (_Float16*)h_out, gamma, beta, (const _Float16*)Bw, bias,
+        (_Float16*)C, (const int*)ids, (const _Float16*)tok,
+        (const _Float16*)posemb, (const int*)pos, (_Float16*)kcache,
+        (_Float16*)vcache, M, N, K, heads, smax, eps);
+  };
+  using I1 = std::integral_constant<int, 1>;
+  using I2 = std::integral_constant<int, 2>;
+  using I3 = std::integral_constant<int, 3>;
+  using E0 = std::integral_constant<int, 0>;
+  using E1 = std::integral_constant<int, 1>;
+  using E2 = std::integral_constant<int, 2>;
+  if (pro == 1 && epi == 0) L(I1{}, E0{});
+  else if (pro == 1 && epi == 1) L(I1{}, E1{});
+  else if (pro == 1 && epi == 2) L(I1{}, E2{});
+  else if (pro == 2 && epi == 0) L(I2{}, E0{});
+  else if (pro == 2 && epi == 1) L(I2{}, E1{});
+  else if (pro == 2 && epi == 2) L(I2{}, E2{});
+  else if (pro == 3 && epi == 2) L(I3{}, E2{});
+  else if (pro == 3 && epi == 0) L(I3{}, E0{});
+  else throw std::runtime_error("decode_gemm_fused: bad pro/epi combo");
+}
+
 }  // namespace trtlab

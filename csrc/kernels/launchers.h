@@ -120,6 +120,16 @@ void launch_chunk_embed(const void* ids, const void* tok, const void* posemb,
                         void* out, const void* pos, int B, int K, int smax,
                         int hidden, hipStream_t stream);
 void launch_advance_pos(void* pos, int B, int smax, hipStream_t stream);
+// fused decode GEMM: prologue 1=LN 2=ADD_LN(+h_out) 3=EMBED_LN; epilogue
+// 0=bias 1=bias+gelu 2=bias+KV-scatter (see decode.hip)
+void launch_decode_gemm_fused(int pro, int epi, const void* x, const void* r,
+                              void* h_out, const float* gamma,
+                              const float* beta, const void* Bw,
+                              const float* bias, void* C, const void* ids,
+                              const void* tok, const void* posemb,
+                              const void* pos, void* kcache, void* vcache,
+                              int M, int N, int K, int heads, int smax,
+                              float eps, hipStream_t stream);
 
 void launch_embedding(int dtype, const void* ids, const void* tok,
                       const void* pos, const void* seg, const void* segids,
