@@ -445,7 +445,8 @@ __global__ __launch_bounds__(256) void decode_gemm_fused_kernel(
   // stage A' (normalized) + B for tile t into slot
   auto stage = [&](int t, int slot) {
     uint32_t base = lds0 + slot * kBuf;
-    // B via async glds (weights)
+    char* abase = smem + slot * kBuf;  // generic pointer for ds_writes
+    // B via async glds (weights; the 32-bit lds0 form is glds-only)
     stage_tile<_Float16, BN>(Bw + (int64_t)n0 * K + (int64_t)t * 64, K, n0,
                              N, base + kABytes, tid);
     // A' via VALU normalize + ds_write (2 chunks of 8 elems per thread)
@@ -466,8 +467,7 @@ __global__ __launch_bounds__(256) void decode_gemm_fused_kernel(
         float b = beta ? beta[kk + j] : 0.0f;
         o8[j] = (_Float16)(((float)v8[j] - mean) * inv * g + b);
       }
-      *(short8v*)((char*)(uintptr_t)(base + row * 128 +
-                                     (cb ^ ((row & 7) << 4)))) =
+      *(short8v*)(abase + row * 128 + (cb ^ ((row & 7) << 4))) =
           *(const short8v*)o8;
     }
   };
