@@ -33,12 +33,19 @@ class DecodeSession:
     def __init__(self, graph, batch: int, smax: int = 1024, device: int = 0,
                  capture: bool = True, lm_head: bool = False,
                  fused: bool = False):
-        """fused=True (needs batch <= 64): horizontal kernel fusion for
-        the latency-bound step — LN / residual-add / embed prologues and
-        KV-scatter / GeLU epilogues fold into the small-M GEMMs
-        (csrc decode_gemm_fused), cutting ~8 kernels/layer to 5. The
-        residual stream ping-pongs between two buffers (the fused ADD_LN
-        writes the NEW stream while other blocks still read the old)."""
+        """fused=True (needs batch <= 64): EXPERIMENTAL horizontal kernel
+        fusion for the latency-bound step — LN / residual-add / embed
+        prologues and KV-scatter / GeLU epilogues fold into the small-M
+        GEMMs (csrc decode_gemm_fused), cutting ~8 kernels/layer to 5.
+        MEASURED on MI355X (profiles/README r2): the fused GEMMs lose
+        split-K + the deep staging pipeline and only fill cdiv(N,64)
+        workgroups, so the step is currently ~2x SLOWER than the unfused
+        autotuned path (0.86 -> 1.63 ms b8) despite 40% fewer kernels —
+        default OFF; numerics are verified (test_fused_decode_matches_
+        unfused). Making the fused GEMM pipeline-deep is the round-3
+        follow-up. The residual stream ping-pongs between two buffers
+        (the fused ADD_LN writes the NEW stream while other blocks still
+        read the old)."""
         import torch
 
         from trtlab_amd import native
