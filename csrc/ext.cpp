@@ -5,6 +5,9 @@
 #include <pybind11/pybind11.h>
 #include <pybind11/stl.h>
 
+#include <chrono>
+#include <thread>
+
 #include "kernels/launchers.h"
 #include "runtime/comm.h"
 #include "runtime/runtime.h"
@@ -212,6 +215,39 @@ PYBIND11_MODULE(_C, m) {
         d["histogram"] = h;
         return d;
       });
+
+  // hybrid futex mutex/cv stress (host-only test hook): N threads bang a
+  // shared counter under HybridMutex; returns (counter, ms). A lost
+  // update proves a broken lock, so the test asserts exactly T*I.
+  mem.def("hybrid_mutex_stress", [](int threads, int iters) {
+    HybridMutex mu;
+    HybridCondition cv;
+    long counter = 0;
+    bool go = false;
+    std::vector<std::thread> ts;
+    auto t0 = std::chrono::steady_clock::now();
+    for (int t = 0; t < threads; ++t)
+      ts.emplace_back([&] {
+        {
+          HybridLock g(mu);
+          while (!go) cv.wait(mu);
+        }
+        for (int i = 0; i < iters; ++i) {
+          HybridLock g(mu);
+          ++counter;
+        }
+      });
+    {
+      HybridLock g(mu);
+      go = true;
+      cv.notify_all();
+    }
+    for (auto& th : ts) th.join();
+    double ms = std::chrono::duration<double, std::milli>(
+                    std::chrono::steady_clock::now() - t0)
+                    .count();
+    return py::make_tuple(counter, ms);
+  }, py::arg("threads") = 8, py::arg("iters") = 100000);
 
   py::class_<BlockPool>(mem, "BlockPool")
       .def(py::init<size_t, int, int>(), py::arg("block_bytes"),

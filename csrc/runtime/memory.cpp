@@ -92,18 +92,18 @@ BlockPool::BlockPool(size_t block_bytes, int count, int device)
 BlockPool::~BlockPool() { device_free(base_, block_bytes_ * total_); }
 
 void* BlockPool::acquire() {
-  std::lock_guard<std::mutex> g(mu_);
+  HybridLock g(mu_);
   if (free_.empty()) return nullptr;  // caller blocks at the Python Pool level
   void* p = free_.back();
   free_.pop_back();
   return p;
 }
 void BlockPool::release(void* p) {
-  std::lock_guard<std::mutex> g(mu_);
+  HybridLock g(mu_);
   free_.push_back((char*)p);
 }
 int BlockPool::available() {
-  std::lock_guard<std::mutex> g(mu_);
+  HybridLock g(mu_);
   return (int)free_.size();
 }
 

@@ -1,6 +1,7 @@
 // trtlab_amd — native runtime: memory primitives + graph-captured executor.
 #pragma once
 #include "../common.h"
+#include "hybrid.h"
 
 #include <array>
 #include <atomic>
@@ -39,7 +40,10 @@ class BlockPool {
   char* base_;
   int total_;
   std::vector<char*> free_;
-  std::mutex mu_;
+  // hybrid spin-then-futex lock (reference hybrid_mutex role): the pool's
+  // acquire/release critical sections are a few instructions — spinning
+  // keeps the hot dispatch path out of the kernel
+  HybridMutex mu_;
 };
 
 // Growing best-fit device allocator for dynamic workloads (multi-model

@@ -166,3 +166,15 @@ def test_tracked_allocator_composes():
     assert a.tracker.total_allocs == 4
     assert a.tracker.in_use == 0
     assert a.tracker.high_water == (1 + 2 + 4 + 8)
+
+
+def test_hybrid_futex_mutex_stress():
+    """Hybrid spin-then-futex mutex + condvar (reference hybrid_mutex.h
+    role): 8 threads x 200k guarded increments must lose no updates, and
+    the condvar start-gate must release every thread."""
+    import trtlab_amd
+
+    C = trtlab_amd.native()
+    n, ms = C.memory.hybrid_mutex_stress(8, 200000)
+    assert n == 8 * 200000
+    assert ms < 60000
