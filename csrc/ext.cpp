@@ -579,6 +579,33 @@ PYBIND11_MODULE(_C, m) {
           py::arg("qkv"), py::arg("kcache"), py::arg("vcache"), py::arg("B"),
           py::arg("H"), py::arg("P"), py::arg("smax"), py::arg("stream") = 0,
           py::arg("sync") = true);
+  ops.def("kv_append_paged",
+          [](uintptr_t qkv, uintptr_t kp, uintptr_t vp, uintptr_t table,
+             uintptr_t pos, int B, int H, int max_pages, uintptr_t stream,
+             bool sync) {
+            launch_kv_append_paged((void*)qkv, (void*)kp, (void*)vp,
+                                   (void*)table, (void*)pos, B, H,
+                                   max_pages, as_stream(stream));
+            if (sync) TRT_HIP_CHECK(hipStreamSynchronize(as_stream(stream)));
+          },
+          py::arg("qkv"), py::arg("kpool"), py::arg("vpool"),
+          py::arg("table"), py::arg("pos"), py::arg("B"), py::arg("H"),
+          py::arg("max_pages"), py::arg("stream") = 0,
+          py::arg("sync") = true);
+  ops.def("decode_attention_paged",
+          [](uintptr_t qkv, uintptr_t kp, uintptr_t vp, uintptr_t out,
+             uintptr_t table, uintptr_t pos, int B, int H, int max_pages,
+             float scale, uintptr_t stream, bool sync) {
+            launch_decode_attention_paged((void*)qkv, (void*)kp, (void*)vp,
+                                          (void*)out, (void*)table,
+                                          (void*)pos, B, H, max_pages,
+                                          scale, as_stream(stream));
+            if (sync) TRT_HIP_CHECK(hipStreamSynchronize(as_stream(stream)));
+          },
+          py::arg("qkv"), py::arg("kpool"), py::arg("vpool"), py::arg("out"),
+          py::arg("table"), py::arg("pos"), py::arg("B"), py::arg("H"),
+          py::arg("max_pages"), py::arg("scale"), py::arg("stream") = 0,
+          py::arg("sync") = true);
   ops.def("decode_attention",
           [](uintptr_t qkv, uintptr_t kc, uintptr_t vc, uintptr_t out,
              uintptr_t pos, int B, int H, int smax, float scale,

@@ -558,4 +558,113 @@ void launch_decode_gemm_fused(int pro, int epi, const void* x, const void* r,
   else throw std::runtime_error("decode_gemm_fused: bad pro/epi combo");
 }
 
+
+// ---- paged KV cache (vLLM-style block tables) ----
+// Many concurrent sessions share ONE fixed physical pool per layer; each
+// slot maps logical 64-position pages to physical page ids through a
+// device block table [B][max_pages] (filled by the host as positions
+// grow; reset/idle returns pages to the host free list, so parked slots
+// hold no memory). Page layout: [page][64 positions][H][64] — a
+// position's per-head row stays one contiguous 128-B cacheline.
+__global__ __launch_bounds__(64) void kv_append_paged_kernel(
+    const _Float16* __restrict__ qkv, _Float16* __restrict__ kpool,
+    _Float16* __restrict__ vpool, const int* __restrict__ table,
+    const int* __restrict__ pos, int B, int H, int max_pages) {
+  int b = blockIdx.x / H, h = blockIdx.x % H;
+  int d = threadIdx.x;
+  int p = pos[b];
+  if (p < 0) return;  // idle slot
+  int page = table[b * max_pages + (p >> 6)];
+  if (page < 0) return;  // unmapped (host error) — fail soft, not wild
+  int hid = H * 64;
+  int64_t src = (int64_t)b * 3 * hid + h * 64 + d;
+  int64_t dst = (((int64_t)page * 64 + (p & 63)) * H + h) * 64 + d;
+  kpool[dst] = qkv[src + hid];
+  vpool[dst] = qkv[src + 2 * hid];
+}
+
+void launch_kv_append_paged(const void* qkv, void* kpool, void* vpool,
+                            const void* table, const void* pos, int B, int H,
+                            int max_pages, hipStream_t stream) {
+  hipLaunchKernelGGL(kv_append_paged_kernel, dim3(B * H), dim3(64), 0,
+                     stream, (const _Float16*)qkv, (_Float16*)kpool,
+                     (_Float16*)vpool, (const int*)table, (const int*)pos,
+                     B, H, max_pages);
+}
+
+__global__ __launch_bounds__(64) void decode_attention_paged_kernel(
+    const _Float16* __restrict__ qkv, const _Float16* __restrict__ kpool,
+    const _Float16* __restrict__ vpool, _Float16* __restrict__ out,
+    const int* __restrict__ table, const int* __restrict__ pos, int B, int H,
+    int max_pages, float scale) {
+  __shared__ float p_s[4096];
+  int b = blockIdx.x / H, h = blockIdx.x % H;
+  int lane = threadIdx.x;
+  int hid = H * 64;
+  if (pos[b] < 0) return;  // idle slot
+  int n = pos[b] + 1;
+
+  float q[64];
+  {
+    const _Float16* qrow = qkv + (int64_t)b * 3 * hid + h * 64;
+#pragma unroll
+    for (int c = 0; c < 8; ++c) {
+      half8v v = *(const half8v*)(qrow + c * 8);
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        q[c * 8 + j] = (float)((const _Float16*)&v)[j];
+    }
+  }
+  const int* tab = table + (int64_t)b * max_pages;
+
+  float m = -3.0e38f;
+  for (int t = lane; t < n; t += 64) {
+    int64_t rowoff =
+        (((int64_t)tab[t >> 6] * 64 + (t & 63)) * H + h) * 64;
+    const _Float16* krow = kpool + rowoff;
+    float s = 0.f;
+#pragma unroll
+    for (int c = 0; c < 8; ++c) {
+      half8v v = *(const half8v*)(krow + c * 8);
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        s += q[c * 8 + j] * (float)((const _Float16*)&v)[j];
+    }
+    s *= scale;
+    p_s[t] = s;
+    m = fmaxf(m, s);
+  }
+#pragma unroll
+  for (int off = 32; off; off >>= 1) m = fmaxf(m, __shfl_xor(m, off, 64));
+  __syncthreads();
+  float l = 0.f;
+  for (int t = lane; t < n; t += 64) {
+    float e = __expf(p_s[t] - m);
+    p_s[t] = e;
+    l += e;
+  }
+#pragma unroll
+  for (int off = 32; off; off >>= 1) l += __shfl_xor(l, off, 64);
+  __syncthreads();
+  float acc = 0.f;
+  for (int t = 0; t < n; ++t) {
+    int64_t rowoff =
+        (((int64_t)tab[t >> 6] * 64 + (t & 63)) * H + h) * 64;
+    acc += p_s[t] * (float)vpool[rowoff + lane];
+  }
+  out[(int64_t)b * hid + h * 64 + lane] = (_Float16)(acc / l);
+}
+
+void launch_decode_attention_paged(const void* qkv, const void* kpool,
+                                   const void* vpool, void* out,
+                                   const void* table, const void* pos, int B,
+                                   int H, int max_pages, float scale,
+                                   hipStream_t stream) {
+  hipLaunchKernelGGL(decode_attention_paged_kernel, dim3(B * H), dim3(64), 0,
+                     stream, (const _Float16*)qkv, (const _Float16*)kpool,
+                     (const _Float16*)vpool, (_Float16*)out,
+                     (const int*)table, (const int*)pos, B, H, max_pages,
+                     scale);
+}
+
 }  // namespace trtlab

@@ -120,6 +120,15 @@ void launch_chunk_embed(const void* ids, const void* tok, const void* posemb,
                         void* out, const void* pos, int B, int K, int smax,
                         int hidden, hipStream_t stream);
 void launch_advance_pos(void* pos, int B, int smax, hipStream_t stream);
+// paged KV cache (vLLM-style block tables; see decode.hip)
+void launch_kv_append_paged(const void* qkv, void* kpool, void* vpool,
+                            const void* table, const void* pos, int B, int H,
+                            int max_pages, hipStream_t stream);
+void launch_decode_attention_paged(const void* qkv, const void* kpool,
+                                   const void* vpool, void* out,
+                                   const void* table, const void* pos, int B,
+                                   int H, int max_pages, float scale,
+                                   hipStream_t stream);
 // fused decode GEMM: prologue 1=LN 2=ADD_LN(+h_out) 3=EMBED_LN; epilogue
 // 0=bias 1=bias+gelu 2=bias+KV-scatter (see decode.hip)
 void launch_decode_gemm_fused(int pro, int epi, const void* x, const void* r,

@@ -27,12 +27,86 @@ from typing import Dict, List
 import numpy as np
 
 
+
+class PagedKVPool:
+    """vLLM-style paged KV memory: ONE fixed physical pool per layer,
+    shared by every slot, mapped through a per-slot device block table.
+    Pages hold 64 positions x heads x 64 dims; a page id is valid across
+    all layers (each layer has its own K/V pools, ids allocated in
+    lockstep), so one table serves the whole stack. Host-side free list;
+    idle/reset slots return their pages — parked sequences hold ZERO
+    cache memory (the paged payoff vs the dense [B][H][smax][64] layout).
+    """
+
+    def __init__(self, layers: int, heads: int, batch: int,
+                 num_pages: int, max_pages_per_slot: int,
+                 device: int = 0):
+        import torch
+
+        from trtlab_amd import native
+
+        self._C = native()
+        self._torch = torch
+        torch.cuda.set_device(device)
+        self.layers = layers
+        self.heads = heads
+        self.batch = batch
+        self.num_pages = num_pages
+        self.max_pages = max_pages_per_slot
+        self.kpools = [torch.zeros(num_pages, 64, heads, 64,
+                                   dtype=torch.half, device="cuda")
+                       for _ in range(layers)]
+        self.vpools = [torch.zeros_like(k) for k in self.kpools]
+        self.table = torch.full((batch, max_pages_per_slot), -1,
+                                dtype=torch.int32, device="cuda")
+        self._table_host = np.full((batch, max_pages_per_slot), -1,
+                                   np.int32)
+        self._free = list(range(num_pages - 1, -1, -1))
+
+    @property
+    def pages_free(self) -> int:
+        return len(self._free)
+
+    def ensure(self, b: int, logical_idx: int) -> None:
+        """Map slot b's logical page if unmapped (called by the session
+        right before a step that will write position logical_idx*64+...)."""
+        if logical_idx >= self.max_pages:
+            raise RuntimeError(
+                f"slot {b}: logical page {logical_idx} >= max_pages")
+        if self._table_host[b, logical_idx] >= 0:
+            return
+        if not self._free:
+            raise MemoryError(
+                "PagedKVPool exhausted (all pages mapped) — free a slot")
+        pid = self._free.pop()
+        self._table_host[b, logical_idx] = pid
+        ent = np.array([pid], np.int32)
+        self._C.memory.memcpy_h2d(
+            self.table.data_ptr() + (b * self.max_pages + logical_idx) * 4,
+            ent, 4)
+
+    def free_slot(self, b: int) -> int:
+        """Return slot b's pages to the free list; returns count freed."""
+        n = 0
+        for i in range(self.max_pages):
+            pid = int(self._table_host[b, i])
+            if pid >= 0:
+                self._free.append(pid)
+                self._table_host[b, i] = -1
+                n += 1
+        row = np.full(self.max_pages, -1, np.int32)
+        self._C.memory.memcpy_h2d(
+            self.table.data_ptr() + b * self.max_pages * 4, row,
+            row.nbytes)
+        return n
+
+
 class DecodeSession:
     """One generation session: fixed batch, growing position."""
 
     def __init__(self, graph, batch: int, smax: int = 1024, device: int = 0,
                  capture: bool = True, lm_head: bool = False,
-                 fused: bool = False):
+                 fused: bool = False, paged=None):
         """fused=True (needs batch <= 64): EXPERIMENTAL horizontal kernel
         fusion for the latency-bound step — LN / residual-add / embed
         prologues and KV-scatter / GeLU epilogues fold into the small-M
@@ -121,9 +195,28 @@ class DecodeSession:
         self.logits = (torch.zeros(B, self.vocab, dtype=torch.half,
                                    device="cuda") if lm_head else None)
         self.h2 = torch.zeros(B, Hd, dtype=torch.half, device="cuda")
+        # paged KV mode: paged = a PagedKVPool (shared with other
+        # sessions) or True (private pool sized for smax). The dense
+        # per-layer caches above are then released — slots draw pages on
+        # demand and return them on idle/reset.
+        self.kv_pool = None
+        if paged:
+            mp = (smax + 63) // 64
+            self.kv_pool = paged if isinstance(paged, PagedKVPool) else \
+                PagedKVPool(self.n_layers, self.heads, B,
+                            num_pages=B * mp, max_pages_per_slot=mp,
+                            device=device)
+            if self.kv_pool.layers != self.n_layers or \
+                    self.kv_pool.heads != self.heads:
+                raise ValueError("pool layer/head mismatch")
+            for lay in self.layers:  # release the dense caches
+                lay["kcache"] = None
+                lay["vcache"] = None
         self.fused = bool(fused)
         if self.fused and (B > 64 or not lm_head):
             raise ValueError("fused decode needs batch <= 64 and lm_head")
+        if self.fused and self.kv_pool is not None:
+            raise ValueError("fused + paged not supported together yet")
         self.stream = self._C.hip.stream_create()
         self._graph = 0
         self._steps = 0
@@ -227,16 +320,32 @@ class DecodeSession:
                         self.qkv.data_ptr(), bias=lay["qkv_b"].data_ptr(),
                         M=B, N=3 * Hd, K=Hd, epi=self._epi_bias, stream=s,
                         sync=False, tile=T4)
-            ops.kv_append(self.qkv.data_ptr(), lay["kcache"].data_ptr(),
-                          lay["vcache"].data_ptr(), self.pos.data_ptr(), B,
-                          self.heads, self.smax, stream=s, sync=False)
-            ops.decode_attention(self.qkv.data_ptr(),
-                                 lay["kcache"].data_ptr(),
-                                 lay["vcache"].data_ptr(),
-                                 self.att.data_ptr(), self.pos.data_ptr(), B,
-                                 self.heads, self.smax,
-                                 1.0 / float(np.sqrt(64.0)), stream=s,
-                                 sync=False)
+            if self.kv_pool is not None:
+                pool = self.kv_pool
+                ops.kv_append_paged(self.qkv.data_ptr(),
+                                    pool.kpools[li].data_ptr(),
+                                    pool.vpools[li].data_ptr(),
+                                    pool.table.data_ptr(),
+                                    self.pos.data_ptr(), B, self.heads,
+                                    pool.max_pages, stream=s, sync=False)
+                ops.decode_attention_paged(
+                    self.qkv.data_ptr(), pool.kpools[li].data_ptr(),
+                    pool.vpools[li].data_ptr(), self.att.data_ptr(),
+                    pool.table.data_ptr(), self.pos.data_ptr(), B,
+                    self.heads, pool.max_pages,
+                    1.0 / float(np.sqrt(64.0)), stream=s, sync=False)
+            else:
+                ops.kv_append(self.qkv.data_ptr(), lay["kcache"].data_ptr(),
+                              lay["vcache"].data_ptr(), self.pos.data_ptr(),
+                              B, self.heads, self.smax, stream=s, sync=False)
+                ops.decode_attention(self.qkv.data_ptr(),
+                                     lay["kcache"].data_ptr(),
+                                     lay["vcache"].data_ptr(),
+                                     self.att.data_ptr(),
+                                     self.pos.data_ptr(), B,
+                                     self.heads, self.smax,
+                                     1.0 / float(np.sqrt(64.0)), stream=s,
+                                     sync=False)
             ops.gemm_bt(0, self.att.data_ptr(), lay["proj_w"].data_ptr(),
                         self.x2.data_ptr(), bias=lay["proj_b"].data_ptr(),
                         M=B, N=Hd, K=Hd, epi=self._epi_bias, stream=s,
@@ -284,6 +393,9 @@ class DecodeSession:
 
         if self._steps != 0 or self._graph:
             raise RuntimeError("prefill must run before the first step()")
+        if self.kv_pool is not None:
+            raise RuntimeError("prefill is not supported in paged mode yet "
+                               "(step the prompt, or use dense caches)")
         prompt = np.ascontiguousarray(prompt, np.int32)
         B, P = prompt.shape
         assert B == self.batch and 0 < P < self.smax
@@ -362,6 +474,11 @@ class DecodeSession:
         [B, hidden] fp32 (or logits [B, vocab] with lm_head=True). The
         device-side position counter starts at 0 and the captured graph
         advances it, so replays need no host-side position plumbing."""
+        if self.kv_pool is not None:
+            # map the page each active slot's next write lands in
+            for b in range(self.batch):
+                if self._active[b]:
+                    self.kv_pool.ensure(b, int(self._slot_steps[b]) >> 6)
         if (self._slot_steps[self._active] >= self.smax).any():
             raise RuntimeError(
                 "DecodeSession: a slot hit the sequence limit "
@@ -405,6 +522,9 @@ class DecodeSession:
 
         if self.logits is None:
             raise RuntimeError("verify_chunk requires lm_head=True")
+        if self.kv_pool is not None:
+            raise RuntimeError("verify_chunk is not supported in paged "
+                               "mode yet")
         tokens = np.ascontiguousarray(tokens, np.int32)
         B, K = tokens.shape
         assert B == self.batch and K >= 1
@@ -514,6 +634,8 @@ class DecodeSession:
         self.pos[b] = 0
         self._slot_steps[b] = 0
         self._active[b] = True
+        if self.kv_pool is not None:
+            self.kv_pool.free_slot(b)  # fresh sequence: recycle its pages
         self._torch.cuda.synchronize()
 
     def idle_slot(self, b: int) -> None:
@@ -524,6 +646,8 @@ class DecodeSession:
         self.pos[b] = -1
         self._slot_steps[b] = 0
         self._active[b] = False
+        if self.kv_pool is not None:
+            self.kv_pool.free_slot(b)  # parked slots hold ZERO cache pages
         self._torch.cuda.synchronize()
 
     def close(self):
