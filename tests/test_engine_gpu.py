@@ -1097,20 +1097,13 @@ def test_paged_kv_matches_dense_and_recycles_pages():
     # parked slot returns its pages; the other slots keep decoding
     paged.idle_slot(2)
     assert total - pool.pages_free == (B - 1) * 2
-    nxt = paged.step(toks[0])
-    np.testing.assert_allclose(nxt[0], dense_step_ref(g, B, toks, 71)[0],
-                               rtol=1, atol=1e9)  # shape/finite sanity only
+    nxt = paged.step(toks[0])  # remaining slots keep decoding
     assert np.isfinite(nxt).all()
     # reset: fresh sequence reuses recycled pages from position 0
     paged.reset_slot(2)
     out = paged.step(toks[1])
     assert np.isfinite(out[2]).all()
     paged.close()
-
-
-def dense_step_ref(g, B, toks, n):
-    # placeholder helper for the sanity branch above
-    return np.zeros((B, 1), np.float32)
 
 
 def test_paged_pool_shared_across_sessions_and_exhaustion():
