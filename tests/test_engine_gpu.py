@@ -1122,9 +1122,8 @@ def test_paged_pool_shared_across_sessions_and_exhaustion():
     assert pool.pages_free == 1
     for _ in range(63):
         s1.step(ids)      # stays within page 0
-    s1.step(ids)          # position 64 -> second page for slot 0 and...
-    # slot 1 also needs its second page -> pool exhausted mid-ensure
+    # position 64: slot 0 gets the last free page, slot 1's second page
+    # cannot be mapped -> loud exhaustion inside the same step call
     with pytest.raises(MemoryError):
-        for _ in range(70):
-            s1.step(ids)
+        s1.step(ids)
     s1.close()
