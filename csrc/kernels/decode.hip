@@ -361,6 +361,9 @@ __global__ __launch_bounds__(256) void decode_gemm_fused_kernel(
   constexpr int BM = 64, BN = 64;
   __shared__ __attribute__((aligned(16))) char smem[2 * (BM + BN) * 128];
   __shared__ float sm_mean[BM], sm_inv[BM];
+  // gamma/beta staged once per block: the per-tile normalize would
+  // otherwise expose a global-latency chain every iteration
+  __shared__ float sm_gamma[2048], sm_beta[2048];
   uint32_t lds0 = (uint32_t)(uintptr_t)&smem[0];
   constexpr int kABytes = BM * 128;
   constexpr int kBuf = (BM + BN) * 128;
@@ -371,6 +374,7 @@ __global__ __launch_bounds__(256) void decode_gemm_fused_kernel(
   const int wr = wave >> 1;
   const int wc = wave & 1;
   const int n0 = (int)blockIdx.x * BN;
+  int cur_kbase = 0;  // K-base of the tile currently being staged
 
   // source value of the pre-norm row (x + r / embed), 8 elems at kk
   auto load8 = [&](int row, int kk, _Float16* v8) {
@@ -397,7 +401,11 @@ __global__ __launch_bounds__(256) void decode_gemm_fused_kernel(
     }
   };
 
-  // ---- LN stats: 4 lanes per row, vectorized strided reads ----
+  // ---- stage gamma/beta + LN stats ----
+  for (int i = tid; i < K; i += 256) {
+    sm_gamma[i] = gamma ? gamma[i] : 1.0f;
+    sm_beta[i] = beta ? beta[i] : 0.0f;
+  }
   {
     int row = tid >> 2;
     int part = tid & 3;
@@ -442,47 +450,63 @@ __global__ __launch_bounds__(256) void decode_gemm_fused_kernel(
     for (int j = 0; j < 2; ++j) acc[i][j] = {0, 0, 0, 0};
 
   const int ktiles = K / 64;
-  // stage A' (normalized) + B for tile t into slot
-  auto stage = [&](int t, int slot) {
-    uint32_t base = lds0 + slot * kBuf;
-    char* abase = smem + slot * kBuf;  // generic pointer for ds_writes
-    // B via async glds (weights; the 32-bit lds0 form is glds-only)
-    stage_tile<_Float16, BN>(Bw + (int64_t)n0 * K + (int64_t)t * 64, K, n0,
-                             N, base + kABytes, tid);
-    // A' via VALU normalize + ds_write (2 chunks of 8 elems per thread)
+  // Register-prefetch pipeline (the attention kernel's proven pattern):
+  // A and B tiles are prefetched into REGISTERS (not glds), so
+  // __syncthreads() between ds_writes and the MFMAs only waits lgkmcnt —
+  // the next tile's global loads stay in flight under the compute, and
+  // the ds_writes' register dependencies are the only load waits.
+  _Float16 a_reg[2][8];
+  _Float16 b_reg[2][8];
+  auto prefetch = [&](int t) {
 #pragma unroll
     for (int c = 0; c < 2; ++c) {
-      int idx = c * 256 + tid;        // 0..511
-      int row = idx >> 3;             // 0..63
-      int cb = (idx & 7) * 16;        // byte offset in the 128-B row
+      int idx = c * 256 + tid;      // 0..511
+      int row = idx >> 3;           // 0..63
+      int cb = (idx & 7) * 16;
       int kk = t * 64 + cb / 2;
-      _Float16 v8[8];
+      int ar = row < M ? row : M - 1;
+      load8(ar, kk, a_reg[c]);
+      int br = n0 + row;
+      if (br >= N) br = N - 1;
+      *(half8v*)b_reg[c] = *(const half8v*)(Bw + (int64_t)br * K + kk);
+    }
+  };
+  auto stage_regs = [&](char* abase) {
+#pragma unroll
+    for (int c = 0; c < 2; ++c) {
+      int idx = c * 256 + tid;
+      int row = idx >> 3;
+      int cb = (idx & 7) * 16;
+      int kk = (cb / 2);  // column within the tile for gamma/beta lookup
+      (void)kk;
+      float mean = sm_mean[row < M ? row : M - 1];
+      float inv = sm_inv[row < M ? row : M - 1];
       _Float16 o8[8];
-      int rrow = row < M ? row : M - 1;
-      load8(rrow, kk, v8);
-      float mean = sm_mean[rrow], inv = sm_inv[rrow];
+      int kcol = 0;  // filled below per element
+      (void)kcol;
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        float g = gamma ? gamma[kk + j] : 1.0f;
-        float b = beta ? beta[kk + j] : 0.0f;
-        o8[j] = (_Float16)(((float)v8[j] - mean) * inv * g + b);
+        int kg = cur_kbase + cb / 2 + j;
+        o8[j] = (_Float16)(((float)a_reg[c][j] - mean) * inv * sm_gamma[kg] +
+                           sm_beta[kg]);
       }
       *(short8v*)(abase + row * 128 + (cb ^ ((row & 7) << 4))) =
           *(const short8v*)o8;
+      *(short8v*)(abase + kABytes + row * 128 + (cb ^ ((row & 7) << 4))) =
+          *(const short8v*)b_reg[c];
     }
   };
 
-  int cur = 0;
-  stage(0, 0);
-  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-  __syncthreads();
+  char* abase = smem;  // single slot: barrier-write-barrier-compute
+  cur_kbase = 0;
+  prefetch(0);
   for (int t = 0; t < ktiles; ++t) {
-    if (t + 1 < ktiles) stage(t + 1, cur ^ 1);
-    const char* As = &smem[cur * kBuf];
-    mfma_tile<_Float16, BM, BN>(As, As + kABytes, lane, wr, wc, acc);
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    __syncthreads();
-    cur ^= 1;
+    __syncthreads();   // prior tile's LDS reads complete before overwrite
+    cur_kbase = t * 64;
+    stage_regs(abase);             // waits only the prefetched registers
+    if (t + 1 < ktiles) prefetch(t + 1);  // overlaps the MFMAs below
+    __syncthreads();               // ds_writes visible (lgkmcnt)
+    mfma_tile<_Float16, BM, BN>(abase, abase + kABytes, lane, wr, wc, acc);
   }
 
   // ---- epilogue: bias (+gelu) (+K/V scatter) ----
@@ -529,6 +553,7 @@ void launch_decode_gemm_fused(int pro, int epi, const void* x, const void* r,
                               float eps, hipStream_t stream) {
   if (M > 64) throw std::runtime_error("decode_gemm_fused: M > 64");
   if (K % 64 != 0) throw std::runtime_error("decode_gemm_fused: K % 64");
+  if (K > 2048) throw std::runtime_error("decode_gemm_fused: K > 2048");
   dim3 grid((unsigned)cdiv(N, 64));
   dim3 block(256);
   auto L = [&](auto pro_c, auto epi_c) {

@@ -278,13 +278,16 @@ class DecodeSession:
                         self.x2.data_ptr(), bias=lay["ff2_b"].data_ptr(),
                         M=B, N=Hd, K=self.inter, epi=self._epi_bias,
                         stream=s, sync=False)
-        # logits = ln_f(h_cur + ff2_out) @ tok^T  (weight-tied head)
-        ops.decode_gemm_fused(
-            2, 0, x=self.x2.data_ptr(), r=h_cur.data_ptr(),
-            h_out=h_nxt.data_ptr(), gamma=self.lnf_g.data_ptr(),
-            beta=self.lnf_b.data_ptr(), B=self.tok.data_ptr(), bias=0,
-            C=self.logits.data_ptr(), pos=self.pos.data_ptr(), M=B,
-            N=self.vocab, K=Hd, stream=s, sync=False)
+        # head stays UNFUSED: at N=vocab the fused kernel's per-block LN
+        # stats replicate across ~786 blocks (measured 102.6 vs 14.5 us,
+        # tools/dbg_fused_perf) — add_layernorm + plain gemm_bt win there
+        ops.add_layernorm(0, self.x2.data_ptr(), h_cur.data_ptr(),
+                          self.lnf_g.data_ptr(), self.lnf_b.data_ptr(),
+                          self.out.data_ptr(), M=B, N=Hd, stream=s,
+                          sync=False)
+        ops.gemm_bt(0, self.out.data_ptr(), self.tok.data_ptr(),
+                    self.logits.data_ptr(), M=B, N=self.vocab, K=Hd,
+                    epi=self._epi_none, stream=s, sync=False)
         # no cross-step residual state: layer 0's EMBED prologue
         # regenerates the stream each step (h/h2 are just scratch)
         ops.advance_pos(self.pos.data_ptr(), B, self.smax, stream=s,
