@@ -37,6 +37,7 @@ OpDesc op_from_dict(const py::dict& d) {
   o.in2_off = gl("in2_off", -1);
   o.out_off = gl("out_off", -1);
   o.out2_off = gl("out2_off", -1);
+  o.out3_off = gl("out3_off", -1);
   o.w_off = gl("w_off", -1);
   o.w2_off = gl("w2_off", -1);
   o.scale_off = gl("scale_off", -1);

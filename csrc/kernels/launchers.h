@@ -36,15 +36,20 @@ void launch_softmax_rows(int dtype, const void* in, void* out, int M, int N,
                          int64_t ld, hipStream_t stream);
 // q_out (optional): fused fp8-e4m3 quantized copy of the output row at
 // scale q_scale (producer-side quantization for fp8 projections).
+// mx_codes/mx_scales + mx_mode (4 = MXFP4, 8 = MXFP8): producer-fused
+// OCP MX quantization of the normalized row (codes + e8m0 block scales),
+// layouts identical to quantize_mxfp4/8.
 void launch_layernorm(int dtype, const void* in, const float* gamma,
                       const float* beta, void* out, int M, int N, int64_t ld,
                       float eps, hipStream_t stream, void* q_out = nullptr,
-                      float q_scale = 0.f);
+                      float q_scale = 0.f, void* mx_codes = nullptr,
+                      void* mx_scales = nullptr, int mx_mode = 0);
 void launch_add_layernorm(int dtype, const void* x, const void* res,
                           const float* gamma, const float* beta, void* out,
                           void* sum_out, int M, int N, int64_t ld, float eps,
                           hipStream_t stream, void* q_out = nullptr,
-                          float q_scale = 0.f);
+                          float q_scale = 0.f, void* mx_codes = nullptr,
+                          void* mx_scales = nullptr, int mx_mode = 0);
 
 void launch_elementwise(int dtype, int op, const void* a, const void* b,
                         void* out, int64_t n, hipStream_t stream);

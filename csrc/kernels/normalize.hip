@@ -86,15 +86,73 @@ __device__ __forceinline__ void store8_fp8(unsigned char* p,
   *(uint2*)p = *(const uint2*)out;
 }
 
+// Producer-fused OCP MX quantization of one normalized 8-elem chunk:
+// the 32-element MX block spans the QUAD of adjacent lanes owning chunks
+// 4q..4q+3, so the block amax is a 2-step shfl_xor quad-reduce. Code
+// packing and e8m0 scale bias match quantize_mxfp4/8_kernel exactly
+// (gemm_mx.hip) so the scaled-MFMA GEMMs consume either producer.
+__device__ __forceinline__ void store8_mx(
+    unsigned char* __restrict__ codes, unsigned char* __restrict__ scales,
+    int chunk, const float* v8, int mx_mode /*4 or 8*/, int lane) {
+  float amax = 0.f;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) amax = fmaxf(amax, fabsf(v8[j]));
+  amax = fmaxf(amax, __shfl_xor(amax, 1, 64));
+  amax = fmaxf(amax, __shfl_xor(amax, 2, 64));
+  int bias = (mx_mode == 4) ? 2 : 8;  // e2m1 emax 2, e4m3 emax 8
+  int e = (amax > 0.f) ? (int)floorf(log2f(amax)) - bias : 0;
+  e = e < -127 ? -127 : (e > 127 ? 127 : e);
+  if ((lane & 3) == 0) scales[chunk >> 2] = (unsigned char)(e + 127);
+  float inv = exp2f((float)-e);
+  if (mx_mode == 8) {
+    unsigned char out[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      __hip_fp8_e4m3 q = __hip_fp8_e4m3(
+          fminf(fmaxf(v8[j] * inv, -448.f), 448.f));
+      out[j] = *(const unsigned char*)&q;
+    }
+    *(float2v*)(codes + chunk * 8) = *(const float2v*)out;
+  } else {
+    unsigned char out[4];
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      unsigned char byte = 0;
+#pragma unroll
+      for (int h = 0; h < 2; ++h) {
+        float q = v8[2 * j + h] * inv;
+        float a = fabsf(q);
+        int m;
+        if (a < 0.25f) m = 0;
+        else if (a < 0.75f) m = 1;
+        else if (a < 1.25f) m = 2;
+        else if (a < 1.75f) m = 3;
+        else if (a < 2.5f) m = 4;
+        else if (a < 3.5f) m = 5;
+        else if (a < 5.0f) m = 6;
+        else m = 7;
+        unsigned char code = (unsigned char)(q < 0.f ? (m | 8) : m);
+        byte |= (unsigned char)(code << (4 * h));
+      }
+      out[j] = byte;
+    }
+    *(float*)(codes + chunk * 4) = *(const float*)out;
+  }
+}
+
 // layernorm core shared by the plain and residual-add variants. q_out, if
 // set, receives the fp8-e4m3 quantized row (value / q_scale) — the fused
-// producer-side quantization for fp8 transformer projections.
+// producer-side quantization for fp8 transformer projections. mx_codes/
+// mx_scales (mx_mode 4/8) instead emit the OCP MX row (codes + e8m0
+// per-32-block scales) — producer-fused MX quantization.
 template <typename T, bool ADD>
 __device__ __forceinline__ void layernorm_row(
     const T* __restrict__ src, const T* __restrict__ res,
     const float* __restrict__ gamma, const float* __restrict__ beta,
     T* __restrict__ dst, T* __restrict__ sum_out, int N, float eps, int lane,
-    unsigned char* __restrict__ q_out = nullptr, float q_inv_scale = 1.0f) {
+    unsigned char* __restrict__ q_out = nullptr, float q_inv_scale = 1.0f,
+    unsigned char* __restrict__ mx_codes = nullptr,
+    unsigned char* __restrict__ mx_scales = nullptr, int mx_mode = 0) {
   float v[kMaxChunks][8];
   int nc = 0;
   float s = 0.f;
@@ -139,6 +197,7 @@ __device__ __forceinline__ void layernorm_row(
     }
     store8(dst + c * 8, v[nc]);
     if (q_out) store8_fp8(q_out + c * 8, v[nc], q_inv_scale);
+    if (mx_codes) store8_mx(mx_codes, mx_scales, c, v[nc], mx_mode, lane);
   }
 }
 
@@ -149,14 +208,23 @@ __global__ void layernorm_kernel(const T* __restrict__ in,
                                  T* __restrict__ out,
                                  unsigned char* __restrict__ q_out, int M,
                                  int N, int64_t ld, float eps,
-                                 float q_inv_scale) {
+                                 float q_inv_scale,
+                                 unsigned char* __restrict__ mx_codes,
+                                 unsigned char* __restrict__ mx_scales,
+                                 int mx_mode) {
   int lane = threadIdx.x & 63;
   int row = (blockIdx.x * (blockDim.x >> 6)) + (threadIdx.x >> 6);
   if (row >= M) return;
+  int64_t cstride = (mx_mode == 4) ? ld / 2 : ld;
   layernorm_row<T, false>(in + (int64_t)row * ld, nullptr, gamma, beta,
                           out + (int64_t)row * ld, nullptr, N, eps, lane,
                           q_out ? q_out + (int64_t)row * ld : nullptr,
-                          q_inv_scale);
+                          q_inv_scale,
+                          mx_codes ? mx_codes + (int64_t)row * cstride
+                                   : nullptr,
+                          mx_scales ? mx_scales + (int64_t)row * (ld / 32)
+                                    : nullptr,
+                          mx_mode);
 }
 
 template <typename T>
@@ -168,16 +236,25 @@ __global__ void add_layernorm_kernel(const T* __restrict__ x,
                                      T* __restrict__ sum_out,
                                      unsigned char* __restrict__ q_out, int M,
                                      int N, int64_t ld, float eps,
-                                     float q_inv_scale) {
+                                     float q_inv_scale,
+                                     unsigned char* __restrict__ mx_codes,
+                                     unsigned char* __restrict__ mx_scales,
+                                     int mx_mode) {
   int lane = threadIdx.x & 63;
   int row = (blockIdx.x * (blockDim.x >> 6)) + (threadIdx.x >> 6);
   if (row >= M) return;
+  int64_t cstride = (mx_mode == 4) ? ld / 2 : ld;
   layernorm_row<T, true>(x + (int64_t)row * ld, res + (int64_t)row * ld,
                          gamma, beta, out + (int64_t)row * ld,
                          sum_out ? sum_out + (int64_t)row * ld : nullptr, N,
                          eps, lane,
                          q_out ? q_out + (int64_t)row * ld : nullptr,
-                         q_inv_scale);
+                         q_inv_scale,
+                         mx_codes ? mx_codes + (int64_t)row * cstride
+                                  : nullptr,
+                         mx_scales ? mx_scales + (int64_t)row * (ld / 32)
+                                   : nullptr,
+                         mx_mode);
 }
 
 static inline dim3 rows_grid(int M) { return dim3((unsigned)cdiv(M, 4)); }
@@ -197,36 +274,49 @@ void launch_softmax_rows(int dtype, const void* in, void* out, int M, int N,
 void launch_layernorm(int dtype, const void* in, const float* gamma,
                       const float* beta, void* out, int M, int N, int64_t ld,
                       float eps, hipStream_t stream, void* q_out,
-                      float q_scale) {
+                      float q_scale, void* mx_codes, void* mx_scales,
+                      int mx_mode) {
   if (N > 2048 || N % 8 != 0) throw std::runtime_error("layernorm: bad N");
+  if (mx_mode && N % 32 != 0)
+    throw std::runtime_error("layernorm mx: N % 32 != 0");
   float inv = q_scale != 0.f ? 1.0f / q_scale : 1.0f;
   if (dtype == 0)
     hipLaunchKernelGGL((layernorm_kernel<_Float16>), rows_grid(M), dim3(256), 0,
                        stream, (const _Float16*)in, gamma, beta, (_Float16*)out,
-                       (unsigned char*)q_out, M, N, ld, eps, inv);
+                       (unsigned char*)q_out, M, N, ld, eps, inv,
+                       (unsigned char*)mx_codes, (unsigned char*)mx_scales,
+                       mx_mode);
   else
     hipLaunchKernelGGL((layernorm_kernel<__bf16>), rows_grid(M), dim3(256), 0,
                        stream, (const __bf16*)in, gamma, beta, (__bf16*)out,
-                       (unsigned char*)q_out, M, N, ld, eps, inv);
+                       (unsigned char*)q_out, M, N, ld, eps, inv,
+                       (unsigned char*)mx_codes, (unsigned char*)mx_scales,
+                       mx_mode);
 }
 
 void launch_add_layernorm(int dtype, const void* x, const void* res,
                           const float* gamma, const float* beta, void* out,
                           void* sum_out, int M, int N, int64_t ld, float eps,
-                          hipStream_t stream, void* q_out, float q_scale) {
+                          hipStream_t stream, void* q_out, float q_scale,
+                          void* mx_codes, void* mx_scales, int mx_mode) {
   if (N > 2048 || N % 8 != 0) throw std::runtime_error("add_layernorm: bad N");
+  if (mx_mode && N % 32 != 0)
+    throw std::runtime_error("add_layernorm mx: N % 32 != 0");
   float inv = q_scale != 0.f ? 1.0f / q_scale : 1.0f;
   if (dtype == 0)
     hipLaunchKernelGGL((add_layernorm_kernel<_Float16>), rows_grid(M),
                        dim3(256), 0, stream, (const _Float16*)x,
                        (const _Float16*)res, gamma, beta, (_Float16*)out,
                        (_Float16*)sum_out, (unsigned char*)q_out, M, N, ld,
-                       eps, inv);
+                       eps, inv, (unsigned char*)mx_codes,
+                       (unsigned char*)mx_scales, mx_mode);
   else
     hipLaunchKernelGGL((add_layernorm_kernel<__bf16>), rows_grid(M), dim3(256),
                        0, stream, (const __bf16*)x, (const __bf16*)res, gamma,
                        beta, (__bf16*)out, (__bf16*)sum_out,
-                       (unsigned char*)q_out, M, N, ld, eps, inv);
+                       (unsigned char*)q_out, M, N, ld, eps, inv,
+                       (unsigned char*)mx_codes, (unsigned char*)mx_scales,
+                       mx_mode);
 }
 
 }  // namespace trtlab

@@ -199,16 +199,22 @@ void ExecutionContext::enqueue_all(hipStream_t s) {
                             op.N, os);
         break;
       case kLayerNorm:
-        // out2 (optional) = fused fp8 copy at q_scale
+        // epi 0: out2 (optional) = fused fp8 copy at q_scale.
+        // epi 4/8: producer-fused MX — out2 = codes, out3 = e8m0 scales.
         launch_layernorm(op.dtype, A(op.in_off), Fp(op.scale_off),
                          Fp(op.bias_off), A(op.out_off), op.M, op.N, op.N,
-                         op.eps, os, A(op.out2_off), op.q_scale);
+                         op.eps, os,
+                         op.epi ? nullptr : A(op.out2_off), op.q_scale,
+                         op.epi ? A(op.out2_off) : nullptr, A(op.out3_off),
+                         op.epi);
         break;
       case kAddLayerNorm:
         launch_add_layernorm(op.dtype, A(op.in_off), A(op.in2_off),
                              Fp(op.scale_off), Fp(op.bias_off), A(op.out_off),
                              nullptr /*sum_out*/, op.M, op.N, op.N, op.eps, os,
-                             A(op.out2_off) /*fused fp8 copy*/, op.q_scale);
+                             op.epi ? nullptr : A(op.out2_off), op.q_scale,
+                             op.epi ? A(op.out2_off) : nullptr,
+                             A(op.out3_off), op.epi);
         break;
       case kElementwise:
         launch_elementwise(op.dtype, op.epi, A(op.in_off), A(op.in2_off),

@@ -123,8 +123,9 @@ struct OpDesc {
   int kind = 0;
   int dtype = 0;  // 0 fp16, 1 bf16
   int epi = 0;    // epilogue enum / elementwise op code
-  // arena offsets in bytes (-1 = absent)
+  // arena offsets in bytes (-1 = absent); out3 = MX block-scales output
   int64_t in_off = -1, in2_off = -1, out_off = -1, out2_off = -1;
+  int64_t out3_off = -1;
   // weight-blob offsets in bytes (-1 = absent)
   int64_t w_off = -1, scale_off = -1, bias_off = -1;
   int64_t w2_off = -1;  // second weight slab (MX weight scales)

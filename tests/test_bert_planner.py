@@ -89,7 +89,15 @@ def test_mxfp4_plan_structure():
     plan = Planner(dtype=DT_MX4).compile(g)
     kinds = [d["kind"] for d in plan.ops]
     assert kinds.count(K_GEMM_MX4) == 8  # 4 gemms x 2 layers
-    assert kinds.count(K_QUANT_MX4) == 8
+    # producer fusion: quants whose row comes from a layernorm are folded
+    # INTO that LN (epi = 4 marks the MX-emitting LN); only rows produced
+    # by non-LN ops (graph input, attention, ff1-gelu) keep a standalone
+    # quantize — 2-layer BERT: 3 fused (l0.ff1, l1.qkv, l1.ff1), 5 left
+    assert kinds.count(K_QUANT_MX4) == 5
+    fused_ln = [d for d in plan.ops
+                if d["kind"] in (5, 6) and d.get("epi") == 4]
+    assert len(fused_ln) == 3
+    assert all(d["out2_off"] >= 0 and d["out3_off"] >= 0 for d in fused_ln)
     assert kinds.count(K_GEMM) == 0
     mx = [d for d in plan.ops if d["kind"] == K_GEMM_MX4]
     assert all(d["w2_off"] >= 0 and "in2_off" in d for d in mx)
@@ -116,7 +124,10 @@ def test_mxfp8_plan_structure():
     g = build_bert(batch=2, seq=128, layers=1, seed=0)
     plan = Planner(dtype=DT_MX8).compile(g)
     kinds = [d["kind"] for d in plan.ops]
-    assert kinds.count(K_GEMM_MX8) == 4 and kinds.count(K_QUANT_MX8) == 4
+    assert kinds.count(K_GEMM_MX8) == 4
+    assert kinds.count(K_QUANT_MX8) == 3  # ff1's quant fused into add_ln
+    assert sum(1 for d in plan.ops
+               if d["kind"] in (5, 6) and d.get("epi") == 8) == 1
     assert kinds.count(K_GEMM) == 0
     x = np.random.RandomState(3).randn(*plan.input_shape).astype(
         np.float32) * 0.5

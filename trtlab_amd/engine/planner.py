@@ -334,7 +334,28 @@ class Planner:
             g.params["mx_wscales"] = wsc  # e8m0 [N, K/32] -> w2_off
             new_ops.append(q)
             new_ops.append(g)
-        exec_ops[:] = new_ops
+        # ---- producer fusion: a quant whose input row is produced by a
+        # layernorm/add_layernorm is folded INTO that LN (the kernel emits
+        # the MX codes + e8m0 scales alongside its fp16 output), dropping
+        # the standalone quantize launch (round-2 MX lever).
+        producers = {}
+        for op in new_ops:
+            producers[op.output] = op
+        fused_ops = []
+        mode = 4 if fp4 else 8
+        for op in new_ops:
+            if op.kind == kq:
+                prod = producers.get(op.inputs[0])
+                if prod is not None and prod.kind in (K_LAYERNORM,
+                                                      K_ADD_LAYERNORM) \
+                        and "mx_out" not in prod.params:
+                    prod.params["mx_out"] = op.output
+                    prod.params["mx_scales"] = op.params["q_out"]
+                    prod.params["mx_mode"] = mode
+                    producers[op.output] = prod
+                    continue  # quant op absorbed
+            fused_ops.append(op)
+        exec_ops[:] = fused_ops
 
     # --------------------------------------------------------------- plan
     def compile(self, g: Graph) -> EnginePlan:
@@ -511,6 +532,9 @@ class Planner:
             touch(op.output, i)
             if "q_out" in op.params:  # fused fp8 second output
                 touch(op.params["q_out"], i)
+            if "mx_out" in op.params:  # producer-fused MX outputs
+                touch(op.params["mx_out"], i)
+                touch(op.params["mx_scales"], i)
         output_name = exec_ops[-1].output
         # N output bindings: the final op's output is the primary binding
         # [0]; tensors pinned by g.mark_output follow. All live to the end
@@ -588,17 +612,19 @@ class Planner:
             elif op.kind == K_SOFTMAX:
                 m, ncol = shapes[op.inputs[0]]
                 d.update(kind=K_SOFTMAX, M=m, N=ncol)
-            elif op.kind == K_LAYERNORM:
+            elif op.kind in (K_LAYERNORM, K_ADD_LAYERNORM):
                 m, ncol = shapes[op.inputs[0]]
-                d.update(kind=K_LAYERNORM, M=m, N=ncol, eps=op.params["eps"],
-                         out2_off=offsets.get(op.params.get("q_out"), -1),
-                         q_scale=op.params.get("q_scale", 0.0))
-            elif op.kind == K_ADD_LAYERNORM:
-                m, ncol = shapes[op.inputs[0]]
-                d.update(kind=K_ADD_LAYERNORM, M=m, N=ncol,
-                         eps=op.params["eps"],
-                         out2_off=offsets.get(op.params.get("q_out"), -1),
-                         q_scale=op.params.get("q_scale", 0.0))
+                d.update(kind=op.kind, M=m, N=ncol, eps=op.params["eps"])
+                if "mx_out" in op.params:
+                    # producer-fused MX: epi = mode (4/8), out2 = codes,
+                    # out3 = e8m0 block scales
+                    d.update(epi=op.params["mx_mode"],
+                             out2_off=offsets[op.params["mx_out"]],
+                             out3_off=offsets[op.params["mx_scales"]])
+                else:
+                    d.update(out2_off=offsets.get(op.params.get("q_out"),
+                                                  -1),
+                             q_scale=op.params.get("q_scale", 0.0))
             elif op.kind == K_ELEMENTWISE:
                 n = 1
                 for s_ in shapes[op.output]:

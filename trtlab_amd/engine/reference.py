@@ -144,21 +144,23 @@ def run_reference(plan: EnginePlan, input_nhwc, return_all: bool = False):
             t[op.output] = x.reshape(nb, h * w, c).mean(dim=1)
         elif op.kind == K_SOFTMAX:
             t[op.output] = F.softmax(x, dim=-1)
-        elif op.kind == K_LAYERNORM:
+        elif op.kind in (K_LAYERNORM, K_ADD_LAYERNORM):
+            src = x if op.kind == K_LAYERNORM else x + t[op.inputs[1]]
             g_ = torch.from_numpy(op.scale)
             b_ = torch.from_numpy(op.bias)
-            y = F.layer_norm(x, (x.shape[-1],), g_, b_, d["eps"])
+            y = F.layer_norm(src, (src.shape[-1],), g_, b_, d["eps"])
             t[op.output] = y
             if op.params.get("q_out"):
                 t[op.params["q_out"]] = _fp8_round(y / d["q_scale"])
-        elif op.kind == K_ADD_LAYERNORM:
-            s = x + t[op.inputs[1]]
-            g_ = torch.from_numpy(op.scale)
-            b_ = torch.from_numpy(op.bias)
-            y = F.layer_norm(s, (s.shape[-1],), g_, b_, d["eps"])
-            t[op.output] = y
-            if op.params.get("q_out"):
-                t[op.params["q_out"]] = _fp8_round(y / d["q_scale"])
+            if op.params.get("mx_out"):  # producer-fused MX quantization
+                from trtlab_amd.engine.mx import (quantize_mxfp4,
+                                                  quantize_mxfp8)
+
+                qf = (quantize_mxfp4 if op.params["mx_mode"] == 4
+                      else quantize_mxfp8)
+                codes, scales = qf(y.numpy().astype(np.float32))
+                t[op.params["mx_out"]] = torch.from_numpy(codes)
+                t[op.params["mx_scales"]] = torch.from_numpy(scales)
         elif op.kind == K_ELEMENTWISE:
             code = d["epi"]
             if code == 0:
