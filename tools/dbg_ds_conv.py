@@ -44,4 +44,4 @@ for name, nb, h, w, cin, cout, s in shapes:
     best.sort()
     row = " ".join(f"t{t}={u:6.1f}" for u, t in sorted(best, key=lambda p: p[1]))
     print(f"{name:26s} {row}  best t{best[0][1]} {best[0][0]:.1f} us "
-          f"({gb/best[0][0]*1e6:.2f} TB/s)", flush=True)
+          f"({gb/best[0][0]*1e6:.0f} GB/s)", flush=True)
