@@ -562,64 +562,65 @@ PYBIND11_MODULE(_C, m) {
           py::arg("stream") = 0, py::arg("sync") = true);
   ops.def("kv_append",
           [](uintptr_t qkv, uintptr_t kc, uintptr_t vc, uintptr_t pos, int B,
-             int H, int smax, uintptr_t stream, bool sync) {
+             int H, int smax, uintptr_t stream, bool sync, int D) {
             launch_kv_append((void*)qkv, (void*)kc, (void*)vc, (void*)pos, B,
-                             H, smax, as_stream(stream));
+                             H, smax, as_stream(stream), D);
             if (sync) TRT_HIP_CHECK(hipStreamSynchronize(as_stream(stream)));
           },
           py::arg("qkv"), py::arg("kcache"), py::arg("vcache"),
           py::arg("pos"), py::arg("B"), py::arg("H"), py::arg("smax"),
-          py::arg("stream") = 0, py::arg("sync") = true);
+          py::arg("stream") = 0, py::arg("sync") = true,
+          py::arg("D") = 64);
   ops.def("kv_append_range",
           [](uintptr_t qkv, uintptr_t kc, uintptr_t vc, int B, int H, int P,
-             int smax, uintptr_t stream, bool sync) {
+             int smax, uintptr_t stream, bool sync, int D) {
             launch_kv_append_range((void*)qkv, (void*)kc, (void*)vc, B, H, P,
-                                   smax, as_stream(stream));
+                                   smax, as_stream(stream), D);
             if (sync) TRT_HIP_CHECK(hipStreamSynchronize(as_stream(stream)));
           },
           py::arg("qkv"), py::arg("kcache"), py::arg("vcache"), py::arg("B"),
           py::arg("H"), py::arg("P"), py::arg("smax"), py::arg("stream") = 0,
-          py::arg("sync") = true);
+          py::arg("sync") = true, py::arg("D") = 64);
   ops.def("kv_append_paged",
           [](uintptr_t qkv, uintptr_t kp, uintptr_t vp, uintptr_t table,
              uintptr_t pos, int B, int H, int max_pages, uintptr_t stream,
-             bool sync) {
+             bool sync, int D) {
             launch_kv_append_paged((void*)qkv, (void*)kp, (void*)vp,
                                    (void*)table, (void*)pos, B, H,
-                                   max_pages, as_stream(stream));
+                                   max_pages, as_stream(stream), D);
             if (sync) TRT_HIP_CHECK(hipStreamSynchronize(as_stream(stream)));
           },
           py::arg("qkv"), py::arg("kpool"), py::arg("vpool"),
           py::arg("table"), py::arg("pos"), py::arg("B"), py::arg("H"),
           py::arg("max_pages"), py::arg("stream") = 0,
-          py::arg("sync") = true);
+          py::arg("sync") = true, py::arg("D") = 64);
   ops.def("decode_attention_paged",
           [](uintptr_t qkv, uintptr_t kp, uintptr_t vp, uintptr_t out,
              uintptr_t table, uintptr_t pos, int B, int H, int max_pages,
-             float scale, uintptr_t stream, bool sync) {
+             float scale, uintptr_t stream, bool sync, int D) {
             launch_decode_attention_paged((void*)qkv, (void*)kp, (void*)vp,
                                           (void*)out, (void*)table,
                                           (void*)pos, B, H, max_pages,
-                                          scale, as_stream(stream));
+                                          scale, as_stream(stream), D);
             if (sync) TRT_HIP_CHECK(hipStreamSynchronize(as_stream(stream)));
           },
           py::arg("qkv"), py::arg("kpool"), py::arg("vpool"), py::arg("out"),
           py::arg("table"), py::arg("pos"), py::arg("B"), py::arg("H"),
           py::arg("max_pages"), py::arg("scale"), py::arg("stream") = 0,
-          py::arg("sync") = true);
+          py::arg("sync") = true, py::arg("D") = 64);
   ops.def("decode_attention",
           [](uintptr_t qkv, uintptr_t kc, uintptr_t vc, uintptr_t out,
              uintptr_t pos, int B, int H, int smax, float scale,
-             uintptr_t stream, bool sync) {
+             uintptr_t stream, bool sync, int D) {
             launch_decode_attention((void*)qkv, (void*)kc, (void*)vc,
                                     (void*)out, (void*)pos, B, H, smax,
-                                    scale, as_stream(stream));
+                                    scale, as_stream(stream), D);
             if (sync) TRT_HIP_CHECK(hipStreamSynchronize(as_stream(stream)));
           },
           py::arg("qkv"), py::arg("kcache"), py::arg("vcache"),
           py::arg("out"), py::arg("pos"), py::arg("B"), py::arg("H"),
           py::arg("smax"), py::arg("scale"), py::arg("stream") = 0,
-          py::arg("sync") = true);
+          py::arg("sync") = true, py::arg("D") = 64);
   ops.def("decode_embed",
           [](uintptr_t ids, uintptr_t tok, uintptr_t pe, uintptr_t out,
              uintptr_t pos, int B, int hidden, uintptr_t stream, bool sync) {
@@ -656,28 +657,30 @@ PYBIND11_MODULE(_C, m) {
           py::arg("stream") = 0, py::arg("sync") = true);
   ops.def("kv_append_chunk",
           [](uintptr_t qkv, uintptr_t kc, uintptr_t vc, uintptr_t pos, int B,
-             int H, int K, int smax, uintptr_t stream, bool sync) {
+             int H, int K, int smax, uintptr_t stream, bool sync, int D) {
             launch_kv_append_chunk((void*)qkv, (void*)kc, (void*)vc,
                                    (void*)pos, B, H, K, smax,
-                                   as_stream(stream));
+                                   as_stream(stream), D);
             if (sync) TRT_HIP_CHECK(hipStreamSynchronize(as_stream(stream)));
           },
           py::arg("qkv"), py::arg("kcache"), py::arg("vcache"),
           py::arg("pos"), py::arg("B"), py::arg("H"), py::arg("K"),
-          py::arg("smax"), py::arg("stream") = 0, py::arg("sync") = true);
+          py::arg("smax"), py::arg("stream") = 0, py::arg("sync") = true,
+          py::arg("D") = 64);
   ops.def("chunk_attention",
           [](uintptr_t qkv, uintptr_t kc, uintptr_t vc, uintptr_t out,
              uintptr_t pos, int B, int H, int K, int smax, float scale,
-             uintptr_t stream, bool sync) {
+             uintptr_t stream, bool sync, int D) {
             launch_chunk_attention((void*)qkv, (void*)kc, (void*)vc,
                                    (void*)out, (void*)pos, B, H, K, smax,
-                                   scale, as_stream(stream));
+                                   scale, as_stream(stream), D);
             if (sync) TRT_HIP_CHECK(hipStreamSynchronize(as_stream(stream)));
           },
           py::arg("qkv"), py::arg("kcache"), py::arg("vcache"),
           py::arg("out"), py::arg("pos"), py::arg("B"), py::arg("H"),
           py::arg("K"), py::arg("smax"), py::arg("scale"),
-          py::arg("stream") = 0, py::arg("sync") = true);
+          py::arg("stream") = 0, py::arg("sync") = true,
+          py::arg("D") = 64);
   ops.def("chunk_embed",
           [](uintptr_t ids, uintptr_t tok, uintptr_t pe, uintptr_t out,
              uintptr_t pos, int B, int K, int smax, int hidden,

@@ -26,24 +26,25 @@ namespace trtlab {
 __global__ __launch_bounds__(64) void kv_append_kernel(
     const _Float16* __restrict__ qkv, _Float16* __restrict__ kcache,
     _Float16* __restrict__ vcache, const int* __restrict__ pos, int B, int H,
-    int smax) {
+    int smax, int D) {
   int b = blockIdx.x / H, h = blockIdx.x % H;
-  int d = threadIdx.x;  // 0..63
   int p = pos[b];
   if (p < 0) return;  // idle slot
-  int hid = H * 64;
-  int64_t src = (int64_t)b * 3 * hid + h * 64 + d;
-  int64_t dst = (((int64_t)b * H + h) * smax + p) * 64 + d;
-  kcache[dst] = qkv[src + hid];
-  vcache[dst] = qkv[src + 2 * hid];
+  int hid = H * D;
+  for (int d = threadIdx.x; d < D; d += 64) {
+    int64_t src = (int64_t)b * 3 * hid + h * D + d;
+    int64_t dst = (((int64_t)b * H + h) * smax + p) * D + d;
+    kcache[dst] = qkv[src + hid];
+    vcache[dst] = qkv[src + 2 * hid];
+  }
 }
 
 void launch_kv_append(const void* qkv, void* kcache, void* vcache,
                       const void* pos, int B, int H, int smax,
-                      hipStream_t stream) {
+                      hipStream_t stream, int D) {
   hipLaunchKernelGGL(kv_append_kernel, dim3(B * H), dim3(64), 0, stream,
                      (const _Float16*)qkv, (_Float16*)kcache,
-                     (_Float16*)vcache, (const int*)pos, B, H, smax);
+                     (_Float16*)vcache, (const int*)pos, B, H, smax, D);
 }
 
 // Batch-scatter a full prompt's K/V head rows (qkv [B*P, 3*H*64], rows
@@ -52,24 +53,25 @@ void launch_kv_append(const void* qkv, void* kcache, void* vcache,
 // kernels once, then decode continues from position P.
 __global__ __launch_bounds__(64) void kv_append_range_kernel(
     const _Float16* __restrict__ qkv, _Float16* __restrict__ kcache,
-    _Float16* __restrict__ vcache, int B, int H, int P, int smax) {
+    _Float16* __restrict__ vcache, int B, int H, int P, int smax, int D) {
   int p = blockIdx.x % P;
   int bh = blockIdx.x / P;
   int b = bh / H, h = bh % H;
-  int d = threadIdx.x;
-  int hid = H * 64;
-  int64_t src = ((int64_t)b * P + p) * 3 * hid + h * 64 + d;
-  int64_t dst = (((int64_t)b * H + h) * smax + p) * 64 + d;
-  kcache[dst] = qkv[src + hid];
-  vcache[dst] = qkv[src + 2 * hid];
+  int hid = H * D;
+  for (int d = threadIdx.x; d < D; d += 64) {
+    int64_t src = ((int64_t)b * P + p) * 3 * hid + h * D + d;
+    int64_t dst = (((int64_t)b * H + h) * smax + p) * D + d;
+    kcache[dst] = qkv[src + hid];
+    vcache[dst] = qkv[src + 2 * hid];
+  }
 }
 
 void launch_kv_append_range(const void* qkv, void* kcache, void* vcache,
                             int B, int H, int P, int smax,
-                            hipStream_t stream) {
+                            hipStream_t stream, int D) {
   hipLaunchKernelGGL(kv_append_range_kernel, dim3(B * H * P), dim3(64), 0,
                      stream, (const _Float16*)qkv, (_Float16*)kcache,
-                     (_Float16*)vcache, B, H, P, smax);
+                     (_Float16*)vcache, B, H, P, smax, D);
 }
 
 // Single-query attention against the cache: out[b, h*64+d] =
@@ -77,6 +79,7 @@ void launch_kv_append_range(const void* qkv, void* kcache, void* vcache,
 // keys lane, lane+64, ... for the score pass (its K rows are whole
 // 128-byte cachelines), then output element d = lane for the PV pass with
 // the probabilities broadcast through LDS. Smax <= 4096.
+template <int D>  // head_dim 64 or 128
 __global__ __launch_bounds__(64) void decode_attention_kernel(
     const _Float16* __restrict__ qkv, const _Float16* __restrict__ kcache,
     const _Float16* __restrict__ vcache, _Float16* __restrict__ out,
@@ -84,35 +87,32 @@ __global__ __launch_bounds__(64) void decode_attention_kernel(
   __shared__ float p_s[4096];
   int b = blockIdx.x / H, h = blockIdx.x % H;
   int lane = threadIdx.x;
-  int hid = H * 64;
+  int hid = H * D;
   if (pos[b] < 0) return;  // idle slot: no scan, stale output row unused
   int n = pos[b] + 1;  // keys 0..pos[b] (this step's K already appended)
 
-  // q for this head, one element per lane
-  float q[64];
+  // q for this head (fp16 registers: D=128 stays within budget)
+  _Float16 q[D];
   {
-    const _Float16* qrow = qkv + (int64_t)b * 3 * hid + h * 64;
+    const _Float16* qrow = qkv + (int64_t)b * 3 * hid + h * D;
 #pragma unroll
-    for (int c = 0; c < 8; ++c) {
-      half8v v = *(const half8v*)(qrow + c * 8);
-#pragma unroll
-      for (int j = 0; j < 8; ++j) q[c * 8 + j] = (float)((const _Float16*)&v)[j];
-    }
+    for (int c = 0; c < D / 8; ++c)
+      *(half8v*)(q + c * 8) = *(const half8v*)(qrow + c * 8);
   }
-  const _Float16* K = kcache + ((int64_t)b * H + h) * smax * 64;
-  const _Float16* V = vcache + ((int64_t)b * H + h) * smax * 64;
+  const _Float16* K = kcache + ((int64_t)b * H + h) * smax * D;
+  const _Float16* V = vcache + ((int64_t)b * H + h) * smax * D;
 
   // scores for this lane's keys
   float m = -3.0e38f;
   for (int t = lane; t < n; t += 64) {
-    const _Float16* krow = K + (int64_t)t * 64;
+    const _Float16* krow = K + (int64_t)t * D;
     float s = 0.f;
 #pragma unroll
-    for (int c = 0; c < 8; ++c) {
+    for (int c = 0; c < D / 8; ++c) {
       half8v v = *(const half8v*)(krow + c * 8);
 #pragma unroll
       for (int j = 0; j < 8; ++j)
-        s += q[c * 8 + j] * (float)((const _Float16*)&v)[j];
+        s += (float)q[c * 8 + j] * (float)((const _Float16*)&v)[j];
     }
     s *= scale;
     p_s[t] = s;
@@ -131,22 +131,41 @@ __global__ __launch_bounds__(64) void decode_attention_kernel(
   for (int off = 32; off; off >>= 1) l += __shfl_xor(l, off, 64);
   __syncthreads();
 
-  // PV: output element d = lane
-  float acc = 0.f;
-  for (int t = 0; t < n; ++t) acc += p_s[t] * (float)V[(int64_t)t * 64 + lane];
-  out[(int64_t)b * hid + h * 64 + lane] = (_Float16)(acc / l);
+  // PV: output elements d = lane (+64 for D=128)
+  float acc[D / 64];
+#pragma unroll
+  for (int j = 0; j < D / 64; ++j) acc[j] = 0.f;
+  for (int t = 0; t < n; ++t) {
+    const _Float16* vrow = V + (int64_t)t * D;
+    float p = p_s[t];
+#pragma unroll
+    for (int j = 0; j < D / 64; ++j)
+      acc[j] += p * (float)vrow[j * 64 + lane];
+  }
+#pragma unroll
+  for (int j = 0; j < D / 64; ++j)
+    out[(int64_t)b * hid + h * D + j * 64 + lane] =
+        (_Float16)(acc[j] / l);
 }
 
 void launch_decode_attention(const void* qkv, const void* kcache,
                              const void* vcache, void* out, const void* pos,
                              int B, int H, int smax, float scale,
-                             hipStream_t stream) {
+                             hipStream_t stream, int D) {
   if (smax > 4096)
     throw std::runtime_error("decode_attention: smax > 4096 unsupported");
-  hipLaunchKernelGGL(decode_attention_kernel, dim3(B * H), dim3(64), 0,
-                     stream, (const _Float16*)qkv, (const _Float16*)kcache,
-                     (const _Float16*)vcache, (_Float16*)out, (const int*)pos,
-                     B, H, smax, scale);
+  if (D == 128)
+    hipLaunchKernelGGL(decode_attention_kernel<128>, dim3(B * H), dim3(64),
+                       0, stream, (const _Float16*)qkv,
+                       (const _Float16*)kcache, (const _Float16*)vcache,
+                       (_Float16*)out, (const int*)pos, B, H, smax, scale);
+  else if (D == 64)
+    hipLaunchKernelGGL(decode_attention_kernel<64>, dim3(B * H), dim3(64), 0,
+                       stream, (const _Float16*)qkv, (const _Float16*)kcache,
+                       (const _Float16*)vcache, (_Float16*)out,
+                       (const int*)pos, B, H, smax, scale);
+  else
+    throw std::runtime_error("decode_attention: head_dim must be 64 or 128");
 }
 
 // ---- speculative-decoding verification chunk kernels ----
@@ -159,34 +178,36 @@ void launch_decode_attention(const void* qkv, const void* kcache,
 __global__ __launch_bounds__(64) void kv_append_chunk_kernel(
     const _Float16* __restrict__ qkv, _Float16* __restrict__ kcache,
     _Float16* __restrict__ vcache, const int* __restrict__ pos, int B, int H,
-    int K, int smax) {
+    int K, int smax, int D) {
   int q = blockIdx.x % K;
   int bh = blockIdx.x / K;
   int b = bh / H, h = bh % H;
-  int d = threadIdx.x;
   int p0 = pos[b];
   if (p0 < 0) return;  // idle slot
   int p = p0 + q;
   if (p >= smax) p = smax - 1;
-  int hid = H * 64;
-  int64_t src = ((int64_t)b * K + q) * 3 * hid + h * 64 + d;
-  int64_t dst = (((int64_t)b * H + h) * smax + p) * 64 + d;
-  kcache[dst] = qkv[src + hid];
-  vcache[dst] = qkv[src + 2 * hid];
+  int hid = H * D;
+  for (int d = threadIdx.x; d < D; d += 64) {
+    int64_t src = ((int64_t)b * K + q) * 3 * hid + h * D + d;
+    int64_t dst = (((int64_t)b * H + h) * smax + p) * D + d;
+    kcache[dst] = qkv[src + hid];
+    vcache[dst] = qkv[src + 2 * hid];
+  }
 }
 
 void launch_kv_append_chunk(const void* qkv, void* kcache, void* vcache,
                             const void* pos, int B, int H, int K, int smax,
-                            hipStream_t stream) {
+                            hipStream_t stream, int D) {
   hipLaunchKernelGGL(kv_append_chunk_kernel, dim3(B * H * K), dim3(64), 0,
                      stream, (const _Float16*)qkv, (_Float16*)kcache,
-                     (_Float16*)vcache, (const int*)pos, B, H, K, smax);
+                     (_Float16*)vcache, (const int*)pos, B, H, K, smax, D);
 }
 
 // Multi-query single-head attention against the cache: query row (b, q)
 // attends keys 0 .. pos[b]+q (its own K already appended). One wave per
 // (b, q, h) — the chunk's queries are causal within the chunk by
 // construction of their lengths.
+template <int D>  // head_dim 64 or 128
 __global__ __launch_bounds__(64) void chunk_attention_kernel(
     const _Float16* __restrict__ qkv, const _Float16* __restrict__ kcache,
     const _Float16* __restrict__ vcache, _Float16* __restrict__ out,
@@ -197,37 +218,32 @@ __global__ __launch_bounds__(64) void chunk_attention_kernel(
   int bh = blockIdx.x / K;
   int b = bh / H, h = bh % H;
   int lane = threadIdx.x;
-  int hid = H * 64;
+  int hid = H * D;
   int p0 = pos[b];
   if (p0 < 0) return;  // idle slot
   int n = p0 + q + 1;
   if (n > smax) n = smax;
 
-  float qv[64];
+  _Float16 qv[D];
   {
-    const _Float16* qrow =
-        qkv + ((int64_t)b * K + q) * 3 * hid + h * 64;
+    const _Float16* qrow = qkv + ((int64_t)b * K + q) * 3 * hid + h * D;
 #pragma unroll
-    for (int c = 0; c < 8; ++c) {
-      half8v v = *(const half8v*)(qrow + c * 8);
-#pragma unroll
-      for (int j = 0; j < 8; ++j)
-        qv[c * 8 + j] = (float)((const _Float16*)&v)[j];
-    }
+    for (int c = 0; c < D / 8; ++c)
+      *(half8v*)(qv + c * 8) = *(const half8v*)(qrow + c * 8);
   }
-  const _Float16* Kc = kcache + ((int64_t)b * H + h) * smax * 64;
-  const _Float16* Vc = vcache + ((int64_t)b * H + h) * smax * 64;
+  const _Float16* Kc = kcache + ((int64_t)b * H + h) * smax * D;
+  const _Float16* Vc = vcache + ((int64_t)b * H + h) * smax * D;
 
   float m = -3.0e38f;
   for (int t = lane; t < n; t += 64) {
-    const _Float16* krow = Kc + (int64_t)t * 64;
+    const _Float16* krow = Kc + (int64_t)t * D;
     float s = 0.f;
 #pragma unroll
-    for (int c = 0; c < 8; ++c) {
+    for (int c = 0; c < D / 8; ++c) {
       half8v v = *(const half8v*)(krow + c * 8);
 #pragma unroll
       for (int j = 0; j < 8; ++j)
-        s += qv[c * 8 + j] * (float)((const _Float16*)&v)[j];
+        s += (float)qv[c * 8 + j] * (float)((const _Float16*)&v)[j];
     }
     s *= scale;
     p_s[t] = s;
@@ -245,22 +261,42 @@ __global__ __launch_bounds__(64) void chunk_attention_kernel(
 #pragma unroll
   for (int off = 32; off; off >>= 1) l += __shfl_xor(l, off, 64);
   __syncthreads();
-  float acc = 0.f;
-  for (int t = 0; t < n; ++t)
-    acc += p_s[t] * (float)Vc[(int64_t)t * 64 + lane];
-  out[((int64_t)b * K + q) * hid + h * 64 + lane] = (_Float16)(acc / l);
+  float acc[D / 64];
+#pragma unroll
+  for (int j = 0; j < D / 64; ++j) acc[j] = 0.f;
+  for (int t = 0; t < n; ++t) {
+    const _Float16* vrow = Vc + (int64_t)t * D;
+    float p = p_s[t];
+#pragma unroll
+    for (int j = 0; j < D / 64; ++j)
+      acc[j] += p * (float)vrow[j * 64 + lane];
+  }
+#pragma unroll
+  for (int j = 0; j < D / 64; ++j)
+    out[((int64_t)b * K + q) * hid + h * D + j * 64 + lane] =
+        (_Float16)(acc[j] / l);
 }
 
 void launch_chunk_attention(const void* qkv, const void* kcache,
                             const void* vcache, void* out, const void* pos,
                             int B, int H, int K, int smax, float scale,
-                            hipStream_t stream) {
+                            hipStream_t stream, int D) {
   if (smax > 4096)
     throw std::runtime_error("chunk_attention: smax > 4096 unsupported");
-  hipLaunchKernelGGL(chunk_attention_kernel, dim3(B * H * K), dim3(64), 0,
-                     stream, (const _Float16*)qkv, (const _Float16*)kcache,
-                     (const _Float16*)vcache, (_Float16*)out,
-                     (const int*)pos, B, H, K, smax, scale);
+  if (D == 128)
+    hipLaunchKernelGGL(chunk_attention_kernel<128>, dim3(B * H * K),
+                       dim3(64), 0, stream, (const _Float16*)qkv,
+                       (const _Float16*)kcache, (const _Float16*)vcache,
+                       (_Float16*)out, (const int*)pos, B, H, K, smax,
+                       scale);
+  else if (D == 64)
+    hipLaunchKernelGGL(chunk_attention_kernel<64>, dim3(B * H * K), dim3(64),
+                       0, stream, (const _Float16*)qkv,
+                       (const _Float16*)kcache, (const _Float16*)vcache,
+                       (_Float16*)out, (const int*)pos, B, H, K, smax,
+                       scale);
+  else
+    throw std::runtime_error("chunk_attention: head_dim must be 64 or 128");
 }
 
 // Chunk embedding: out[b*K + q] = tok[ids[b*K + q]] + posemb[pos[b] + q].
@@ -594,29 +630,31 @@ void launch_decode_gemm_fused(int pro, int epi, const void* x, const void* r,
 __global__ __launch_bounds__(64) void kv_append_paged_kernel(
     const _Float16* __restrict__ qkv, _Float16* __restrict__ kpool,
     _Float16* __restrict__ vpool, const int* __restrict__ table,
-    const int* __restrict__ pos, int B, int H, int max_pages) {
+    const int* __restrict__ pos, int B, int H, int max_pages, int D) {
   int b = blockIdx.x / H, h = blockIdx.x % H;
-  int d = threadIdx.x;
   int p = pos[b];
   if (p < 0) return;  // idle slot
   int page = table[b * max_pages + (p >> 6)];
   if (page < 0) return;  // unmapped (host error) — fail soft, not wild
-  int hid = H * 64;
-  int64_t src = (int64_t)b * 3 * hid + h * 64 + d;
-  int64_t dst = (((int64_t)page * 64 + (p & 63)) * H + h) * 64 + d;
-  kpool[dst] = qkv[src + hid];
-  vpool[dst] = qkv[src + 2 * hid];
+  int hid = H * D;
+  for (int d = threadIdx.x; d < D; d += 64) {
+    int64_t src = (int64_t)b * 3 * hid + h * D + d;
+    int64_t dst = (((int64_t)page * 64 + (p & 63)) * H + h) * D + d;
+    kpool[dst] = qkv[src + hid];
+    vpool[dst] = qkv[src + 2 * hid];
+  }
 }
 
 void launch_kv_append_paged(const void* qkv, void* kpool, void* vpool,
                             const void* table, const void* pos, int B, int H,
-                            int max_pages, hipStream_t stream) {
+                            int max_pages, hipStream_t stream, int D) {
   hipLaunchKernelGGL(kv_append_paged_kernel, dim3(B * H), dim3(64), 0,
                      stream, (const _Float16*)qkv, (_Float16*)kpool,
                      (_Float16*)vpool, (const int*)table, (const int*)pos,
-                     B, H, max_pages);
+                     B, H, max_pages, D);
 }
 
+template <int D>  // head_dim 64 or 128
 __global__ __launch_bounds__(64) void decode_attention_paged_kernel(
     const _Float16* __restrict__ qkv, const _Float16* __restrict__ kpool,
     const _Float16* __restrict__ vpool, _Float16* __restrict__ out,
@@ -625,35 +663,30 @@ __global__ __launch_bounds__(64) void decode_attention_paged_kernel(
   __shared__ float p_s[4096];
   int b = blockIdx.x / H, h = blockIdx.x % H;
   int lane = threadIdx.x;
-  int hid = H * 64;
+  int hid = H * D;
   if (pos[b] < 0) return;  // idle slot
   int n = pos[b] + 1;
 
-  float q[64];
+  _Float16 q[D];
   {
-    const _Float16* qrow = qkv + (int64_t)b * 3 * hid + h * 64;
+    const _Float16* qrow = qkv + (int64_t)b * 3 * hid + h * D;
 #pragma unroll
-    for (int c = 0; c < 8; ++c) {
-      half8v v = *(const half8v*)(qrow + c * 8);
-#pragma unroll
-      for (int j = 0; j < 8; ++j)
-        q[c * 8 + j] = (float)((const _Float16*)&v)[j];
-    }
+    for (int c = 0; c < D / 8; ++c)
+      *(half8v*)(q + c * 8) = *(const half8v*)(qrow + c * 8);
   }
   const int* tab = table + (int64_t)b * max_pages;
 
   float m = -3.0e38f;
   for (int t = lane; t < n; t += 64) {
-    int64_t rowoff =
-        (((int64_t)tab[t >> 6] * 64 + (t & 63)) * H + h) * 64;
+    int64_t rowoff = (((int64_t)tab[t >> 6] * 64 + (t & 63)) * H + h) * D;
     const _Float16* krow = kpool + rowoff;
     float s = 0.f;
 #pragma unroll
-    for (int c = 0; c < 8; ++c) {
+    for (int c = 0; c < D / 8; ++c) {
       half8v v = *(const half8v*)(krow + c * 8);
 #pragma unroll
       for (int j = 0; j < 8; ++j)
-        s += q[c * 8 + j] * (float)((const _Float16*)&v)[j];
+        s += (float)q[c * 8 + j] * (float)((const _Float16*)&v)[j];
     }
     s *= scale;
     p_s[t] = s;
@@ -671,25 +704,41 @@ __global__ __launch_bounds__(64) void decode_attention_paged_kernel(
 #pragma unroll
   for (int off = 32; off; off >>= 1) l += __shfl_xor(l, off, 64);
   __syncthreads();
-  float acc = 0.f;
+  float acc[D / 64];
+#pragma unroll
+  for (int j = 0; j < D / 64; ++j) acc[j] = 0.f;
   for (int t = 0; t < n; ++t) {
-    int64_t rowoff =
-        (((int64_t)tab[t >> 6] * 64 + (t & 63)) * H + h) * 64;
-    acc += p_s[t] * (float)vpool[rowoff + lane];
+    int64_t rowoff = (((int64_t)tab[t >> 6] * 64 + (t & 63)) * H + h) * D;
+    float p = p_s[t];
+#pragma unroll
+    for (int j = 0; j < D / 64; ++j)
+      acc[j] += p * (float)vpool[rowoff + j * 64 + lane];
   }
-  out[(int64_t)b * hid + h * 64 + lane] = (_Float16)(acc / l);
+#pragma unroll
+  for (int j = 0; j < D / 64; ++j)
+    out[(int64_t)b * hid + h * D + j * 64 + lane] = (_Float16)(acc[j] / l);
 }
 
 void launch_decode_attention_paged(const void* qkv, const void* kpool,
                                    const void* vpool, void* out,
                                    const void* table, const void* pos, int B,
                                    int H, int max_pages, float scale,
-                                   hipStream_t stream) {
-  hipLaunchKernelGGL(decode_attention_paged_kernel, dim3(B * H), dim3(64), 0,
-                     stream, (const _Float16*)qkv, (const _Float16*)kpool,
-                     (const _Float16*)vpool, (_Float16*)out,
-                     (const int*)table, (const int*)pos, B, H, max_pages,
-                     scale);
+                                   hipStream_t stream, int D) {
+  if (D == 128)
+    hipLaunchKernelGGL(decode_attention_paged_kernel<128>, dim3(B * H),
+                       dim3(64), 0, stream, (const _Float16*)qkv,
+                       (const _Float16*)kpool, (const _Float16*)vpool,
+                       (_Float16*)out, (const int*)table, (const int*)pos,
+                       B, H, max_pages, scale);
+  else if (D == 64)
+    hipLaunchKernelGGL(decode_attention_paged_kernel<64>, dim3(B * H),
+                       dim3(64), 0, stream, (const _Float16*)qkv,
+                       (const _Float16*)kpool, (const _Float16*)vpool,
+                       (_Float16*)out, (const int*)table, (const int*)pos,
+                       B, H, max_pages, scale);
+  else
+    throw std::runtime_error(
+        "decode_attention_paged: head_dim must be 64 or 128");
 }
 
 }  // namespace trtlab

@@ -102,25 +102,25 @@ void launch_quantize_mxfp8(const void* x, void* codes, void* scales,
 // ---- incremental decode (KV cache; see csrc/kernels/decode.hip) ----
 void launch_kv_append(const void* qkv, void* kcache, void* vcache,
                       const void* pos, int B, int H, int smax,
-                      hipStream_t stream);
+                      hipStream_t stream, int D = 64);
 void launch_kv_append_range(const void* qkv, void* kcache, void* vcache,
                             int B, int H, int P, int smax,
-                            hipStream_t stream);
+                            hipStream_t stream, int D = 64);
 void launch_decode_attention(const void* qkv, const void* kcache,
                              const void* vcache, void* out, const void* pos,
                              int B, int H, int smax, float scale,
-                             hipStream_t stream);
+                             hipStream_t stream, int D = 64);
 void launch_decode_embed(const void* ids, const void* tok, const void* posemb,
                          void* out, const void* pos, int B, int hidden,
                          hipStream_t stream);
 // speculative-decoding verification chunk (see decode.hip)
 void launch_kv_append_chunk(const void* qkv, void* kcache, void* vcache,
                             const void* pos, int B, int H, int K, int smax,
-                            hipStream_t stream);
+                            hipStream_t stream, int D = 64);
 void launch_chunk_attention(const void* qkv, const void* kcache,
                             const void* vcache, void* out, const void* pos,
                             int B, int H, int K, int smax, float scale,
-                            hipStream_t stream);
+                            hipStream_t stream, int D = 64);
 void launch_chunk_embed(const void* ids, const void* tok, const void* posemb,
                         void* out, const void* pos, int B, int K, int smax,
                         int hidden, hipStream_t stream);
@@ -128,12 +128,12 @@ void launch_advance_pos(void* pos, int B, int smax, hipStream_t stream);
 // paged KV cache (vLLM-style block tables; see decode.hip)
 void launch_kv_append_paged(const void* qkv, void* kpool, void* vpool,
                             const void* table, const void* pos, int B, int H,
-                            int max_pages, hipStream_t stream);
+                            int max_pages, hipStream_t stream, int D = 64);
 void launch_decode_attention_paged(const void* qkv, const void* kpool,
                                    const void* vpool, void* out,
                                    const void* table, const void* pos, int B,
                                    int H, int max_pages, float scale,
-                                   hipStream_t stream);
+                                   hipStream_t stream, int D = 64);
 // fused decode GEMM: prologue 1=LN 2=ADD_LN(+h_out) 3=EMBED_LN; epilogue
 // 0=bias 1=bias+gelu 2=bias+KV-scatter (see decode.hip)
 void launch_decode_gemm_fused(int pro, int epi, const void* x, const void* r,

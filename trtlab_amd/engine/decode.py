@@ -40,7 +40,7 @@ class PagedKVPool:
 
     def __init__(self, layers: int, heads: int, batch: int,
                  num_pages: int, max_pages_per_slot: int,
-                 device: int = 0):
+                 device: int = 0, head_dim: int = 64):
         import torch
 
         from trtlab_amd import native
@@ -53,7 +53,8 @@ class PagedKVPool:
         self.batch = batch
         self.num_pages = num_pages
         self.max_pages = max_pages_per_slot
-        self.kpools = [torch.zeros(num_pages, 64, heads, 64,
+        self.head_dim = head_dim
+        self.kpools = [torch.zeros(num_pages, 64, heads, head_dim,
                                    dtype=torch.half, device="cuda")
                        for _ in range(layers)]
         self.vpools = [torch.zeros_like(k) for k in self.kpools]
@@ -160,6 +161,10 @@ class DecodeSession:
         while f"l{li}_qkv" in nodes:
             att = nodes[f"l{li}_att"]
             self.heads = att.attrs["heads"]
+            self.hd = att.attrs.get("head_dim",
+                                    self.hidden // self.heads)
+            if self.hd not in (64, 128):
+                raise ValueError("decode: head_dim must be 64 or 128")
             lay = {}
             for key, nm in (("ln1", f"l{li}_ln1"), ("ln2", f"l{li}_ln2")):
                 lay[key + "_g"] = dev32(nodes[nm].attrs["gamma"])
@@ -168,7 +173,7 @@ class DecodeSession:
                 n = nodes[f"l{li}_{key}"]
                 lay[key + "_w"] = dev16(n.attrs["weight"])
                 lay[key + "_b"] = dev32(n.attrs["bias"])
-            lay["kcache"] = torch.zeros(batch, self.heads, smax, 64,
+            lay["kcache"] = torch.zeros(batch, self.heads, smax, self.hd,
                                         dtype=torch.half, device="cuda")
             lay["vcache"] = torch.zeros_like(lay["kcache"])
             self.layers.append(lay)
@@ -205,9 +210,10 @@ class DecodeSession:
             self.kv_pool = paged if isinstance(paged, PagedKVPool) else \
                 PagedKVPool(self.n_layers, self.heads, B,
                             num_pages=B * mp, max_pages_per_slot=mp,
-                            device=device)
+                            device=device, head_dim=self.hd)
             if self.kv_pool.layers != self.n_layers or \
-                    self.kv_pool.heads != self.heads:
+                    self.kv_pool.heads != self.heads or \
+                    getattr(self.kv_pool, "head_dim", 64) != self.hd:
                 raise ValueError("pool layer/head mismatch")
             for lay in self.layers:  # release the dense caches
                 lay["kcache"] = None
@@ -215,6 +221,8 @@ class DecodeSession:
         self.fused = bool(fused)
         if self.fused and (B > 64 or not lm_head):
             raise ValueError("fused decode needs batch <= 64 and lm_head")
+        if self.fused and self.hd != 64:
+            raise ValueError("fused decode supports head_dim 64 only")
         if self.fused and self.kv_pool is not None:
             raise ValueError("fused + paged not supported together yet")
         self.stream = self._C.hip.stream_create()
@@ -229,7 +237,7 @@ class DecodeSession:
         C, s = self._C, self.stream
         B, Hd = self.batch, self.hidden
         ops = C.ops
-        scale = 1.0 / float(np.sqrt(64.0))
+        scale = 1.0 / float(np.sqrt(float(self.hd)))
         h_cur, h_nxt = self.h, self.h2
         for li, lay in enumerate(self.layers):
             if li == 0:  # embed + ln1 + qkv + kv scatter, persists h
@@ -330,25 +338,28 @@ class DecodeSession:
                                     pool.vpools[li].data_ptr(),
                                     pool.table.data_ptr(),
                                     self.pos.data_ptr(), B, self.heads,
-                                    pool.max_pages, stream=s, sync=False)
+                                    pool.max_pages, stream=s, sync=False,
+                                    D=self.hd)
                 ops.decode_attention_paged(
                     self.qkv.data_ptr(), pool.kpools[li].data_ptr(),
                     pool.vpools[li].data_ptr(), self.att.data_ptr(),
                     pool.table.data_ptr(), self.pos.data_ptr(), B,
                     self.heads, pool.max_pages,
-                    1.0 / float(np.sqrt(64.0)), stream=s, sync=False)
+                    1.0 / float(np.sqrt(float(self.hd))), stream=s,
+                    sync=False, D=self.hd)
             else:
                 ops.kv_append(self.qkv.data_ptr(), lay["kcache"].data_ptr(),
                               lay["vcache"].data_ptr(), self.pos.data_ptr(),
-                              B, self.heads, self.smax, stream=s, sync=False)
+                              B, self.heads, self.smax, stream=s, sync=False,
+                              D=self.hd)
                 ops.decode_attention(self.qkv.data_ptr(),
                                      lay["kcache"].data_ptr(),
                                      lay["vcache"].data_ptr(),
                                      self.att.data_ptr(),
                                      self.pos.data_ptr(), B,
                                      self.heads, self.smax,
-                                     1.0 / float(np.sqrt(64.0)), stream=s,
-                                     sync=False)
+                                     1.0 / float(np.sqrt(float(self.hd))),
+                                     stream=s, sync=False, D=self.hd)
             ops.gemm_bt(0, self.att.data_ptr(), lay["proj_w"].data_ptr(),
                         self.x2.data_ptr(), bias=lay["proj_b"].data_ptr(),
                         M=B, N=Hd, K=Hd, epi=self._epi_bias, stream=s,
@@ -430,9 +441,10 @@ class DecodeSession:
                         sync=False)
             ops.kv_append_range(qkv.data_ptr(), lay["kcache"].data_ptr(),
                                 lay["vcache"].data_ptr(), B, self.heads, Pp,
-                                self.smax, stream=s, sync=False)
+                                self.smax, stream=s, sync=False, D=self.hd)
             ops.attention(0, qkv.data_ptr(), x2.data_ptr(), B, Pp,
-                          self.heads, 64, 1.0 / float(np.sqrt(64.0)),
+                          self.heads, self.hd,
+                          1.0 / float(np.sqrt(float(self.hd))),
                           stream=s, sync=False, causal=1)
             ops.gemm_bt(0, x2.data_ptr(), lay["proj_w"].data_ptr(),
                         x.data_ptr(), bias=lay["proj_b"].data_ptr(), M=M,
@@ -570,14 +582,14 @@ class DecodeSession:
                                 lay["kcache"].data_ptr(),
                                 lay["vcache"].data_ptr(),
                                 self.pos.data_ptr(), B, self.heads, K,
-                                self.smax, stream=s, sync=False)
+                                self.smax, stream=s, sync=False, D=self.hd)
             ops.chunk_attention(cb["qkv"].data_ptr(),
                                 lay["kcache"].data_ptr(),
                                 lay["vcache"].data_ptr(),
                                 cb["att"].data_ptr(), self.pos.data_ptr(),
                                 B, self.heads, K, self.smax,
-                                1.0 / float(np.sqrt(64.0)), stream=s,
-                                sync=False)
+                                1.0 / float(np.sqrt(float(self.hd))),
+                                stream=s, sync=False, D=self.hd)
             ops.gemm_bt(0, cb["att"].data_ptr(), lay["proj_w"].data_ptr(),
                         cb["x2"].data_ptr(), bias=lay["proj_b"].data_ptr(),
                         M=M, N=Hd, K=Hd, epi=self._epi_bias, stream=s,

@@ -24,8 +24,8 @@ def build_gpt2(batch: int = 8, seq: int = 1024, hidden: int = 768,
     """embeddings=True: int32 token ids input [B*S] -> tok+pos gather.
     Otherwise pre-embedded hidden states [B*S, hidden] fp16."""
     assert seq >= 1  # attention streams key tiles at ANY sequence length
-    assert hidden % heads == 0 and hidden // heads == 64, \
-        "attention kernel: head_dim must be 64"
+    assert hidden % heads == 0 and hidden // heads in (64, 128), \
+        "attention kernels: head_dim must be 64 or 128"
     inter = hidden * 4
     rng = np.random.RandomState(seed)
 
