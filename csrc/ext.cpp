@@ -443,6 +443,36 @@ PYBIND11_MODULE(_C, m) {
           py::arg("dtype"), py::arg("op"), py::arg("a"), py::arg("b") = 0,
           py::arg("out") = 0, py::arg("n") = 0, py::arg("stream") = 0,
           py::arg("sync") = true);
+  ops.def("rmsnorm",
+          [](int dtype, uintptr_t in, uintptr_t gamma, uintptr_t out, int M,
+             int N, float eps, uintptr_t stream, bool sync) {
+            launch_rmsnorm(dtype, (void*)in, (float*)gamma, (void*)out, M,
+                           N, N, eps, as_stream(stream));
+            if (sync) TRT_HIP_CHECK(hipStreamSynchronize(as_stream(stream)));
+          },
+          py::arg("dtype"), py::arg("in"), py::arg("gamma"), py::arg("out"),
+          py::arg("M"), py::arg("N"), py::arg("eps") = 1e-5f,
+          py::arg("stream") = 0, py::arg("sync") = true);
+  ops.def("silu_mul",
+          [](int dtype, uintptr_t a, uintptr_t b, uintptr_t out, int64_t n,
+             uintptr_t stream, bool sync) {
+            launch_silu_mul(dtype, (void*)a, (void*)b, (void*)out, n,
+                            as_stream(stream));
+            if (sync) TRT_HIP_CHECK(hipStreamSynchronize(as_stream(stream)));
+          },
+          py::arg("dtype"), py::arg("a"), py::arg("b"), py::arg("out"),
+          py::arg("n"), py::arg("stream") = 0, py::arg("sync") = true);
+  ops.def("rope",
+          [](int dtype, uintptr_t qkv, uintptr_t pos, int M, int S, int H,
+             int D, float theta, uintptr_t stream, bool sync) {
+            launch_rope(dtype, (void*)qkv, (void*)pos, M, S, H, D, theta,
+                        as_stream(stream));
+            if (sync) TRT_HIP_CHECK(hipStreamSynchronize(as_stream(stream)));
+          },
+          py::arg("dtype"), py::arg("qkv"), py::arg("pos") = 0,
+          py::arg("M") = 0, py::arg("S") = 0, py::arg("H") = 0,
+          py::arg("D") = 0, py::arg("theta") = 10000.0f,
+          py::arg("stream") = 0, py::arg("sync") = true);
   ops.def("clip",
           [](int dtype, uintptr_t in, uintptr_t out, int64_t n, float mn,
              float mx, uintptr_t stream, bool sync) {

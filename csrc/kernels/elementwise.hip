@@ -354,5 +354,85 @@ void launch_copy2d(int dtype, const void* src, void* dst, int64_t M, int C,
                        ldd);
 }
 
+
+// ---- SwiGLU gate: out = silu(a) * b (LLaMA FFN) ----
+template <typename T>
+__global__ void silu_mul_kernel(const T* __restrict__ a,
+                                const T* __restrict__ b,
+                                T* __restrict__ out, int64_t n8) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n8;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    short4v a0 = *(const short4v*)((const T*)a + i * 8);
+    short4v a1 = *(const short4v*)((const T*)a + i * 8 + 4);
+    short4v b0 = *(const short4v*)((const T*)b + i * 8);
+    short4v b1 = *(const short4v*)((const T*)b + i * 8 + 4);
+    T o[8];
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      float x0 = (float)((const T*)&a0)[j];
+      float x1 = (float)((const T*)&a1)[j];
+      o[j] = (T)(x0 / (1.f + __expf(-x0)) * (float)((const T*)&b0)[j]);
+      o[4 + j] = (T)(x1 / (1.f + __expf(-x1)) * (float)((const T*)&b1)[j]);
+    }
+    *(short4v*)(out + i * 8) = *(const short4v*)&o[0];
+    *(short4v*)(out + i * 8 + 4) = *(const short4v*)&o[4];
+  }
+}
+
+void launch_silu_mul(int dtype, const void* a, const void* b, void* out,
+                     int64_t n, hipStream_t stream) {
+  if (n % 8 != 0) throw std::runtime_error("silu_mul: n % 8 != 0");
+  int blocks = ew_blocks(n / 8);
+  if (dtype == 0)
+    hipLaunchKernelGGL((silu_mul_kernel<_Float16>), dim3(blocks), dim3(256),
+                       0, stream, (const _Float16*)a, (const _Float16*)b,
+                       (_Float16*)out, n / 8);
+  else
+    hipLaunchKernelGGL((silu_mul_kernel<__bf16>), dim3(blocks), dim3(256), 0,
+                       stream, (const __bf16*)a, (const __bf16*)b,
+                       (__bf16*)out, n / 8);
+}
+
+// ---- rotary position embedding (RoPE, LLaMA-style interleaved-half) ----
+// In-place on the q and k column blocks of the fused qkv rows
+// [M = B*S, 3*H*D]: for head h, pair (d, d + D/2), angle =
+// pos * theta^(-2d/D). Row positions: pos = row % S (full-sequence
+// forward; decode uses per-slot device pos via rope_pos).
+template <typename T>
+__global__ void rope_kernel(T* __restrict__ qkv,
+                            const int* __restrict__ pos_dev, int M, int S,
+                            int H, int D, float theta, int K /*B for dev*/) {
+  int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int half = D / 2;
+  int64_t per_row = (int64_t)2 * H * half;  // q and k rotate, v untouched
+  if (idx >= (int64_t)M * per_row) return;
+  int64_t row = idx / per_row;
+  int r = (int)(idx - row * per_row);
+  int qk = r / (H * half);        // 0 = q block, 1 = k block
+  int rr = r % (H * half);
+  int h = rr / half, d = rr % half;
+  int p = pos_dev ? pos_dev[row] : (int)(row % S);
+  if (p < 0) return;  // idle slot (decode)
+  int hid = H * D;
+  T* base = qkv + row * (int64_t)3 * hid + qk * hid + h * D;
+  float ang = (float)p * __powf(theta, -2.0f * (float)d / (float)D);
+  float c, sn;
+  __sincosf(ang, &sn, &c);
+  float x0 = (float)base[d], x1 = (float)base[d + half];
+  base[d] = (T)(x0 * c - x1 * sn);
+  base[d + half] = (T)(x0 * sn + x1 * c);
+}
+
+void launch_rope(int dtype, void* qkv, const void* pos_dev, int M, int S,
+                 int H, int D, float theta, hipStream_t stream) {
+  if (D % 2 != 0) throw std::runtime_error("rope: D % 2 != 0");
+  int64_t total = (int64_t)M * 2 * H * (D / 2);
+  int blocks = (int)std::min<int64_t>((total + 255) / 256, 16384);
+  if (dtype != 0) throw std::runtime_error("rope: fp16 only");
+  hipLaunchKernelGGL((rope_kernel<_Float16>), dim3(blocks), dim3(256), 0,
+                     stream, (_Float16*)qkv, (const int*)pos_dev, M, S, H, D,
+                     theta, 0);
+}
+
 }  // namespace trtlab
 

@@ -51,6 +51,17 @@ void launch_add_layernorm(int dtype, const void* x, const void* res,
                           float q_scale = 0.f, void* mx_codes = nullptr,
                           void* mx_scales = nullptr, int mx_mode = 0);
 
+void launch_rmsnorm(int dtype, const void* in, const float* gamma, void* out,
+                    int M, int N, int64_t ld, float eps, hipStream_t stream);
+void launch_add_rmsnorm(int dtype, const void* x, const void* res,
+                        const float* gamma, void* out, void* sum_out, int M,
+                        int N, int64_t ld, float eps, hipStream_t stream);
+void launch_silu_mul(int dtype, const void* a, const void* b, void* out,
+                     int64_t n, hipStream_t stream);
+// RoPE on the q/k blocks of fused qkv rows; pos_dev = per-row device
+// positions (decode) or null (pos = row %% S, full-sequence forward)
+void launch_rope(int dtype, void* qkv, const void* pos_dev, int M, int S,
+                 int H, int D, float theta, hipStream_t stream);
 void launch_elementwise(int dtype, int op, const void* a, const void* b,
                         void* out, int64_t n, hipStream_t stream);
 void launch_clip(int dtype, const void* in, void* out, int64_t n, float mn,

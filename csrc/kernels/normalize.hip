@@ -319,4 +319,114 @@ void launch_add_layernorm(int dtype, const void* x, const void* res,
                        mx_mode);
 }
 
+
+// ---- RMSNorm (LLaMA-family): out = x / rms(x) * gamma ----
+// Same one-wave-per-row structure as layernorm; no mean subtraction, no
+// beta (the LLaMA norm has none).
+template <typename T>
+__global__ void rmsnorm_kernel(const T* __restrict__ in,
+                               const float* __restrict__ gamma,
+                               T* __restrict__ out, int M, int N, int64_t ld,
+                               float eps) {
+  int lane = threadIdx.x & 63;
+  int row = (blockIdx.x * (blockDim.x >> 6)) + (threadIdx.x >> 6);
+  if (row >= M) return;
+  const T* src = in + (int64_t)row * ld;
+  T* dst = out + (int64_t)row * ld;
+  float v[kMaxChunks][8];
+  int nc = 0;
+  float ss = 0.f;
+  for (int c = lane; c * 8 < N; c += 64, ++nc) {
+    load8(src + c * 8, v[nc]);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) ss += v[nc][j] * v[nc][j];
+  }
+  ss = wave_reduce_sum(ss);
+  float r = rsqrtf(ss / (float)N + eps);
+  nc = 0;
+  for (int c = lane; c * 8 < N; c += 64, ++nc) {
+    float4v g0 = *(const float4v*)(gamma + c * 8);
+    float4v g1 = *(const float4v*)(gamma + c * 8 + 4);
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      v[nc][j] = v[nc][j] * r * ((const float*)&g0)[j];
+      v[nc][4 + j] = v[nc][4 + j] * r * ((const float*)&g1)[j];
+    }
+    store8(dst + c * 8, v[nc]);
+  }
+}
+
+// add + rmsnorm (residual stream variant; sum_out = updated residual)
+template <typename T>
+__global__ void add_rmsnorm_kernel(const T* __restrict__ x,
+                                   const T* __restrict__ res,
+                                   const float* __restrict__ gamma,
+                                   T* __restrict__ out,
+                                   T* __restrict__ sum_out, int M, int N,
+                                   int64_t ld, float eps) {
+  int lane = threadIdx.x & 63;
+  int row = (blockIdx.x * (blockDim.x >> 6)) + (threadIdx.x >> 6);
+  if (row >= M) return;
+  const T* xs = x + (int64_t)row * ld;
+  const T* rs = res + (int64_t)row * ld;
+  T* dst = out + (int64_t)row * ld;
+  T* sm = sum_out ? sum_out + (int64_t)row * ld : nullptr;
+  float v[kMaxChunks][8];
+  int nc = 0;
+  float ss = 0.f;
+  for (int c = lane; c * 8 < N; c += 64, ++nc) {
+    float a[8], b[8];
+    load8(xs + c * 8, a);
+    load8(rs + c * 8, b);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      v[nc][j] = a[j] + b[j];
+      ss += v[nc][j] * v[nc][j];
+    }
+    if (sm) store8(sm + c * 8, v[nc]);
+  }
+  ss = wave_reduce_sum(ss);
+  float r = rsqrtf(ss / (float)N + eps);
+  nc = 0;
+  for (int c = lane; c * 8 < N; c += 64, ++nc) {
+    float4v g0 = *(const float4v*)(gamma + c * 8);
+    float4v g1 = *(const float4v*)(gamma + c * 8 + 4);
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      v[nc][j] = v[nc][j] * r * ((const float*)&g0)[j];
+      v[nc][4 + j] = v[nc][4 + j] * r * ((const float*)&g1)[j];
+    }
+    store8(dst + c * 8, v[nc]);
+  }
+}
+
+void launch_rmsnorm(int dtype, const void* in, const float* gamma, void* out,
+                    int M, int N, int64_t ld, float eps,
+                    hipStream_t stream) {
+  if (N > 2048 || N % 8 != 0) throw std::runtime_error("rmsnorm: bad N");
+  if (dtype == 0)
+    hipLaunchKernelGGL((rmsnorm_kernel<_Float16>), rows_grid(M), dim3(256),
+                       0, stream, (const _Float16*)in, gamma,
+                       (_Float16*)out, M, N, ld, eps);
+  else
+    hipLaunchKernelGGL((rmsnorm_kernel<__bf16>), rows_grid(M), dim3(256), 0,
+                       stream, (const __bf16*)in, gamma, (__bf16*)out, M, N,
+                       ld, eps);
+}
+
+void launch_add_rmsnorm(int dtype, const void* x, const void* res,
+                        const float* gamma, void* out, void* sum_out, int M,
+                        int N, int64_t ld, float eps, hipStream_t stream) {
+  if (N > 2048 || N % 8 != 0) throw std::runtime_error("add_rmsnorm: bad N");
+  if (dtype == 0)
+    hipLaunchKernelGGL((add_rmsnorm_kernel<_Float16>), rows_grid(M),
+                       dim3(256), 0, stream, (const _Float16*)x,
+                       (const _Float16*)res, gamma, (_Float16*)out,
+                       (_Float16*)sum_out, M, N, ld, eps);
+  else
+    hipLaunchKernelGGL((add_rmsnorm_kernel<__bf16>), rows_grid(M), dim3(256),
+                       0, stream, (const __bf16*)x, (const __bf16*)res,
+                       gamma, (__bf16*)out, (__bf16*)sum_out, M, N, ld, eps);
+}
+
 }  // namespace trtlab

@@ -275,6 +275,19 @@ void ExecutionContext::enqueue_all(hipStream_t s) {
                          Wp(op.scale_off), Wp(op.bias_off), A(op.in2_off),
                          A(op.out_off), op.M, op.S, op.N, os);
         break;
+      case kRMSNorm:
+        launch_rmsnorm(op.dtype, A(op.in_off), Fp(op.scale_off),
+                       A(op.out_off), op.M, op.N, op.N, op.eps, os);
+        break;
+      case kSiluMul:
+        launch_silu_mul(op.dtype, A(op.in_off), A(op.in2_off), A(op.out_off),
+                        op.n_elems, os);
+        break;
+      case kRope:
+        // in-place on the (arena-aliased) qkv buffer; eps carries theta
+        launch_rope(op.dtype, A(op.out_off), nullptr, op.M, op.S, op.NH,
+                    op.HD, op.eps, os);
+        break;
       case kClip:
         // res_scale = min bound, q_scale = max bound
         launch_clip(op.dtype, A(op.in_off), A(op.out_off), op.n_elems,

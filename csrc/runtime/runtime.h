@@ -117,6 +117,9 @@ enum OpKind : int {
   kClip = 19,        // out = min(max(x, res_scale), q_scale) — ONNX Clip
   kTranspose2D = 20, // out[N][M] = in[M][N]^T (tiled LDS transpose)
   kCopy2D = 21,      // dst[m][epi + c] = src[m][c], dst row stride Cout
+  kRMSNorm = 22,     // LLaMA norm: x / rms(x) * gamma (no mean, no beta)
+  kSiluMul = 23,     // SwiGLU gate: silu(a) * b
+  kRope = 24,        // rotary embedding in-place on qkv q/k blocks
 };
 
 struct OpDesc {

@@ -192,6 +192,28 @@ class Graph:
                                else np.asarray(seg_table, np.float32),
                                seq=s_), name)
 
+    def rmsnorm(self, x: str, gamma, eps: float = 1e-5,
+                name: Optional[str] = None) -> str:
+        """LLaMA norm: x / rms(x) * gamma (no mean subtraction/beta)."""
+        return self._emit("rmsnorm", [x], self.tensors[x].shape,
+                          dict(gamma=np.asarray(gamma, np.float32),
+                               eps=eps), name)
+
+    def silu_mul(self, a: str, b: str, name: Optional[str] = None) -> str:
+        """SwiGLU gate: silu(a) * b (LLaMA FFN)."""
+        assert self.tensors[a].shape == self.tensors[b].shape
+        return self._emit("silu_mul", [a, b], self.tensors[a].shape, {},
+                          name)
+
+    def rope(self, qkv: str, heads: int, seq: int, theta: float = 10000.0,
+             name: Optional[str] = None) -> str:
+        """Rotary position embedding applied to the q/k blocks of fused
+        qkv rows (in place — the planner aliases output and input in the
+        arena). LLaMA-style half-split rotation, angle base `theta`."""
+        return self._emit("rope", [qkv], self.tensors[qkv].shape,
+                          dict(heads=heads, seq=seq, theta=float(theta)),
+                          name)
+
     def clip(self, x: str, mn: float, mx: float,
              name: Optional[str] = None) -> str:
         """out = min(max(x, mn), mx) (ONNX Clip with arbitrary bounds)."""

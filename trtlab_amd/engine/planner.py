@@ -56,6 +56,9 @@ K_GEMM_MX8 = 18  # MXFP8 x MXFP8 scaled-MFMA GEMM (fp16 out + epilogue)
 K_CLIP = 19  # out = min(max(x, mn), mx) — ONNX Clip with arbitrary bounds
 K_TRANSPOSE2D = 20  # out[N][M] = in[M][N]^T (tiled LDS transpose kernel)
 K_COPY2D = 21  # dst[m][coff + c] = src[m][c] — Concat lowering
+K_RMSNORM = 22  # LLaMA norm: x / rms(x) * gamma
+K_SILU_MUL = 23  # SwiGLU gate: silu(a) * b
+K_ROPE = 24  # rotary embedding, in-place on qkv (arena-aliased output)
 
 
 def _bf16_bits(arr: np.ndarray) -> np.ndarray:
@@ -201,6 +204,17 @@ class Planner:
                 exec_ops.append(op)
             elif n.kind == "attention":
                 exec_ops.append(ExecOp(K_ATTENTION, n.name, [n.inputs[0]],
+                                       n.output, dict(n.attrs)))
+            elif n.kind == "rmsnorm":
+                op = ExecOp(K_RMSNORM, n.name, [n.inputs[0]], n.output,
+                            dict(eps=n.attrs["eps"]))
+                op.scale = n.attrs["gamma"].astype(np.float32)
+                exec_ops.append(op)
+            elif n.kind == "silu_mul":
+                exec_ops.append(ExecOp(K_SILU_MUL, n.name, list(n.inputs),
+                                       n.output, {}))
+            elif n.kind == "rope":
+                exec_ops.append(ExecOp(K_ROPE, n.name, [n.inputs[0]],
                                        n.output, dict(n.attrs)))
             elif n.kind == "clip":
                 exec_ops.append(ExecOp(K_CLIP, n.name, [n.inputs[0]],
@@ -559,6 +573,21 @@ class Planner:
                 n *= d
             return n * itemsize.get(t, 2)
 
+        # rope outputs alias their inputs (in-place rotation): merge the
+        # two tensors' live intervals onto the INPUT and copy its offset
+        # to the output after planning
+        rope_alias: Dict[str, str] = {}
+        for op in exec_ops:
+            if op.kind == K_ROPE:
+                src, dst = op.inputs[0], op.output
+                # follow chains (rope of rope never happens, but be safe)
+                src = rope_alias.get(src, src)
+                rope_alias[dst] = src
+                s0a, e0a = tensors_used[src]
+                s0b, e0b = tensors_used[dst]
+                tensors_used[src] = (min(s0a, s0b), max(e0a, e0b))
+                del tensors_used[dst]
+
         arena = ArenaPlanner()
         for t, (s0, e0) in tensors_used.items():
             if self.reuse:
@@ -566,6 +595,8 @@ class Planner:
             else:
                 arena.add(t, nbytes_of(t), 0, len(exec_ops))
         offsets, arena_bytes = arena.plan()
+        for dst, src in rope_alias.items():
+            offsets[dst] = offsets[src]
 
         # ---- emit op dicts ----
         op_dicts: List[Dict[str, Any]] = []
@@ -664,6 +695,22 @@ class Planner:
                          epi=op.params.get("out_dtype", 0),
                          q_scale=op.params.get("q_scale", 0.0),
                          causal=1 if op.params.get("causal") else 0)
+            elif op.kind == K_RMSNORM:
+                m, ncol = shapes[op.inputs[0]]
+                d.update(kind=K_RMSNORM, M=m, N=ncol, eps=op.params["eps"])
+            elif op.kind == K_SILU_MUL:
+                n_ = 1
+                for s_ in shapes[op.output]:
+                    n_ *= s_
+                d.update(kind=K_SILU_MUL, n_elems=n_)
+            elif op.kind == K_ROPE:
+                m, n3 = shapes[op.inputs[0]]
+                heads = op.params["heads"]
+                hd = n3 // (3 * heads)
+                # eps carries theta; the op runs IN PLACE on the aliased
+                # arena buffer (out_off == in_off by construction below)
+                d.update(kind=K_ROPE, M=m, S=op.params["seq"], NH=heads,
+                         HD=hd, eps=op.params["theta"])
             elif op.kind == K_CLIP:
                 n = 1
                 for s_ in shapes[op.output]:

@@ -15,7 +15,7 @@ import torch
 import torch.nn.functional as F
 
 from trtlab_amd.engine.planner import (
-    K_CLIP, K_COPY2D, K_TRANSPOSE2D,
+    K_CLIP, K_COPY2D, K_TRANSPOSE2D, K_RMSNORM, K_SILU_MUL, K_ROPE,
     EnginePlan, K_ADD_LAYERNORM, K_ATTENTION, K_AVGPOOL, K_CHANNEL_PAD, K_CONV,
     K_DEQUANT, K_ELEMENTWISE, K_GAVGPOOL, K_GEMM, K_LAYERNORM, K_MAXPOOL,
     K_EMBEDDING, K_GEMM_MX4, K_GEMM_MX8, K_QUANT_MX4, K_QUANT_MX8,
@@ -215,6 +215,34 @@ def run_reference(plan: EnginePlan, input_nhwc, return_all: bool = False):
             y = (att @ v).permute(0, 2, 1, 3).reshape(b * s, hid)
             if d.get("epi") == 3:  # fused fp8 output
                 y = _fp8_round(y / d["q_scale"])
+            t[op.output] = y
+        elif op.kind == K_RMSNORM:
+            g_ = torch.from_numpy(op.scale)
+            r = torch.rsqrt(x.pow(2).mean(-1, keepdim=True) + d["eps"])
+            t[op.output] = x * r * g_
+        elif op.kind == K_SILU_MUL:
+            t[op.output] = F.silu(x) * t[op.inputs[1]]
+        elif op.kind == K_ROPE:
+            m, n3 = x.shape
+            heads, hd, seq = d["NH"], d["HD"], d["S"]
+            half = hd // 2
+            theta = d["eps"]
+            posv = torch.arange(m) % seq
+            dvec = torch.arange(half, dtype=torch.float64)
+            ang = posv[:, None].double() * theta ** (-2.0 * dvec / hd)
+            cos = torch.cos(ang).float()  # [m, half]
+            sin = torch.sin(ang).float()
+            y = x.clone()
+            hid = heads * hd
+            for blk in range(2):  # q then k; v untouched
+                base = blk * hid
+                v = x[:, base:base + hid].reshape(m, heads, hd)
+                x0 = v[..., :half]
+                x1 = v[..., half:]
+                r0 = x0 * cos[:, None, :] - x1 * sin[:, None, :]
+                r1 = x0 * sin[:, None, :] + x1 * cos[:, None, :]
+                y[:, base:base + hid] = torch.cat([r0, r1], -1).reshape(
+                    m, hid)
             t[op.output] = y
         elif op.kind == K_CLIP:
             t[op.output] = torch.clamp(x, d["res_scale"], d["q_scale"])

@@ -4,3 +4,4 @@ synthetic-data protocol, matching the reference's models/README.md:4-8
 from trtlab_amd.models.bert import build_bert  # noqa: F401
 from trtlab_amd.models.gpt2 import build_gpt2  # noqa: F401
 from trtlab_amd.models.resnet import build_resnet  # noqa: F401
+from trtlab_amd.models.llama import build_llama  # noqa: F401
