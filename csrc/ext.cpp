@@ -453,6 +453,19 @@ PYBIND11_MODULE(_C, m) {
           py::arg("dtype"), py::arg("in"), py::arg("gamma"), py::arg("out"),
           py::arg("M"), py::arg("N"), py::arg("eps") = 1e-5f,
           py::arg("stream") = 0, py::arg("sync") = true);
+  ops.def("add_rmsnorm",
+          [](int dtype, uintptr_t x, uintptr_t res, uintptr_t gamma,
+             uintptr_t out, uintptr_t sum_out, int M, int N, float eps,
+             uintptr_t stream, bool sync) {
+            launch_add_rmsnorm(dtype, (void*)x, (void*)res, (float*)gamma,
+                               (void*)out, (void*)sum_out, M, N, N, eps,
+                               as_stream(stream));
+            if (sync) TRT_HIP_CHECK(hipStreamSynchronize(as_stream(stream)));
+          },
+          py::arg("dtype"), py::arg("x"), py::arg("res"), py::arg("gamma"),
+          py::arg("out"), py::arg("sum_out") = 0, py::arg("M") = 0,
+          py::arg("N") = 0, py::arg("eps") = 1e-5f, py::arg("stream") = 0,
+          py::arg("sync") = true);
   ops.def("silu_mul",
           [](int dtype, uintptr_t a, uintptr_t b, uintptr_t out, int64_t n,
              uintptr_t stream, bool sync) {
@@ -464,15 +477,16 @@ PYBIND11_MODULE(_C, m) {
           py::arg("n"), py::arg("stream") = 0, py::arg("sync") = true);
   ops.def("rope",
           [](int dtype, uintptr_t qkv, uintptr_t pos, int M, int S, int H,
-             int D, float theta, uintptr_t stream, bool sync) {
+             int D, float theta, uintptr_t stream, bool sync, int chunk) {
             launch_rope(dtype, (void*)qkv, (void*)pos, M, S, H, D, theta,
-                        as_stream(stream));
+                        as_stream(stream), chunk);
             if (sync) TRT_HIP_CHECK(hipStreamSynchronize(as_stream(stream)));
           },
           py::arg("dtype"), py::arg("qkv"), py::arg("pos") = 0,
           py::arg("M") = 0, py::arg("S") = 0, py::arg("H") = 0,
           py::arg("D") = 0, py::arg("theta") = 10000.0f,
-          py::arg("stream") = 0, py::arg("sync") = true);
+          py::arg("stream") = 0, py::arg("sync") = true,
+          py::arg("chunk") = 0);
   ops.def("clip",
           [](int dtype, uintptr_t in, uintptr_t out, int64_t n, float mn,
              float mx, uintptr_t stream, bool sync) {

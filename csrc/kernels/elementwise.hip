@@ -411,8 +411,15 @@ __global__ void rope_kernel(T* __restrict__ qkv,
   int qk = r / (H * half);        // 0 = q block, 1 = k block
   int rr = r % (H * half);
   int h = rr / half, d = rr % half;
-  int p = pos_dev ? pos_dev[row] : (int)(row % S);
-  if (p < 0) return;  // idle slot (decode)
+  int p;
+  if (!pos_dev)
+    p = (int)(row % S);                       // full-sequence forward
+  else if (K > 0)
+    p = pos_dev[row / K] + (int)(row % K);    // verify chunk: K rows/slot
+  else
+    p = pos_dev[row];                         // decode step: one row/slot
+  if (p < 0 || (K > 0 && pos_dev[row / K] < 0))
+    return;  // idle slot (decode)
   int hid = H * D;
   T* base = qkv + row * (int64_t)3 * hid + qk * hid + h * D;
   float ang = (float)p * __powf(theta, -2.0f * (float)d / (float)D);
@@ -424,14 +431,14 @@ __global__ void rope_kernel(T* __restrict__ qkv,
 }
 
 void launch_rope(int dtype, void* qkv, const void* pos_dev, int M, int S,
-                 int H, int D, float theta, hipStream_t stream) {
+                 int H, int D, float theta, hipStream_t stream, int chunk) {
   if (D % 2 != 0) throw std::runtime_error("rope: D % 2 != 0");
   int64_t total = (int64_t)M * 2 * H * (D / 2);
   int blocks = (int)std::min<int64_t>((total + 255) / 256, 16384);
   if (dtype != 0) throw std::runtime_error("rope: fp16 only");
   hipLaunchKernelGGL((rope_kernel<_Float16>), dim3(blocks), dim3(256), 0,
                      stream, (_Float16*)qkv, (const int*)pos_dev, M, S, H, D,
-                     theta, 0);
+                     theta, chunk);
 }
 
 }  // namespace trtlab

@@ -58,10 +58,13 @@ void launch_add_rmsnorm(int dtype, const void* x, const void* res,
                         int N, int64_t ld, float eps, hipStream_t stream);
 void launch_silu_mul(int dtype, const void* a, const void* b, void* out,
                      int64_t n, hipStream_t stream);
-// RoPE on the q/k blocks of fused qkv rows; pos_dev = per-row device
-// positions (decode) or null (pos = row %% S, full-sequence forward)
+// RoPE on the q/k blocks of fused qkv rows. pos_dev = null: pos = row % S
+// (full-sequence forward). pos_dev set, chunk = 0: pos = pos_dev[row]
+// (decode step, M = B slots). chunk = K > 0: pos = pos_dev[row / K] +
+// row % K (speculative verify chunk, rows grouped K-per-slot).
 void launch_rope(int dtype, void* qkv, const void* pos_dev, int M, int S,
-                 int H, int D, float theta, hipStream_t stream);
+                 int H, int D, float theta, hipStream_t stream,
+                 int chunk = 0);
 void launch_elementwise(int dtype, int op, const void* a, const void* b,
                         void* out, int64_t n, hipStream_t stream);
 void launch_clip(int dtype, const void* in, void* out, int64_t n, float mn,

@@ -1,4 +1,4 @@
-"""Incremental (KV-cache) decoding for GPT-2-family models.
+"""Incremental (KV-cache) decoding for GPT-2- and LLaMA-family models.
 
 Beyond-reference capability (the CUDA reference served static TensorRT
 engines only): a `DecodeSession` holds per-layer K/V caches resident in
@@ -15,10 +15,14 @@ buffer (csrc/ext.cpp test_scratch); the captured graph bakes its address,
 so other raw-op users must not force a regrow while a session is live
 (engine-managed contexts are unaffected).
 
-Weights come from a `build_gpt2(embeddings=True)` IR graph (the same
-random-init builder the full-sequence engine uses, so prefill/decode can
-be cross-checked). Prefill here is sequential priming (step the prompt
-token by token); fused prefill-into-cache is round-2 work.
+Weights come from a `build_gpt2(embeddings=True)` or `build_llama()` IR
+graph (the same random-init builders the full-sequence engine uses, so
+prefill/decode can be cross-checked). The recipe is detected from the
+node names: `l0_rms1` present -> LLaMA (RMSNorm + RoPE + SwiGLU, no
+biases; RoPE reads the device position counter, so the captured decode
+graph stays position-free); otherwise GPT-2 (LayerNorm + learned pos
+emb + GELU). prefill()/step()/verify_chunk() and the paged KV pool work
+for both.
 """
 from __future__ import annotations
 
@@ -138,7 +142,12 @@ class DecodeSession:
         self._epi_none = EPI_NONE
 
         # ---- pull weights out of the IR graph by node kind/name ----
+        # Two layer recipes share the session plumbing (KV caches, paged
+        # pool, graph capture, position counters, spec-decode chunks):
+        #   gpt2  — LayerNorm + learned pos emb + GELU MLP (l{i}_ln1 ...)
+        #   llama — RMSNorm + RoPE + SwiGLU, no biases   (l{i}_rms1 ...)
         nodes = {n.name: n for n in graph.nodes}
+        self.arch = "llama" if "l0_rms1" in nodes else "gpt2"
         emb = next(n for n in graph.nodes if n.kind == "embedding")
         self.hidden = emb.attrs["tok"].shape[1]
         self.vocab = emb.attrs["tok"].shape[0]
@@ -166,23 +175,40 @@ class DecodeSession:
             if self.hd not in (64, 128):
                 raise ValueError("decode: head_dim must be 64 or 128")
             lay = {}
-            for key, nm in (("ln1", f"l{li}_ln1"), ("ln2", f"l{li}_ln2")):
-                lay[key + "_g"] = dev32(nodes[nm].attrs["gamma"])
-                lay[key + "_b"] = dev32(nodes[nm].attrs["beta"])
-            for key in ("qkv", "proj", "ff1", "ff2"):
-                n = nodes[f"l{li}_{key}"]
-                lay[key + "_w"] = dev16(n.attrs["weight"])
-                lay[key + "_b"] = dev32(n.attrs["bias"])
+            if self.arch == "llama":
+                lay["rms1_g"] = dev32(nodes[f"l{li}_rms1"].attrs["gamma"])
+                lay["rms2_g"] = dev32(nodes[f"l{li}_rms2"].attrs["gamma"])
+                self.theta = float(
+                    nodes[f"l{li}_rope"].attrs.get("theta", 10000.0))
+                for key in ("qkv", "proj", "gate", "up", "down"):
+                    lay[key + "_w"] = dev16(nodes[f"l{li}_{key}"]
+                                            .attrs["weight"])
+            else:
+                for key, nm in (("ln1", f"l{li}_ln1"), ("ln2", f"l{li}_ln2")):
+                    lay[key + "_g"] = dev32(nodes[nm].attrs["gamma"])
+                    lay[key + "_b"] = dev32(nodes[nm].attrs["beta"])
+                for key in ("qkv", "proj", "ff1", "ff2"):
+                    n = nodes[f"l{li}_{key}"]
+                    lay[key + "_w"] = dev16(n.attrs["weight"])
+                    lay[key + "_b"] = dev32(n.attrs["bias"])
             lay["kcache"] = torch.zeros(batch, self.heads, smax, self.hd,
                                         dtype=torch.half, device="cuda")
             lay["vcache"] = torch.zeros_like(lay["kcache"])
             self.layers.append(lay)
             li += 1
         self.n_layers = li
-        self.inter = self.layers[0]["ff1_w"].shape[0]
-        gf = nodes["ln_f"]
-        self.lnf_g = dev32(gf.attrs["gamma"])
-        self.lnf_b = dev32(gf.attrs["beta"])
+        if self.arch == "llama":
+            if self.hidden > 2048:
+                raise ValueError("llama decode: hidden > 2048 (rmsnorm "
+                                 "kernel row limit)")
+            self.inter = self.layers[0]["gate_w"].shape[0]
+            self.lnf_g = dev32(nodes["rms_f"].attrs["gamma"])
+            self.lnf_b = None
+        else:
+            self.inter = self.layers[0]["ff1_w"].shape[0]
+            gf = nodes["ln_f"]
+            self.lnf_g = dev32(gf.attrs["gamma"])
+            self.lnf_b = dev32(gf.attrs["beta"])
         self.lm_head = lm_head  # logits = h @ tok^T (weight tying)
 
         B, Hd = batch, self.hidden
@@ -196,6 +222,10 @@ class DecodeSession:
         self.att = torch.zeros(B, Hd, dtype=torch.half, device="cuda")
         self.x2 = torch.zeros(B, Hd, dtype=torch.half, device="cuda")
         self.ff = torch.zeros(B, self.inter, dtype=torch.half, device="cuda")
+        # SwiGLU needs gate AND up live at once (silu(gate) * up)
+        self.ff2 = (torch.zeros(B, self.inter, dtype=torch.half,
+                                device="cuda")
+                    if self.arch == "llama" else None)
         self.out = torch.zeros(B, Hd, dtype=torch.half, device="cuda")
         self.logits = (torch.zeros(B, self.vocab, dtype=torch.half,
                                    device="cuda") if lm_head else None)
@@ -219,6 +249,8 @@ class DecodeSession:
                 lay["kcache"] = None
                 lay["vcache"] = None
         self.fused = bool(fused)
+        if self.fused and self.arch != "gpt2":
+            raise ValueError("fused decode supports the gpt2 recipe only")
         if self.fused and (B > 64 or not lm_head):
             raise ValueError("fused decode needs batch <= 64 and lm_head")
         if self.fused and self.hd != 64:
@@ -301,7 +333,102 @@ class DecodeSession:
         ops.advance_pos(self.pos.data_ptr(), B, self.smax, stream=s,
                         sync=False)
 
+    def _kv_attn(self, li: int, lay: Dict):
+        """Shared per-layer KV append + attention (dense or paged)."""
+        ops, s = self._C.ops, self.stream
+        B = self.batch
+        scale = 1.0 / float(np.sqrt(float(self.hd)))
+        if self.kv_pool is not None:
+            pool = self.kv_pool
+            ops.kv_append_paged(self.qkv.data_ptr(),
+                                pool.kpools[li].data_ptr(),
+                                pool.vpools[li].data_ptr(),
+                                pool.table.data_ptr(),
+                                self.pos.data_ptr(), B, self.heads,
+                                pool.max_pages, stream=s, sync=False,
+                                D=self.hd)
+            ops.decode_attention_paged(
+                self.qkv.data_ptr(), pool.kpools[li].data_ptr(),
+                pool.vpools[li].data_ptr(), self.att.data_ptr(),
+                pool.table.data_ptr(), self.pos.data_ptr(), B,
+                self.heads, pool.max_pages, scale, stream=s,
+                sync=False, D=self.hd)
+        else:
+            ops.kv_append(self.qkv.data_ptr(), lay["kcache"].data_ptr(),
+                          lay["vcache"].data_ptr(), self.pos.data_ptr(),
+                          B, self.heads, self.smax, stream=s, sync=False,
+                          D=self.hd)
+            ops.decode_attention(self.qkv.data_ptr(),
+                                 lay["kcache"].data_ptr(),
+                                 lay["vcache"].data_ptr(),
+                                 self.att.data_ptr(),
+                                 self.pos.data_ptr(), B,
+                                 self.heads, self.smax, scale,
+                                 stream=s, sync=False, D=self.hd)
+
+    def _enqueue_llama(self):
+        """One LLaMA decode step (pos-relative, captured like the gpt2
+        recipe): embed -> [per layer: rmsnorm -> qkv gemm -> rope(pos) ->
+        kv_append -> decode_attention -> proj -> add_rmsnorm(res fused) ->
+        gate/up gemms -> silu_mul -> down -> add_rmsnorm] -> rms_f ->
+        [lm head] -> advance_pos. No biases anywhere (LLaMA); RoPE reads
+        the device position counter directly (pos_dev[row], idle slots
+        skip), so the captured graph stays position-free."""
+        C, s = self._C, self.stream
+        B, Hd = self.batch, self.hidden
+        T4 = 0
+        ops = C.ops
+        ops.decode_embed(self.ids.data_ptr(), self.tok.data_ptr(),
+                         self.posemb.data_ptr(), self.h.data_ptr(),
+                         self.pos.data_ptr(), B, Hd, stream=s, sync=False)
+        ops.rmsnorm(0, self.h.data_ptr(),
+                    self.layers[0]["rms1_g"].data_ptr(), self.x.data_ptr(),
+                    B, Hd, stream=s, sync=False)
+        for li, lay in enumerate(self.layers):
+            ops.gemm_bt(0, self.x.data_ptr(), lay["qkv_w"].data_ptr(),
+                        self.qkv.data_ptr(), M=B, N=3 * Hd, K=Hd,
+                        epi=self._epi_none, stream=s, sync=False, tile=T4)
+            ops.rope(0, self.qkv.data_ptr(), pos=self.pos.data_ptr(), M=B,
+                     S=self.smax, H=self.heads, D=self.hd, theta=self.theta,
+                     stream=s, sync=False)
+            self._kv_attn(li, lay)
+            ops.gemm_bt(0, self.att.data_ptr(), lay["proj_w"].data_ptr(),
+                        self.x2.data_ptr(), M=B, N=Hd, K=Hd,
+                        epi=self._epi_none, stream=s, sync=False, tile=T4)
+            # h += proj; x = rms2(h)  (one kernel, sum_out = new residual)
+            ops.add_rmsnorm(0, self.x2.data_ptr(), self.h.data_ptr(),
+                            lay["rms2_g"].data_ptr(), self.x.data_ptr(),
+                            sum_out=self.h.data_ptr(), M=B, N=Hd,
+                            stream=s, sync=False)
+            ops.gemm_bt(0, self.x.data_ptr(), lay["gate_w"].data_ptr(),
+                        self.ff.data_ptr(), M=B, N=self.inter, K=Hd,
+                        epi=self._epi_none, stream=s, sync=False, tile=T4)
+            ops.gemm_bt(0, self.x.data_ptr(), lay["up_w"].data_ptr(),
+                        self.ff2.data_ptr(), M=B, N=self.inter, K=Hd,
+                        epi=self._epi_none, stream=s, sync=False, tile=T4)
+            ops.silu_mul(0, self.ff.data_ptr(), self.ff2.data_ptr(),
+                         self.ff.data_ptr(), B * self.inter, stream=s,
+                         sync=False)
+            ops.gemm_bt(0, self.ff.data_ptr(), lay["down_w"].data_ptr(),
+                        self.x2.data_ptr(), M=B, N=Hd, K=self.inter,
+                        epi=self._epi_none, stream=s, sync=False, tile=T4)
+            nxt = (self.layers[li + 1] if li + 1 < self.n_layers else None)
+            gptr = (nxt["rms1_g"] if nxt else self.lnf_g).data_ptr()
+            dst = (self.x if nxt else self.out).data_ptr()
+            ops.add_rmsnorm(0, self.x2.data_ptr(), self.h.data_ptr(), gptr,
+                            dst, sum_out=self.h.data_ptr(), M=B, N=Hd,
+                            stream=s, sync=False)
+        if self.logits is not None:
+            ops.gemm_bt(0, self.out.data_ptr(), self.tok.data_ptr(),
+                        self.logits.data_ptr(), M=B, N=self.vocab, K=Hd,
+                        epi=self._epi_none, stream=s, sync=False)
+        ops.advance_pos(self.pos.data_ptr(), B, self.smax, stream=s,
+                        sync=False)
+
     def _enqueue(self):
+        if getattr(self, "arch", "gpt2") == "llama":
+            self._enqueue_llama()
+            return
         if getattr(self, "fused", False):
             self._enqueue_fused()
             return
@@ -331,35 +458,7 @@ class DecodeSession:
                         self.qkv.data_ptr(), bias=lay["qkv_b"].data_ptr(),
                         M=B, N=3 * Hd, K=Hd, epi=self._epi_bias, stream=s,
                         sync=False, tile=T4)
-            if self.kv_pool is not None:
-                pool = self.kv_pool
-                ops.kv_append_paged(self.qkv.data_ptr(),
-                                    pool.kpools[li].data_ptr(),
-                                    pool.vpools[li].data_ptr(),
-                                    pool.table.data_ptr(),
-                                    self.pos.data_ptr(), B, self.heads,
-                                    pool.max_pages, stream=s, sync=False,
-                                    D=self.hd)
-                ops.decode_attention_paged(
-                    self.qkv.data_ptr(), pool.kpools[li].data_ptr(),
-                    pool.vpools[li].data_ptr(), self.att.data_ptr(),
-                    pool.table.data_ptr(), self.pos.data_ptr(), B,
-                    self.heads, pool.max_pages,
-                    1.0 / float(np.sqrt(float(self.hd))), stream=s,
-                    sync=False, D=self.hd)
-            else:
-                ops.kv_append(self.qkv.data_ptr(), lay["kcache"].data_ptr(),
-                              lay["vcache"].data_ptr(), self.pos.data_ptr(),
-                              B, self.heads, self.smax, stream=s, sync=False,
-                              D=self.hd)
-                ops.decode_attention(self.qkv.data_ptr(),
-                                     lay["kcache"].data_ptr(),
-                                     lay["vcache"].data_ptr(),
-                                     self.att.data_ptr(),
-                                     self.pos.data_ptr(), B,
-                                     self.heads, self.smax,
-                                     1.0 / float(np.sqrt(float(self.hd))),
-                                     stream=s, sync=False, D=self.hd)
+            self._kv_attn(li, lay)
             ops.gemm_bt(0, self.att.data_ptr(), lay["proj_w"].data_ptr(),
                         self.x2.data_ptr(), bias=lay["proj_b"].data_ptr(),
                         M=B, N=Hd, K=Hd, epi=self._epi_bias, stream=s,
@@ -414,6 +513,11 @@ class DecodeSession:
         B, P = prompt.shape
         assert B == self.batch and 0 < P < self.smax
         Pp = (P + 127) // 128 * 128
+        if Pp > self.posemb.shape[0]:
+            raise RuntimeError(
+                f"prefill: padded prompt ({Pp}) exceeds the position table "
+                f"({int(self.posemb.shape[0])}) — shorten the prompt or "
+                "build the model with a longer seq")
         ids = np.zeros((B, Pp), np.int32)
         ids[:, :P] = prompt
         M = B * Pp
@@ -426,48 +530,101 @@ class DecodeSession:
         x2 = torch.empty(M, Hd, dtype=torch.half, device="cuda")
         qkv = torch.empty(M, 3 * Hd, dtype=torch.half, device="cuda")
         ff = torch.empty(M, inter, dtype=torch.half, device="cuda")
+        ff2 = (torch.empty(M, inter, dtype=torch.half, device="cuda")
+               if self.arch == "llama" else None)
         torch.cuda.synchronize()
 
+        scale = 1.0 / float(np.sqrt(float(self.hd)))
         ops.embedding(0, dids.data_ptr(), self.tok.data_ptr(),
                       self.posemb.data_ptr(), out=h.data_ptr(), M=M, S=Pp,
                       H=Hd, stream=s, sync=False)
-        ops.layernorm(0, h.data_ptr(), self.layers[0]["ln1_g"].data_ptr(),
-                      self.layers[0]["ln1_b"].data_ptr(), x.data_ptr(), M,
-                      Hd, stream=s, sync=False)
-        for li, lay in enumerate(self.layers):
-            ops.gemm_bt(0, x.data_ptr(), lay["qkv_w"].data_ptr(),
-                        qkv.data_ptr(), bias=lay["qkv_b"].data_ptr(), M=M,
-                        N=3 * Hd, K=Hd, epi=self._epi_bias, stream=s,
-                        sync=False)
-            ops.kv_append_range(qkv.data_ptr(), lay["kcache"].data_ptr(),
-                                lay["vcache"].data_ptr(), B, self.heads, Pp,
-                                self.smax, stream=s, sync=False, D=self.hd)
-            ops.attention(0, qkv.data_ptr(), x2.data_ptr(), B, Pp,
-                          self.heads, self.hd,
-                          1.0 / float(np.sqrt(float(self.hd))),
-                          stream=s, sync=False, causal=1)
-            ops.gemm_bt(0, x2.data_ptr(), lay["proj_w"].data_ptr(),
-                        x.data_ptr(), bias=lay["proj_b"].data_ptr(), M=M,
-                        N=Hd, K=Hd, epi=self._epi_bias, stream=s, sync=False)
-            ops.add_layernorm(0, x.data_ptr(), h.data_ptr(),
-                              lay["ln2_g"].data_ptr(),
-                              lay["ln2_b"].data_ptr(), x2.data_ptr(),
-                              sum_out=h.data_ptr(), M=M, N=Hd, stream=s,
-                              sync=False)
-            ops.gemm_bt(0, x2.data_ptr(), lay["ff1_w"].data_ptr(),
-                        ff.data_ptr(), bias=lay["ff1_b"].data_ptr(), M=M,
-                        N=inter, K=Hd, epi=self._epi_gelu, stream=s,
-                        sync=False)
-            ops.gemm_bt(0, ff.data_ptr(), lay["ff2_w"].data_ptr(),
-                        x2.data_ptr(), bias=lay["ff2_b"].data_ptr(), M=M,
-                        N=Hd, K=inter, epi=self._epi_bias, stream=s,
-                        sync=False)
-            nxt = (self.layers[li + 1] if li + 1 < self.n_layers else None)
-            gptr = (nxt["ln1_g"] if nxt else self.lnf_g).data_ptr()
-            bptr = (nxt["ln1_b"] if nxt else self.lnf_b).data_ptr()
-            ops.add_layernorm(0, x2.data_ptr(), h.data_ptr(), gptr, bptr,
-                              x.data_ptr(), sum_out=h.data_ptr(), M=M, N=Hd,
-                              stream=s, sync=False)
+        if self.arch == "llama":
+            ops.rmsnorm(0, h.data_ptr(),
+                        self.layers[0]["rms1_g"].data_ptr(), x.data_ptr(),
+                        M, Hd, stream=s, sync=False)
+            for li, lay in enumerate(self.layers):
+                ops.gemm_bt(0, x.data_ptr(), lay["qkv_w"].data_ptr(),
+                            qkv.data_ptr(), M=M, N=3 * Hd, K=Hd,
+                            epi=self._epi_none, stream=s, sync=False)
+                # full-sequence rope: pos = row % Pp (exact for rows < P;
+                # padded tail is causal-masked / overwritten before read)
+                ops.rope(0, qkv.data_ptr(), M=M, S=Pp, H=self.heads,
+                         D=self.hd, theta=self.theta, stream=s, sync=False)
+                ops.kv_append_range(qkv.data_ptr(),
+                                    lay["kcache"].data_ptr(),
+                                    lay["vcache"].data_ptr(), B, self.heads,
+                                    Pp, self.smax, stream=s, sync=False,
+                                    D=self.hd)
+                ops.attention(0, qkv.data_ptr(), x2.data_ptr(), B, Pp,
+                              self.heads, self.hd, scale, stream=s,
+                              sync=False, causal=1)
+                ops.gemm_bt(0, x2.data_ptr(), lay["proj_w"].data_ptr(),
+                            x.data_ptr(), M=M, N=Hd, K=Hd,
+                            epi=self._epi_none, stream=s, sync=False)
+                ops.add_rmsnorm(0, x.data_ptr(), h.data_ptr(),
+                                lay["rms2_g"].data_ptr(), x2.data_ptr(),
+                                sum_out=h.data_ptr(), M=M, N=Hd, stream=s,
+                                sync=False)
+                ops.gemm_bt(0, x2.data_ptr(), lay["gate_w"].data_ptr(),
+                            ff.data_ptr(), M=M, N=inter, K=Hd,
+                            epi=self._epi_none, stream=s, sync=False)
+                ops.gemm_bt(0, x2.data_ptr(), lay["up_w"].data_ptr(),
+                            ff2.data_ptr(), M=M, N=inter, K=Hd,
+                            epi=self._epi_none, stream=s, sync=False)
+                ops.silu_mul(0, ff.data_ptr(), ff2.data_ptr(),
+                             ff.data_ptr(), M * inter, stream=s, sync=False)
+                ops.gemm_bt(0, ff.data_ptr(), lay["down_w"].data_ptr(),
+                            x2.data_ptr(), M=M, N=Hd, K=inter,
+                            epi=self._epi_none, stream=s, sync=False)
+                nxt = (self.layers[li + 1] if li + 1 < self.n_layers
+                       else None)
+                gptr = (nxt["rms1_g"] if nxt else self.lnf_g).data_ptr()
+                ops.add_rmsnorm(0, x2.data_ptr(), h.data_ptr(), gptr,
+                                x.data_ptr(), sum_out=h.data_ptr(), M=M,
+                                N=Hd, stream=s, sync=False)
+        else:
+            ops.layernorm(0, h.data_ptr(),
+                          self.layers[0]["ln1_g"].data_ptr(),
+                          self.layers[0]["ln1_b"].data_ptr(), x.data_ptr(),
+                          M, Hd, stream=s, sync=False)
+            for li, lay in enumerate(self.layers):
+                ops.gemm_bt(0, x.data_ptr(), lay["qkv_w"].data_ptr(),
+                            qkv.data_ptr(), bias=lay["qkv_b"].data_ptr(),
+                            M=M, N=3 * Hd, K=Hd, epi=self._epi_bias,
+                            stream=s, sync=False)
+                ops.kv_append_range(qkv.data_ptr(),
+                                    lay["kcache"].data_ptr(),
+                                    lay["vcache"].data_ptr(), B, self.heads,
+                                    Pp, self.smax, stream=s, sync=False,
+                                    D=self.hd)
+                ops.attention(0, qkv.data_ptr(), x2.data_ptr(), B, Pp,
+                              self.heads, self.hd, scale,
+                              stream=s, sync=False, causal=1)
+                ops.gemm_bt(0, x2.data_ptr(), lay["proj_w"].data_ptr(),
+                            x.data_ptr(), bias=lay["proj_b"].data_ptr(),
+                            M=M, N=Hd, K=Hd, epi=self._epi_bias, stream=s,
+                            sync=False)
+                ops.add_layernorm(0, x.data_ptr(), h.data_ptr(),
+                                  lay["ln2_g"].data_ptr(),
+                                  lay["ln2_b"].data_ptr(), x2.data_ptr(),
+                                  sum_out=h.data_ptr(), M=M, N=Hd, stream=s,
+                                  sync=False)
+                ops.gemm_bt(0, x2.data_ptr(), lay["ff1_w"].data_ptr(),
+                            ff.data_ptr(), bias=lay["ff1_b"].data_ptr(),
+                            M=M, N=inter, K=Hd, epi=self._epi_gelu,
+                            stream=s, sync=False)
+                ops.gemm_bt(0, ff.data_ptr(), lay["ff2_w"].data_ptr(),
+                            x2.data_ptr(), bias=lay["ff2_b"].data_ptr(),
+                            M=M, N=Hd, K=inter, epi=self._epi_bias,
+                            stream=s, sync=False)
+                nxt = (self.layers[li + 1] if li + 1 < self.n_layers
+                       else None)
+                gptr = (nxt["ln1_g"] if nxt else self.lnf_g).data_ptr()
+                bptr = (nxt["ln1_b"] if nxt else self.lnf_b).data_ptr()
+                ops.add_layernorm(0, x2.data_ptr(), h.data_ptr(), gptr,
+                                  bptr, x.data_ptr(),
+                                  sum_out=h.data_ptr(), M=M, N=Hd,
+                                  stream=s, sync=False)
         self._C.hip.stream_synchronize(s)
         # x holds ln_f(h) for every position; hand back the last real one
         last = x.reshape(B, Pp, Hd)[:, P - 1].contiguous()
@@ -559,61 +716,128 @@ class DecodeSession:
                 qkv=torch.zeros(M, 3 * Hd, dtype=torch.half, device="cuda"),
                 att=torch.zeros(M, Hd, dtype=torch.half, device="cuda"),
                 ff=torch.zeros(M, inter, dtype=torch.half, device="cuda"),
+                ff2=(torch.zeros(M, inter, dtype=torch.half, device="cuda")
+                     if self.arch == "llama" else None),
                 logits=torch.zeros(M, self.vocab, dtype=torch.half,
                                    device="cuda"),
             )
             self._chunk_bufs = cb
         self._C.memory.memcpy_h2d(cb["ids"].data_ptr(), tokens.reshape(-1),
                                   tokens.nbytes)
+        scale = 1.0 / float(np.sqrt(float(self.hd)))
         ops.chunk_embed(cb["ids"].data_ptr(), self.tok.data_ptr(),
                         self.posemb.data_ptr(), cb["h"].data_ptr(),
                         self.pos.data_ptr(), B, K, self.smax, Hd, stream=s,
                         sync=False)
-        ops.layernorm(0, cb["h"].data_ptr(),
-                      self.layers[0]["ln1_g"].data_ptr(),
-                      self.layers[0]["ln1_b"].data_ptr(),
-                      cb["x"].data_ptr(), M, Hd, stream=s, sync=False)
-        for li, lay in enumerate(self.layers):
-            ops.gemm_bt(0, cb["x"].data_ptr(), lay["qkv_w"].data_ptr(),
-                        cb["qkv"].data_ptr(), bias=lay["qkv_b"].data_ptr(),
-                        M=M, N=3 * Hd, K=Hd, epi=self._epi_bias, stream=s,
-                        sync=False)
-            ops.kv_append_chunk(cb["qkv"].data_ptr(),
-                                lay["kcache"].data_ptr(),
-                                lay["vcache"].data_ptr(),
-                                self.pos.data_ptr(), B, self.heads, K,
-                                self.smax, stream=s, sync=False, D=self.hd)
-            ops.chunk_attention(cb["qkv"].data_ptr(),
-                                lay["kcache"].data_ptr(),
-                                lay["vcache"].data_ptr(),
-                                cb["att"].data_ptr(), self.pos.data_ptr(),
-                                B, self.heads, K, self.smax,
-                                1.0 / float(np.sqrt(float(self.hd))),
-                                stream=s, sync=False, D=self.hd)
-            ops.gemm_bt(0, cb["att"].data_ptr(), lay["proj_w"].data_ptr(),
-                        cb["x2"].data_ptr(), bias=lay["proj_b"].data_ptr(),
-                        M=M, N=Hd, K=Hd, epi=self._epi_bias, stream=s,
-                        sync=False)
-            ops.add_layernorm(0, cb["x2"].data_ptr(), cb["h"].data_ptr(),
-                              lay["ln2_g"].data_ptr(),
-                              lay["ln2_b"].data_ptr(), cb["x"].data_ptr(),
-                              sum_out=cb["h"].data_ptr(), M=M, N=Hd,
-                              stream=s, sync=False)
-            ops.gemm_bt(0, cb["x"].data_ptr(), lay["ff1_w"].data_ptr(),
-                        cb["ff"].data_ptr(), bias=lay["ff1_b"].data_ptr(),
-                        M=M, N=inter, K=Hd, epi=self._epi_gelu, stream=s,
-                        sync=False)
-            ops.gemm_bt(0, cb["ff"].data_ptr(), lay["ff2_w"].data_ptr(),
-                        cb["x2"].data_ptr(), bias=lay["ff2_b"].data_ptr(),
-                        M=M, N=Hd, K=inter, epi=self._epi_bias, stream=s,
-                        sync=False)
-            nxt = (self.layers[li + 1] if li + 1 < self.n_layers else None)
-            gptr = (nxt["ln1_g"] if nxt else self.lnf_g).data_ptr()
-            bptr = (nxt["ln1_b"] if nxt else self.lnf_b).data_ptr()
-            ops.add_layernorm(0, cb["x2"].data_ptr(), cb["h"].data_ptr(),
-                              gptr, bptr, cb["x"].data_ptr(),
-                              sum_out=cb["h"].data_ptr(), M=M, N=Hd,
-                              stream=s, sync=False)
+        if self.arch == "llama":
+            ops.rmsnorm(0, cb["h"].data_ptr(),
+                        self.layers[0]["rms1_g"].data_ptr(),
+                        cb["x"].data_ptr(), M, Hd, stream=s, sync=False)
+            for li, lay in enumerate(self.layers):
+                ops.gemm_bt(0, cb["x"].data_ptr(), lay["qkv_w"].data_ptr(),
+                            cb["qkv"].data_ptr(), M=M, N=3 * Hd, K=Hd,
+                            epi=self._epi_none, stream=s, sync=False)
+                # chunk rope: row b*K+i rotates at pos[b] + i
+                ops.rope(0, cb["qkv"].data_ptr(), pos=self.pos.data_ptr(),
+                         M=M, S=self.smax, H=self.heads, D=self.hd,
+                         theta=self.theta, stream=s, sync=False, chunk=K)
+                ops.kv_append_chunk(cb["qkv"].data_ptr(),
+                                    lay["kcache"].data_ptr(),
+                                    lay["vcache"].data_ptr(),
+                                    self.pos.data_ptr(), B, self.heads, K,
+                                    self.smax, stream=s, sync=False,
+                                    D=self.hd)
+                ops.chunk_attention(cb["qkv"].data_ptr(),
+                                    lay["kcache"].data_ptr(),
+                                    lay["vcache"].data_ptr(),
+                                    cb["att"].data_ptr(),
+                                    self.pos.data_ptr(), B, self.heads, K,
+                                    self.smax, scale, stream=s, sync=False,
+                                    D=self.hd)
+                ops.gemm_bt(0, cb["att"].data_ptr(),
+                            lay["proj_w"].data_ptr(), cb["x2"].data_ptr(),
+                            M=M, N=Hd, K=Hd, epi=self._epi_none, stream=s,
+                            sync=False)
+                ops.add_rmsnorm(0, cb["x2"].data_ptr(), cb["h"].data_ptr(),
+                                lay["rms2_g"].data_ptr(),
+                                cb["x"].data_ptr(),
+                                sum_out=cb["h"].data_ptr(), M=M, N=Hd,
+                                stream=s, sync=False)
+                ops.gemm_bt(0, cb["x"].data_ptr(), lay["gate_w"].data_ptr(),
+                            cb["ff"].data_ptr(), M=M, N=inter, K=Hd,
+                            epi=self._epi_none, stream=s, sync=False)
+                ops.gemm_bt(0, cb["x"].data_ptr(), lay["up_w"].data_ptr(),
+                            cb["ff2"].data_ptr(), M=M, N=inter, K=Hd,
+                            epi=self._epi_none, stream=s, sync=False)
+                ops.silu_mul(0, cb["ff"].data_ptr(), cb["ff2"].data_ptr(),
+                             cb["ff"].data_ptr(), M * inter, stream=s,
+                             sync=False)
+                ops.gemm_bt(0, cb["ff"].data_ptr(),
+                            lay["down_w"].data_ptr(), cb["x2"].data_ptr(),
+                            M=M, N=Hd, K=inter, epi=self._epi_none,
+                            stream=s, sync=False)
+                nxt = (self.layers[li + 1] if li + 1 < self.n_layers
+                       else None)
+                gptr = (nxt["rms1_g"] if nxt else self.lnf_g).data_ptr()
+                ops.add_rmsnorm(0, cb["x2"].data_ptr(), cb["h"].data_ptr(),
+                                gptr, cb["x"].data_ptr(),
+                                sum_out=cb["h"].data_ptr(), M=M, N=Hd,
+                                stream=s, sync=False)
+        else:
+            ops.layernorm(0, cb["h"].data_ptr(),
+                          self.layers[0]["ln1_g"].data_ptr(),
+                          self.layers[0]["ln1_b"].data_ptr(),
+                          cb["x"].data_ptr(), M, Hd, stream=s, sync=False)
+            for li, lay in enumerate(self.layers):
+                ops.gemm_bt(0, cb["x"].data_ptr(), lay["qkv_w"].data_ptr(),
+                            cb["qkv"].data_ptr(),
+                            bias=lay["qkv_b"].data_ptr(),
+                            M=M, N=3 * Hd, K=Hd, epi=self._epi_bias,
+                            stream=s, sync=False)
+                ops.kv_append_chunk(cb["qkv"].data_ptr(),
+                                    lay["kcache"].data_ptr(),
+                                    lay["vcache"].data_ptr(),
+                                    self.pos.data_ptr(), B, self.heads, K,
+                                    self.smax, stream=s, sync=False,
+                                    D=self.hd)
+                ops.chunk_attention(cb["qkv"].data_ptr(),
+                                    lay["kcache"].data_ptr(),
+                                    lay["vcache"].data_ptr(),
+                                    cb["att"].data_ptr(),
+                                    self.pos.data_ptr(),
+                                    B, self.heads, K, self.smax, scale,
+                                    stream=s, sync=False, D=self.hd)
+                ops.gemm_bt(0, cb["att"].data_ptr(),
+                            lay["proj_w"].data_ptr(), cb["x2"].data_ptr(),
+                            bias=lay["proj_b"].data_ptr(),
+                            M=M, N=Hd, K=Hd, epi=self._epi_bias, stream=s,
+                            sync=False)
+                ops.add_layernorm(0, cb["x2"].data_ptr(),
+                                  cb["h"].data_ptr(),
+                                  lay["ln2_g"].data_ptr(),
+                                  lay["ln2_b"].data_ptr(),
+                                  cb["x"].data_ptr(),
+                                  sum_out=cb["h"].data_ptr(), M=M, N=Hd,
+                                  stream=s, sync=False)
+                ops.gemm_bt(0, cb["x"].data_ptr(), lay["ff1_w"].data_ptr(),
+                            cb["ff"].data_ptr(),
+                            bias=lay["ff1_b"].data_ptr(),
+                            M=M, N=inter, K=Hd, epi=self._epi_gelu,
+                            stream=s, sync=False)
+                ops.gemm_bt(0, cb["ff"].data_ptr(), lay["ff2_w"].data_ptr(),
+                            cb["x2"].data_ptr(),
+                            bias=lay["ff2_b"].data_ptr(),
+                            M=M, N=Hd, K=inter, epi=self._epi_bias,
+                            stream=s, sync=False)
+                nxt = (self.layers[li + 1] if li + 1 < self.n_layers
+                       else None)
+                gptr = (nxt["ln1_g"] if nxt else self.lnf_g).data_ptr()
+                bptr = (nxt["ln1_b"] if nxt else self.lnf_b).data_ptr()
+                ops.add_layernorm(0, cb["x2"].data_ptr(),
+                                  cb["h"].data_ptr(),
+                                  gptr, bptr, cb["x"].data_ptr(),
+                                  sum_out=cb["h"].data_ptr(), M=M, N=Hd,
+                                  stream=s, sync=False)
         ops.gemm_bt(0, cb["x"].data_ptr(), self.tok.data_ptr(),
                     cb["logits"].data_ptr(), M=M, N=self.vocab, K=Hd,
                     epi=self._epi_none, stream=s, sync=False)
