@@ -366,6 +366,23 @@ PYBIND11_MODULE(_C, m) {
           py::arg("sw") = 1, py::arg("ph") = 0, py::arg("pw") = 0,
           py::arg("epi") = 0, py::arg("stream") = 0, py::arg("sync") = true,
           py::arg("tile") = 0, py::arg("res_scale") = 1.0f);
+  ops.def("bottleneck_tail",
+          [](int dtype, uintptr_t in, uintptr_t W1, uintptr_t W2,
+             uintptr_t out, uintptr_t s1, uintptr_t b1, uintptr_t s2,
+             uintptr_t b2, uintptr_t residual, uintptr_t zero_page, int Nb,
+             int H, int W, int Cm, int Co, uintptr_t stream, bool sync) {
+            launch_bottleneck_tail(dtype, (void*)in, (void*)W1, (void*)W2,
+                                   (void*)out, (float*)s1, (float*)b1,
+                                   (float*)s2, (float*)b2, (void*)residual,
+                                   (void*)zero_page, Nb, H, W, Cm, Co,
+                                   as_stream(stream));
+            if (sync) TRT_HIP_CHECK(hipStreamSynchronize(as_stream(stream)));
+          },
+          py::arg("dtype"), py::arg("in"), py::arg("W1"), py::arg("W2"),
+          py::arg("out"), py::arg("s1"), py::arg("b1"), py::arg("s2"),
+          py::arg("b2"), py::arg("residual"), py::arg("zero_page"),
+          py::arg("Nb"), py::arg("H"), py::arg("W"), py::arg("Cm"),
+          py::arg("Co"), py::arg("stream") = 0, py::arg("sync") = true);
   ops.def("maxpool2d",
           [](int dtype, uintptr_t in, uintptr_t out, int Nb, int H, int W,
              int C, int KH, int KW, int sh, int sw, int ph, int pw,

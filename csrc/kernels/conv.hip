@@ -235,6 +235,206 @@ __global__ __launch_bounds__(256) void conv_smallk_kernel(
   }
 }
 
+// ---- fused bottleneck tail: conv3x3+BN+ReLU -> conv1x1+BN+res+ReLU ----
+// The ResNet bottleneck interior (3x3 s1 p1, width CM) immediately feeds a
+// 1x1 expand (CM -> Co) whose only input is the 3x3 output. One workgroup
+// computes the ENTIRE channel extent (BN1 = CM) of the 3x3 for a 64-row
+// spatial tile, applies BN+ReLU, re-stages the tile T[64][CM] into LDS in
+// the standard swizzled A-tile layout, and runs the 1x1 GEMM (K = CM) from
+// LDS — the intermediate tensor never touches HBM and one kernel launch
+// (~10-15 us at these latency-bound b8 shapes) disappears per pair.
+// CM in {64, 128} (ResNet-50 stages 1-2; wider stages would overflow the
+// 64 KB static-LDS T tile at full width). fp16 only (the headline dtype).
+template <typename T, int CM, bool DEEP, int BM = 64>
+__global__ __launch_bounds__(256) void bottleneck_tail_kernel(
+    const T* __restrict__ in, const T* __restrict__ W1,
+    const T* __restrict__ W2, T* __restrict__ out,
+    const float* __restrict__ s1, const float* __restrict__ b1,
+    const float* __restrict__ s2, const float* __restrict__ b2,
+    const T* __restrict__ residual, const T* __restrict__ zero_page,
+    const ConvParams p, int Co) {
+  constexpr int BNC = 64;                      // GEMM1 column chunk
+  constexpr int NB1 = CM / BNC;                // chunks across the 3x3 width
+  constexpr int kABytes = BM * 128;
+  constexpr int kBuf = (BM + BNC) * 128;       // one GEMM1 staging slot
+  constexpr int NBUF = DEEP ? 4 : 2;
+  constexpr int kT2 = CM / 64;                 // GEMM2 K-tiles
+  constexpr int kTBlk = BM * 128;              // one T K-tile block
+  constexpr int kTBytes = kT2 * kTBlk;         // T tile (BM x CM fp16)
+  constexpr int kSmem =
+      (NBUF * kBuf > kTBytes + 8192) ? NBUF * kBuf : kTBytes + 8192;
+  __shared__ __attribute__((aligned(16))) char smem[kSmem];
+  uint32_t lds0 = (uint32_t)(uintptr_t)&smem[0];
+
+  uint32_t tile = xcd_swizzle(blockIdx.x, gridDim.x);
+  int m0 = (int)tile * BM;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wr = wave >> 1;
+  const int wc = wave & 1;
+
+  // ---- GEMM1: T = relu(bn1(conv3x3(in))), 64-col chunks so the staging
+  // slot stays 16 KB and the 4-deep counted pipeline fits 64 KB LDS (the
+  // grid-starved stage-2 shape needs it; accumulators for every chunk
+  // stay live in VGPRs until the shared T-write below) ----
+  typename Mfma16x16x32<T>::accv acc1[NB1][BM / 32][BNC / 32];
+#pragma unroll
+  for (int c = 0; c < NB1; ++c)
+#pragma unroll
+    for (int i = 0; i < BM / 32; ++i)
+#pragma unroll
+      for (int j = 0; j < BNC / 32; ++j) acc1[c][i][j] = {0, 0, 0, 0};
+
+  const int ktiles = p.K / kTileElems<T>;
+  for (int nc = 0; nc < NB1; ++nc) {
+    auto stage1 = [&](int t, int slot) {
+      uint32_t base = lds0 + slot * kBuf;
+      stage_conv_a<T, BM>(in, zero_page, p, m0, t * kTileElems<T>, base,
+                          tid);
+      stage_tile<T, BNC>(W1 + (int64_t)nc * BNC * p.K +
+                             (int64_t)t * kTileElems<T>,
+                         p.K, nc * BNC, CM, base + kABytes, tid);
+    };
+    if constexpr (DEEP) {
+      constexpr int G = BM / 32 + BNC / 32;
+      for (int i = 0; i < 3 && i < ktiles; ++i) stage1(i, i);
+      for (int t = 0; t < ktiles; ++t) {
+        int ahead = ktiles - 1 - t;
+        if (ahead > 2) ahead = 2;
+        wait_tiles_inflight<G>(ahead);
+        __builtin_amdgcn_s_barrier();
+        if (t + 3 < ktiles) stage1(t + 3, (t + 3) & 3);
+        const char* As = &smem[(t & 3) * kBuf];
+        mfma_tile<T, BM, BNC>(As, As + kABytes, lane, wr, wc, acc1[nc]);
+      }
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      __syncthreads();
+    } else {
+      stage1(0, 0);
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      __syncthreads();
+      int cur = 0;
+      for (int t = 0; t < ktiles; ++t) {
+        if (t + 1 < ktiles) stage1(t + 1, cur ^ 1);
+        const char* As = &smem[cur * kBuf];
+        mfma_tile<T, BM, BNC>(As, As + kABytes, lane, wr, wc, acc1[nc]);
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+        __syncthreads();
+        cur ^= 1;
+      }
+    }
+  }
+
+  // ---- T -> LDS in swizzled A-tile layout (GEMM1 LDS slots are dead) ----
+#pragma unroll
+  for (int nc = 0; nc < NB1; ++nc) {
+#pragma unroll
+    for (int i = 0; i < BM / 32; ++i) {
+#pragma unroll
+      for (int j = 0; j < BNC / 32; ++j) {
+        int col = nc * BNC + wc * (BNC / 2) + j * 16 + (lane & 15);
+        float sc = s1[col], bi = b1[col];
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          uint32_t row =
+              (uint32_t)(wr * (BM / 2) + i * 16 + ((lane >> 4) << 2) + r);
+          float v = fmaxf((float)acc1[nc][i][j][r] * sc + bi, 0.0f);
+          uint32_t kb = (uint32_t)(col & 63) * (uint32_t)sizeof(T);
+          uint32_t off = (uint32_t)(col >> 6) * (uint32_t)kTBlk +
+                         (row << 7) + (kb ^ ((row & 7) << 4));
+          *(T*)(smem + off) = (T)v;
+        }
+      }
+    }
+  }
+  __syncthreads();
+
+  // ---- GEMM2: out = relu(bn2(T @ W2^T) + residual), N2 chunks of 64 ----
+  const char* Tlds = smem;
+  char* B2s = smem + kTBytes;
+  for (int c2 = 0; c2 < Co / 64; ++c2) {
+#pragma unroll
+    for (int t = 0; t < kT2; ++t)
+      stage_tile<T, 64>(W2 + (int64_t)c2 * 64 * CM + t * kTileElems<T>, CM,
+                        c2 * 64, Co, (uint32_t)(uintptr_t)(B2s + t * 8192),
+                        tid);
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+    typename Mfma16x16x32<T>::accv acc2[BM / 32][2];
+#pragma unroll
+    for (int i = 0; i < BM / 32; ++i)
+#pragma unroll
+      for (int j = 0; j < 2; ++j) acc2[i][j] = {0, 0, 0, 0};
+#pragma unroll
+    for (int t = 0; t < kT2; ++t)
+      mfma_tile<T, BM, 64>(Tlds + t * kTBlk, B2s + t * 8192, lane, wr, wc,
+                           acc2);
+    store_epilogue<T, Epi::kScaleBiasAddRelu, BM, 64>(
+        acc2, out, Co, m0, c2 * 64, p.M, Co, s2, b2, residual, p.res_scale,
+        lane, wr, wc);
+    __syncthreads();  // before the next chunk overwrites B2s
+  }
+}
+
+void launch_bottleneck_tail(int dtype, const void* in, const void* W1,
+                            const void* W2, void* out, const float* s1,
+                            const float* b1, const float* s2, const float* b2,
+                            const void* residual, const void* zero_page,
+                            int Nb, int H, int W, int Cm, int Co,
+                            hipStream_t stream) {
+  if (dtype != 0)
+    throw std::runtime_error("bottleneck_tail: fp16 only");
+  if (Cm != 64 && Cm != 128)
+    throw std::runtime_error("bottleneck_tail: Cm must be 64 or 128");
+  if (Co % 64 != 0) throw std::runtime_error("bottleneck_tail: Co % 64");
+  ConvParams p;
+  p.res_scale = 1.0f;
+  p.Nb = Nb; p.H = H; p.W = W; p.C = Cm;
+  p.Cout = Cm; p.KH = 3; p.KW = 3;
+  p.sh = 1; p.sw = 1; p.ph = 1; p.pw = 1;
+  p.OH = H; p.OW = W;
+  p.M = Nb * H * W;
+  p.Kreal = 9 * Cm;
+  p.K = p.Kreal;  // 9*Cm is a multiple of 64 for Cm in {64,128}
+  p.d_ohw = make_fastdiv((uint32_t)(p.OH * p.OW));
+  p.d_ow = make_fastdiv((uint32_t)p.OW);
+  p.d_c = make_fastdiv((uint32_t)Cm);
+  p.d_kw = make_fastdiv((uint32_t)p.KW);
+  // grid-starved shapes (stage 2: 98 WGs at BM=64) halve the M-tile to
+  // double the workgroup count; the deep pipeline follows the same gate
+  // as the standalone conv.
+  int bm = (cdiv(p.M, 64) >= 256) ? 64 : 32;
+  dim3 grid((unsigned)cdiv(p.M, bm));
+  bool deep = want_deep_pipe(grid.x, p.K / 64);
+  auto go = [&](auto cm, auto dp, auto bmv) {
+    constexpr int CMv = decltype(cm)::value;
+    constexpr bool DPv = decltype(dp)::value;
+    constexpr int BMv = decltype(bmv)::value;
+    hipLaunchKernelGGL((bottleneck_tail_kernel<_Float16, CMv, DPv, BMv>),
+                       grid, dim3(256), 0, stream, (const _Float16*)in,
+                       (const _Float16*)W1, (const _Float16*)W2,
+                       (_Float16*)out, s1, b1, s2, b2,
+                       (const _Float16*)residual, (const _Float16*)zero_page,
+                       p, Co);
+  };
+  auto c64 = std::integral_constant<int, 64>{};
+  auto c128 = std::integral_constant<int, 128>{};
+  auto b64 = std::integral_constant<int, 64>{};
+  auto b32 = std::integral_constant<int, 32>{};
+  if (Cm == 64) {
+    if (bm == 64) { deep ? go(c64, std::true_type{}, b64)
+                         : go(c64, std::false_type{}, b64); }
+    else { deep ? go(c64, std::true_type{}, b32)
+                : go(c64, std::false_type{}, b32); }
+  } else {
+    if (bm == 64) { deep ? go(c128, std::true_type{}, b64)
+                         : go(c128, std::false_type{}, b64); }
+    else { deep ? go(c128, std::true_type{}, b32)
+                : go(c128, std::false_type{}, b32); }
+  }
+}
+
 template <typename T>
 static void launch_conv2d_t(const void* in, const void* Wt, void* out,
                             const float* scale, const float* bias,

@@ -65,6 +65,15 @@ void launch_silu_mul(int dtype, const void* a, const void* b, void* out,
 void launch_rope(int dtype, void* qkv, const void* pos_dev, int M, int S,
                  int H, int D, float theta, hipStream_t stream,
                  int chunk = 0);
+// Fused ResNet bottleneck tail: conv3x3(s1,p1,Cm)+BN+ReLU feeding
+// conv1x1(Cm->Co)+BN+residual+ReLU in ONE kernel (intermediate stays in
+// LDS). Cm in {64,128}; fp16 only.
+void launch_bottleneck_tail(int dtype, const void* in, const void* W1,
+                            const void* W2, void* out, const float* s1,
+                            const float* b1, const float* s2, const float* b2,
+                            const void* residual, const void* zero_page,
+                            int Nb, int H, int W, int Cm, int Co,
+                            hipStream_t stream);
 void launch_elementwise(int dtype, int op, const void* a, const void* b,
                         void* out, int64_t n, hipStream_t stream);
 void launch_clip(int dtype, const void* in, void* out, int64_t n, float mn,

@@ -173,6 +173,17 @@ void ExecutionContext::enqueue_all(hipStream_t s) {
                       op.KH, op.KW, op.sh, op.sw, op.ph, op.pw, op.epi, os,
                       op.tile, op.fork ? scratch2_ : scratch_, op.res_scale);
         break;
+      case kBtail:
+        // fused bottleneck tail: W1 at w_off, W2 at w2_off; the fp32
+        // scale/bias blobs hold [s1 | s2] / [b1 | b2] (s2 at + C floats)
+        launch_bottleneck_tail(op.dtype, A(op.in_off), Wp(op.w_off),
+                               Wp(op.w2_off), A(op.out_off),
+                               Fp(op.scale_off), Fp(op.bias_off),
+                               Fp(op.scale_off) + op.C,
+                               Fp(op.bias_off) + op.C, A(op.in2_off),
+                               eng_->zero_page(), op.Nb, op.H, op.W, op.C,
+                               op.Cout, os);
+        break;
       case kGemmBt:
         launch_gemm_bt(op.dtype, A(op.in_off), Wp(op.w_off), A(op.out_off),
                        Fp(op.scale_off), Fp(op.bias_off), A(op.in2_off),

@@ -120,6 +120,7 @@ enum OpKind : int {
   kRMSNorm = 22,     // LLaMA norm: x / rms(x) * gamma (no mean, no beta)
   kSiluMul = 23,     // SwiGLU gate: silu(a) * b
   kRope = 24,        // rotary embedding in-place on qkv q/k blocks
+  kBtail = 25,       // fused bottleneck tail: 3x3+BN+ReLU -> 1x1+BN+res+ReLU
 };
 
 struct OpDesc {

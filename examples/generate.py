@@ -1,11 +1,12 @@
 #!/usr/bin/env python3
-"""GPT-2-style greedy generation on the incremental decode path (GPU).
+"""Greedy generation on the incremental decode path (GPU) — GPT-2 or LLaMA.
 
 Random-init weights (no network for checkpoints), so the "text" is noise —
 this demonstrates the serving mechanics: one hipGraph replay per token
 against resident KV caches, with logits from the weight-tied lm head.
 
     python examples/generate.py --batch 4 --prompt-len 16 --new-tokens 32
+    python examples/generate.py --arch llama --hidden 1024 --layers 8
 """
 import argparse
 import os
@@ -17,12 +18,15 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import numpy as np
 
 from trtlab_amd.engine.decode import DecodeSession
-from trtlab_amd.models import build_gpt2
+from trtlab_amd.models import build_gpt2, build_llama
 
 
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--batch", type=int, default=4)
+    ap.add_argument("--arch", choices=("gpt2", "llama"), default="gpt2")
+    ap.add_argument("--hidden", type=int, default=0,
+                    help="hidden size (default: 768 gpt2 / 1024 llama)")
     ap.add_argument("--layers", type=int, default=12)
     ap.add_argument("--prompt-len", type=int, default=16)
     ap.add_argument("--new-tokens", type=int, default=32)
@@ -33,12 +37,21 @@ def main():
     ap.add_argument("--draft-layers", type=int, default=2)
     args = ap.parse_args()
 
-    g = build_gpt2(batch=args.batch, seq=1024, layers=args.layers, seed=0,
-                   embeddings=True)
+    def build(layers):
+        if args.arch == "llama":
+            hid = args.hidden or 1024
+            return build_llama(batch=args.batch, seq=1024, hidden=hid,
+                               layers=layers, heads=hid // 128, seed=0)
+        return build_gpt2(batch=args.batch, seq=1024, layers=layers,
+                          hidden=args.hidden or 768, seed=0,
+                          embeddings=True)
+
+    g = build(args.layers)
+    vocab = 32000 if args.arch == "llama" else 50257
     sess = DecodeSession(g, batch=args.batch, smax=1024, lm_head=True)
 
     rng = np.random.RandomState(0)
-    prompt = rng.randint(1, 50257,
+    prompt = rng.randint(1, vocab,
                          (args.batch, args.prompt_len)).astype(np.int32)
     logits = sess.prefill(prompt)  # one fused pass fills the KV caches
 
@@ -48,8 +61,7 @@ def main():
         # the target verifies each chunk in ONE forward
         from trtlab_amd.engine.decode import SpeculativeDecoder
 
-        gd = build_gpt2(batch=args.batch, seq=1024,
-                        layers=args.draft_layers, seed=0, embeddings=True)
+        gd = build(args.draft_layers)
         draft = DecodeSession(gd, batch=args.batch, smax=1024,
                               capture=False, lm_head=True)
         draft.prefill(prompt)

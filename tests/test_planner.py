@@ -21,17 +21,25 @@ def rn50_plan():
 def test_resnet50_fusion_counts(rn50_plan):
     g, plan = rn50_plan
     kinds = [d["kind"] for d in plan.ops]
-    assert kinds.count(K_CONV) == 53  # 1 stem + 16*3 bottleneck + 4 downsample
+    # 53 conv layers total (1 stem + 16*3 bottleneck + 4 downsample); the
+    # 3 stage-1 bottleneck tails (3x3 + 1x1-expand, width 64) fuse into
+    # K_BTAIL pairs, so the kernel inventory is 47 convs + 3 fused pairs
+    from trtlab_amd.engine.planner import K_BTAIL
+    assert kinds.count(K_BTAIL) == 3
+    assert kinds.count(K_CONV) + 2 * kinds.count(K_BTAIL) == 53
     assert kinds.count(K_CHANNEL_PAD) == 1
     assert kinds.count(K_GEMM) == 1
     # every bottleneck's last conv carries the fused residual-add + relu
+    # (13 standalone + 3 inside the fused tails)
     adds = [d for d in plan.ops if d.get("epi") == EPI_SCALE_BIAS_ADD_RELU]
-    assert len(adds) == 16
+    assert len(adds) + kinds.count(K_BTAIL) == 16
     # downsample convs are plain scale+bias
     plain = [d for d in plan.ops if d.get("epi") == EPI_SCALE_BIAS]
     assert len(plain) == 4
     relu = [d for d in plan.ops if d.get("epi") == EPI_SCALE_BIAS_RELU]
-    assert len(relu) == 33  # stem + 2 per bottleneck
+    # stem + 2 per bottleneck, minus the 3 BN+ReLU 3x3s absorbed into
+    # the fused stage-1 tails
+    assert len(relu) + kinds.count(K_BTAIL) == 33
 
 
 def test_conv_k_padding(rn50_plan):
@@ -256,7 +264,8 @@ def test_fork_join_downsample_marking():
     assert kinds == "fjfjfjfj"  # strictly alternating pairs
     for f, j in zip(forks, joins):
         assert j > f + 1  # there is work to overlap
+        from trtlab_amd.engine.planner import K_BTAIL
         assert plan.ops[f]["kind"] == K_CONV
-        assert plan.ops[j]["kind"] == K_CONV
+        assert plan.ops[j]["kind"] in (K_CONV, K_BTAIL)
         # the join consumes the forked output as its residual
         assert plan.ops[j]["in2_off"] == plan.ops[f]["out_off"]

@@ -16,7 +16,7 @@ import torch.nn.functional as F
 
 from trtlab_amd.engine.planner import (
     K_CLIP, K_COPY2D, K_TRANSPOSE2D, K_RMSNORM, K_SILU_MUL, K_ROPE,
-    EnginePlan, K_ADD_LAYERNORM, K_ATTENTION, K_AVGPOOL, K_CHANNEL_PAD, K_CONV,
+    EnginePlan, K_ADD_LAYERNORM, K_ATTENTION, K_AVGPOOL, K_BTAIL, K_CHANNEL_PAD, K_CONV,
     K_DEQUANT, K_ELEMENTWISE, K_GAVGPOOL, K_GEMM, K_LAYERNORM, K_MAXPOOL,
     K_EMBEDDING, K_GEMM_MX4, K_GEMM_MX8, K_QUANT_MX4, K_QUANT_MX8,
     K_QUANTIZE, K_SEQLENS, K_SOFTMAX,
@@ -93,6 +93,25 @@ def run_reference(plan: EnginePlan, input_nhwc, return_all: bool = False):
             elif d["dtype"] == 3:  # fp8: emulate the e4m3 output store
                 y = _fp8_round(y)
             t[op.output] = y
+        elif op.kind == K_BTAIL:
+            # fused bottleneck tail: conv3x3+BN+ReLU then 1x1+BN+res+ReLU
+            # (weights pre-packed: w = [64, 576] bt-flat, w2 = [Co, 64];
+            # scale/bias = [s1 | s2] / [b1 | b2])
+            nb, h, w, c = x.shape
+            wt = torch.from_numpy(op.w.astype(np.float32)) \
+                .reshape(c, 3, 3, c).permute(0, 3, 1, 2)
+            mid = F.conv2d(x.permute(0, 3, 1, 2), wt, stride=1,
+                           padding=1).permute(0, 2, 3, 1)
+            sc = torch.from_numpy(op.scale)
+            bi = torch.from_numpy(op.bias)
+            mid = torch.relu(mid * sc[:c] + bi[:c])
+            w2 = torch.from_numpy(
+                op.params["w2"].astype(np.float32))  # [Co, c]
+            co = d["Cout"]
+            acc = mid.reshape(-1, c) @ w2.t()
+            res = t[op.inputs[1]].reshape(-1, co)
+            y = torch.relu(acc * sc[c:] + bi[c:] + res)
+            t[op.output] = y.reshape(nb, h, w, co)
         elif op.kind == K_EMBEDDING:
             ids = x.long()
             tok = torch.from_numpy(op.w.astype(np.float32))
