@@ -5,6 +5,7 @@ import os
 import subprocess
 import sys
 import tempfile
+import time
 from pathlib import Path
 
 import numpy as np
@@ -140,3 +141,54 @@ def test_replica_group_least_outstanding_and_failover():
     g2.mark_unhealthy(0)
     with pytest.raises(RuntimeError):
         g2.next_index()
+
+
+def _tcp_rdzv_worker(rank, world, port, q):
+    import os
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    from trtlab_amd.parallel import exchange_unique_id_tcp
+
+    uid = bytes(range(128)) if rank == 0 else None
+    got = exchange_unique_id_tcp(rank, world, uid, addr="127.0.0.1",
+                                 port=port, timeout=30.0)
+    q.put((rank, got))
+
+
+def test_tcp_uid_rendezvous_three_ranks():
+    """Multi-node bring-up channel: rank 0 serves the 128-byte RCCL uid
+    over TCP (MASTER_PORT+1); peers connect with retry. Pure transport
+    test (no GPU): three processes, all must read rank 0's exact bytes —
+    including a peer that starts BEFORE the server is listening."""
+    import multiprocessing as mp
+
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = 29871
+    world = 3
+    # start a non-zero rank FIRST to exercise the connect-retry path
+    p1 = ctx.Process(target=_tcp_rdzv_worker, args=(1, world, port, q))
+    p1.start()
+    time.sleep(0.3)
+    p0 = ctx.Process(target=_tcp_rdzv_worker, args=(0, world, port, q))
+    p2 = ctx.Process(target=_tcp_rdzv_worker, args=(2, world, port, q))
+    p0.start()
+    p2.start()
+    got = {}
+    for _ in range(world):
+        r, uid = q.get(timeout=60)
+        got[r] = uid
+    for p in (p0, p1, p2):
+        p.join(timeout=30)
+        assert p.exitcode == 0
+    expect = bytes(range(128))
+    assert got == {0: expect, 1: expect, 2: expect}
+
+
+def test_tcp_uid_rendezvous_timeout():
+    """A peer with no rank 0 to reach fails with a clear TimeoutError."""
+    from trtlab_amd.parallel import exchange_unique_id_tcp
+
+    with pytest.raises(TimeoutError, match="rank 0"):
+        exchange_unique_id_tcp(1, 2, addr="127.0.0.1", port=29899,
+                               timeout=1.0)

@@ -76,6 +76,71 @@ def exchange_unique_id(rank: int, world: int, uid: Optional[bytes] = None,
     raise TimeoutError(f"RCCL rendezvous timed out waiting for {path}")
 
 
+def exchange_unique_id_tcp(rank: int, world: int,
+                           uid: Optional[bytes] = None,
+                           addr: Optional[str] = None,
+                           port: Optional[int] = None,
+                           timeout: float = 300.0) -> bytes:
+    """TCP unique-id rendezvous for MULTI-NODE launches (the file path
+    above needs a shared filesystem; this needs only the MASTER_ADDR
+    reachability torchrun already assumes). rank 0 serves the 128-byte
+    RCCL uid on (addr, port); every other rank connects with retry,
+    sends its rank (4 bytes, for logging/validation), and reads the uid.
+    Port defaults to MASTER_PORT+1 so torchrun's own TCPStore on
+    MASTER_PORT is not disturbed."""
+    import socket
+    import struct
+
+    addr = addr or os.environ.get("MASTER_ADDR", "127.0.0.1")
+    port = port or int(os.environ.get("MASTER_PORT", "29500")) + 1
+    deadline = time.monotonic() + timeout
+    if rank == 0:
+        if uid is None or len(uid) != _UID_BYTES:
+            raise ValueError("rank 0 must supply the 128-byte uid")
+        srv = socket.socket(socket.AF_INET, socket.SOCK_STREAM)
+        srv.setsockopt(socket.SOL_SOCKET, socket.SO_REUSEADDR, 1)
+        srv.bind(("0.0.0.0", port))
+        srv.listen(world)
+        srv.settimeout(1.0)
+        served: set = set()
+        try:
+            while len(served) < world - 1:
+                if time.monotonic() > deadline:
+                    raise TimeoutError(
+                        f"uid rendezvous: {len(served)}/{world - 1} peers "
+                        f"after {timeout:.0f}s")
+                try:
+                    conn, _ = srv.accept()
+                except socket.timeout:
+                    continue
+                with conn:
+                    conn.settimeout(10.0)
+                    peer = struct.unpack("!i", conn.recv(4))[0]
+                    conn.sendall(uid)
+                    served.add(peer)
+        finally:
+            srv.close()
+        return uid
+    last_err: Optional[Exception] = None
+    while time.monotonic() < deadline:
+        try:
+            with socket.create_connection((addr, port), timeout=5.0) as c:
+                c.sendall(struct.pack("!i", rank))
+                data = b""
+                while len(data) < _UID_BYTES:
+                    chunk = c.recv(_UID_BYTES - len(data))
+                    if not chunk:
+                        raise ConnectionError("short uid read")
+                    data += chunk
+                return data
+        except OSError as e:  # rank 0 not listening yet
+            last_err = e
+            time.sleep(0.2)
+    raise TimeoutError(
+        f"uid rendezvous: could not reach rank 0 at {addr}:{port} "
+        f"({last_err})")
+
+
 class Communicator:
     """Owned RCCL communicator (one per process; ranks may share a GPU —
     that is how the 2-rank proof runs within a 1-GPU lease).
@@ -88,10 +153,14 @@ class Communicator:
                  world: Optional[int] = None, device: Optional[int] = None,
                  uid: Optional[bytes] = None,
                  rendezvous_path: Optional[str] = None,
-                 force: bool = False):
+                 force: bool = False, rendezvous: str = "auto"):
         """force=True builds the native RCCL communicator even at world
         size 1 (a real self-clique: ncclCommInitRank + collectives execute
         on the GPU). Default world-1 behavior is no-op passthrough.
+
+        rendezvous: "file" (single node, shared /tmp), "tcp" (multi-node:
+        uid served from rank 0 on MASTER_PORT+1), or "auto" — tcp when
+        MASTER_ADDR points off-host, else file. TRTLAB_RDZV overrides.
 
         NOTE: this RCCL build rejects two ranks on ONE device
         ("Duplicate GPU detected", ncclInvalidUsage) — a genuine N>1
@@ -115,8 +184,17 @@ class Communicator:
         if uid is None:
             uid = C.comm.unique_id() if self.rank == 0 else None
             if self.world > 1:
-                uid = exchange_unique_id(self.rank, self.world, uid,
-                                         path=rendezvous_path)
+                mode = os.environ.get("TRTLAB_RDZV", rendezvous)
+                if mode == "auto":
+                    master = os.environ.get("MASTER_ADDR", "127.0.0.1")
+                    local = master in ("127.0.0.1", "localhost", "::1",
+                                       os.uname().nodename)
+                    mode = "file" if local else "tcp"
+                if mode == "tcp":
+                    uid = exchange_unique_id_tcp(self.rank, self.world, uid)
+                else:
+                    uid = exchange_unique_id(self.rank, self.world, uid,
+                                             path=rendezvous_path)
         self._comm = C.comm.Communicator(self.rank, self.world, uid,
                                          self.device)
 
