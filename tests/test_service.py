@@ -321,3 +321,32 @@ def test_trtis_surface_roundtrip():
         assert resp.meta_data.output[0].raw.byte_size == out.nbytes
     finally:
         server.shutdown()
+
+
+def test_trtis_model_config_generator():
+    """model_config.pbtxt generation from a compiled plan (reference
+    12_ConfigGenerator role): bindings, dtype mapping, batch-dim
+    stripping, instance group and dynamic batching stanzas."""
+    from trtlab_amd.engine.planner import Planner
+    from trtlab_amd.models import build_bert
+    from trtlab_amd.rpc.trtis import model_config_pbtxt
+
+    g = build_bert(batch=2, seq=64, layers=1, seed=0,
+                   embeddings=True, mask_input=True)
+    plan = Planner().compile(g)
+    txt = model_config_pbtxt(plan, "bert", max_batch_size=16, instances=3,
+                             preferred_batch_sizes=(4, 8),
+                             queue_delay_us=200)
+    assert 'name: "bert"' in txt
+    assert "max_batch_size: 16" in txt
+    assert txt.count("input {") == len(plan.inputs)
+    assert txt.count("output {") == len(plan.outputs)
+    assert "TYPE_INT32" in txt          # token-id inputs
+    assert "TYPE_FP16" in txt           # hidden outputs
+    assert "count: 3" in txt and "KIND_GPU" in txt
+    assert "preferred_batch_size: 4" in txt
+    assert "max_queue_delay_microseconds: 200" in txt
+    # max_batch strips the leading batch dim from a [M, H] binding
+    out_b = plan.outputs[0]
+    dims = [str(d) for d in out_b["shape"][1:]]
+    assert f"dims: [ {', '.join(dims)} ]" in txt

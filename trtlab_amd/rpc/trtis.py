@@ -231,3 +231,45 @@ class TrtisService:
             meta_data=hdr, raw_output=raw_out, batch_id=request.batch_id,
             compute_time=compute_s,
             request_time=time.monotonic() - t_start)
+
+
+# --------------------------------------------------------------- config gen
+_TRTIS_DT = {"f16": "TYPE_FP16", "bf16": "TYPE_BF16", "f32": "TYPE_FP32",
+             "i32": "TYPE_INT32", "i8": "TYPE_INT8"}
+
+
+def model_config_pbtxt(plan, name: str, max_batch_size: int = 0,
+                       instances: int = 1,
+                       preferred_batch_sizes=(), queue_delay_us: int = 0,
+                       ) -> str:
+    """Generate a TRTIS `model_config.pbtxt` for a compiled EnginePlan —
+    the reference ships a standalone ConfigGenerator app for this
+    (examples/12_ConfigGenerator; model_config.proto in
+    11_Protos). Bindings come from the plan's typed input/output tables;
+    dims drop the leading batch dim when max_batch_size > 0 (the TRTIS
+    convention). preferred_batch_sizes/queue_delay_us emit the
+    dynamic_batching stanza used with the batching server."""
+    def stanza(kind, b):
+        dims = list(b["shape"])
+        if max_batch_size > 0 and len(dims) > 1:
+            dims = dims[1:]
+        dd = ", ".join(str(int(d)) for d in dims)
+        return (f"{kind} {{\n  name: \"{b['name']}\"\n"
+                f"  data_type: {_TRTIS_DT[b['dtype']]}\n"
+                f"  dims: [ {dd} ]\n}}\n")
+
+    out = [f"name: \"{name}\"",
+           "platform: \"trtlab_amd\"",
+           f"max_batch_size: {max_batch_size}", ""]
+    for b in plan.inputs:
+        out.append(stanza("input", b))
+    for b in plan.outputs:
+        out.append(stanza("output", b))
+    out.append("instance_group {\n  count: %d\n  kind: KIND_GPU\n}\n"
+               % instances)
+    if preferred_batch_sizes:
+        pb = "\n".join(f"  preferred_batch_size: {int(p)}"
+                       for p in preferred_batch_sizes)
+        out.append("dynamic_batching {\n%s\n  max_queue_delay_microseconds:"
+                   " %d\n}\n" % (pb, queue_delay_us))
+    return "\n".join(out)
