@@ -332,7 +332,7 @@ def test_trtis_model_config_generator():
     from trtlab_amd.rpc.trtis import model_config_pbtxt
 
     g = build_bert(batch=2, seq=64, layers=1, seed=0,
-                   embeddings=True, mask_input=True)
+                   embeddings=True, varlen=True, mask_input=True)
     plan = Planner().compile(g)
     txt = model_config_pbtxt(plan, "bert", max_batch_size=16, instances=3,
                              preferred_batch_sizes=(4, 8),
