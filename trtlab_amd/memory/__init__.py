@@ -13,6 +13,8 @@ trtlab/tensorrt/src/workspace.cc:40-41).
 """
 from __future__ import annotations
 
+import os
+
 import threading
 
 import numpy as np
@@ -144,6 +146,62 @@ def device_bytes_in_use() -> int:
 # Blocking block pool (reference pool.h v4 + cuda allocator pools):
 # fixed-size device blocks, Pop blocks until a block is free, releases
 # return the block — the concurrency-limiting primitive for Buffers.
+class FirstTouchBuffer:
+    """Host buffer whose pages are faulted in ("first-touched") from a
+    thread pinned to a chosen NUMA node's CPUs, so the kernel places them
+    in that node's memory — the reference's first_touch_allocator
+    (core/memory/first_touch_allocator.h:35) for staging buffers feeding
+    a GPU on a specific socket. Pure malloc + sched_setaffinity; pinned
+    registration is the caller's choice (PinnedBuffer already faults via
+    hipHostMalloc on the calling thread).
+
+    node=-1 touches on the calling thread (plain first-touch)."""
+
+    PAGE = 4096
+
+    def __init__(self, nbytes: int, node: int = -1, topology=None):
+        import ctypes
+        import threading
+
+        self.nbytes = int(nbytes)
+        self._buf = (ctypes.c_char * self.nbytes)()
+        self.ptr = ctypes.addressof(self._buf)
+        self.node = node
+        self.touched_on: List[int] = []
+
+        def touch():
+            if node >= 0:
+                if topology is not None:
+                    cpus = topology.nearest_cpus(node)
+                else:
+                    from trtlab_amd.core.numa import NumaTopology
+
+                    cpus = NumaTopology().nearest_cpus(node)
+                if cpus:
+                    try:
+                        os.sched_setaffinity(0, cpus)
+                        self.touched_on = sorted(
+                            os.sched_getaffinity(0))
+                    except OSError:
+                        pass  # restricted affinity mask (containers)
+            mv = memoryview(self._buf).cast("B")
+            for off in range(0, self.nbytes, self.PAGE):
+                mv[off] = 0
+            if self.nbytes:
+                mv[-1] = 0
+
+        if node >= 0:
+            t = threading.Thread(target=touch, name="first-touch")
+            t.start()
+            t.join()
+        else:
+            touch()
+
+    def close(self):
+        self._buf = None
+        self.ptr = 0
+
+
 class BlockingBlockPool:
     def __init__(self, block_bytes: int, count: int, device: int = 0):
         from trtlab_amd import native
