@@ -1,0 +1,193 @@
+"""Streaming token-generation service: one request up, a token stream down.
+
+The RPC shape mirrors the reference's streaming lifecycle + single-up/
+multiple-down client (nvrpc life_cycle_streaming.h:61,
+client_single_up_multiple_down.h:43); what runs behind it is
+beyond-reference: a continuous-batching engine loop over an incremental
+DecodeSession — every replayed step advances ALL live streams one token,
+new requests claim idle slots mid-flight (their prompt tokens are fed
+one per step while other slots keep generating), finished slots park via
+idle_slot() and return to the free list.
+
+The engine only needs the DecodeSession surface (batch, step(ids) ->
+logits, reset_slot, idle_slot), so tests drive it with a deterministic
+fake on CPU and the same service serves GPT-2 or LLaMA sessions on GPU
+(examples/generation_server.py).
+"""
+from __future__ import annotations
+
+import asyncio
+from typing import List, Optional
+
+import numpy as np
+from google.protobuf import descriptor_pb2, descriptor_pool, message_factory
+
+_T = descriptor_pb2.FieldDescriptorProto
+
+_f = descriptor_pb2.FileDescriptorProto()
+_f.name = "trtlab_amd/rpc/generation.proto"
+_f.package = "trtlab.gen"
+_f.syntax = "proto3"
+
+_req = _f.message_type.add()
+_req.name = "GenerateRequest"
+for name, num, ft, rep in (("prompt", 1, _T.TYPE_INT32, True),
+                           ("max_tokens", 2, _T.TYPE_INT32, False)):
+    fd = _req.field.add()
+    fd.name = name
+    fd.number = num
+    fd.type = ft
+    fd.label = _T.LABEL_REPEATED if rep else _T.LABEL_OPTIONAL
+
+_tok = _f.message_type.add()
+_tok.name = "GenerateToken"
+for name, num, ft in (("token", 1, _T.TYPE_INT32),
+                      ("slot", 2, _T.TYPE_INT32),
+                      ("index", 3, _T.TYPE_INT32),
+                      ("done", 4, _T.TYPE_BOOL)):
+    fd = _tok.field.add()
+    fd.name = name
+    fd.number = num
+    fd.type = ft
+    fd.label = _T.LABEL_OPTIONAL
+
+_pool = descriptor_pool.Default()
+_fd = _pool.Add(_f)
+GenerateRequest = message_factory.GetMessageClass(
+    _fd.message_types_by_name["GenerateRequest"])
+GenerateToken = message_factory.GetMessageClass(
+    _fd.message_types_by_name["GenerateToken"])
+
+
+class GenerationEngine:
+    """Continuous-batching loop over a lockstep decode session."""
+
+    def __init__(self, session, eos: int = -1):
+        self.session = session
+        self.B = session.batch
+        self.eos = eos
+        self._slots: List[Optional[dict]] = [None] * self.B
+        self._free: List[int] = list(range(self.B))
+        self._wake = asyncio.Event()
+        self._freed = asyncio.Event()
+        self._task: Optional[asyncio.Task] = None
+        self.steps = 0  # total engine steps (observability / tests)
+        # a fresh DecodeSession has every slot ACTIVE at pos 0; park them
+        # all so unclaimed slots cost nothing and never hit the sequence
+        # limit while other slots generate (submit() re-activates)
+        for b in range(self.B):
+            self.session.idle_slot(b)
+
+    def ensure_started(self) -> None:
+        if self._task is None or self._task.done():
+            self._task = asyncio.get_running_loop().create_task(self._loop())
+
+    async def stop(self) -> None:
+        if self._task is not None:
+            self._task.cancel()
+            try:
+                await self._task
+            except asyncio.CancelledError:
+                pass
+            self._task = None
+
+    async def submit(self, prompt: List[int],
+                     max_tokens: int) -> tuple:
+        """Claim a slot (waiting for one to free if the batch is full),
+        prime it with the prompt, and return (slot, token queue). The
+        queue yields ints and a final None sentinel."""
+        if not prompt:
+            raise ValueError("empty prompt")
+        smax = getattr(self.session, "smax", 1 << 30)
+        if len(prompt) + max_tokens >= smax:
+            raise ValueError(
+                f"prompt+max_tokens ({len(prompt)}+{max_tokens}) exceeds "
+                f"the session window ({smax})")
+        while not self._free:
+            self._freed.clear()
+            await self._freed.wait()
+        b = self._free.pop()
+        self.session.reset_slot(b)
+        q: asyncio.Queue = asyncio.Queue()
+        self._slots[b] = dict(q=q, prompt=list(prompt), pi=0,
+                              remaining=int(max_tokens), cur=0)
+        self._wake.set()
+        return b, q
+
+    def _finish(self, b: int) -> None:
+        self._slots[b]["q"].put_nowait(None)
+        self._slots[b] = None
+        self.session.idle_slot(b)
+        self._free.append(b)
+        self._freed.set()
+
+    def _emit(self, b: int, logits_row: np.ndarray) -> None:
+        st = self._slots[b]
+        tok = int(np.argmax(logits_row))
+        st["cur"] = tok
+        st["q"].put_nowait(tok)
+        st["remaining"] -= 1
+        if st["remaining"] <= 0 or tok == self.eos:
+            self._finish(b)
+
+    async def _loop(self) -> None:
+        loop = asyncio.get_running_loop()
+        while True:
+            active = [b for b in range(self.B) if self._slots[b]]
+            if not active:
+                self._wake.clear()
+                await self._wake.wait()
+                continue
+            ids = np.zeros(self.B, np.int32)
+            for b in active:
+                st = self._slots[b]
+                ids[b] = (st["prompt"][st["pi"]]
+                          if st["pi"] < len(st["prompt"]) else st["cur"])
+            logits = await loop.run_in_executor(None, self.session.step,
+                                                ids)
+            self.steps += 1
+            for b in active:
+                st = self._slots[b]
+                if st is None:
+                    continue
+                if st["pi"] < len(st["prompt"]):
+                    st["pi"] += 1
+                    if st["pi"] == len(st["prompt"]):
+                        self._emit(b, logits[b])  # first generated token
+                else:
+                    self._emit(b, logits[b])
+
+
+class GenerationService:
+    """`trtlab.gen.Generation/Generate` — single request up, greedy token
+    stream down, over the shared continuous-batching engine."""
+
+    def __init__(self, session, eos: int = -1,
+                 name: str = "trtlab.gen.Generation"):
+        from trtlab_amd.rpc.server import StreamingService
+
+        self.engine = GenerationEngine(session, eos=eos)
+        svc = StreamingService(name)
+        svc.register_streaming("Generate", self._generate, GenerateRequest,
+                               GenerateToken)
+        self.service = svc
+
+    async def _generate(self, request_iter, context, resources):
+        self.engine.ensure_started()
+        req = await request_iter.__anext__()  # single-up
+        try:
+            b, q = await self.engine.submit(list(req.prompt),
+                                            req.max_tokens or 16)
+        except ValueError as e:
+            import grpc
+
+            await context.abort(grpc.StatusCode.INVALID_ARGUMENT, str(e))
+            return
+        i = 0
+        while True:
+            tok = await q.get()
+            if tok is None:
+                yield GenerateToken(slot=b, index=i, done=True)
+                return
+            yield GenerateToken(token=tok, slot=b, index=i)
+            i += 1
