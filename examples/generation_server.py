@@ -56,8 +56,10 @@ def client(args):
         response_deserializer=GenerateToken.FromString)
     t0 = time.perf_counter()
     toks = []
-    for resp in call(iter([GenerateRequest(prompt=args.prompt,
-                                           max_tokens=args.new_tokens)])):
+    req = GenerateRequest(prompt=args.prompt, max_tokens=args.new_tokens,
+                          temperature=args.temperature, top_k=args.top_k,
+                          top_p=args.top_p, seed=args.seed)
+    for resp in call(iter([req])):
         if not resp.done:
             toks.append(resp.token)
             print(f"  token[{resp.index}] = {resp.token} "
@@ -78,6 +80,10 @@ def main():
     ap.add_argument("--port", type=int, default=50055)
     ap.add_argument("--prompt", type=int, nargs="*", default=[11, 42, 7])
     ap.add_argument("--new-tokens", type=int, default=32)
+    ap.add_argument("--temperature", type=float, default=0.0)
+    ap.add_argument("--top-k", type=int, default=0)
+    ap.add_argument("--top-p", type=float, default=0.0)
+    ap.add_argument("--seed", type=int, default=0)
     args = ap.parse_args()
     client(args) if args.client else serve(args)
 

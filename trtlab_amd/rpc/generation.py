@@ -32,7 +32,11 @@ _f.syntax = "proto3"
 _req = _f.message_type.add()
 _req.name = "GenerateRequest"
 for name, num, ft, rep in (("prompt", 1, _T.TYPE_INT32, True),
-                           ("max_tokens", 2, _T.TYPE_INT32, False)):
+                           ("max_tokens", 2, _T.TYPE_INT32, False),
+                           ("temperature", 3, _T.TYPE_FLOAT, False),
+                           ("top_k", 4, _T.TYPE_INT32, False),
+                           ("top_p", 5, _T.TYPE_FLOAT, False),
+                           ("seed", 6, _T.TYPE_INT32, False)):
     fd = _req.field.add()
     fd.name = name
     fd.number = num
@@ -91,11 +95,17 @@ class GenerationEngine:
                 pass
             self._task = None
 
-    async def submit(self, prompt: List[int],
-                     max_tokens: int) -> tuple:
+    async def submit(self, prompt: List[int], max_tokens: int,
+                     temperature: float = 0.0, top_k: int = 0,
+                     top_p: float = 0.0, seed: int = 0) -> tuple:
         """Claim a slot (waiting for one to free if the batch is full),
         prime it with the prompt, and return (slot, token queue). The
-        queue yields ints and a final None sentinel."""
+        queue yields ints and a final None sentinel.
+
+        temperature <= 0 decodes greedily; otherwise logits/T are
+        softmaxed and sampled, restricted to the top_k highest (0 = all)
+        and the top_p nucleus (0 = all), from a per-slot RNG seeded by
+        `seed` so streams are reproducible."""
         if not prompt:
             raise ValueError("empty prompt")
         smax = getattr(self.session, "smax", 1 << 30)
@@ -110,7 +120,10 @@ class GenerationEngine:
         self.session.reset_slot(b)
         q: asyncio.Queue = asyncio.Queue()
         self._slots[b] = dict(q=q, prompt=list(prompt), pi=0,
-                              remaining=int(max_tokens), cur=0)
+                              remaining=int(max_tokens), cur=0,
+                              temperature=float(temperature),
+                              top_k=int(top_k), top_p=float(top_p),
+                              rng=np.random.RandomState(seed or None))
         self._wake.set()
         return b, q
 
@@ -121,9 +134,31 @@ class GenerationEngine:
         self._free.append(b)
         self._freed.set()
 
+    @staticmethod
+    def _sample(logits: np.ndarray, temperature: float, top_k: int,
+                top_p: float, rng) -> int:
+        if temperature <= 0.0:
+            return int(np.argmax(logits))
+        x = logits.astype(np.float64) / temperature
+        x -= x.max()
+        p = np.exp(x)
+        p /= p.sum()
+        order = np.argsort(-p)
+        keep = len(order)
+        if top_k > 0:
+            keep = min(keep, top_k)
+        if 0.0 < top_p < 1.0:
+            c = np.cumsum(p[order])
+            # smallest prefix whose mass reaches top_p (always >= 1)
+            keep = min(keep, int(np.searchsorted(c, top_p) + 1))
+        idx = order[:keep]
+        pk = p[idx] / p[idx].sum()
+        return int(rng.choice(idx, p=pk))
+
     def _emit(self, b: int, logits_row: np.ndarray) -> None:
         st = self._slots[b]
-        tok = int(np.argmax(logits_row))
+        tok = self._sample(logits_row, st["temperature"], st["top_k"],
+                           st["top_p"], st["rng"])
         st["cur"] = tok
         st["q"].put_nowait(tok)
         st["remaining"] -= 1
@@ -176,8 +211,10 @@ class GenerationService:
         self.engine.ensure_started()
         req = await request_iter.__anext__()  # single-up
         try:
-            b, q = await self.engine.submit(list(req.prompt),
-                                            req.max_tokens or 16)
+            b, q = await self.engine.submit(
+                list(req.prompt), req.max_tokens or 16,
+                temperature=req.temperature, top_k=req.top_k,
+                top_p=req.top_p, seed=req.seed)
         except ValueError as e:
             import grpc
 
