@@ -176,6 +176,8 @@ class DecodeSession:
                 raise ValueError("decode: head_dim must be 64 or 128")
             lay = {}
             if self.arch == "llama":
+                self.rms_eps = float(
+                    nodes[f"l{li}_rms1"].attrs.get("eps", 1e-5))
                 lay["rms1_g"] = dev32(nodes[f"l{li}_rms1"].attrs["gamma"])
                 lay["rms2_g"] = dev32(nodes[f"l{li}_rms2"].attrs["gamma"])
                 self.theta = float(
@@ -383,7 +385,7 @@ class DecodeSession:
                          self.pos.data_ptr(), B, Hd, stream=s, sync=False)
         ops.rmsnorm(0, self.h.data_ptr(),
                     self.layers[0]["rms1_g"].data_ptr(), self.x.data_ptr(),
-                    B, Hd, stream=s, sync=False)
+                    B, Hd, eps=self.rms_eps, stream=s, sync=False)
         for li, lay in enumerate(self.layers):
             ops.gemm_bt(0, self.x.data_ptr(), lay["qkv_w"].data_ptr(),
                         self.qkv.data_ptr(), M=B, N=3 * Hd, K=Hd,
@@ -399,7 +401,7 @@ class DecodeSession:
             ops.add_rmsnorm(0, self.x2.data_ptr(), self.h.data_ptr(),
                             lay["rms2_g"].data_ptr(), self.x.data_ptr(),
                             sum_out=self.h.data_ptr(), M=B, N=Hd,
-                            stream=s, sync=False)
+                            eps=self.rms_eps, stream=s, sync=False)
             ops.gemm_bt(0, self.x.data_ptr(), lay["gate_w"].data_ptr(),
                         self.ff.data_ptr(), M=B, N=self.inter, K=Hd,
                         epi=self._epi_none, stream=s, sync=False, tile=T4)
@@ -417,7 +419,7 @@ class DecodeSession:
             dst = (self.x if nxt else self.out).data_ptr()
             ops.add_rmsnorm(0, self.x2.data_ptr(), self.h.data_ptr(), gptr,
                             dst, sum_out=self.h.data_ptr(), M=B, N=Hd,
-                            stream=s, sync=False)
+                            eps=self.rms_eps, stream=s, sync=False)
         if self.logits is not None:
             ops.gemm_bt(0, self.out.data_ptr(), self.tok.data_ptr(),
                         self.logits.data_ptr(), M=B, N=self.vocab, K=Hd,
@@ -541,7 +543,7 @@ class DecodeSession:
         if self.arch == "llama":
             ops.rmsnorm(0, h.data_ptr(),
                         self.layers[0]["rms1_g"].data_ptr(), x.data_ptr(),
-                        M, Hd, stream=s, sync=False)
+                        M, Hd, eps=self.rms_eps, stream=s, sync=False)
             for li, lay in enumerate(self.layers):
                 ops.gemm_bt(0, x.data_ptr(), lay["qkv_w"].data_ptr(),
                             qkv.data_ptr(), M=M, N=3 * Hd, K=Hd,
@@ -563,7 +565,7 @@ class DecodeSession:
                             epi=self._epi_none, stream=s, sync=False)
                 ops.add_rmsnorm(0, x.data_ptr(), h.data_ptr(),
                                 lay["rms2_g"].data_ptr(), x2.data_ptr(),
-                                sum_out=h.data_ptr(), M=M, N=Hd, stream=s,
+                                sum_out=h.data_ptr(), M=M, N=Hd, eps=self.rms_eps, stream=s,
                                 sync=False)
                 ops.gemm_bt(0, x2.data_ptr(), lay["gate_w"].data_ptr(),
                             ff.data_ptr(), M=M, N=inter, K=Hd,
@@ -581,7 +583,7 @@ class DecodeSession:
                 gptr = (nxt["rms1_g"] if nxt else self.lnf_g).data_ptr()
                 ops.add_rmsnorm(0, x2.data_ptr(), h.data_ptr(), gptr,
                                 x.data_ptr(), sum_out=h.data_ptr(), M=M,
-                                N=Hd, stream=s, sync=False)
+                                N=Hd, eps=self.rms_eps, stream=s, sync=False)
         else:
             ops.layernorm(0, h.data_ptr(),
                           self.layers[0]["ln1_g"].data_ptr(),
@@ -732,7 +734,7 @@ class DecodeSession:
         if self.arch == "llama":
             ops.rmsnorm(0, cb["h"].data_ptr(),
                         self.layers[0]["rms1_g"].data_ptr(),
-                        cb["x"].data_ptr(), M, Hd, stream=s, sync=False)
+                        cb["x"].data_ptr(), M, Hd, eps=self.rms_eps, stream=s, sync=False)
             for li, lay in enumerate(self.layers):
                 ops.gemm_bt(0, cb["x"].data_ptr(), lay["qkv_w"].data_ptr(),
                             cb["qkv"].data_ptr(), M=M, N=3 * Hd, K=Hd,
@@ -762,7 +764,7 @@ class DecodeSession:
                                 lay["rms2_g"].data_ptr(),
                                 cb["x"].data_ptr(),
                                 sum_out=cb["h"].data_ptr(), M=M, N=Hd,
-                                stream=s, sync=False)
+                                eps=self.rms_eps, stream=s, sync=False)
                 ops.gemm_bt(0, cb["x"].data_ptr(), lay["gate_w"].data_ptr(),
                             cb["ff"].data_ptr(), M=M, N=inter, K=Hd,
                             epi=self._epi_none, stream=s, sync=False)
@@ -782,7 +784,7 @@ class DecodeSession:
                 ops.add_rmsnorm(0, cb["x2"].data_ptr(), cb["h"].data_ptr(),
                                 gptr, cb["x"].data_ptr(),
                                 sum_out=cb["h"].data_ptr(), M=M, N=Hd,
-                                stream=s, sync=False)
+                                eps=self.rms_eps, stream=s, sync=False)
         else:
             ops.layernorm(0, cb["h"].data_ptr(),
                           self.layers[0]["ln1_g"].data_ptr(),
