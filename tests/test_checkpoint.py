@@ -17,10 +17,12 @@ H, LAYERS, HEADS, VOCAB, INTER = 256, 2, 2, 300, 448
 EPS, THETA = 1e-6, 10000.0
 
 
-def _mk_checkpoint(tmpdir):
+def _mk_checkpoint(tmpdir, kvh=HEADS):
     from safetensors.numpy import save_file
 
     rng = np.random.RandomState(7)
+    hd = H // HEADS
+    kv_rows = kvh * hd
 
     def w(o, i):
         return (rng.randn(o, i) / np.sqrt(i)).astype(np.float32)
@@ -36,8 +38,8 @@ def _mk_checkpoint(tmpdir):
         st[p + "post_attention_layernorm.weight"] = \
             rng.uniform(0.9, 1.1, H).astype(np.float32)
         for nm, shape in (("self_attn.q_proj", (H, H)),
-                          ("self_attn.k_proj", (H, H)),
-                          ("self_attn.v_proj", (H, H)),
+                          ("self_attn.k_proj", (kv_rows, H)),
+                          ("self_attn.v_proj", (kv_rows, H)),
                           ("self_attn.o_proj", (H, H)),
                           ("mlp.gate_proj", (INTER, H)),
                           ("mlp.up_proj", (INTER, H)),
@@ -46,14 +48,16 @@ def _mk_checkpoint(tmpdir):
     save_file(st, os.path.join(tmpdir, "model.safetensors"))
     with open(os.path.join(tmpdir, "config.json"), "w") as f:
         json.dump({"num_attention_heads": HEADS,
-                   "num_key_value_heads": HEADS,
+                   "num_key_value_heads": kvh,
                    "rope_theta": THETA, "rms_norm_eps": EPS}, f)
     return st
 
 
-def _hf_forward(st, ids, seq):
-    """Independent HF-semantics oracle (fp32 torch)."""
+def _hf_forward(st, ids, seq, kvh=HEADS):
+    """Independent HF-semantics oracle (fp32 torch; GQA when
+    kvh < HEADS — query head h attends kv head h // (HEADS//kvh))."""
     hd = H // HEADS
+    grp = HEADS // kvh
 
     def rms(x, g):
         v = x / torch.sqrt((x * x).mean(-1, keepdim=True) + EPS)
@@ -77,15 +81,17 @@ def _hf_forward(st, ids, seq):
         p = f"model.layers.{li}."
         x = rms(h, st[p + "input_layernorm.weight"])
         q = lin(x, p + "self_attn.q_proj.weight").view(seq, HEADS, hd)
-        k = lin(x, p + "self_attn.k_proj.weight").view(seq, HEADS, hd)
-        v = lin(x, p + "self_attn.v_proj.weight").view(seq, HEADS, hd)
+        k = lin(x, p + "self_attn.k_proj.weight").view(seq, kvh, hd)
+        v = lin(x, p + "self_attn.v_proj.weight").view(seq, kvh, hd)
         q, k = rope(q, pos), rope(k, pos)
         att = torch.zeros(seq, HEADS, hd)
         mask = torch.tril(torch.ones(seq, seq, dtype=torch.bool))
         for hh in range(HEADS):
-            sc = (q[:, hh] @ k[:, hh].t()) / np.sqrt(hd)
+            kk = k[:, hh // grp]
+            vv = v[:, hh // grp]
+            sc = (q[:, hh] @ kk.t()) / np.sqrt(hd)
             sc = sc.masked_fill(~mask, float("-inf"))
-            att[:, hh] = torch.softmax(sc, -1) @ v[:, hh]
+            att[:, hh] = torch.softmax(sc, -1) @ vv
         h = h + lin(att.reshape(seq, H), p + "self_attn.o_proj.weight")
         x = rms(h, st[p + "post_attention_layernorm.weight"])
         gate = lin(x, p + "mlp.gate_proj.weight")
@@ -107,12 +113,18 @@ def test_safetensors_llama_matches_hf_semantics(tmp_path):
     assert err < 2e-3, err
 
 
-def test_safetensors_llama_gqa_rejected(tmp_path):
-    _mk_checkpoint(str(tmp_path))
-    with open(tmp_path / "config.json", "w") as f:
-        json.dump({"num_attention_heads": 2, "num_key_value_heads": 1}, f)
-    with pytest.raises(ValueError, match="GQA"):
-        build_llama_from_safetensors(str(tmp_path), seq=16)
+def test_safetensors_llama_gqa_matches_hf_semantics(tmp_path):
+    """GQA (1 kv head serving 2 query heads) loads via exact kv-head
+    replication and matches the true grouped-attention oracle."""
+    st = _mk_checkpoint(str(tmp_path), kvh=1)
+    seq = 20
+    g = build_llama_from_safetensors(str(tmp_path), batch=1, seq=seq)
+    plan = Planner().compile(g)
+    ids = np.random.RandomState(5).randint(0, VOCAB, seq).astype(np.int32)
+    out = run_reference(plan, ids)
+    ref = _hf_forward(st, torch.from_numpy(ids).long(), seq, kvh=1)
+    err = np.abs(out - ref).max() / max(np.abs(ref).max(), 1e-6)
+    assert err < 2e-3, err
 
 
 def test_safetensors_llama_decode_session_compat(tmp_path):
