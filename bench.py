@@ -80,7 +80,7 @@ def main():
     ap.add_argument("--contexts", type=int, default=3)
     ap.add_argument("--model", default="resnet50",
                     choices=["resnet50", "resnet101", "resnet152", "bert",
-                             "bert-large", "gpt2"])
+                             "bert-large", "gpt2", "llama"])
     ap.add_argument("--dtype", default="fp16",
                     choices=["fp16", "bf16", "int8", "fp8", "mxfp4", "mxfp8"])
     ap.add_argument("--no-autotune", action="store_true",
@@ -138,6 +138,14 @@ def main():
 
         g = build_gpt2(batch=args.batch, seq=1024, layers=12, seed=0)
         cfg_extra = {"seq_len": 1024, "hidden": 768, "layers": 12,
+                     "phase": "prefill"}
+    elif args.model == "llama":
+        # LLaMA-architecture prefill (RMSNorm + RoPE + SwiGLU, hd128)
+        from trtlab_amd.models import build_llama
+
+        g = build_llama(batch=args.batch, seq=512, hidden=2048, layers=8,
+                        heads=16, seed=0)
+        cfg_extra = {"seq_len": 512, "hidden": 2048, "layers": 8,
                      "phase": "prefill"}
     else:
         depth = int(args.model.replace("resnet", ""))

@@ -18,11 +18,19 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 def serve(args):
     from trtlab_amd.engine.decode import DecodeSession
-    from trtlab_amd.models import build_gpt2, build_llama
+    from trtlab_amd.models import (build_gpt2, build_llama,
+                                   build_llama_from_safetensors)
     from trtlab_amd.rpc.generation import GenerationService
     from trtlab_amd.rpc.server import Server
 
-    if args.arch == "llama":
+    if args.ckpt:
+        # real HF weights (single file or sharded dir), arch-selected
+        from trtlab_amd.models import build_gpt2_from_safetensors
+
+        loader = (build_llama_from_safetensors if args.arch == "llama"
+                  else build_gpt2_from_safetensors)
+        g = loader(args.ckpt, batch=args.batch, seq=args.smax)
+    elif args.arch == "llama":
         hid = args.hidden or 1024
         g = build_llama(batch=args.batch, seq=args.smax, hidden=hid,
                         layers=args.layers, heads=hid // 128, seed=0)
@@ -73,6 +81,8 @@ def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--client", action="store_true")
     ap.add_argument("--arch", choices=("gpt2", "llama"), default="gpt2")
+    ap.add_argument("--ckpt", default="",
+                    help="HF LLaMA safetensors checkpoint (file or dir)")
     ap.add_argument("--batch", type=int, default=8)
     ap.add_argument("--layers", type=int, default=8)
     ap.add_argument("--hidden", type=int, default=0)

@@ -140,3 +140,87 @@ def test_safetensors_llama_decode_session_compat(tmp_path):
                      "gate", "up", "down"):
             assert f"l{li}_{suff}" in names, suff
     assert "rms_f" in names and "embed" in names
+
+
+def test_safetensors_gpt2_matches_hf_semantics(tmp_path):
+    """GPT-2 loader: Conv1D transposition + tanh-GELU + learned
+    positions vs an independent HF-semantics oracle."""
+    from safetensors.numpy import save_file
+
+    from trtlab_amd.models import build_gpt2_from_safetensors
+
+    Hh, LAY, HD, V, P = 128, 2, 2, 200, 64  # head_dim 64
+    rng = np.random.RandomState(11)
+    st = {"transformer.wte.weight":
+          (rng.randn(V, Hh) * 0.05).astype(np.float32),
+          "transformer.wpe.weight":
+          (rng.randn(P, Hh) * 0.05).astype(np.float32),
+          "transformer.ln_f.weight":
+          rng.uniform(0.9, 1.1, Hh).astype(np.float32),
+          "transformer.ln_f.bias":
+          (rng.randn(Hh) * 0.02).astype(np.float32)}
+    for li in range(LAY):
+        p = f"transformer.h.{li}."
+        for nm, shp in (("ln_1.weight", (Hh,)), ("ln_1.bias", (Hh,)),
+                        ("ln_2.weight", (Hh,)), ("ln_2.bias", (Hh,)),
+                        ("attn.c_attn.weight", (Hh, 3 * Hh)),
+                        ("attn.c_attn.bias", (3 * Hh,)),
+                        ("attn.c_proj.weight", (Hh, Hh)),
+                        ("attn.c_proj.bias", (Hh,)),
+                        ("mlp.c_fc.weight", (Hh, 4 * Hh)),
+                        ("mlp.c_fc.bias", (4 * Hh,)),
+                        ("mlp.c_proj.weight", (4 * Hh, Hh)),
+                        ("mlp.c_proj.bias", (Hh,))):
+            scale = 0.02 if nm.endswith("bias") else 1 / np.sqrt(shp[0])
+            st[p + nm] = (rng.randn(*shp) * scale).astype(np.float32)
+        st[p + "ln_1.weight"] = rng.uniform(0.9, 1.1, Hh).astype(np.float32)
+        st[p + "ln_2.weight"] = rng.uniform(0.9, 1.1, Hh).astype(np.float32)
+    save_file(st, os.path.join(str(tmp_path), "model.safetensors"))
+    with open(tmp_path / "config.json", "w") as f:
+        json.dump({"n_head": HD}, f)
+
+    seq = 16
+    g = build_gpt2_from_safetensors(str(tmp_path), batch=1, seq=seq)
+    plan = Planner().compile(g)
+    ids = np.random.RandomState(4).randint(0, V, seq).astype(np.int32)
+    out = run_reference(plan, ids)
+
+    # --- independent HF-GPT2 oracle (Conv1D: x @ W) ---
+    def t(k):
+        return torch.from_numpy(st["transformer." + k])
+
+    def ln(x, w, b):
+        mu = x.mean(-1, keepdim=True)
+        var = x.var(-1, unbiased=False, keepdim=True)
+        return (x - mu) / torch.sqrt(var + 1e-5) * t(w) + t(b)
+
+    def gelu_new(x):
+        return 0.5 * x * (1 + torch.tanh(
+            np.sqrt(2 / np.pi) * (x + 0.044715 * x ** 3)))
+
+    hid = t("wte.weight")[torch.from_numpy(ids).long()] + \
+        t("wpe.weight")[:seq]
+    hd = Hh // HD
+    mask = torch.tril(torch.ones(seq, seq, dtype=torch.bool))
+    for li in range(LAY):
+        p = f"h.{li}."
+        x = ln(hid, p + "ln_1.weight", p + "ln_1.bias")
+        qkv = x @ t(p + "attn.c_attn.weight") + t(p + "attn.c_attn.bias")
+        q, k, v = qkv.split(Hh, dim=-1)
+        q = q.view(seq, HD, hd)
+        k = k.view(seq, HD, hd)
+        v = v.view(seq, HD, hd)
+        att = torch.zeros(seq, HD, hd)
+        for hh in range(HD):
+            sc = (q[:, hh] @ k[:, hh].t()) / np.sqrt(hd)
+            sc = sc.masked_fill(~mask, float("-inf"))
+            att[:, hh] = torch.softmax(sc, -1) @ v[:, hh]
+        hid = hid + att.reshape(seq, Hh) @ t(p + "attn.c_proj.weight") + \
+            t(p + "attn.c_proj.bias")
+        x = ln(hid, p + "ln_2.weight", p + "ln_2.bias")
+        ff = gelu_new(x @ t(p + "mlp.c_fc.weight") + t(p + "mlp.c_fc.bias"))
+        hid = hid + ff @ t(p + "mlp.c_proj.weight") + \
+            t(p + "mlp.c_proj.bias")
+    ref = ln(hid, "ln_f.weight", "ln_f.bias").numpy()
+    err = np.abs(out - ref).max() / max(np.abs(ref).max(), 1e-6)
+    assert err < 2e-3, err

@@ -137,3 +137,66 @@ def build_llama_from_safetensors(path: str, batch: int = 1,
         h = g.add(h, down, name=f"l{li}_res2")
     g.rmsnorm(h, state["model.norm.weight"], eps=eps, name="rms_f")
     return g
+
+
+def build_gpt2_from_safetensors(path: str, batch: int = 1,
+                                seq: int = 256,
+                                heads: Optional[int] = None) -> Graph:
+    """Load HF GPT-2 weights (`GPT2LMHeadModel` safetensors) into the
+    engine IR. HF GPT-2 uses Conv1D (x @ W, weight [in, out]) where the
+    engine's gemm computes x @ W^T — every c_attn/c_proj/c_fc weight is
+    transposed on load. Names may carry the `transformer.` prefix.
+
+    Mapping: wte/wpe -> embedding tables; h.{i}.ln_1/ln_2 -> layernorms;
+    h.{i}.attn.c_attn -> fused qkv; attn.c_proj -> proj; mlp.c_fc/c_proj
+    -> ff1/ff2 (GELU between, HF gelu_new == the engine's tanh GELU);
+    ln_f -> final layernorm."""
+    state = _load_state(path)
+    cfg = _read_config(path)
+    pre = "transformer." if any(k.startswith("transformer.")
+                                for k in state) else ""
+
+    def get(name):
+        return state[pre + name]
+
+    heads = heads or cfg.get("n_head")
+    if heads is None:
+        raise ValueError("pass heads= (no config.json with n_head)")
+    tok = get("wte.weight")                 # [V, H]
+    posw = get("wpe.weight")                # [P, H]
+    vocab, hidden = tok.shape
+    if hidden // heads not in (64, 128):
+        raise ValueError(f"head_dim {hidden // heads} unsupported")
+    if seq > posw.shape[0]:
+        raise ValueError(f"seq {seq} > position table {posw.shape[0]}")
+    layers = 0
+    while f"{pre}h.{layers}.ln_1.weight" in state:
+        layers += 1
+    if layers == 0:
+        raise ValueError("no h.* blocks in checkpoint")
+
+    g = Graph(f"gpt2_hf_h{hidden}_l{layers}_s{seq}_b{batch}")
+    m = batch * seq
+    ids = g.input((m,), name="token_ids", dtype="i32")
+    h = g.embedding(ids, tok, posw[:seq], name="embed")
+    for li in range(layers):
+        p = f"h.{li}."
+        x = g.layernorm(h, get(p + "ln_1.weight"), get(p + "ln_1.bias"),
+                        name=f"l{li}_ln1")
+        qkv = g.gemm(x, get(p + "attn.c_attn.weight").T.copy(),
+                     get(p + "attn.c_attn.bias"), name=f"l{li}_qkv")
+        att = g.attention(qkv, heads=heads, seq=seq, causal=True,
+                          name=f"l{li}_att")
+        proj = g.gemm(att, get(p + "attn.c_proj.weight").T.copy(),
+                      get(p + "attn.c_proj.bias"), name=f"l{li}_proj")
+        h = g.add(h, proj, name=f"l{li}_res1")
+        x = g.layernorm(h, get(p + "ln_2.weight"), get(p + "ln_2.bias"),
+                        name=f"l{li}_ln2")
+        ff1 = g.gemm(x, get(p + "mlp.c_fc.weight").T.copy(),
+                     get(p + "mlp.c_fc.bias"), name=f"l{li}_ff1")
+        ff1 = g.gelu(ff1, name=f"l{li}_gelu")
+        ff2 = g.gemm(ff1, get(p + "mlp.c_proj.weight").T.copy(),
+                     get(p + "mlp.c_proj.bias"), name=f"l{li}_ff2")
+        h = g.add(h, ff2, name=f"l{li}_res2")
+    g.layernorm(h, get("ln_f.weight"), get("ln_f.bias"), name="ln_f")
+    return g
