@@ -1274,3 +1274,42 @@ def test_llama_prefill_matches_sequential():
         assert np.abs(sa - sb).max() / max(np.abs(sa).max(), 1e-6) < 0.05, t
     a.close()
     b.close()
+
+
+def test_speculative_decoding_invariant_llama():
+    """Spec decode on the LLaMA recipe (chunk-strided RoPE in the
+    verifier): output identical to target-only greedy with both an
+    identical draft (acceptance ~1) and a cross-architecture draft."""
+    from trtlab_amd.engine.decode import DecodeSession, SpeculativeDecoder
+    from trtlab_amd.models import build_llama
+
+    g = build_llama(batch=2, seq=96, hidden=512, layers=2, heads=4,
+                    seed=0, vocab=2000)
+    g_same = build_llama(batch=2, seq=96, hidden=512, layers=2, heads=4,
+                         seed=0, vocab=2000)
+    g_diff = build_llama(batch=2, seq=96, hidden=512, layers=1, heads=4,
+                         seed=5, vocab=2000)
+    seed_tok = np.array([17, 23], np.int32)
+    STEPS = 10
+
+    base = DecodeSession(g, batch=2, smax=96, capture=False, lm_head=True)
+    cur = seed_tok
+    ref = []
+    for _ in range(STEPS):
+        lg = base.step(cur)
+        cur = lg.argmax(-1).astype(np.int32)
+        ref.append(cur)
+    base.close()
+    ref = np.stack(ref, axis=1)
+
+    for gd, expect_high in ((g_same, True), (g_diff, False)):
+        t = DecodeSession(g, batch=2, smax=96, capture=False, lm_head=True)
+        d = DecodeSession(gd, batch=2, smax=96, capture=False,
+                          lm_head=True)
+        sd = SpeculativeDecoder(t, d, k=3)
+        toks, rate = sd.generate(seed_tok, STEPS)
+        np.testing.assert_array_equal(toks, ref)
+        if expect_high:
+            assert rate > 0.9, rate
+        t.close()
+        d.close()
