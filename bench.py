@@ -170,7 +170,10 @@ def main():
     ctxs = [eng.create_context(capture=True) for _ in range(args.contexts)]
 
     rng = np.random.RandomState(123 + rank)
-    batch = (rng.randn(*plan.input_shape) * 0.5).astype(np.float16)
+    if plan.inputs[0]["dtype"] == "i32":  # token-id models (llama)
+        batch = rng.randint(1, 30000, plan.input_shape).astype(np.int32)
+    else:
+        batch = (rng.randn(*plan.input_shape) * 0.5).astype(np.float16)
     if args.dtype == "bf16":
         # bf16 bindings carry bit patterns (numpy has no bf16 dtype)
         from trtlab_amd.engine.planner import _bf16_bits

@@ -233,7 +233,11 @@ __device__ __forceinline__ void store_epilogue(
 template <int G>
 __device__ __forceinline__ void wait_tiles_inflight(int ahead) {
   if (ahead >= 2) {
-    if constexpr (G == 4)
+    if constexpr (G == 2)
+      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    else if constexpr (G == 3)
+      asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+    else if constexpr (G == 4)
       asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
     else if constexpr (G == 6)
       asm volatile("s_waitcnt vmcnt(12)" ::: "memory");
@@ -246,7 +250,11 @@ __device__ __forceinline__ void wait_tiles_inflight(int ahead) {
     else
       asm volatile("s_waitcnt vmcnt(16)" ::: "memory");
   } else if (ahead == 1) {
-    if constexpr (G == 4)
+    if constexpr (G == 2)
+      asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+    else if constexpr (G == 3)
+      asm volatile("s_waitcnt vmcnt(3)" ::: "memory");
+    else if constexpr (G == 4)
       asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
     else if constexpr (G == 6)
       asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
