@@ -173,6 +173,14 @@ void ExecutionContext::enqueue_all(hipStream_t s) {
                       op.KH, op.KW, op.sh, op.sw, op.ph, op.pw, op.epi, os,
                       op.tile, op.fork ? scratch2_ : scratch_, op.res_scale);
         break;
+      case kConst:
+        // device constant: blob -> arena slot (captured in the graph)
+        TRT_HIP_CHECK(hipMemcpyAsync(A(op.out_off), Wp(op.w_off),
+                                     (size_t)op.n_elems,
+                                     hipMemcpyDeviceToDevice, os));
+        break;
+      case kView:
+        break;  // arena alias — out_off == in_off, nothing to launch
       case kBtail:
         // fused bottleneck tail: W1 at w_off, W2 at w2_off; the fp32
         // scale/bias blobs hold [s1 | s2] / [b1 | b2] (s2 at + C floats)

@@ -121,6 +121,8 @@ enum OpKind : int {
   kSiluMul = 23,     // SwiGLU gate: silu(a) * b
   kRope = 24,        // rotary embedding in-place on qkv q/k blocks
   kBtail = 25,       // fused bottleneck tail: 3x3+BN+ReLU -> 1x1+BN+res+ReLU
+  kConst = 26,       // weight-blob constant -> arena tensor (D2D copy)
+  kView = 27,        // zero-copy reshape (arena alias; no kernel)
 };
 
 struct OpDesc {

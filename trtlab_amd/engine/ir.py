@@ -214,6 +214,28 @@ class Graph:
                           dict(heads=heads, seq=seq, theta=float(theta)),
                           name)
 
+    def view(self, x: str, shape: Tuple[int, ...],
+             name: Optional[str] = None) -> str:
+        """Zero-copy reshape: same row-major bytes under a new shape
+        (e.g. NHWC patch grid [B, g, g, H] -> token rows [B*g*g, H]).
+        Lowered to an arena alias — no kernel, no copy."""
+        src = self.tensors[x]
+        n = 1
+        for d in shape:
+            n *= int(d)
+        assert n == src.numel, f"view: numel {n} != {src.numel}"
+        return self._emit("view", [x], tuple(int(d) for d in shape), {},
+                          name)
+
+    def constant(self, value: np.ndarray,
+                 name: Optional[str] = None) -> str:
+        """Device-resident constant tensor (packed into the weight blob,
+        copied into its arena slot inside the captured graph) — e.g. ViT
+        position embeddings added to the patch tokens."""
+        value = np.asarray(value, np.float32)
+        return self._emit("constant", [], value.shape, dict(value=value),
+                          name)
+
     def clip(self, x: str, mn: float, mx: float,
              name: Optional[str] = None) -> str:
         """out = min(max(x, mn), mx) (ONNX Clip with arbitrary bounds)."""

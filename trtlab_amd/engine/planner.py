@@ -60,6 +60,8 @@ K_RMSNORM = 22  # LLaMA norm: x / rms(x) * gamma
 K_SILU_MUL = 23  # SwiGLU gate: silu(a) * b
 K_ROPE = 24  # rotary embedding, in-place on qkv (arena-aliased output)
 K_BTAIL = 25  # fused bottleneck tail: conv3x3+BN+ReLU -> 1x1+BN+res+ReLU
+K_CONST = 26  # weight-blob constant -> arena tensor (one D2D copy)
+K_VIEW = 27   # zero-copy reshape: output aliases the input's arena bytes
 
 
 def _bf16_bits(arr: np.ndarray) -> np.ndarray:
@@ -239,6 +241,20 @@ class Planner:
                         K_COPY2D, f"{n.name}_part{ci}", [t], n.output,
                         dict(coff=coff, C=c, ldd=ctot)))
                     coff += c
+            elif n.kind == "view":
+                # zero-copy reshape (row-major layouts agree); lowered to
+                # an arena alias like rope's in-place output
+                op = ExecOp(K_VIEW, n.name, [n.inputs[0]], n.output,
+                            dict(shape=g.tensors[n.output].shape))
+                exec_ops.append(op)
+            elif n.kind == "constant":
+                # device-resident constant (e.g. ViT position embeddings):
+                # packed into the weight blob, copied D2D into its arena
+                # slot each run (captured in the graph)
+                op = ExecOp(K_CONST, n.name, [], n.output,
+                            dict(shape=g.tensors[n.output].shape))
+                op.w = n.attrs["value"].astype(np.float16)
+                exec_ops.append(op)
             elif n.kind == "flatten":
                 raise ValueError("flatten should be a view, not a node")
             else:
@@ -645,7 +661,7 @@ class Planner:
         # to the output after planning
         rope_alias: Dict[str, str] = {}
         for op in exec_ops:
-            if op.kind == K_ROPE:
+            if op.kind in (K_ROPE, K_VIEW):
                 src, dst = op.inputs[0], op.output
                 # follow chains (rope of rope never happens, but be safe)
                 src = rope_alias.get(src, src)
@@ -675,7 +691,8 @@ class Planner:
                 else self.dtype)
             d: Dict[str, Any] = dict(dtype=op_dtype, w_off=w_off,
                                      scale_off=s_off, bias_off=b_off,
-                                     in_off=offsets[op.inputs[0]],
+                                     in_off=(offsets[op.inputs[0]]
+                                             if op.inputs else -1),
                                      out_off=offsets[op.output],
                                      fork=op.params.get("fork", 0),
                                      join=op.params.get("join", 0))
@@ -691,6 +708,10 @@ class Planner:
                          sh=op.params["stride"], sw=op.params["stride"],
                          ph=op.params["padding"], pw=op.params["padding"],
                          res_scale=op.params.get("res_scale", 1.0))
+            elif op.kind == K_VIEW:
+                d.update(kind=K_VIEW)
+            elif op.kind == K_CONST:
+                d.update(kind=K_CONST, n_elems=nbytes_of(op.output))
             elif op.kind == K_BTAIL:
                 ish = shapes[op.inputs[0]]
                 d.update(kind=K_BTAIL, Nb=ish[0], H=ish[1], W=ish[2],

@@ -15,8 +15,8 @@ import torch
 import torch.nn.functional as F
 
 from trtlab_amd.engine.planner import (
-    K_CLIP, K_COPY2D, K_TRANSPOSE2D, K_RMSNORM, K_SILU_MUL, K_ROPE,
-    EnginePlan, K_ADD_LAYERNORM, K_ATTENTION, K_AVGPOOL, K_BTAIL, K_CHANNEL_PAD, K_CONV,
+    K_CLIP, K_COPY2D, K_TRANSPOSE2D, K_RMSNORM, K_SILU_MUL, K_ROPE, K_VIEW,
+    EnginePlan, K_ADD_LAYERNORM, K_ATTENTION, K_AVGPOOL, K_BTAIL, K_CHANNEL_PAD, K_CONST, K_CONV,
     K_DEQUANT, K_ELEMENTWISE, K_GAVGPOOL, K_GEMM, K_LAYERNORM, K_MAXPOOL,
     K_EMBEDDING, K_GEMM_MX4, K_GEMM_MX8, K_QUANT_MX4, K_QUANT_MX8,
     K_QUANTIZE, K_SEQLENS, K_SOFTMAX,
@@ -59,7 +59,7 @@ def run_reference(plan: EnginePlan, input_nhwc, return_all: bool = False):
              torch.from_numpy(np.ascontiguousarray(input_nhwc)).float()}
     # shapes registry from the planner's op dicts + exec op metadata
     for op, d in zip(plan.exec_ops, plan.ops):
-        x = t[op.inputs[0]]
+        x = t[op.inputs[0]] if op.inputs else None
         if op.kind == K_CHANNEL_PAD:
             cin, cpad = d["C"], d["Cout"]
             flat = x.reshape(-1, cin)
@@ -93,6 +93,11 @@ def run_reference(plan: EnginePlan, input_nhwc, return_all: bool = False):
             elif d["dtype"] == 3:  # fp8: emulate the e4m3 output store
                 y = _fp8_round(y)
             t[op.output] = y
+        elif op.kind == K_VIEW:
+            t[op.output] = x.reshape(op.params["shape"])
+        elif op.kind == K_CONST:
+            t[op.output] = torch.from_numpy(
+                op.w.astype(np.float32)).reshape(op.params["shape"])
         elif op.kind == K_BTAIL:
             # fused bottleneck tail: conv3x3+BN+ReLU then 1x1+BN+res+ReLU
             # (weights pre-packed: w = [64, 576] bt-flat, w2 = [Co, 64];
