@@ -80,7 +80,7 @@ def main():
     ap.add_argument("--contexts", type=int, default=3)
     ap.add_argument("--model", default="resnet50",
                     choices=["resnet50", "resnet101", "resnet152", "bert",
-                             "bert-large", "gpt2", "llama"])
+                             "bert-large", "gpt2", "llama", "vit"])
     ap.add_argument("--dtype", default="fp16",
                     choices=["fp16", "bf16", "int8", "fp8", "mxfp4", "mxfp8"])
     ap.add_argument("--no-autotune", action="store_true",
@@ -139,6 +139,13 @@ def main():
         g = build_gpt2(batch=args.batch, seq=1024, layers=12, seed=0)
         cfg_extra = {"seq_len": 1024, "hidden": 768, "layers": 12,
                      "phase": "prefill"}
+    elif args.model == "vit":
+        # ViT-B/16: conv patch embed + 12-layer encoder (196 tokens)
+        from trtlab_amd.models import build_vit
+
+        g = build_vit(batch=args.batch, image=224, patch=16, seed=0)
+        cfg_extra = {"image": 224, "patch": 16, "hidden": 768,
+                     "layers": 12}
     elif args.model == "llama":
         # LLaMA-architecture prefill (RMSNorm + RoPE + SwiGLU, hd128)
         from trtlab_amd.models import build_llama

@@ -280,3 +280,36 @@ def test_generation_stream_sampled_reproducible():
         assert a != c  # T=5 over 20 candidates: collision ~impossible
     finally:
         srv.shutdown()
+
+
+def test_generation_client_disconnect_frees_slot():
+    """A client that drops mid-stream must not leak its slot: the engine
+    reclaims it and a follow-up stream completes normally."""
+
+    class SlowFake(FakeSession):
+        def step(self, ids):
+            import time as _t
+
+            _t.sleep(0.01)
+            return super().step(ids)
+
+    sess = SlowFake(1)  # ONE slot: a leak would deadlock the next stream
+    gen = GenerationService(sess)
+    srv = Server("127.0.0.1:0")
+    srv.register_service(gen.service)
+    srv.async_start()
+    try:
+        ch = grpc.insecure_channel(f"127.0.0.1:{srv.port}")
+        call = ch.stream_stream(
+            "/trtlab.gen.Generation/Generate",
+            request_serializer=lambda m: m.SerializeToString(),
+            response_deserializer=GenerateToken.FromString)
+        stream = call(iter([GenerateRequest(prompt=[5], max_tokens=200)]))
+        next(stream)  # one token arrives, then the client walks away
+        stream.cancel()
+        ch.close()
+        # the single slot must come back: the next stream completes
+        toks, done = _collect(srv.port, [9], 5)
+        assert done and toks == expected_chain([9], 5)
+    finally:
+        srv.shutdown()

@@ -67,13 +67,15 @@ class GenerationEngine:
     """Continuous-batching loop over a lockstep decode session."""
 
     def __init__(self, session, eos: int = -1):
+        import collections
+
         self.session = session
         self.B = session.batch
         self.eos = eos
         self._slots: List[Optional[dict]] = [None] * self.B
         self._free: List[int] = list(range(self.B))
+        self._pending: collections.deque = collections.deque()
         self._wake = asyncio.Event()
-        self._freed = asyncio.Event()
         self._task: Optional[asyncio.Task] = None
         self.steps = 0  # total engine steps (observability / tests)
         # a fresh DecodeSession has every slot ACTIVE at pos 0; park them
@@ -113,26 +115,34 @@ class GenerationEngine:
             raise ValueError(
                 f"prompt+max_tokens ({len(prompt)}+{max_tokens}) exceeds "
                 f"the session window ({smax})")
-        while not self._free:
-            self._freed.clear()
-            await self._freed.wait()
-        b = self._free.pop()
-        self.session.reset_slot(b)
-        q: asyncio.Queue = asyncio.Queue()
-        self._slots[b] = dict(q=q, prompt=list(prompt), pi=0,
-                              remaining=int(max_tokens), cur=0,
-                              temperature=float(temperature),
-                              top_k=int(top_k), top_p=float(top_p),
-                              rng=np.random.RandomState(seed or None))
+        # the LOOP is the sole owner of session state: submissions queue
+        # and are admitted between steps (reset_slot must never race an
+        # in-flight replay that is reading this slot's position)
+        fut = asyncio.get_running_loop().create_future()
+        self._pending.append((dict(q=asyncio.Queue(), prompt=list(prompt),
+                                   pi=0, remaining=int(max_tokens), cur=0,
+                                   temperature=float(temperature),
+                                   top_k=int(top_k), top_p=float(top_p),
+                                   rng=np.random.RandomState(seed or None)),
+                              fut))
         self._wake.set()
-        return b, q
+        return await fut
 
     def _finish(self, b: int) -> None:
         self._slots[b]["q"].put_nowait(None)
         self._slots[b] = None
         self.session.idle_slot(b)
         self._free.append(b)
-        self._freed.set()
+
+    def cancel(self, b: int, q: asyncio.Queue) -> None:
+        """Mark slot b for release (client disconnected mid-stream);
+        the loop reclaims it between steps. The q identity guards
+        against cancelling a slot that already finished and was handed
+        to another stream."""
+        st = self._slots[b]
+        if st is not None and st["q"] is q:
+            st["dead"] = True
+            self._wake.set()
 
     @staticmethod
     def _sample(logits: np.ndarray, temperature: float, top_k: int,
@@ -168,14 +178,31 @@ class GenerationEngine:
     async def _loop(self) -> None:
         loop = asyncio.get_running_loop()
         while True:
+            # reclaim cancelled slots, then admit queued submissions —
+            # all session mutations happen HERE, between steps
+            for b in range(self.B):
+                st = self._slots[b]
+                if st is not None and st.get("dead"):
+                    self._finish(b)
+            while self._pending and self._free:
+                st, fut = self._pending.popleft()
+                if fut.cancelled():
+                    continue
+                b = self._free.pop()
+                self.session.reset_slot(b)
+                self._slots[b] = st
+                fut.set_result((b, st["q"]))
             active = [b for b in range(self.B) if self._slots[b]]
             if not active:
                 self._wake.clear()
+                if self._pending:  # waiting on a free slot, not on work
+                    self._wake.set()
                 await self._wake.wait()
                 continue
             ids = np.zeros(self.B, np.int32)
+            snap = {}
             for b in active:
-                st = self._slots[b]
+                st = snap[b] = self._slots[b]
                 ids[b] = (st["prompt"][st["pi"]]
                           if st["pi"] < len(st["prompt"]) else st["cur"])
             logits = await loop.run_in_executor(None, self.session.step,
@@ -183,7 +210,10 @@ class GenerationEngine:
             self.steps += 1
             for b in active:
                 st = self._slots[b]
-                if st is None:
+                # identity check: a stream cancelled during the step may
+                # have freed the slot AND a new submit() re-claimed it —
+                # this step's logits belong to the OLD stream's token
+                if st is None or st is not snap[b]:
                     continue
                 if st["pi"] < len(st["prompt"]):
                     st["pi"] += 1
@@ -221,10 +251,15 @@ class GenerationService:
             await context.abort(grpc.StatusCode.INVALID_ARGUMENT, str(e))
             return
         i = 0
-        while True:
-            tok = await q.get()
-            if tok is None:
-                yield GenerateToken(slot=b, index=i, done=True)
-                return
-            yield GenerateToken(token=tok, slot=b, index=i)
-            i += 1
+        try:
+            while True:
+                tok = await q.get()
+                if tok is None:
+                    yield GenerateToken(slot=b, index=i, done=True)
+                    return
+                yield GenerateToken(token=tok, slot=b, index=i)
+                i += 1
+        finally:
+            # client gone mid-stream (cancellation / disconnect): free
+            # the slot instead of generating into a dead queue
+            self.engine.cancel(b, q)
