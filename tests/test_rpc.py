@@ -256,3 +256,45 @@ def test_streaming_call_write_queue(echo_server):
         assert call.status.result(timeout=5) is True
     finally:
         c.close()
+
+
+def test_forwarding_middleman_relays_and_measures():
+    """Middleman (reference 04_Middleman role): raw-bytes unary relay —
+    client -> middleman -> backend echo — returns the backend's exact
+    response and records hop latency."""
+    import grpc
+
+    from trtlab_amd.rpc.middleman import ForwardingService, _RawBytes
+    from trtlab_amd.rpc.server import AsyncService, Server
+
+    # backend: an echo service that tags responses
+    backend_svc = AsyncService("trtlab.Echo")
+
+    async def echo(request, context, resources):
+        return _RawBytes(b"echo:" + request.data)
+
+    backend_svc.register_unary("Ping", echo, _RawBytes, _RawBytes)
+    backend = Server("127.0.0.1:0")
+    backend.register_service(backend_svc)
+    backend.async_start()
+
+    fwd = ForwardingService(f"127.0.0.1:{backend.port}",
+                            service="trtlab.Echo", method="Ping")
+    front = Server("127.0.0.1:0")
+    front.register_service(fwd.service)
+    front.async_start()
+    try:
+        ch = grpc.insecure_channel(f"127.0.0.1:{front.port}")
+        call = ch.unary_unary("/trtlab.Echo/Ping",
+                              request_serializer=lambda b: b,
+                              response_deserializer=lambda b: b)
+        for i in range(5):
+            out = call(f"m{i}".encode())
+            assert out == f"echo:m{i}".encode()
+        ch.close()
+        s = fwd.stats()
+        assert s["requests"] == 5 and s["p50_ms"] > 0
+    finally:
+        fwd.close()
+        front.shutdown()
+        backend.shutdown()
