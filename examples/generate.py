@@ -62,11 +62,13 @@ def main():
         from trtlab_amd.engine.decode import SpeculativeDecoder
 
         gd = build(args.draft_layers)
+        # capture=True: verification chunks + draft steps replay as
+        # hipGraphs (per chunk size) instead of eager per-kernel launches
         draft = DecodeSession(gd, batch=args.batch, smax=1024,
-                              capture=False, lm_head=True)
+                              capture=True, lm_head=True)
         draft.prefill(prompt)
         target = DecodeSession(g, batch=args.batch, smax=1024,
-                               capture=False, lm_head=True)
+                               capture=True, lm_head=True)
         target.prefill(prompt)
         sd = SpeculativeDecoder(target, draft, k=args.speculative)
         t0 = time.perf_counter()

@@ -1313,3 +1313,29 @@ def test_speculative_decoding_invariant_llama():
             assert rate > 0.9, rate
         t.close()
         d.close()
+
+
+def test_captured_verify_chunk_matches_eager():
+    """hipGraph-captured chunked verification (per chunk size) replays
+    identically to the eager path across positions and chunk sizes."""
+    from trtlab_amd.engine.decode import DecodeSession
+    from trtlab_amd.models import build_gpt2
+
+    g = build_gpt2(batch=2, seq=64, layers=2, seed=0, embeddings=True)
+    rng = np.random.RandomState(3)
+    toks4 = rng.randint(1, 50257, (3, 2, 4)).astype(np.int32)
+    toks1 = rng.randint(1, 50257, (3, 2, 1)).astype(np.int32)
+
+    cap = DecodeSession(g, batch=2, smax=64, capture=True, lm_head=True)
+    eag = DecodeSession(g, batch=2, smax=64, capture=False, lm_head=True)
+    for r in range(3):  # r=0 captures, r>0 replays; interleave K=4 and 1
+        for toks in (toks4[r], toks1[r]):
+            a = cap.verify_chunk(toks)
+            b = eag.verify_chunk(toks)
+            np.testing.assert_allclose(a, b, rtol=2e-2, atol=2e-2)
+            k = toks.shape[1]
+            cap.add_pos(np.full(2, k, np.int64))
+            eag.add_pos(np.full(2, k, np.int64))
+    assert set(cap._chunk_graphs) == {4, 1}
+    cap.close()
+    eag.close()

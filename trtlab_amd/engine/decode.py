@@ -685,47 +685,13 @@ class DecodeSession:
                 if self.logits is not None else out)
 
     # ------------------------------------------- speculative verification
-    def verify_chunk(self, tokens: np.ndarray) -> np.ndarray:
-        """Score K proposed tokens per slot in ONE chunked forward
-        (speculative decoding's verification step): tokens [B, K] are
-        consumed at positions pos[b]..pos[b]+K-1 (their K/V overwrite any
-        stale entries), and the TARGET logits for each position come back
-        as [B, K, vocab] fp32. pos is NOT advanced — call add_pos() with
-        the per-slot accepted counts. Requires lm_head=True."""
-        import torch
-
-        if self.logits is None:
-            raise RuntimeError("verify_chunk requires lm_head=True")
-        if self.kv_pool is not None:
-            raise RuntimeError("verify_chunk is not supported in paged "
-                               "mode yet")
-        tokens = np.ascontiguousarray(tokens, np.int32)
-        B, K = tokens.shape
-        assert B == self.batch and K >= 1
+    def _enqueue_chunk(self, cb: Dict, B: int, K: int) -> None:
+        """Record one chunked-verification pass's kernels on the session
+        stream (no syncs — capturable). All position dependence reads the
+        device counter, so a captured chunk graph replays at any pos."""
         M = B * K
         Hd, inter = self.hidden, self.inter
         ops, s = self._C.ops, self.stream
-        torch = self._torch
-
-        cb = getattr(self, "_chunk_bufs", None)
-        if cb is None or cb["K"] != K:
-            cb = dict(
-                K=K,
-                ids=torch.zeros(M, dtype=torch.int32, device="cuda"),
-                h=torch.zeros(M, Hd, dtype=torch.half, device="cuda"),
-                x=torch.zeros(M, Hd, dtype=torch.half, device="cuda"),
-                x2=torch.zeros(M, Hd, dtype=torch.half, device="cuda"),
-                qkv=torch.zeros(M, 3 * Hd, dtype=torch.half, device="cuda"),
-                att=torch.zeros(M, Hd, dtype=torch.half, device="cuda"),
-                ff=torch.zeros(M, inter, dtype=torch.half, device="cuda"),
-                ff2=(torch.zeros(M, inter, dtype=torch.half, device="cuda")
-                     if self.arch == "llama" else None),
-                logits=torch.zeros(M, self.vocab, dtype=torch.half,
-                                   device="cuda"),
-            )
-            self._chunk_bufs = cb
-        self._C.memory.memcpy_h2d(cb["ids"].data_ptr(), tokens.reshape(-1),
-                                  tokens.nbytes)
         scale = 1.0 / float(np.sqrt(float(self.hd)))
         ops.chunk_embed(cb["ids"].data_ptr(), self.tok.data_ptr(),
                         self.posemb.data_ptr(), cb["h"].data_ptr(),
@@ -843,7 +809,71 @@ class DecodeSession:
         ops.gemm_bt(0, cb["x"].data_ptr(), self.tok.data_ptr(),
                     cb["logits"].data_ptr(), M=M, N=self.vocab, K=Hd,
                     epi=self._epi_none, stream=s, sync=False)
-        self._C.hip.stream_synchronize(s)
+
+    def verify_chunk(self, tokens: np.ndarray) -> np.ndarray:
+        """Score K proposed tokens per slot in ONE chunked forward
+        (speculative decoding's verification step): tokens [B, K] are
+        consumed at positions pos[b]..pos[b]+K-1 (their K/V overwrite any
+        stale entries), and the TARGET logits for each position come back
+        as [B, K, vocab] fp32. pos is NOT advanced — call add_pos() with
+        the per-slot accepted counts. Requires lm_head=True.
+
+        With capture=True the pass is hipGraph-captured PER CHUNK SIZE on
+        first use and replayed afterwards (the measured spec-decode cost
+        was eager per-kernel launch overhead; all position dependence is
+        device-side, so one graph serves every position)."""
+        import torch
+
+        if self.logits is None:
+            raise RuntimeError("verify_chunk requires lm_head=True")
+        if self.kv_pool is not None:
+            raise RuntimeError("verify_chunk is not supported in paged "
+                               "mode yet")
+        tokens = np.ascontiguousarray(tokens, np.int32)
+        B, K = tokens.shape
+        assert B == self.batch and K >= 1
+        M = B * K
+        Hd, inter = self.hidden, self.inter
+        s = self.stream
+
+        cbs = getattr(self, "_chunk_bufs_by_k", None)
+        if cbs is None:
+            cbs = self._chunk_bufs_by_k = {}
+            self._chunk_graphs = {}
+        cb = cbs.get(K)
+        if cb is None:
+            cb = cbs[K] = dict(
+                K=K,
+                ids=torch.zeros(M, dtype=torch.int32, device="cuda"),
+                h=torch.zeros(M, Hd, dtype=torch.half, device="cuda"),
+                x=torch.zeros(M, Hd, dtype=torch.half, device="cuda"),
+                x2=torch.zeros(M, Hd, dtype=torch.half, device="cuda"),
+                qkv=torch.zeros(M, 3 * Hd, dtype=torch.half, device="cuda"),
+                att=torch.zeros(M, Hd, dtype=torch.half, device="cuda"),
+                ff=torch.zeros(M, inter, dtype=torch.half, device="cuda"),
+                ff2=(torch.zeros(M, inter, dtype=torch.half, device="cuda")
+                     if self.arch == "llama" else None),
+                logits=torch.zeros(M, self.vocab, dtype=torch.half,
+                                   device="cuda"),
+            )
+        self._C.memory.memcpy_h2d(cb["ids"].data_ptr(), tokens.reshape(-1),
+                                  tokens.nbytes)
+        if self.capture:
+            gph = self._chunk_graphs.get(K)
+            if gph is None:
+                # first call: eager warm-up does the real work, then a
+                # fresh pass is RECORDED (capture does not execute)
+                self._enqueue_chunk(cb, B, K)
+                self._C.hip.stream_synchronize(s)
+                self._C.hip.stream_begin_capture(s)
+                self._enqueue_chunk(cb, B, K)
+                self._chunk_graphs[K] = self._C.hip.stream_end_capture(s)
+            else:
+                self._C.hip.graph_launch(gph, s)
+                self._C.hip.stream_synchronize(s)
+        else:
+            self._enqueue_chunk(cb, B, K)
+            self._C.hip.stream_synchronize(s)
         return cb["logits"].float().cpu().numpy().reshape(B, K, self.vocab)
 
     def add_pos(self, counts: np.ndarray) -> None:
@@ -895,6 +925,10 @@ class DecodeSession:
         if self._graph:
             self._C.hip.graph_destroy(self._graph)
             self._graph = 0
+        for g in getattr(self, "_chunk_graphs", {}).values():
+            self._C.hip.graph_destroy(g)
+        if getattr(self, "_chunk_graphs", None):
+            self._chunk_graphs = {}
 
     def __del__(self):
         try:
