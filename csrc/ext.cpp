@@ -504,6 +504,15 @@ PYBIND11_MODULE(_C, m) {
           py::arg("D") = 0, py::arg("theta") = 10000.0f,
           py::arg("stream") = 0, py::arg("sync") = true,
           py::arg("chunk") = 0);
+  ops.def("argmax_rows",
+          [](uintptr_t x, uintptr_t out, int M, int V, uintptr_t stream,
+             bool sync) {
+            launch_argmax_rows((void*)x, (void*)out, M, V,
+                               as_stream(stream));
+            if (sync) TRT_HIP_CHECK(hipStreamSynchronize(as_stream(stream)));
+          },
+          py::arg("x"), py::arg("out"), py::arg("M"), py::arg("V"),
+          py::arg("stream") = 0, py::arg("sync") = true);
   ops.def("clip",
           [](int dtype, uintptr_t in, uintptr_t out, int64_t n, float mn,
              float mx, uintptr_t stream, bool sync) {

@@ -441,5 +441,74 @@ void launch_rope(int dtype, void* qkv, const void* pos_dev, int M, int S,
                      theta, chunk);
 }
 
+// ---- row-wise argmax: fp16 logits [M, V] -> int32 index [M] ----
+// One BLOCK (4 waves) per row, 16-B vector loads (greedy decoding head:
+// 4 B/row D2H instead of the whole logits matrix; at decode M is tiny,
+// so per-row parallelism matters more than row count). Ties resolve to
+// the LOWEST index, matching numpy argmax so the speculative-decode
+// invariance holds exactly.
+__global__ __launch_bounds__(256) void argmax_rows_kernel(
+    const _Float16* __restrict__ x, int* __restrict__ out, int M, int V) {
+  int row = blockIdx.x;
+  if (row >= M) return;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63, wave = tid >> 6;
+  const _Float16* r = x + (int64_t)row * V;
+  float best = -1e30f;
+  int bi = 0x7fffffff;
+  const int nv8 = V >> 3;
+  for (int i = tid; i < nv8; i += 256) {
+    short4v v0 = *(const short4v*)(r + i * 8);
+    short4v v1 = *(const short4v*)(r + i * 8 + 4);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float v = (float)((const _Float16*)(j < 4 ? (const void*)&v0
+                                                : (const void*)&v1))[j & 3];
+      int c = i * 8 + j;
+      if (v > best || (v == best && c < bi)) {
+        best = v;
+        bi = c;
+      }
+    }
+  }
+  for (int c = (nv8 << 3) + tid; c < V; c += 256) {
+    float v = (float)r[c];
+    if (v > best || (v == best && c < bi)) {
+      best = v;
+      bi = c;
+    }
+  }
+#pragma unroll
+  for (int off = 32; off; off >>= 1) {
+    float ob = __shfl_xor(best, off, 64);
+    int oi = __shfl_xor(bi, off, 64);
+    if (ob > best || (ob == best && oi < bi)) {
+      best = ob;
+      bi = oi;
+    }
+  }
+  __shared__ float wb[4];
+  __shared__ int wi[4];
+  if (lane == 0) {
+    wb[wave] = best;
+    wi[wave] = bi;
+  }
+  __syncthreads();
+  if (tid == 0) {
+    for (int w = 1; w < 4; ++w)
+      if (wb[w] > wb[0] || (wb[w] == wb[0] && wi[w] < wi[0])) {
+        wb[0] = wb[w];
+        wi[0] = wi[w];
+      }
+    out[row] = wi[0];
+  }
+}
+
+void launch_argmax_rows(const void* x, void* out, int M, int V,
+                        hipStream_t stream) {
+  hipLaunchKernelGGL(argmax_rows_kernel, dim3((unsigned)M), dim3(256), 0,
+                     stream, (const _Float16*)x, (int*)out, M, V);
+}
+
 }  // namespace trtlab
 

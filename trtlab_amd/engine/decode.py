@@ -809,8 +809,11 @@ class DecodeSession:
         ops.gemm_bt(0, cb["x"].data_ptr(), self.tok.data_ptr(),
                     cb["logits"].data_ptr(), M=M, N=self.vocab, K=Hd,
                     epi=self._epi_none, stream=s, sync=False)
+        ops.argmax_rows(cb["logits"].data_ptr(), cb["gids"].data_ptr(),
+                        M, self.vocab, stream=s, sync=False)
 
-    def verify_chunk(self, tokens: np.ndarray) -> np.ndarray:
+    def verify_chunk(self, tokens: np.ndarray,
+                     greedy: bool = False) -> np.ndarray:
         """Score K proposed tokens per slot in ONE chunked forward
         (speculative decoding's verification step): tokens [B, K] are
         consumed at positions pos[b]..pos[b]+K-1 (their K/V overwrite any
@@ -855,6 +858,7 @@ class DecodeSession:
                      if self.arch == "llama" else None),
                 logits=torch.zeros(M, self.vocab, dtype=torch.half,
                                    device="cuda"),
+                gids=torch.zeros(M, dtype=torch.int32, device="cuda"),
             )
         self._C.memory.memcpy_h2d(cb["ids"].data_ptr(), tokens.reshape(-1),
                                   tokens.nbytes)
@@ -874,13 +878,21 @@ class DecodeSession:
         else:
             self._enqueue_chunk(cb, B, K)
             self._C.hip.stream_synchronize(s)
+        if greedy:
+            return cb["gids"].cpu().numpy().reshape(B, K)
         return cb["logits"].float().cpu().numpy().reshape(B, K, self.vocab)
+
+    def get_pos(self) -> np.ndarray:
+        """Per-slot positions from the HOST mirror (no D2H round-trip:
+        device pos only changes via step()'s advance kernel — mirrored
+        by _slot_steps — or via these host-side setters)."""
+        return np.where(self._active, self._slot_steps, -1).astype(np.int32)
 
     def add_pos(self, counts: np.ndarray) -> None:
         """Advance each slot's position by counts[b] (speculative
         acceptance: the chunk's first counts[b] tokens are now consumed)."""
         counts = np.asarray(counts, np.int64)
-        cur = self.pos.cpu().numpy().astype(np.int64)
+        cur = self.get_pos().astype(np.int64)
         new = np.where(cur >= 0, np.minimum(cur + counts, self.smax - 1),
                        cur)
         self.pos.copy_(self._torch.from_numpy(new.astype(np.int32)).cuda())
@@ -969,31 +981,34 @@ class SpeculativeDecoder:
         [B]. Returns (tokens [B, steps] int32, acceptance_rate)."""
         B, k = self.target.batch, self.k
         seed = np.ascontiguousarray(seed, np.int32)
-        lt = self.target.verify_chunk(seed[:, None])[:, 0]  # consume seed
+        # ids-only verification (greedy=True): argmax happens on-device,
+        # so only token ids ever cross PCIe (B*k ints vs B*k*vocab floats)
+        gt = self.target.verify_chunk(seed[:, None], greedy=True)[:, 0]
         self.target.add_pos(np.ones(B, np.int64))
-        ld = self.draft.verify_chunk(seed[:, None])[:, 0]
+        gd = self.draft.verify_chunk(seed[:, None], greedy=True)[:, 0]
         self.draft.add_pos(np.ones(B, np.int64))
 
         out = [[] for _ in range(B)]
         while min(len(o) for o in out) < steps:
             # ---- draft chain: k greedy proposals ----
             props = np.zeros((B, k), np.int32)
-            cur = ld.argmax(-1).astype(np.int32)
+            cur = gd.astype(np.int32)
             for i in range(k):
                 props[:, i] = cur
-                ld = self.draft.verify_chunk(cur[:, None])[:, 0]
+                gd = self.draft.verify_chunk(cur[:, None],
+                                             greedy=True)[:, 0]
                 self.draft.add_pos(np.ones(B, np.int64))
-                cur = ld.argmax(-1).astype(np.int32)
+                cur = gd.astype(np.int32)
             # ---- target verifies the whole chunk at once ----
-            tl = self.target.verify_chunk(props)  # [B, k, vocab]
+            tg = self.target.verify_chunk(props, greedy=True)  # [B, k]
             accept = np.zeros(B, np.int64)
             corr = np.zeros(B, np.int32)
             for b in range(B):
-                g = int(lt[b].argmax())
+                g = int(gt[b])
                 a = 0
                 while a < k and props[b, a] == g:
                     out[b].append(g)
-                    g = int(tl[b, a].argmax())
+                    g = int(tg[b, a])
                     a += 1
                 accept[b] = a
                 corr[b] = g          # target's own next token
@@ -1003,12 +1018,11 @@ class SpeculativeDecoder:
             # ---- advance target past the accepted prefix, consume the
             # correction token (its K/V overwrites the rejected slot) ----
             self.target.add_pos(accept)
-            lt = self.target.verify_chunk(corr[:, None])[:, 0]
+            gt = self.target.verify_chunk(corr[:, None], greedy=True)[:, 0]
             self.target.add_pos(np.ones(B, np.int64))
             # ---- roll the draft back to the target's frontier ----
-            tpos = self.target.pos.cpu().numpy().astype(np.int32)
-            self.draft.set_pos(tpos - 1)
-            ld = self.draft.verify_chunk(corr[:, None])[:, 0]
+            self.draft.set_pos(self.target.get_pos() - 1)
+            gd = self.draft.verify_chunk(corr[:, None], greedy=True)[:, 0]
             self.draft.add_pos(np.ones(B, np.int64))
         rate = self.accepted / max(self.proposed, 1)
         return (np.array([o[:steps] for o in out], np.int32), rate)
