@@ -318,3 +318,35 @@ def test_gpt2_onnx_roundtrip():
     o1 = run_reference(plan1, ids)
     o2 = run_reference(plan2, ids)
     assert np.allclose(o1, o2, atol=2e-3), np.abs(o1 - o2).max()
+
+
+def test_vit_onnx_roundtrip():
+    """ViT exports (Reshape-as-view, position constant as initializer)
+    and re-imports to an identical-output graph."""
+    from trtlab_amd.engine.onnx_io import export_onnx, import_onnx
+    from trtlab_amd.models import build_vit
+
+    g = build_vit(batch=1, image=64, patch=16, hidden=256, layers=1,
+                  heads=4, classes=10, seed=0)
+    g2 = import_onnx(export_onnx(g))
+    p1, p2 = Planner().compile(g), Planner().compile(g2)
+    x = (np.random.RandomState(0).randn(*p1.input_shape) * 0.5).astype(
+        np.float32)
+    a, b = run_reference(p1, x), run_reference(p2, x)
+    assert np.abs(a - b).max() / max(np.abs(a).max(), 1e-6) < 1e-5
+
+
+def test_llama_onnx_roundtrip():
+    """LLaMA exports via custom-domain TrtlabRMSNorm/SiluMul/Rope ops
+    and re-imports to an identical-output graph."""
+    from trtlab_amd.engine.onnx_io import export_onnx, import_onnx
+    from trtlab_amd.models import build_llama
+
+    g = build_llama(batch=1, seq=32, hidden=256, layers=1, heads=2,
+                    seed=0, vocab=500)
+    g2 = import_onnx(export_onnx(g))
+    p1, p2 = Planner().compile(g), Planner().compile(g2)
+    ids = np.random.RandomState(1).randint(
+        0, 500, p1.input_shape).astype(np.int32)
+    a, b = run_reference(p1, ids), run_reference(p2, ids)
+    assert np.abs(a - b).max() / max(np.abs(a).max(), 1e-6) < 1e-5

@@ -166,7 +166,13 @@ def import_onnx(data: bytes, batch: Optional[int] = None,
         elif op == "Gelu":
             out = g.gelu(x)
         elif op == "Add":
-            out = g.add(x, remap.get(ins[1], ins[1]))
+            other = remap.get(ins[1], ins[1])
+            if other in inits and other not in g.tensors:
+                # constant operand (e.g. ViT position embeddings):
+                # materialize it as a device-resident constant tensor
+                other = g.constant(inits[other].astype(np.float32),
+                                   name=other + "_const")
+            out = g.add(x, other)
         elif op == "AveragePool":
             ks = attrs.get("kernel_shape", [2, 2])
             strides = attrs.get("strides", [1, 1])
@@ -182,9 +188,19 @@ def import_onnx(data: bytes, batch: Optional[int] = None,
         elif op == "GlobalAveragePool":
             out = g.global_avgpool(x)
         elif op in ("Flatten", "Reshape", "Squeeze", "Unsqueeze"):
-            # shape plumbing between GAP and Gemm (torchvision-style heads);
-            # the IR's gavgpool output is already [N, C], so these are views
-            out = x
+            # shape plumbing. When the target shape is a static
+            # initializer and actually differs, emit a zero-copy view
+            # (ViT token flatten / pool reshape); otherwise pass through
+            # (torchvision GAP->Gemm heads where the IR is already 2-D).
+            tgt = None
+            if op == "Reshape" and len(ins) > 1 and ins[1] in inits:
+                tgt = [int(v) for v in
+                       np.asarray(inits[ins[1]]).reshape(-1)]
+            cur = list(g.tensors[x].shape)
+            if tgt and tgt != cur and all(d > 0 for d in tgt):
+                out = g.view(x, tuple(tgt))
+            else:
+                out = x
         elif op in ("Identity", "Dropout"):
             out = x  # inference mode: both are pass-through
         elif op == "Clip":
@@ -259,6 +275,15 @@ def import_onnx(data: bytes, batch: Optional[int] = None,
                               causal=bool(attrs.get("causal", 0)),
                               varlen=bool(attrs.get("varlen", 0)),
                               pad_id=int(attrs.get("pad_id", 0)))
+        elif op == "TrtlabRMSNorm":
+            out = g.rmsnorm(x, inits[ins[1]].astype(np.float32),
+                            eps=float(attrs.get("epsilon", 1e-5)))
+        elif op == "TrtlabSiluMul":
+            out = g.silu_mul(x, remap.get(ins[1], ins[1]))
+        elif op == "TrtlabRope":
+            out = g.rope(x, heads=int(attrs["heads"]),
+                         seq=int(attrs["seq"]),
+                         theta=float(attrs.get("theta", 10000.0)))
         else:
             raise ValueError(f"ONNX op {op} not supported by the importer")
         remap[outs[0]] = out
@@ -416,6 +441,28 @@ def export_onnx(g: Graph) -> bytes:
                            _attr_i("causal", 1 if a.get("causal") else 0),
                            _attr_i("varlen", 1 if a.get("varlen") else 0),
                            _attr_i("pad_id", a.get("pad_id", 0)))
+        elif n.kind == "view":
+            # Reshape with the target shape as an int64 initializer
+            shp = np.asarray(g.tensors[n.output].shape, np.int64)
+            nodes += _node("Reshape", [n.inputs[0], add_init(shp)],
+                           [n.output])
+        elif n.kind == "constant":
+            # the constant tensor becomes a graph initializer named like
+            # the node's output (standard ONNX constant-folding shape)
+            inits += w.f_bytes(_GRAPH_INIT,
+                               _tensor_bytes(n.output, n.attrs["value"]))
+        elif n.kind == "rmsnorm":
+            nodes += _node("TrtlabRMSNorm",
+                           [n.inputs[0], add_init(n.attrs["gamma"])],
+                           [n.output], _attr_f("epsilon", n.attrs["eps"]))
+        elif n.kind == "silu_mul":
+            nodes += _node("TrtlabSiluMul", list(n.inputs), [n.output])
+        elif n.kind == "rope":
+            a = n.attrs
+            nodes += _node("TrtlabRope", [n.inputs[0]], [n.output],
+                           _attr_i("heads", a["heads"]),
+                           _attr_i("seq", a["seq"]),
+                           _attr_f("theta", a["theta"]))
         else:
             raise ValueError(f"export: unsupported node kind {n.kind}")
 
