@@ -219,3 +219,51 @@ def test_two_level_execution_limiter():
         f.result(timeout=30)
     assert peak[0] == 1  # global limiter held launches to one at a time
     mgr.shutdown()
+
+
+def test_trace2chrome_converter(tmp_path):
+    """rocprofv3 kernel-trace CSV -> chrome://tracing JSON (aux tracing
+    subsystem): events keep name/timestamps, lanes map to queues, and
+    column-name variants across rocprof versions are tolerated."""
+    import csv
+    import json
+    import sys
+    from pathlib import Path
+
+    sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
+    from tools.trace2chrome import convert
+
+    p = tmp_path / "k.csv"
+    with open(p, "w", newline="") as f:
+        w = csv.DictWriter(f, fieldnames=[
+            "Kind", "Agent_Id", "Queue_Id", "Kernel_Name",
+            "Start_Timestamp", "End_Timestamp"])
+        w.writeheader()
+        w.writerow(dict(Kind="KERNEL_DISPATCH", Agent_Id="1", Queue_Id="2",
+                        Kernel_Name="trtlab::conv_igemm_kernel",
+                        Start_Timestamp="1000", End_Timestamp="31000"))
+        w.writerow(dict(Kind="KERNEL_DISPATCH", Agent_Id="1", Queue_Id="3",
+                        Kernel_Name="trtlab::attention_kernel",
+                        Start_Timestamp="31000", End_Timestamp="32000"))
+        w.writerow(dict(Kind="HEADERLESS", Agent_Id="", Queue_Id="",
+                        Kernel_Name="", Start_Timestamp="",
+                        End_Timestamp=""))  # malformed row skipped
+    out = tmp_path / "t.json"
+    n = convert(str(p), str(out))
+    assert n == 2
+    data = json.load(open(out))
+    ev = data["traceEvents"]
+    assert ev[0]["name"].startswith("trtlab::conv")
+    assert ev[0]["ts"] == 1.0 and ev[0]["dur"] == 30.0  # ns -> us
+    assert ev[0]["pid"] == "gpu1" and ev[0]["tid"] == "queue2"
+    assert ev[1]["tid"] == "queue3"
+
+    # BeginNs/EndNs variant (older rocprof)
+    p2 = tmp_path / "k2.csv"
+    with open(p2, "w", newline="") as f:
+        w = csv.DictWriter(f, fieldnames=["Name", "BeginNs", "EndNs",
+                                          "queue-id", "gpu-id"])
+        w.writeheader()
+        w.writerow(dict(Name="k", BeginNs="5000", EndNs="6000",
+                        **{"queue-id": "0", "gpu-id": "0"}))
+    assert convert(str(p2), str(tmp_path / "t2.json")) == 1
