@@ -82,8 +82,8 @@ def main():
         out = [toks]
         t0 = time.perf_counter()
         for _ in range(args.new_tokens - 1):
-            logits = sess.step(toks)
-            toks = np.argmax(logits, axis=1).astype(np.int32)
+            # in-graph argmax head: B ints over PCIe instead of logits
+            toks = sess.step(toks, return_ids=True).astype(np.int32)
             out.append(toks)
         dt = time.perf_counter() - t0
         seqs = np.stack(out, axis=1)

@@ -1339,3 +1339,28 @@ def test_captured_verify_chunk_matches_eager():
     assert set(cap._chunk_graphs) == {4, 1}
     cap.close()
     eag.close()
+
+
+def test_step_return_ids_matches_argmax():
+    """In-graph greedy head: step(return_ids=True) equals host argmax of
+    the logits for both recipes, captured and eager."""
+    from trtlab_amd.engine.decode import DecodeSession
+    from trtlab_amd.models import build_gpt2, build_llama
+
+    for g in (build_gpt2(batch=2, seq=32, layers=2, seed=0,
+                         embeddings=True),
+              build_llama(batch=2, seq=32, hidden=512, layers=2, heads=4,
+                          seed=0, vocab=3000)):
+        for cap in (True, False):
+            a = DecodeSession(g, batch=2, smax=32, capture=cap,
+                              lm_head=True)
+            b = DecodeSession(g, batch=2, smax=32, capture=cap,
+                              lm_head=True)
+            toks = np.array([5, 9], np.int32)
+            for t in range(4):
+                ids = a.step(toks, return_ids=True)
+                lg = b.step(toks)
+                np.testing.assert_array_equal(ids, lg.argmax(-1))
+                toks = ids.astype(np.int32)
+            a.close()
+            b.close()

@@ -231,6 +231,12 @@ class DecodeSession:
         self.out = torch.zeros(B, Hd, dtype=torch.half, device="cuda")
         self.logits = (torch.zeros(B, self.vocab, dtype=torch.half,
                                    device="cuda") if lm_head else None)
+        # device-side greedy head: argmax ids land here every step (the
+        # argmax kernel is captured in the step graph; step(return_ids=
+        # True) then moves B ints instead of B*vocab logits over PCIe)
+        self.gids = (torch.zeros(B, dtype=torch.int32, device="cuda")
+                     if lm_head else None)
+        self.supports_ids = lm_head
         self.h2 = torch.zeros(B, Hd, dtype=torch.half, device="cuda")
         # paged KV mode: paged = a PagedKVPool (shared with other
         # sessions) or True (private pool sized for smax). The dense
@@ -330,6 +336,8 @@ class DecodeSession:
         ops.gemm_bt(0, self.out.data_ptr(), self.tok.data_ptr(),
                     self.logits.data_ptr(), M=B, N=self.vocab, K=Hd,
                     epi=self._epi_none, stream=s, sync=False)
+        ops.argmax_rows(self.logits.data_ptr(), self.gids.data_ptr(),
+                        B, self.vocab, stream=s, sync=False)
         # no cross-step residual state: layer 0's EMBED prologue
         # regenerates the stream each step (h/h2 are just scratch)
         ops.advance_pos(self.pos.data_ptr(), B, self.smax, stream=s,
@@ -424,6 +432,8 @@ class DecodeSession:
             ops.gemm_bt(0, self.out.data_ptr(), self.tok.data_ptr(),
                         self.logits.data_ptr(), M=B, N=self.vocab, K=Hd,
                         epi=self._epi_none, stream=s, sync=False)
+            ops.argmax_rows(self.logits.data_ptr(), self.gids.data_ptr(),
+                            B, self.vocab, stream=s, sync=False)
         ops.advance_pos(self.pos.data_ptr(), B, self.smax, stream=s,
                         sync=False)
 
@@ -491,6 +501,8 @@ class DecodeSession:
             ops.gemm_bt(0, self.out.data_ptr(), self.tok.data_ptr(),
                         self.logits.data_ptr(), M=B, N=self.vocab, K=Hd,
                         epi=self._epi_none, stream=s, sync=False)
+            ops.argmax_rows(self.logits.data_ptr(), self.gids.data_ptr(),
+                            B, self.vocab, stream=s, sync=False)
         ops.advance_pos(self.pos.data_ptr(), B, self.smax, stream=s,
                         sync=False)
 
@@ -643,11 +655,18 @@ class DecodeSession:
             return self.logits.float().cpu().numpy()
         return last.float().cpu().numpy()
 
-    def step(self, ids: np.ndarray) -> np.ndarray:
+    def step(self, ids: np.ndarray,
+             return_ids: bool = False) -> np.ndarray:
         """Feed one token per sequence; returns the final hidden state
         [B, hidden] fp32 (or logits [B, vocab] with lm_head=True). The
         device-side position counter starts at 0 and the captured graph
-        advances it, so replays need no host-side position plumbing."""
+        advances it, so replays need no host-side position plumbing.
+
+        return_ids=True (lm_head only): return the greedy argmax token
+        per slot [B] int32 instead of logits — the argmax runs inside
+        the captured graph, so only B ints cross PCIe per step."""
+        if return_ids and self.logits is None:
+            raise RuntimeError("return_ids requires lm_head=True")
         if self.kv_pool is not None:
             # map the page each active slot's next write lands in
             for b in range(self.batch):
@@ -671,6 +690,8 @@ class DecodeSession:
                 self._graph = self._C.hip.stream_end_capture(self.stream)
                 self._steps += 1
                 self._slot_steps += 1
+                if return_ids:
+                    return self.gids.cpu().numpy()
                 out = self.out.float().cpu().numpy()
                 return (self.logits.float().cpu().numpy()
                         if self.logits is not None else out)
@@ -680,6 +701,8 @@ class DecodeSession:
         self._C.hip.stream_synchronize(self.stream)
         self._steps += 1
         self._slot_steps += 1
+        if return_ids:
+            return self.gids.cpu().numpy()
         out = self.out.float().cpu().numpy()
         return (self.logits.float().cpu().numpy()
                 if self.logits is not None else out)

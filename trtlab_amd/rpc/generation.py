@@ -165,10 +165,11 @@ class GenerationEngine:
         pk = p[idx] / p[idx].sum()
         return int(rng.choice(idx, p=pk))
 
-    def _emit(self, b: int, logits_row: np.ndarray) -> None:
+    def _emit(self, b: int, logits_row, tok: int = -1) -> None:
         st = self._slots[b]
-        tok = self._sample(logits_row, st["temperature"], st["top_k"],
-                           st["top_p"], st["rng"])
+        if tok < 0:
+            tok = self._sample(logits_row, st["temperature"], st["top_k"],
+                               st["top_p"], st["rng"])
         st["cur"] = tok
         st["q"].put_nowait(tok)
         st["remaining"] -= 1
@@ -205,8 +206,18 @@ class GenerationEngine:
                 st = snap[b] = self._slots[b]
                 ids[b] = (st["prompt"][st["pi"]]
                           if st["pi"] < len(st["prompt"]) else st["cur"])
-            logits = await loop.run_in_executor(None, self.session.step,
-                                                ids)
+            # greedy fast path: when every live stream decodes greedily
+            # and the session has the in-graph argmax head, only B ints
+            # cross PCIe per step (B*vocab logits otherwise)
+            ids_only = (getattr(self.session, "supports_ids", False) and
+                        all(snap[b]["temperature"] <= 0.0 for b in active))
+            if ids_only:
+                gids = await loop.run_in_executor(
+                    None, lambda: self.session.step(ids, return_ids=True))
+                logits = None
+            else:
+                logits = await loop.run_in_executor(None,
+                                                    self.session.step, ids)
             self.steps += 1
             for b in active:
                 st = self._slots[b]
@@ -215,12 +226,14 @@ class GenerationEngine:
                 # this step's logits belong to the OLD stream's token
                 if st is None or st is not snap[b]:
                     continue
+                row = None if ids_only else logits[b]
+                tok = int(gids[b]) if ids_only else -1
                 if st["pi"] < len(st["prompt"]):
                     st["pi"] += 1
                     if st["pi"] == len(st["prompt"]):
-                        self._emit(b, logits[b])  # first generated token
+                        self._emit(b, row, tok)  # first generated token
                 else:
-                    self._emit(b, logits[b])
+                    self._emit(b, row, tok)
 
 
 class GenerationService:
