@@ -651,6 +651,51 @@ PYBIND11_MODULE(_C, m) {
           py::arg("qkv"), py::arg("kcache"), py::arg("vcache"), py::arg("B"),
           py::arg("H"), py::arg("P"), py::arg("smax"), py::arg("stream") = 0,
           py::arg("sync") = true, py::arg("D") = 64);
+  ops.def("kv_append_chunk_paged",
+          [](uintptr_t qkv, uintptr_t kpool, uintptr_t vpool,
+             uintptr_t table, uintptr_t pos, int B, int H, int K,
+             int max_pages, uintptr_t stream, bool sync, int D) {
+            launch_kv_append_chunk_paged((void*)qkv, (void*)kpool,
+                                         (void*)vpool, (void*)table,
+                                         (void*)pos, B, H, K, max_pages,
+                                         as_stream(stream), D);
+            if (sync) TRT_HIP_CHECK(hipStreamSynchronize(as_stream(stream)));
+          },
+          py::arg("qkv"), py::arg("kpool"), py::arg("vpool"),
+          py::arg("table"), py::arg("pos"), py::arg("B"), py::arg("H"),
+          py::arg("K"), py::arg("max_pages"), py::arg("stream") = 0,
+          py::arg("sync") = true, py::arg("D") = 64);
+  ops.def("kv_append_range_paged",
+          [](uintptr_t qkv, uintptr_t kpool, uintptr_t vpool,
+             uintptr_t table, int B, int H, int P, int max_pages,
+             uintptr_t stream, bool sync, int D) {
+            launch_kv_append_range_paged((void*)qkv, (void*)kpool,
+                                         (void*)vpool, (void*)table, B, H,
+                                         P, max_pages, as_stream(stream),
+                                         D);
+            if (sync) TRT_HIP_CHECK(hipStreamSynchronize(as_stream(stream)));
+          },
+          py::arg("qkv"), py::arg("kpool"), py::arg("vpool"),
+          py::arg("table"), py::arg("B"), py::arg("H"), py::arg("P"),
+          py::arg("max_pages"), py::arg("stream") = 0,
+          py::arg("sync") = true, py::arg("D") = 64);
+  ops.def("chunk_attention_paged",
+          [](uintptr_t qkv, uintptr_t kpool, uintptr_t vpool, uintptr_t out,
+             uintptr_t table, uintptr_t pos, int B, int H, int K,
+             int max_pages, float scale, uintptr_t stream, bool sync,
+             int D) {
+            launch_chunk_attention_paged((void*)qkv, (void*)kpool,
+                                         (void*)vpool, (void*)out,
+                                         (void*)table, (void*)pos, B, H, K,
+                                         max_pages, scale,
+                                         as_stream(stream), D);
+            if (sync) TRT_HIP_CHECK(hipStreamSynchronize(as_stream(stream)));
+          },
+          py::arg("qkv"), py::arg("kpool"), py::arg("vpool"),
+          py::arg("out"), py::arg("table"), py::arg("pos"), py::arg("B"),
+          py::arg("H"), py::arg("K"), py::arg("max_pages"),
+          py::arg("scale"), py::arg("stream") = 0, py::arg("sync") = true,
+          py::arg("D") = 64);
   ops.def("kv_append_paged",
           [](uintptr_t qkv, uintptr_t kp, uintptr_t vp, uintptr_t table,
              uintptr_t pos, int B, int H, int max_pages, uintptr_t stream,

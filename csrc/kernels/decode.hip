@@ -645,6 +645,162 @@ __global__ __launch_bounds__(64) void kv_append_paged_kernel(
   }
 }
 
+// ---- paged variants of the chunk / prefill cache writers + reader ----
+// (speculative verification and fused prefill for paged sessions: same
+// math as the dense chunk kernels, addresses resolved through the
+// per-slot page table; pages host-mapped before launch.)
+__global__ __launch_bounds__(64) void kv_append_chunk_paged_kernel(
+    const _Float16* __restrict__ qkv, _Float16* __restrict__ kpool,
+    _Float16* __restrict__ vpool, const int* __restrict__ table,
+    const int* __restrict__ pos, int B, int H, int K, int max_pages,
+    int D) {
+  int q = blockIdx.x % K;
+  int bh = blockIdx.x / K;
+  int b = bh / H, h = bh % H;
+  int p0 = pos[b];
+  if (p0 < 0) return;  // idle slot
+  int p = p0 + q;
+  int page = table[b * max_pages + (p >> 6)];
+  if (page < 0) return;  // unmapped (host error) — fail soft
+  int hid = H * D;
+  for (int d = threadIdx.x; d < D; d += 64) {
+    int64_t src = ((int64_t)b * K + q) * 3 * hid + h * D + d;
+    int64_t dst = (((int64_t)page * 64 + (p & 63)) * H + h) * D + d;
+    kpool[dst] = qkv[src + hid];
+    vpool[dst] = qkv[src + 2 * hid];
+  }
+}
+
+void launch_kv_append_chunk_paged(const void* qkv, void* kpool, void* vpool,
+                                  const void* table, const void* pos, int B,
+                                  int H, int K, int max_pages,
+                                  hipStream_t stream, int D) {
+  hipLaunchKernelGGL(kv_append_chunk_paged_kernel, dim3(B * H * K),
+                     dim3(64), 0, stream, (const _Float16*)qkv,
+                     (_Float16*)kpool, (_Float16*)vpool, (const int*)table,
+                     (const int*)pos, B, H, K, max_pages, D);
+}
+
+__global__ __launch_bounds__(64) void kv_append_range_paged_kernel(
+    const _Float16* __restrict__ qkv, _Float16* __restrict__ kpool,
+    _Float16* __restrict__ vpool, const int* __restrict__ table, int B,
+    int H, int P, int max_pages, int D) {
+  int p = blockIdx.x % P;
+  int bh = blockIdx.x / P;
+  int b = bh / H, h = bh % H;
+  int page = table[b * max_pages + (p >> 6)];
+  if (page < 0) return;
+  int hid = H * D;
+  for (int d = threadIdx.x; d < D; d += 64) {
+    int64_t src = ((int64_t)b * P + p) * 3 * hid + h * D + d;
+    int64_t dst = (((int64_t)page * 64 + (p & 63)) * H + h) * D + d;
+    kpool[dst] = qkv[src + hid];
+    vpool[dst] = qkv[src + 2 * hid];
+  }
+}
+
+void launch_kv_append_range_paged(const void* qkv, void* kpool, void* vpool,
+                                  const void* table, int B, int H, int P,
+                                  int max_pages, hipStream_t stream, int D) {
+  hipLaunchKernelGGL(kv_append_range_paged_kernel, dim3(B * H * P),
+                     dim3(64), 0, stream, (const _Float16*)qkv,
+                     (_Float16*)kpool, (_Float16*)vpool, (const int*)table,
+                     B, H, P, max_pages, D);
+}
+
+template <int D>  // head_dim 64 or 128
+__global__ __launch_bounds__(64) void chunk_attention_paged_kernel(
+    const _Float16* __restrict__ qkv, const _Float16* __restrict__ kpool,
+    const _Float16* __restrict__ vpool, _Float16* __restrict__ out,
+    const int* __restrict__ table, const int* __restrict__ pos, int B,
+    int H, int K, int max_pages, float scale) {
+  __shared__ float p_s[4096];
+  int q = blockIdx.x % K;
+  int bh = blockIdx.x / K;
+  int b = bh / H, h = bh % H;
+  int lane = threadIdx.x;
+  int hid = H * D;
+  int p0 = pos[b];
+  if (p0 < 0) return;  // idle slot
+  int n = p0 + q + 1;
+
+  _Float16 qv[D];
+  {
+    const _Float16* qrow = qkv + ((int64_t)b * K + q) * 3 * hid + h * D;
+#pragma unroll
+    for (int c = 0; c < D / 8; ++c)
+      *(half8v*)(qv + c * 8) = *(const half8v*)(qrow + c * 8);
+  }
+  const int* trow = table + b * max_pages;
+
+  float m = -3.0e38f;
+  for (int t = lane; t < n; t += 64) {
+    int page = trow[t >> 6];
+    const _Float16* krow =
+        kpool + (((int64_t)page * 64 + (t & 63)) * H + h) * D;
+    float sc = 0.f;
+#pragma unroll
+    for (int c = 0; c < D / 8; ++c) {
+      half8v v = *(const half8v*)(krow + c * 8);
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        sc += (float)qv[c * 8 + j] * (float)((const _Float16*)&v)[j];
+    }
+    sc *= scale;
+    p_s[t] = sc;
+    m = fmaxf(m, sc);
+  }
+#pragma unroll
+  for (int off = 32; off; off >>= 1) m = fmaxf(m, __shfl_xor(m, off, 64));
+  __syncthreads();
+  float l = 0.f;
+  for (int t = lane; t < n; t += 64) {
+    float e = __expf(p_s[t] - m);
+    p_s[t] = e;
+    l += e;
+  }
+#pragma unroll
+  for (int off = 32; off; off >>= 1) l += __shfl_xor(l, off, 64);
+  __syncthreads();
+  float acc[D / 64];
+#pragma unroll
+  for (int j = 0; j < D / 64; ++j) acc[j] = 0.f;
+  for (int t = 0; t < n; ++t) {
+    int page = trow[t >> 6];
+    const _Float16* vrow =
+        vpool + (((int64_t)page * 64 + (t & 63)) * H + h) * D;
+    float pp = p_s[t];
+#pragma unroll
+    for (int j = 0; j < D / 64; ++j)
+      acc[j] += pp * (float)vrow[j * 64 + lane];
+  }
+#pragma unroll
+  for (int j = 0; j < D / 64; ++j)
+    out[((int64_t)b * K + q) * hid + h * D + j * 64 + lane] =
+        (_Float16)(acc[j] / l);
+}
+
+void launch_chunk_attention_paged(const void* qkv, const void* kpool,
+                                  const void* vpool, void* out,
+                                  const void* table, const void* pos, int B,
+                                  int H, int K, int max_pages, float scale,
+                                  hipStream_t stream, int D) {
+  if (D == 128)
+    hipLaunchKernelGGL(chunk_attention_paged_kernel<128>, dim3(B * H * K),
+                       dim3(64), 0, stream, (const _Float16*)qkv,
+                       (const _Float16*)kpool, (const _Float16*)vpool,
+                       (_Float16*)out, (const int*)table, (const int*)pos,
+                       B, H, K, max_pages, scale);
+  else if (D == 64)
+    hipLaunchKernelGGL(chunk_attention_paged_kernel<64>, dim3(B * H * K),
+                       dim3(64), 0, stream, (const _Float16*)qkv,
+                       (const _Float16*)kpool, (const _Float16*)vpool,
+                       (_Float16*)out, (const int*)table, (const int*)pos,
+                       B, H, K, max_pages, scale);
+  else
+    throw std::runtime_error("chunk_attention_paged: head_dim 64/128");
+}
+
 void launch_kv_append_paged(const void* qkv, void* kpool, void* vpool,
                             const void* table, const void* pos, int B, int H,
                             int max_pages, hipStream_t stream, int D) {

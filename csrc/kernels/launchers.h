@@ -152,6 +152,19 @@ void launch_chunk_embed(const void* ids, const void* tok, const void* posemb,
                         int hidden, hipStream_t stream);
 void launch_advance_pos(void* pos, int B, int smax, hipStream_t stream);
 // paged KV cache (vLLM-style block tables; see decode.hip)
+void launch_kv_append_chunk_paged(const void* qkv, void* kpool, void* vpool,
+                                  const void* table, const void* pos, int B,
+                                  int H, int K, int max_pages,
+                                  hipStream_t stream, int D = 64);
+void launch_kv_append_range_paged(const void* qkv, void* kpool, void* vpool,
+                                  const void* table, int B, int H, int P,
+                                  int max_pages, hipStream_t stream,
+                                  int D = 64);
+void launch_chunk_attention_paged(const void* qkv, const void* kpool,
+                                  const void* vpool, void* out,
+                                  const void* table, const void* pos, int B,
+                                  int H, int K, int max_pages, float scale,
+                                  hipStream_t stream, int D = 64);
 void launch_kv_append_paged(const void* qkv, void* kpool, void* vpool,
                             const void* table, const void* pos, int B, int H,
                             int max_pages, hipStream_t stream, int D = 64);

@@ -506,6 +506,23 @@ class DecodeSession:
         ops.advance_pos(self.pos.data_ptr(), B, self.smax, stream=s,
                         sync=False)
 
+    def _kv_range(self, li: int, lay: Dict, qkv_ptr: int, B: int,
+                  Pp: int) -> None:
+        """Prefill cache writer (dense or paged)."""
+        ops, s = self._C.ops, self.stream
+        if self.kv_pool is not None:
+            pool = self.kv_pool
+            ops.kv_append_range_paged(qkv_ptr, pool.kpools[li].data_ptr(),
+                                      pool.vpools[li].data_ptr(),
+                                      pool.table.data_ptr(), B, self.heads,
+                                      Pp, pool.max_pages, stream=s,
+                                      sync=False, D=self.hd)
+        else:
+            ops.kv_append_range(qkv_ptr, lay["kcache"].data_ptr(),
+                                lay["vcache"].data_ptr(), B, self.heads,
+                                Pp, self.smax, stream=s, sync=False,
+                                D=self.hd)
+
     def prefill(self, prompt: np.ndarray) -> np.ndarray:
         """Fill the KV caches from a whole prompt [B, P] in ONE pass through
         the full-sequence kernels (causal online-softmax attention), then
@@ -520,9 +537,6 @@ class DecodeSession:
 
         if self._steps != 0 or self._graph:
             raise RuntimeError("prefill must run before the first step()")
-        if self.kv_pool is not None:
-            raise RuntimeError("prefill is not supported in paged mode yet "
-                               "(step the prompt, or use dense caches)")
         prompt = np.ascontiguousarray(prompt, np.int32)
         B, P = prompt.shape
         assert B == self.batch and 0 < P < self.smax
@@ -534,6 +548,13 @@ class DecodeSession:
                 "build the model with a longer seq")
         ids = np.zeros((B, Pp), np.int32)
         ids[:, :P] = prompt
+        if self.kv_pool is not None:
+            # map pages covering the PADDED prompt (the tail rows are
+            # written by kv_append_range_paged and overwritten by later
+            # decode steps before they are ever attended)
+            for b in range(B):
+                for li in range((Pp - 1) // 64 + 1):
+                    self.kv_pool.ensure(b, li)
         M = B * Pp
         Hd, inter = self.hidden, self.inter
         ops, s = self._C.ops, self.stream
@@ -564,11 +585,7 @@ class DecodeSession:
                 # padded tail is causal-masked / overwritten before read)
                 ops.rope(0, qkv.data_ptr(), M=M, S=Pp, H=self.heads,
                          D=self.hd, theta=self.theta, stream=s, sync=False)
-                ops.kv_append_range(qkv.data_ptr(),
-                                    lay["kcache"].data_ptr(),
-                                    lay["vcache"].data_ptr(), B, self.heads,
-                                    Pp, self.smax, stream=s, sync=False,
-                                    D=self.hd)
+                self._kv_range(li, lay, qkv.data_ptr(), B, Pp)
                 ops.attention(0, qkv.data_ptr(), x2.data_ptr(), B, Pp,
                               self.heads, self.hd, scale, stream=s,
                               sync=False, causal=1)
@@ -606,11 +623,7 @@ class DecodeSession:
                             qkv.data_ptr(), bias=lay["qkv_b"].data_ptr(),
                             M=M, N=3 * Hd, K=Hd, epi=self._epi_bias,
                             stream=s, sync=False)
-                ops.kv_append_range(qkv.data_ptr(),
-                                    lay["kcache"].data_ptr(),
-                                    lay["vcache"].data_ptr(), B, self.heads,
-                                    Pp, self.smax, stream=s, sync=False,
-                                    D=self.hd)
+                self._kv_range(li, lay, qkv.data_ptr(), B, Pp)
                 ops.attention(0, qkv.data_ptr(), x2.data_ptr(), B, Pp,
                               self.heads, self.hd, scale,
                               stream=s, sync=False, causal=1)
@@ -708,6 +721,42 @@ class DecodeSession:
                 if self.logits is not None else out)
 
     # ------------------------------------------- speculative verification
+    def _kv_attn_chunk(self, li: int, lay: Dict, cb: Dict, B: int,
+                       K: int) -> None:
+        """Per-layer chunk KV append + multi-query attention (dense or
+        paged — the paged path resolves rows through the page table)."""
+        ops, s = self._C.ops, self.stream
+        scale = 1.0 / float(np.sqrt(float(self.hd)))
+        if self.kv_pool is not None:
+            pool = self.kv_pool
+            ops.kv_append_chunk_paged(cb["qkv"].data_ptr(),
+                                      pool.kpools[li].data_ptr(),
+                                      pool.vpools[li].data_ptr(),
+                                      pool.table.data_ptr(),
+                                      self.pos.data_ptr(), B, self.heads,
+                                      K, pool.max_pages, stream=s,
+                                      sync=False, D=self.hd)
+            ops.chunk_attention_paged(cb["qkv"].data_ptr(),
+                                      pool.kpools[li].data_ptr(),
+                                      pool.vpools[li].data_ptr(),
+                                      cb["att"].data_ptr(),
+                                      pool.table.data_ptr(),
+                                      self.pos.data_ptr(), B, self.heads,
+                                      K, pool.max_pages, scale, stream=s,
+                                      sync=False, D=self.hd)
+        else:
+            ops.kv_append_chunk(cb["qkv"].data_ptr(),
+                                lay["kcache"].data_ptr(),
+                                lay["vcache"].data_ptr(),
+                                self.pos.data_ptr(), B, self.heads, K,
+                                self.smax, stream=s, sync=False, D=self.hd)
+            ops.chunk_attention(cb["qkv"].data_ptr(),
+                                lay["kcache"].data_ptr(),
+                                lay["vcache"].data_ptr(),
+                                cb["att"].data_ptr(), self.pos.data_ptr(),
+                                B, self.heads, K, self.smax, scale,
+                                stream=s, sync=False, D=self.hd)
+
     def _enqueue_chunk(self, cb: Dict, B: int, K: int) -> None:
         """Record one chunked-verification pass's kernels on the session
         stream (no syncs — capturable). All position dependence reads the
@@ -732,19 +781,7 @@ class DecodeSession:
                 ops.rope(0, cb["qkv"].data_ptr(), pos=self.pos.data_ptr(),
                          M=M, S=self.smax, H=self.heads, D=self.hd,
                          theta=self.theta, stream=s, sync=False, chunk=K)
-                ops.kv_append_chunk(cb["qkv"].data_ptr(),
-                                    lay["kcache"].data_ptr(),
-                                    lay["vcache"].data_ptr(),
-                                    self.pos.data_ptr(), B, self.heads, K,
-                                    self.smax, stream=s, sync=False,
-                                    D=self.hd)
-                ops.chunk_attention(cb["qkv"].data_ptr(),
-                                    lay["kcache"].data_ptr(),
-                                    lay["vcache"].data_ptr(),
-                                    cb["att"].data_ptr(),
-                                    self.pos.data_ptr(), B, self.heads, K,
-                                    self.smax, scale, stream=s, sync=False,
-                                    D=self.hd)
+                self._kv_attn_chunk(li, lay, cb, B, K)
                 ops.gemm_bt(0, cb["att"].data_ptr(),
                             lay["proj_w"].data_ptr(), cb["x2"].data_ptr(),
                             M=M, N=Hd, K=Hd, epi=self._epi_none, stream=s,
@@ -785,19 +822,7 @@ class DecodeSession:
                             bias=lay["qkv_b"].data_ptr(),
                             M=M, N=3 * Hd, K=Hd, epi=self._epi_bias,
                             stream=s, sync=False)
-                ops.kv_append_chunk(cb["qkv"].data_ptr(),
-                                    lay["kcache"].data_ptr(),
-                                    lay["vcache"].data_ptr(),
-                                    self.pos.data_ptr(), B, self.heads, K,
-                                    self.smax, stream=s, sync=False,
-                                    D=self.hd)
-                ops.chunk_attention(cb["qkv"].data_ptr(),
-                                    lay["kcache"].data_ptr(),
-                                    lay["vcache"].data_ptr(),
-                                    cb["att"].data_ptr(),
-                                    self.pos.data_ptr(),
-                                    B, self.heads, K, self.smax, scale,
-                                    stream=s, sync=False, D=self.hd)
+                self._kv_attn_chunk(li, lay, cb, B, K)
                 ops.gemm_bt(0, cb["att"].data_ptr(),
                             lay["proj_w"].data_ptr(), cb["x2"].data_ptr(),
                             bias=lay["proj_b"].data_ptr(),
@@ -852,11 +877,15 @@ class DecodeSession:
 
         if self.logits is None:
             raise RuntimeError("verify_chunk requires lm_head=True")
-        if self.kv_pool is not None:
-            raise RuntimeError("verify_chunk is not supported in paged "
-                               "mode yet")
         tokens = np.ascontiguousarray(tokens, np.int32)
         B, K = tokens.shape
+        if self.kv_pool is not None:
+            # map every page the chunk's writes land in (pos..pos+K-1)
+            for b in range(self.batch):
+                if self._active[b]:
+                    p0 = int(self._slot_steps[b])
+                    for li in range(p0 >> 6, ((p0 + K - 1) >> 6) + 1):
+                        self.kv_pool.ensure(b, li)
         assert B == self.batch and K >= 1
         M = B * K
         Hd, inter = self.hidden, self.inter
