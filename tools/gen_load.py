@@ -25,6 +25,7 @@ def main():
     ap.add_argument("--hidden", type=int, default=1024)
     ap.add_argument("--prompt-len", type=int, default=16)
     ap.add_argument("--new-tokens", type=int, default=64)
+    ap.add_argument("--inline-step", action="store_true")
     args = ap.parse_args()
 
     from trtlab_amd.engine.decode import DecodeSession
@@ -36,7 +37,7 @@ def main():
     g = build_llama(batch=args.batch, seq=1024, hidden=args.hidden,
                     layers=args.layers, heads=args.hidden // 128, seed=0)
     sess = DecodeSession(g, batch=args.batch, smax=1024, lm_head=True)
-    svc = GenerationService(sess)
+    svc = GenerationService(sess, inline_step=args.inline_step)
     srv = Server("127.0.0.1:0")
     srv.register_service(svc.service)
     srv.async_start()

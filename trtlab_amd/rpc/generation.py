@@ -66,12 +66,21 @@ GenerateToken = message_factory.GetMessageClass(
 class GenerationEngine:
     """Continuous-batching loop over a lockstep decode session."""
 
-    def __init__(self, session, eos: int = -1):
+    def __init__(self, session, eos: int = -1, inline_step: bool = False):
+        """inline_step=True runs session.step() directly on the event
+        loop instead of a worker thread. MEASURED COUNTERPRODUCTIVE
+        (tools/gen_load.py, MI355X, 16 streams/8 slots): 5,491 -> 1,229
+        tok/s. The worker-thread hop YIELDS the event loop for the whole
+        GPU step — exactly when new streams get admitted — so inlining
+        starves admission and the loop burns steps on half-empty batches
+        (1,185 steps vs the 159 ideal). Kept as an off-default knob for
+        single-stream latency experiments."""
         import collections
 
         self.session = session
         self.B = session.batch
         self.eos = eos
+        self.inline_step = inline_step
         self._slots: List[Optional[dict]] = [None] * self.B
         self._free: List[int] = list(range(self.B))
         self._pending: collections.deque = collections.deque()
@@ -212,12 +221,19 @@ class GenerationEngine:
             ids_only = (getattr(self.session, "supports_ids", False) and
                         all(snap[b]["temperature"] <= 0.0 for b in active))
             if ids_only:
-                gids = await loop.run_in_executor(
-                    None, lambda: self.session.step(ids, return_ids=True))
+                if self.inline_step:
+                    gids = self.session.step(ids, return_ids=True)
+                else:
+                    gids = await loop.run_in_executor(
+                        None,
+                        lambda: self.session.step(ids, return_ids=True))
                 logits = None
             else:
-                logits = await loop.run_in_executor(None,
-                                                    self.session.step, ids)
+                if self.inline_step:
+                    logits = self.session.step(ids)
+                else:
+                    logits = await loop.run_in_executor(
+                        None, self.session.step, ids)
             self.steps += 1
             for b in active:
                 st = self._slots[b]
@@ -241,10 +257,12 @@ class GenerationService:
     stream down, over the shared continuous-batching engine."""
 
     def __init__(self, session, eos: int = -1,
-                 name: str = "trtlab.gen.Generation"):
+                 name: str = "trtlab.gen.Generation",
+                 inline_step: bool = False):
         from trtlab_amd.rpc.server import StreamingService
 
-        self.engine = GenerationEngine(session, eos=eos)
+        self.engine = GenerationEngine(session, eos=eos,
+                                       inline_step=inline_step)
         svc = StreamingService(name)
         svc.register_streaming("Generate", self._generate, GenerateRequest,
                                GenerateToken)
