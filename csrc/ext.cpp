@@ -513,6 +513,17 @@ PYBIND11_MODULE(_C, m) {
           },
           py::arg("x"), py::arg("out"), py::arg("M"), py::arg("V"),
           py::arg("stream") = 0, py::arg("sync") = true);
+  ops.def("gumbel_argmax_rows",
+          [](uintptr_t x, uintptr_t out, uintptr_t temps, uintptr_t seeds,
+             uintptr_t pos, int M, int V, uintptr_t stream, bool sync) {
+            launch_gumbel_argmax_rows((void*)x, (void*)out, (void*)temps,
+                                      (void*)seeds, (void*)pos, M, V,
+                                      as_stream(stream));
+            if (sync) TRT_HIP_CHECK(hipStreamSynchronize(as_stream(stream)));
+          },
+          py::arg("x"), py::arg("out"), py::arg("temps") = 0,
+          py::arg("seeds") = 0, py::arg("pos") = 0, py::arg("M") = 0,
+          py::arg("V") = 0, py::arg("stream") = 0, py::arg("sync") = true);
   ops.def("clip",
           [](int dtype, uintptr_t in, uintptr_t out, int64_t n, float mn,
              float mx, uintptr_t stream, bool sync) {

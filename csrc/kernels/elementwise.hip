@@ -510,5 +510,85 @@ void launch_argmax_rows(const void* x, void* out, int M, int V,
                      stream, (const _Float16*)x, (int*)out, M, V);
 }
 
+// ---- device-side categorical sampling via the Gumbel-max trick ----
+// argmax(logits/T + G_c) with G_c = -log(-log(u_c)) samples EXACTLY from
+// softmax(logits/T) — so temperature sampling reuses the argmax shape
+// and only row winners (4 B each) leave the GPU. u_c comes from a
+// counter-based splitmix64 keyed on (seed[row], pos[row], c): pos is the
+// DEVICE position counter, so every decode step draws fresh noise even
+// inside a captured graph. temps[row] <= 0 degrades to plain greedy
+// argmax (one kernel serves mixed greedy/sampled batches).
+__device__ __forceinline__ uint64_t splitmix64(uint64_t z) {
+  z += 0x9e3779b97f4a7c15ull;
+  z = (z ^ (z >> 30)) * 0xbf58476d1ce4e5b9ull;
+  z = (z ^ (z >> 27)) * 0x94d049bb133111ebull;
+  return z ^ (z >> 31);
+}
+
+__global__ __launch_bounds__(256) void gumbel_argmax_rows_kernel(
+    const _Float16* __restrict__ x, int* __restrict__ out,
+    const float* __restrict__ temps, const int* __restrict__ seeds,
+    const int* __restrict__ pos, int M, int V) {
+  int row = blockIdx.x;
+  if (row >= M) return;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63, wave = tid >> 6;
+  const _Float16* r = x + (int64_t)row * V;
+  const float T = temps ? temps[row] : 0.0f;
+  const float invT = (T > 0.0f) ? 1.0f / T : 0.0f;
+  const uint64_t key =
+      ((uint64_t)(uint32_t)(seeds ? seeds[row] : 0) << 32) ^
+      (uint64_t)(uint32_t)(pos ? pos[row] : 0) ^
+      ((uint64_t)(uint32_t)row << 20);
+  float best = -1e30f;
+  int bi = 0x7fffffff;
+  for (int c = tid; c < V; c += 256) {
+    float v = (float)r[c];
+    if (T > 0.0f) {
+      uint64_t h = splitmix64(key ^ (uint64_t)c);
+      // u in (0, 1): top 24 bits, never exactly 0
+      float u = ((float)(uint32_t)(h >> 40) + 0.5f) * (1.0f / 16777216.f);
+      v = v * invT - __logf(-__logf(u));
+    }
+    if (v > best || (v == best && c < bi)) {
+      best = v;
+      bi = c;
+    }
+  }
+#pragma unroll
+  for (int off = 32; off; off >>= 1) {
+    float ob = __shfl_xor(best, off, 64);
+    int oi = __shfl_xor(bi, off, 64);
+    if (ob > best || (ob == best && oi < bi)) {
+      best = ob;
+      bi = oi;
+    }
+  }
+  __shared__ float wb[4];
+  __shared__ int wi[4];
+  if (lane == 0) {
+    wb[wave] = best;
+    wi[wave] = bi;
+  }
+  __syncthreads();
+  if (tid == 0) {
+    for (int w = 1; w < 4; ++w)
+      if (wb[w] > wb[0] || (wb[w] == wb[0] && wi[w] < wi[0])) {
+        wb[0] = wb[w];
+        wi[0] = wi[w];
+      }
+    out[row] = wi[0];
+  }
+}
+
+void launch_gumbel_argmax_rows(const void* x, void* out, const void* temps,
+                               const void* seeds, const void* pos, int M,
+                               int V, hipStream_t stream) {
+  hipLaunchKernelGGL(gumbel_argmax_rows_kernel, dim3((unsigned)M),
+                     dim3(256), 0, stream, (const _Float16*)x, (int*)out,
+                     (const float*)temps, (const int*)seeds,
+                     (const int*)pos, M, V);
+}
+
 }  // namespace trtlab
 

@@ -79,6 +79,11 @@ void launch_elementwise(int dtype, int op, const void* a, const void* b,
 // Row-wise argmax (greedy decode head): fp16 [M, V] -> int32 [M].
 void launch_argmax_rows(const void* x, void* out, int M, int V,
                         hipStream_t stream);
+// Gumbel-max categorical sampling from softmax(logits/temps[row]);
+// temps[row] <= 0 -> plain argmax. Noise keyed (seeds[row], pos[row], c).
+void launch_gumbel_argmax_rows(const void* x, void* out, const void* temps,
+                               const void* seeds, const void* pos, int M,
+                               int V, hipStream_t stream);
 void launch_clip(int dtype, const void* in, void* out, int64_t n, float mn,
                  float mx, hipStream_t stream);
 void launch_transpose2d(int dtype, const void* in, void* out, int M, int N,

@@ -934,6 +934,35 @@ class DecodeSession:
             return cb["gids"].cpu().numpy().reshape(B, K)
         return cb["logits"].float().cpu().numpy().reshape(B, K, self.vocab)
 
+    def sample_tokens(self, temps: np.ndarray,
+                      seeds: np.ndarray) -> np.ndarray:
+        """Device-side temperature sampling of the LAST step's logits
+        (Gumbel-max: an exact softmax(logits/T) categorical draw per
+        slot; temps[b] <= 0 degrades to greedy argmax). Noise is keyed
+        (seeds[b], device pos[b], index), so consecutive steps draw
+        fresh noise and a (seed, position) pair replays identically.
+        Only B ints cross PCIe."""
+        if self.logits is None:
+            raise RuntimeError("sample_tokens requires lm_head=True")
+        torch = self._torch
+        if getattr(self, "_temps_dev", None) is None:
+            self._temps_dev = torch.zeros(self.batch, dtype=torch.float32,
+                                          device="cuda")
+            self._seeds_dev = torch.zeros(self.batch, dtype=torch.int32,
+                                          device="cuda")
+        self._C.memory.memcpy_h2d(self._temps_dev.data_ptr(),
+                                  np.ascontiguousarray(temps, np.float32),
+                                  self.batch * 4)
+        self._C.memory.memcpy_h2d(self._seeds_dev.data_ptr(),
+                                  np.ascontiguousarray(seeds, np.int32),
+                                  self.batch * 4)
+        self._C.ops.gumbel_argmax_rows(
+            self.logits.data_ptr(), self.gids.data_ptr(),
+            temps=self._temps_dev.data_ptr(),
+            seeds=self._seeds_dev.data_ptr(), pos=self.pos.data_ptr(),
+            M=self.batch, V=self.vocab, stream=self.stream, sync=True)
+        return self.gids.cpu().numpy()
+
     def get_pos(self) -> np.ndarray:
         """Per-slot positions from the HOST mirror (no D2H round-trip:
         device pos only changes via step()'s advance kernel — mirrored

@@ -132,6 +132,7 @@ class GenerationEngine:
                                    pi=0, remaining=int(max_tokens), cur=0,
                                    temperature=float(temperature),
                                    top_k=int(top_k), top_p=float(top_p),
+                                   seed=int(seed),
                                    rng=np.random.RandomState(seed or None)),
                               fut))
         self._wake.set()
@@ -215,18 +216,31 @@ class GenerationEngine:
                 st = snap[b] = self._slots[b]
                 ids[b] = (st["prompt"][st["pi"]]
                           if st["pi"] < len(st["prompt"]) else st["cur"])
-            # greedy fast path: when every live stream decodes greedily
-            # and the session has the in-graph argmax head, only B ints
-            # cross PCIe per step (B*vocab logits otherwise)
+            # device fast path: greedy uses the in-graph argmax head;
+            # plain-temperature sampling uses the Gumbel-max kernel on
+            # the resident logits — either way only B ints cross PCIe.
+            # top-k / nucleus streams still need their logits rows.
             ids_only = (getattr(self.session, "supports_ids", False) and
-                        all(snap[b]["temperature"] <= 0.0 for b in active))
+                        all(snap[b]["top_k"] == 0 and
+                            snap[b]["top_p"] == 0.0 for b in active))
             if ids_only:
+                sampled = any(snap[b]["temperature"] > 0.0 for b in active)
+                temps = np.zeros(self.B, np.float32)
+                seeds = np.zeros(self.B, np.int32)
+                for b in active:
+                    temps[b] = max(snap[b]["temperature"], 0.0)
+                    seeds[b] = snap[b].get("seed", 0)
+
+                def _dev_step():
+                    g = self.session.step(ids, return_ids=True)
+                    if sampled:
+                        g = self.session.sample_tokens(temps, seeds)
+                    return g
+
                 if self.inline_step:
-                    gids = self.session.step(ids, return_ids=True)
+                    gids = _dev_step()
                 else:
-                    gids = await loop.run_in_executor(
-                        None,
-                        lambda: self.session.step(ids, return_ids=True))
+                    gids = await loop.run_in_executor(None, _dev_step)
                 logits = None
             else:
                 if self.inline_step:
