@@ -438,12 +438,6 @@ class DecodeSession:
                         sync=False)
 
     def _enqueue(self):
-        if getattr(self, "arch", "gpt2") == "llama":
-            self._enqueue_llama()
-            return
-        if getattr(self, "fused", False):
-            self._enqueue_fused()
-            return
         """Record one decode step's kernels on self.stream (pos-relative:
         kv_append/decode_attention/embed all read the device counter).
 
@@ -454,6 +448,12 @@ class DecodeSession:
         disabled on the tiny M=B gemms via the tile hint (the fp32-slab
         reduce kernel doubled the gemm count for no win at this floor).
         """
+        if getattr(self, "arch", "gpt2") == "llama":
+            self._enqueue_llama()
+            return
+        if getattr(self, "fused", False):
+            self._enqueue_fused()
+            return
         C, s = self._C, self.stream
         B, Hd = self.batch, self.hidden
         T4 = 0  # heuristic tiles (split-K allowed)
