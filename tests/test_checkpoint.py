@@ -224,3 +224,30 @@ def test_safetensors_gpt2_matches_hf_semantics(tmp_path):
     ref = ln(hid, "ln_f.weight", "ln_f.bias").numpy()
     err = np.abs(out - ref).max() / max(np.abs(ref).max(), 1e-6)
     assert err < 2e-3, err
+
+
+def test_safetensors_sharded_checkpoint(tmp_path):
+    """Sharded checkpoints (multiple *.safetensors in a dir) merge into
+    one state dict and load identically to the single-file form."""
+    from safetensors.numpy import load_file, save_file
+
+    st = _mk_checkpoint(str(tmp_path))
+    # re-save as two shards + remove the single file
+    keys = sorted(st)
+    half = len(keys) // 2
+    save_file({k: st[k] for k in keys[:half]},
+              os.path.join(str(tmp_path), "model-00001-of-00002"
+                           ".safetensors"))
+    save_file({k: st[k] for k in keys[half:]},
+              os.path.join(str(tmp_path), "model-00002-of-00002"
+                           ".safetensors"))
+    single = os.path.join(str(tmp_path), "model.safetensors")
+    ref_state = load_file(single)
+    os.remove(single)
+
+    g = build_llama_from_safetensors(str(tmp_path), batch=1, seq=16)
+    plan = Planner().compile(g)
+    ids = np.random.RandomState(2).randint(0, VOCAB, 16).astype(np.int32)
+    out = run_reference(plan, ids)
+    ref = _hf_forward(ref_state, torch.from_numpy(ids).long(), 16)
+    assert np.abs(out - ref).max() / max(np.abs(ref).max(), 1e-6) < 2e-3

@@ -1399,3 +1399,34 @@ def test_paged_chunk_and_prefill_match_dense():
         np.testing.assert_array_equal(ga, gb)
         dense.close()
         paged.close()
+
+
+def test_speculative_decoding_paged_target():
+    """Spec decode with a PAGED-cache target (paged prefill-free chunked
+    verification): output identical to the dense target-only baseline."""
+    from trtlab_amd.engine.decode import DecodeSession, SpeculativeDecoder
+    from trtlab_amd.models import build_gpt2
+
+    g = build_gpt2(batch=2, seq=96, layers=2, seed=0, embeddings=True)
+    gd = build_gpt2(batch=2, seq=96, layers=1, seed=4, embeddings=True)
+    seed_tok = np.array([31, 8], np.int32)
+    STEPS = 8
+
+    base = DecodeSession(g, batch=2, smax=96, capture=False, lm_head=True)
+    cur = seed_tok
+    ref = []
+    for _ in range(STEPS):
+        lg = base.step(cur)
+        cur = lg.argmax(-1).astype(np.int32)
+        ref.append(cur)
+    base.close()
+    ref = np.stack(ref, axis=1)
+
+    t = DecodeSession(g, batch=2, smax=96, capture=False, lm_head=True,
+                      paged=True)
+    d = DecodeSession(gd, batch=2, smax=96, capture=False, lm_head=True)
+    sd = SpeculativeDecoder(t, d, k=3)
+    toks, _ = sd.generate(seed_tok, STEPS)
+    np.testing.assert_array_equal(toks, ref)
+    t.close()
+    d.close()
