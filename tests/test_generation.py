@@ -313,3 +313,18 @@ def test_generation_client_disconnect_frees_slot():
         assert done and toks == expected_chain([9], 5)
     finally:
         srv.shutdown()
+
+
+def test_engine_stats_counters():
+    """Engine observability: steps / tokens_out / streams / slot gauges
+    reflect the run (tokens/steps ~ packing efficiency)."""
+    sess, gen, srv = _serve(batch=2)
+    try:
+        _collect(srv.port, [5], 6)
+        _collect(srv.port, [9, 2], 4)
+        st = gen.engine.stats()
+        assert st["tokens_out"] == 10 and st["streams"] == 2
+        assert st["active"] == 0 and st["free"] == 2
+        assert st["pending"] == 0 and st["steps"] >= 7
+    finally:
+        srv.shutdown()

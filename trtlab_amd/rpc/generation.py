@@ -86,7 +86,9 @@ class GenerationEngine:
         self._pending: collections.deque = collections.deque()
         self._wake = asyncio.Event()
         self._task: Optional[asyncio.Task] = None
-        self.steps = 0  # total engine steps (observability / tests)
+        self.steps = 0       # total engine steps (observability / tests)
+        self.tokens_out = 0  # total tokens emitted across all streams
+        self.streams = 0     # total streams admitted
         # a fresh DecodeSession has every slot ACTIVE at pos 0; park them
         # all so unclaimed slots cost nothing and never hit the sequence
         # limit while other slots generate (submit() re-activates)
@@ -138,6 +140,14 @@ class GenerationEngine:
         self._wake.set()
         return await fut
 
+    def stats(self) -> dict:
+        """Engine counters for observability (tokens/steps ratio ==
+        continuous-batching packing efficiency x batch)."""
+        return dict(steps=self.steps, tokens_out=self.tokens_out,
+                    streams=self.streams,
+                    active=sum(1 for s_ in self._slots if s_ is not None),
+                    free=len(self._free), pending=len(self._pending))
+
     def _finish(self, b: int) -> None:
         self._slots[b]["q"].put_nowait(None)
         self._slots[b] = None
@@ -182,6 +192,7 @@ class GenerationEngine:
                                st["top_p"], st["rng"])
         st["cur"] = tok
         st["q"].put_nowait(tok)
+        self.tokens_out += 1
         st["remaining"] -= 1
         if st["remaining"] <= 0 or tok == self.eos:
             self._finish(b)
@@ -202,6 +213,7 @@ class GenerationEngine:
                 b = self._free.pop()
                 self.session.reset_slot(b)
                 self._slots[b] = st
+                self.streams += 1
                 fut.set_result((b, st["q"]))
             active = [b for b in range(self.B) if self._slots[b]]
             if not active:
