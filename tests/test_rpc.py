@@ -298,3 +298,59 @@ def test_forwarding_middleman_relays_and_measures():
         fwd.close()
         front.shutdown()
         backend.shutdown()
+
+
+def test_balanced_client_spreads_and_fails_over():
+    """Client-side LB (reference 99_LoadBalancer role, in-process): load
+    spreads across healthy replicas; killing one evicts it after a
+    transport failure and every later call lands on the survivor;
+    application errors do NOT evict."""
+    import grpc as _grpc
+
+    from trtlab_amd.rpc.balancer import BalancedClient
+    from trtlab_amd.rpc.middleman import _RawBytes
+    from trtlab_amd.rpc.server import AsyncService, Server
+
+    def make_server(tag):
+        svc = AsyncService("trtlab.Echo")
+
+        async def echo(request, context, resources):
+            if request.data == b"boom":
+                await context.abort(_grpc.StatusCode.INVALID_ARGUMENT,
+                                    "bad request")
+            return _RawBytes(tag + request.data)
+
+        svc.register_unary("Ping", echo, _RawBytes, _RawBytes)
+        srv = Server("127.0.0.1:0")
+        srv.register_service(svc)
+        srv.async_start()
+        return srv
+
+    s1, s2 = make_server(b"a:"), make_server(b"b:")
+    lb = BalancedClient([f"127.0.0.1:{s1.port}", f"127.0.0.1:{s2.port}"],
+                        service="trtlab.Echo", method="Ping",
+                        cooldown_s=30.0)
+    try:
+        seen = set()
+        for i in range(8):
+            seen.add(bytes(lb.call(f"m{i}".encode(), timeout=10))[:2])
+        assert seen == {b"a:", b"b:"}  # both replicas served
+
+        # application error: surfaced, backend stays healthy
+        import pytest as _pytest
+        with _pytest.raises(_grpc.RpcError) as ei:
+            lb.call(b"boom", timeout=10)
+        assert ei.value.code() == _grpc.StatusCode.INVALID_ARGUMENT
+        assert not any(v["down"] for v in lb.stats().values())
+
+        # kill one replica: next calls retry onto the survivor
+        s1.shutdown()
+        for i in range(6):
+            out = bytes(lb.call(f"k{i}".encode(), timeout=10))
+            assert out.startswith(b"b:")
+        st = lb.stats()
+        downs = [a for a, v in st.items() if v["down"]]
+        assert len(downs) == 1 and str(s1.port) in downs[0]
+    finally:
+        lb.close()
+        s2.shutdown()
