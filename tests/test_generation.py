@@ -328,3 +328,40 @@ def test_engine_stats_counters():
         assert st["pending"] == 0 and st["steps"] >= 7
     finally:
         srv.shutdown()
+
+
+def test_admission_queue_bound():
+    """max_waiting bounds the admission queue: with 1 slot busy and 1
+    stream waiting, a third gets RESOURCE_EXHAUSTED immediately instead
+    of queueing unboundedly."""
+    import time as _t
+
+    class SlowFake(FakeSession):
+        def step(self, ids):
+            import time as _tt
+
+            _tt.sleep(0.02)
+            return super().step(ids)
+
+    sess = SlowFake(1)
+    gen = GenerationService(sess, max_waiting=1)
+    srv = Server("127.0.0.1:0")
+    srv.register_service(gen.service)
+    srv.async_start()
+    try:
+        from concurrent.futures import ThreadPoolExecutor
+
+        with ThreadPoolExecutor(3) as ex:
+            f1 = ex.submit(_collect, srv.port, [5], 40)  # occupies the slot
+            _t.sleep(0.15)
+            f2 = ex.submit(_collect, srv.port, [6], 5)   # waits (1 queued)
+            _t.sleep(0.15)
+            with pytest.raises(grpc.RpcError) as ei:
+                _collect(srv.port, [7], 5)               # over the bound
+            assert ei.value.code() == grpc.StatusCode.RESOURCE_EXHAUSTED
+            toks1, done1 = f1.result(60)
+            toks2, done2 = f2.result(60)
+        assert done1 and toks1 == expected_chain([5], 40)
+        assert done2 and toks2 == expected_chain([6], 5)
+    finally:
+        srv.shutdown()

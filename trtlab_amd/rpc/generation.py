@@ -66,7 +66,8 @@ GenerateToken = message_factory.GetMessageClass(
 class GenerationEngine:
     """Continuous-batching loop over a lockstep decode session."""
 
-    def __init__(self, session, eos: int = -1, inline_step: bool = False):
+    def __init__(self, session, eos: int = -1, inline_step: bool = False,
+                 max_waiting: int = 0):
         """inline_step=True runs session.step() directly on the event
         loop instead of a worker thread. MEASURED COUNTERPRODUCTIVE
         (tools/gen_load.py, MI355X, 16 streams/8 slots): 5,491 -> 1,229
@@ -81,6 +82,10 @@ class GenerationEngine:
         self.B = session.batch
         self.eos = eos
         self.inline_step = inline_step
+        # admission control: > 0 bounds the wait queue — beyond it,
+        # submit() raises OverflowError (the service maps it to
+        # RESOURCE_EXHAUSTED) instead of queueing unboundedly
+        self.max_waiting = int(max_waiting)
         self._slots: List[Optional[dict]] = [None] * self.B
         self._free: List[int] = list(range(self.B))
         self._pending: collections.deque = collections.deque()
@@ -126,6 +131,9 @@ class GenerationEngine:
             raise ValueError(
                 f"prompt+max_tokens ({len(prompt)}+{max_tokens}) exceeds "
                 f"the session window ({smax})")
+        if self.max_waiting > 0 and len(self._pending) >= self.max_waiting:
+            raise OverflowError(
+                f"admission queue full ({self.max_waiting} waiting)")
         # the LOOP is the sole owner of session state: submissions queue
         # and are admitted between steps (reset_slot must never race an
         # in-flight replay that is reading this slot's position)
@@ -284,11 +292,12 @@ class GenerationService:
 
     def __init__(self, session, eos: int = -1,
                  name: str = "trtlab.gen.Generation",
-                 inline_step: bool = False):
+                 inline_step: bool = False, max_waiting: int = 0):
         from trtlab_amd.rpc.server import StreamingService
 
         self.engine = GenerationEngine(session, eos=eos,
-                                       inline_step=inline_step)
+                                       inline_step=inline_step,
+                                       max_waiting=max_waiting)
         svc = StreamingService(name)
         svc.register_streaming("Generate", self._generate, GenerateRequest,
                                GenerateToken)
@@ -306,6 +315,11 @@ class GenerationService:
             import grpc
 
             await context.abort(grpc.StatusCode.INVALID_ARGUMENT, str(e))
+            return
+        except OverflowError as e:
+            import grpc
+
+            await context.abort(grpc.StatusCode.RESOURCE_EXHAUSTED, str(e))
             return
         i = 0
         try:
