@@ -524,6 +524,18 @@ PYBIND11_MODULE(_C, m) {
           py::arg("x"), py::arg("out"), py::arg("temps") = 0,
           py::arg("seeds") = 0, py::arg("pos") = 0, py::arg("M") = 0,
           py::arg("V") = 0, py::arg("stream") = 0, py::arg("sync") = true);
+  ops.def("channel_affine",
+          [](int dtype, uintptr_t x, uintptr_t out, uintptr_t s,
+             uintptr_t b, int64_t M, int C, bool relu, uintptr_t stream,
+             bool sync) {
+            launch_channel_affine(dtype, (void*)x, (void*)out, (float*)s,
+                                  (float*)b, M, C, relu,
+                                  as_stream(stream));
+            if (sync) TRT_HIP_CHECK(hipStreamSynchronize(as_stream(stream)));
+          },
+          py::arg("dtype"), py::arg("x"), py::arg("out"), py::arg("s"),
+          py::arg("b"), py::arg("M"), py::arg("C"), py::arg("relu") = false,
+          py::arg("stream") = 0, py::arg("sync") = true);
   ops.def("clip",
           [](int dtype, uintptr_t in, uintptr_t out, int64_t n, float mn,
              float mx, uintptr_t stream, bool sync) {

@@ -57,6 +57,35 @@ __global__ void channel_pad_kernel(const T* __restrict__ in, T* __restrict__ out
   }
 }
 
+// Per-channel affine (+ optional ReLU): out[m, c] = x[m, c]*s[c] + b[c].
+// Standalone batchnorm whose producer is not a conv (DenseNet's
+// pre-activation BN after a concat; ONNX BatchNormalization on non-conv
+// inputs) — the folded (scale, bias) come from the host. C % 8 == 0.
+template <typename T, bool RELU>
+__global__ void channel_affine_kernel(const T* __restrict__ x,
+                                      T* __restrict__ out,
+                                      const float* __restrict__ s,
+                                      const float* __restrict__ b,
+                                      int64_t n8, int C8) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n8;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int c8 = (int)(i % C8) * 8;
+    short4v v0 = *(const short4v*)(x + i * 8);
+    short4v v1 = *(const short4v*)(x + i * 8 + 4);
+    T r[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float v = (float)((const T*)(j < 4 ? (const void*)&v0
+                                         : (const void*)&v1))[j & 3];
+      v = v * s[c8 + j] + b[c8 + j];
+      if constexpr (RELU) v = fmaxf(v, 0.0f);
+      r[j] = (T)v;
+    }
+    *(short4v*)(out + i * 8) = *(const short4v*)&r[0];
+    *(short4v*)(out + i * 8 + 4) = *(const short4v*)&r[4];
+  }
+}
+
 // fp32 -> fp16/bf16 and back (bindings staging, tests)
 template <typename T>
 __global__ void cast_from_f32_kernel(const float* __restrict__ in,
@@ -99,6 +128,27 @@ void launch_elementwise(int dtype, int op, const void* a, const void* b,
   };
   if (dtype == 0) dis(_Float16{});
   else dis(__bf16{});
+}
+
+void launch_channel_affine(int dtype, const void* x, void* out,
+                           const float* s, const float* b, int64_t M, int C,
+                           bool relu, hipStream_t stream) {
+  if (C % 8 != 0)
+    throw std::runtime_error("channel_affine: C % 8 != 0");
+  int64_t n8 = M * (int64_t)(C / 8);
+  int blocks = ew_blocks(n8);
+  auto l = [&](auto t, auto r) {
+    using T = decltype(t);
+    hipLaunchKernelGGL((channel_affine_kernel<T, decltype(r)::value>),
+                       dim3(blocks), dim3(256), 0, stream, (const T*)x,
+                       (T*)out, s, b, n8, C / 8);
+  };
+  if (dtype == 0) {
+    relu ? l(_Float16{}, std::true_type{})
+         : l(_Float16{}, std::false_type{});
+  } else {
+    relu ? l(__bf16{}, std::true_type{}) : l(__bf16{}, std::false_type{});
+  }
 }
 
 void launch_channel_pad(int dtype, const void* in, void* out, int64_t M,

@@ -76,6 +76,10 @@ void launch_bottleneck_tail(int dtype, const void* in, const void* W1,
                             hipStream_t stream);
 void launch_elementwise(int dtype, int op, const void* a, const void* b,
                         void* out, int64_t n, hipStream_t stream);
+// Per-channel affine (+ReLU): standalone batchnorm (DenseNet pre-act).
+void launch_channel_affine(int dtype, const void* x, void* out,
+                           const float* s, const float* b, int64_t M, int C,
+                           bool relu, hipStream_t stream);
 // Row-wise argmax (greedy decode head): fp16 [M, V] -> int32 [M].
 void launch_argmax_rows(const void* x, void* out, int M, int V,
                         hipStream_t stream);

@@ -181,6 +181,11 @@ void ExecutionContext::enqueue_all(hipStream_t s) {
         break;
       case kView:
         break;  // arena alias — out_off == in_off, nothing to launch
+      case kChAffine:
+        launch_channel_affine(op.dtype, A(op.in_off), A(op.out_off),
+                              Fp(op.scale_off), Fp(op.bias_off),
+                              op.n_elems, op.C, op.epi != 0, os);
+        break;
       case kBtail:
         // fused bottleneck tail: W1 at w_off, W2 at w2_off; the fp32
         // scale/bias blobs hold [s1 | s2] / [b1 | b2] (s2 at + C floats)

@@ -123,6 +123,7 @@ enum OpKind : int {
   kBtail = 25,       // fused bottleneck tail: 3x3+BN+ReLU -> 1x1+BN+res+ReLU
   kConst = 26,       // weight-blob constant -> arena tensor (D2D copy)
   kView = 27,        // zero-copy reshape (arena alias; no kernel)
+  kChAffine = 28,    // out[m,c] = x[m,c]*s[c]+b[c] (+ReLU via epi=1)
 };
 
 struct OpDesc {

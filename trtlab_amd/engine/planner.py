@@ -62,6 +62,7 @@ K_ROPE = 24  # rotary embedding, in-place on qkv (arena-aliased output)
 K_BTAIL = 25  # fused bottleneck tail: conv3x3+BN+ReLU -> 1x1+BN+res+ReLU
 K_CONST = 26  # weight-blob constant -> arena tensor (one D2D copy)
 K_VIEW = 27   # zero-copy reshape: output aliases the input's arena bytes
+K_CHAFF = 28  # per-channel affine (+ReLU): standalone (pre-act) batchnorm
 
 
 def _bf16_bits(arr: np.ndarray) -> np.ndarray:
@@ -166,7 +167,24 @@ class Planner:
             elif n.kind == "gemm":
                 exec_ops.append(self._fuse_gemm(g, n, consumed, single_user))
             elif n.kind == "batchnorm":
-                raise ValueError(f"unfused batchnorm {n.name} (expected after conv2d)")
+                # standalone BN (producer is not a conv — DenseNet's
+                # pre-activation blocks after concat): fold the stats
+                # into a per-channel affine; absorb a following relu
+                a = n.attrs
+                sc = (a["gamma"] /
+                      np.sqrt(a["var"] + a["eps"])).astype(np.float32)
+                bi = (a["beta"] - a["mean"] * sc).astype(np.float32)
+                out = n.output
+                relu = 0
+                r = single_user(out)
+                if r is not None and r.kind == "relu":
+                    consumed.add(r.name)
+                    out = r.output
+                    relu = 1
+                op = ExecOp(K_CHAFF, n.name, [n.inputs[0]], out,
+                            dict(relu=relu))
+                op.scale, op.bias = sc, bi
+                exec_ops.append(op)
             elif n.kind == "relu":
                 exec_ops.append(ExecOp(K_ELEMENTWISE, n.name, [n.inputs[0]],
                                        n.output, dict(op=0)))
@@ -708,6 +726,13 @@ class Planner:
                          sh=op.params["stride"], sw=op.params["stride"],
                          ph=op.params["padding"], pw=op.params["padding"],
                          res_scale=op.params.get("res_scale", 1.0))
+            elif op.kind == K_CHAFF:
+                sh = shapes[op.inputs[0]]
+                m = 1
+                for dd in sh[:-1]:
+                    m *= dd
+                d.update(kind=K_CHAFF, epi=op.params["relu"], n_elems=m,
+                         C=sh[-1])
             elif op.kind == K_VIEW:
                 d.update(kind=K_VIEW)
             elif op.kind == K_CONST:

@@ -15,7 +15,7 @@ import torch
 import torch.nn.functional as F
 
 from trtlab_amd.engine.planner import (
-    K_CLIP, K_COPY2D, K_TRANSPOSE2D, K_RMSNORM, K_SILU_MUL, K_ROPE, K_VIEW,
+    K_CHAFF, K_CLIP, K_COPY2D, K_TRANSPOSE2D, K_RMSNORM, K_SILU_MUL, K_ROPE, K_VIEW,
     EnginePlan, K_ADD_LAYERNORM, K_ATTENTION, K_AVGPOOL, K_BTAIL, K_CHANNEL_PAD, K_CONST, K_CONV,
     K_DEQUANT, K_ELEMENTWISE, K_GAVGPOOL, K_GEMM, K_LAYERNORM, K_MAXPOOL,
     K_EMBEDDING, K_GEMM_MX4, K_GEMM_MX8, K_QUANT_MX4, K_QUANT_MX8,
@@ -93,6 +93,9 @@ def run_reference(plan: EnginePlan, input_nhwc, return_all: bool = False):
             elif d["dtype"] == 3:  # fp8: emulate the e4m3 output store
                 y = _fp8_round(y)
             t[op.output] = y
+        elif op.kind == K_CHAFF:
+            y = x * torch.from_numpy(op.scale) + torch.from_numpy(op.bias)
+            t[op.output] = torch.relu(y) if d["epi"] else y
         elif op.kind == K_VIEW:
             t[op.output] = x.reshape(op.params["shape"])
         elif op.kind == K_CONST:
