@@ -85,3 +85,18 @@ def test_densenet_engine_matches_reference():
     ref = run_reference(plan, x)
     err = np.abs(out - ref).max() / max(np.abs(ref).max(), 1e-6)
     assert err < 0.08, err
+
+
+def test_densenet_onnx_roundtrip():
+    """DenseNet exports (Conv/BN/Relu/Concat/AveragePool) and re-imports
+    to an identical-output graph — including standalone (non-conv-fed)
+    BatchNormalization, which lowers to the channel-affine op."""
+    from trtlab_amd.engine.onnx_io import export_onnx, import_onnx
+
+    g = build_densenet(**CFG)
+    g2 = import_onnx(export_onnx(g))
+    p1, p2 = Planner().compile(g), Planner().compile(g2)
+    x = (np.random.RandomState(1).randn(*p1.input_shape) * 0.5).astype(
+        np.float32)
+    a, b = run_reference(p1, x), run_reference(p2, x)
+    assert np.abs(a - b).max() / max(np.abs(a).max(), 1e-6) < 1e-5
