@@ -269,3 +269,43 @@ def test_fork_join_downsample_marking():
         assert plan.ops[j]["kind"] in (K_CONV, K_BTAIL)
         # the join consumes the forked output as its residual
         assert plan.ops[j]["in2_off"] == plan.ops[f]["out_off"]
+
+
+def test_btail_fusion_gating():
+    """Bottleneck-tail fusion applies to fp16 plans only (the kernel is
+    fp16), is disabled by btail_fusion=False, and never fuses when the
+    3x3 output is a pinned engine output."""
+    from trtlab_amd.engine.planner import DT_BF16, DT_I8, K_BTAIL, Planner
+    from trtlab_amd.models import build_resnet
+
+    g = build_resnet(50, batch=1, image=64, seed=0)
+    fp16 = Planner().compile(g)
+    assert sum(1 for d in fp16.ops if d["kind"] == K_BTAIL) == 3
+
+    off = Planner(btail_fusion=False).compile(
+        build_resnet(50, batch=1, image=64, seed=0))
+    assert sum(1 for d in off.ops if d["kind"] == K_BTAIL) == 0
+
+    bf16 = Planner(dtype=DT_BF16).compile(
+        build_resnet(50, batch=1, image=64, seed=0))
+    assert sum(1 for d in bf16.ops if d["kind"] == K_BTAIL) == 0
+
+    i8 = Planner(dtype=DT_I8).compile(
+        build_resnet(50, batch=1, image=64, seed=0))
+    assert sum(1 for d in i8.ops if d["kind"] == K_BTAIL) == 0
+
+    # pinning a 3x3 output chain as an engine output blocks that pair's
+    # fusion (the intermediate must stay addressable): pin the RELU that
+    # follows the first 64-wide 3x3 conv (the tensor the fusion would
+    # otherwise swallow into LDS)
+    g2 = build_resnet(50, batch=1, image=64, seed=0)
+    first3 = next(n for n in g2.nodes
+                  if n.kind == "conv2d" and
+                  n.attrs["weight"].shape == (64, 64, 3, 3))
+    relu = next(n for n in g2.nodes
+                if n.kind == "relu" and len(n.inputs) == 1 and
+                any(m.kind == "batchnorm" and m.inputs[0] == first3.output
+                    and n.inputs[0] == m.output for m in g2.nodes))
+    g2.mark_output(relu.output)
+    pinned = Planner().compile(g2)
+    assert sum(1 for d in pinned.ops if d["kind"] == K_BTAIL) == 2
