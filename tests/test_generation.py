@@ -365,3 +365,43 @@ def test_admission_queue_bound():
         assert done2 and toks2 == expected_chain([6], 5)
     finally:
         srv.shutdown()
+
+
+def test_host_sampler_properties():
+    """Host-side sampler (used for top-k / nucleus streams) invariants,
+    property-style over random logits: greedy reductions, candidate-set
+    containment, nucleus prefix bound, seed determinism."""
+    from trtlab_amd.rpc.generation import GenerationEngine
+
+    s = GenerationEngine._sample
+    rng = np.random.RandomState(0)
+    for trial in range(25):
+        v = rng.randint(4, 40)
+        logits = rng.randn(v).astype(np.float32) * rng.uniform(0.5, 3)
+        greedy = int(np.argmax(logits))
+        # temperature 0 and top_k=1 are greedy regardless of seeds
+        assert s(logits, 0.0, 0, 0.0, np.random.RandomState(trial)) \
+            == greedy
+        assert s(logits, 2.0, 1, 0.0, np.random.RandomState(trial)) \
+            == greedy
+        # top-k draws stay inside the k best
+        k = rng.randint(1, v)
+        topk = set(np.argsort(-logits)[:k].tolist())
+        for seed in range(5):
+            assert s(logits, 1.0, k, 0.0,
+                     np.random.RandomState(seed)) in topk
+        # nucleus: the drawn index is inside the smallest prefix whose
+        # probability mass reaches p
+        p = float(rng.uniform(0.2, 0.95))
+        x = logits.astype(np.float64)
+        prob = np.exp(x - x.max())
+        prob /= prob.sum()
+        order = np.argsort(-prob)
+        keep = int(np.searchsorted(np.cumsum(prob[order]), p) + 1)
+        nucleus = set(order[:keep].tolist())
+        for seed in range(5):
+            assert s(logits, 1.0, 0, p,
+                     np.random.RandomState(seed)) in nucleus
+        # same seed -> same draw
+        assert s(logits, 0.7, 0, 0.9, np.random.RandomState(42)) == \
+            s(logits, 0.7, 0, 0.9, np.random.RandomState(42))
