@@ -354,3 +354,36 @@ def test_balanced_client_spreads_and_fails_over():
     finally:
         lb.close()
         s2.shutdown()
+
+
+def test_balanced_client_all_down_reprobe():
+    """When every backend is cooling down, the balancer re-probes the
+    one whose cooldown expires soonest instead of failing fast — a
+    transient full outage recovers without client-side restarts."""
+    from trtlab_amd.rpc.balancer import BalancedClient
+    from trtlab_amd.rpc.middleman import _RawBytes
+    from trtlab_amd.rpc.server import AsyncService, Server
+
+    svc = AsyncService("trtlab.Echo")
+
+    async def echo(request, context, resources):
+        return _RawBytes(b"ok:" + request.data)
+
+    svc.register_unary("Ping", echo, _RawBytes, _RawBytes)
+    srv = Server("127.0.0.1:0")
+    srv.register_service(svc)
+    srv.async_start()
+    # second backend never existed: its port is closed
+    lb = BalancedClient([f"127.0.0.1:{srv.port}", "127.0.0.1:1"],
+                        service="trtlab.Echo", method="Ping",
+                        cooldown_s=60.0)
+    try:
+        # force both into cooldown: dead backend by calling it, live one
+        # artificially
+        for be in lb._backends:
+            be.down_until = __import__("time").monotonic() + 60.0
+        out = bytes(lb.call(b"x", timeout=10))
+        assert out == b"ok:x"  # re-probe found the healthy one (retries)
+    finally:
+        lb.close()
+        srv.shutdown()

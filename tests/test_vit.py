@@ -120,3 +120,24 @@ def test_vit_engine_matches_reference():
     ref = run_reference(plan, x)
     err = np.abs(out - ref).max() / max(np.abs(ref).max(), 1e-6)
     assert err < 0.08, err
+
+
+def test_llama_plan_roundtrip(tmp_path):
+    """LLaMA plans (rope arena aliasing, rmsnorm/silu ops) survive
+    save_plan/load_plan byte-exactly."""
+    import os
+
+    from trtlab_amd.engine.plan_io import load_plan, save_plan
+    from trtlab_amd.models import build_llama
+
+    g = build_llama(batch=1, seq=32, hidden=256, layers=1, heads=2,
+                    seed=0, vocab=400)
+    plan = Planner().compile(g)
+    pth = os.path.join(str(tmp_path), "llama.npz")
+    save_plan(plan, pth)
+    p2 = load_plan(pth)
+    ids = np.random.RandomState(2).randint(
+        0, 400, plan.input_shape).astype(np.int32)
+    a = run_reference(plan, ids)
+    b = run_reference(p2, ids)
+    assert np.abs(a - b).max() < 1e-6
